@@ -1,0 +1,150 @@
+/* rw_stream.h — C-ABI executor boundary for the RisingWave stream hot path.
+ *
+ * This is the drop-in boundary of SURVEY.md §8b: each entry point replaces
+ * the construction/driving of a reference executor:
+ *
+ *   rw_hash_agg_create  ⇔ HashAggExecutorBuilder::new_boxed_executor
+ *                         (src/stream/src/from_proto/hash_agg.rs:52-120;
+ *                          node params proto/stream_plan.proto:548-560)
+ *   rw_hash_join_create ⇔ HashJoinExecutorBuilder::new_boxed_executor
+ *                         (src/stream/src/from_proto/hash_join.rs:37-229;
+ *                          node params proto/stream_plan.proto:639-682)
+ *   *_push_chunk        ⇔ Message::Chunk into Execute::execute's stream
+ *                         (executor/mod.rs:244-291,1332-1349); for the join,
+ *                         `side` selects the barrier-aligned input
+ *                         (executor/barrier_align.rs:45-160)
+ *   *_flush             ⇔ Message::Barrier(epoch): state commit + (agg)
+ *                         change emission (hash_agg.rs:651-658,
+ *                          hash_join.rs:737-748; StateTable::commit
+ *                          state_table.rs:1718)
+ *   *_poll              ⇔ the Message::Chunk outputs yielded since the last
+ *                         poll, in yield order; NULL when drained
+ *
+ * Errors: negative return = error code; rw_last_error() gives a message
+ * (⇔ StreamExecutorResult). Ownership: chunks passed in are COPIED by the
+ * callee during the call; chunks returned by *_poll are callee-allocated and
+ * must be freed with rw_chunk_free.
+ */
+#ifndef RW_STREAM_H
+#define RW_STREAM_H
+
+#include "rw_chunk.h"
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* ----- errors ----- */
+#define RW_OK 0
+#define RW_E_INVAL (-1)
+#define RW_E_OVERFLOW (-2)       /* checked_add/checked_sub failed (general.rs:28-41) */
+#define RW_E_INCONSISTENT (-3)   /* e.g. duplicate pk insert (join/hash_join.rs:760-799) */
+#define RW_E_NOGPU (-4)          /* product library without a visible GPU */
+#define RW_E_INTERNAL (-5)
+
+const char* rw_last_error(void);
+void rw_chunk_free(RwChunk* c);
+
+/* ----- HashAgg ----- */
+
+/* Agg kinds on the path (expr/impl/src/aggregate/general.rs):
+ * count(*) :154-157 (counts rows, no arg) · count(col) (skips NULL arg)
+ * sum      :18-41 (checked, NULL-skipping; NULL result when no input)
+ * sum0     :27 (init 0, used for 2-phase count)
+ * min/max  :90-125 (value state when append-only, else materialized input —
+ *           chosen as in frontend generic/agg.rs:394-461) */
+enum RwAggKind {
+    RW_AGG_COUNT_STAR = 0,
+    RW_AGG_COUNT = 1,
+    RW_AGG_SUM = 2,
+    RW_AGG_SUM0 = 3,
+    RW_AGG_MIN = 4,
+    RW_AGG_MAX = 5,
+};
+
+typedef struct RwAggCall {
+    uint8_t kind;     /* RwAggKind */
+    int32_t arg;      /* input column index, -1 for count(*) */
+    uint8_t ret_type; /* RwTypeId of the output */
+} RwAggCall;
+
+typedef struct RwHashAggDesc {
+    uint32_t n_input_cols;
+    const uint8_t* input_types; /* RwTypeId per input column */
+    uint32_t n_group_key;
+    const uint32_t* group_key_indices;
+    uint32_t n_calls;
+    const RwAggCall* calls;
+    uint32_t row_count_index; /* index of the count(*) call (hash_agg.rs:97-98) */
+    uint32_t n_stream_key;    /* input stream key, orders materialized input */
+    const uint32_t* stream_key;
+    uint32_t chunk_size;      /* output chunk rows (config/mod.rs:222-224) */
+    uint8_t append_only;      /* value-state min/max allowed */
+} RwHashAggDesc;
+
+void* rw_hash_agg_create(const RwHashAggDesc* desc);
+int rw_hash_agg_push_chunk(void* h, const RwChunk* chunk);
+int rw_hash_agg_flush(void* h, uint64_t epoch); /* barrier */
+RwChunk* rw_hash_agg_poll(void* h);
+void rw_hash_agg_destroy(void* h);
+
+/* ----- HashJoin ----- */
+
+/* Join type constants, same values as executor/join/mod.rs:42-52. */
+enum RwJoinType {
+    RW_JOIN_INNER = 0,
+    RW_JOIN_LEFT_OUTER = 1,
+    RW_JOIN_RIGHT_OUTER = 2,
+    RW_JOIN_FULL_OUTER = 3,
+    RW_JOIN_LEFT_SEMI = 4,
+    RW_JOIN_LEFT_ANTI = 5,
+    RW_JOIN_RIGHT_SEMI = 6,
+    RW_JOIN_RIGHT_ANTI = 7,
+};
+
+/* Non-equi condition restricted to the comparison shapes the reference tests
+ * and the q7 plan use (a NonStrictExpression comparing two columns of the
+ * concatenated row, hash_join.rs:1381-1403). */
+enum RwCmpOp {
+    RW_CMP_LT = 0,
+    RW_CMP_LE = 1,
+    RW_CMP_GT = 2,
+    RW_CMP_GE = 3,
+};
+
+typedef struct RwHashJoinDesc {
+    uint8_t join_type;   /* RwJoinType */
+    uint8_t append_only; /* append_only_optimize (hash_join.rs:186-187) */
+    uint32_t n_key;
+    const uint32_t* key_l; /* join key column indices per side */
+    const uint32_t* key_r;
+    const uint8_t* null_safe; /* n_key flags (stream_plan.proto:648) */
+    uint32_t n_cols_l;
+    const uint8_t* types_l;
+    uint32_t n_cols_r;
+    const uint8_t* types_r;
+    uint32_t n_pk_l; /* deduped pk indices = JoinParams.deduped_pk_indices */
+    const uint32_t* pk_l;
+    uint32_t n_pk_r;
+    const uint32_t* pk_r;
+    uint32_t n_output; /* output_indices into [left cols ‖ right cols] */
+    const uint32_t* output_indices;
+    uint8_t has_cond; /* optional non-equi condition */
+    uint8_t cond_op;  /* RwCmpOp */
+    uint32_t cond_l;  /* column indices into the concatenated row */
+    uint32_t cond_r;
+    uint32_t chunk_size;
+} RwHashJoinDesc;
+
+enum RwJoinSide { RW_SIDE_LEFT = 0, RW_SIDE_RIGHT = 1 };
+
+void* rw_hash_join_create(const RwHashJoinDesc* desc);
+int rw_hash_join_push_chunk(void* h, int side, const RwChunk* chunk);
+int rw_hash_join_flush(void* h, uint64_t epoch); /* aligned barrier */
+RwChunk* rw_hash_join_poll(void* h);
+void rw_hash_join_destroy(void* h);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* RW_STREAM_H */

@@ -1,0 +1,325 @@
+// oracle/common.hpp — shared datum/row/chunk machinery for the CPU oracle.
+//
+// ORACLE — TEST INFRASTRUCTURE ONLY. This library is a CPU restatement of the
+// reference executors used to pin parity. Only tests/,
+// __graft_entry__.smoke() and bench.py's cpu_baseline leg may load it. It is
+// never the product path.
+//
+// Restates (semantics; file:line cite the reference under /root/reference):
+//  - Datum equality/order: ScalarImpl PartialEq / DefaultOrd with NULLs
+//    largest (common/src/util/sort_util.rs:67-74, memcmp_encoding.rs:31-52).
+//  - StreamChunkBuilder::append_iter_inner chunking incl. the U-pair
+//    no-split rule (common/src/array/stream_chunk_builder.rs:188-218).
+//  - StreamChunk::eliminate_adjacent_noop_update
+//    (common/src/array/stream_chunk.rs:331-384).
+#pragma once
+#include <cassert>
+#include <cmath>
+#include <cstdint>
+#include <cstring>
+#include <map>
+#include <memory>
+#include <string>
+#include <vector>
+
+#include "../include/rw_chunk.h"
+
+namespace orc {
+
+struct Datum {
+    bool null = true;
+    union {
+        int64_t i;
+        double d;
+    };
+    Datum() : null(true), i(0) {}
+    static Datum of_i(int64_t v) {
+        Datum x;
+        x.null = false;
+        x.i = v;
+        return x;
+    }
+    static Datum of_d(double v) {
+        Datum x;
+        x.null = false;
+        x.d = v;
+        return x;
+    }
+};
+
+inline bool type_is_float(uint8_t t) { return t == RW_T_F64 || t == RW_T_F32; }
+
+// ScalarImpl PartialEq: floats are OrderedFloat (NaN == NaN).
+inline bool datum_eq(const Datum& a, const Datum& b, uint8_t t) {
+    if (a.null || b.null) return a.null == b.null;
+    if (type_is_float(t)) {
+        return a.d == b.d || (std::isnan(a.d) && std::isnan(b.d));
+    }
+    return a.i == b.i;
+}
+
+// DefaultOrd: total order, NULLs largest, NaN largest among floats
+// (types/ord.rs DefaultOrd; sort_util.rs NullsAre::Largest default).
+inline int datum_cmp(const Datum& a, const Datum& b, uint8_t t) {
+    if (a.null || b.null) {
+        if (a.null && b.null) return 0;
+        return a.null ? 1 : -1; // NULL largest
+    }
+    if (type_is_float(t)) {
+        bool an = std::isnan(a.d), bn = std::isnan(b.d);
+        if (an || bn) return an == bn ? 0 : (an ? 1 : -1);
+        return a.d < b.d ? -1 : (a.d > b.d ? 1 : 0);
+    }
+    return a.i < b.i ? -1 : (a.i > b.i ? 1 : 0);
+}
+
+using Row = std::vector<Datum>;
+
+inline bool row_eq(const Row& a, const Row& b, const std::vector<uint8_t>& types) {
+    if (a.size() != b.size()) return false;
+    for (size_t i = 0; i < a.size(); i++)
+        if (!datum_eq(a[i], b[i], types[i])) return false;
+    return true;
+}
+
+// Order spec for a sequence of datums (e.g. minput cache key, join pk).
+struct OrderCol {
+    uint8_t type;
+    bool desc = false; // nulls largest in the VALUE order; desc reverses values
+};
+
+// memcomparable order: per column, compare by (asc|desc) with NULLs largest
+// IN THE STORED ORDER — i.e. for DESC columns NULL sorts FIRST
+// (nulls_are_first = desc && nulls_are_largest, sort_util.rs:208-211).
+inline int ordered_cmp(const Row& a, const Row& b, const std::vector<OrderCol>& order) {
+    assert(a.size() == b.size() && a.size() == order.size());
+    for (size_t i = 0; i < a.size(); i++) {
+        int c = datum_cmp(a[i], b[i], order[i].type);
+        if (order[i].desc) c = -c;
+        if (c) return c;
+    }
+    return 0;
+}
+
+struct RowOrderLess {
+    std::vector<OrderCol> order;
+    bool operator()(const Row& a, const Row& b) const { return ordered_cmp(a, b, order) < 0; }
+};
+
+// ----- owned chunk (builder output / C-ABI marshalling) -----
+
+struct OwnedChunk {
+    std::vector<uint8_t> ops;
+    std::vector<uint8_t> vis; // empty = all visible
+    std::vector<uint8_t> types;
+    std::vector<std::vector<Datum>> cols; // column-major
+    size_t n_rows() const { return ops.size(); }
+    Row row(size_t r) const {
+        Row out(cols.size());
+        for (size_t c = 0; c < cols.size(); c++) out[c] = cols[c][r];
+        return out;
+    }
+    bool visible(size_t r) const { return vis.empty() || vis[r]; }
+};
+
+// View over an incoming C chunk.
+struct ChunkView {
+    const RwChunk* c;
+    uint8_t op(size_t r) const { return c->ops[r]; }
+    bool visible(size_t r) const { return c->vis == nullptr || c->vis[r]; }
+    size_t n_rows() const { return c->n_rows; }
+    size_t n_cols() const { return c->n_cols; }
+    uint8_t type(size_t col) const { return c->cols[col].type; }
+    Datum at(size_t r, size_t col) const {
+        const RwColumn& cc = c->cols[col];
+        if (!cc.valid[r]) return Datum();
+        switch (cc.type) {
+            case RW_T_I64:
+            case RW_T_TS: return Datum::of_i(((const int64_t*)cc.data)[r]);
+            case RW_T_I32: return Datum::of_i(((const int32_t*)cc.data)[r]);
+            case RW_T_BOOL: return Datum::of_i(((const uint8_t*)cc.data)[r]);
+            case RW_T_F64: return Datum::of_d(((const double*)cc.data)[r]);
+            case RW_T_F32: return Datum::of_d(((const float*)cc.data)[r]);
+            default: return Datum();
+        }
+    }
+    Row row(size_t r) const {
+        Row out(n_cols());
+        for (size_t col = 0; col < n_cols(); col++) out[col] = at(r, col);
+        return out;
+    }
+};
+
+// StreamChunkBuilder (stream_chunk_builder.rs:188-218): yields a chunk when
+// size reaches max_chunk_size, EXCEPT when the just-appended op is
+// UpdateDelete — then it waits for the paired UpdateInsert (max+1 rows).
+struct ChunkBuilder {
+    size_t max_chunk_size;
+    std::vector<uint8_t> types;
+    OwnedChunk cur;
+    ChunkBuilder(size_t max_size, std::vector<uint8_t> ts)
+        : max_chunk_size(std::max<size_t>(max_size, 1)), types(std::move(ts)) {
+        reset();
+    }
+    void reset() {
+        cur = OwnedChunk();
+        cur.types = types;
+        cur.cols.assign(types.size(), {});
+    }
+    // returns true if a full chunk was produced into `out`
+    bool append_row(uint8_t op, const Row& row, std::unique_ptr<OwnedChunk>* out) {
+        assert(row.size() == types.size());
+        cur.ops.push_back(op);
+        for (size_t i = 0; i < row.size(); i++) cur.cols[i].push_back(row[i]);
+        size_t size = cur.ops.size();
+        if ((size == max_chunk_size && op != RW_OP_UPDATE_DELETE) || size > max_chunk_size) {
+            *out = take();
+            return true;
+        }
+        return false;
+    }
+    std::unique_ptr<OwnedChunk> take() {
+        if (cur.ops.empty()) return nullptr;
+        auto out = std::make_unique<OwnedChunk>(std::move(cur));
+        reset();
+        return out;
+    }
+};
+
+// StreamChunk::eliminate_adjacent_noop_update (stream_chunk.rs:331-384).
+inline void eliminate_adjacent_noop_update(OwnedChunk& c) {
+    size_t len = c.n_rows();
+    if (c.vis.empty()) c.vis.assign(len, 1);
+    long prev_r = -1;
+    for (size_t curr = 0; curr < len; curr++) {
+        if (!c.vis[curr]) continue;
+        bool matched = false;
+        if (prev_r >= 0) {
+            uint8_t po = c.ops[prev_r], co = c.ops[curr];
+            bool del_then_ins = (po == RW_OP_UPDATE_DELETE || po == RW_OP_DELETE) &&
+                                (co == RW_OP_UPDATE_INSERT || co == RW_OP_INSERT);
+            bool ins_then_del = (po == RW_OP_UPDATE_INSERT || po == RW_OP_INSERT) &&
+                                (co == RW_OP_UPDATE_DELETE || co == RW_OP_DELETE);
+            if ((del_then_ins || ins_then_del) &&
+                row_eq(c.row(prev_r), c.row(curr), c.types)) {
+                c.vis[prev_r] = 0;
+                c.vis[curr] = 0;
+                prev_r = -1;
+                matched = true;
+            }
+        }
+        if (!matched) prev_r = (long)curr;
+    }
+    // Normalize half-visible update pairs.
+    for (size_t idx = 0; idx + 1 < len; idx++) {
+        if (c.ops[idx] == RW_OP_UPDATE_DELETE && c.ops[idx + 1] == RW_OP_UPDATE_INSERT) {
+            bool dv = c.vis[idx], iv = c.vis[idx + 1];
+            if (dv && !iv) c.ops[idx] = RW_OP_DELETE;
+            else if (!dv && iv) c.ops[idx + 1] = RW_OP_INSERT;
+        }
+    }
+    // drop the vis vector if everything is visible (cosmetic)
+    bool all = true;
+    for (auto v : c.vis)
+        if (!v) { all = false; break; }
+    if (all) c.vis.clear();
+}
+
+// ----- C-ABI marshalling of an OwnedChunk -----
+
+inline RwChunk* chunk_to_c(const OwnedChunk& oc) {
+    size_t n = oc.n_rows(), m = oc.cols.size();
+    // single allocation block layout: RwChunk, RwColumn[m], then arrays
+    auto* ch = new RwChunk();
+    auto* cols = new RwColumn[m];
+    auto* ops = new uint8_t[n];
+    std::memcpy(ops, oc.ops.data(), n);
+    uint8_t* vis = nullptr;
+    if (!oc.vis.empty()) {
+        vis = new uint8_t[n];
+        std::memcpy(vis, oc.vis.data(), n);
+    }
+    for (size_t c = 0; c < m; c++) {
+        uint8_t t = oc.types[c];
+        auto* valid = new uint8_t[n];
+        void* data = nullptr;
+        switch (t) {
+            case RW_T_I64:
+            case RW_T_TS: {
+                auto* p = new int64_t[n];
+                for (size_t r = 0; r < n; r++) {
+                    valid[r] = !oc.cols[c][r].null;
+                    p[r] = valid[r] ? oc.cols[c][r].i : 0;
+                }
+                data = p;
+                break;
+            }
+            case RW_T_I32: {
+                auto* p = new int32_t[n];
+                for (size_t r = 0; r < n; r++) {
+                    valid[r] = !oc.cols[c][r].null;
+                    p[r] = valid[r] ? (int32_t)oc.cols[c][r].i : 0;
+                }
+                data = p;
+                break;
+            }
+            case RW_T_BOOL: {
+                auto* p = new uint8_t[n];
+                for (size_t r = 0; r < n; r++) {
+                    valid[r] = !oc.cols[c][r].null;
+                    p[r] = valid[r] ? (uint8_t)oc.cols[c][r].i : 0;
+                }
+                data = p;
+                break;
+            }
+            case RW_T_F64: {
+                auto* p = new double[n];
+                for (size_t r = 0; r < n; r++) {
+                    valid[r] = !oc.cols[c][r].null;
+                    p[r] = valid[r] ? oc.cols[c][r].d : 0;
+                }
+                data = p;
+                break;
+            }
+            case RW_T_F32: {
+                auto* p = new float[n];
+                for (size_t r = 0; r < n; r++) {
+                    valid[r] = !oc.cols[c][r].null;
+                    p[r] = valid[r] ? (float)oc.cols[c][r].d : 0;
+                }
+                data = p;
+                break;
+            }
+        }
+        cols[c].type = t;
+        cols[c].valid = valid;
+        cols[c].data = data;
+    }
+    ch->n_rows = (uint32_t)n;
+    ch->n_cols = (uint32_t)m;
+    ch->ops = ops;
+    ch->vis = vis;
+    ch->cols = cols;
+    return ch;
+}
+
+inline void chunk_free_c(RwChunk* ch) {
+    if (!ch) return;
+    for (uint32_t c = 0; c < ch->n_cols; c++) {
+        delete[] ch->cols[c].valid;
+        switch (ch->cols[c].type) {
+            case RW_T_I64:
+            case RW_T_TS: delete[] (int64_t*)ch->cols[c].data; break;
+            case RW_T_I32: delete[] (int32_t*)ch->cols[c].data; break;
+            case RW_T_BOOL: delete[] (uint8_t*)ch->cols[c].data; break;
+            case RW_T_F64: delete[] (double*)ch->cols[c].data; break;
+            case RW_T_F32: delete[] (float*)ch->cols[c].data; break;
+        }
+    }
+    delete[] ch->cols;
+    delete[] ch->ops;
+    delete[] ch->vis;
+    delete ch;
+}
+
+} // namespace orc

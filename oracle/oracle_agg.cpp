@@ -1,0 +1,360 @@
+// oracle/oracle_agg.cpp — CPU restatement of HashAggExecutor.
+// ORACLE — TEST INFRASTRUCTURE ONLY (see common.hpp header note).
+//
+// Restates (reference under /root/reference):
+//  - HashAggExecutor::apply_chunk / flush_data
+//    (src/stream/src/executor/aggregate/hash_agg.rs:332-514). Per-row apply in
+//    row order is equivalent to the reference's per-group visibility loop
+//    (:241-258,366-398): each row belongs to exactly one group and a group's
+//    state depends only on its own rows' order.
+//  - AggGroup::get_outputs / build_outputs_change with OnlyOutputIfHasInput
+//    (aggregate/agg_group.rs:131-165,431-467,545-610): reset value states at
+//    row_count==0, 0→n Insert, n→0 Delete(prev row), changed U−/U+,
+//    unchanged nothing; negative row_count clamps to 0 (:55-79).
+//  - Value-state aggregates (expr/impl/src/aggregate/general.rs): sum with
+//    checked add/sub (:18-41), count ±1 (:154-162), append-only min/max
+//    (:90-125). NULL inputs are skipped by sum/count(col)/min/max; count(*)
+//    counts rows. Sum state stays Some(0) if all inputs retracted (macro
+//    Option<S> semantics) — outputs 0, not NULL, while row_count > 0.
+//  - Materialized-input min/max (aggregate/minput.rs:170-248): state ordered
+//    by [value ASC(min)/DESC(max), stream_key ASC], NULLs largest
+//    (test_utils/agg_executor.rs:72-121); output = first entry's value.
+#include <stdexcept>
+#include <unordered_map>
+
+#include "../include/rw_stream.h"
+#include "common.hpp"
+
+namespace orc {
+
+thread_local std::string g_err;
+
+struct ValueState {
+    // count / sum: has=false ⇔ Rust state None (sum) — count init 0 has=true
+    bool has = false;
+    bool is_float = false;
+    int64_t i = 0;
+    double d = 0;
+};
+
+struct MInputState {
+    // ordered multiset: key = [value, stream_key...], value = arg datum
+    std::map<Row, Datum, RowOrderLess> entries;
+    MInputState() = default;
+    explicit MInputState(RowOrderLess less) : entries(std::move(less)) {}
+};
+
+struct AggGroupState {
+    std::vector<ValueState> vstates;
+    std::vector<MInputState> mstates; // indexed per call; unused for value calls
+    bool has_prev = false;
+    Row prev_outputs; // agg outputs only (no group key)
+};
+
+struct KeyHash {
+    size_t operator()(const Row& r) const {
+        size_t h = 0xcbf29ce484222325ULL;
+        for (auto& d : r) {
+            uint64_t v = d.null ? 0x9e3779b97f4a7c15ULL : (uint64_t)d.i;
+            h ^= v + 0x9e3779b97f4a7c15ULL + (h << 6) + (h >> 2);
+        }
+        return h;
+    }
+};
+struct KeyEq {
+    const std::vector<uint8_t>* types;
+    bool operator()(const Row& a, const Row& b) const {
+        for (size_t i = 0; i < a.size(); i++)
+            if (!datum_eq(a[i], b[i], (*types)[i])) return false;
+        return true;
+    }
+};
+
+struct HashAggOracle {
+    // descriptor
+    std::vector<uint8_t> input_types;
+    std::vector<uint32_t> group_key;
+    std::vector<RwAggCall> calls;
+    uint32_t row_count_index;
+    std::vector<uint32_t> stream_key;
+    size_t chunk_size;
+    bool append_only;
+    std::vector<uint8_t> group_key_types;
+    std::vector<uint8_t> out_types; // group key types ++ ret types
+    std::vector<bool> call_is_minput;
+
+    std::unordered_map<Row, AggGroupState, KeyHash, KeyEq> groups;
+    std::vector<Row> dirty_order; // first-touch order of dirty group keys
+    std::unordered_map<Row, bool, KeyHash, KeyEq> dirty;
+
+    ChunkBuilder builder;
+    std::vector<std::unique_ptr<OwnedChunk>> outputs;
+
+    HashAggOracle(const RwHashAggDesc* d, std::vector<uint8_t> out_ts)
+        : builder(d->chunk_size, out_ts) {
+        input_types.assign(d->input_types, d->input_types + d->n_input_cols);
+        group_key.assign(d->group_key_indices, d->group_key_indices + d->n_group_key);
+        calls.assign(d->calls, d->calls + d->n_calls);
+        row_count_index = d->row_count_index;
+        stream_key.assign(d->stream_key, d->stream_key + d->n_stream_key);
+        chunk_size = d->chunk_size;
+        append_only = d->append_only;
+        for (auto k : group_key) group_key_types.push_back(input_types[k]);
+        out_types = out_ts;
+        for (auto& c : calls) {
+            bool minput = (c.kind == RW_AGG_MIN || c.kind == RW_AGG_MAX) && !append_only;
+            call_is_minput.push_back(minput);
+        }
+        groups = decltype(groups)(16, KeyHash{}, KeyEq{&group_key_types});
+        dirty = decltype(dirty)(16, KeyHash{}, KeyEq{&group_key_types});
+    }
+
+    RowOrderLess minput_order(const RwAggCall& c) const {
+        // pk of the materialized-input table: value (ASC min / DESC max),
+        // then stream key ASC (agg_executor.rs:90-105)
+        RowOrderLess less;
+        less.order.push_back({input_types[c.arg], c.kind == RW_AGG_MAX});
+        for (auto sk : stream_key) less.order.push_back({input_types[sk], false});
+        return less;
+    }
+
+    AggGroupState& touch(const Row& key) {
+        auto it = groups.find(key);
+        if (it == groups.end()) {
+            AggGroupState g;
+            for (auto& c : calls) {
+                ValueState v;
+                if (c.kind == RW_AGG_COUNT_STAR || c.kind == RW_AGG_COUNT ||
+                    c.kind == RW_AGG_SUM0) {
+                    v.has = true; // init_state = 0
+                }
+                v.is_float = (c.ret_type == RW_T_F64 || c.ret_type == RW_T_F32);
+                g.vstates.push_back(v);
+                g.mstates.emplace_back(minput_order(c));
+            }
+            it = groups.emplace(key, std::move(g)).first;
+        }
+        if (!dirty.count(key)) {
+            dirty.emplace(key, true);
+            dirty_order.push_back(key);
+        }
+        return it->second;
+    }
+
+    int apply_row(AggGroupState& g, const ChunkView& cv, size_t r, bool retract) {
+        for (size_t ci = 0; ci < calls.size(); ci++) {
+            const auto& c = calls[ci];
+            if (call_is_minput[ci]) {
+                // minput materializes the row keyed by [value, stream_key...]
+                Row key;
+                key.push_back(cv.at(r, c.arg));
+                for (auto sk : stream_key) key.push_back(cv.at(r, sk));
+                auto& m = g.mstates[ci].entries;
+                if (!retract) {
+                    m.emplace(std::move(key), cv.at(r, c.arg));
+                } else {
+                    auto it = m.find(key);
+                    if (it != m.end()) m.erase(it);
+                }
+                continue;
+            }
+            auto& v = g.vstates[ci];
+            switch (c.kind) {
+                case RW_AGG_COUNT_STAR:
+                    v.i += retract ? -1 : 1; // general.rs:159-162 (unchecked)
+                    break;
+                case RW_AGG_COUNT: {
+                    Datum a = cv.at(r, c.arg);
+                    if (!a.null) v.i += retract ? -1 : 1; // :154-157
+                    break;
+                }
+                case RW_AGG_SUM:
+                case RW_AGG_SUM0: {
+                    Datum a = cv.at(r, c.arg);
+                    if (a.null) break;
+                    if (v.is_float) {
+                        double x = type_is_float(input_types[c.arg]) ? a.d : (double)a.i;
+                        v.d = retract ? v.d - x : v.d + x;
+                        v.has = true;
+                    } else {
+                        int64_t out;
+                        bool ovf = retract ? __builtin_sub_overflow(v.i, a.i, &out)
+                                           : __builtin_add_overflow(v.i, a.i, &out);
+                        if (ovf) {
+                            g_err = "sum out of range (general.rs:33-40)";
+                            return RW_E_OVERFLOW;
+                        }
+                        v.i = out;
+                        v.has = true;
+                    }
+                    break;
+                }
+                case RW_AGG_MIN:
+                case RW_AGG_MAX: {
+                    // value state — append-only only (agg_state.rs:49-56)
+                    Datum a = cv.at(r, c.arg);
+                    if (a.null) break;
+                    uint8_t t = input_types[c.arg];
+                    Datum curd;
+                    if (v.has) {
+                        curd = v.is_float ? Datum::of_d(v.d) : Datum::of_i(v.i);
+                        int cmpv = datum_cmp(a, curd, t);
+                        bool take = (c.kind == RW_AGG_MIN) ? (cmpv < 0) : (cmpv > 0);
+                        if (!take) break;
+                    }
+                    v.has = true;
+                    if (type_is_float(t)) v.d = a.d;
+                    else v.i = a.i;
+                    break;
+                }
+            }
+        }
+        return RW_OK;
+    }
+
+    int push_chunk(const RwChunk* chunk) {
+        ChunkView cv{chunk};
+        for (size_t r = 0; r < cv.n_rows(); r++) {
+            if (!cv.visible(r)) continue;
+            Row key;
+            key.reserve(group_key.size());
+            for (auto k : group_key) key.push_back(cv.at(r, k));
+            auto& g = touch(key);
+            uint8_t op = cv.op(r);
+            bool retract = (op == RW_OP_DELETE || op == RW_OP_UPDATE_DELETE);
+            int rc = apply_row(g, cv, r, retract);
+            if (rc != RW_OK) return rc;
+        }
+        return RW_OK;
+    }
+
+    int64_t row_count_of(const Row& outputs) const {
+        // row_count_of (agg_group.rs:55-79): clamp negatives to 0
+        const Datum& d = outputs[row_count_index];
+        if (d.null) return 0; // "should not be NULL"; tolerate like release mode
+        return d.i < 0 ? 0 : d.i;
+    }
+
+    // get_outputs (agg_group.rs:431-467)
+    Row get_outputs(AggGroupState& g) {
+        // current row count from the count(*) value state
+        int64_t rc = g.vstates[row_count_index].i;
+        if (rc < 0) rc = 0;
+        if (rc == 0) {
+            // reset value states only (agg_state.rs:149-155)
+            for (size_t ci = 0; ci < calls.size(); ci++) {
+                if (call_is_minput[ci]) continue;
+                ValueState v;
+                const auto& c = calls[ci];
+                if (c.kind == RW_AGG_COUNT_STAR || c.kind == RW_AGG_COUNT ||
+                    c.kind == RW_AGG_SUM0)
+                    v.has = true;
+                v.is_float = g.vstates[ci].is_float;
+                g.vstates[ci] = v;
+            }
+        }
+        Row out(calls.size());
+        for (size_t ci = 0; ci < calls.size(); ci++) {
+            const auto& c = calls[ci];
+            if (call_is_minput[ci]) {
+                const auto& m = g.mstates[ci].entries;
+                out[ci] = m.empty() ? Datum() : m.begin()->second;
+                continue;
+            }
+            const auto& v = g.vstates[ci];
+            switch (c.kind) {
+                case RW_AGG_COUNT_STAR:
+                case RW_AGG_COUNT:
+                case RW_AGG_SUM0: out[ci] = Datum::of_i(v.i); break;
+                case RW_AGG_SUM:
+                    out[ci] = v.has ? (v.is_float ? Datum::of_d(v.d) : Datum::of_i(v.i))
+                                    : Datum();
+                    break;
+                case RW_AGG_MIN:
+                case RW_AGG_MAX:
+                    if (!v.has) out[ci] = Datum();
+                    else out[ci] = v.is_float ? Datum::of_d(v.d) : Datum::of_i(v.i);
+                    break;
+            }
+        }
+        return out;
+    }
+
+    void emit(uint8_t op, const Row& key, const Row& outputs) {
+        Row row;
+        row.reserve(key.size() + outputs.size());
+        for (auto& d : key) row.push_back(d);
+        for (auto& d : outputs) row.push_back(d);
+        std::unique_ptr<OwnedChunk> full;
+        if (builder.append_row(op, row, &full)) outputs_push(std::move(full));
+    }
+    void outputs_push(std::unique_ptr<OwnedChunk> c) {
+        if (c) outputs.push_back(std::move(c));
+    }
+
+    // flush_data, emit-on-update branch (hash_agg.rs:475-501)
+    int flush(uint64_t /*epoch*/) {
+        for (auto& key : dirty_order) {
+            auto& g = groups[key];
+            Row curr = get_outputs(g);
+            int64_t prev_rc = g.has_prev ? row_count_of(g.prev_outputs) : 0;
+            int64_t curr_rc = row_count_of(curr);
+            // OnlyOutputIfHasInput::infer_change_type (agg_group.rs:131-165)
+            if (prev_rc == 0 && curr_rc == 0) {
+                // nothing
+            } else if (prev_rc == 0) {
+                emit(RW_OP_INSERT, key, curr);
+                g.prev_outputs = curr;
+                g.has_prev = true;
+            } else if (curr_rc == 0) {
+                emit(RW_OP_DELETE, key, g.prev_outputs);
+                g.has_prev = false;
+                g.prev_outputs.clear();
+            } else {
+                std::vector<uint8_t> call_types;
+                for (auto& c : calls) call_types.push_back(c.ret_type);
+                if (!row_eq(g.prev_outputs, curr, call_types)) {
+                    emit(RW_OP_UPDATE_DELETE, key, g.prev_outputs);
+                    emit(RW_OP_UPDATE_INSERT, key, curr);
+                    g.prev_outputs = curr;
+                }
+            }
+        }
+        dirty_order.clear();
+        dirty.clear();
+        outputs_push(builder.take());
+        return RW_OK;
+    }
+
+    RwChunk* poll() {
+        if (outputs.empty()) return nullptr;
+        auto c = std::move(outputs.front());
+        outputs.erase(outputs.begin());
+        return chunk_to_c(*c);
+    }
+};
+
+} // namespace orc
+
+using namespace orc;
+
+extern "C" {
+
+const char* rw_last_error(void) { return g_err.c_str(); }
+void rw_chunk_free(RwChunk* c) { chunk_free_c(c); }
+
+void* rw_hash_agg_create(const RwHashAggDesc* d) {
+    std::vector<uint8_t> out_types;
+    for (uint32_t i = 0; i < d->n_group_key; i++)
+        out_types.push_back(d->input_types[d->group_key_indices[i]]);
+    for (uint32_t i = 0; i < d->n_calls; i++) out_types.push_back(d->calls[i].ret_type);
+    return new HashAggOracle(d, out_types);
+}
+int rw_hash_agg_push_chunk(void* h, const RwChunk* c) {
+    return ((HashAggOracle*)h)->push_chunk(c);
+}
+int rw_hash_agg_flush(void* h, uint64_t epoch) { return ((HashAggOracle*)h)->flush(epoch); }
+RwChunk* rw_hash_agg_poll(void* h) { return ((HashAggOracle*)h)->poll(); }
+void rw_hash_agg_destroy(void* h) { delete (HashAggOracle*)h; }
+
+} // extern "C"

@@ -1,0 +1,350 @@
+// oracle/oracle_join.cpp — CPU restatement of HashJoinExecutor.
+// ORACLE — TEST INFRASTRUCTURE ONLY (see common.hpp header note).
+//
+// Restates (reference under /root/reference):
+//  - eq_join_oneside / handle_match_rows / handle_match_row
+//    (src/stream/src/executor/hash_join.rs:949-1374): per visible probe row —
+//    null-safe NeverMatch check (:1004-1016; NeverMatch rows are forwarded
+//    and NOT written to own state, :1143-1152), matched-row loop in
+//    memcomparable deduped-pk order (cache BTreeMap join/join_row_set.rs /
+//    state-table prefix scan join/hash_join.rs:263-288 iterate identically),
+//    non-equi condition (:1294-1302, NULL ⇒ false :1388-1403), degree
+//    bookkeeping: on Insert emit BEFORE the matched row's degree increment,
+//    on Delete emit AFTER the decrement (:1311-1342; update_degree
+//    join/hash_join.rs:355-380), probe-row forwarding by its own match count
+//    (:1216-1227), append-only optimize: delete the single matched row and
+//    skip own insert (:1241-1245,1359-1364).
+//  - Output op selection per join type: JoinChunkBuilder
+//    (join/builder.rs:153-318) incl. outer-side NULL transitions and
+//    eliminate_adjacent_noop_update on every yielded chunk (:166-168).
+//  - State mutation: JoinHashMap::insert/delete keyed by (join key,
+//    memcmp deduped pk) (join/hash_join.rs:578-681). Duplicate-pk inserts
+//    overwrite (non-strict consistency mode, src/stream/src/consistency.rs).
+#include <algorithm>
+#include <map>
+#include <memory>
+#include <string>
+#include <vector>
+
+#include "../include/rw_stream.h"
+#include "common.hpp"
+
+namespace orc {
+
+extern thread_local std::string g_err; // defined in oracle_agg.cpp
+
+// join type predicates (executor/join/mod.rs:103-165)
+static bool is_outer_side(uint8_t t, int side) {
+    return t == RW_JOIN_FULL_OUTER || (t == RW_JOIN_LEFT_OUTER && side == RW_SIDE_LEFT) ||
+           (t == RW_JOIN_RIGHT_OUTER && side == RW_SIDE_RIGHT);
+}
+static bool outer_side_null(uint8_t t, int side) {
+    return t == RW_JOIN_FULL_OUTER || (t == RW_JOIN_LEFT_OUTER && side == RW_SIDE_RIGHT) ||
+           (t == RW_JOIN_RIGHT_OUTER && side == RW_SIDE_LEFT);
+}
+static bool forward_exactly_once(uint8_t t, int side) {
+    return ((t == RW_JOIN_LEFT_SEMI || t == RW_JOIN_LEFT_ANTI) && side == RW_SIDE_LEFT) ||
+           ((t == RW_JOIN_RIGHT_SEMI || t == RW_JOIN_RIGHT_ANTI) && side == RW_SIDE_RIGHT);
+}
+static bool only_forward_matched_side(uint8_t t, int side) {
+    return ((t == RW_JOIN_LEFT_SEMI || t == RW_JOIN_LEFT_ANTI) && side == RW_SIDE_RIGHT) ||
+           ((t == RW_JOIN_RIGHT_SEMI || t == RW_JOIN_RIGHT_ANTI) && side == RW_SIDE_LEFT);
+}
+static bool is_semi(uint8_t t) { return t == RW_JOIN_LEFT_SEMI || t == RW_JOIN_RIGHT_SEMI; }
+static bool is_anti(uint8_t t) { return t == RW_JOIN_LEFT_ANTI || t == RW_JOIN_RIGHT_ANTI; }
+static bool need_degree(uint8_t t, int side) {
+    // need_left_degree / need_right_degree (join/mod.rs:153-165)
+    if (side == RW_SIDE_LEFT)
+        return t == RW_JOIN_FULL_OUTER || t == RW_JOIN_LEFT_OUTER || t == RW_JOIN_LEFT_ANTI ||
+               t == RW_JOIN_LEFT_SEMI;
+    return t == RW_JOIN_FULL_OUTER || t == RW_JOIN_RIGHT_OUTER || t == RW_JOIN_RIGHT_ANTI ||
+           t == RW_JOIN_RIGHT_SEMI;
+}
+
+struct JoinEntry {
+    Row row;
+    uint64_t degree = 0;
+};
+
+struct JoinSideState {
+    std::vector<uint32_t> key_idx;
+    std::vector<uint32_t> pk_idx; // deduped pk indices into the input row
+    std::vector<uint8_t> types;
+    bool need_deg = false;
+    // jk → (pk → entry); both in memcomparable (ASC, NULLs largest) order
+    std::map<Row, std::map<Row, JoinEntry, RowOrderLess>, RowOrderLess> table;
+    RowOrderLess pk_less;
+};
+
+// The join-type/side-dependent output builder (join/builder.rs:153-318).
+struct JoinOutBuilder {
+    ChunkBuilder b;
+    std::vector<std::pair<uint32_t, uint32_t>> update_to_output, matched_to_output;
+    std::vector<std::unique_ptr<OwnedChunk>>* outputs;
+    JoinOutBuilder(size_t chunk_size, std::vector<uint8_t> out_types,
+                   std::vector<std::pair<uint32_t, uint32_t>> u2o,
+                   std::vector<std::pair<uint32_t, uint32_t>> m2o,
+                   std::vector<std::unique_ptr<OwnedChunk>>* out)
+        // chunk size forced >= 2 (join/builder.rs:44-47)
+        : b(std::max<size_t>(chunk_size, 2), out_types),
+          update_to_output(std::move(u2o)),
+          matched_to_output(std::move(m2o)),
+          outputs(out) {}
+
+    size_t width() const { return b.types.size(); }
+    void post(std::unique_ptr<OwnedChunk> c) {
+        if (!c) return;
+        eliminate_adjacent_noop_update(*c); // JoinChunkBuilder::post_process
+        outputs->push_back(std::move(c));
+    }
+    void append_row(uint8_t op, const Row& upd, const Row& match) {
+        Row out(width());
+        for (auto& [ui, oi] : update_to_output) out[oi] = upd[ui];
+        for (auto& [mi, oi] : matched_to_output) out[oi] = match[mi];
+        std::unique_ptr<OwnedChunk> full;
+        if (b.append_row(op, out, &full)) post(std::move(full));
+    }
+    void append_row_update(uint8_t op, const Row& upd) {
+        Row out(width()); // other side NULL
+        for (auto& [ui, oi] : update_to_output) out[oi] = upd[ui];
+        std::unique_ptr<OwnedChunk> full;
+        if (b.append_row(op, out, &full)) post(std::move(full));
+    }
+    void append_row_matched(uint8_t op, const Row& match) {
+        Row out(width());
+        for (auto& [mi, oi] : matched_to_output) out[oi] = match[mi];
+        std::unique_ptr<OwnedChunk> full;
+        if (b.append_row(op, out, &full)) post(std::move(full));
+    }
+    void take() { post(b.take()); }
+};
+
+struct HashJoinOracle {
+    RwHashJoinDesc d_store;
+    uint8_t T;
+    bool append_only;
+    std::vector<uint8_t> null_safe;
+    std::vector<uint8_t> types_l, types_r, concat_types, out_types;
+    std::vector<uint32_t> output_indices;
+    JoinSideState side[2];
+    std::vector<std::pair<uint32_t, uint32_t>> l2o, m_r2o; // left/right → output
+    std::vector<std::unique_ptr<OwnedChunk>> outputs;
+
+    HashJoinOracle(const RwHashJoinDesc* d) {
+        T = d->join_type;
+        append_only = d->append_only;
+        null_safe.assign(d->null_safe, d->null_safe + d->n_key);
+        types_l.assign(d->types_l, d->types_l + d->n_cols_l);
+        types_r.assign(d->types_r, d->types_r + d->n_cols_r);
+        concat_types = types_l;
+        concat_types.insert(concat_types.end(), types_r.begin(), types_r.end());
+        output_indices.assign(d->output_indices, d->output_indices + d->n_output);
+        for (auto i : output_indices) out_types.push_back(concat_types[i]);
+        d_store = *d;
+
+        side[0].key_idx.assign(d->key_l, d->key_l + d->n_key);
+        side[1].key_idx.assign(d->key_r, d->key_r + d->n_key);
+        side[0].pk_idx.assign(d->pk_l, d->pk_l + d->n_pk_l);
+        side[1].pk_idx.assign(d->pk_r, d->pk_r + d->n_pk_r);
+        side[0].types = types_l;
+        side[1].types = types_r;
+        side[0].need_deg = need_degree(T, RW_SIDE_LEFT);
+        side[1].need_deg = need_degree(T, RW_SIDE_RIGHT);
+        for (int s = 0; s < 2; s++) {
+            RowOrderLess key_less, pk_less;
+            for (auto k : side[s].key_idx) key_less.order.push_back({side[s].types[k], false});
+            for (auto p : side[s].pk_idx) pk_less.order.push_back({side[s].types[p], false});
+            side[s].table = decltype(side[s].table)(key_less);
+            side[s].pk_less = pk_less;
+        }
+        // get_i2o_mapping (join/builder.rs:63-81)
+        for (uint32_t oi = 0; oi < output_indices.size(); oi++) {
+            uint32_t idx = output_indices[oi];
+            if (idx < types_l.size()) l2o.push_back({idx, oi});
+            else m_r2o.push_back({idx - (uint32_t)types_l.size(), oi});
+        }
+    }
+
+    bool cond_ok(const Row& upd, const Row& match, int probe_side) const {
+        if (!d_store.has_cond) return true;
+        // row_concat (hash_join.rs:917-932): left part at 0, right at n_cols_l
+        const Row& lrow = probe_side == RW_SIDE_LEFT ? upd : match;
+        const Row& rrow = probe_side == RW_SIDE_LEFT ? match : upd;
+        auto at = [&](uint32_t i) -> const Datum& {
+            return i < types_l.size() ? lrow[i] : rrow[i - types_l.size()];
+        };
+        const Datum& a = at(d_store.cond_l);
+        const Datum& b = at(d_store.cond_r);
+        if (a.null || b.null) return false; // NULL comparison ⇒ false
+        uint8_t t = concat_types[d_store.cond_l];
+        int c = datum_cmp(a, b, t);
+        switch (d_store.cond_op) {
+            case RW_CMP_LT: return c < 0;
+            case RW_CMP_LE: return c <= 0;
+            case RW_CMP_GT: return c > 0;
+            case RW_CMP_GE: return c >= 0;
+        }
+        return false;
+    }
+
+    int push_chunk(int S, const RwChunk* chunk) {
+        ChunkView cv{chunk};
+        JoinSideState& upd_side = side[S];
+        JoinSideState& match_side = side[1 - S];
+        // builder: update = probe side mapping, matched = other side
+        JoinOutBuilder b(d_store.chunk_size, out_types, S == RW_SIDE_LEFT ? l2o : m_r2o,
+                         S == RW_SIDE_LEFT ? m_r2o : l2o, &outputs);
+
+        for (size_t r = 0; r < cv.n_rows(); r++) {
+            if (!cv.visible(r)) continue;
+            uint8_t in_op = cv.op(r);
+            bool is_insert = (in_op == RW_OP_INSERT || in_op == RW_OP_UPDATE_INSERT);
+            uint8_t op = is_insert ? RW_OP_INSERT : RW_OP_DELETE; // U± downgraded
+            Row row = cv.row(r);
+
+            // null-safe check (hash_join.rs:1004-1016)
+            bool never_match = false;
+            Row key(upd_side.key_idx.size());
+            for (size_t i = 0; i < upd_side.key_idx.size(); i++) {
+                key[i] = row[upd_side.key_idx[i]];
+                if (key[i].null && !null_safe[i]) never_match = true;
+            }
+            if (never_match) {
+                if (is_anti(T) && forward_exactly_once(T, S)) b.append_row_update(op, row);
+                else if (is_outer_side(T, S)) b.append_row_update(op, row);
+                continue; // no state write (hash_join.rs:1143-1152)
+            }
+
+            uint64_t degree = 0;
+            bool have_ao_match = false;
+            Row ao_match_pk;
+            auto mit = match_side.table.find(key);
+            if (mit != match_side.table.end()) {
+                for (auto& [mpk, entry] : mit->second) {
+                    if (cond_ok(row, entry.row, S)) {
+                        degree += 1;
+                        if (is_insert) {
+                            // emit BEFORE degree update (hash_join.rs:1311-1317)
+                            if (!forward_exactly_once(T, S)) with_match(b, op, row, entry);
+                            if (match_side.need_deg) entry.degree += 1;
+                        } else {
+                            // degree update BEFORE emit (hash_join.rs:1334-1342)
+                            if (match_side.need_deg) entry.degree -= 1;
+                            if (!forward_exactly_once(T, S)) with_match(b, op, row, entry);
+                        }
+                    }
+                    if (append_only) {
+                        // hash_join.rs:1359-1364: jk ⊇ pk ⇒ at most one match
+                        have_ao_match = true;
+                        ao_match_pk = mpk;
+                    }
+                }
+            }
+
+            if (degree == 0) {
+                // forward_if_not_matched (join/builder.rs:303-312)
+                if ((is_anti(T) && forward_exactly_once(T, S)) || is_outer_side(T, S))
+                    b.append_row_update(op, row);
+            } else if (is_semi(T) && forward_exactly_once(T, S)) {
+                // forward_exactly_once_if_matched (join/builder.rs:287-300)
+                b.append_row_update(op, row);
+            }
+
+            if (append_only && have_ao_match && is_insert) {
+                // delete matched row, skip own insert (hash_join.rs:1241-1245)
+                mit->second.erase(ao_match_pk);
+                if (mit->second.empty()) match_side.table.erase(mit);
+                continue;
+            }
+
+            // own-state update (join/hash_join.rs:578-681)
+            Row pk(upd_side.pk_idx.size());
+            for (size_t i = 0; i < upd_side.pk_idx.size(); i++) pk[i] = row[upd_side.pk_idx[i]];
+            if (is_insert) {
+                auto& m = side_entry(upd_side, key);
+                m[pk] = JoinEntry{row, degree};
+            } else {
+                auto it = upd_side.table.find(key);
+                if (it != upd_side.table.end()) {
+                    it->second.erase(pk);
+                    if (it->second.empty()) upd_side.table.erase(it);
+                }
+            }
+        }
+        b.take();
+        return RW_OK;
+    }
+
+    std::map<Row, JoinEntry, RowOrderLess>& side_entry(JoinSideState& s, const Row& key) {
+        auto it = s.table.find(key);
+        if (it == s.table.end())
+            it = s.table.emplace(key, std::map<Row, JoinEntry, RowOrderLess>(s.pk_less)).first;
+        return it->second;
+    }
+
+    // with_match_on_insert / with_match_on_delete (join/builder.rs:173-284).
+    // `op` is RW_OP_INSERT or RW_OP_DELETE; `entry.degree` is pre-increment
+    // for insert, post-decrement for delete, so is_zero_degree == (degree==0).
+    void with_match(JoinOutBuilder& b, uint8_t op, const Row& row, const JoinEntry& entry) {
+        bool zero = entry.degree == 0;
+        if (op == RW_OP_INSERT) {
+            if (is_anti(T)) {
+                if (zero && only_forward_matched_side(T, cur_side))
+                    b.append_row_matched(RW_OP_DELETE, entry.row);
+            } else if (is_semi(T)) {
+                if (zero && only_forward_matched_side(T, cur_side))
+                    b.append_row_matched(RW_OP_INSERT, entry.row);
+            } else if (zero && outer_side_null(T, cur_side)) {
+                b.append_row_matched(RW_OP_DELETE, entry.row);
+                b.append_row(RW_OP_INSERT, row, entry.row);
+            } else {
+                b.append_row(RW_OP_INSERT, row, entry.row);
+            }
+        } else {
+            if (is_anti(T)) {
+                if (zero && only_forward_matched_side(T, cur_side))
+                    b.append_row_matched(RW_OP_INSERT, entry.row);
+            } else if (is_semi(T)) {
+                if (zero && only_forward_matched_side(T, cur_side))
+                    b.append_row_matched(RW_OP_DELETE, entry.row);
+            } else if (zero && outer_side_null(T, cur_side)) {
+                b.append_row(RW_OP_DELETE, row, entry.row);
+                b.append_row_matched(RW_OP_INSERT, entry.row);
+            } else {
+                b.append_row(RW_OP_DELETE, row, entry.row);
+            }
+        }
+    }
+
+    int cur_side = 0;
+    int push(int S, const RwChunk* chunk) {
+        cur_side = S;
+        return push_chunk(S, chunk);
+    }
+
+    RwChunk* poll() {
+        if (outputs.empty()) return nullptr;
+        auto c = std::move(outputs.front());
+        outputs.erase(outputs.begin());
+        return chunk_to_c(*c);
+    }
+};
+
+} // namespace orc
+
+using namespace orc;
+
+extern "C" {
+
+void* rw_hash_join_create(const RwHashJoinDesc* d) { return new HashJoinOracle(d); }
+int rw_hash_join_push_chunk(void* h, int side, const RwChunk* c) {
+    return ((HashJoinOracle*)h)->push(side, c);
+}
+int rw_hash_join_flush(void* h, uint64_t) {
+    (void)h; // state commit is a no-op for the in-memory oracle
+    return RW_OK;
+}
+RwChunk* rw_hash_join_poll(void* h) { return ((HashJoinOracle*)h)->poll(); }
+void rw_hash_join_destroy(void* h) { delete (HashJoinOracle*)h; }
+
+} // extern "C"

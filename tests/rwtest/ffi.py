@@ -1,0 +1,403 @@
+"""ctypes bindings for the rw_stream C-ABI (include/rw_stream.h).
+
+Drives either the CPU oracle (oracle/liboracle.so — test infrastructure) or
+the product library (risingwave_amd/librw_amd.so — the MI355X path). The two
+export identical symbols; tests load each via its own CDLL handle.
+"""
+import ctypes as C
+import os
+
+import numpy as np
+
+REPO = os.path.dirname(os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+# type ids (rw_chunk.h)
+T_I64, T_I32, T_F64, T_F32, T_BOOL, T_TS = 0, 1, 2, 3, 4, 5
+OP_INSERT, OP_DELETE, OP_UPDATE_DELETE, OP_UPDATE_INSERT = 0, 1, 2, 3
+OP_BY_TOKEN = {"+": OP_INSERT, "-": OP_DELETE, "U-": OP_UPDATE_DELETE, "U+": OP_UPDATE_INSERT}
+TOKEN_BY_OP = {v: k for k, v in OP_BY_TOKEN.items()}
+
+NP_BY_TYPE = {
+    T_I64: np.int64,
+    T_I32: np.int32,
+    T_F64: np.float64,
+    T_F32: np.float32,
+    T_BOOL: np.uint8,
+    T_TS: np.int64,
+}
+
+AGG_COUNT_STAR, AGG_COUNT, AGG_SUM, AGG_SUM0, AGG_MIN, AGG_MAX = 0, 1, 2, 3, 4, 5
+
+JOIN_INNER, JOIN_LEFT_OUTER, JOIN_RIGHT_OUTER, JOIN_FULL_OUTER = 0, 1, 2, 3
+JOIN_LEFT_SEMI, JOIN_LEFT_ANTI, JOIN_RIGHT_SEMI, JOIN_RIGHT_ANTI = 4, 5, 6, 7
+CMP_LT, CMP_LE, CMP_GT, CMP_GE = 0, 1, 2, 3
+SIDE_LEFT, SIDE_RIGHT = 0, 1
+
+
+class RwColumn(C.Structure):
+    _fields_ = [("type", C.c_uint8), ("valid", C.c_void_p), ("data", C.c_void_p)]
+
+
+class RwChunkC(C.Structure):
+    _fields_ = [
+        ("n_rows", C.c_uint32),
+        ("n_cols", C.c_uint32),
+        ("ops", C.POINTER(C.c_uint8)),
+        ("vis", C.POINTER(C.c_uint8)),
+        ("cols", C.POINTER(RwColumn)),
+    ]
+
+
+class RwAggCall(C.Structure):
+    _fields_ = [("kind", C.c_uint8), ("arg", C.c_int32), ("ret_type", C.c_uint8)]
+
+
+class RwHashAggDesc(C.Structure):
+    _fields_ = [
+        ("n_input_cols", C.c_uint32),
+        ("input_types", C.POINTER(C.c_uint8)),
+        ("n_group_key", C.c_uint32),
+        ("group_key_indices", C.POINTER(C.c_uint32)),
+        ("n_calls", C.c_uint32),
+        ("calls", C.POINTER(RwAggCall)),
+        ("row_count_index", C.c_uint32),
+        ("n_stream_key", C.c_uint32),
+        ("stream_key", C.POINTER(C.c_uint32)),
+        ("chunk_size", C.c_uint32),
+        ("append_only", C.c_uint8),
+    ]
+
+
+class RwHashJoinDesc(C.Structure):
+    _fields_ = [
+        ("join_type", C.c_uint8),
+        ("append_only", C.c_uint8),
+        ("n_key", C.c_uint32),
+        ("key_l", C.POINTER(C.c_uint32)),
+        ("key_r", C.POINTER(C.c_uint32)),
+        ("null_safe", C.POINTER(C.c_uint8)),
+        ("n_cols_l", C.c_uint32),
+        ("types_l", C.POINTER(C.c_uint8)),
+        ("n_cols_r", C.c_uint32),
+        ("types_r", C.POINTER(C.c_uint8)),
+        ("n_pk_l", C.c_uint32),
+        ("pk_l", C.POINTER(C.c_uint32)),
+        ("n_pk_r", C.c_uint32),
+        ("pk_r", C.POINTER(C.c_uint32)),
+        ("n_output", C.c_uint32),
+        ("output_indices", C.POINTER(C.c_uint32)),
+        ("has_cond", C.c_uint8),
+        ("cond_op", C.c_uint8),
+        ("cond_l", C.c_uint32),
+        ("cond_r", C.c_uint32),
+        ("chunk_size", C.c_uint32),
+    ]
+
+
+def _u32arr(vals):
+    return (C.c_uint32 * len(vals))(*vals)
+
+
+def _u8arr(vals):
+    return (C.c_uint8 * len(vals))(*vals)
+
+
+class Chunk:
+    """Python-side chunk: ops list, vis list (or None), typed numpy columns."""
+
+    def __init__(self, types, ops, cols, valids, vis=None):
+        self.types = list(types)
+        self.ops = np.asarray(ops, dtype=np.uint8)
+        self.cols = [np.ascontiguousarray(c, dtype=NP_BY_TYPE[t]) for c, t in zip(cols, types)]
+        self.valids = [np.ascontiguousarray(v, dtype=np.uint8) for v in valids]
+        self.vis = None if vis is None else np.asarray(vis, dtype=np.uint8)
+
+    @property
+    def n_rows(self):
+        return len(self.ops)
+
+    def to_c(self):
+        n = self.n_rows
+        cols = (RwColumn * len(self.cols))()
+        self._keep = []
+        for i, (t, d, v) in enumerate(zip(self.types, self.cols, self.valids)):
+            cols[i].type = t
+            cols[i].valid = v.ctypes.data
+            cols[i].data = d.ctypes.data
+            self._keep.extend([d, v])
+        ch = RwChunkC()
+        ch.n_rows = n
+        ch.n_cols = len(self.cols)
+        ops = self.ops
+        ch.ops = C.cast(ops.ctypes.data, C.POINTER(C.c_uint8))
+        if self.vis is not None:
+            ch.vis = C.cast(self.vis.ctypes.data, C.POINTER(C.c_uint8))
+        else:
+            ch.vis = None
+        ch.cols = cols
+        self._keep.extend([ops, cols, self.vis])
+        return ch
+
+    def visible_rows(self):
+        """Yield (op_token, row-tuple) for visible rows; NULL -> None."""
+        for r in range(self.n_rows):
+            if self.vis is not None and not self.vis[r]:
+                continue
+            vals = []
+            for t, d, v in zip(self.types, self.cols, self.valids):
+                if not v[r]:
+                    vals.append(None)
+                elif t in (T_F64, T_F32):
+                    vals.append(float(d[r]))
+                else:
+                    vals.append(int(d[r]))
+            yield (TOKEN_BY_OP[int(self.ops[r])], tuple(vals))
+
+
+TYPE_BY_TOKEN = {"I": T_I64, "i": T_I32, "F": T_F64, "f": T_F32, "B": T_BOOL, "TS": T_TS}
+
+
+def from_pretty(s):
+    """Parse the reference test fixture format
+    (common/src/array/data_chunk.rs:646-760 / stream_chunk.rs from_pretty):
+    header of type tokens, then rows `<op> v v v [D]`, `.` = NULL."""
+    lines = [l.strip() for l in s.strip().split("\n") if l.strip()]
+    types = [TYPE_BY_TOKEN[tok] for tok in lines[0].split()]
+    ops, vis = [], []
+    cols = [[] for _ in types]
+    valids = [[] for _ in types]
+    for line in lines[1:]:
+        toks = line.split()
+        op = toks[0]
+        if op in ("U-", "U+", "+", "-"):
+            ops.append(OP_BY_TOKEN[op])
+            toks = toks[1:]
+        else:  # data-chunk style: all inserts
+            ops.append(OP_INSERT)
+        row_vis = 1
+        if toks and toks[-1] == "D":
+            row_vis = 0
+            toks = toks[:-1]
+        vis.append(row_vis)
+        assert len(toks) == len(types), f"bad row: {line}"
+        for i, tok in enumerate(toks):
+            if tok == ".":
+                valids[i].append(0)
+                cols[i].append(0)
+            else:
+                valids[i].append(1)
+                cols[i].append(float(tok) if types[i] in (T_F64, T_F32) else int(tok))
+    if all(v for v in vis):
+        vis = None
+    return Chunk(types, ops, cols, valids, vis)
+
+
+class Lib:
+    """Wrapper over one .so implementing the rw_stream C-ABI."""
+
+    def __init__(self, path):
+        self.lib = C.CDLL(path)
+        L = self.lib
+        L.rw_hash_agg_create.restype = C.c_void_p
+        L.rw_hash_agg_create.argtypes = [C.POINTER(RwHashAggDesc)]
+        L.rw_hash_agg_push_chunk.restype = C.c_int
+        L.rw_hash_agg_push_chunk.argtypes = [C.c_void_p, C.POINTER(RwChunkC)]
+        L.rw_hash_agg_flush.restype = C.c_int
+        L.rw_hash_agg_flush.argtypes = [C.c_void_p, C.c_uint64]
+        L.rw_hash_agg_poll.restype = C.POINTER(RwChunkC)
+        L.rw_hash_agg_poll.argtypes = [C.c_void_p]
+        L.rw_hash_agg_destroy.argtypes = [C.c_void_p]
+        L.rw_hash_join_create.restype = C.c_void_p
+        L.rw_hash_join_create.argtypes = [C.POINTER(RwHashJoinDesc)]
+        L.rw_hash_join_push_chunk.restype = C.c_int
+        L.rw_hash_join_push_chunk.argtypes = [C.c_void_p, C.c_int, C.POINTER(RwChunkC)]
+        L.rw_hash_join_flush.restype = C.c_int
+        L.rw_hash_join_flush.argtypes = [C.c_void_p, C.c_uint64]
+        L.rw_hash_join_poll.restype = C.POINTER(RwChunkC)
+        L.rw_hash_join_poll.argtypes = [C.c_void_p]
+        L.rw_hash_join_destroy.argtypes = [C.c_void_p]
+        L.rw_chunk_free.argtypes = [C.POINTER(RwChunkC)]
+        L.rw_last_error.restype = C.c_char_p
+
+    def last_error(self):
+        return self.lib.rw_last_error().decode()
+
+    def _read_chunk(self, p):
+        ch = p.contents
+        n, m = ch.n_rows, ch.n_cols
+        types, cols, valids = [], [], []
+        for ci in range(m):
+            col = ch.cols[ci]
+            t = col.type
+            types.append(t)
+            dt = NP_BY_TYPE[t]
+            nbytes = n * np.dtype(dt).itemsize
+            data = np.frombuffer(C.string_at(col.data, nbytes), dtype=dt).copy() if n else np.array([], dt)
+            valid = np.frombuffer(C.string_at(col.valid, n), dtype=np.uint8).copy() if n else np.array([], np.uint8)
+            cols.append(data)
+            valids.append(valid)
+        ops = np.frombuffer(C.string_at(ch.ops, n), dtype=np.uint8).copy() if n else np.array([], np.uint8)
+        vis = None
+        if ch.vis:
+            vis = np.frombuffer(C.string_at(ch.vis, n), dtype=np.uint8).copy()
+        self.lib.rw_chunk_free(p)
+        return Chunk(types, ops, cols, valids, vis)
+
+
+class HashAgg:
+    def __init__(self, lib: Lib, input_types, group_key, calls, row_count_index,
+                 stream_key=(), chunk_size=1024, append_only=False):
+        """calls: list of (kind, arg, ret_type)."""
+        self.lib = lib
+        d = RwHashAggDesc()
+        d.n_input_cols = len(input_types)
+        self._it = _u8arr(input_types)
+        d.input_types = self._it
+        d.n_group_key = len(group_key)
+        self._gk = _u32arr(group_key)
+        d.group_key_indices = self._gk
+        d.n_calls = len(calls)
+        self._calls = (RwAggCall * len(calls))()
+        for i, (k, a, rt) in enumerate(calls):
+            self._calls[i].kind = k
+            self._calls[i].arg = a
+            self._calls[i].ret_type = rt
+        d.calls = self._calls
+        d.row_count_index = row_count_index
+        d.n_stream_key = len(stream_key)
+        self._sk = _u32arr(stream_key)
+        d.stream_key = self._sk
+        d.chunk_size = chunk_size
+        d.append_only = 1 if append_only else 0
+        self.h = lib.lib.rw_hash_agg_create(C.byref(d))
+        assert self.h
+
+    def push(self, chunk: Chunk):
+        c = chunk.to_c()
+        rc = self.lib.lib.rw_hash_agg_push_chunk(self.h, C.byref(c))
+        if rc != 0:
+            raise RuntimeError(f"push failed {rc}: {self.lib.last_error()}")
+
+    def flush(self, epoch):
+        rc = self.lib.lib.rw_hash_agg_flush(self.h, epoch)
+        if rc != 0:
+            raise RuntimeError(f"flush failed {rc}: {self.lib.last_error()}")
+
+    def poll_all(self):
+        out = []
+        while True:
+            p = self.lib.lib.rw_hash_agg_poll(self.h)
+            if not p:
+                return out
+            out.append(self.lib._read_chunk(p))
+
+    def close(self):
+        if self.h:
+            self.lib.lib.rw_hash_agg_destroy(self.h)
+            self.h = None
+
+
+class HashJoin:
+    def __init__(self, lib: Lib, join_type, types_l, types_r, key_l, key_r,
+                 pk_l, pk_r, output_indices=None, null_safe=None, cond=None,
+                 chunk_size=1024, append_only=False):
+        """cond: (op, cond_l, cond_r) into the concatenated row, or None."""
+        self.lib = lib
+        d = RwHashJoinDesc()
+        d.join_type = join_type
+        d.append_only = 1 if append_only else 0
+        d.n_key = len(key_l)
+        self._kl = _u32arr(key_l)
+        self._kr = _u32arr(key_r)
+        d.key_l, d.key_r = self._kl, self._kr
+        ns = null_safe or [0] * len(key_l)
+        self._ns = _u8arr(ns)
+        d.null_safe = self._ns
+        d.n_cols_l = len(types_l)
+        self._tl = _u8arr(types_l)
+        d.types_l = self._tl
+        d.n_cols_r = len(types_r)
+        self._tr = _u8arr(types_r)
+        d.types_r = self._tr
+        d.n_pk_l = len(pk_l)
+        self._pl = _u32arr(pk_l)
+        d.pk_l = self._pl
+        d.n_pk_r = len(pk_r)
+        self._pr = _u32arr(pk_r)
+        d.pk_r = self._pr
+        if output_indices is None:
+            if join_type in (JOIN_LEFT_SEMI, JOIN_LEFT_ANTI):
+                output_indices = list(range(len(types_l)))
+            elif join_type in (JOIN_RIGHT_SEMI, JOIN_RIGHT_ANTI):
+                output_indices = list(range(len(types_l), len(types_l) + len(types_r)))
+            else:
+                output_indices = list(range(len(types_l) + len(types_r)))
+        d.n_output = len(output_indices)
+        self._oi = _u32arr(output_indices)
+        d.output_indices = self._oi
+        if cond is not None:
+            d.has_cond = 1
+            d.cond_op, d.cond_l, d.cond_r = cond
+        else:
+            d.has_cond = 0
+            d.cond_op = d.cond_l = d.cond_r = 0
+        d.chunk_size = chunk_size
+        self.h = lib.lib.rw_hash_join_create(C.byref(d))
+        assert self.h
+
+    def push(self, side, chunk: Chunk):
+        c = chunk.to_c()
+        rc = self.lib.lib.rw_hash_join_push_chunk(self.h, side, C.byref(c))
+        if rc != 0:
+            raise RuntimeError(f"push failed {rc}: {self.lib.last_error()}")
+
+    def flush(self, epoch):
+        rc = self.lib.lib.rw_hash_join_flush(self.h, epoch)
+        if rc != 0:
+            raise RuntimeError(f"flush failed {rc}: {self.lib.last_error()}")
+
+    def poll_all(self):
+        out = []
+        while True:
+            p = self.lib.lib.rw_hash_join_poll(self.h)
+            if not p:
+                return out
+            out.append(self.lib._read_chunk(p))
+
+    def close(self):
+        if self.h:
+            self.lib.lib.rw_hash_join_destroy(self.h)
+            self.h = None
+
+
+def rows_multiset(chunks):
+    """All visible (op, row) across chunks, sorted — the per-epoch multiset
+    (parity bar of SURVEY.md: tests/integration_tests/snapshot.rs sorts too)."""
+    rows = []
+    for c in chunks:
+        rows.extend(c.visible_rows())
+    key = lambda r: (r[0], tuple((v is None, v if v is not None else 0) for v in r[1]))
+    return sorted(rows, key=key)
+
+
+def rows_ordered(chunks):
+    """All visible (op, row) across chunks, in yield order (order-exact tests)."""
+    rows = []
+    for c in chunks:
+        rows.extend(c.visible_rows())
+    return rows
+
+
+_oracle = None
+
+
+def oracle():
+    global _oracle
+    if _oracle is None:
+        path = os.path.join(REPO, "oracle", "liboracle.so")
+        if not os.path.exists(path):
+            import subprocess
+
+            subprocess.run(["make", "-C", os.path.join(REPO, "oracle")], check=True)
+        _oracle = Lib(path)
+    return _oracle

@@ -1,0 +1,208 @@
+"""Oracle HashAgg vs the reference's own golden vectors.
+
+Fixtures transcribed verbatim from
+/root/reference/src/stream/tests/integration_tests/hash_agg.rs
+(test_hash_agg_count_sum :22-97, test_hash_agg_min :100-174,
+test_hash_agg_min_append_only :177-256). The reference snapshots are taken
+with sort_chunk(true) (snapshot.rs:29-40), so comparison is the per-epoch
+row multiset — same bar as upstream.
+"""
+from rwtest import ffi
+from rwtest.ffi import (
+    AGG_COUNT_STAR, AGG_MIN, AGG_SUM, T_I64, from_pretty, oracle, rows_multiset,
+)
+
+
+def run_epochs(agg, epochs):
+    """epochs: list of lists of chunks; returns list of per-epoch multisets."""
+    out = []
+    for i, chunks in enumerate(epochs):
+        for c in chunks:
+            agg.push(c)
+        agg.flush(i + 1)
+        out.append(rows_multiset(agg.poll_all()))
+    agg.close()
+    return out
+
+
+def expect(rows):
+    key = lambda r: (r[0], tuple((v is None, v if v is not None else 0) for v in r[1]))
+    return sorted(rows, key=key)
+
+
+def test_hash_agg_count_sum():
+    # hash_agg.rs:22-97
+    agg = ffi.HashAgg(
+        oracle(),
+        input_types=[T_I64, T_I64, T_I64],
+        group_key=[0],
+        calls=[(AGG_COUNT_STAR, -1, T_I64), (AGG_SUM, 1, T_I64), (AGG_SUM, 2, T_I64)],
+        row_count_index=0,
+    )
+    e1 = from_pretty(
+        """ I I I
+        + 1 1 1
+        + 2 2 2
+        + 2 2 2"""
+    )
+    e2 = from_pretty(
+        """ I I I
+        - 1 1 1
+        - 2 2 2 D
+        - 2 2 2
+        + 3 3 3"""
+    )
+    out = run_epochs(agg, [[e1], [e2]])
+    assert out[0] == expect([("+", (1, 1, 1, 1)), ("+", (2, 2, 4, 4))])
+    assert out[1] == expect(
+        [
+            ("+", (3, 1, 3, 3)),
+            ("-", (1, 1, 1, 1)),
+            ("U-", (2, 2, 4, 4)),
+            ("U+", (2, 1, 2, 2)),
+        ]
+    )
+
+
+def test_hash_agg_min():
+    # hash_agg.rs:100-174 — retractable min via materialized-input state
+    agg = ffi.HashAgg(
+        oracle(),
+        input_types=[T_I64, T_I64, T_I64],
+        group_key=[0],
+        calls=[(AGG_COUNT_STAR, -1, T_I64), (AGG_MIN, 1, T_I64)],
+        row_count_index=0,
+        stream_key=[2],
+    )
+    e1 = from_pretty(
+        """ I     I    I
+        + 1   233 1001
+        + 1 23333 1002
+        + 2  2333 1003"""
+    )
+    e2 = from_pretty(
+        """ I     I    I
+        - 1   233 1001
+        - 1 23333 1002 D
+        - 2  2333 1003"""
+    )
+    out = run_epochs(agg, [[e1], [e2]])
+    assert out[0] == expect([("+", (1, 2, 233)), ("+", (2, 1, 2333))])
+    assert out[1] == expect(
+        [("-", (2, 1, 2333)), ("U-", (1, 2, 233)), ("U+", (1, 1, 23333))]
+    )
+
+
+def test_hash_agg_min_append_only():
+    # hash_agg.rs:177-256 — append-only min via value state
+    agg = ffi.HashAgg(
+        oracle(),
+        input_types=[T_I64, T_I64, T_I64],
+        group_key=[0],
+        calls=[(AGG_COUNT_STAR, -1, T_I64), (AGG_MIN, 1, T_I64)],
+        row_count_index=0,
+        stream_key=[2],
+        append_only=True,
+    )
+    e1 = from_pretty(
+        """ I  I  I
+        + 2 5  1000
+        + 1 15 1001
+        + 1 8  1002
+        + 2 5  1003
+        + 2 10 1004"""
+    )
+    e2 = from_pretty(
+        """ I  I  I
+        + 1 20 1005
+        + 1 1  1006
+        + 2 10 1007
+        + 2 20 1008"""
+    )
+    out = run_epochs(agg, [[e1], [e2]])
+    assert out[0] == expect([("+", (1, 2, 8)), ("+", (2, 3, 5))])
+    assert out[1] == expect(
+        [("U-", (1, 2, 8)), ("U-", (2, 3, 5)), ("U+", (1, 4, 1)), ("U+", (2, 5, 5))]
+    )
+
+
+def test_unchanged_group_emits_nothing():
+    # OnlyOutputIfHasInput (agg_group.rs:154-163): update with equal
+    # prev/curr row emits nothing.
+    agg = ffi.HashAgg(
+        oracle(),
+        input_types=[T_I64, T_I64],
+        group_key=[0],
+        calls=[(AGG_COUNT_STAR, -1, T_I64), (AGG_SUM, 1, T_I64)],
+        row_count_index=0,
+    )
+    e1 = from_pretty(" I I\n + 1 5\n + 1 3")
+    # +2 then -2 on the sum column, count unchanged net? count changes 2->...
+    # use a chunk whose net effect is zero on every output
+    e2 = from_pretty(" I I\n + 1 7\n - 1 7")
+    out = run_epochs(agg, [[e1], [e2]])
+    assert out[0] == expect([("+", (1, 2, 8))])
+    assert out[1] == []
+
+
+def test_delete_then_reinsert_group():
+    # group drops to 0 (Delete emitted with prev values), then reappears
+    # (Insert) — agg_group.rs:141-153 + reset-on-zero (agg_group.rs:431-445)
+    agg = ffi.HashAgg(
+        oracle(),
+        input_types=[T_I64, T_I64],
+        group_key=[0],
+        calls=[(AGG_COUNT_STAR, -1, T_I64), (AGG_SUM, 1, T_I64)],
+        row_count_index=0,
+    )
+    e1 = from_pretty(" I I\n + 7 10")
+    e2 = from_pretty(" I I\n - 7 10")
+    e3 = from_pretty(" I I\n + 7 99")
+    out = run_epochs(agg, [[e1], [e2], [e3]])
+    assert out[0] == expect([("+", (7, 1, 10))])
+    assert out[1] == expect([("-", (7, 1, 10))])
+    assert out[2] == expect([("+", (7, 1, 99))])
+
+
+def test_sum_null_skip_and_null_result():
+    # sum skips NULL inputs; count(col) counts non-NULL (general.rs:127-157)
+    agg = ffi.HashAgg(
+        oracle(),
+        input_types=[T_I64, T_I64],
+        group_key=[0],
+        calls=[
+            (ffi.AGG_COUNT_STAR, -1, T_I64),
+            (ffi.AGG_COUNT, 1, T_I64),
+            (AGG_SUM, 1, T_I64),
+        ],
+        row_count_index=0,
+    )
+    e1 = from_pretty(" I I\n + 1 .\n + 1 5")
+    out = run_epochs(agg, [[e1]])
+    assert out[0] == expect([("+", (1, 2, 1, 5))])
+
+
+def test_multi_chunk_epoch_and_chunking():
+    # several chunks per epoch; output respects chunk_size with U-pairs unsplit
+    agg = ffi.HashAgg(
+        oracle(),
+        input_types=[T_I64, T_I64],
+        group_key=[0],
+        calls=[(AGG_COUNT_STAR, -1, T_I64), (AGG_SUM, 1, T_I64)],
+        row_count_index=0,
+        chunk_size=3,
+    )
+    e1 = [from_pretty(" I I\n + %d 1" % k) for k in range(5)]
+    out1 = run_epochs_keep(agg, [e1])
+    assert out1[0] == expect([("+", (k, 1, 1)) for k in range(5)])
+    agg.close()
+
+
+def run_epochs_keep(agg, epochs):
+    out = []
+    for i, chunks in enumerate(epochs):
+        for c in chunks:
+            agg.push(c)
+        agg.flush(i + 1)
+        out.append(rows_multiset(agg.poll_all()))
+    return out
