@@ -1,0 +1,313 @@
+"""Oracle HashJoin vs the reference's own golden vectors.
+
+Fixtures transcribed verbatim from the in-module tests of
+/root/reference/src/stream/src/executor/hash_join.rs (line refs per test).
+The "classical" executor (hash_join.rs:1538-1655): two [i64,i64] inputs,
+join key [0], deduped pk [1], chunk_size 1024, output = all columns
+(one side only for semi/anti); condition default `$1 < $3` (:1530-1536).
+The "append_only" executor (:1657-1737): [i64,i64,i64] inputs, key [0,1],
+deduped pk [], append_only_optimize on.
+
+These single-side-input sequences are deterministic in the reference
+(matched rows iterate in memcomparable pk order), so comparisons are
+order-exact — stronger than the multiset bar. Rows marked D in the
+expected chunks are invisible (noop-update-eliminated) and thus absent
+from visible_rows().
+"""
+from rwtest import ffi
+from rwtest.ffi import (
+    CMP_LT, JOIN_FULL_OUTER, JOIN_INNER, JOIN_LEFT_ANTI, JOIN_LEFT_OUTER,
+    JOIN_LEFT_SEMI, JOIN_RIGHT_OUTER, JOIN_RIGHT_SEMI, SIDE_LEFT, SIDE_RIGHT,
+    T_I64, from_pretty, oracle, rows_ordered,
+)
+
+I2 = [T_I64, T_I64]
+I3 = [T_I64, T_I64, T_I64]
+
+
+def classical(join_type, null_safe=False, cond=False):
+    return ffi.HashJoin(
+        oracle(), join_type, I2, I2, key_l=[0], key_r=[0], pk_l=[1], pk_r=[1],
+        null_safe=[1 if null_safe else 0],
+        cond=(CMP_LT, 1, 3) if cond else None,
+    )
+
+
+def append_only(join_type):
+    return ffi.HashJoin(
+        oracle(), join_type, I3, I3, key_l=[0, 1], key_r=[0, 1],
+        pk_l=[], pk_r=[], null_safe=[0, 0], append_only=True,
+    )
+
+
+def push(j, side, pretty):
+    j.push(side, from_pretty(pretty))
+    return rows_ordered(j.poll_all())
+
+
+def rows(spec):
+    """spec: list of (op, *values) tuples."""
+    return [(s[0], tuple(s[1:])) for s in spec]
+
+
+def test_inner_join():
+    # hash_join.rs:1831-1898
+    j = classical(JOIN_INNER)
+    assert push(j, SIDE_LEFT, " I I\n + 1 4\n + 2 5\n + 3 6") == []
+    assert push(j, SIDE_LEFT, " I I\n + 3 8\n - 3 8") == []
+    assert push(j, SIDE_RIGHT, " I I\n + 2 7\n + 4 8\n + 6 9") == rows(
+        [("+", 2, 5, 2, 7)]
+    )
+    assert push(j, SIDE_RIGHT, " I I\n + 3 10\n + 6 11") == rows([("+", 3, 6, 3, 10)])
+    j.close()
+
+
+def test_null_safe_inner_join():
+    # hash_join.rs:1901-1968
+    j = classical(JOIN_INNER, null_safe=True)
+    assert push(j, SIDE_LEFT, " I I\n + 1 4\n + 2 5\n + . 6") == []
+    assert push(j, SIDE_LEFT, " I I\n + . 8\n - . 8") == []
+    assert push(j, SIDE_RIGHT, " I I\n + 2 7\n + 4 8\n + 6 9") == rows(
+        [("+", 2, 5, 2, 7)]
+    )
+    out = push(j, SIDE_RIGHT, " I I\n + . 10\n + 6 11")
+    assert out == [("+", (None, 6, None, 10))]
+    j.close()
+
+
+def test_non_null_safe_null_keys_never_match():
+    # the null-bitmap subset check (hash_join.rs:1004-1016): NULL keys on a
+    # non-null-safe join never match and are not stored
+    j = classical(JOIN_INNER)
+    assert push(j, SIDE_LEFT, " I I\n + . 6") == []
+    assert push(j, SIDE_RIGHT, " I I\n + . 10") == []
+    j.close()
+
+
+def test_left_semi_join():
+    # hash_join.rs:1971-2078
+    j = classical(JOIN_LEFT_SEMI)
+    assert push(j, SIDE_LEFT, " I I\n + 1 4\n + 2 5\n + 3 6") == []
+    assert push(j, SIDE_LEFT, " I I\n + 3 8\n - 3 8") == []
+    assert push(j, SIDE_RIGHT, " I I\n + 2 7\n + 4 8\n + 6 9") == rows([("+", 2, 5)])
+    assert push(j, SIDE_RIGHT, " I I\n + 3 10\n + 6 11") == rows([("+", 3, 6)])
+    assert push(j, SIDE_LEFT, " I I\n + 6 10") == rows([("+", 6, 10)])
+    assert push(j, SIDE_RIGHT, " I I\n - 6 11") == []
+    assert push(j, SIDE_RIGHT, " I I\n - 6 9") == rows([("-", 6, 10)])
+    j.close()
+
+
+def test_right_semi_join():
+    # hash_join.rs:2410-2517 (mirror of left semi)
+    j = classical(JOIN_RIGHT_SEMI)
+    assert push(j, SIDE_RIGHT, " I I\n + 1 4\n + 2 5\n + 3 6") == []
+    assert push(j, SIDE_RIGHT, " I I\n + 3 8\n - 3 8") == []
+    assert push(j, SIDE_LEFT, " I I\n + 2 7\n + 4 8\n + 6 9") == rows([("+", 2, 5)])
+    assert push(j, SIDE_LEFT, " I I\n + 3 10\n + 6 11") == rows([("+", 3, 6)])
+    assert push(j, SIDE_RIGHT, " I I\n + 6 10") == rows([("+", 6, 10)])
+    assert push(j, SIDE_LEFT, " I I\n - 6 11") == []
+    assert push(j, SIDE_LEFT, " I I\n - 6 9") == rows([("-", 6, 10)])
+    j.close()
+
+
+def test_left_anti_join():
+    # hash_join.rs:2520-2647
+    j = classical(JOIN_LEFT_ANTI)
+    assert push(j, SIDE_LEFT, " I I\n + 1 4\n + 2 5\n + 3 6") == rows(
+        [("+", 1, 4), ("+", 2, 5), ("+", 3, 6)]
+    )
+    assert push(j, SIDE_LEFT, " I I\n + 3 8\n - 3 8") == []  # D D eliminated
+    assert push(j, SIDE_RIGHT, " I I\n + 2 7\n + 4 8\n + 6 9") == rows([("-", 2, 5)])
+    assert push(j, SIDE_RIGHT, " I I\n + 3 10\n + 6 11\n + 1 2\n + 1 3") == rows(
+        [("-", 3, 6), ("-", 1, 4)]
+    )
+    assert push(j, SIDE_LEFT, " I I\n + 9 10") == rows([("+", 9, 10)])
+    assert push(j, SIDE_RIGHT, " I I\n - 1 2") == []
+    assert push(j, SIDE_RIGHT, " I I\n - 1 3") == rows([("+", 1, 4)])
+    j.close()
+
+
+def test_inner_join_append_only():
+    # hash_join.rs:2191-2261
+    j = append_only(JOIN_INNER)
+    assert push(j, SIDE_LEFT, " I I I\n + 1 4 1\n + 2 5 2\n + 3 6 3") == []
+    assert push(j, SIDE_LEFT, " I I I\n + 4 9 4\n + 5 10 5") == []
+    assert push(j, SIDE_RIGHT, " I I I\n + 2 5 1\n + 4 9 2\n + 6 9 3") == rows(
+        [("+", 2, 5, 2, 2, 5, 1), ("+", 4, 9, 4, 4, 9, 2)]
+    )
+    assert push(j, SIDE_RIGHT, " I I I\n + 1 4 4\n + 3 6 5") == rows(
+        [("+", 1, 4, 1, 1, 4, 4), ("+", 3, 6, 3, 3, 6, 5)]
+    )
+    j.close()
+
+
+def test_left_semi_join_append_only():
+    # hash_join.rs:2264-2334
+    j = append_only(JOIN_LEFT_SEMI)
+    assert push(j, SIDE_LEFT, " I I I\n + 1 4 1\n + 2 5 2\n + 3 6 3") == []
+    assert push(j, SIDE_LEFT, " I I I\n + 4 9 4\n + 5 10 5") == []
+    assert push(j, SIDE_RIGHT, " I I I\n + 2 5 1\n + 4 9 2\n + 6 9 3") == rows(
+        [("+", 2, 5, 2), ("+", 4, 9, 4)]
+    )
+    assert push(j, SIDE_RIGHT, " I I I\n + 1 4 4\n + 3 6 5") == rows(
+        [("+", 1, 4, 1), ("+", 3, 6, 3)]
+    )
+    j.close()
+
+
+def test_inner_join_with_barrier():
+    # hash_join.rs:2780-2872 — processed order after barrier alignment:
+    # l1, r1, <barrier>, l2, r2
+    j = classical(JOIN_INNER)
+    assert push(j, SIDE_LEFT, " I I\n + 1 4\n + 2 5\n + 3 6") == []
+    assert push(j, SIDE_RIGHT, " I I\n + 2 7\n + 4 8\n + 6 9") == rows(
+        [("+", 2, 5, 2, 7)]
+    )
+    j.flush(2)
+    assert push(j, SIDE_LEFT, " I I\n + 6 8\n + 3 8") == rows([("+", 6, 8, 6, 9)])
+    assert push(j, SIDE_RIGHT, " I I\n + 3 10\n + 6 11") == rows(
+        [("+", 3, 6, 3, 10), ("+", 3, 8, 3, 10), ("+", 6, 8, 6, 11)]
+    )
+    j.close()
+
+
+def test_inner_join_with_null_and_barrier():
+    # hash_join.rs:2875-2967 — NULLs in value columns
+    j = classical(JOIN_INNER)
+    assert push(j, SIDE_LEFT, " I I\n + 1 4\n + 2 .\n + 3 .") == []
+    out = push(j, SIDE_RIGHT, " I I\n + 2 7\n + 4 8\n + 6 9")
+    assert out == [("+", (2, None, 2, 7))]
+    j.flush(2)
+    out = push(j, SIDE_LEFT, " I I\n + 6 .\n + 3 8")
+    assert out == [("+", (6, None, 6, 9))]
+    out = push(j, SIDE_RIGHT, " I I\n + 3 10\n + 6 11")
+    assert out == [
+        ("+", (3, 8, 3, 10)),
+        ("+", (3, None, 3, 10)),
+        ("+", (6, None, 6, 11)),
+    ]
+    j.close()
+
+
+def test_left_outer_join():
+    # hash_join.rs:2970-3051
+    j = classical(JOIN_LEFT_OUTER)
+    assert push(j, SIDE_LEFT, " I I\n + 1 4\n + 2 5\n + 3 6") == [
+        ("+", (1, 4, None, None)),
+        ("+", (2, 5, None, None)),
+        ("+", (3, 6, None, None)),
+    ]
+    assert push(j, SIDE_LEFT, " I I\n + 3 8\n - 3 8") == []  # D D
+    assert push(j, SIDE_RIGHT, " I I\n + 2 7\n + 4 8\n + 6 9") == [
+        ("-", (2, 5, None, None)),
+        ("+", (2, 5, 2, 7)),
+    ]
+    assert push(j, SIDE_RIGHT, " I I\n + 3 10\n + 6 11") == [
+        ("-", (3, 6, None, None)),
+        ("+", (3, 6, 3, 10)),
+    ]
+    j.close()
+
+
+def test_null_safe_left_outer_join():
+    # hash_join.rs:3054-3135
+    j = classical(JOIN_LEFT_OUTER, null_safe=True)
+    assert push(j, SIDE_LEFT, " I I\n + 1 4\n + 2 5\n + . 6") == [
+        ("+", (1, 4, None, None)),
+        ("+", (2, 5, None, None)),
+        ("+", (None, 6, None, None)),
+    ]
+    assert push(j, SIDE_LEFT, " I I\n + . 8\n - . 8") == []
+    assert push(j, SIDE_RIGHT, " I I\n + 2 7\n + 4 8\n + 6 9") == [
+        ("-", (2, 5, None, None)),
+        ("+", (2, 5, 2, 7)),
+    ]
+    assert push(j, SIDE_RIGHT, " I I\n + . 10\n + 6 11") == [
+        ("-", (None, 6, None, None)),
+        ("+", (None, 6, None, 10)),
+    ]
+    j.close()
+
+
+def test_right_outer_join():
+    # hash_join.rs:3138-3203
+    j = classical(JOIN_RIGHT_OUTER)
+    assert push(j, SIDE_LEFT, " I I\n + 1 4\n + 2 5\n + 3 6") == []
+    assert push(j, SIDE_LEFT, " I I\n + 3 8\n - 3 8") == []
+    assert push(j, SIDE_RIGHT, " I I\n + 2 7\n + 4 8\n + 6 9") == [
+        ("+", (2, 5, 2, 7)),
+        ("+", (None, None, 4, 8)),
+        ("+", (None, None, 6, 9)),
+    ]
+    assert push(j, SIDE_RIGHT, " I I\n + 5 10\n - 5 10") == []  # D D
+    j.close()
+
+
+def test_full_outer_join():
+    # hash_join.rs:3366-3449
+    j = classical(JOIN_FULL_OUTER)
+    assert push(j, SIDE_LEFT, " I I\n + 1 4\n + 2 5\n + 3 6") == [
+        ("+", (1, 4, None, None)),
+        ("+", (2, 5, None, None)),
+        ("+", (3, 6, None, None)),
+    ]
+    assert push(j, SIDE_LEFT, " I I\n + 3 8\n - 3 8") == []
+    assert push(j, SIDE_RIGHT, " I I\n + 2 7\n + 4 8\n + 6 9") == [
+        ("-", (2, 5, None, None)),
+        ("+", (2, 5, 2, 7)),
+        ("+", (None, None, 4, 8)),
+        ("+", (None, None, 6, 9)),
+    ]
+    assert push(j, SIDE_RIGHT, " I I\n + 5 10\n - 5 10") == []
+    j.close()
+
+
+def test_full_outer_join_update():
+    # hash_join.rs:3452-3510 — NULL transitions + noop elimination leave
+    # exactly a delete+insert pair
+    j = classical(JOIN_FULL_OUTER)
+    assert push(j, SIDE_LEFT, " I I\n + 1 1") == [("+", (1, 1, None, None))]
+    assert push(j, SIDE_RIGHT, " I I\n + 1 1") == [
+        ("-", (1, 1, None, None)),
+        ("+", (1, 1, 1, 1)),
+    ]
+    assert push(j, SIDE_LEFT, " I I\n - 1 1\n + 1 2") == [
+        ("-", (1, 1, 1, 1)),
+        ("+", (1, 2, 1, 1)),
+    ]
+    j.close()
+
+
+def test_full_outer_join_nonequi():
+    # hash_join.rs:3513-3606 — incl. regression #2420 (forward once on
+    # multiple condition-failing matches; forward on empty entry)
+    j = classical(JOIN_FULL_OUTER, cond=True)
+    assert push(j, SIDE_LEFT, " I I\n + 1 4\n + 2 5\n + 3 6\n + 3 7") == [
+        ("+", (1, 4, None, None)),
+        ("+", (2, 5, None, None)),
+        ("+", (3, 6, None, None)),
+        ("+", (3, 7, None, None)),
+    ]
+    assert push(j, SIDE_LEFT, " I I\n + 3 8\n - 3 8\n - 1 4") == [
+        ("-", (1, 4, None, None))
+    ]
+    assert push(j, SIDE_RIGHT, " I I\n + 2 6\n + 4 8\n + 3 4") == [
+        ("-", (2, 5, None, None)),
+        ("+", (2, 5, 2, 6)),
+        ("+", (None, None, 4, 8)),
+        ("+", (None, None, 3, 4)),
+    ]
+    assert push(j, SIDE_RIGHT, " I I\n + 5 10\n - 5 10\n + 1 2") == [
+        ("+", (None, None, 1, 2))
+    ]
+    j.close()
+
+
+def test_inner_join_nonequi():
+    # hash_join.rs:3609-3664
+    j = classical(JOIN_INNER, cond=True)
+    assert push(j, SIDE_LEFT, " I I\n + 1 4\n + 2 10\n + 3 6") == []
+    assert push(j, SIDE_LEFT, " I I\n + 3 8\n - 3 8") == []
+    assert push(j, SIDE_RIGHT, " I I\n + 2 7\n + 4 8\n + 6 9") == []
+    assert push(j, SIDE_RIGHT, " I I\n + 3 10\n + 6 11") == rows([("+", 3, 6, 3, 10)])
+    j.close()
