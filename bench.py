@@ -1,0 +1,293 @@
+#!/usr/bin/env python3
+"""bench.py — measures the hot path per the driver contract.
+
+Workload (BASELINE.json configs[1], the single-GPU config the metric is
+quoted on): Nexmark q7 windowed hash-agg — append-only HashAgg
+group_key=[window $expr1], aggs=[max(price), count] (reference plan
+`nexmark.yaml` q7 block) — on synthetic bid-shaped chunks: price ~
+uniform[1,1e7) i64, date_time monotone in 10s windows, all-Insert ops,
+seeded. A step = one pass of the apply kernel over one 1M-row batch
+(= 256 ingested 4K-row chunks staged within the epoch)
+already resident in HBM; a checkpoint barrier (flush) fires every
+--barrier-every steps inside the timed region.
+
+Parity gate: before timing, a small q7 run is compared row-for-row
+(multiset per epoch) against the CPU oracle; a mismatch aborts the bench.
+
+Multi-GPU (--gpus N under torchrun): weak scaling, rank-local window key
+ranges (the upstream vnode exchange has already routed keys by window —
+DESIGN.md §7); no data-path collective; barrier + max-over-ranks timing.
+"""
+import argparse
+import ctypes
+import json
+import os
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+sys.path.insert(0, os.path.join(REPO, "tests"))
+
+import numpy as np
+
+
+class KernelStats(ctypes.Structure):
+    _fields_ = [
+        ("launches", ctypes.c_uint64),
+        ("total_ms", ctypes.c_double),
+        ("rows", ctypes.c_uint64),
+    ]
+
+
+CHUNK_ROWS = 4096
+CHUNKS_PER_BATCH = 256  # 1M rows per step
+WINDOW_US = 10_000_000  # 10s tumble (q7)
+# algorithmic bytes per input row for the q7 agg_apply kernel: key 8 + price 8
+# + op 1 + 2 validity bytes = 19 B of mandatory input-stream traffic (the few
+# window slots stay cache-resident; SURVEY §8d's 48 B/row includes an
+# HBM-resident probe, reported separately via rocprof traffic)
+BYTES_PER_ROW = 19
+HBM_PEAK_GBS = 8000.0  # spec peak (MI355X_MICROARCH.md)
+
+
+def make_q7_chunk(ffi, rng, n, window_base, n_windows):
+    # date_time-derived window key: monotone within the chunk
+    w = window_base + np.sort(rng.integers(0, n_windows, n)) * WINDOW_US
+    price = rng.integers(1, 10**7, n)
+    return ffi.Chunk(
+        [ffi.T_I64, ffi.T_I64],
+        np.zeros(n, np.uint8),
+        [w, price],
+        [np.ones(n, np.uint8), np.ones(n, np.uint8)],
+    )
+
+
+def parity_gate(ffi, gpu_lib, rng):
+    from rwtest.ffi import AGG_COUNT_STAR, AGG_MAX, T_I64, oracle, rows_multiset
+
+    calls = [(AGG_MAX, 1, T_I64), (AGG_COUNT_STAR, -1, T_I64)]
+    execs = []
+    for lib in (gpu_lib, oracle()):
+        execs.append(ffi.HashAgg(lib, [T_I64, T_I64], [0], calls, 1, append_only=True))
+    for epoch in range(3):
+        chunks = [make_q7_chunk(ffi, rng, CHUNK_ROWS, 0, 32) for _ in range(4)]
+        outs = []
+        for a in execs:
+            for c in chunks:
+                a.push(c)
+            a.flush(epoch + 1)
+            outs.append(rows_multiset(a.poll_all()))
+        if outs[0] != outs[1]:
+            raise SystemExit("PARITY GATE FAILED: GPU != oracle on q7 sample")
+    for a in execs:
+        a.close()
+
+
+def cpu_baseline(ffi, rng, target_seconds=10.0):
+    """Time the oracle (the CPU restatement, kind 'port') on the same q7
+    workload, bounded sample, single thread."""
+    from rwtest.ffi import AGG_COUNT_STAR, AGG_MAX, T_I64, oracle
+
+    calls = [(AGG_MAX, 1, T_I64), (AGG_COUNT_STAR, -1, T_I64)]
+    agg = ffi.HashAgg(oracle(), [T_I64, T_I64], [0], calls, 1, append_only=True)
+    chunk = make_q7_chunk(ffi, rng, CHUNK_ROWS, 0, 32)
+    # warm + calibrate
+    agg.push(chunk)
+    t0 = time.perf_counter()
+    agg.push(chunk)
+    per_chunk = time.perf_counter() - t0
+    n = max(8, min(int(target_seconds / max(per_chunk, 1e-9)), 200_000))
+    t0 = time.perf_counter()
+    for i in range(n):
+        agg.push(chunk)
+        if (i + 1) % 64 == 0:
+            agg.flush(i)
+            agg.poll_all()
+    dt = time.perf_counter() - t0
+    agg.close()
+    rows = n * CHUNK_ROWS
+    return {
+        "value": rows / dt,
+        "unit": "rows/s",
+        "cores": 1,
+        "kind": "port",
+        "sample": f"{rows} q7 rows ({dt:.1f}s single-thread oracle, flush every 64 chunks)",
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=30)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--barrier-every", type=int, default=16)
+    ap.add_argument("--windows-per-epoch", type=int, default=64)
+    ap.add_argument("--seed", type=int, default=1)
+    ap.add_argument("--skip-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    if world > 1:
+        os.environ.setdefault("HIP_VISIBLE_DEVICES", str(local_rank))
+
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+
+        dist = dist_mod
+        backend = "nccl" if _cuda_available() else "gloo"
+        dist.init_process_group(backend=backend)
+
+    from rwtest import ffi
+
+    import risingwave_amd
+
+    risingwave_amd.load_library()
+    gpu_lib = ffi.Lib(risingwave_amd.lib_path())
+    L = gpu_lib.lib
+    L.rw_agg_bench_preload.restype = ctypes.c_void_p
+    L.rw_agg_bench_preload.argtypes = [ctypes.c_void_p, ctypes.POINTER(ffi.RwChunkC)]
+    L.rw_agg_bench_apply.restype = ctypes.c_int
+    L.rw_agg_bench_apply.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
+    L.rw_agg_sync.restype = ctypes.c_int
+    L.rw_agg_sync.argtypes = [ctypes.c_void_p]
+    L.rw_agg_kernel_stats.argtypes = [ctypes.c_void_p, ctypes.POINTER(KernelStats)]
+    L.rw_agg_stats_reset.argtypes = [ctypes.c_void_p]
+
+    rng = np.random.default_rng(args.seed + rank)
+
+    # ---- parity gate (rank 0 only; cheap) ----
+    if rank == 0:
+        parity_gate(ffi, gpu_lib, np.random.default_rng(99))
+
+    # ---- build the timed executor + preloaded batches ----
+    from rwtest.ffi import AGG_COUNT_STAR, AGG_MAX, T_I64
+
+    calls = [(AGG_MAX, 1, T_I64), (AGG_COUNT_STAR, -1, T_I64)]
+    agg = ffi.HashAgg(gpu_lib, [T_I64, T_I64], [0], calls, 1, append_only=True,
+                      state_capacity_hint=1 << 20)
+
+    batch_rows = CHUNK_ROWS * CHUNKS_PER_BATCH
+    # rank-local window space (weak scaling: upstream exchange already routed
+    # window keys)
+    window_base = rank * 1_000_000 * WINDOW_US
+    n_batches = 4  # distinct resident batches cycled through the steps
+    batches = []
+    for b in range(n_batches):
+        c = make_q7_chunk(ffi, rng, batch_rows,
+                          window_base + b * args.windows_per_epoch * WINDOW_US,
+                          args.windows_per_epoch)
+        cc = c.to_c()
+        h = L.rw_agg_bench_preload(agg.h, ctypes.byref(cc))
+        assert h, gpu_lib.last_error()
+        batches.append(h)
+
+    def step(i):
+        rc = L.rw_agg_bench_apply(agg.h, batches[i % n_batches])
+        assert rc == 0, gpu_lib.last_error()
+        if (i + 1) % args.barrier_every == 0:
+            agg.flush(i)
+            agg.poll_all()
+
+    # ---- warmup ----
+    for i in range(args.warmup):
+        step(i)
+    rc = L.rw_agg_sync(agg.h)
+    assert rc == 0, gpu_lib.last_error()
+    L.rw_agg_stats_reset(agg.h)
+
+    if dist:
+        dist.barrier()
+    _dev_sync()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(i)
+    rc = L.rw_agg_sync(agg.h)
+    assert rc == 0, gpu_lib.last_error()
+    _dev_sync()
+    elapsed = time.perf_counter() - t0
+    if dist:
+        import torch
+
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ks = KernelStats()
+    L.rw_agg_kernel_stats(agg.h, ctypes.byref(ks))
+
+    if rank == 0:
+        total_rows = args.steps * batch_rows * world
+        value = total_rows / elapsed
+        ms_per_step = elapsed * 1000.0 / args.steps
+        # roofline: algorithmic bytes per apply launch ÷ measured launch time
+        # (HIP events on the executor's stream, inside the C library)
+        avg_launch_ms = ks.total_ms / max(ks.launches, 1)
+        achieved_gbs = (BYTES_PER_ROW * batch_rows) / (avg_launch_ms * 1e-3) / 1e9
+        result = {
+            "metric": "input rows/sec/GPU on Nexmark q7 stream",
+            "value": value,
+            "unit": "rows/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # no published number in-repo
+            "dtype": "int64",
+            "data": "synthetic",
+            "config": {
+                "workload": "nexmark_q7",
+                "chunk_rows": CHUNK_ROWS,
+                "chunks_per_step": CHUNKS_PER_BATCH,
+                "windows_per_epoch": args.windows_per_epoch,
+                "barrier_every_steps": args.barrier_every,
+                "agg": "max(price), count group by 10s window (append-only)",
+                "parallelism": f"dp{world}",
+            },
+            "roofline": {
+                "bound": "hbm",
+                "achieved": achieved_gbs,
+                "peak": HBM_PEAK_GBS,
+                "unit": "GB/s",
+                "frac": achieved_gbs / HBM_PEAK_GBS,
+                "traffic": None,  # filled from rocprofv3 PMC runs (profiles/)
+            },
+            "cpu_baseline": None,
+        }
+        if not args.skip_cpu_baseline and world == 1:
+            result["cpu_baseline"] = cpu_baseline(ffi, np.random.default_rng(5))
+        print(json.dumps(result))
+
+    agg.close()
+    if dist:
+        dist.destroy_process_group()
+
+
+def _cuda_available():
+    try:
+        import torch
+
+        return torch.cuda.is_available()
+    except Exception:
+        return False
+
+
+def _dev_sync():
+    # rw_agg_sync drains the executor's stream; also sync the device per the
+    # driver contract when torch sees the GPU
+    try:
+        import torch
+
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+    except Exception:
+        pass
+
+
+if __name__ == "__main__":
+    main()

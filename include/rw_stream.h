@@ -80,6 +80,8 @@ typedef struct RwHashAggDesc {
     const uint32_t* stream_key;
     uint32_t chunk_size;      /* output chunk rows (config/mod.rs:222-224) */
     uint8_t append_only;      /* value-state min/max allowed */
+    uint64_t state_capacity_hint; /* expected group count (0 = default);
+                                     GPU sizes the HBM table from this */
 } RwHashAggDesc;
 
 void* rw_hash_agg_create(const RwHashAggDesc* desc);
@@ -134,6 +136,8 @@ typedef struct RwHashJoinDesc {
     uint32_t cond_l;  /* column indices into the concatenated row */
     uint32_t cond_r;
     uint32_t chunk_size;
+    uint64_t state_capacity_hint; /* expected distinct keys per side (0 = default) */
+    uint64_t row_capacity_hint;   /* expected resident rows per side (0 = default) */
 } RwHashJoinDesc;
 
 enum RwJoinSide { RW_SIDE_LEFT = 0, RW_SIDE_RIGHT = 1 };

@@ -3,7 +3,9 @@ import sys
 
 import pytest
 
-sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+_TESTS = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, _TESTS)
+sys.path.insert(0, os.path.dirname(_TESTS))  # repo root, for risingwave_amd
 
 
 def pytest_configure(config):

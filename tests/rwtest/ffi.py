@@ -65,6 +65,7 @@ class RwHashAggDesc(C.Structure):
         ("stream_key", C.POINTER(C.c_uint32)),
         ("chunk_size", C.c_uint32),
         ("append_only", C.c_uint8),
+        ("state_capacity_hint", C.c_uint64),
     ]
 
 
@@ -91,6 +92,8 @@ class RwHashJoinDesc(C.Structure):
         ("cond_l", C.c_uint32),
         ("cond_r", C.c_uint32),
         ("chunk_size", C.c_uint32),
+        ("state_capacity_hint", C.c_uint64),
+        ("row_capacity_hint", C.c_uint64),
     ]
 
 
@@ -246,7 +249,8 @@ class Lib:
 
 class HashAgg:
     def __init__(self, lib: Lib, input_types, group_key, calls, row_count_index,
-                 stream_key=(), chunk_size=1024, append_only=False):
+                 stream_key=(), chunk_size=1024, append_only=False,
+                 state_capacity_hint=0):
         """calls: list of (kind, arg, ret_type)."""
         self.lib = lib
         d = RwHashAggDesc()
@@ -269,8 +273,10 @@ class HashAgg:
         d.stream_key = self._sk
         d.chunk_size = chunk_size
         d.append_only = 1 if append_only else 0
+        d.state_capacity_hint = state_capacity_hint
         self.h = lib.lib.rw_hash_agg_create(C.byref(d))
-        assert self.h
+        if not self.h:
+            raise RuntimeError(f"rw_hash_agg_create failed: {lib.last_error()}")
 
     def push(self, chunk: Chunk):
         c = chunk.to_c()
@@ -300,7 +306,8 @@ class HashAgg:
 class HashJoin:
     def __init__(self, lib: Lib, join_type, types_l, types_r, key_l, key_r,
                  pk_l, pk_r, output_indices=None, null_safe=None, cond=None,
-                 chunk_size=1024, append_only=False):
+                 chunk_size=1024, append_only=False, state_capacity_hint=0,
+                 row_capacity_hint=0):
         """cond: (op, cond_l, cond_r) into the concatenated row, or None."""
         self.lib = lib
         d = RwHashJoinDesc()
@@ -342,8 +349,11 @@ class HashJoin:
             d.has_cond = 0
             d.cond_op = d.cond_l = d.cond_r = 0
         d.chunk_size = chunk_size
+        d.state_capacity_hint = state_capacity_hint
+        d.row_capacity_hint = row_capacity_hint
         self.h = lib.lib.rw_hash_join_create(C.byref(d))
-        assert self.h
+        if not self.h:
+            raise RuntimeError(f"rw_hash_join_create failed: {lib.last_error()}")
 
     def push(self, side, chunk: Chunk):
         c = chunk.to_c()
