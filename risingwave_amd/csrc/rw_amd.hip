@@ -24,6 +24,7 @@
 #include <cstdio>
 #include <cstring>
 #include <string>
+#include <unordered_set>
 #include <vector>
 
 #include "../../include/rw_stream.h"
@@ -190,70 +191,131 @@ struct AggTableDev {
 // (get_group_visibilities/touch_agg_groups collapse to find-or-insert),
 // then AggregateFunction::update per call (general.rs:18-41,154-162) via
 // atomics — order-free for count/sum/min/max over the epoch's row multiset.
+// Wave-level segmented pre-aggregation: q7 input is date_time-monotone, so
+// consecutive rows (= consecutive lanes) share the window slot. Each
+// contiguous equal-slot run inside a wave is reduced with shuffles and its
+// TAIL lane issues ONE atomic per call — cutting hot-slot atomics by up to
+// 64× (the unsorted case degrades to runs of 1 = plain per-lane atomics,
+// still correct). count/sum combine by +, min/max by min/max, null-presence
+// by OR; all order-free, so the result is exactly the reference's.
 __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, int KW, int n_calls,
                                  AggCallDev c0, AggCallDev c1, AggCallDev c2,
                                  AggCallDev c3) {
     AggCallDev calls[4] = {c0, c1, c2, c3};
     uint32_t stride = gridDim.x * blockDim.x;
-    for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < b.n_rows;
-         r += stride) {
-        if (b.vis && !b.vis[r]) continue;
-        int64_t kw[MAX_KW];
-        uint8_t nullmask = 0;
-        for (int i = 0; i < KW; i++) {
-            bool valid = b.col_valid[i][r];
-            kw[i] = valid ? b.col_vals[i][r] : 0;
-            nullmask |= (!valid) << i;
+    uint32_t iters = (b.n_rows + stride - 1) / stride;
+    int lane = threadIdx.x & 63;
+    size_t cap = (size_t)t.cap_mask + 1;
+    const uint32_t SLOT_NONE = (uint32_t)-1;
+
+    for (uint32_t it = 0; it < iters; it++) {
+        uint32_t r = it * stride + blockIdx.x * blockDim.x + threadIdx.x;
+        bool active = (r < b.n_rows) && !(b.vis && !b.vis[r]);
+        uint32_t slot = SLOT_NONE;
+        long long v[4];
+        uint8_t hasmask = 0;
+        if (active) {
+            int64_t kw[MAX_KW];
+            uint8_t nullmask = 0;
+            for (int i = 0; i < KW; i++) {
+                bool valid = b.col_valid[i][r];
+                kw[i] = valid ? b.col_vals[i][r] : 0;
+                nullmask |= (!valid) << i;
+            }
+            slot = table_find_or_insert(t.state, t.keys, t.key_nulls, t.cap_mask,
+                                        kw, nullmask, KW);
+            if (slot == SLOT_NONE) atomicExch(&t.counters[2], 1u); // table full
         }
-        uint32_t slot = table_find_or_insert(t.state, t.keys, t.key_nulls,
-                                             t.cap_mask, kw, nullmask, KW);
-        if (slot == (uint32_t)-1) {
-            atomicExch(&t.counters[2], 1u); // table full
-            continue;
-        }
-        uint8_t op = b.ops[r];
-        bool retract = (op == RW_OP_DELETE || op == RW_OP_UPDATE_DELETE);
-        long long sign = retract ? -1 : 1;
-        size_t cap = (size_t)t.cap_mask + 1;
+        // per-lane contributions (identity when inactive / NULL arg)
+        uint8_t op = active && slot != SLOT_NONE ? b.ops[r] : RW_OP_INSERT;
+        long long sign =
+            (op == RW_OP_DELETE || op == RW_OP_UPDATE_DELETE) ? -1 : 1;
+        bool contributing = active && slot != SLOT_NONE;
         for (int ci = 0; ci < n_calls; ci++) {
             const AggCallDev& c = calls[ci];
-            long long* acc = t.acc + (size_t)ci * cap;
-            uint8_t* has = t.has + (size_t)ci * cap;
+            int col = KW + ci;
+            bool arg_valid = contributing && b.col_valid[col][r];
             switch (c.kind) {
-                case RW_AGG_COUNT_STAR:
-                    atomic_add_i64(&acc[slot], sign);
-                    break;
-                case RW_AGG_COUNT: {
-                    int col = KW + ci;
-                    if (b.col_valid[col][r]) atomic_add_i64(&acc[slot], sign);
-                    break;
-                }
+                case RW_AGG_COUNT_STAR: v[ci] = contributing ? sign : 0; break;
+                case RW_AGG_COUNT: v[ci] = arg_valid ? sign : 0; break;
                 case RW_AGG_SUM:
-                case RW_AGG_SUM0: {
-                    int col = KW + ci;
-                    if (b.col_valid[col][r]) {
-                        atomic_add_i64(&acc[slot], sign * b.col_vals[col][r]);
-                        if (!has[slot]) has[slot] = 1; // benign race
-                    }
+                case RW_AGG_SUM0:
+                    v[ci] = arg_valid ? sign * b.col_vals[col][r] : 0;
+                    if (arg_valid) hasmask |= 1 << ci;
                     break;
-                }
                 case RW_AGG_MIN:
-                case RW_AGG_MAX: {
-                    int col = KW + ci;
-                    if (b.col_valid[col][r]) {
-                        long long v = b.col_vals[col][r];
-                        if (c.kind == RW_AGG_MIN) atomic_min_i64(&acc[slot], v);
-                        else atomic_max_i64(&acc[slot], v);
-                        if (!has[slot]) has[slot] = 1;
-                    }
+                    v[ci] = arg_valid ? b.col_vals[col][r] : INT64_MAX;
+                    if (arg_valid) hasmask |= 1 << ci;
                     break;
-                }
+                case RW_AGG_MAX:
+                    v[ci] = arg_valid ? b.col_vals[col][r] : INT64_MIN;
+                    if (arg_valid) hasmask |= 1 << ci;
+                    break;
             }
         }
-        // dirty tracking (hash_agg.rs group_change_set)
-        if (atomicCAS(&t.dirty_flag[slot], 0u, 1u) == 0u) {
-            uint32_t i = atomicAdd(&t.counters[0], 1u);
-            t.dirty_list[i] = slot;
+        if (!contributing) slot = SLOT_NONE;
+        // segmented inclusive reduction over contiguous equal-slot runs
+        uint32_t prev_slot = (uint32_t)__shfl_up((int)slot, 1);
+        uint64_t heads =
+            __ballot(lane == 0 || prev_slot != slot || slot == SLOT_NONE);
+        uint64_t le_mask = heads & (~0ULL >> (63 - lane));
+        int run_start = 63 - __clzll(le_mask | 1ULL); // lane 0 is always a head
+        int run_pos = lane - run_start;
+        for (int d = 1; d < 64; d <<= 1) {
+            long long ov[4];
+            uint8_t ohas;
+            for (int ci = 0; ci < n_calls; ci++) ov[ci] = __shfl_up(v[ci], d);
+            ohas = (uint8_t)__shfl_up((int)hasmask, d);
+            if (run_pos >= d) {
+                for (int ci = 0; ci < n_calls; ci++) {
+                    switch (calls[ci].kind) {
+                        case RW_AGG_MIN: v[ci] = v[ci] < ov[ci] ? v[ci] : ov[ci]; break;
+                        case RW_AGG_MAX: v[ci] = v[ci] > ov[ci] ? v[ci] : ov[ci]; break;
+                        default: v[ci] += ov[ci];
+                    }
+                }
+                hasmask |= ohas;
+            }
+        }
+        uint32_t next_slot = (uint32_t)__shfl_down((int)slot, 1);
+        bool tail = contributing && (lane == 63 || next_slot != slot);
+        if (tail) {
+            for (int ci = 0; ci < n_calls; ci++) {
+                long long* acc = t.acc + (size_t)ci * cap;
+                uint8_t* has = t.has + (size_t)ci * cap;
+                switch (calls[ci].kind) {
+                    case RW_AGG_COUNT_STAR:
+                    case RW_AGG_COUNT:
+                        if (v[ci]) atomic_add_i64(&acc[slot], v[ci]);
+                        break;
+                    case RW_AGG_SUM:
+                    case RW_AGG_SUM0:
+                        if (hasmask & (1 << ci)) {
+                            atomic_add_i64(&acc[slot], v[ci]);
+                            if (!has[slot]) has[slot] = 1; // benign race
+                        }
+                        break;
+                    case RW_AGG_MIN:
+                        if (hasmask & (1 << ci)) {
+                            atomic_min_i64(&acc[slot], v[ci]);
+                            if (!has[slot]) has[slot] = 1;
+                        }
+                        break;
+                    case RW_AGG_MAX:
+                        if (hasmask & (1 << ci)) {
+                            atomic_max_i64(&acc[slot], v[ci]);
+                            if (!has[slot]) has[slot] = 1;
+                        }
+                        break;
+                }
+            }
+            // dirty tracking (hash_agg.rs group_change_set), read-first to
+            // keep the hot line in shared state
+            if (t.dirty_flag[slot] == 0 &&
+                atomicCAS(&t.dirty_flag[slot], 0u, 1u) == 0u) {
+                uint32_t i = atomicAdd(&t.counters[0], 1u);
+                t.dirty_list[i] = slot;
+            }
         }
     }
 }
@@ -886,9 +948,9 @@ __device__ __forceinline__ bool join_cond_ok(const JoinMeta& m, int probe_side,
 // probe the match side + update own side, one thread per probe row
 __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
                                   JoinSideDev match, JoinMeta m, int S,
-                                  JoinOutDev out) {
+                                  JoinOutDev out, uint32_t r0, uint32_t r1) {
     uint32_t stride = gridDim.x * blockDim.x;
-    for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < b.n_rows;
+    for (uint32_t r = r0 + blockIdx.x * blockDim.x + threadIdx.x; r < r1;
          r += stride) {
         if (b.vis && !b.vis[r]) continue;
         uint8_t op_in = b.ops[r];
@@ -1149,8 +1211,8 @@ struct HashJoin {
         return RW_OK;
     }
 
-    int probe(int s, const JoinBatchDev& b, bool timed) {
-        uint32_t blocks = (b.n_rows + 255) / 256;
+    int probe(int s, const JoinBatchDev& b, bool timed, uint32_t r0, uint32_t r1) {
+        uint32_t blocks = (r1 - r0 + 255) / 256;
         if (blocks > 2048) blocks = 2048;
         if (!blocks) blocks = 1;
         hipEvent_t e0 = nullptr, e1 = nullptr;
@@ -1160,7 +1222,7 @@ struct HashJoin {
             HIP_TRY(hipEventRecord(e0, stream));
         }
         join_probe_kernel<<<blocks, 256, 0, stream>>>(b, side[s], side[1 - s], m, s,
-                                                      out);
+                                                      out, r0, r1);
         if (timed) {
             HIP_TRY(hipEventRecord(e1, stream));
             HIP_TRY(hipEventSynchronize(e1));
@@ -1223,13 +1285,71 @@ struct HashJoin {
         return RW_OK;
     }
 
+    // The parallel kernel keeps the reference's sequential per-chunk
+    // semantics except when a DELETE targets a row also INSERTed in the same
+    // chunk (hash_join.rs processes rows in order, :993). Detect those rows
+    // host-side and run each such delete in its own single-row segment, in
+    // row order — stream ordering makes earlier segments' state visible.
+    std::vector<uint32_t> conflict_segments(int s, const RwChunk* c) {
+        std::vector<uint32_t> bounds; // segment start offsets (0 implied)
+        bool any_delete = false;
+        for (uint32_t r = 0; r < c->n_rows; r++) {
+            uint8_t op = c->ops[r];
+            if (op == RW_OP_DELETE || op == RW_OP_UPDATE_DELETE) {
+                any_delete = true;
+                break;
+            }
+        }
+        if (!any_delete) return bounds;
+        auto row_key = [&](uint32_t r) {
+            std::string k;
+            k.reserve(m.n_cols[s] * 9);
+            for (int ci = 0; ci < m.n_cols[s]; ci++) {
+                uint8_t valid = c->cols[ci].valid[r];
+                k.push_back((char)valid);
+                int64_t v = valid ? ((const int64_t*)c->cols[ci].data)[r] : 0;
+                k.append((const char*)&v, 8);
+            }
+            return k;
+        };
+        std::unordered_multiset<std::string> inserts;
+        for (uint32_t r = 0; r < c->n_rows; r++) {
+            if (c->vis && !c->vis[r]) continue;
+            uint8_t op = c->ops[r];
+            if (op == RW_OP_INSERT || op == RW_OP_UPDATE_INSERT)
+                inserts.insert(row_key(r));
+        }
+        if (inserts.empty()) return bounds;
+        for (uint32_t r = 0; r < c->n_rows; r++) {
+            if (c->vis && !c->vis[r]) continue;
+            uint8_t op = c->ops[r];
+            if ((op == RW_OP_DELETE || op == RW_OP_UPDATE_DELETE) &&
+                inserts.count(row_key(r))) {
+                bounds.push_back(r);     // segment ends before the delete
+                bounds.push_back(r + 1); // the delete runs alone
+            }
+        }
+        return bounds;
+    }
+
     int push_chunk(int s, const RwChunk* c) {
         if (s != 0 && s != 1) FAIL(RW_E_INVAL, "bad side");
         JoinBatchDev b;
         int rc = upload(s, c, &b);
         if (rc != RW_OK) return rc;
-        rc = probe(s, b, true);
-        if (rc != RW_OK) return rc;
+        std::vector<uint32_t> bounds = conflict_segments(s, c);
+        uint32_t start = 0;
+        for (uint32_t bnd : bounds) {
+            if (bnd > start) {
+                rc = probe(s, b, true, start, bnd);
+                if (rc != RW_OK) return rc;
+            }
+            start = bnd;
+        }
+        if (start < b.n_rows) {
+            rc = probe(s, b, true, start, b.n_rows);
+            if (rc != RW_OK) return rc;
+        }
         HIP_TRY(hipStreamSynchronize(stream));
         return drain_output();
     }
