@@ -125,6 +125,10 @@ def main():
     ap.add_argument("--windows-per-epoch", type=int, default=64)
     ap.add_argument("--seed", type=int, default=1)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
+    ap.add_argument("--workload", choices=["q7", "q8"], default="q7",
+                    help="q7 = windowed hash-agg (BASELINE configs[1], the "
+                         "default the driver measures); q8 = stream-stream "
+                         "hash-join, 10M-key build side (configs[2])")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -163,6 +167,12 @@ def main():
     if rank == 0:
         parity_gate(ffi, gpu_lib, np.random.default_rng(99))
 
+    if args.workload == "q8":
+        bench_q8(args, ffi, gpu_lib, rng, rank, world, dist)
+        if dist:
+            dist.destroy_process_group()
+        return
+
     # ---- build the timed executor + preloaded batches ----
     from rwtest.ffi import AGG_COUNT_STAR, AGG_MAX, T_I64
 
@@ -174,7 +184,10 @@ def main():
     # rank-local window space (weak scaling: upstream exchange already routed
     # window keys)
     window_base = rank * 1_000_000 * WINDOW_US
-    n_batches = 4  # distinct resident batches cycled through the steps
+    # 16 × 19 MB ≈ 300 MB of resident input > the 256 MB Infinity Cache, so
+    # the timed region streams from HBM (L3-masking gotcha,
+    # cdna_hip_programming.md §2)
+    n_batches = 16
     batches = []
     for b in range(n_batches):
         c = make_q7_chunk(ffi, rng, batch_rows,
@@ -266,6 +279,188 @@ def main():
     agg.close()
     if dist:
         dist.destroy_process_group()
+
+
+def bench_q8(args, ffi, gpu_lib, rng, rank, world, dist):
+    """Nexmark q8 stream-stream hash-join (BASELINE configs[2]): build side =
+    10M distinct person ids resident in HBM, probe = auction.seller
+    zipf(1.1) over the same id space; join key = (id, window_start,
+    window_end) = Key256 class (SURVEY §8d). value = probe input rows/s;
+    join outputs stay in HBM for the device-resident downstream operator."""
+    import ctypes
+
+    from rwtest.ffi import JOIN_INNER, SIDE_LEFT, SIDE_RIGHT, T_I64, T_TS
+
+    L = gpu_lib.lib
+    L.rw_join_bench_preload.restype = ctypes.c_void_p
+    L.rw_join_bench_preload.argtypes = [ctypes.c_void_p, ctypes.c_int,
+                                        ctypes.POINTER(ffi.RwChunkC)]
+    L.rw_join_bench_apply.restype = ctypes.c_int
+    L.rw_join_bench_apply.argtypes = [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p]
+    L.rw_join_bench_drain.restype = ctypes.c_longlong
+    L.rw_join_bench_drain.argtypes = [ctypes.c_void_p]
+    L.rw_join_kernel_stats.argtypes = [ctypes.c_void_p, ctypes.POINTER(KernelStats)]
+
+    BUILD_KEYS = 10_000_000
+    batch_rows = CHUNK_ROWS * CHUNKS_PER_BATCH
+    t4 = [T_I64, T_TS, T_TS, T_I64]
+    j = ffi.HashJoin(gpu_lib, JOIN_INNER, t4, t4, key_l=[0, 1, 2],
+                     key_r=[0, 1, 2], pk_l=[3], pk_r=[3],
+                     state_capacity_hint=1 << 24,
+                     row_capacity_hint=BUILD_KEYS + (args.steps + args.warmup + 4)
+                     * batch_rows + 1_000_000)
+
+    WS = 1_000 * WINDOW_US
+    ones = lambda n: np.ones(n, np.uint8)
+
+    def preload(side, ids, rowid0):
+        n = len(ids)
+        c = ffi.Chunk(t4, np.zeros(n, np.uint8),
+                      [ids, np.full(n, WS), np.full(n, WS + WINDOW_US),
+                       np.arange(rowid0, rowid0 + n)],
+                      [ones(n)] * 4)
+        cc = c.to_c()
+        h = L.rw_join_bench_preload(j.h, side, ctypes.byref(cc))
+        assert h, gpu_lib.last_error()
+        return h
+
+    # build side (right): 10M distinct person ids, ingested untimed
+    rowid = 0
+    for lo in range(0, BUILD_KEYS, batch_rows):
+        ids = np.arange(lo, min(lo + batch_rows, BUILD_KEYS), dtype=np.int64)
+        h = preload(SIDE_RIGHT, ids, rowid)
+        rowid += len(ids)
+        rc = L.rw_join_bench_apply(j.h, SIDE_RIGHT, h)
+        assert rc == 0, gpu_lib.last_error()
+        assert L.rw_join_bench_drain(j.h) >= 0, gpu_lib.last_error()
+
+    # probe batches (left): zipf(1.1) sellers over the same id space
+    # (8 × ~36 MB ≈ 290 MB resident > L3, so probes stream from HBM)
+    n_batches = 8
+    batches = []
+    for b in range(n_batches):
+        z = rng.zipf(1.1, batch_rows)
+        ids = ((z - 1) % BUILD_KEYS).astype(np.int64)
+        batches.append(preload(SIDE_LEFT, ids, rowid))
+        rowid += batch_rows
+
+    def step(i):
+        rc = L.rw_join_bench_apply(j.h, SIDE_LEFT, batches[i % n_batches])
+        assert rc == 0, gpu_lib.last_error()
+        n = L.rw_join_bench_drain(j.h)
+        assert n >= 0, gpu_lib.last_error()
+        return n
+
+    for i in range(args.warmup):
+        step(i)
+    j.lib.lib.rw_join_stats_reset(j.h)
+    if dist:
+        dist.barrier()
+    t0 = time.perf_counter()
+    matches = 0
+    for i in range(args.steps):
+        matches += step(i)
+    elapsed = time.perf_counter() - t0
+    if dist:
+        import torch
+
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ks = KernelStats()
+    L.rw_join_kernel_stats(j.h, ctypes.byref(ks))
+    if rank == 0:
+        total_rows = args.steps * batch_rows * world
+        avg_launch_ms = ks.total_ms / max(ks.launches, 1)
+        # SURVEY §8d: ≈128 B algorithmic per probe row at match-rate 1
+        achieved_gbs = (128 * batch_rows) / (avg_launch_ms * 1e-3) / 1e9
+        result = {
+            "metric": "input rows/sec/GPU on Nexmark q8 stream",
+            "value": total_rows / elapsed,
+            "unit": "rows/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed * 1000.0 / args.steps,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "int64",
+            "data": "synthetic",
+            "config": {
+                "workload": "nexmark_q8",
+                "build_keys": BUILD_KEYS,
+                "probe_rows_per_step": batch_rows,
+                "probe_dist": "zipf(1.1)",
+                "join_key": "(id i64, ws ts, we ts)",
+                "match_rate": matches / max(args.steps * batch_rows, 1),
+                "parallelism": f"dp{world}",
+            },
+            "roofline": {
+                "bound": "hbm",
+                "achieved": achieved_gbs,
+                "peak": HBM_PEAK_GBS,
+                "unit": "GB/s",
+                "frac": achieved_gbs / HBM_PEAK_GBS,
+                "traffic": None,
+            },
+            "cpu_baseline": None,
+        }
+        if not args.skip_cpu_baseline and world == 1:
+            result["cpu_baseline"] = cpu_baseline_q8(ffi, np.random.default_rng(5))
+        print(json.dumps(result))
+    j.close()
+
+
+def cpu_baseline_q8(ffi, rng, target_seconds=8.0):
+    """Oracle join on a scaled-down q8 sample: build 200k keys, probe 4K-row
+    chunks with the same zipf shape; single thread, kind 'port'."""
+    from rwtest.ffi import JOIN_INNER, SIDE_LEFT, SIDE_RIGHT, T_I64, T_TS, oracle
+
+    BUILD = 200_000
+    t4 = [T_I64, T_TS, T_TS, T_I64]
+    j = ffi.HashJoin(oracle(), JOIN_INNER, t4, t4, key_l=[0, 1, 2],
+                     key_r=[0, 1, 2], pk_l=[3], pk_r=[3])
+    WS = 1_000 * WINDOW_US
+    ones = np.ones(CHUNK_ROWS, np.uint8)
+
+    def chunk(ids, rowid0):
+        n = len(ids)
+        return ffi.Chunk(t4, np.zeros(n, np.uint8),
+                         [ids, np.full(n, WS), np.full(n, WS + WINDOW_US),
+                          np.arange(rowid0, rowid0 + n)],
+                         [np.ones(n, np.uint8)] * 4)
+
+    rowid = 0
+    for lo in range(0, BUILD, CHUNK_ROWS):
+        ids = np.arange(lo, min(lo + CHUNK_ROWS, BUILD), dtype=np.int64)
+        j.push(SIDE_RIGHT, chunk(ids, rowid))
+        j.poll_all()
+        rowid += len(ids)
+
+    z = rng.zipf(1.1, CHUNK_ROWS)
+    probe = chunk(((z - 1) % BUILD).astype(np.int64), rowid)
+    t0 = time.perf_counter()
+    j.push(SIDE_LEFT, probe)
+    j.poll_all()
+    per_chunk = time.perf_counter() - t0
+    n = max(8, min(int(target_seconds / max(per_chunk, 1e-9)), 100_000))
+    t0 = time.perf_counter()
+    for i in range(n):
+        j.push(SIDE_LEFT, probe)
+        j.poll_all()
+    dt = time.perf_counter() - t0
+    j.close()
+    rows = n * CHUNK_ROWS
+    return {
+        "value": rows / dt,
+        "unit": "rows/s",
+        "cores": 1,
+        "kind": "port",
+        "sample": f"{rows} q8 probe rows over a 200k-key build side "
+                  f"({dt:.1f}s single-thread oracle)",
+    }
 
 
 def _cuda_available():

@@ -67,7 +67,7 @@ __device__ __forceinline__ uint64_t mix64(uint64_t x) {
     return x ^ (x >> 31);
 }
 
-__device__ __forceinline__ uint64_t hash_key(const int64_t* kw, uint8_t nullmask,
+__device__ __forceinline__ uint64_t hash_key(const int64_t* kw, uint32_t nullmask,
                                              int KW) {
     uint64_t h = 0x20210401u ^ (uint64_t)nullmask * 0x9e3779b97f4a7c15ULL;
     for (int i = 0; i < KW; i++) h = mix64(h ^ (uint64_t)kw[i]);
@@ -90,32 +90,54 @@ __device__ __forceinline__ void atomic_max_i64(long long* p, long long v) {
 #define SLOT_CLAIMED 1u
 #define SLOT_READY 2u
 
+// All cross-lane-shared table words use relaxed AGENT-scope atomics (sc1:
+// L1-bypassing, write-through to the coherent point) — the guide's R1/R2
+// hand-off forms. The claim path is R1: sc1 key stores -> s_waitcnt vmcnt(0)
+// drain -> sc1 READY store; readers use sc1 loads throughout, which per the
+// microarch visibility table replace the acquire when the producer stored
+// sc1. No acquire/release fences anywhere on the hot path.
+#define RLX __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT
+
+__device__ __forceinline__ uint32_t ld_u32(const uint32_t* p) {
+    return __hip_atomic_load((uint32_t*)p, RLX);
+}
+__device__ __forceinline__ void st_u32(uint32_t* p, uint32_t v) {
+    __hip_atomic_store(p, v, RLX);
+}
+__device__ __forceinline__ int64_t ld_i64(const int64_t* p) {
+    return (int64_t)__hip_atomic_load((unsigned long long*)p, RLX);
+}
+__device__ __forceinline__ void st_i64(int64_t* p, int64_t v) {
+    __hip_atomic_store((unsigned long long*)p, (unsigned long long)v, RLX);
+}
+
 // Find-or-insert into an open-addressed key table (linear probe).
-// Returns slot index, or (uint32_t)-1 on table-full.
+// key_nulls is a u32-per-slot null mask. Returns slot, or -1 on table-full.
 __device__ __forceinline__ uint32_t table_find_or_insert(
-    uint32_t* state, int64_t* keys, uint8_t* key_nulls, uint32_t cap_mask,
-    const int64_t* kw, uint8_t nullmask, int KW) {
+    uint32_t* state, int64_t* keys, uint32_t* key_nulls, uint32_t cap_mask,
+    const int64_t* kw, uint32_t nullmask, int KW) {
     uint32_t slot = (uint32_t)(hash_key(kw, nullmask, KW) & cap_mask);
     for (uint32_t probes = 0; probes <= cap_mask; probes++) {
-        uint32_t st = __hip_atomic_load(&state[slot], __ATOMIC_ACQUIRE,
-                                        __HIP_MEMORY_SCOPE_AGENT);
+        uint32_t st = ld_u32(&state[slot]);
         if (st == SLOT_EMPTY) {
             uint32_t prev = atomicCAS(&state[slot], SLOT_EMPTY, SLOT_CLAIMED);
             if (prev == SLOT_EMPTY) {
-                for (int i = 0; i < KW; i++) keys[(size_t)slot * KW + i] = kw[i];
-                key_nulls[slot] = nullmask;
-                __hip_atomic_store(&state[slot], SLOT_READY, __ATOMIC_RELEASE,
-                                   __HIP_MEMORY_SCOPE_AGENT);
+                for (int i = 0; i < KW; i++) st_i64(&keys[(size_t)slot * KW + i], kw[i]);
+                st_u32(&key_nulls[slot], nullmask);
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); // R1 drain
+                st_u32(&state[slot], SLOT_READY);
                 return slot;
             }
             st = prev;
         }
-        while (st == SLOT_CLAIMED)
-            st = __hip_atomic_load(&state[slot], __ATOMIC_ACQUIRE,
-                                   __HIP_MEMORY_SCOPE_AGENT);
+        while (st == SLOT_CLAIMED) {
+            __builtin_amdgcn_s_sleep(1);
+            st = ld_u32(&state[slot]);
+        }
         // st == SLOT_READY
-        bool eq = key_nulls[slot] == nullmask;
-        for (int i = 0; eq && i < KW; i++) eq = keys[(size_t)slot * KW + i] == kw[i];
+        bool eq = ld_u32(&key_nulls[slot]) == nullmask;
+        for (int i = 0; eq && i < KW; i++)
+            eq = ld_i64(&keys[(size_t)slot * KW + i]) == kw[i];
         if (eq) return slot;
         slot = (slot + 1) & cap_mask;
     }
@@ -125,18 +147,17 @@ __device__ __forceinline__ uint32_t table_find_or_insert(
 // Find-only (no insert). Returns slot or -1.
 __device__ __forceinline__ uint32_t table_find(const uint32_t* state,
                                                const int64_t* keys,
-                                               const uint8_t* key_nulls,
+                                               const uint32_t* key_nulls,
                                                uint32_t cap_mask, const int64_t* kw,
-                                               uint8_t nullmask, int KW) {
+                                               uint32_t nullmask, int KW) {
     uint32_t slot = (uint32_t)(hash_key(kw, nullmask, KW) & cap_mask);
     for (uint32_t probes = 0; probes <= cap_mask; probes++) {
-        uint32_t st = __hip_atomic_load(&state[slot], __ATOMIC_ACQUIRE,
-                                        __HIP_MEMORY_SCOPE_AGENT);
+        uint32_t st = ld_u32(&state[slot]);
         if (st == SLOT_EMPTY) return (uint32_t)-1;
         if (st == SLOT_READY) {
-            bool eq = key_nulls[slot] == nullmask;
+            bool eq = ld_u32(&key_nulls[slot]) == nullmask;
             for (int i = 0; eq && i < KW; i++)
-                eq = keys[(size_t)slot * KW + i] == kw[i];
+                eq = ld_i64(&keys[(size_t)slot * KW + i]) == kw[i];
             if (eq) return slot;
         }
         slot = (slot + 1) & cap_mask;
@@ -169,7 +190,7 @@ struct AggBatch {
 struct AggTableDev {
     uint32_t* state;
     int64_t* keys;      // [cap * KW]
-    uint8_t* key_nulls; // [cap]
+    uint32_t* key_nulls; // [cap]
     long long* acc;     // [n_calls][cap]
     uint8_t* has;       // [n_calls][cap] — any non-null input applied
     long long* prev;    // [n_calls][cap]
@@ -211,26 +232,48 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, int KW, int n_calls,
     for (uint32_t it = 0; it < iters; it++) {
         uint32_t r = it * stride + blockIdx.x * blockDim.x + threadIdx.x;
         bool active = (r < b.n_rows) && !(b.vis && !b.vis[r]);
-        uint32_t slot = SLOT_NONE;
-        long long v[4];
-        uint8_t hasmask = 0;
+        int64_t kw[MAX_KW];
+        uint32_t nullmask = 0;
+        for (int i = 0; i < KW; i++) kw[i] = 0;
         if (active) {
-            int64_t kw[MAX_KW];
-            uint8_t nullmask = 0;
             for (int i = 0; i < KW; i++) {
                 bool valid = b.col_valid[i][r];
                 kw[i] = valid ? b.col_vals[i][r] : 0;
                 nullmask |= (!valid) << i;
             }
+        }
+        // run-head detection by neighbor key compare: only one lane per
+        // contiguous equal-key run probes the table (q7 input is
+        // window-sorted, so runs are long; unsorted degrades to per-lane)
+        uint64_t act_b = __ballot(active);
+        bool same = true;
+        for (int i = 0; i < KW; i++) {
+            int64_t pk = __shfl_up(kw[i], 1);
+            same = same && (pk == kw[i]);
+        }
+        uint32_t pn = (uint32_t)__shfl_up((int)nullmask, 1);
+        same = same && (pn == nullmask);
+        bool prev_active = lane > 0 && ((act_b >> (lane - 1)) & 1);
+        bool head = active && !(same && prev_active);
+        uint32_t slot = SLOT_NONE;
+        if (head) {
             slot = table_find_or_insert(t.state, t.keys, t.key_nulls, t.cap_mask,
                                         kw, nullmask, KW);
             if (slot == SLOT_NONE) atomicExch(&t.counters[2], 1u); // table full
         }
+        uint64_t heads_b = __ballot(head);
+        uint64_t le_mask = heads_b & (~0ULL >> (63 - lane));
+        int run_start = 63 - __clzll(le_mask | 1ULL);
+        if (active && !head) slot = (uint32_t)__shfl((int)slot, run_start);
+        int run_pos = lane - run_start;
+
         // per-lane contributions (identity when inactive / NULL arg)
         uint8_t op = active && slot != SLOT_NONE ? b.ops[r] : RW_OP_INSERT;
         long long sign =
             (op == RW_OP_DELETE || op == RW_OP_UPDATE_DELETE) ? -1 : 1;
         bool contributing = active && slot != SLOT_NONE;
+        long long v[4];
+        uint8_t hasmask = 0;
         for (int ci = 0; ci < n_calls; ci++) {
             const AggCallDev& c = calls[ci];
             int col = KW + ci;
@@ -253,14 +296,8 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, int KW, int n_calls,
                     break;
             }
         }
-        if (!contributing) slot = SLOT_NONE;
-        // segmented inclusive reduction over contiguous equal-slot runs
-        uint32_t prev_slot = (uint32_t)__shfl_up((int)slot, 1);
-        uint64_t heads =
-            __ballot(lane == 0 || prev_slot != slot || slot == SLOT_NONE);
-        uint64_t le_mask = heads & (~0ULL >> (63 - lane));
-        int run_start = 63 - __clzll(le_mask | 1ULL); // lane 0 is always a head
-        int run_pos = lane - run_start;
+        // segmented inclusive reduction over the runs; the run TAIL issues
+        // one atomic per call (order-free combines: +, min, max, OR)
         for (int d = 1; d < 64; d <<= 1) {
             long long ov[4];
             uint8_t ohas;
@@ -277,8 +314,10 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, int KW, int n_calls,
                 hasmask |= ohas;
             }
         }
-        uint32_t next_slot = (uint32_t)__shfl_down((int)slot, 1);
-        bool tail = contributing && (lane == 63 || next_slot != slot);
+        // a run ends before a head, before a non-contributing lane, or at 63
+        uint64_t cont_b = __ballot(contributing);
+        uint64_t ends_after = (heads_b | ~cont_b) >> 1 | (1ULL << 63);
+        bool tail = contributing && ((ends_after >> lane) & 1);
         if (tail) {
             for (int ci = 0; ci < n_calls; ci++) {
                 long long* acc = t.acc + (size_t)ci * cap;
@@ -310,8 +349,8 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, int KW, int n_calls,
                 }
             }
             // dirty tracking (hash_agg.rs group_change_set), read-first to
-            // keep the hot line in shared state
-            if (t.dirty_flag[slot] == 0 &&
+            // keep the hot line shared
+            if (ld_u32(&t.dirty_flag[slot]) == 0 &&
                 atomicCAS(&t.dirty_flag[slot], 0u, 1u) == 0u) {
                 uint32_t i = atomicAdd(&t.counters[0], 1u);
                 t.dirty_list[i] = slot;
@@ -518,7 +557,7 @@ struct HashAgg {
         HIP_TRY(hipMalloc(&t.state, cap * 4));
         HIP_TRY(hipMemset(t.state, 0, cap * 4));
         HIP_TRY(hipMalloc(&t.keys, cap * KW * 8));
-        HIP_TRY(hipMalloc(&t.key_nulls, cap));
+        HIP_TRY(hipMalloc(&t.key_nulls, cap * 4));
         HIP_TRY(hipMalloc(&t.acc, cap * n_calls * 8));
         HIP_TRY(hipMalloc(&t.has, cap * n_calls));
         HIP_TRY(hipMemset(t.has, 0, cap * n_calls));
@@ -875,7 +914,7 @@ struct JoinSideDev {
     // key table
     uint32_t* state;
     int64_t* keys;      // [cap * KW]
-    uint8_t* key_nulls; // [cap]
+    uint32_t* key_nulls; // [cap]
     uint32_t* head;     // [cap] chain head row index, UINT32_MAX = none
     uint32_t cap_mask;
     // row store (SoA)
@@ -945,47 +984,88 @@ __device__ __forceinline__ bool join_cond_ok(const JoinMeta& m, int probe_side,
     return false;
 }
 
-// probe the match side + update own side, one thread per probe row
+// probe the match side + update own side, one thread per probe row.
+// Emission reserves output rows with ONE wave-level atomicAdd per iteration
+// (per-lane match counts -> wave prefix sum -> lane-0 cursor add), then a
+// second chain walk writes the rows: the match side's state is immutable
+// during a non-append-only launch (probes read the OTHER side; own-side
+// writes never alias it), so the two walks agree. The append-only path
+// mutates the match side (kill-on-match) and keeps per-match atomics.
 __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
                                   JoinSideDev match, JoinMeta m, int S,
                                   JoinOutDev out, uint32_t r0, uint32_t r1) {
     uint32_t stride = gridDim.x * blockDim.x;
-    for (uint32_t r = r0 + blockIdx.x * blockDim.x + threadIdx.x; r < r1;
-         r += stride) {
-        if (b.vis && !b.vis[r]) continue;
-        uint8_t op_in = b.ops[r];
+    uint32_t n = r1 - r0;
+    uint32_t iters = (n + stride - 1) / stride;
+    int lane = threadIdx.x & 63;
+
+    for (uint32_t it = 0; it < iters; it++) {
+        uint32_t r = r0 + it * stride + blockIdx.x * blockDim.x + threadIdx.x;
+        bool active = (r < r1) && !(b.vis && !b.vis[r]);
+        uint8_t op_in = active ? b.ops[r] : RW_OP_INSERT;
         bool is_insert = (op_in == RW_OP_INSERT || op_in == RW_OP_UPDATE_INSERT);
         uint8_t op = is_insert ? RW_OP_INSERT : RW_OP_DELETE;
         int64_t kw[MAX_KW];
-        uint8_t nullmask = 0;
-        for (int i = 0; i < m.KW; i++) {
-            uint8_t col = m.key_cols[S][i];
-            bool valid = b.col_valid[col][r];
-            kw[i] = valid ? b.col_vals[col][r] : 0;
-            nullmask |= (!valid) << i;
+        uint32_t nullmask = 0;
+        if (active) {
+            for (int i = 0; i < m.KW; i++) {
+                uint8_t col = m.key_cols[S][i];
+                bool valid = b.col_valid[col][r];
+                kw[i] = valid ? b.col_vals[col][r] : 0;
+                nullmask |= (!valid) << i;
+            }
+            // null-safe NeverMatch (hash_join.rs:1004-1016): inner join
+            // forwards nothing and writes no state
+            if (nullmask & ~(uint32_t)m.null_safe_mask) active = false;
         }
-        // null-safe NeverMatch (hash_join.rs:1004-1016): inner join forwards
-        // nothing and writes no state
-        if (nullmask & ~m.null_safe_mask) continue;
 
-        uint32_t mslot = table_find(match.state, match.keys, match.key_nulls,
-                                    match.cap_mask, kw, nullmask, m.KW);
+        uint32_t mslot = UINT32_MAX;
+        uint32_t my_n = 0;
         uint32_t matched_row = UINT32_MAX;
-        if (mslot != UINT32_MAX) {
-            uint32_t row = __hip_atomic_load(&match.head[mslot], __ATOMIC_ACQUIRE,
-                                             __HIP_MEMORY_SCOPE_AGENT);
-            while (row != UINT32_MAX) {
-                if (match.alive[row] && join_cond_ok(m, S, b, r, match, row)) {
-                    // emit concat row (JoinStreamChunkBuilder::append_row,
-                    // join/builder.rs:87-106)
-                    uint32_t orow = atomicAdd(&out.counters[0], 1u);
-                    if (orow >= out.cap) {
-                        atomicExch(&out.counters[1], 1u);
-                    } else {
+        if (active) {
+            mslot = table_find(match.state, match.keys, match.key_nulls,
+                               match.cap_mask, kw, nullmask, m.KW);
+            if (mslot != UINT32_MAX) {
+                uint32_t row = ld_u32(&match.head[mslot]);
+                while (row != UINT32_MAX) {
+                    if (ld_u32(&match.alive[row]) &&
+                        join_cond_ok(m, S, b, r, match, row)) {
+                        my_n++;
+                        matched_row = row;
+                    }
+                    row = match.next[row];
+                }
+            }
+        }
+
+        if (!m.append_only) {
+            // wave-aggregated output reservation
+            uint32_t incl = my_n;
+            for (int d = 1; d < 64; d <<= 1) {
+                uint32_t o = __shfl_up(incl, d);
+                if (lane >= d) incl += o;
+            }
+            uint32_t total = (uint32_t)__shfl((int)incl, 63);
+            uint32_t base = 0;
+            if (lane == 0 && total)
+                base = atomicAdd(&out.counters[0], total);
+            base = (uint32_t)__shfl((int)base, 0);
+            uint32_t my_base = base + incl - my_n;
+            if (total && base + total > out.cap) {
+                if (lane == 0) atomicExch(&out.counters[1], 1u); // overflow
+            } else if (my_n) {
+                // second walk: emit (JoinStreamChunkBuilder::append_row,
+                // join/builder.rs:87-106)
+                uint32_t row = ld_u32(&match.head[mslot]);
+                uint32_t k = 0;
+                while (row != UINT32_MAX && k < my_n) {
+                    if (ld_u32(&match.alive[row]) &&
+                        join_cond_ok(m, S, b, r, match, row)) {
+                        uint32_t orow = my_base + k;
                         out.ops[orow] = op;
-                        for (int k = 0; k < m.n_out; k++) {
-                            bool from_probe = (int)m.out_src[k] == S;
-                            uint8_t col = m.out_col[k];
+                        for (int c = 0; c < m.n_out; c++) {
+                            bool from_probe = (int)m.out_src[c] == S;
+                            uint8_t col = m.out_col[c];
                             int64_t v;
                             uint8_t valid;
                             if (from_probe) {
@@ -995,8 +1075,40 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
                                 valid = match.col_valid[col][row];
                                 v = match.col_vals[col][row];
                             }
-                            out.vals[(size_t)orow * m.n_out + k] = valid ? v : 0;
-                            out.nulls[(size_t)orow * m.n_out + k] = !valid;
+                            out.vals[(size_t)orow * m.n_out + c] = valid ? v : 0;
+                            out.nulls[(size_t)orow * m.n_out + c] = !valid;
+                        }
+                        k++;
+                    }
+                    row = match.next[row];
+                }
+            }
+        } else if (active && mslot != UINT32_MAX) {
+            // append-only path: <=1 match (jk superset of pk); per-match
+            // cursor atomics, single walk with kill
+            uint32_t row = ld_u32(&match.head[mslot]);
+            while (row != UINT32_MAX) {
+                if (ld_u32(&match.alive[row]) &&
+                    join_cond_ok(m, S, b, r, match, row)) {
+                    uint32_t orow = atomicAdd(&out.counters[0], 1u);
+                    if (orow >= out.cap) {
+                        atomicExch(&out.counters[1], 1u);
+                    } else {
+                        out.ops[orow] = op;
+                        for (int c = 0; c < m.n_out; c++) {
+                            bool from_probe = (int)m.out_src[c] == S;
+                            uint8_t col = m.out_col[c];
+                            int64_t v;
+                            uint8_t valid;
+                            if (from_probe) {
+                                valid = b.col_valid[col][r];
+                                v = b.col_vals[col][r];
+                            } else {
+                                valid = match.col_valid[col][row];
+                                v = match.col_vals[col][row];
+                            }
+                            out.vals[(size_t)orow * m.n_out + c] = valid ? v : 0;
+                            out.nulls[(size_t)orow * m.n_out + c] = !valid;
                         }
                     }
                     matched_row = row;
@@ -1005,10 +1117,12 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
             }
         }
 
+        if (!active) continue;
+
         if (m.append_only && is_insert && matched_row != UINT32_MAX) {
-            // append-only optimize: jk ⊇ pk ⇒ single match; delete it and
-            // skip own insert (hash_join.rs:1241-1245)
-            match.alive[matched_row] = 0;
+            // append-only optimize: delete the single matched row and skip
+            // own insert (hash_join.rs:1241-1245)
+            st_u32(&match.alive[matched_row], 0);
             continue;
         }
 
@@ -1027,40 +1141,41 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
                 continue;
             }
             for (int c = 0; c < m.n_cols[S]; c++) {
-                own.col_vals[c][row] = b.col_vals[c][r];
+                st_i64(&own.col_vals[c][row], b.col_vals[c][r]);
                 own.col_valid[c][row] = b.col_valid[c][r];
             }
-            own.alive[row] = 1;
-            // publish the row before linking it (agent-scope release so a
-            // later launch's chain walk sees complete rows; within one launch
-            // probes never read own-side rows)
-            __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
-            own.next[row] = atomicExch(&own.head[own_slot], row);
+            st_u32(&own.alive[row], 1);
+            // lock-free chain push: set next BEFORE publishing the row (a
+            // same-launch delete may walk this chain), R1-draining the row
+            // payload ahead of the CAS publish
+            uint32_t old_head = ld_u32(&own.head[own_slot]);
+            for (;;) {
+                st_u32(&own.next[row], old_head);
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); // R1 drain
+                uint32_t prev = atomicCAS(&own.head[own_slot], old_head, row);
+                if (prev == old_head) break;
+                old_head = prev;
+            }
         } else {
             // delete own row (join/hash_join.rs:659-681 deletes by deduped
-            // pk; we compare the FULL row so that a same-chunk U−/U+ pair —
-            // same stream key, different values — deletes exactly the old
-            // row even though the kernel is parallel; for pk-unique state
-            // the two comparisons are equivalent)
+            // pk; FULL-row compare keeps same-chunk U-pairs race-free, and
+            // the CAS claim makes two identical deletes kill two distinct
+            // identical rows)
             uint32_t own_slot = table_find(own.state, own.keys, own.key_nulls,
                                            own.cap_mask, kw, nullmask, m.KW);
             if (own_slot == UINT32_MAX) continue;
-            uint32_t row = own.head[own_slot];
+            uint32_t row = ld_u32(&own.head[own_slot]);
             while (row != UINT32_MAX) {
-                if (own.alive[row]) {
+                if (ld_u32(&own.alive[row])) {
                     bool eq = true;
                     for (int c = 0; eq && c < m.n_cols[S]; c++) {
                         uint8_t va = b.col_valid[c][r], vb = own.col_valid[c][row];
                         eq = (va == vb) &&
-                             (!va || b.col_vals[c][r] == own.col_vals[c][row]);
+                             (!va || b.col_vals[c][r] == ld_i64(&own.col_vals[c][row]));
                     }
-                    // CAS-claim so two identical deletes in one chunk kill
-                    // two distinct identical rows, never the same one twice
-                    if (eq && atomicCAS(&own.alive[row], 1u, 0u) == 1u) {
-                        break;
-                    }
+                    if (eq && atomicCAS(&own.alive[row], 1u, 0u) == 1u) break;
                 }
-                row = own.next[row];
+                row = ld_u32(&own.next[row]);
             }
         }
     }
@@ -1148,7 +1263,7 @@ struct HashJoin {
             HIP_TRY(hipMalloc(&js.state, (size_t)cap * 4));
             HIP_TRY(hipMemset(js.state, 0, (size_t)cap * 4));
             HIP_TRY(hipMalloc(&js.keys, (size_t)cap * m.KW * 8));
-            HIP_TRY(hipMalloc(&js.key_nulls, cap));
+            HIP_TRY(hipMalloc(&js.key_nulls, (size_t)cap * 4));
             HIP_TRY(hipMalloc(&js.head, (size_t)cap * 4));
             HIP_TRY(hipMemset(js.head, 0xFF, (size_t)cap * 4));
             js.row_cap = (uint32_t)row_cap;
@@ -1416,6 +1531,51 @@ int rw_hash_join_push_chunk(void* h, int side, const RwChunk* c) {
 int rw_hash_join_flush(void* h, uint64_t epoch) { return ((HashJoin*)h)->flush(epoch); }
 RwChunk* rw_hash_join_poll(void* h) { return ((HashJoin*)h)->poll(); }
 void rw_hash_join_destroy(void* h) { delete (HashJoin*)h; }
+
+// --- bench support: device-resident probe batches; outputs stay in HBM
+// (the downstream operator is device-resident — DESIGN.md §5; the
+// PCIe-inclusive ingest rate is reported separately) ---
+
+void* rw_join_bench_preload(void* h, int side, const RwChunk* c) {
+    auto* j = (HashJoin*)h;
+    auto* b = new JoinBatchDev{};
+    uint32_t n = c->n_rows;
+    for (int ci = 0; ci < j->m.n_cols[side]; ci++) {
+        if (hipMalloc(&b->col_vals[ci], (size_t)n * 8) != hipSuccess) return nullptr;
+        if (hipMalloc(&b->col_valid[ci], n) != hipSuccess) return nullptr;
+        hipMemcpy(b->col_vals[ci], c->cols[ci].data, (size_t)n * 8,
+                  hipMemcpyHostToDevice);
+        hipMemcpy(b->col_valid[ci], c->cols[ci].valid, n, hipMemcpyHostToDevice);
+    }
+    if (hipMalloc(&b->ops, n) != hipSuccess) return nullptr;
+    hipMemcpy(b->ops, c->ops, n, hipMemcpyHostToDevice);
+    b->vis = nullptr;
+    b->n_rows = n;
+    return b;
+}
+
+// launch the probe for a preloaded batch; outputs accumulate in the device
+// buffer until rw_join_bench_drain
+int rw_join_bench_apply(void* h, int side, void* batch) {
+    auto* j = (HashJoin*)h;
+    return j->probe(side, *(JoinBatchDev*)batch, true, 0,
+                    ((JoinBatchDev*)batch)->n_rows);
+}
+
+// device-side drain: return match count, check errors, reset the cursor
+long long rw_join_bench_drain(void* h) {
+    auto* j = (HashJoin*)h;
+    if (hipStreamSynchronize(j->stream) != hipSuccess) return -1;
+    uint32_t ctr[2];
+    if (hipMemcpy(ctr, j->out.counters, 8, hipMemcpyDeviceToHost) != hipSuccess)
+        return -1;
+    if (ctr[1] != 0) {
+        g_err = "join bench overflow/full (code " + std::to_string(ctr[1]) + ")";
+        return -(long long)ctr[1] - 1;
+    }
+    hipMemset(j->out.counters, 0, 8);
+    return (long long)ctr[0];
+}
 
 int rw_join_kernel_stats(void* h, RwKernelStats* out) {
     auto* j = (HashJoin*)h;
