@@ -22,6 +22,7 @@
 
 #include <cstdint>
 #include <cstdio>
+#include <cstdlib>
 #include <cstring>
 #include <string>
 #include <unordered_set>
@@ -221,7 +222,7 @@ struct AggTableDev {
 // by OR; all order-free, so the result is exactly the reference's.
 __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, int KW, int n_calls,
                                  AggCallDev c0, AggCallDev c1, AggCallDev c2,
-                                 AggCallDev c3) {
+                                 AggCallDev c3, int mode) {
     AggCallDev calls[4] = {c0, c1, c2, c3};
     uint32_t stride = gridDim.x * blockDim.x;
     uint32_t iters = (b.n_rows + stride - 1) / stride;
@@ -255,6 +256,7 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, int KW, int n_calls,
         same = same && (pn == nullmask);
         bool prev_active = lane > 0 && ((act_b >> (lane - 1)) & 1);
         bool head = active && !(same && prev_active);
+        if (mode == 2) head = active; // debug: no dedupe, per-lane probe
         uint32_t slot = SLOT_NONE;
         if (head) {
             slot = table_find_or_insert(t.state, t.keys, t.key_nulls, t.cap_mask,
@@ -295,6 +297,46 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, int KW, int n_calls,
                     if (arg_valid) hasmask |= 1 << ci;
                     break;
             }
+        }
+        if (mode == 1) {
+            // debug fallback: per-lane atomics, no wave aggregation
+            if (contributing) {
+                for (int ci = 0; ci < n_calls; ci++) {
+                    long long* acc = t.acc + (size_t)ci * cap;
+                    uint8_t* has = t.has + (size_t)ci * cap;
+                    switch (calls[ci].kind) {
+                        case RW_AGG_COUNT_STAR:
+                        case RW_AGG_COUNT:
+                            if (v[ci]) atomic_add_i64(&acc[slot], v[ci]);
+                            break;
+                        case RW_AGG_SUM:
+                        case RW_AGG_SUM0:
+                            if (hasmask & (1 << ci)) {
+                                atomic_add_i64(&acc[slot], v[ci]);
+                                if (!has[slot]) has[slot] = 1;
+                            }
+                            break;
+                        case RW_AGG_MIN:
+                            if (hasmask & (1 << ci)) {
+                                atomic_min_i64(&acc[slot], v[ci]);
+                                if (!has[slot]) has[slot] = 1;
+                            }
+                            break;
+                        case RW_AGG_MAX:
+                            if (hasmask & (1 << ci)) {
+                                atomic_max_i64(&acc[slot], v[ci]);
+                                if (!has[slot]) has[slot] = 1;
+                            }
+                            break;
+                    }
+                }
+                if (ld_u32(&t.dirty_flag[slot]) == 0 &&
+                    atomicCAS(&t.dirty_flag[slot], 0u, 1u) == 0u) {
+                    uint32_t i = atomicAdd(&t.counters[0], 1u);
+                    t.dirty_list[i] = slot;
+                }
+            }
+            continue;
         }
         // segmented inclusive reduction over the runs; the run TAIL issues
         // one atomic per call (order-free combines: +, min, max, OR)
@@ -506,6 +548,7 @@ struct HashAgg {
     uint64_t apply_launches = 0, apply_rows = 0;
     // pending outputs
     std::vector<RwChunk*> outq;
+    int debug_mode = 0; // RW_AGG_DEBUG_MODE: 1 per-lane atomics, 2 no-dedupe
 
     int grid_for(uint32_t work) const {
         uint32_t blocks = (work + 255) / 256;
@@ -520,6 +563,7 @@ struct HashAgg {
 
     int init(const RwHashAggDesc* d) {
         if (!gpu_ok()) FAIL(RW_E_NOGPU, "risingwave_amd: no GPU visible (product path has no CPU fallback)");
+        if (const char* m = getenv("RW_AGG_DEBUG_MODE")) debug_mode = atoi(m);
         desc = *d;
         input_types.assign(d->input_types, d->input_types + d->n_input_cols);
         group_key.assign(d->group_key_indices, d->group_key_indices + d->n_group_key);
@@ -658,7 +702,7 @@ struct HashAgg {
             HIP_TRY(hipEventRecord(e0, stream));
         }
         agg_apply_kernel<<<grid_for(b.n_rows), 256, 0, stream>>>(
-            b, t, KW, n_calls, cd(0), cd(1), cd(2), cd(3));
+            b, t, KW, n_calls, cd(0), cd(1), cd(2), cd(3), debug_mode);
         if (timed) {
             HIP_TRY(hipEventRecord(e1, stream));
             HIP_TRY(hipEventSynchronize(e1));
@@ -883,11 +927,167 @@ int rw_agg_kernel_stats(void* h, RwKernelStats* out) {
     return RW_OK;
 }
 
+// debug: scan the device table for duplicate READY keys and report counts
+typedef struct {
+    uint64_t ready_slots;
+    uint64_t dup_keys;
+    uint64_t dirty_count;
+    uint64_t out_cursor;
+} RwAggDebug;
+
+int rw_agg_debug_scan(void* h, RwAggDebug* out) {
+    auto* agg = (HashAgg*)h;
+    size_t cap = (size_t)agg->t.cap_mask + 1;
+    std::vector<uint32_t> state(cap);
+    std::vector<int64_t> keys(cap * agg->KW);
+    std::vector<uint32_t> knulls(cap);
+    uint32_t ctr[3];
+    hipDeviceSynchronize();
+    HIP_TRY(hipMemcpy(state.data(), agg->t.state, cap * 4, hipMemcpyDeviceToHost));
+    HIP_TRY(hipMemcpy(keys.data(), agg->t.keys, cap * agg->KW * 8,
+                      hipMemcpyDeviceToHost));
+    HIP_TRY(hipMemcpy(knulls.data(), agg->t.key_nulls, cap * 4,
+                      hipMemcpyDeviceToHost));
+    HIP_TRY(hipMemcpy(ctr, agg->t.counters, 12, hipMemcpyDeviceToHost));
+    std::unordered_multiset<std::string> seen;
+    uint64_t ready = 0, dups = 0;
+    for (size_t sl = 0; sl < cap; sl++) {
+        if (state[sl] != SLOT_READY) continue;
+        ready++;
+        std::string k((const char*)&keys[sl * agg->KW], agg->KW * 8);
+        k.append((const char*)&knulls[sl], 4);
+        if (seen.count(k)) dups++;
+        seen.insert(k);
+    }
+    out->ready_slots = ready;
+    out->dup_keys = dups;
+    out->dirty_count = ctr[0];
+    out->out_cursor = ctr[1];
+    return RW_OK;
+}
+
 int rw_agg_stats_reset(void* h) {
     auto* agg = (HashAgg*)h;
     agg->apply_launches = 0;
     agg->apply_ms_total = 0;
     agg->apply_rows = 0;
+    return RW_OK;
+}
+
+} // extern "C"
+
+// ---------------------------------------------------------------------------
+// Vnode hashing (VirtualNode::compute_chunk, consistent_hash/vnode.rs:146-181)
+// ---------------------------------------------------------------------------
+// vnode = IEEE CRC-32 (crc32fast) of the row's dist-key datums fed exactly
+// as hash_datum does (native-endian primitive bytes; NULL = u32 0xfffffff0,
+// array/mod.rs:99) mod vnode_count. Oracle counterpart:
+// oracle/oracle_dispatch.cpp; pinned against zlib.crc32 in tests.
+
+__constant__ uint32_t g_crc_table[256];
+static bool g_crc_table_init = false;
+
+__device__ __forceinline__ uint32_t crc32_bytes(uint32_t crc, const uint8_t* p,
+                                                int n) {
+    for (int i = 0; i < n; i++)
+        crc = g_crc_table[(crc ^ p[i]) & 0xFF] ^ (crc >> 8);
+    return crc;
+}
+
+struct VnodeBatch {
+    int64_t* col_vals[MAX_KW];
+    uint8_t* col_valid[MAX_KW];
+    uint8_t* col_types[1]; // unused placeholder alignment
+    uint32_t n_rows;
+};
+
+__global__ void vnode_kernel(VnodeBatch b, int n_keys, uint32_t vnode_count,
+                             uint16_t* out, uint8_t t0, uint8_t t1, uint8_t t2,
+                             uint8_t t3) {
+    uint8_t types[4] = {t0, t1, t2, t3};
+    uint32_t stride = gridDim.x * blockDim.x;
+    for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < b.n_rows;
+         r += stride) {
+        uint32_t crc = 0xFFFFFFFFu;
+        for (int k = 0; k < n_keys; k++) {
+            if (!b.col_valid[k][r]) {
+                uint32_t sentinel = 0xfffffff0u;
+                crc = crc32_bytes(crc, (const uint8_t*)&sentinel, 4);
+            } else {
+                int64_t v = b.col_vals[k][r];
+                int nbytes = (types[k] == RW_T_I32) ? 4 : 8;
+                if (types[k] == RW_T_I32) {
+                    int32_t v32 = (int32_t)v;
+                    crc = crc32_bytes(crc, (const uint8_t*)&v32, 4);
+                } else {
+                    crc = crc32_bytes(crc, (const uint8_t*)&v, 8);
+                }
+                (void)nbytes;
+            }
+        }
+        out[r] = (uint16_t)((uint64_t)(crc ^ 0xFFFFFFFFu) % vnode_count);
+    }
+}
+
+extern "C" {
+
+typedef struct {
+    uint32_t n_keys;
+    const uint32_t* key_indices;
+    uint32_t vnode_count;
+} RwVnodeDesc;
+
+int rw_vnode_compute(const RwVnodeDesc* d, const RwChunk* chunk, uint16_t* out) {
+    if (!gpu_ok()) FAIL(RW_E_NOGPU, "no GPU visible");
+    if (d->n_keys < 1 || d->n_keys > MAX_KW) FAIL(RW_E_INVAL, "n_keys");
+    if (!g_crc_table_init) {
+        uint32_t tab[256];
+        for (uint32_t i = 0; i < 256; i++) {
+            uint32_t c = i;
+            for (int k = 0; k < 8; k++)
+                c = (c & 1) ? 0xEDB88320u ^ (c >> 1) : c >> 1;
+            tab[i] = c;
+        }
+        HIP_TRY(hipMemcpyToSymbol(HIP_SYMBOL(g_crc_table), tab, sizeof tab));
+        g_crc_table_init = true;
+    }
+    uint32_t n = chunk->n_rows;
+    VnodeBatch b{};
+    uint8_t types[4] = {0, 0, 0, 0};
+    for (uint32_t k = 0; k < d->n_keys; k++) {
+        const RwColumn& c = chunk->cols[d->key_indices[k]];
+        types[k] = c.type;
+        if (c.type == RW_T_BOOL) FAIL(RW_E_INVAL, "bool vnode key unsupported on GPU");
+        HIP_TRY(hipMalloc(&b.col_vals[k], (size_t)n * 8));
+        HIP_TRY(hipMalloc(&b.col_valid[k], n));
+        if (c.type == RW_T_I32 || c.type == RW_T_F32) {
+            // widen to i64 storage but hash native width
+            std::vector<int64_t> tmp(n);
+            for (uint32_t r = 0; r < n; r++)
+                tmp[r] = (int64_t)(uint32_t)((const uint32_t*)c.data)[r];
+            HIP_TRY(hipMemcpy(b.col_vals[k], tmp.data(), (size_t)n * 8,
+                              hipMemcpyHostToDevice));
+        } else {
+            HIP_TRY(hipMemcpy(b.col_vals[k], c.data, (size_t)n * 8,
+                              hipMemcpyHostToDevice));
+        }
+        HIP_TRY(hipMemcpy(b.col_valid[k], c.valid, n, hipMemcpyHostToDevice));
+    }
+    b.n_rows = n;
+    uint16_t* dout;
+    HIP_TRY(hipMalloc(&dout, (size_t)n * 2));
+    uint32_t blocks = (n + 255) / 256;
+    if (blocks > 2048) blocks = 2048;
+    if (!blocks) blocks = 1;
+    vnode_kernel<<<blocks, 256>>>(b, (int)d->n_keys, d->vnode_count, dout,
+                                  types[0], types[1], types[2], types[3]);
+    HIP_TRY(hipDeviceSynchronize());
+    HIP_TRY(hipMemcpy(out, dout, (size_t)n * 2, hipMemcpyDeviceToHost));
+    for (uint32_t k = 0; k < d->n_keys; k++) {
+        hipFree(b.col_vals[k]);
+        hipFree(b.col_valid[k]);
+    }
+    hipFree(dout);
     return RW_OK;
 }
 

@@ -278,3 +278,20 @@ def test_join_q8_shape_multiword_key():
             (i % 2, mk_chunk(t4, np.zeros(n, np.uint8), [ids, ws, we, rid]))
         )
     run_join_and_compare(g, o, pushes)
+
+
+def test_vnode_gpu_matches_oracle():
+    # GPU crc32 vnode kernel vs the oracle (itself pinned against zlib in
+    # test_vnode_dispatch.py)
+    from test_vnode_dispatch import compute_vnodes
+
+    rng = np.random.default_rng(21)
+    n = 4096
+    keys = rng.integers(-(2**62), 2**62, n)
+    ws = rng.integers(0, 1 << 40, n)
+    valid = (rng.random(n) > 0.1).astype(np.uint8)
+    c = ffi.Chunk([T_I64, T_TS], np.zeros(n, np.uint8), [keys, ws],
+                  [valid, np.ones(n, np.uint8)])
+    got_gpu = compute_vnodes(gpu(), c, [0, 1])
+    got_orc = compute_vnodes(ffi.oracle(), c, [0, 1])
+    assert got_gpu == got_orc
