@@ -266,7 +266,11 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, int KW, int n_calls,
         uint64_t heads_b = __ballot(head);
         uint64_t le_mask = heads_b & (~0ULL >> (63 - lane));
         int run_start = 63 - __clzll(le_mask | 1ULL);
-        if (active && !head) slot = (uint32_t)__shfl((int)slot, run_start);
+        // the shuffle MUST be executed by all lanes: ds_bpermute returns
+        // undefined data when the SOURCE lane (the head) is exec-masked off,
+        // so a divergent `if (!head) slot = shfl(...)` reads garbage
+        uint32_t bcast_slot = (uint32_t)__shfl((int)slot, run_start);
+        if (active && !head) slot = bcast_slot;
         int run_pos = lane - run_start;
 
         // per-lane contributions (identity when inactive / NULL arg)
@@ -963,6 +967,36 @@ int rw_agg_debug_scan(void* h, RwAggDebug* out) {
     out->dup_keys = dups;
     out->dirty_count = ctr[0];
     out->out_cursor = ctr[1];
+    return RW_OK;
+}
+
+// debug: dump dirty_list entries -> (slot, state, key0, acc[0], acc[1])
+int rw_agg_debug_dirty(void* h, uint32_t max_n, uint32_t* slots, uint32_t* states,
+                       int64_t* key0, long long* acc0, long long* acc1,
+                       uint32_t* n_out) {
+    auto* agg = (HashAgg*)h;
+    size_t cap = (size_t)agg->t.cap_mask + 1;
+    hipDeviceSynchronize();
+    uint32_t ctr[3];
+    HIP_TRY(hipMemcpy(ctr, agg->t.counters, 12, hipMemcpyDeviceToHost));
+    uint32_t n = ctr[0] < max_n ? ctr[0] : max_n;
+    std::vector<uint32_t> dl(n);
+    if (n) HIP_TRY(hipMemcpy(dl.data(), agg->t.dirty_list, n * 4, hipMemcpyDeviceToHost));
+    std::vector<uint32_t> st(cap);
+    std::vector<int64_t> keys(cap * agg->KW);
+    std::vector<long long> acc(cap * agg->n_calls);
+    HIP_TRY(hipMemcpy(st.data(), agg->t.state, cap * 4, hipMemcpyDeviceToHost));
+    HIP_TRY(hipMemcpy(keys.data(), agg->t.keys, cap * agg->KW * 8, hipMemcpyDeviceToHost));
+    HIP_TRY(hipMemcpy(acc.data(), agg->t.acc, cap * agg->n_calls * 8, hipMemcpyDeviceToHost));
+    for (uint32_t i = 0; i < n; i++) {
+        uint32_t sl = dl[i];
+        slots[i] = sl;
+        states[i] = st[sl];
+        key0[i] = keys[(size_t)sl * agg->KW];
+        acc0[i] = acc[sl];
+        acc1[i] = agg->n_calls > 1 ? acc[cap + sl] : 0;
+    }
+    *n_out = n;
     return RW_OK;
 }
 

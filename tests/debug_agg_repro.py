@@ -50,6 +50,7 @@ def main():
         d = Dbg()
         gpu.lib.rw_agg_debug_scan(g.h, ctypes.byref(d))
         print(f"epoch {epoch+1} PRE-FLUSH: ready={d.ready_slots} dup_keys={d.dup_keys} dirty={d.dirty_count} out_cursor={d.out_cursor}")
+        dump_dirty(gpu, g)
         g.flush(epoch + 1)
         o.flush(epoch + 1)
         gpu.lib.rw_agg_debug_scan(g.h, ctypes.byref(d))
@@ -67,6 +68,32 @@ def main():
             print("  missing in GPU:", list(missing.items())[:5])
     g.close()
     o.close()
+
+
+def dump_dirty(gpu, g):
+    import ctypes as C
+
+    gpu.lib.rw_agg_debug_dirty.argtypes = [C.c_void_p, C.c_uint32,
+        C.POINTER(C.c_uint32), C.POINTER(C.c_uint32), C.POINTER(C.c_int64),
+        C.POINTER(C.c_longlong), C.POINTER(C.c_longlong), C.POINTER(C.c_uint32)]
+    N = 4096
+    slots = (C.c_uint32 * N)(); states = (C.c_uint32 * N)()
+    key0 = (C.c_int64 * N)(); a0 = (C.c_longlong * N)(); a1 = (C.c_longlong * N)()
+    n = C.c_uint32()
+    gpu.lib.rw_agg_debug_dirty(g.h, N, slots, states, key0, a0, a1, C.byref(n))
+    rows = [(slots[i], states[i], key0[i], a0[i], a1[i]) for i in range(n.value)]
+    # report duplicate slots and non-READY entries
+    from collections import Counter
+
+    cnt = Counter(r[0] for r in rows)
+    dups = {s: c for s, c in cnt.items() if c > 1}
+    nonready = [r for r in rows if r[1] != 2]
+    keycnt = Counter(r[2] for r in rows)
+    dupkeys = {k: c for k, c in keycnt.items() if c > 1}
+    print(f"  dirty={n.value} dup_slots={dups} nonready={nonready[:5]} dup_keys_in_dirty={dupkeys}")
+    for r in rows:
+        if r[2] in dupkeys:
+            print("   entry:", r)
 
 
 if __name__ == "__main__":
