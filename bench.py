@@ -129,6 +129,10 @@ def main():
                     help="q7 = windowed hash-agg (BASELINE configs[1], the "
                          "default the driver measures); q8 = stream-stream "
                          "hash-join, 10M-key build side (configs[2])")
+    ap.add_argument("--exchange", choices=["auto", "on", "off"], default="auto",
+                    help="vnode partition + RCCL all-to-all-v before the agg "
+                         "(the reference's HashDataDispatcher hop, SURVEY "
+                         "§8e). auto = on when WORLD_SIZE > 1")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -142,8 +146,9 @@ def main():
         import torch.distributed as dist_mod
 
         dist = dist_mod
-        backend = "nccl" if _cuda_available() else "gloo"
-        dist.init_process_group(backend=backend)
+        # gloo for control (timing barriers, uniqueId broadcast); the DATA
+        # plane collective is our own RCCL communicator (rw_exchange)
+        dist.init_process_group(backend="gloo")
 
     from rwtest import ffi
 
@@ -180,10 +185,21 @@ def main():
     agg = ffi.HashAgg(gpu_lib, [T_I64, T_I64], [0], calls, 1, append_only=True,
                       state_capacity_hint=1 << 20)
 
+    use_exchange = args.exchange == "on" or (args.exchange == "auto" and world > 1)
+    exch = None
+    if use_exchange:
+        exch = setup_exchange(ffi, rank, world, dist)
+        if exch is None:
+            print(f"# rank {rank}: RCCL exchange init failed "
+                  f"({'see stderr' if rank == 0 else ''}) — falling back to "
+                  f"pre-routed weak scaling", file=sys.stderr)
+            use_exchange = False
+
     batch_rows = CHUNK_ROWS * CHUNKS_PER_BATCH
-    # rank-local window space (weak scaling: upstream exchange already routed
-    # window keys)
-    window_base = rank * 1_000_000 * WINDOW_US
+    # Without exchange: rank-local window space (weak scaling, upstream
+    # exchange already routed keys — DESIGN.md §7). With exchange: a GLOBAL
+    # window space; the RCCL all-to-all routes each window to its vnode owner.
+    window_base = 0 if use_exchange else rank * 1_000_000 * WINDOW_US
     # 16 × 19 MB ≈ 300 MB of resident input > the 256 MB Infinity Cache, so
     # the timed region streams from HBM (L3-masking gotcha,
     # cdna_hip_programming.md §2)
@@ -198,9 +214,19 @@ def main():
         assert h, gpu_lib.last_error()
         batches.append(h)
 
+    if use_exchange:
+        payload_cap = int(batch_rows * (1 + 2 * 9) * 4)  # 4x headroom for skew
+        xb = exch.make_buffers(payload_cap)
+
     def step(i):
-        rc = L.rw_agg_bench_apply(agg.h, batches[i % n_batches])
-        assert rc == 0, gpu_lib.last_error()
+        if use_exchange:
+            recv_blocks = exch.run(agg.h, batches[i % n_batches], xb)
+            rc = L.rw_agg_apply_payload(
+                agg.h, xb.recv, recv_blocks, world, 2)
+            assert rc == 0, gpu_lib.last_error()
+        else:
+            rc = L.rw_agg_bench_apply(agg.h, batches[i % n_batches])
+            assert rc == 0, gpu_lib.last_error()
         if (i + 1) % args.barrier_every == 0:
             agg.flush(i)
             agg.poll_all()
@@ -281,6 +307,96 @@ def main():
         dist.destroy_process_group()
 
 
+class ExchangeCtx:
+    """Wrapper over librw_exchange.so: vnode partition kernel + RCCL
+    all-to-all-v (DESIGN.md §7 / SURVEY §8e)."""
+
+    class Buffers:
+        pass
+
+    def __init__(self, lib, h, world):
+        self.lib = lib
+        self.h = h
+        self.world = world
+
+    def make_buffers(self, cap):
+        L = self.lib
+        L.rw_xbuf_alloc.restype = ctypes.c_void_p
+        L.rw_xbuf_alloc.argtypes = [ctypes.c_uint64]
+        b = self.Buffers()
+        b.send = L.rw_xbuf_alloc(cap)
+        b.recv = L.rw_xbuf_alloc(cap)
+        b.cap = cap
+        assert b.send and b.recv
+        return b
+
+    def run(self, agg_h, batch, xb, gpu_lib=None):
+        L = self.lib
+        import risingwave_amd
+
+        A = ctypes.CDLL(risingwave_amd.lib_path())
+        NSLOT = 12  # MAX_KW + MAX_CALLS
+        vals = (ctypes.c_void_p * NSLOT)()
+        valids = (ctypes.c_void_p * NSLOT)()
+        ops = ctypes.c_void_p()
+        nrows = ctypes.c_uint32()
+        A.rw_agg_batch_ptrs.argtypes = [ctypes.c_void_p,
+                                        ctypes.POINTER(ctypes.c_void_p),
+                                        ctypes.POINTER(ctypes.c_void_p),
+                                        ctypes.POINTER(ctypes.c_void_p),
+                                        ctypes.POINTER(ctypes.c_uint32)]
+        A.rw_agg_batch_ptrs(batch, vals, valids, ctypes.byref(ops),
+                            ctypes.byref(nrows))
+        key_cols = (ctypes.c_uint32 * 1)(0)  # batch slot 0 = window key
+        send_counts = (ctypes.c_uint64 * self.world)()
+        recv_counts = (ctypes.c_uint64 * self.world)()
+        L.rw_exchange_run.restype = ctypes.c_int
+        rc = L.rw_exchange_run(
+            self.h, ctypes.cast(vals, ctypes.c_void_p),
+            ctypes.cast(valids, ctypes.c_void_p), ops, nrows, 2, key_cols, 1,
+            256, ctypes.c_void_p(xb.send), ctypes.c_uint64(xb.cap),
+            ctypes.c_void_p(xb.recv), ctypes.c_uint64(xb.cap), send_counts,
+            recv_counts)
+        if rc != 0:
+            L.rw_exchange_last_error.restype = ctypes.c_char_p
+            raise RuntimeError(f"exchange failed {rc}: "
+                               f"{L.rw_exchange_last_error().decode()}")
+        return recv_counts
+
+    def stats(self):
+        ms = ctypes.c_double()
+        n = ctypes.c_uint64()
+        self.lib.rw_exchange_stats.argtypes = [ctypes.c_void_p,
+                                               ctypes.POINTER(ctypes.c_double),
+                                               ctypes.POINTER(ctypes.c_uint64)]
+        self.lib.rw_exchange_stats(self.h, ctypes.byref(ms), ctypes.byref(n))
+        return ms.value, n.value
+
+
+def setup_exchange(ffi, rank, world, dist):
+    try:
+        lib = ctypes.CDLL(os.path.join(REPO, "risingwave_amd",
+                                       "librw_exchange.so"))
+    except OSError as e:
+        print(f"# librw_exchange.so load failed: {e}", file=sys.stderr)
+        return None
+    lib.rw_exchange_unique_id_size.restype = ctypes.c_int
+    sz = lib.rw_exchange_unique_id_size()
+    uid = (ctypes.c_uint8 * sz)()
+    if rank == 0:
+        if lib.rw_exchange_get_unique_id(uid) != 0:
+            return None
+    if world > 1 and dist is not None:
+        obj = [bytes(bytearray(uid))] if rank == 0 else [None]
+        dist.broadcast_object_list(obj, src=0)
+        uid = (ctypes.c_uint8 * sz).from_buffer_copy(obj[0])
+    lib.rw_exchange_create.restype = ctypes.c_void_p
+    h = lib.rw_exchange_create(world, rank, uid)
+    if not h:
+        return None
+    return ExchangeCtx(lib, h, world)
+
+
 def bench_q8(args, ffi, gpu_lib, rng, rank, world, dist):
     """Nexmark q8 stream-stream hash-join (BASELINE configs[2]): build side =
     10M distinct person ids resident in HBM, probe = auction.seller
@@ -353,7 +469,8 @@ def bench_q8(args, ffi, gpu_lib, rng, rank, world, dist):
 
     for i in range(args.warmup):
         step(i)
-    j.lib.lib.rw_join_stats_reset(j.h)
+    L.rw_join_stats_reset.argtypes = [ctypes.c_void_p]
+    L.rw_join_stats_reset(j.h)
     if dist:
         dist.barrier()
     t0 = time.perf_counter()

@@ -220,9 +220,10 @@ struct AggTableDev {
 // 64× (the unsorted case degrades to runs of 1 = plain per-lane atomics,
 // still correct). count/sum combine by +, min/max by min/max, null-presence
 // by OR; all order-free, so the result is exactly the reference's.
-__global__ void agg_apply_kernel(AggBatch b, AggTableDev t, int KW, int n_calls,
-                                 AggCallDev c0, AggCallDev c1, AggCallDev c2,
-                                 AggCallDev c3, int mode) {
+template <int KW, int n_calls>
+__global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
+                                 AggCallDev c1, AggCallDev c2, AggCallDev c3,
+                                 int mode) {
     AggCallDev calls[4] = {c0, c1, c2, c3};
     uint32_t stride = gridDim.x * blockDim.x;
     uint32_t iters = (b.n_rows + stride - 1) / stride;
@@ -698,6 +699,33 @@ struct HashAgg {
         return RW_OK;
     }
 
+    void launch_apply(const AggBatch& b) {
+        int grid = grid_for(b.n_rows);
+        auto a0 = cd(0), a1 = cd(1), a2 = cd(2), a3 = cd(3);
+        #define RW_LAUNCH(kw, nc)                                             \
+            agg_apply_kernel<kw, nc><<<grid, 256, 0, stream>>>(               \
+                b, t, a0, a1, a2, a3, debug_mode)
+        switch (KW * 8 + n_calls) {
+            case 1 * 8 + 1: RW_LAUNCH(1, 1); break;
+            case 1 * 8 + 2: RW_LAUNCH(1, 2); break;
+            case 1 * 8 + 3: RW_LAUNCH(1, 3); break;
+            case 1 * 8 + 4: RW_LAUNCH(1, 4); break;
+            case 2 * 8 + 1: RW_LAUNCH(2, 1); break;
+            case 2 * 8 + 2: RW_LAUNCH(2, 2); break;
+            case 2 * 8 + 3: RW_LAUNCH(2, 3); break;
+            case 2 * 8 + 4: RW_LAUNCH(2, 4); break;
+            case 3 * 8 + 1: RW_LAUNCH(3, 1); break;
+            case 3 * 8 + 2: RW_LAUNCH(3, 2); break;
+            case 3 * 8 + 3: RW_LAUNCH(3, 3); break;
+            case 3 * 8 + 4: RW_LAUNCH(3, 4); break;
+            case 4 * 8 + 1: RW_LAUNCH(4, 1); break;
+            case 4 * 8 + 2: RW_LAUNCH(4, 2); break;
+            case 4 * 8 + 3: RW_LAUNCH(4, 3); break;
+            case 4 * 8 + 4: RW_LAUNCH(4, 4); break;
+        }
+        #undef RW_LAUNCH
+    }
+
     int apply(const AggBatch& b, bool timed) {
         hipEvent_t e0 = nullptr, e1 = nullptr;
         if (timed) {
@@ -705,8 +733,7 @@ struct HashAgg {
             HIP_TRY(hipEventCreate(&e1));
             HIP_TRY(hipEventRecord(e0, stream));
         }
-        agg_apply_kernel<<<grid_for(b.n_rows), 256, 0, stream>>>(
-            b, t, KW, n_calls, cd(0), cd(1), cd(2), cd(3), debug_mode);
+        launch_apply(b);
         if (timed) {
             HIP_TRY(hipEventRecord(e1, stream));
             HIP_TRY(hipEventSynchronize(e1));
@@ -923,6 +950,48 @@ typedef struct {
     uint64_t rows;
 } RwKernelStats;
 
+// expose a preloaded batch's device pointers (for the exchange path)
+int rw_agg_batch_ptrs(void* batch, const int64_t** vals, const uint8_t** valids,
+                      const uint8_t** ops, uint32_t* n_rows) {
+    auto* b = (AggBatch*)batch;
+    for (int i = 0; i < MAX_KW + MAX_CALLS; i++) {
+        vals[i] = b->col_vals[i];
+        valids[i] = b->col_valid[i];
+    }
+    *ops = b->ops;
+    *n_rows = b->n_rows;
+    return RW_OK;
+}
+
+// apply an exchange payload (device-resident; per-block layout
+// vals[col][n] ∥ valid[col][n] ∥ ops[n], cols in the executor's batch slot
+// order: group key cols then call args) — one apply launch per block
+int rw_agg_apply_payload(void* h, const uint8_t* payload,
+                         const uint64_t* block_rows, int n_blocks, int n_cols) {
+    auto* agg = (HashAgg*)h;
+    if (n_cols != agg->KW + agg->n_calls)
+        FAIL(RW_E_INVAL, "payload n_cols %d != %d", n_cols, agg->KW + agg->n_calls);
+    uint64_t row_bytes = 1 + (uint64_t)n_cols * 9;
+    uint64_t off = 0;
+    for (int bi = 0; bi < n_blocks; bi++) {
+        uint64_t n = block_rows[bi];
+        if (!n) continue;
+        AggBatch b{};
+        for (int c = 0; c < n_cols; c++) {
+            b.col_vals[c] = (int64_t*)(payload + off + (uint64_t)c * n * 8);
+            b.col_valid[c] =
+                (uint8_t*)(payload + off + (uint64_t)n_cols * n * 8 + (uint64_t)c * n);
+        }
+        b.ops = (uint8_t*)(payload + off + (uint64_t)n_cols * n * 9);
+        b.vis = nullptr;
+        b.n_rows = (uint32_t)n;
+        int rc = agg->apply(b, true);
+        if (rc != RW_OK) return rc;
+        off += (n * row_bytes + 7) & ~7ull;
+    }
+    return RW_OK;
+}
+
 int rw_agg_kernel_stats(void* h, RwKernelStats* out) {
     auto* agg = (HashAgg*)h;
     out->launches = agg->apply_launches;
@@ -1122,6 +1191,78 @@ int rw_vnode_compute(const RwVnodeDesc* d, const RwChunk* chunk, uint16_t* out) 
         hipFree(b.col_valid[k]);
     }
     hipFree(dout);
+    return RW_OK;
+}
+
+typedef struct {
+    RwVnodeDesc v;
+    uint32_t n_outputs;
+    const uint32_t* vnode_to_output;
+} RwDispatchDesc;
+
+// HashDataDispatcher::dispatch_data (dispatch.rs:949-1050): vnodes from the
+// GPU kernel; the tiny per-row visibility/op rewrite is host-side (the
+// device-side dense compaction lives in rw_exchange.hip's partition path).
+int rw_dispatch_compute(const RwDispatchDesc* d, const RwChunk* chunk,
+                        RwChunk** outs) {
+    uint32_t n = chunk->n_rows;
+    std::vector<uint16_t> vnodes(n);
+    int rc = rw_vnode_compute(&d->v, chunk, vnodes.data());
+    if (rc != RW_OK) return rc;
+
+    std::vector<uint8_t> ops(chunk->ops, chunk->ops + n);
+    auto visible = [&](uint32_t r) { return !chunk->vis || chunk->vis[r]; };
+    auto datum_eq_i64 = [&](uint32_t col, uint32_t a, uint32_t b) {
+        const RwColumn& c = chunk->cols[col];
+        if (c.valid[a] != c.valid[b]) return false;
+        if (!c.valid[a]) return true;
+        switch (c.type) {
+            case RW_T_I32: return ((const int32_t*)c.data)[a] == ((const int32_t*)c.data)[b];
+            default: return ((const int64_t*)c.data)[a] == ((const int64_t*)c.data)[b];
+        }
+    };
+    long last_ud = -1;
+    for (uint32_t r = 0; r < n; r++) {
+        if (!visible(r)) continue;
+        if (ops[r] == RW_OP_UPDATE_DELETE) {
+            last_ud = (long)r;
+        } else if (ops[r] == RW_OP_UPDATE_INSERT && last_ud >= 0) {
+            bool changed = false;
+            for (uint32_t k = 0; k < d->v.n_keys && !changed; k++)
+                changed = !datum_eq_i64(d->v.key_indices[k], (uint32_t)last_ud, r);
+            if (changed) {
+                ops[last_ud] = RW_OP_DELETE;
+                ops[r] = RW_OP_INSERT;
+            }
+            last_ud = -1;
+        }
+    }
+
+    for (uint32_t o = 0; o < d->n_outputs; o++) {
+        auto* ch = new RwChunk();
+        auto* cols = new RwColumn[chunk->n_cols];
+        auto* op_arr = new uint8_t[n];
+        auto* vis = new uint8_t[n];
+        memcpy(op_arr, ops.data(), n);
+        for (uint32_t r = 0; r < n; r++)
+            vis[r] = visible(r) && d->vnode_to_output[vnodes[r]] == o ? 1 : 0;
+        for (uint32_t c = 0; c < chunk->n_cols; c++) {
+            uint32_t sz = rw_type_size(chunk->cols[c].type);
+            auto* data = new int64_t[n]; // freed as int64_t* by rw_chunk_free
+            auto* valid = new uint8_t[n];
+            memcpy(data, chunk->cols[c].data, (size_t)n * sz);
+            memcpy(valid, chunk->cols[c].valid, n);
+            cols[c].type = chunk->cols[c].type;
+            cols[c].valid = valid;
+            cols[c].data = data;
+        }
+        ch->n_rows = n;
+        ch->n_cols = chunk->n_cols;
+        ch->ops = op_arr;
+        ch->vis = vis;
+        ch->cols = cols;
+        outs[o] = ch;
+    }
     return RW_OK;
 }
 
