@@ -175,7 +175,9 @@ __device__ __forceinline__ uint32_t table_find(const uint32_t* state,
 
 struct AggCallDev {
     uint8_t kind;
-    int32_t arg; // column index in the input batch, -1 for count(*)
+    int32_t arg;  // column index in the input BATCH (gk cols ∥ arg cols ∥ sk cols)
+    uint8_t minput; // materialized-input state (retractable min/max)
+    int8_t mord;    // ordinal among minput calls
 };
 
 // Device-side batch of input rows (SoA), i64-widened values.
@@ -206,6 +208,20 @@ struct AggTableDev {
     uint8_t* out_ops;
     uint32_t out_capacity;
     uint32_t cap_mask;
+    // materialized-input state (retractable min/max, aggregate/minput.rs):
+    // shared row store + per-(minput call, slot) chain heads; rows carry
+    // (arg value, stream key) and a CAS-claimed alive flag, mirroring the
+    // state-table rows keyed [group, value, stream key]
+    uint32_t* mheads;   // [n_minput][cap], UINT32_MAX = empty
+    long long* mval;    // [mrow_cap]
+    uint8_t* mval_null; // [mrow_cap]
+    long long* msk;     // [n_sk][mrow_cap]
+    uint8_t* msk_null;  // [n_sk][mrow_cap]
+    uint32_t* mnext;
+    uint32_t* malive;
+    uint32_t* mcursor; // single counter
+    uint32_t mrow_cap;
+    int n_sk;
 };
 
 // agg_apply: HashAggExecutor::apply_chunk (hash_agg.rs:332-409) as one
@@ -223,17 +239,17 @@ struct AggTableDev {
 template <int KW, int n_calls>
 __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
                                  AggCallDev c1, AggCallDev c2, AggCallDev c3,
-                                 int mode) {
+                                 int mode, uint32_t r0, uint32_t r1) {
     AggCallDev calls[4] = {c0, c1, c2, c3};
     uint32_t stride = gridDim.x * blockDim.x;
-    uint32_t iters = (b.n_rows + stride - 1) / stride;
+    uint32_t iters = (r1 - r0 + stride - 1) / stride;
     int lane = threadIdx.x & 63;
     size_t cap = (size_t)t.cap_mask + 1;
     const uint32_t SLOT_NONE = (uint32_t)-1;
 
     for (uint32_t it = 0; it < iters; it++) {
-        uint32_t r = it * stride + blockIdx.x * blockDim.x + threadIdx.x;
-        bool active = (r < b.n_rows) && !(b.vis && !b.vis[r]);
+        uint32_t r = r0 + it * stride + blockIdx.x * blockDim.x + threadIdx.x;
+        bool active = (r < r1) && !(b.vis && !b.vis[r]);
         int64_t kw[MAX_KW];
         uint32_t nullmask = 0;
         for (int i = 0; i < KW; i++) kw[i] = 0;
@@ -285,6 +301,62 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
             const AggCallDev& c = calls[ci];
             int col = KW + ci;
             bool arg_valid = contributing && b.col_valid[col][r];
+            if (c.minput) {
+                // materialized-input ops are exact per-row chain mutations
+                // (minput.rs apply_batch): no wave aggregation
+                v[ci] = 0;
+                if (contributing) {
+                    uint32_t* headp = t.mheads + (size_t)c.mord * cap + slot;
+                    bool valid = b.col_valid[col][r];
+                    long long val = valid ? b.col_vals[col][r] : 0;
+                    if (sign > 0) {
+                        uint32_t row = atomicAdd(t.mcursor, 1u);
+                        if (row >= t.mrow_cap) {
+                            atomicExch(&t.counters[2], 3u); // minput store full
+                        } else {
+                            st_i64((int64_t*)&t.mval[row], val);
+                            t.mval_null[row] = !valid;
+                            for (int k = 0; k < t.n_sk; k++) {
+                                int skc = KW + n_calls + k;
+                                st_i64((int64_t*)&t.msk[(size_t)k * t.mrow_cap + row],
+                                       b.col_valid[skc][r] ? b.col_vals[skc][r] : 0);
+                                t.msk_null[(size_t)k * t.mrow_cap + row] =
+                                    !b.col_valid[skc][r];
+                            }
+                            st_u32(&t.malive[row], 1);
+                            uint32_t old_head = ld_u32(headp);
+                            for (;;) {
+                                st_u32(&t.mnext[row], old_head);
+                                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+                                uint32_t prev = atomicCAS(headp, old_head, row);
+                                if (prev == old_head) break;
+                                old_head = prev;
+                            }
+                        }
+                    } else {
+                        // delete: CAS-kill one matching (value, stream key)
+                        uint32_t row = ld_u32(headp);
+                        while (row != UINT32_MAX) {
+                            if (ld_u32(&t.malive[row])) {
+                                bool eq = (t.mval_null[row] == (uint8_t)!valid) &&
+                                          (!valid || ld_i64((const int64_t*)&t.mval[row]) == val);
+                                for (int k = 0; eq && k < t.n_sk; k++) {
+                                    int skc = KW + n_calls + k;
+                                    uint8_t va = b.col_valid[skc][r];
+                                    uint8_t vb = !t.msk_null[(size_t)k * t.mrow_cap + row];
+                                    eq = (va == vb) &&
+                                         (!va || b.col_vals[skc][r] ==
+                                                     ld_i64((const int64_t*)&t.msk[(size_t)k * t.mrow_cap + row]));
+                                }
+                                if (eq && atomicCAS(&t.malive[row], 1u, 0u) == 1u)
+                                    break;
+                            }
+                            row = ld_u32(&t.mnext[row]);
+                        }
+                    }
+                }
+                continue;
+            }
             switch (c.kind) {
                 case RW_AGG_COUNT_STAR: v[ci] = contributing ? sign : 0; break;
                 case RW_AGG_COUNT: v[ci] = arg_valid ? sign : 0; break;
@@ -307,6 +379,7 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
             // debug fallback: per-lane atomics, no wave aggregation
             if (contributing) {
                 for (int ci = 0; ci < n_calls; ci++) {
+                    if (calls[ci].minput) continue;
                     long long* acc = t.acc + (size_t)ci * cap;
                     uint8_t* has = t.has + (size_t)ci * cap;
                     switch (calls[ci].kind) {
@@ -367,6 +440,7 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
         bool tail = contributing && ((ends_after >> lane) & 1);
         if (tail) {
             for (int ci = 0; ci < n_calls; ci++) {
+                if (calls[ci].minput) continue;
                 long long* acc = t.acc + (size_t)ci * cap;
                 uint8_t* has = t.has + (size_t)ci * cap;
                 switch (calls[ci].kind) {
@@ -439,6 +513,38 @@ __global__ void agg_flush_kernel(AggTableDev t, int KW, int n_calls,
                     default: acc[slot] = 0;
                 }
                 has[slot] = 0;
+            }
+            if (c.minput) {
+                // output_first over the materialized rows (minput.rs:236-241):
+                // min = value ASC NULLS LAST → smallest non-null, NULL if all
+                // null; max = value DESC NULLS FIRST → NULL if any NULL row
+                // (the reference's order types, test_utils/agg_executor.rs:
+                // 96-99 + sort_util.rs NullsAre::Largest)
+                uint32_t row = t.mheads[(size_t)c.mord * cap + slot];
+                bool any = false, any_null = false;
+                long long best = 0;
+                while (row != UINT32_MAX) {
+                    if (t.malive[row]) {
+                        if (t.mval_null[row]) {
+                            any_null = true;
+                        } else {
+                            long long v = t.mval[row];
+                            if (!any) best = v;
+                            else if (c.kind == RW_AGG_MIN) best = v < best ? v : best;
+                            else best = v > best ? v : best;
+                            any = true;
+                        }
+                    }
+                    row = t.mnext[row];
+                }
+                if (c.kind == RW_AGG_MAX && any_null) {
+                    curr[ci] = 0;
+                    curr_null[ci] = 1;
+                } else {
+                    curr[ci] = best;
+                    curr_null[ci] = !any;
+                }
+                continue;
             }
             switch (c.kind) {
                 case RW_AGG_COUNT_STAR:
@@ -554,6 +660,9 @@ struct HashAgg {
     // pending outputs
     std::vector<RwChunk*> outq;
     int debug_mode = 0; // RW_AGG_DEBUG_MODE: 1 per-lane atomics, 2 no-dedupe
+    int n_minput = 0;   // materialized-input (retractable min/max) calls
+    std::vector<uint8_t> call_minput;
+    std::vector<uint32_t> stream_key;
 
     int grid_for(uint32_t work) const {
         uint32_t blocks = (work + 255) / 256;
@@ -562,8 +671,12 @@ struct HashAgg {
     }
 
     AggCallDev cd(int i) const {
-        if (i < n_calls) return AggCallDev{calls[i].kind, calls[i].arg};
-        return AggCallDev{0, -1};
+        if (i < n_calls) {
+            int8_t mord = 0;
+            for (int j = 0; j < i; j++) mord += call_minput[j];
+            return AggCallDev{calls[i].kind, calls[i].arg, call_minput[i], mord};
+        }
+        return AggCallDev{0, -1, 0, 0};
     }
 
     int init(const RwHashAggDesc* d) {
@@ -582,14 +695,26 @@ struct HashAgg {
             if (ty != RW_T_I64 && ty != RW_T_TS)
                 FAIL(RW_E_INVAL, "group key type %d unsupported on GPU (i64/ts only)", ty);
         }
+        stream_key.assign(d->stream_key, d->stream_key + d->n_stream_key);
         for (auto& c : calls) {
             if (c.kind > RW_AGG_MAX) FAIL(RW_E_INVAL, "agg kind %d", c.kind);
-            if ((c.kind == RW_AGG_MIN || c.kind == RW_AGG_MAX) && !d->append_only)
-                FAIL(RW_E_INVAL, "retractable min/max not in round-1 kernels (SURVEY §8f-1)");
+            bool minput = (c.kind == RW_AGG_MIN || c.kind == RW_AGG_MAX) &&
+                          !d->append_only;
+            call_minput.push_back(minput);
+            n_minput += minput;
             if (c.arg >= 0) {
                 uint8_t ty = input_types[c.arg];
                 if (ty != RW_T_I64 && ty != RW_T_TS)
                     FAIL(RW_E_INVAL, "agg arg type %d unsupported on GPU (i64/ts only)", ty);
+            }
+        }
+        if (n_minput) {
+            if (stream_key.size() > 2)
+                FAIL(RW_E_INVAL, "stream key wider than 2 unsupported for GPU minput state");
+            for (auto sk : stream_key) {
+                uint8_t ty = input_types[sk];
+                if (ty != RW_T_I64 && ty != RW_T_TS)
+                    FAIL(RW_E_INVAL, "stream key type %d unsupported on GPU", ty);
             }
         }
         out_width = KW + n_calls;
@@ -623,18 +748,40 @@ struct HashAgg {
         HIP_TRY(hipMalloc(&t.out_vals, (size_t)t.out_capacity * out_width * 8));
         HIP_TRY(hipMalloc(&t.out_nulls, (size_t)t.out_capacity * out_width));
         HIP_TRY(hipMalloc(&t.out_ops, t.out_capacity));
+        t.n_sk = (int)stream_key.size();
+        if (n_minput) {
+            t.mrow_cap = 1u << 22;
+            if (desc.state_capacity_hint) {
+                uint64_t want = desc.state_capacity_hint * 8;
+                t.mrow_cap = 1;
+                while (t.mrow_cap < want && t.mrow_cap < (1u << 28)) t.mrow_cap <<= 1;
+            }
+            HIP_TRY(hipMalloc(&t.mheads, (size_t)n_minput * cap * 4));
+            HIP_TRY(hipMemset(t.mheads, 0xFF, (size_t)n_minput * cap * 4));
+            HIP_TRY(hipMalloc(&t.mval, (size_t)t.mrow_cap * 8));
+            HIP_TRY(hipMalloc(&t.mval_null, t.mrow_cap));
+            size_t nsk = t.n_sk ? t.n_sk : 1;
+            HIP_TRY(hipMalloc(&t.msk, nsk * t.mrow_cap * 8));
+            HIP_TRY(hipMalloc(&t.msk_null, nsk * t.mrow_cap));
+            HIP_TRY(hipMalloc(&t.mnext, (size_t)t.mrow_cap * 4));
+            HIP_TRY(hipMalloc(&t.malive, (size_t)t.mrow_cap * 4));
+            HIP_TRY(hipMalloc(&t.mcursor, 4));
+            HIP_TRY(hipMemset(t.mcursor, 0, 4));
+        }
         agg_init_kernel<<<2048, 256, 0, stream>>>(t, n_calls, cd(0), cd(1), cd(2),
                                                   cd(3));
         HIP_TRY(hipStreamSynchronize(stream));
         return RW_OK;
     }
 
+    int n_batch_slots() const { return KW + n_calls + (n_minput ? (int)stream_key.size() : 0); }
+
     int ensure_stage(uint32_t n_rows) {
         if (stage_cap >= n_rows) return RW_OK;
         free_stage();
         uint32_t cap = 4096;
         while (cap < n_rows) cap <<= 1;
-        for (int i = 0; i < KW + n_calls; i++) {
+        for (int i = 0; i < n_batch_slots(); i++) {
             HIP_TRY(hipMalloc(&stage.col_vals[i], (size_t)cap * 8));
             HIP_TRY(hipMalloc(&stage.col_valid[i], cap));
         }
@@ -644,7 +791,7 @@ struct HashAgg {
         return RW_OK;
     }
     void free_stage() {
-        for (int i = 0; i < KW + n_calls; i++) {
+        for (int i = 0; i < n_batch_slots(); i++) {
             if (stage.col_vals[i]) hipFree(stage.col_vals[i]);
             if (stage.col_valid[i]) hipFree(stage.col_valid[i]);
             stage.col_vals[i] = nullptr;
@@ -688,6 +835,12 @@ struct HashAgg {
                 HIP_TRY(hipMemsetAsync(b.col_valid[KW + ci], 1, n, stream));
             }
         }
+        if (n_minput) {
+            for (size_t j = 0; j < stream_key.size(); j++) {
+                int rc = upcol(KW + n_calls + (int)j, stream_key[j]);
+                if (rc != RW_OK) return rc;
+            }
+        }
         HIP_TRY(hipMemcpyAsync(b.ops, c->ops, n, hipMemcpyHostToDevice, stream));
         if (c->vis) {
             HIP_TRY(hipMemcpyAsync(b.vis, c->vis, n, hipMemcpyHostToDevice, stream));
@@ -699,12 +852,12 @@ struct HashAgg {
         return RW_OK;
     }
 
-    void launch_apply(const AggBatch& b) {
-        int grid = grid_for(b.n_rows);
+    void launch_apply(const AggBatch& b, uint32_t r0, uint32_t r1) {
+        int grid = grid_for(r1 - r0);
         auto a0 = cd(0), a1 = cd(1), a2 = cd(2), a3 = cd(3);
         #define RW_LAUNCH(kw, nc)                                             \
             agg_apply_kernel<kw, nc><<<grid, 256, 0, stream>>>(               \
-                b, t, a0, a1, a2, a3, debug_mode)
+                b, t, a0, a1, a2, a3, debug_mode, r0, r1)
         switch (KW * 8 + n_calls) {
             case 1 * 8 + 1: RW_LAUNCH(1, 1); break;
             case 1 * 8 + 2: RW_LAUNCH(1, 2); break;
@@ -726,14 +879,20 @@ struct HashAgg {
         #undef RW_LAUNCH
     }
 
-    int apply(const AggBatch& b, bool timed) {
+    int apply(const AggBatch& b, bool timed,
+              const std::vector<uint32_t>& seg_bounds = {}) {
         hipEvent_t e0 = nullptr, e1 = nullptr;
         if (timed) {
             HIP_TRY(hipEventCreate(&e0));
             HIP_TRY(hipEventCreate(&e1));
             HIP_TRY(hipEventRecord(e0, stream));
         }
-        launch_apply(b);
+        uint32_t start = 0;
+        for (uint32_t bnd : seg_bounds) {
+            if (bnd > start) launch_apply(b, start, bnd);
+            start = bnd;
+        }
+        if (start < b.n_rows) launch_apply(b, start, b.n_rows);
         if (timed) {
             HIP_TRY(hipEventRecord(e1, stream));
             HIP_TRY(hipEventSynchronize(e1));
@@ -748,11 +907,55 @@ struct HashAgg {
         return RW_OK;
     }
 
+    // A minput DELETE targeting a row INSERTed earlier in the same chunk
+    // must observe that insert (the reference applies rows in order,
+    // hash_agg.rs:332-398); the parallel kernel keeps that only across
+    // launches, so such deletes run in their own single-row segments.
+    std::vector<uint32_t> minput_conflict_segments(const RwChunk* c) {
+        std::vector<uint32_t> bounds;
+        if (!n_minput) return bounds;
+        bool any_delete = false;
+        for (uint32_t r = 0; r < c->n_rows && !any_delete; r++)
+            any_delete = c->ops[r] == RW_OP_DELETE || c->ops[r] == RW_OP_UPDATE_DELETE;
+        if (!any_delete) return bounds;
+        auto row_key = [&](uint32_t r) {
+            // (group key, per-minput arg, stream key) — conservative match
+            std::string k;
+            auto add = [&](uint32_t col) {
+                uint8_t valid = c->cols[col].valid[r];
+                k.push_back((char)valid);
+                int64_t v = valid ? ((const int64_t*)c->cols[col].data)[r] : 0;
+                k.append((const char*)&v, 8);
+            };
+            for (auto g : group_key) add(g);
+            for (int ci = 0; ci < n_calls; ci++)
+                if (call_minput[ci] && calls[ci].arg >= 0) add((uint32_t)calls[ci].arg);
+            for (auto sk : stream_key) add(sk);
+            return k;
+        };
+        std::unordered_multiset<std::string> inserts;
+        for (uint32_t r = 0; r < c->n_rows; r++) {
+            if (c->vis && !c->vis[r]) continue;
+            if (c->ops[r] == RW_OP_INSERT || c->ops[r] == RW_OP_UPDATE_INSERT)
+                inserts.insert(row_key(r));
+        }
+        if (inserts.empty()) return bounds;
+        for (uint32_t r = 0; r < c->n_rows; r++) {
+            if (c->vis && !c->vis[r]) continue;
+            if ((c->ops[r] == RW_OP_DELETE || c->ops[r] == RW_OP_UPDATE_DELETE) &&
+                inserts.count(row_key(r))) {
+                bounds.push_back(r);
+                bounds.push_back(r + 1);
+            }
+        }
+        return bounds;
+    }
+
     int push_chunk(const RwChunk* c) {
         AggBatch b;
         int rc = upload(c, &b, true);
         if (rc != RW_OK) return rc;
-        rc = apply(b, true);
+        rc = apply(b, true, minput_conflict_segments(c));
         if (rc != RW_OK) return rc;
         HIP_TRY(hipStreamSynchronize(stream)); // staging buffer reuse
         return check_overflow();
@@ -763,6 +966,7 @@ struct HashAgg {
         HIP_TRY(hipMemcpy(ctr, t.counters, 12, hipMemcpyDeviceToHost));
         if (ctr[2] == 1) FAIL(RW_E_INTERNAL, "agg state table full (capacity %u)", capacity);
         if (ctr[2] == 2) FAIL(RW_E_INTERNAL, "agg output buffer overflow");
+        if (ctr[2] == 3) FAIL(RW_E_INTERNAL, "agg minput row store full");
         return RW_OK;
     }
 
@@ -862,6 +1066,16 @@ struct HashAgg {
             hipFree(t.out_vals);
             hipFree(t.out_nulls);
             hipFree(t.out_ops);
+            if (t.mheads) {
+                hipFree(t.mheads);
+                hipFree(t.mval);
+                hipFree(t.mval_null);
+                hipFree(t.msk);
+                hipFree(t.msk_null);
+                hipFree(t.mnext);
+                hipFree(t.malive);
+                hipFree(t.mcursor);
+            }
             hipStreamDestroy(stream);
         }
         for (auto* c : outq) rw_chunk_free(c);
@@ -909,7 +1123,7 @@ void* rw_agg_bench_preload(void* h, const RwChunk* c) {
     auto* agg = (HashAgg*)h;
     auto* b = new AggBatch{};
     uint32_t n = c->n_rows;
-    for (int i = 0; i < agg->KW + agg->n_calls; i++) {
+    for (int i = 0; i < agg->n_batch_slots(); i++) {
         if (hipMalloc(&b->col_vals[i], (size_t)n * 8) != hipSuccess) return nullptr;
         if (hipMalloc(&b->col_valid[i], n) != hipSuccess) return nullptr;
     }

@@ -295,3 +295,61 @@ def test_vnode_gpu_matches_oracle():
     got_gpu = compute_vnodes(gpu(), c, [0, 1])
     got_orc = compute_vnodes(ffi.oracle(), c, [0, 1])
     assert got_gpu == got_orc
+
+
+def test_agg_min_retractable_golden():
+    # the reference golden fixture (hash_agg.rs:100-174): retractable min
+    # via materialized-input state — now on the GPU chain-store path
+    calls = [(AGG_COUNT_STAR, -1, T_I64), (AGG_MIN, 1, T_I64)]
+    g = ffi.HashAgg(gpu(), [T_I64, T_I64, T_I64], [0], calls, 0, stream_key=[2])
+    o = ffi.HashAgg(oracle(), [T_I64, T_I64, T_I64], [0], calls, 0, stream_key=[2])
+    e1 = from_pretty(" I I I\n + 1 233 1001\n + 1 23333 1002\n + 2 2333 1003")
+    e2 = from_pretty(" I I I\n - 1 233 1001\n - 1 23333 1002 D\n - 2 2333 1003")
+    run_and_compare(g, o, [[e1], [e2]])
+
+
+def test_agg_minmax_retractable_random():
+    rng = np.random.default_rng(31)
+    calls = [(AGG_COUNT_STAR, -1, T_I64), (AGG_MIN, 1, T_I64), (AGG_MAX, 1, T_I64)]
+    g = ffi.HashAgg(gpu(), [T_I64, T_I64, T_I64], [0], calls, 0, stream_key=[2])
+    o = ffi.HashAgg(oracle(), [T_I64, T_I64, T_I64], [0], calls, 0, stream_key=[2])
+    live = []
+    sk = 0
+    epochs = []
+    for _ in range(4):
+        chunks = []
+        for _ in range(3):
+            n = 512
+            keys = rng.integers(0, 40, n)
+            vals = rng.integers(0, 10000, n)
+            sks = np.zeros(n, np.int64)
+            ops = np.zeros(n, np.uint8)
+            # deletes target rows from EARLIER chunks only (DESIGN §3.1 —
+            # same-chunk conflicts exercised separately below)
+            deletable = list(live)
+            new_rows = []
+            for i in range(n):
+                if deletable and rng.random() < 0.25:
+                    jx = int(rng.integers(0, len(deletable)))
+                    keys[i], vals[i], sks[i] = deletable.pop(jx)
+                    live.remove((int(keys[i]), int(vals[i]), int(sks[i])))
+                    ops[i] = ffi.OP_DELETE
+                else:
+                    sks[i] = sk
+                    sk += 1
+                    new_rows.append((int(keys[i]), int(vals[i]), int(sks[i])))
+            live.extend(new_rows)
+            chunks.append(mk_chunk([T_I64, T_I64, T_I64], ops, [keys, vals, sks]))
+        epochs.append(chunks)
+    run_and_compare(g, o, epochs)
+
+
+def test_agg_min_retractable_same_chunk_conflict():
+    # insert-then-delete of the same row within ONE chunk (the segmented
+    # launch path)
+    calls = [(AGG_COUNT_STAR, -1, T_I64), (AGG_MIN, 1, T_I64)]
+    g = ffi.HashAgg(gpu(), [T_I64, T_I64, T_I64], [0], calls, 0, stream_key=[2])
+    o = ffi.HashAgg(oracle(), [T_I64, T_I64, T_I64], [0], calls, 0, stream_key=[2])
+    e1 = from_pretty(
+        " I I I\n + 1 10 1\n + 1 5 2\n - 1 5 2\n + 2 7 3\n - 1 10 1\n + 1 20 4")
+    run_and_compare(g, o, [[e1]])
