@@ -351,6 +351,12 @@ class ExchangeCtx:
         send_counts = (ctypes.c_uint64 * self.world)()
         recv_counts = (ctypes.c_uint64 * self.world)()
         L.rw_exchange_run.restype = ctypes.c_int
+        L.rw_exchange_run.argtypes = [
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+            ctypes.c_uint32, ctypes.c_int, ctypes.POINTER(ctypes.c_uint32),
+            ctypes.c_int, ctypes.c_uint32, ctypes.c_void_p, ctypes.c_uint64,
+            ctypes.c_void_p, ctypes.c_uint64, ctypes.POINTER(ctypes.c_uint64),
+            ctypes.POINTER(ctypes.c_uint64)]
         rc = L.rw_exchange_run(
             self.h, ctypes.cast(vals, ctypes.c_void_p),
             ctypes.cast(valids, ctypes.c_void_p), ops, nrows, 2, key_cols, 1,

@@ -247,6 +247,15 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
     size_t cap = (size_t)t.cap_mask + 1;
     const uint32_t SLOT_NONE = (uint32_t)-1;
 
+    // per-lane probe memo: with sorted input a head's key rarely changes
+    // between grid-stride iterations — skip the (fabric-bound) table probe
+    // and dirty re-check when it doesn't
+    int64_t memo_kw[KW];
+    for (int i = 0; i < KW; i++) memo_kw[i] = 0;
+    uint32_t memo_null = 0;
+    uint32_t memo_slot = (uint32_t)-1;
+    bool memo_dirtied = false;
+
     for (uint32_t it = 0; it < iters; it++) {
         uint32_t r = r0 + it * stride + blockIdx.x * blockDim.x + threadIdx.x;
         bool active = (r < r1) && !(b.vis && !b.vis[r]);
@@ -276,9 +285,29 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
         if (mode == 2) head = active; // debug: no dedupe, per-lane probe
         uint32_t slot = SLOT_NONE;
         if (head) {
-            slot = table_find_or_insert(t.state, t.keys, t.key_nulls, t.cap_mask,
-                                        kw, nullmask, KW);
-            if (slot == SLOT_NONE) atomicExch(&t.counters[2], 1u); // table full
+            bool memo_hit = memo_slot != SLOT_NONE && nullmask == memo_null;
+            for (int i = 0; i < KW; i++) memo_hit = memo_hit && kw[i] == memo_kw[i];
+            if (memo_hit) {
+                slot = memo_slot;
+            } else {
+                slot = table_find_or_insert(t.state, t.keys, t.key_nulls,
+                                            t.cap_mask, kw, nullmask, KW);
+                if (slot == SLOT_NONE) atomicExch(&t.counters[2], 1u); // full
+                memo_slot = slot;
+                memo_null = nullmask;
+                for (int i = 0; i < KW; i++) memo_kw[i] = kw[i];
+                memo_dirtied = false;
+            }
+            // dirty tracking moved to the head (it owns the memo); exactly-
+            // once overall is still the CAS's job
+            if (slot != SLOT_NONE && !memo_dirtied) {
+                if (ld_u32(&t.dirty_flag[slot]) == 0 &&
+                    atomicCAS(&t.dirty_flag[slot], 0u, 1u) == 0u) {
+                    uint32_t i = atomicAdd(&t.counters[0], 1u);
+                    t.dirty_list[i] = slot;
+                }
+                memo_dirtied = true;
+            }
         }
         uint64_t heads_b = __ballot(head);
         uint64_t le_mask = heads_b & (~0ULL >> (63 - lane));
@@ -468,13 +497,6 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
                         }
                         break;
                 }
-            }
-            // dirty tracking (hash_agg.rs group_change_set), read-first to
-            // keep the hot line shared
-            if (ld_u32(&t.dirty_flag[slot]) == 0 &&
-                atomicCAS(&t.dirty_flag[slot], 0u, 1u) == 0u) {
-                uint32_t i = atomicAdd(&t.counters[0], 1u);
-                t.dirty_list[i] = slot;
             }
         }
     }
