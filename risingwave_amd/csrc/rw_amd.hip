@@ -1568,6 +1568,28 @@ struct JoinOutDev {
     uint32_t cap;
 };
 
+// Find-only with PLAIN (cached) loads — valid only for a side that is not
+// mutated during this launch: cross-launch visibility comes from the kernel
+// boundary, and cached loads fetch whole lines once instead of per-word
+// fabric transactions.
+__device__ __forceinline__ uint32_t table_find_cached(
+    const uint32_t* state, const int64_t* keys, const uint32_t* key_nulls,
+    uint32_t cap_mask, const int64_t* kw, uint32_t nullmask, int KW) {
+    uint32_t slot = (uint32_t)(hash_key(kw, nullmask, KW) & cap_mask);
+    for (uint32_t probes = 0; probes <= cap_mask; probes++) {
+        uint32_t st = state[slot];
+        if (st == SLOT_EMPTY) return (uint32_t)-1;
+        if (st == SLOT_READY) {
+            bool eq = key_nulls[slot] == nullmask;
+            for (int i = 0; eq && i < KW; i++)
+                eq = keys[(size_t)slot * KW + i] == kw[i];
+            if (eq) return slot;
+        }
+        slot = (slot + 1) & cap_mask;
+    }
+    return (uint32_t)-1;
+}
+
 __device__ __forceinline__ bool join_cond_ok(const JoinMeta& m, int probe_side,
                                              const JoinBatchDev& b, uint32_t r,
                                              const JoinSideDev& match,
@@ -1634,13 +1656,19 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
         uint32_t my_n = 0;
         uint32_t matched_row = UINT32_MAX;
         if (active) {
-            mslot = table_find(match.state, match.keys, match.key_nulls,
-                               match.cap_mask, kw, nullmask, m.KW);
-            if (mslot != UINT32_MAX) {
-                uint32_t row = ld_u32(&match.head[mslot]);
+            if (m.append_only) {
+                // append-only mutates the match side in-launch: sc1 path
+                mslot = table_find(match.state, match.keys, match.key_nulls,
+                                   match.cap_mask, kw, nullmask, m.KW);
+            } else {
+                mslot = table_find_cached(match.state, match.keys,
+                                          match.key_nulls, match.cap_mask, kw,
+                                          nullmask, m.KW);
+            }
+            if (mslot != UINT32_MAX && !m.append_only) {
+                uint32_t row = match.head[mslot];
                 while (row != UINT32_MAX) {
-                    if (ld_u32(&match.alive[row]) &&
-                        join_cond_ok(m, S, b, r, match, row)) {
+                    if (match.alive[row] && join_cond_ok(m, S, b, r, match, row)) {
                         my_n++;
                         matched_row = row;
                     }
@@ -1667,10 +1695,10 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
             } else if (my_n) {
                 // second walk: emit (JoinStreamChunkBuilder::append_row,
                 // join/builder.rs:87-106)
-                uint32_t row = ld_u32(&match.head[mslot]);
+                uint32_t row = match.head[mslot];
                 uint32_t k = 0;
                 while (row != UINT32_MAX && k < my_n) {
-                    if (ld_u32(&match.alive[row]) &&
+                    if (match.alive[row] &&
                         join_cond_ok(m, S, b, r, match, row)) {
                         uint32_t orow = my_base + k;
                         out.ops[orow] = op;
@@ -1694,7 +1722,8 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
                     row = match.next[row];
                 }
             }
-        } else if (active && mslot != UINT32_MAX) {
+        }
+        if (m.append_only && active && mslot != UINT32_MAX) {
             // append-only path: <=1 match (jk superset of pk); per-match
             // cursor atomics, single walk with kill
             uint32_t row = ld_u32(&match.head[mslot]);
