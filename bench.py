@@ -165,6 +165,12 @@ def main():
     L.rw_agg_sync.argtypes = [ctypes.c_void_p]
     L.rw_agg_kernel_stats.argtypes = [ctypes.c_void_p, ctypes.POINTER(KernelStats)]
     L.rw_agg_stats_reset.argtypes = [ctypes.c_void_p]
+    L.rw_agg_apply_payload.restype = ctypes.c_int
+    L.rw_agg_apply_payload.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                       ctypes.POINTER(ctypes.c_uint64),
+                                       ctypes.c_int, ctypes.c_int]
+    L.rw_agg_n_batch_slots.restype = ctypes.c_int
+    L.rw_agg_n_batch_slots.argtypes = [ctypes.c_void_p]
 
     rng = np.random.default_rng(args.seed + rank)
 
@@ -220,9 +226,11 @@ def main():
 
     def step(i):
         if use_exchange:
-            recv_blocks = exch.run(agg.h, batches[i % n_batches], xb)
+            nslots = L.rw_agg_n_batch_slots(agg.h)
+            recv_blocks = exch.run(agg.h, batches[i % n_batches], xb,
+                                   n_cols=nslots)
             rc = L.rw_agg_apply_payload(
-                agg.h, xb.recv, recv_blocks, world, 2)
+                agg.h, ctypes.c_void_p(xb.recv), recv_blocks, world, nslots)
             assert rc == 0, gpu_lib.last_error()
         else:
             rc = L.rw_agg_bench_apply(agg.h, batches[i % n_batches])
@@ -330,7 +338,7 @@ class ExchangeCtx:
         assert b.send and b.recv
         return b
 
-    def run(self, agg_h, batch, xb, gpu_lib=None):
+    def run(self, agg_h, batch, xb, n_cols=2):
         L = self.lib
         import risingwave_amd
 
@@ -359,8 +367,8 @@ class ExchangeCtx:
             ctypes.POINTER(ctypes.c_uint64)]
         rc = L.rw_exchange_run(
             self.h, ctypes.cast(vals, ctypes.c_void_p),
-            ctypes.cast(valids, ctypes.c_void_p), ops, nrows, 2, key_cols, 1,
-            256, ctypes.c_void_p(xb.send), ctypes.c_uint64(xb.cap),
+            ctypes.cast(valids, ctypes.c_void_p), ops, nrows, n_cols, key_cols,
+            1, 256, ctypes.c_void_p(xb.send), ctypes.c_uint64(xb.cap),
             ctypes.c_void_p(xb.recv), ctypes.c_uint64(xb.cap), send_counts,
             recv_counts)
         if rc != 0:

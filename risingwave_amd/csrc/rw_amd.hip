@@ -1186,6 +1186,8 @@ typedef struct {
     uint64_t rows;
 } RwKernelStats;
 
+int rw_agg_n_batch_slots(void* h) { return ((HashAgg*)h)->n_batch_slots(); }
+
 // expose a preloaded batch's device pointers (for the exchange path)
 int rw_agg_batch_ptrs(void* batch, const int64_t** vals, const uint8_t** valids,
                       const uint8_t** ops, uint32_t* n_rows) {

@@ -41,11 +41,15 @@ def main():
     print("3: buffers")
     xb = exch.make_buffers(n * 32 * 4)
     print("4: run exchange")
-    recv_counts = exch.run(agg.h, batch, xb)
+    L.rw_agg_n_batch_slots.restype = ctypes.c_int
+    L.rw_agg_n_batch_slots.argtypes = [ctypes.c_void_p]
+    nslots = L.rw_agg_n_batch_slots(agg.h)
+    recv_counts = exch.run(agg.h, batch, xb, n_cols=nslots)
     print("   recv:", list(recv_counts))
     assert sum(recv_counts) == n
     print("5: apply payload")
-    rc = L.rw_agg_apply_payload(agg.h, ctypes.c_void_p(xb.recv), recv_counts, 1, 2)
+    rc = L.rw_agg_apply_payload(agg.h, ctypes.c_void_p(xb.recv), recv_counts, 1,
+                                nslots)
     assert rc == 0, gpu.last_error()
     print("6: flush + compare vs oracle")
     agg.flush(1)
