@@ -284,6 +284,14 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
         bool head = active && !(same && prev_active);
         if (mode == 2) head = active; // debug: no dedupe, per-lane probe
         uint32_t slot = SLOT_NONE;
+        if (mode == 4) { // debug: loads only (no probe/scan/atomics)
+            long long acc4 = nullmask;
+            for (int i = 0; i < KW; i++) acc4 += kw[i];
+            for (int ci = 0; ci < n_calls; ci++)
+                if (active && calls[ci].arg >= 0) acc4 += b.col_vals[KW + ci][r];
+            if (acc4 == 0x7fffffffffffffffLL) t.counters[2] = 9;
+            continue;
+        }
         if (head) {
             bool memo_hit = memo_slot != SLOT_NONE && nullmask == memo_null;
             for (int i = 0; i < KW; i++) memo_hit = memo_hit && kw[i] == memo_kw[i];
@@ -467,6 +475,10 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
         uint64_t cont_b = __ballot(contributing);
         uint64_t ends_after = (heads_b | ~cont_b) >> 1 | (1ULL << 63);
         bool tail = contributing && ((ends_after >> lane) & 1);
+        if (mode == 3) { // debug: skip the tail atomics
+            if (tail && v[0] == 0x7fffffffffffffffLL) t.counters[2] = 9;
+            continue;
+        }
         if (tail) {
             for (int ci = 0; ci < n_calls; ci++) {
                 if (calls[ci].minput) continue;
