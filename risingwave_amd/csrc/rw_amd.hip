@@ -180,14 +180,18 @@ struct AggCallDev {
     int8_t mord;    // ordinal among minput calls
 };
 
-// Device-side batch of input rows (SoA), i64-widened values.
+// Device-side batch of input rows (SoA), i64-widened values. `stride` (in
+// elements) lets the batch view row-major buffers — e.g. the join's output
+// block, where column c of row r sits at vals[r*n_out + c] — without a
+// repack; `valid_inverted` views a null-flag array as validity.
 struct AggBatch {
     int64_t* col_vals[MAX_KW + MAX_CALLS]; // group key cols then arg cols
     uint8_t* col_valid[MAX_KW + MAX_CALLS];
     uint8_t* ops;
     uint8_t* vis; // may be null
     uint32_t n_rows;
-    uint32_t capacity;
+    uint32_t stride = 1;
+    uint8_t valid_inverted = 0;
 };
 
 struct AggTableDev {
@@ -262,10 +266,11 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
         int64_t kw[MAX_KW];
         uint32_t nullmask = 0;
         for (int i = 0; i < KW; i++) kw[i] = 0;
+        size_t rs = (size_t)r * b.stride;
         if (active) {
             for (int i = 0; i < KW; i++) {
-                bool valid = b.col_valid[i][r];
-                kw[i] = valid ? b.col_vals[i][r] : 0;
+                bool valid = b.col_valid[i][rs] ^ b.valid_inverted;
+                kw[i] = valid ? b.col_vals[i][rs] : 0;
                 nullmask |= (!valid) << i;
             }
         }
@@ -288,7 +293,7 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
             long long acc4 = nullmask;
             for (int i = 0; i < KW; i++) acc4 += kw[i];
             for (int ci = 0; ci < n_calls; ci++)
-                if (active && calls[ci].arg >= 0) acc4 += b.col_vals[KW + ci][r];
+                if (active && calls[ci].arg >= 0) acc4 += b.col_vals[KW + ci][rs];
             if (acc4 == 0x7fffffffffffffffLL) t.counters[2] = 9;
             continue;
         }
@@ -337,15 +342,15 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
         for (int ci = 0; ci < n_calls; ci++) {
             const AggCallDev& c = calls[ci];
             int col = KW + ci;
-            bool arg_valid = contributing && b.col_valid[col][r];
+            bool arg_valid = contributing && (b.col_valid[col][rs] ^ b.valid_inverted);
             if (c.minput) {
                 // materialized-input ops are exact per-row chain mutations
                 // (minput.rs apply_batch): no wave aggregation
                 v[ci] = 0;
                 if (contributing) {
                     uint32_t* headp = t.mheads + (size_t)c.mord * cap + slot;
-                    bool valid = b.col_valid[col][r];
-                    long long val = valid ? b.col_vals[col][r] : 0;
+                    bool valid = b.col_valid[col][rs] ^ b.valid_inverted;
+                    long long val = valid ? b.col_vals[col][rs] : 0;
                     if (sign > 0) {
                         uint32_t row = atomicAdd(t.mcursor, 1u);
                         if (row >= t.mrow_cap) {
@@ -355,10 +360,10 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
                             t.mval_null[row] = !valid;
                             for (int k = 0; k < t.n_sk; k++) {
                                 int skc = KW + n_calls + k;
+                                bool skv = b.col_valid[skc][rs] ^ b.valid_inverted;
                                 st_i64((int64_t*)&t.msk[(size_t)k * t.mrow_cap + row],
-                                       b.col_valid[skc][r] ? b.col_vals[skc][r] : 0);
-                                t.msk_null[(size_t)k * t.mrow_cap + row] =
-                                    !b.col_valid[skc][r];
+                                       skv ? b.col_vals[skc][rs] : 0);
+                                t.msk_null[(size_t)k * t.mrow_cap + row] = !skv;
                             }
                             st_u32(&t.malive[row], 1);
                             uint32_t old_head = ld_u32(headp);
@@ -379,10 +384,10 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
                                           (!valid || ld_i64((const int64_t*)&t.mval[row]) == val);
                                 for (int k = 0; eq && k < t.n_sk; k++) {
                                     int skc = KW + n_calls + k;
-                                    uint8_t va = b.col_valid[skc][r];
+                                    uint8_t va = b.col_valid[skc][rs] ^ b.valid_inverted;
                                     uint8_t vb = !t.msk_null[(size_t)k * t.mrow_cap + row];
                                     eq = (va == vb) &&
-                                         (!va || b.col_vals[skc][r] ==
+                                         (!va || b.col_vals[skc][rs] ==
                                                      ld_i64((const int64_t*)&t.msk[(size_t)k * t.mrow_cap + row]));
                                 }
                                 if (eq && atomicCAS(&t.malive[row], 1u, 0u) == 1u)
@@ -399,15 +404,15 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
                 case RW_AGG_COUNT: v[ci] = arg_valid ? sign : 0; break;
                 case RW_AGG_SUM:
                 case RW_AGG_SUM0:
-                    v[ci] = arg_valid ? sign * b.col_vals[col][r] : 0;
+                    v[ci] = arg_valid ? sign * b.col_vals[col][rs] : 0;
                     if (arg_valid) hasmask |= 1 << ci;
                     break;
                 case RW_AGG_MIN:
-                    v[ci] = arg_valid ? b.col_vals[col][r] : INT64_MAX;
+                    v[ci] = arg_valid ? b.col_vals[col][rs] : INT64_MAX;
                     if (arg_valid) hasmask |= 1 << ci;
                     break;
                 case RW_AGG_MAX:
-                    v[ci] = arg_valid ? b.col_vals[col][r] : INT64_MIN;
+                    v[ci] = arg_valid ? b.col_vals[col][rs] : INT64_MIN;
                     if (arg_valid) hasmask |= 1 << ci;
                     break;
             }
@@ -1847,6 +1852,7 @@ struct HashJoin {
     JoinBatchDev stage[2]{};
     uint32_t stage_cap[2] = {0, 0};
     std::vector<RwChunk*> outq;
+    uint8_t* zeros = nullptr; // all-zero null flags for pipeline dummy cols
     double probe_ms_total = 0;
     uint64_t probe_launches = 0, probe_rows = 0;
 
@@ -2164,6 +2170,7 @@ struct HashJoin {
             hipFree(out.ops);
             hipFree(out.counters);
         }
+        if (zeros) hipFree(zeros);
         if (stream) hipStreamDestroy(stream);
         for (auto* c : outq) rw_chunk_free(c);
     }
@@ -2229,6 +2236,67 @@ long long rw_join_bench_drain(void* h) {
     }
     hipMemset(j->out.counters, 0, 8);
     return (long long)ctr[0];
+}
+
+// Device-side pipeline hop: apply the join's accumulated output buffer
+// (row-major [row][n_out]) directly as agg input via a strided AggBatch
+// view — the fragment edge agg(join(...)) of the TPC-H q3 MV without
+// leaving HBM. The agg must have been created with the join's output
+// schema as its input schema. Resets the join's output cursor.
+int rw_agg_apply_joinout(void* agg_h, void* join_h) {
+    auto* agg = (HashAgg*)agg_h;
+    auto* j = (HashJoin*)join_h;
+    if (hipStreamSynchronize(j->stream) != hipSuccess)
+        FAIL(RW_E_INTERNAL, "join sync failed");
+    uint32_t ctr[2];
+    HIP_TRY(hipMemcpy(ctr, j->out.counters, 8, hipMemcpyDeviceToHost));
+    if (ctr[1] != 0) FAIL(RW_E_INTERNAL, "join overflow %u before pipeline hop", ctr[1]);
+    uint32_t n = ctr[0];
+    if (!n) return RW_OK;
+    if (!j->zeros) {
+        HIP_TRY(hipMalloc(&j->zeros, (size_t)j->out.cap * j->m.n_out));
+        HIP_TRY(hipMemset(j->zeros, 0, (size_t)j->out.cap * j->m.n_out));
+    }
+    AggBatch b{};
+    b.stride = (uint32_t)j->m.n_out;
+    b.valid_inverted = 1;
+    auto bind = [&](int slot, int src_col) {
+        b.col_vals[slot] = j->out.vals + src_col;
+        b.col_valid[slot] = j->out.nulls + src_col;
+    };
+    for (int i = 0; i < agg->KW; i++) {
+        if ((int)agg->group_key[i] >= j->m.n_out)
+            FAIL(RW_E_INVAL, "agg group key %u outside join output", agg->group_key[i]);
+        bind(i, (int)agg->group_key[i]);
+    }
+    for (int ci = 0; ci < agg->n_calls; ci++) {
+        if (agg->calls[ci].arg >= 0) {
+            if (agg->calls[ci].arg >= j->m.n_out)
+                FAIL(RW_E_INVAL, "agg arg outside join output");
+            bind(agg->KW + ci, agg->calls[ci].arg);
+        } else {
+            b.col_vals[agg->KW + ci] = j->out.vals; // unread
+            b.col_valid[agg->KW + ci] = j->zeros;   // inverted -> all valid
+        }
+    }
+    if (agg->n_minput) {
+        for (size_t k = 0; k < agg->stream_key.size(); k++) {
+            if ((int)agg->stream_key[k] >= j->m.n_out)
+                FAIL(RW_E_INVAL, "agg stream key outside join output");
+            bind(agg->KW + agg->n_calls + (int)k, (int)agg->stream_key[k]);
+        }
+    }
+    b.ops = j->out.ops;
+    b.vis = nullptr;
+    b.n_rows = n;
+    // run the agg on ITS stream after the join's work is drained (synced
+    // above), then reset the join cursor
+    int rc = agg->apply(b, true);
+    if (rc != RW_OK) return rc;
+    if (hipStreamSynchronize(agg->stream) != hipSuccess)
+        FAIL(RW_E_INTERNAL, "agg sync failed");
+    HIP_TRY(hipMemset(j->out.counters, 0, 8));
+    return agg->check_overflow();
 }
 
 int rw_join_kernel_stats(void* h, RwKernelStats* out) {

@@ -121,6 +121,12 @@ struct Exchange {
     int rank = 0, n_ranks = 1;
     double exch_ms = 0;
     uint64_t exch_launches = 0;
+    // persistent scratch (allocated on first run, grown as needed)
+    uint32_t* d_dest = nullptr;
+    uint32_t dest_cap = 0;
+    unsigned long long *d_counts = nullptr, *d_offsets = nullptr,
+                       *d_cursors = nullptr, *d_count_mat = nullptr;
+    hipEvent_t e0 = nullptr, e1 = nullptr;
 };
 
 extern "C" {
@@ -166,6 +172,15 @@ void rw_exchange_destroy(void* h) {
     auto* x = (Exchange*)h;
     if (!x) return;
     if (x->comm) ncclCommDestroy(x->comm);
+    if (x->d_dest) hipFree(x->d_dest);
+    if (x->d_counts) {
+        hipFree(x->d_counts);
+        hipFree(x->d_offsets);
+        hipFree(x->d_cursors);
+        hipFree(x->d_count_mat);
+        hipEventDestroy(x->e0);
+        hipEventDestroy(x->e1);
+    }
     if (x->stream) hipStreamDestroy(x->stream);
     delete x;
 }
@@ -193,12 +208,22 @@ int rw_exchange_run(void* h, const int64_t* const* col_vals,
     b.ops = ops;
     b.n_rows = n_rows;
 
-    uint32_t* d_dest;
-    unsigned long long *d_counts, *d_offsets, *d_cursors;
-    XHIP(hipMalloc(&d_dest, (size_t)n_rows * 4));
-    XHIP(hipMalloc(&d_counts, R * 8));
-    XHIP(hipMalloc(&d_offsets, R * 8));
-    XHIP(hipMalloc(&d_cursors, R * 8));
+    if (x->dest_cap < n_rows) {
+        if (x->d_dest) hipFree(x->d_dest);
+        XHIP(hipMalloc(&x->d_dest, (size_t)n_rows * 4));
+        x->dest_cap = n_rows;
+    }
+    if (!x->d_counts) {
+        XHIP(hipMalloc(&x->d_counts, R * 8));
+        XHIP(hipMalloc(&x->d_offsets, R * 8));
+        XHIP(hipMalloc(&x->d_cursors, R * 8));
+        XHIP(hipMalloc(&x->d_count_mat, R * 8));
+        XHIP(hipEventCreate(&x->e0));
+        XHIP(hipEventCreate(&x->e1));
+    }
+    uint32_t* d_dest = x->d_dest;
+    unsigned long long *d_counts = x->d_counts, *d_offsets = x->d_offsets,
+                       *d_cursors = x->d_cursors;
     XHIP(hipMemsetAsync(d_counts, 0, R * 8, x->stream));
     XHIP(hipMemsetAsync(d_cursors, 0, R * 8, x->stream));
 
@@ -208,9 +233,7 @@ int rw_exchange_run(void* h, const int64_t* const* col_vals,
     uint32_t k[4] = {0, 0, 0, 0};
     for (int i = 0; i < n_keys; i++) k[i] = key_cols[i];
 
-    hipEvent_t e0, e1;
-    XHIP(hipEventCreate(&e0));
-    XHIP(hipEventCreate(&e1));
+    hipEvent_t e0 = x->e0, e1 = x->e1;
     XHIP(hipEventRecord(e0, x->stream));
 
     x_count_kernel<<<blocks, 256, 0, x->stream>>>(b, n_keys, k[0], k[1], k[2],
@@ -238,8 +261,7 @@ int rw_exchange_run(void* h, const int64_t* const* col_vals,
 
     // exchange per-peer row counts (all-to-all of one u64 per peer), then
     // the payload blocks (all-to-all-v), all on the exchange stream
-    unsigned long long* d_count_mat; // recv counts
-    XHIP(hipMalloc(&d_count_mat, R * 8));
+    unsigned long long* d_count_mat = x->d_count_mat;
     XNCCL(ncclGroupStart());
     for (int p = 0; p < R; p++) {
         XNCCL(ncclSend(d_counts + p, 1, ncclUint64, p, x->comm, x->stream));
@@ -275,18 +297,11 @@ int rw_exchange_run(void* h, const int64_t* const* col_vals,
     hipEventElapsedTime(&ms, e0, e1);
     x->exch_ms += ms;
     x->exch_launches++;
-    hipEventDestroy(e0);
-    hipEventDestroy(e1);
 
     for (int p = 0; p < R; p++) {
         send_counts_out[p] = counts[p];
         recv_counts_out[p] = recv_counts[p];
     }
-    hipFree(d_dest);
-    hipFree(d_counts);
-    hipFree(d_offsets);
-    hipFree(d_cursors);
-    hipFree(d_count_mat);
     return 0;
 }
 
