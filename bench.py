@@ -125,10 +125,13 @@ def main():
     ap.add_argument("--windows-per-epoch", type=int, default=64)
     ap.add_argument("--seed", type=int, default=1)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
-    ap.add_argument("--workload", choices=["q7", "q8"], default="q7",
+    ap.add_argument("--workload", choices=["q7", "q8", "q3"], default="q7",
                     help="q7 = windowed hash-agg (BASELINE configs[1], the "
                          "default the driver measures); q8 = stream-stream "
-                         "hash-join, 10M-key build side (configs[2])")
+                         "hash-join, 10M-key build side (configs[2]); q3 = "
+                         "TPC-H-stream join+agg pipeline with insert/delete "
+                         "mix (configs[4], minus the Hummock spill)")
+    ap.add_argument("--q3-orders", type=int, default=100_000_000)
     ap.add_argument("--exchange", choices=["auto", "on", "off"], default="auto",
                     help="vnode partition + RCCL all-to-all-v before the agg "
                          "(the reference's HashDataDispatcher hop, SURVEY "
@@ -180,6 +183,11 @@ def main():
 
     if args.workload == "q8":
         bench_q8(args, ffi, gpu_lib, rng, rank, world, dist)
+        if dist:
+            dist.destroy_process_group()
+        return
+    if args.workload == "q3":
+        bench_q3(args, ffi, gpu_lib, rng, rank, world, dist)
         if dist:
             dist.destroy_process_group()
         return
@@ -547,6 +555,172 @@ def bench_q8(args, ffi, gpu_lib, rng, rank, world, dist):
             result["cpu_baseline"] = cpu_baseline_q8(ffi, np.random.default_rng(5))
         print(json.dumps(result))
     j.close()
+
+
+def bench_q3(args, ffi, gpu_lib, rng, rank, world, dist):
+    """TPC-H-stream q3 (BASELINE configs[4], minus the Hummock spill):
+    orders ⋈ lineitem inner join feeding sum(revenue)/count group by
+    (orderkey, orderdate, shippriority), 100M-key order state in HBM,
+    ~10% Delete mix on the lineitem stream, checkpoint flush every
+    --barrier-every steps. The join's output block feeds the agg in HBM
+    (rw_agg_apply_joinout). value = lineitem input rows/s."""
+    import ctypes
+
+    from rwtest.ffi import AGG_COUNT_STAR, AGG_SUM, JOIN_INNER, SIDE_LEFT, \
+        SIDE_RIGHT, T_I64
+
+    L = gpu_lib.lib
+    L.rw_join_bench_preload.restype = ctypes.c_void_p
+    L.rw_join_bench_preload.argtypes = [ctypes.c_void_p, ctypes.c_int,
+                                        ctypes.POINTER(ffi.RwChunkC)]
+    L.rw_join_bench_apply.restype = ctypes.c_int
+    L.rw_join_bench_apply.argtypes = [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p]
+    L.rw_join_bench_drain.restype = ctypes.c_longlong
+    L.rw_join_bench_drain.argtypes = [ctypes.c_void_p]
+    L.rw_join_kernel_stats.argtypes = [ctypes.c_void_p, ctypes.POINTER(KernelStats)]
+    L.rw_join_stats_reset.argtypes = [ctypes.c_void_p]
+    L.rw_agg_apply_joinout.restype = ctypes.c_int
+    L.rw_agg_apply_joinout.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
+
+    N_ORDERS = args.q3_orders
+    batch_rows = CHUNK_ROWS * CHUNKS_PER_BATCH
+    # lineitem: [l_orderkey, revenue, l_rowid]; orders: [o_orderkey, o_date,
+    # o_prio, o_rowid]; join on orderkey; output = all 7 columns
+    tl = [T_I64, T_I64, T_I64]
+    tr = [T_I64, T_I64, T_I64, T_I64]
+    j = ffi.HashJoin(gpu_lib, JOIN_INNER, tl, tr, key_l=[0], key_r=[0],
+                     pk_l=[2], pk_r=[3],
+                     state_capacity_hint=max(N_ORDERS, 1 << 22),
+                     row_capacity_hint=N_ORDERS +
+                     (args.steps + args.warmup + 4) * batch_rows + 1_000_000)
+    # agg over the join output: group (o_orderkey, o_date, o_prio) = cols
+    # 3,4,5 of the concat row; sum(revenue)=col 1; count(*)
+    agg = ffi.HashAgg(gpu_lib, [T_I64] * 7, [3, 4, 5],
+                      [(AGG_SUM, 1, T_I64), (AGG_COUNT_STAR, -1, T_I64)],
+                      row_count_index=1, state_capacity_hint=1 << 23)
+
+    ones = lambda n: np.ones(n, np.uint8)
+
+    def preload(side, cols, ops=None):
+        n = len(cols[0])
+        c = ffi.Chunk(tl if side == SIDE_LEFT else tr,
+                      np.zeros(n, np.uint8) if ops is None else ops, cols,
+                      [ones(n)] * len(cols))
+        cc = c.to_c()
+        h = L.rw_join_bench_preload(j.h, side, ctypes.byref(cc))
+        assert h, gpu_lib.last_error()
+        return h
+
+    # build orders (untimed)
+    for lo in range(0, N_ORDERS, batch_rows):
+        hi = min(lo + batch_rows, N_ORDERS)
+        ok = np.arange(lo, hi, dtype=np.int64)
+        h = preload(SIDE_RIGHT, [ok, ok % 2557, ok % 3, ok])
+        rc = L.rw_join_bench_apply(j.h, SIDE_RIGHT, h)
+        assert rc == 0, gpu_lib.last_error()
+        assert L.rw_join_bench_drain(j.h) >= 0, gpu_lib.last_error()
+
+    # lineitem batches: ~10% deletes of rows inserted by the previous batch,
+    # orderkeys in a sliding 512k-order window (bounds dirty groups/epoch)
+    n_batches = 8
+    WINDOW = 1 << 19
+    batches = []
+    rowid = 0
+    prev_insert = None
+    for b in range(n_batches):
+        n = batch_rows
+        base = (b * WINDOW) % max(N_ORDERS - WINDOW, 1)
+        ok = base + rng.integers(0, WINDOW, n)
+        rev = rng.integers(1, 100_000, n)
+        rid = np.arange(rowid, rowid + n)
+        rowid += n
+        ops = np.zeros(n, np.uint8)
+        if prev_insert is not None:
+            n_del = n // 10
+            sel = rng.choice(len(prev_insert[0]), n_del, replace=False)
+            ok[:n_del] = prev_insert[0][sel]
+            rev[:n_del] = prev_insert[1][sel]
+            rid[:n_del] = prev_insert[2][sel]
+            ops[:n_del] = ffi.OP_DELETE
+        prev_insert = (ok[n // 10:].copy(), rev[n // 10:].copy(),
+                       rid[n // 10:].copy())
+        batches.append(preload(SIDE_LEFT, [ok, rev, rid], ops))
+
+    def step(i):
+        rc = L.rw_join_bench_apply(j.h, SIDE_LEFT, batches[i % n_batches])
+        assert rc == 0, gpu_lib.last_error()
+        rc = L.rw_agg_apply_joinout(agg.h, j.h)
+        assert rc == 0, gpu_lib.last_error()
+        if (i + 1) % args.barrier_every == 0:
+            agg.flush(i)
+            agg.poll_all()
+
+    for i in range(args.warmup):
+        step(i)
+    L.rw_join_stats_reset(j.h)
+    L.rw_agg_stats_reset(agg.h)
+    if dist:
+        dist.barrier()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(i)
+    L.rw_agg_sync(agg.h)
+    elapsed = time.perf_counter() - t0
+    if dist:
+        import torch
+
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    jks = KernelStats()
+    aks = KernelStats()
+    L.rw_join_kernel_stats(j.h, ctypes.byref(jks))
+    L.rw_agg_kernel_stats(agg.h, ctypes.byref(aks))
+    if rank == 0:
+        total_rows = args.steps * batch_rows * world
+        avg_probe_ms = jks.total_ms / max(jks.launches, 1)
+        achieved = (128 * batch_rows) / (avg_probe_ms * 1e-3) / 1e9
+        result = {
+            "metric": "input rows/sec/GPU on TPC-H-stream q3",
+            "value": total_rows / elapsed,
+            "unit": "rows/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed * 1000.0 / args.steps,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "int64",
+            "data": "synthetic",
+            "config": {
+                "workload": "tpch_q3_stream",
+                "orders": N_ORDERS,
+                "lineitem_rows_per_step": batch_rows,
+                "delete_mix": 0.1,
+                "barrier_every_steps": args.barrier_every,
+                "pipeline": "join(orders ⋈ lineitem) → agg(sum,count by "
+                            "orderkey,date,prio) in HBM",
+                "parallelism": f"dp{world}",
+            },
+            "roofline": {
+                "bound": "hbm",
+                "achieved": achieved,
+                "peak": HBM_PEAK_GBS,
+                "unit": "GB/s",
+                "frac": achieved / HBM_PEAK_GBS,
+                "traffic": None,
+            },
+            "kernels": {
+                "join_probe_avg_ms": avg_probe_ms,
+                "agg_apply_avg_ms": aks.total_ms / max(aks.launches, 1),
+            },
+            "cpu_baseline": None,
+        }
+        print(json.dumps(result))
+    j.close()
+    agg.close()
 
 
 def cpu_baseline_q8(ffi, rng, target_seconds=8.0):

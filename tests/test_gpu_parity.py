@@ -353,3 +353,91 @@ def test_agg_min_retractable_same_chunk_conflict():
     e1 = from_pretty(
         " I I I\n + 1 10 1\n + 1 5 2\n - 1 5 2\n + 2 7 3\n - 1 10 1\n + 1 20 4")
     run_and_compare(g, o, [[e1]])
+
+
+def test_pipeline_join_agg_q3_shape():
+    # the q3 fragment edge: inner join feeding sum/count agg — the GPU runs
+    # the hop device-side (bench path: rw_join_bench_apply keeps outputs in
+    # HBM, rw_agg_apply_joinout consumes them); the oracle pushes the join's
+    # output chunks into its agg. Flush outputs must agree per epoch.
+    import ctypes
+
+    from rwtest.ffi import AGG_COUNT_STAR, AGG_SUM
+
+    rng = np.random.default_rng(41)
+    tl = [T_I64, T_I64, T_I64]
+    tr = [T_I64, T_I64, T_I64, T_I64]
+
+    gl = gpu()
+    L = gl.lib
+    L.rw_join_bench_preload.restype = ctypes.c_void_p
+    L.rw_join_bench_preload.argtypes = [ctypes.c_void_p, ctypes.c_int,
+                                        ctypes.POINTER(ffi.RwChunkC)]
+    L.rw_join_bench_apply.restype = ctypes.c_int
+    L.rw_join_bench_apply.argtypes = [ctypes.c_void_p, ctypes.c_int,
+                                      ctypes.c_void_p]
+    L.rw_join_bench_drain.restype = ctypes.c_longlong
+    L.rw_join_bench_drain.argtypes = [ctypes.c_void_p]
+    L.rw_agg_apply_joinout.restype = ctypes.c_int
+    L.rw_agg_apply_joinout.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
+
+    gj = ffi.HashJoin(gl, JOIN_INNER, tl, tr, key_l=[0], key_r=[0],
+                      pk_l=[2], pk_r=[3])
+    ga = ffi.HashAgg(gl, [T_I64] * 7, [3, 4, 5],
+                     [(AGG_SUM, 1, T_I64), (AGG_COUNT_STAR, -1, T_I64)],
+                     row_count_index=1)
+    oj = ffi.HashJoin(ffi.oracle(), JOIN_INNER, tl, tr, key_l=[0], key_r=[0],
+                      pk_l=[2], pk_r=[3])
+    oa = ffi.HashAgg(ffi.oracle(), [T_I64] * 7, [3, 4, 5],
+                     [(AGG_SUM, 1, T_I64), (AGG_COUNT_STAR, -1, T_I64)],
+                     row_count_index=1)
+
+    def gpu_apply(side, chunk):
+        cc = chunk.to_c()
+        h = L.rw_join_bench_preload(gj.h, side, ctypes.byref(cc))
+        assert h, gl.last_error()
+        assert L.rw_join_bench_apply(gj.h, side, h) == 0, gl.last_error()
+
+    # orders build side
+    n_orders = 2000
+    ok = np.arange(n_orders, dtype=np.int64)
+    orders = mk_chunk(tr, np.zeros(n_orders, np.uint8), [ok, ok % 31, ok % 3, ok])
+    gpu_apply(SIDE_RIGHT, orders)
+    assert L.rw_join_bench_drain(gj.h) == 0, gl.last_error()
+    oj.push(SIDE_RIGHT, orders)
+    assert oj.poll_all() == []
+
+    rowid = 0
+    prev = None
+    for epoch in range(3):
+        for _ in range(2):
+            n = 1024
+            okey = rng.integers(0, n_orders, n)
+            rev = rng.integers(1, 1000, n)
+            rid = np.arange(rowid, rowid + n)
+            rowid += n
+            ops = np.zeros(n, np.uint8)
+            if prev is not None:
+                nd = n // 8
+                sel = rng.choice(len(prev[0]), nd, replace=False)
+                okey[:nd], rev[:nd], rid[:nd] = (prev[0][sel], prev[1][sel],
+                                                 prev[2][sel])
+                ops[:nd] = ffi.OP_DELETE
+            prev = (okey[n // 8:].copy(), rev[n // 8:].copy(),
+                    rid[n // 8:].copy())
+            li = mk_chunk(tl, ops, [okey, rev, rid])
+            gpu_apply(SIDE_LEFT, li)
+            assert L.rw_agg_apply_joinout(ga.h, gj.h) == 0, gl.last_error()
+            oj.push(SIDE_LEFT, li)
+            for oc in oj.poll_all():
+                oa.push(oc)
+        ga.flush(epoch + 1)
+        oa.flush(epoch + 1)
+        mg = rows_multiset(ga.poll_all())
+        mo = rows_multiset(oa.poll_all())
+        assert mg == mo, (f"pipeline epoch {epoch + 1}: GPU {len(mg)} rows "
+                          f"vs oracle {len(mo)}")
+    gj.close()
+    ga.close()
+    oj.close()
+    oa.close()
