@@ -192,6 +192,7 @@ struct AggBatch {
     uint32_t n_rows;
     uint32_t stride = 1;
     uint8_t valid_inverted = 0;
+    uint8_t dense = 0; // all rows visible+Insert+non-null (checked at upload)
 };
 
 struct AggTableDev {
@@ -240,7 +241,7 @@ struct AggTableDev {
 // 64× (the unsorted case degrades to runs of 1 = plain per-lane atomics,
 // still correct). count/sum combine by +, min/max by min/max, null-presence
 // by OR; all order-free, so the result is exactly the reference's.
-template <int KW, int n_calls>
+template <int KW, int n_calls, bool DENSE>
 __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
                                  AggCallDev c1, AggCallDev c2, AggCallDev c3,
                                  int mode, uint32_t r0, uint32_t r1) {
@@ -262,14 +263,14 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
 
     for (uint32_t it = 0; it < iters; it++) {
         uint32_t r = r0 + it * stride + blockIdx.x * blockDim.x + threadIdx.x;
-        bool active = (r < r1) && !(b.vis && !b.vis[r]);
+        bool active = DENSE ? (r < r1) : ((r < r1) && !(b.vis && !b.vis[r]));
         int64_t kw[MAX_KW];
         uint32_t nullmask = 0;
         for (int i = 0; i < KW; i++) kw[i] = 0;
         size_t rs = (size_t)r * b.stride;
         if (active) {
             for (int i = 0; i < KW; i++) {
-                bool valid = b.col_valid[i][rs] ^ b.valid_inverted;
+                bool valid = DENSE || (b.col_valid[i][rs] ^ b.valid_inverted);
                 kw[i] = valid ? b.col_vals[i][rs] : 0;
                 nullmask |= (!valid) << i;
             }
@@ -333,7 +334,8 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
         int run_pos = lane - run_start;
 
         // per-lane contributions (identity when inactive / NULL arg)
-        uint8_t op = active && slot != SLOT_NONE ? b.ops[r] : RW_OP_INSERT;
+        uint8_t op = (!DENSE && active && slot != SLOT_NONE) ? b.ops[r]
+                                                             : RW_OP_INSERT;
         long long sign =
             (op == RW_OP_DELETE || op == RW_OP_UPDATE_DELETE) ? -1 : 1;
         bool contributing = active && slot != SLOT_NONE;
@@ -342,7 +344,9 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
         for (int ci = 0; ci < n_calls; ci++) {
             const AggCallDev& c = calls[ci];
             int col = KW + ci;
-            bool arg_valid = contributing && (b.col_valid[col][rs] ^ b.valid_inverted);
+            bool arg_valid =
+                contributing &&
+                (DENSE || (b.col_valid[col][rs] ^ b.valid_inverted));
             if (c.minput) {
                 // materialized-input ops are exact per-row chain mutations
                 // (minput.rs apply_batch): no wave aggregation
@@ -464,7 +468,7 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
             long long ov[4];
             uint8_t ohas;
             for (int ci = 0; ci < n_calls; ci++) ov[ci] = __shfl_up(v[ci], d);
-            ohas = (uint8_t)__shfl_up((int)hasmask, d);
+            ohas = DENSE ? hasmask : (uint8_t)__shfl_up((int)hasmask, d);
             if (run_pos >= d) {
                 for (int ci = 0; ci < n_calls; ci++) {
                     switch (calls[ci].kind) {
@@ -887,6 +891,23 @@ struct HashAgg {
             b.vis = nullptr;
         }
         b.n_rows = n;
+        // dense fast path: every row visible, Insert, and non-null in every
+        // referenced column — the kernel then skips the three byte streams
+        b.dense = (c->vis == nullptr);
+        for (uint32_t r = 0; b.dense && r < n; r++) b.dense = c->ops[r] == 0;
+        auto col_all_valid = [&](uint32_t col_idx) {
+            const uint8_t* v = c->cols[col_idx].valid;
+            for (uint32_t r = 0; r < n; r++)
+                if (!v[r]) return false;
+            return true;
+        };
+        for (int i = 0; b.dense && i < KW; i++)
+            b.dense = col_all_valid(group_key[i]);
+        for (int ci = 0; b.dense && ci < n_calls; ci++)
+            if (calls[ci].arg >= 0) b.dense = col_all_valid((uint32_t)calls[ci].arg);
+        if (n_minput)
+            for (size_t jx = 0; b.dense && jx < stream_key.size(); jx++)
+                b.dense = col_all_valid(stream_key[jx]);
         *out = b;
         return RW_OK;
     }
@@ -895,8 +916,14 @@ struct HashAgg {
         int grid = grid_for(r1 - r0);
         auto a0 = cd(0), a1 = cd(1), a2 = cd(2), a3 = cd(3);
         #define RW_LAUNCH(kw, nc)                                             \
-            agg_apply_kernel<kw, nc><<<grid, 256, 0, stream>>>(               \
-                b, t, a0, a1, a2, a3, debug_mode, r0, r1)
+            do {                                                              \
+                if (b.dense)                                                  \
+                    agg_apply_kernel<kw, nc, true><<<grid, 256, 0, stream>>>( \
+                        b, t, a0, a1, a2, a3, debug_mode, r0, r1);            \
+                else                                                          \
+                    agg_apply_kernel<kw, nc, false><<<grid, 256, 0, stream>>>(\
+                        b, t, a0, a1, a2, a3, debug_mode, r0, r1);            \
+            } while (0)
         switch (KW * 8 + n_calls) {
             case 1 * 8 + 1: RW_LAUNCH(1, 1); break;
             case 1 * 8 + 2: RW_LAUNCH(1, 2); break;
