@@ -1274,6 +1274,29 @@ int rw_agg_apply_payload(void* h, const uint8_t* payload,
     return RW_OK;
 }
 
+// bench sink: run the flush (change inference + prev update) but leave the
+// emitted records in HBM for the device-resident downstream — returns the
+// emitted row count, resets the cursor. The host-marshalling path
+// (rw_hash_agg_flush + poll) remains the parity-test surface.
+long long rw_agg_flush_device(void* h, uint64_t epoch) {
+    auto* agg = (HashAgg*)h;
+    (void)epoch;
+    agg_flush_kernel<<<2048, 256, 0, agg->stream>>>(
+        agg->t, agg->KW, agg->n_calls, (int)agg->desc.row_count_index,
+        agg->cd(0), agg->cd(1), agg->cd(2), agg->cd(3));
+    if (hipStreamSynchronize(agg->stream) != hipSuccess) return -1;
+    uint32_t ctr[3];
+    if (hipMemcpy(ctr, agg->t.counters, 12, hipMemcpyDeviceToHost) != hipSuccess)
+        return -1;
+    if (ctr[2] != 0) {
+        g_err = "agg overflow (code " + std::to_string(ctr[2]) + ")";
+        return -(long long)ctr[2] - 1;
+    }
+    long long n = ctr[1];
+    if (hipMemset(agg->t.counters, 0, 12) != hipSuccess) return -1;
+    return n;
+}
+
 int rw_agg_kernel_stats(void* h, RwKernelStats* out) {
     auto* agg = (HashAgg*)h;
     out->launches = agg->apply_launches;
