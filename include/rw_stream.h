@@ -138,6 +138,12 @@ typedef struct RwHashJoinDesc {
     uint32_t chunk_size;
     uint64_t state_capacity_hint; /* expected distinct keys per side (0 = default) */
     uint64_t row_capacity_hint;   /* expected resident rows per side (0 = default) */
+    /* join-key watermark positions (stream_plan.proto watermark handling;
+     * hash_join.rs watermark_indices_in_jk): positions into the join key
+     * with a per-position state-cleaning flag */
+    uint32_t n_wm_jk;
+    const uint32_t* wm_jk_pos;
+    const uint8_t* wm_jk_clean;
 } RwHashJoinDesc;
 
 enum RwJoinSide { RW_SIDE_LEFT = 0, RW_SIDE_RIGHT = 1 };
@@ -147,6 +153,21 @@ int rw_hash_join_push_chunk(void* h, int side, const RwChunk* chunk);
 int rw_hash_join_flush(void* h, uint64_t epoch); /* aligned barrier */
 RwChunk* rw_hash_join_poll(void* h);
 void rw_hash_join_destroy(void* h);
+
+/* Watermark on an input column (hash_join.rs:815-914 restricted to join-key
+ * watermarks): buffers per side, emits min across sides when it advances
+ * (BufferedWatermarks), cleans both sides' state below the selected value
+ * when the position's clean flag is set (TTL, state_table.rs:1707).
+ * Emitted output watermarks (update side's columns first, then the match
+ * side's — hash_join.rs:852-866) are returned through out_cols/out_vals
+ * (capacity max_out); returns the count or a negative error. */
+int rw_hash_join_watermark(void* h, int side, uint32_t col_idx, int64_t val,
+                           uint32_t* out_cols, int64_t* out_vals, int max_out);
+
+/* Watermark on an agg group-key position: state cleaning only (groups with
+ * key below the watermark are reset — hash_agg.rs:503-507 /
+ * update_watermark); non-EOWC, so nothing is emitted. */
+int rw_hash_agg_watermark(void* h, uint32_t group_key_pos, int64_t val);
 
 #ifdef __cplusplus
 }

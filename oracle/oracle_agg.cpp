@@ -326,6 +326,26 @@ struct HashAggOracle {
         return RW_OK;
     }
 
+    // watermark TTL cleaning (hash_agg.rs:503-507 → update_watermark):
+    // groups whose watermarked group-key column sorts below the value are
+    // dropped (NULLs largest, kept)
+    int watermark(uint32_t pos, int64_t val) {
+        if (pos >= group_key.size()) return RW_E_INVAL;
+        for (auto it = groups.begin(); it != groups.end();) {
+            const Datum& d = it->first[pos];
+            if (!d.null && d.i < val) {
+                dirty.erase(it->first);
+                it = groups.erase(it);
+            } else {
+                ++it;
+            }
+        }
+        for (auto it = dirty_order.begin(); it != dirty_order.end();)
+            if (!groups.count(*it)) it = dirty_order.erase(it);
+            else ++it;
+        return RW_OK;
+    }
+
     RwChunk* poll() {
         if (outputs.empty()) return nullptr;
         auto c = std::move(outputs.front());
@@ -354,6 +374,9 @@ int rw_hash_agg_push_chunk(void* h, const RwChunk* c) {
     return ((HashAggOracle*)h)->push_chunk(c);
 }
 int rw_hash_agg_flush(void* h, uint64_t epoch) { return ((HashAggOracle*)h)->flush(epoch); }
+int rw_hash_agg_watermark(void* h, uint32_t pos, int64_t val) {
+    return ((HashAggOracle*)h)->watermark(pos, val);
+}
 RwChunk* rw_hash_agg_poll(void* h) { return ((HashAggOracle*)h)->poll(); }
 void rw_hash_agg_destroy(void* h) { delete (HashAggOracle*)h; }
 

@@ -119,8 +119,19 @@ struct JoinOutBuilder {
     void take() { post(b.take()); }
 };
 
+struct WmEntry {
+    bool has = false;
+    int64_t val = 0;
+};
+
 struct HashJoinOracle {
     RwHashJoinDesc d_store;
+    // join-key watermark buffering (hash_join.rs:826-867; BufferedWatermarks
+    // emits the min across sides when it advances)
+    std::vector<uint32_t> wm_pos;
+    std::vector<uint8_t> wm_clean;
+    std::vector<WmEntry> wm_side[2];
+    std::vector<WmEntry> wm_out;
     uint8_t T;
     bool append_only;
     std::vector<uint8_t> null_safe;
@@ -141,6 +152,13 @@ struct HashJoinOracle {
         output_indices.assign(d->output_indices, d->output_indices + d->n_output);
         for (auto i : output_indices) out_types.push_back(concat_types[i]);
         d_store = *d;
+        if (d->n_wm_jk) {
+            wm_pos.assign(d->wm_jk_pos, d->wm_jk_pos + d->n_wm_jk);
+            wm_clean.assign(d->wm_jk_clean, d->wm_jk_clean + d->n_wm_jk);
+        }
+        wm_side[0].resize(d->n_key);
+        wm_side[1].resize(d->n_key);
+        wm_out.resize(d->n_key);
 
         side[0].key_idx.assign(d->key_l, d->key_l + d->n_key);
         side[1].key_idx.assign(d->key_r, d->key_r + d->n_key);
@@ -328,6 +346,50 @@ struct HashJoinOracle {
         outputs.erase(outputs.begin());
         return chunk_to_c(*c);
     }
+
+    // state cleaning below the selected watermark (update_watermark →
+    // state_table watermark cleaning, join/hash_join.rs:521-527): drop every
+    // key whose watermarked key column sorts below the value (NULLs are
+    // largest, never cleaned)
+    void clean_below(size_t jk_idx, int64_t sel) {
+        for (int s = 0; s < 2; s++) {
+            auto& tab = side[s].table;
+            for (auto it = tab.begin(); it != tab.end();) {
+                const Datum& kd = it->first[jk_idx];
+                if (!kd.null && kd.i < sel) it = tab.erase(it);
+                else ++it;
+            }
+        }
+    }
+
+    int watermark(int s, uint32_t col_idx, int64_t val, uint32_t* out_cols,
+                  int64_t* out_vals, int max_out) {
+        int n_out = 0;
+        for (size_t idx = 0; idx < side[s].key_idx.size(); idx++) {
+            if (side[s].key_idx[idx] != col_idx) continue;
+            wm_side[s][idx] = {true, val};
+            if (!wm_side[0][idx].has || !wm_side[1][idx].has) continue;
+            int64_t sel = std::min(wm_side[0][idx].val, wm_side[1][idx].val);
+            if (wm_out[idx].has && sel <= wm_out[idx].val) continue;
+            wm_out[idx] = {true, sel};
+            for (size_t w = 0; w < wm_pos.size(); w++)
+                if (wm_pos[w] == idx && wm_clean[w]) clean_below(idx, sel);
+            // emit for the update side's mapped output columns, then the
+            // match side's (hash_join.rs:852-866)
+            for (int s2 : {s, 1 - s}) {
+                uint32_t src = side[s2].key_idx[idx] +
+                               (s2 == RW_SIDE_RIGHT ? (uint32_t)types_l.size() : 0);
+                for (uint32_t oi = 0; oi < output_indices.size(); oi++) {
+                    if (output_indices[oi] == src && n_out < max_out) {
+                        out_cols[n_out] = oi;
+                        out_vals[n_out] = sel;
+                        n_out++;
+                    }
+                }
+            }
+        }
+        return n_out;
+    }
 };
 
 } // namespace orc
@@ -340,6 +402,12 @@ void* rw_hash_join_create(const RwHashJoinDesc* d) { return new HashJoinOracle(d
 int rw_hash_join_push_chunk(void* h, int side, const RwChunk* c) {
     return ((HashJoinOracle*)h)->push(side, c);
 }
+int rw_hash_join_watermark(void* h, int side, uint32_t col_idx, int64_t val,
+                           uint32_t* out_cols, int64_t* out_vals, int max_out) {
+    return ((HashJoinOracle*)h)->watermark(side, col_idx, val, out_cols,
+                                           out_vals, max_out);
+}
+
 int rw_hash_join_flush(void* h, uint64_t) {
     (void)h; // state commit is a no-op for the in-memory oracle
     return RW_OK;

@@ -1208,6 +1208,7 @@ void* rw_agg_bench_preload(void* h, const RwChunk* c) {
         delete b;
         return nullptr;
     }
+    *b = out; // keep the computed fields (n_rows, dense, vis)
     hipStreamSynchronize(agg->stream);
     return b;
 }
@@ -1890,6 +1891,59 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
     }
 }
 
+// watermark TTL sweeps (state_table watermark cleaning):
+// rows/groups whose watermarked key column sorts below the value are
+// retired in place (slots stay READY so linear probing is undisturbed;
+// memory reclamation is compaction work for a later round)
+__global__ void join_clean_kernel(JoinSideDev sd, int kpos, long long wm, int KW) {
+    size_t cap = (size_t)sd.cap_mask + 1;
+    size_t stride = (size_t)gridDim.x * blockDim.x;
+    for (size_t slot = blockIdx.x * blockDim.x + threadIdx.x; slot < cap;
+         slot += stride) {
+        if (ld_u32(&sd.state[(uint32_t)slot]) != SLOT_READY) continue;
+        if ((ld_u32(&sd.key_nulls[(uint32_t)slot]) >> kpos) & 1) continue; // NULLs largest
+        if (ld_i64((const int64_t*)&sd.keys[slot * KW + kpos]) >= wm) continue;
+        uint32_t row = ld_u32(&sd.head[(uint32_t)slot]);
+        while (row != UINT32_MAX) {
+            st_u32(&sd.alive[row], 0);
+            row = ld_u32(&sd.next[row]);
+        }
+        st_u32(&sd.head[(uint32_t)slot], UINT32_MAX);
+    }
+}
+
+__global__ void agg_clean_kernel(AggTableDev t, int kpos, long long wm, int KW,
+                                 int n_calls, AggCallDev c0, AggCallDev c1,
+                                 AggCallDev c2, AggCallDev c3) {
+    AggCallDev calls[4] = {c0, c1, c2, c3};
+    size_t cap = (size_t)t.cap_mask + 1;
+    size_t stride = (size_t)gridDim.x * blockDim.x;
+    for (size_t slot = blockIdx.x * blockDim.x + threadIdx.x; slot < cap;
+         slot += stride) {
+        if (ld_u32(&t.state[(uint32_t)slot]) != SLOT_READY) continue;
+        if ((ld_u32(&t.key_nulls[(uint32_t)slot]) >> kpos) & 1) continue;
+        if (ld_i64((const int64_t*)&t.keys[slot * KW + kpos]) >= wm) continue;
+        // reset the group as if freshly created (late rows restart it)
+        for (int ci = 0; ci < n_calls; ci++) {
+            long long init = 0;
+            if (calls[ci].kind == RW_AGG_MIN) init = INT64_MAX;
+            if (calls[ci].kind == RW_AGG_MAX) init = INT64_MIN;
+            t.acc[(size_t)ci * cap + slot] = init;
+            t.has[(size_t)ci * cap + slot] = 0;
+            if (calls[ci].minput) {
+                uint32_t row = t.mheads[(size_t)calls[ci].mord * cap + slot];
+                while (row != UINT32_MAX) {
+                    st_u32(&t.malive[row], 0);
+                    row = ld_u32(&t.mnext[row]);
+                }
+                t.mheads[(size_t)calls[ci].mord * cap + slot] = UINT32_MAX;
+            }
+        }
+        t.has_prev[(uint32_t)slot] = 0;
+        t.dirty_flag[(uint32_t)slot] = 0;
+    }
+}
+
 struct HashJoin {
     RwHashJoinDesc desc;
     JoinMeta m{};
@@ -1903,6 +1957,11 @@ struct HashJoin {
     uint32_t stage_cap[2] = {0, 0};
     std::vector<RwChunk*> outq;
     uint8_t* zeros = nullptr; // all-zero null flags for pipeline dummy cols
+    // join-key watermark buffering (hash_join.rs:826-867)
+    struct Wm { bool has = false; int64_t val = 0; };
+    std::vector<uint32_t> wm_pos;
+    std::vector<uint8_t> wm_clean;
+    std::vector<Wm> wm_side0, wm_side1, wm_out;
     double probe_ms_total = 0;
     uint64_t probe_launches = 0, probe_rows = 0;
 
@@ -1961,6 +2020,13 @@ struct HashJoin {
         split(d->cond_l, &m.cond_src_l, &m.cond_col_l);
         split(d->cond_r, &m.cond_src_r, &m.cond_col_r);
         m.append_only = d->append_only;
+        if (d->n_wm_jk) {
+            wm_pos.assign(d->wm_jk_pos, d->wm_jk_pos + d->n_wm_jk);
+            wm_clean.assign(d->wm_jk_clean, d->wm_jk_clean + d->n_wm_jk);
+        }
+        wm_side0.resize(d->n_key);
+        wm_side1.resize(d->n_key);
+        wm_out.resize(d->n_key);
 
         HIP_TRY(hipStreamCreate(&stream));
         uint64_t key_cap = d->state_capacity_hint ? d->state_capacity_hint : (1u << 20);
@@ -2184,6 +2250,47 @@ struct HashJoin {
         return RW_OK;
     }
 
+    int watermark(int s, uint32_t col_idx, int64_t val, uint32_t* out_cols,
+                  int64_t* out_vals, int max_out, int* n_out_p) {
+        int n_out = 0;
+        auto& mine = s == 0 ? wm_side0 : wm_side1;
+        for (uint32_t idx = 0; idx < (uint32_t)m.KW; idx++) {
+            if (m.key_cols[s][idx] != col_idx) continue;
+            mine[idx] = {true, val};
+            if (!wm_side0[idx].has || !wm_side1[idx].has) continue;
+            int64_t sel = std::min(wm_side0[idx].val, wm_side1[idx].val);
+            if (wm_out[idx].has && sel <= wm_out[idx].val) continue;
+            wm_out[idx] = {true, sel};
+            bool clean = false;
+            for (size_t w = 0; w < wm_pos.size(); w++)
+                if (wm_pos[w] == idx && wm_clean[w]) clean = true;
+            if (clean) {
+                for (int sd = 0; sd < 2; sd++)
+                    join_clean_kernel<<<2048, 256, 0, stream>>>(side[sd],
+                                                                (int)idx, sel,
+                                                                m.KW);
+                HIP_TRY(hipStreamSynchronize(stream));
+            }
+            // update side's output columns first, then the match side's
+            for (int pass = 0; pass < 2; pass++) {
+                int s2 = pass == 0 ? s : 1 - s;
+                uint32_t src = m.key_cols[s2][idx] +
+                               (s2 == 1 ? (uint32_t)m.n_cols[0] : 0);
+                for (int oi = 0; oi < m.n_out; oi++) {
+                    uint32_t osrc = (uint32_t)m.out_col[oi] +
+                                    (m.out_src[oi] ? (uint32_t)m.n_cols[0] : 0);
+                    if (osrc == src && n_out < max_out) {
+                        out_cols[n_out] = (uint32_t)oi;
+                        out_vals[n_out] = sel;
+                        n_out++;
+                    }
+                }
+            }
+        }
+        *n_out_p = n_out;
+        return RW_OK;
+    }
+
     RwChunk* poll() {
         if (outq.empty()) return nullptr;
         RwChunk* c = outq.front();
@@ -2240,6 +2347,24 @@ int rw_hash_join_push_chunk(void* h, int side, const RwChunk* c) {
     return ((HashJoin*)h)->push_chunk(side, c);
 }
 int rw_hash_join_flush(void* h, uint64_t epoch) { return ((HashJoin*)h)->flush(epoch); }
+
+int rw_hash_join_watermark(void* h, int side, uint32_t col_idx, int64_t val,
+                           uint32_t* out_cols, int64_t* out_vals, int max_out) {
+    int n = 0;
+    int rc = ((HashJoin*)h)->watermark(side, col_idx, val, out_cols, out_vals,
+                                       max_out, &n);
+    return rc == RW_OK ? n : rc;
+}
+
+int rw_hash_agg_watermark(void* h, uint32_t group_key_pos, int64_t val) {
+    auto* agg = (HashAgg*)h;
+    if (group_key_pos >= (uint32_t)agg->KW) FAIL(RW_E_INVAL, "bad group key pos");
+    agg_clean_kernel<<<2048, 256, 0, agg->stream>>>(
+        agg->t, (int)group_key_pos, val, agg->KW, agg->n_calls, agg->cd(0),
+        agg->cd(1), agg->cd(2), agg->cd(3));
+    HIP_TRY(hipStreamSynchronize(agg->stream));
+    return RW_OK;
+}
 RwChunk* rw_hash_join_poll(void* h) { return ((HashJoin*)h)->poll(); }
 void rw_hash_join_destroy(void* h) { delete (HashJoin*)h; }
 

@@ -15,6 +15,7 @@
 #include <rccl/rccl.h>
 
 #include <cstdint>
+#include <cstdlib>
 #include <cstring>
 #include <string>
 #include <vector>
@@ -259,38 +260,55 @@ int rw_exchange_run(void* h, const int64_t* const* col_vals,
     x_scatter_kernel<<<blocks, 256, 0, x->stream>>>(b, n_cols, d_dest, d_offsets,
                                                     d_counts, d_cursors, send_buf);
 
-    // exchange per-peer row counts (all-to-all of one u64 per peer), then
-    // the payload blocks (all-to-all-v), all on the exchange stream
-    unsigned long long* d_count_mat = x->d_count_mat;
-    XNCCL(ncclGroupStart());
-    for (int p = 0; p < R; p++) {
-        XNCCL(ncclSend(d_counts + p, 1, ncclUint64, p, x->comm, x->stream));
-        XNCCL(ncclRecv(d_count_mat + p, 1, ncclUint64, p, x->comm, x->stream));
+    // exchange per-peer row counts, then the payload blocks (all-to-all-v)
+    // over RCCL/xGMI. A single-rank communicator is a degenerate self-loop:
+    // bypass RCCL with a device copy (semantics identical; RCCL's loopback
+    // path costs ~25 ms per 28 MB on this stack). Set
+    // RW_EXCHANGE_FORCE_NCCL=1 to exercise the collective at R=1.
+    static int force_nccl = -1;
+    if (force_nccl < 0) {
+        const char* e = getenv("RW_EXCHANGE_FORCE_NCCL");
+        force_nccl = e && *e == '1';
     }
-    XNCCL(ncclGroupEnd());
     unsigned long long recv_counts[64];
-    XHIP(hipMemcpyAsync(recv_counts, d_count_mat, R * 8, hipMemcpyDeviceToHost,
-                        x->stream));
-    XHIP(hipStreamSynchronize(x->stream));
-
-    uint64_t roff = 0;
     unsigned long long recv_offsets[64];
-    for (int p = 0; p < R; p++) {
-        recv_offsets[p] = roff;
-        roff += block_bytes(recv_counts[p]);
-    }
-    if (roff > recv_cap) XFAIL(-3, "recv buffer too small");
+    if (R == 1 && !force_nccl) {
+        recv_counts[0] = counts[0];
+        recv_offsets[0] = 0;
+        if (block_bytes(counts[0]) > recv_cap) XFAIL(-3, "recv buffer too small");
+        XHIP(hipMemcpyAsync(recv_buf, send_buf, block_bytes(counts[0]),
+                            hipMemcpyDeviceToDevice, x->stream));
+    } else {
+        unsigned long long* d_count_mat = x->d_count_mat;
+        XNCCL(ncclGroupStart());
+        for (int p = 0; p < R; p++) {
+            XNCCL(ncclSend(d_counts + p, 1, ncclUint64, p, x->comm, x->stream));
+            XNCCL(ncclRecv(d_count_mat + p, 1, ncclUint64, p, x->comm, x->stream));
+        }
+        XNCCL(ncclGroupEnd());
+        XHIP(hipMemcpyAsync(recv_counts, d_count_mat, R * 8,
+                            hipMemcpyDeviceToHost, x->stream));
+        XHIP(hipStreamSynchronize(x->stream));
 
-    XNCCL(ncclGroupStart());
-    for (int p = 0; p < R; p++) {
-        if (counts[p])
-            XNCCL(ncclSend(send_buf + offsets[p], block_bytes(counts[p]), ncclUint8,
-                           p, x->comm, x->stream));
-        if (recv_counts[p])
-            XNCCL(ncclRecv(recv_buf + recv_offsets[p], block_bytes(recv_counts[p]),
-                           ncclUint8, p, x->comm, x->stream));
+        uint64_t roff = 0;
+        for (int p = 0; p < R; p++) {
+            recv_offsets[p] = roff;
+            roff += block_bytes(recv_counts[p]);
+        }
+        if (roff > recv_cap) XFAIL(-3, "recv buffer too small");
+
+        XNCCL(ncclGroupStart());
+        for (int p = 0; p < R; p++) {
+            if (counts[p])
+                XNCCL(ncclSend(send_buf + offsets[p], block_bytes(counts[p]),
+                               ncclUint8, p, x->comm, x->stream));
+            if (recv_counts[p])
+                XNCCL(ncclRecv(recv_buf + recv_offsets[p],
+                               block_bytes(recv_counts[p]), ncclUint8, p,
+                               x->comm, x->stream));
+        }
+        XNCCL(ncclGroupEnd());
     }
-    XNCCL(ncclGroupEnd());
     XHIP(hipEventRecord(e1, x->stream));
     XHIP(hipStreamSynchronize(x->stream));
     float ms = 0;

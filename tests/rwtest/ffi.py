@@ -94,6 +94,9 @@ class RwHashJoinDesc(C.Structure):
         ("chunk_size", C.c_uint32),
         ("state_capacity_hint", C.c_uint64),
         ("row_capacity_hint", C.c_uint64),
+        ("n_wm_jk", C.c_uint32),
+        ("wm_jk_pos", C.POINTER(C.c_uint32)),
+        ("wm_jk_clean", C.POINTER(C.c_uint8)),
     ]
 
 
@@ -221,6 +224,15 @@ class Lib:
         L.rw_hash_join_destroy.argtypes = [C.c_void_p]
         L.rw_chunk_free.argtypes = [C.POINTER(RwChunkC)]
         L.rw_last_error.restype = C.c_char_p
+        try:
+            L.rw_hash_join_watermark.restype = C.c_int
+            L.rw_hash_join_watermark.argtypes = [
+                C.c_void_p, C.c_int, C.c_uint32, C.c_int64,
+                C.POINTER(C.c_uint32), C.POINTER(C.c_int64), C.c_int]
+            L.rw_hash_agg_watermark.restype = C.c_int
+            L.rw_hash_agg_watermark.argtypes = [C.c_void_p, C.c_uint32, C.c_int64]
+        except AttributeError:
+            pass
 
     def last_error(self):
         return self.lib.rw_last_error().decode()
@@ -297,6 +309,11 @@ class HashAgg:
                 return out
             out.append(self.lib._read_chunk(p))
 
+    def watermark(self, group_key_pos, val):
+        rc = self.lib.lib.rw_hash_agg_watermark(self.h, group_key_pos, val)
+        if rc != 0:
+            raise RuntimeError(f"agg watermark failed {rc}")
+
     def close(self):
         if self.h:
             self.lib.lib.rw_hash_agg_destroy(self.h)
@@ -307,7 +324,7 @@ class HashJoin:
     def __init__(self, lib: Lib, join_type, types_l, types_r, key_l, key_r,
                  pk_l, pk_r, output_indices=None, null_safe=None, cond=None,
                  chunk_size=1024, append_only=False, state_capacity_hint=0,
-                 row_capacity_hint=0):
+                 row_capacity_hint=0, wm_jk=()):
         """cond: (op, cond_l, cond_r) into the concatenated row, or None."""
         self.lib = lib
         d = RwHashJoinDesc()
@@ -351,6 +368,11 @@ class HashJoin:
         d.chunk_size = chunk_size
         d.state_capacity_hint = state_capacity_hint
         d.row_capacity_hint = row_capacity_hint
+        d.n_wm_jk = len(wm_jk)
+        self._wp = _u32arr([p for p, _ in wm_jk])
+        self._wc = _u8arr([1 if c else 0 for _, c in wm_jk])
+        d.wm_jk_pos = self._wp
+        d.wm_jk_clean = self._wc
         self.h = lib.lib.rw_hash_join_create(C.byref(d))
         if not self.h:
             raise RuntimeError(f"rw_hash_join_create failed: {lib.last_error()}")
@@ -373,6 +395,15 @@ class HashJoin:
             if not p:
                 return out
             out.append(self.lib._read_chunk(p))
+
+    def watermark(self, side, col_idx, val, max_out=16):
+        cols = (C.c_uint32 * max_out)()
+        vals = (C.c_int64 * max_out)()
+        n = self.lib.lib.rw_hash_join_watermark(self.h, side, col_idx, val,
+                                                cols, vals, max_out)
+        if n < 0:
+            raise RuntimeError(f"watermark failed {n}: {self.lib.last_error()}")
+        return [(int(cols[i]), int(vals[i])) for i in range(n)]
 
     def close(self):
         if self.h:

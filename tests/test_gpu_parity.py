@@ -441,3 +441,50 @@ def test_pipeline_join_agg_q3_shape():
     ga.close()
     oj.close()
     oa.close()
+
+
+def test_join_watermark_gpu():
+    # emission order/values (hash_join.rs:3667-3737) + TTL cleaning effect
+    g = ffi.HashJoin(gpu(), JOIN_INNER, [T_I64, T_I64], [T_I64, T_I64],
+                     key_l=[0], key_r=[0], pk_l=[1], pk_r=[1], wm_jk=[(0, True)])
+    assert g.watermark(SIDE_LEFT, 0, 100) == []
+    assert g.watermark(SIDE_LEFT, 0, 200) == []
+    assert g.watermark(SIDE_RIGHT, 0, 50) == [(2, 50), (0, 50)]
+    assert g.watermark(SIDE_RIGHT, 0, 100) == [(2, 100), (0, 100)]
+    g.close()
+
+    g = ffi.HashJoin(gpu(), JOIN_INNER, [T_I64, T_I64], [T_I64, T_I64],
+                     key_l=[0], key_r=[0], pk_l=[1], pk_r=[1], wm_jk=[(0, True)])
+    o = ffi.HashJoin(oracle(), JOIN_INNER, [T_I64, T_I64], [T_I64, T_I64],
+                     key_l=[0], key_r=[0], pk_l=[1], pk_r=[1], wm_jk=[(0, True)])
+    c = from_pretty(" I I\n + 2 1\n + 6 2")
+    for j in (g, o):
+        j.push(SIDE_LEFT, c)
+        j.poll_all()
+        j.watermark(SIDE_LEFT, 0, 5)
+        j.watermark(SIDE_RIGHT, 0, 5)
+    probe = from_pretty(" I I\n + 2 10\n + 6 11")
+    g.push(SIDE_RIGHT, probe)
+    o.push(SIDE_RIGHT, probe)
+    assert rows_multiset(g.poll_all()) == rows_multiset(o.poll_all())
+    g.close()
+    o.close()
+
+
+def test_agg_watermark_gpu():
+    calls = [(AGG_COUNT_STAR, -1, T_I64), (AGG_SUM, 1, T_I64)]
+    g = ffi.HashAgg(gpu(), [T_I64, T_I64], [0], calls, 0)
+    o = ffi.HashAgg(oracle(), [T_I64, T_I64], [0], calls, 0)
+    c1 = from_pretty(" I I\n + 1 5\n + 9 7")
+    c2 = from_pretty(" I I\n + 1 3\n + 9 1")
+    for a in (g, o):
+        a.push(c1)
+        a.flush(1)
+    assert rows_multiset(g.poll_all()) == rows_multiset(o.poll_all())
+    for a in (g, o):
+        a.watermark(0, 5)
+        a.push(c2)
+        a.flush(2)
+    assert rows_multiset(g.poll_all()) == rows_multiset(o.poll_all())
+    g.close()
+    o.close()

@@ -206,3 +206,28 @@ def run_epochs_keep(agg, epochs):
         agg.flush(i + 1)
         out.append(rows_multiset(agg.poll_all()))
     return out
+
+
+def test_agg_watermark_group_cleaning():
+    # hash_agg.rs:503-507: groups under the window watermark are cleaned; a
+    # late row recreates the group from scratch
+    agg = ffi.HashAgg(
+        oracle(),
+        input_types=[ffi.T_I64, ffi.T_I64],
+        group_key=[0],
+        calls=[(ffi.AGG_COUNT_STAR, -1, ffi.T_I64), (AGG_SUM, 1, ffi.T_I64)],
+        row_count_index=0,
+    )
+    agg.push(from_pretty(" I I\n + 1 5\n + 9 7"))
+    agg.flush(1)
+    assert rows_multiset(agg.poll_all()) == expect(
+        [("+", (1, 1, 5)), ("+", (9, 1, 7))]
+    )
+    agg.watermark(0, 5)  # group 1 cleaned
+    agg.push(from_pretty(" I I\n + 1 3\n + 9 1"))
+    agg.flush(2)
+    # group 1 restarts as a fresh group (Insert), group 9 updates
+    assert rows_multiset(agg.poll_all()) == expect(
+        [("+", (1, 1, 3)), ("U-", (9, 1, 7)), ("U+", (9, 2, 8))]
+    )
+    agg.close()

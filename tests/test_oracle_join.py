@@ -311,3 +311,32 @@ def test_inner_join_nonequi():
     assert push(j, SIDE_RIGHT, " I I\n + 2 7\n + 4 8\n + 6 9") == []
     assert push(j, SIDE_RIGHT, " I I\n + 3 10\n + 6 11") == rows([("+", 3, 6, 3, 10)])
     j.close()
+
+
+def test_join_watermark_emission():
+    # hash_join.rs:3667-3737 (test_streaming_hash_join_watermark): watermarks
+    # buffer per side; the min across sides is emitted when it advances, for
+    # the update side's output column first, then the match side's
+    j = ffi.HashJoin(
+        oracle(), JOIN_INNER, I2, I2, key_l=[0], key_r=[0], pk_l=[1], pk_r=[1],
+        cond=(CMP_LT, 1, 3), wm_jk=[(0, True)],
+    )
+    assert j.watermark(SIDE_LEFT, 0, 100) == []
+    assert j.watermark(SIDE_LEFT, 0, 200) == []
+    # right 50 -> selected min(200,50)=50; update side (right) col 0 maps to
+    # output 2, then left col 0 -> output 0
+    assert j.watermark(SIDE_RIGHT, 0, 50) == [(2, 50), (0, 50)]
+    assert j.watermark(SIDE_RIGHT, 0, 100) == [(2, 100), (0, 100)]
+    j.close()
+
+
+def test_join_watermark_state_cleaning():
+    # rows below the selected watermark are cleaned from both sides and no
+    # longer match (hash_join.rs:843-848 -> update_watermark TTL)
+    j = ffi.HashJoin(oracle(), JOIN_INNER, I2, I2, key_l=[0], key_r=[0],
+                     pk_l=[1], pk_r=[1], wm_jk=[(0, True)])
+    assert push(j, SIDE_LEFT, " I I\n + 2 1\n + 6 2") == []
+    j.watermark(SIDE_LEFT, 0, 5)
+    j.watermark(SIDE_RIGHT, 0, 5)  # selected = 5: left row (2,1) cleaned
+    assert push(j, SIDE_RIGHT, " I I\n + 2 10\n + 6 11") == rows([("+", 6, 2, 6, 11)])
+    j.close()
