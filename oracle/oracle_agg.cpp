@@ -19,9 +19,12 @@
 //  - Materialized-input min/max (aggregate/minput.rs:170-248): state ordered
 //    by [value ASC(min)/DESC(max), stream_key ASC], NULLs largest
 //    (test_utils/agg_executor.rs:72-121); output = first entry's value.
+#include <cstdlib>
+#include <cstring>
 #include <stdexcept>
 #include <unordered_map>
 
+#include "../include/rw_codec.hpp"
 #include "../include/rw_stream.h"
 #include "common.hpp"
 
@@ -89,6 +92,9 @@ struct HashAggOracle {
 
     ChunkBuilder builder;
     std::vector<std::unique_ptr<OwnedChunk>> outputs;
+    // checkpoint spill buffer (§8f-2): intermediate-state-table KV deltas in
+    // memcomparable-key / value-encoded form, accumulated per flush
+    std::vector<uint8_t> spill;
 
     HashAggOracle(const RwHashAggDesc* d, std::vector<uint8_t> out_ts)
         : builder(d->chunk_size, out_ts) {
@@ -280,6 +286,42 @@ struct HashAggOracle {
         return out;
     }
 
+    // append one state-table KV delta (StateTable::commit spill boundary,
+    // state_table.rs:1718): PUT(key,row) for Insert/Update, DELETE(key) for
+    // Delete. Row = group key ++ encoded states (value-state calls carry the
+    // value; materialized-input states encode None — agg_group.rs:417-421).
+    void spill_record(uint8_t put, const Row& key, const Row& outputs) {
+        spill.push_back(put);
+        std::vector<uint8_t> k, v;
+        for (size_t i = 0; i < key.size(); i++) {
+            rwcodec::DatumC d{key[i].null, key[i].i, key[i].d};
+            rwcodec::memcmp_encode_datum(k, group_key_types[i], d, {});
+        }
+        if (put) {
+            for (size_t i = 0; i < key.size(); i++) {
+                rwcodec::DatumC d{key[i].null, key[i].i, key[i].d};
+                rwcodec::value_encode_datum(v, group_key_types[i], d);
+            }
+            for (size_t ci = 0; ci < calls.size(); ci++) {
+                if (call_is_minput[ci]) {
+                    rwcodec::value_encode_datum(v, calls[ci].ret_type,
+                                                {true, 0, 0});
+                } else {
+                    const Datum& o = outputs[ci];
+                    rwcodec::value_encode_datum(v, calls[ci].ret_type,
+                                                {o.null, o.i, o.d});
+                }
+            }
+        }
+        auto put32 = [&](uint32_t x) {
+            for (int b = 0; b < 4; b++) spill.push_back((uint8_t)(x >> (8 * b)));
+        };
+        put32((uint32_t)k.size());
+        spill.insert(spill.end(), k.begin(), k.end());
+        put32((uint32_t)v.size());
+        spill.insert(spill.end(), v.begin(), v.end());
+    }
+
     void emit(uint8_t op, const Row& key, const Row& outputs) {
         Row row;
         row.reserve(key.size() + outputs.size());
@@ -304,10 +346,12 @@ struct HashAggOracle {
                 // nothing
             } else if (prev_rc == 0) {
                 emit(RW_OP_INSERT, key, curr);
+                spill_record(1, key, curr);
                 g.prev_outputs = curr;
                 g.has_prev = true;
             } else if (curr_rc == 0) {
                 emit(RW_OP_DELETE, key, g.prev_outputs);
+                spill_record(0, key, g.prev_outputs);
                 g.has_prev = false;
                 g.prev_outputs.clear();
             } else {
@@ -316,6 +360,7 @@ struct HashAggOracle {
                 if (!row_eq(g.prev_outputs, curr, call_types)) {
                     emit(RW_OP_UPDATE_DELETE, key, g.prev_outputs);
                     emit(RW_OP_UPDATE_INSERT, key, curr);
+                    spill_record(1, key, curr);
                     g.prev_outputs = curr;
                 }
             }
@@ -377,6 +422,18 @@ int rw_hash_agg_flush(void* h, uint64_t epoch) { return ((HashAggOracle*)h)->flu
 int rw_hash_agg_watermark(void* h, uint32_t pos, int64_t val) {
     return ((HashAggOracle*)h)->watermark(pos, val);
 }
+
+// drain the checkpoint spill buffer (caller frees with rw_spill_free)
+int rw_agg_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len) {
+    auto* a = (HashAggOracle*)h;
+    *len = a->spill.size();
+    *buf = (uint8_t*)malloc(a->spill.size() ? a->spill.size() : 1);
+    memcpy(*buf, a->spill.data(), a->spill.size());
+    a->spill.clear();
+    return RW_OK;
+}
+
+void rw_spill_free(uint8_t* buf) { free(buf); }
 RwChunk* rw_hash_agg_poll(void* h) { return ((HashAggOracle*)h)->poll(); }
 void rw_hash_agg_destroy(void* h) { delete (HashAggOracle*)h; }
 

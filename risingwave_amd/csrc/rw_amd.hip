@@ -28,6 +28,7 @@
 #include <unordered_set>
 #include <vector>
 
+#include "../../include/rw_codec.hpp"
 #include "../../include/rw_stream.h"
 
 #define WAVE 64
@@ -702,6 +703,9 @@ struct HashAgg {
     uint64_t apply_launches = 0, apply_rows = 0;
     // pending outputs
     std::vector<RwChunk*> outq;
+    // checkpoint spill buffer (§8f-2): state-table KV deltas, accumulated at
+    // flush, drained by rw_agg_checkpoint_drain
+    std::vector<uint8_t> spill;
     int debug_mode = 0; // RW_AGG_DEBUG_MODE: 1 per-lane atomics, 2 no-dedupe
     int n_minput = 0;   // materialized-input (retractable min/max) calls
     std::vector<uint8_t> call_minput;
@@ -1055,10 +1059,58 @@ struct HashAgg {
             HIP_TRY(hipMemcpy(nulls.data(), t.out_nulls, nulls.size(),
                               hipMemcpyDeviceToHost));
             HIP_TRY(hipMemcpy(ops.data(), t.out_ops, n_out, hipMemcpyDeviceToHost));
+            spill_records(vals, nulls, ops, n_out);
             slice_outputs(vals, nulls, ops, n_out);
         }
         HIP_TRY(hipMemset(t.counters, 0, 12));
         return RW_OK;
+    }
+
+    // Append state-table KV deltas for the flushed records (the spill
+    // boundary of StateTable::commit, state_table.rs:1718): PUT for
+    // Insert/U+, DELETE for Delete; U− rows carry no delta of their own.
+    // Keys are memcomparable (ASC NULLS LAST per group-key column); values
+    // are value-encoded rows of group key ++ encoded states (materialized-
+    // input states encode None, agg_group.rs:417-421).
+    void spill_records(const std::vector<long long>& vals,
+                       const std::vector<uint8_t>& nulls,
+                       const std::vector<uint8_t>& ops, uint32_t n_out) {
+        auto put32 = [&](uint32_t x) {
+            for (int b = 0; b < 4; b++) spill.push_back((uint8_t)(x >> (8 * b)));
+        };
+        for (uint32_t r = 0; r < n_out; r++) {
+            uint8_t op = ops[r];
+            if (op == RW_OP_UPDATE_DELETE) continue;
+            uint8_t put = op != RW_OP_DELETE;
+            spill.push_back(put);
+            std::vector<uint8_t> k, v;
+            for (int i = 0; i < KW; i++) {
+                rwcodec::DatumC d{nulls[(size_t)r * out_width + i] != 0,
+                                  vals[(size_t)r * out_width + i], 0};
+                rwcodec::memcmp_encode_datum(k, out_types[i], d, {});
+            }
+            if (put) {
+                for (int i = 0; i < KW; i++) {
+                    rwcodec::DatumC d{nulls[(size_t)r * out_width + i] != 0,
+                                      vals[(size_t)r * out_width + i], 0};
+                    rwcodec::value_encode_datum(v, out_types[i], d);
+                }
+                for (int ci = 0; ci < n_calls; ci++) {
+                    if (call_minput[ci]) {
+                        rwcodec::value_encode_datum(v, calls[ci].ret_type,
+                                                    {true, 0, 0});
+                    } else {
+                        size_t ix = (size_t)r * out_width + KW + ci;
+                        rwcodec::DatumC d{nulls[ix] != 0, vals[ix], 0};
+                        rwcodec::value_encode_datum(v, calls[ci].ret_type, d);
+                    }
+                }
+            }
+            put32((uint32_t)k.size());
+            spill.insert(spill.end(), k.begin(), k.end());
+            put32((uint32_t)v.size());
+            spill.insert(spill.end(), v.begin(), v.end());
+        }
     }
 
     // Host-side chunking with the U-pair no-split rule
@@ -1375,6 +1427,17 @@ int rw_agg_debug_dirty(void* h, uint32_t max_n, uint32_t* slots, uint32_t* state
     return RW_OK;
 }
 
+int rw_agg_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len) {
+    auto* agg = (HashAgg*)h;
+    *len = agg->spill.size();
+    *buf = (uint8_t*)malloc(agg->spill.size() ? agg->spill.size() : 1);
+    memcpy(*buf, agg->spill.data(), agg->spill.size());
+    agg->spill.clear();
+    return RW_OK;
+}
+
+void rw_spill_free(uint8_t* buf) { free(buf); }
+
 int rw_agg_stats_reset(void* h) {
     auto* agg = (HashAgg*)h;
     agg->apply_launches = 0;
@@ -1396,11 +1459,18 @@ int rw_agg_stats_reset(void* h) {
 __constant__ uint32_t g_crc_table[256];
 static bool g_crc_table_init = false;
 
-__device__ __forceinline__ uint32_t crc32_bytes(uint32_t crc, const uint8_t* p,
-                                                int n) {
+// LDS-staged CRC: divergent indexing of __constant__ memory serializes
+// (each distinct address replays); LDS banks handle it at full rate
+__device__ __forceinline__ uint32_t crc32_bytes(const uint32_t* lut, uint32_t crc,
+                                               const uint8_t* p, int n) {
     for (int i = 0; i < n; i++)
-        crc = g_crc_table[(crc ^ p[i]) & 0xFF] ^ (crc >> 8);
+        crc = lut[(crc ^ p[i]) & 0xFF] ^ (crc >> 8);
     return crc;
+}
+
+__device__ __forceinline__ void stage_crc_lut(uint32_t* lut) {
+    for (int i = threadIdx.x; i < 256; i += blockDim.x) lut[i] = g_crc_table[i];
+    __syncthreads();
 }
 
 struct VnodeBatch {
@@ -1413,6 +1483,8 @@ struct VnodeBatch {
 __global__ void vnode_kernel(VnodeBatch b, int n_keys, uint32_t vnode_count,
                              uint16_t* out, uint8_t t0, uint8_t t1, uint8_t t2,
                              uint8_t t3) {
+    __shared__ uint32_t lut[256];
+    stage_crc_lut(lut);
     uint8_t types[4] = {t0, t1, t2, t3};
     uint32_t stride = gridDim.x * blockDim.x;
     for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < b.n_rows;
@@ -1421,15 +1493,15 @@ __global__ void vnode_kernel(VnodeBatch b, int n_keys, uint32_t vnode_count,
         for (int k = 0; k < n_keys; k++) {
             if (!b.col_valid[k][r]) {
                 uint32_t sentinel = 0xfffffff0u;
-                crc = crc32_bytes(crc, (const uint8_t*)&sentinel, 4);
+                crc = crc32_bytes(lut, crc, (const uint8_t*)&sentinel, 4);
             } else {
                 int64_t v = b.col_vals[k][r];
                 int nbytes = (types[k] == RW_T_I32) ? 4 : 8;
                 if (types[k] == RW_T_I32) {
                     int32_t v32 = (int32_t)v;
-                    crc = crc32_bytes(crc, (const uint8_t*)&v32, 4);
+                    crc = crc32_bytes(lut, crc, (const uint8_t*)&v32, 4);
                 } else {
-                    crc = crc32_bytes(crc, (const uint8_t*)&v, 8);
+                    crc = crc32_bytes(lut, crc, (const uint8_t*)&v, 8);
                 }
                 (void)nbytes;
             }

@@ -48,11 +48,18 @@ static thread_local std::string g_xerr;
 
 __constant__ uint32_t gx_crc_table[256];
 
-__device__ __forceinline__ uint32_t xcrc_bytes(uint32_t crc, const uint8_t* p,
-                                               int n) {
+// LDS-staged CRC: divergent indexing of __constant__ memory serializes
+// (each distinct address replays); LDS banks handle it at full rate
+__device__ __forceinline__ uint32_t xcrc_bytes(const uint32_t* lut, uint32_t crc,
+                                               const uint8_t* p, int n) {
     for (int i = 0; i < n; i++)
-        crc = gx_crc_table[(crc ^ p[i]) & 0xFF] ^ (crc >> 8);
+        crc = lut[(crc ^ p[i]) & 0xFF] ^ (crc >> 8);
     return crc;
+}
+
+__device__ __forceinline__ void stage_crc_lut(uint32_t* lut) {
+    for (int i = threadIdx.x; i < 256; i += blockDim.x) lut[i] = gx_crc_table[i];
+    __syncthreads();
 }
 
 struct XBatch {
@@ -67,6 +74,8 @@ __global__ void x_count_kernel(XBatch b, int n_keys, uint32_t k0, uint32_t k1,
                                uint32_t k2, uint32_t k3, uint32_t vnode_count,
                                int n_ranks, uint32_t* dest_of_row,
                                unsigned long long* counts) {
+    __shared__ uint32_t lut[256];
+    stage_crc_lut(lut);
     uint32_t keys[4] = {k0, k1, k2, k3};
     uint32_t stride = gridDim.x * blockDim.x;
     uint32_t per_rank = vnode_count / n_ranks; // contiguous vnode blocks
@@ -77,10 +86,10 @@ __global__ void x_count_kernel(XBatch b, int n_keys, uint32_t k0, uint32_t k1,
             uint32_t col = keys[k];
             if (!b.col_valid[col][r]) {
                 uint32_t sentinel = 0xfffffff0u;
-                crc = xcrc_bytes(crc, (const uint8_t*)&sentinel, 4);
+                crc = xcrc_bytes(lut, crc, (const uint8_t*)&sentinel, 4);
             } else {
                 int64_t v = b.col_vals[col][r];
-                crc = xcrc_bytes(crc, (const uint8_t*)&v, 8);
+                crc = xcrc_bytes(lut, crc, (const uint8_t*)&v, 8);
             }
         }
         uint32_t vn = (uint32_t)((uint64_t)(crc ^ 0xFFFFFFFFu) % vnode_count);

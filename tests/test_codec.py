@@ -1,0 +1,117 @@
+"""Checkpoint spill encoding (§8f-2 foundation): memcomparable keys +
+value-encoded rows, drained from the agg at flush.
+
+Pinning: the memcomparable crate is a Cargo.lock dependency with no vendored
+source; the reference's own test_memcomparable assertions
+(util/memcmp_encoding.rs:346-430) are transcribed here as ordering
+properties, plus explicit byte vectors derived from the published encoding
+(sign-flipped big-endian + null tag; value encoding = presence byte + LE,
+value_encoding/mod.rs:151-215).
+"""
+import ctypes
+import struct
+
+from rwtest import ffi
+from rwtest.ffi import AGG_COUNT_STAR, AGG_SUM, T_I64, from_pretty, oracle
+
+
+def drain(lib, agg):
+    L = lib.lib
+    L.rw_agg_checkpoint_drain.restype = ctypes.c_int
+    L.rw_agg_checkpoint_drain.argtypes = [ctypes.c_void_p,
+                                          ctypes.POINTER(ctypes.c_void_p),
+                                          ctypes.POINTER(ctypes.c_uint64)]
+    L.rw_spill_free.argtypes = [ctypes.c_void_p]
+    buf = ctypes.c_void_p()
+    ln = ctypes.c_uint64()
+    assert L.rw_agg_checkpoint_drain(agg.h, ctypes.byref(buf), ctypes.byref(ln)) == 0
+    data = ctypes.string_at(buf, ln.value)
+    L.rw_spill_free(buf)
+    return parse(data)
+
+
+def parse(data):
+    recs = []
+    off = 0
+    while off < len(data):
+        put = data[off]
+        off += 1
+        klen = struct.unpack_from("<I", data, off)[0]
+        off += 4
+        k = data[off:off + klen]
+        off += klen
+        vlen = struct.unpack_from("<I", data, off)[0]
+        off += 4
+        v = data[off:off + vlen]
+        off += vlen
+        recs.append((put, k, v))
+    return recs
+
+
+def memcmp_i64(v):
+    return b"\x00" + struct.pack(">Q", (v & (2**64 - 1)) ^ (1 << 63))
+
+
+def value_i64(v):
+    return b"\x01" + struct.pack("<q", v)
+
+
+def run_epoch(lib, chunks):
+    calls = [(AGG_COUNT_STAR, -1, T_I64), (AGG_SUM, 1, T_I64)]
+    agg = ffi.HashAgg(lib, [T_I64, T_I64], [0], calls, 0)
+    for c in chunks:
+        agg.push(c)
+    agg.flush(1)
+    agg.poll_all()
+    recs = drain(lib, agg)
+    agg.close()
+    return agg, recs
+
+
+def test_spill_record_bytes():
+    _, recs = run_epoch(oracle(), [from_pretty(" I I\n + 5 7")])
+    assert len(recs) == 1
+    put, k, v = recs[0]
+    assert put == 1
+    assert k == memcmp_i64(5)
+    # value = group key 5, count 1, sum 7 (value-encoded)
+    assert v == value_i64(5) + value_i64(1) + value_i64(7)
+
+
+def test_spill_delete_record():
+    calls = [(AGG_COUNT_STAR, -1, T_I64), (AGG_SUM, 1, T_I64)]
+    agg = ffi.HashAgg(oracle(), [T_I64, T_I64], [0], calls, 0)
+    agg.push(from_pretty(" I I\n + 3 9"))
+    agg.flush(1)
+    agg.poll_all()
+    drain(oracle(), agg)
+    agg.push(from_pretty(" I I\n - 3 9"))
+    agg.flush(2)
+    agg.poll_all()
+    recs = drain(oracle(), agg)
+    agg.close()
+    assert recs == [(0, memcmp_i64(3), b"")]
+
+
+def test_memcmp_ordering_properties():
+    # util/memcmp_encoding.rs:346-390 (ASC NULLS LAST): the encoded bytes
+    # sort like the values, NULL largest
+    _, recs = run_epoch(
+        oracle(),
+        [from_pretty(" I I\n + -1 1\n + 3874 1\n + 45745 1\n "
+                     "+ -9223372036854775808 1\n + 9223372036854775807 1\n + . 1")],
+    )
+    keys = {tuple(r[1]): None for r in recs}
+    ordered = sorted(keys)
+    import struct as st
+
+    def dec(k):
+        kb = bytes(k)
+        if kb[0] == 1:
+            return None
+        return st.unpack(">Q", kb[1:])[0] ^ (1 << 63)
+
+    vals = [dec(k) for k in ordered]
+    nonnull = [v - 2**64 if v >= 2**63 else v for v in vals if v is not None]
+    assert nonnull == sorted(nonnull)
+    assert vals[-1] is None  # NULL sorts largest

@@ -488,3 +488,37 @@ def test_agg_watermark_gpu():
     assert rows_multiset(g.poll_all()) == rows_multiset(o.poll_all())
     g.close()
     o.close()
+
+
+def test_checkpoint_spill_parity():
+    # §8f-2: GPU and oracle drain byte-identical state-table KV deltas
+    # (sorted by memcomparable key — the KV store's view) for the same epochs
+    from test_codec import drain
+
+    rng = np.random.default_rng(51)
+    calls = [(AGG_COUNT_STAR, -1, T_I64), (AGG_SUM, 1, T_I64)]
+    g = ffi.HashAgg(gpu(), [T_I64, T_I64], [0], calls, 0)
+    o = ffi.HashAgg(ffi.oracle(), [T_I64, T_I64], [0], calls, 0)
+    inserted = []
+    for epoch in range(3):
+        n = 1024
+        keys = rng.integers(0, 200, n)
+        vals = rng.integers(1, 1000, n)
+        ops = np.zeros(n, np.uint8)
+        for i in range(n):
+            if inserted and rng.random() < 0.3:
+                jx = int(rng.integers(0, len(inserted)))
+                keys[i], vals[i] = inserted.pop(jx)
+                ops[i] = ffi.OP_DELETE
+            else:
+                inserted.append((int(keys[i]), int(vals[i])))
+        c = mk_chunk([T_I64, T_I64], ops, [keys, vals])
+        for a in (g, o):
+            a.push(c)
+            a.flush(epoch + 1)
+            a.poll_all()
+        rg = sorted(drain(gpu(), g))
+        ro = sorted(drain(ffi.oracle(), o))
+        assert rg == ro, f"epoch {epoch+1}: {len(rg)} vs {len(ro)} records"
+    g.close()
+    o.close()
