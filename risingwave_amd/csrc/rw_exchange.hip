@@ -15,7 +15,9 @@
 #include <rccl/rccl.h>
 
 #include <cstdint>
+#include <cstdio>
 #include <cstdlib>
+#include <ctime>
 #include <cstring>
 #include <string>
 #include <vector>
@@ -243,16 +245,36 @@ int rw_exchange_run(void* h, const int64_t* const* col_vals,
     uint32_t k[4] = {0, 0, 0, 0};
     for (int i = 0; i < n_keys; i++) k[i] = key_cols[i];
 
+    static int dbg = -1;
+    if (dbg < 0) {
+        const char* e = getenv("RW_EXCHANGE_DEBUG");
+        dbg = e && *e == '1';
+    }
+    struct Phase {
+        const char* name;
+        double t;
+    } phases[8];
+    int np = 0;
+    auto mark = [&](const char* name) {
+        if (!dbg) return;
+        hipStreamSynchronize(x->stream);
+        struct timespec ts;
+        clock_gettime(CLOCK_MONOTONIC, &ts);
+        phases[np++] = {name, ts.tv_sec * 1e3 + ts.tv_nsec / 1e6};
+    };
+    mark("start");
     hipEvent_t e0 = x->e0, e1 = x->e1;
     XHIP(hipEventRecord(e0, x->stream));
 
     x_count_kernel<<<blocks, 256, 0, x->stream>>>(b, n_keys, k[0], k[1], k[2],
                                                   k[3], vnode_count, R, d_dest,
                                                   d_counts);
+    mark("count_kernel");
     unsigned long long counts[64];
     XHIP(hipMemcpyAsync(counts, d_counts, R * 8, hipMemcpyDeviceToHost,
                         x->stream));
     XHIP(hipStreamSynchronize(x->stream));
+    mark("counts_d2h");
 
     // byte offsets of per-destination blocks in send_buf
     uint64_t row_bytes = 1 + (uint64_t)n_cols * 9; // op + per col valid+val
@@ -268,6 +290,7 @@ int rw_exchange_run(void* h, const int64_t* const* col_vals,
                         x->stream));
     x_scatter_kernel<<<blocks, 256, 0, x->stream>>>(b, n_cols, d_dest, d_offsets,
                                                     d_counts, d_cursors, send_buf);
+    mark("scatter");
 
     // exchange per-peer row counts, then the payload blocks (all-to-all-v)
     // over RCCL/xGMI. A single-rank communicator is a degenerate self-loop:
@@ -318,8 +341,14 @@ int rw_exchange_run(void* h, const int64_t* const* col_vals,
         }
         XNCCL(ncclGroupEnd());
     }
+    mark("exchange");
     XHIP(hipEventRecord(e1, x->stream));
     XHIP(hipStreamSynchronize(x->stream));
+    if (dbg && x->exch_launches < 3) {
+        for (int i = 1; i < np; i++)
+            fprintf(stderr, "# exch %s: %.3f ms\n", phases[i].name,
+                    phases[i].t - phases[i - 1].t);
+    }
     float ms = 0;
     hipEventElapsedTime(&ms, e0, e1);
     x->exch_ms += ms;
