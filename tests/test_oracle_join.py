@@ -340,3 +340,65 @@ def test_join_watermark_state_cleaning():
     j.watermark(SIDE_RIGHT, 0, 5)  # selected = 5: left row (2,1) cleaned
     assert push(j, SIDE_RIGHT, " I I\n + 2 10\n + 6 11") == rows([("+", 6, 2, 6, 11)])
     j.close()
+
+
+def test_right_anti_join():
+    # hash_join.rs:2650-2777 (LeftAnti with swapped senders)
+    j = classical(ffi.JOIN_RIGHT_ANTI)
+    assert push(j, SIDE_RIGHT, " I I\n + 1 4\n + 2 5\n + 3 6") == rows(
+        [("+", 1, 4), ("+", 2, 5), ("+", 3, 6)]
+    )
+    assert push(j, SIDE_RIGHT, " I I\n + 3 8\n - 3 8") == []
+    assert push(j, SIDE_LEFT, " I I\n + 2 7\n + 4 8\n + 6 9") == rows([("-", 2, 5)])
+    assert push(j, SIDE_LEFT, " I I\n + 3 10\n + 6 11\n + 1 2\n + 1 3") == rows(
+        [("-", 3, 6), ("-", 1, 4)]
+    )
+    assert push(j, SIDE_RIGHT, " I I\n + 9 10") == rows([("+", 9, 10)])
+    assert push(j, SIDE_LEFT, " I I\n - 1 2") == []
+    assert push(j, SIDE_LEFT, " I I\n - 1 3") == rows([("+", 1, 4)])
+    j.close()
+
+
+def test_left_outer_join_append_only():
+    # hash_join.rs:3206-3292
+    j = append_only(JOIN_LEFT_OUTER)
+    assert push(j, SIDE_LEFT, " I I I\n + 1 4 1\n + 2 5 2\n + 3 6 3") == [
+        ("+", (1, 4, 1, None, None, None)),
+        ("+", (2, 5, 2, None, None, None)),
+        ("+", (3, 6, 3, None, None, None)),
+    ]
+    assert push(j, SIDE_LEFT, " I I I\n + 4 9 4\n + 5 10 5") == [
+        ("+", (4, 9, 4, None, None, None)),
+        ("+", (5, 10, 5, None, None, None)),
+    ]
+    assert push(j, SIDE_RIGHT, " I I I\n + 2 5 1\n + 4 9 2\n + 6 9 3") == [
+        ("-", (2, 5, 2, None, None, None)),
+        ("+", (2, 5, 2, 2, 5, 1)),
+        ("-", (4, 9, 4, None, None, None)),
+        ("+", (4, 9, 4, 4, 9, 2)),
+    ]
+    assert push(j, SIDE_RIGHT, " I I I\n + 1 4 4\n + 3 6 5") == [
+        ("-", (1, 4, 1, None, None, None)),
+        ("+", (1, 4, 1, 1, 4, 4)),
+        ("-", (3, 6, 3, None, None, None)),
+        ("+", (3, 6, 3, 3, 6, 5)),
+    ]
+    j.close()
+
+
+def test_right_outer_join_append_only():
+    # hash_join.rs:3295-3363
+    j = append_only(JOIN_RIGHT_OUTER)
+    assert push(j, SIDE_LEFT, " I I I\n + 1 4 1\n + 2 5 2\n + 3 6 3") == []
+    assert push(j, SIDE_LEFT, " I I I\n + 4 9 4\n + 5 10 5") == []
+    assert push(j, SIDE_RIGHT, " I I I\n + 2 5 1\n + 4 9 2\n + 6 9 3") == [
+        ("+", (2, 5, 2, 2, 5, 1)),
+        ("+", (4, 9, 4, 4, 9, 2)),
+        ("+", (None, None, None, 6, 9, 3)),
+    ]
+    assert push(j, SIDE_RIGHT, " I I I\n + 1 4 4\n + 3 6 5\n + 7 7 6") == [
+        ("+", (1, 4, 1, 1, 4, 4)),
+        ("+", (3, 6, 3, 3, 6, 5)),
+        ("+", (None, None, None, 7, 7, 6)),
+    ]
+    j.close()
