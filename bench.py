@@ -107,29 +107,14 @@ def cpu_baseline(ffi, rng, target_seconds=10.0):
     agg.close()
     n = max(8, min(int(target_seconds / max(per_chunk, 1e-9)), 400_000))
 
-    def worker(tid, out):
-        a = ffi.HashAgg(oracle(), [T_I64, T_I64], [0], calls, 1,
-                        append_only=True)
-        # rank-local window space, as the vnode exchange would route it
-        c = make_q7_chunk(ffi, np.random.default_rng(100 + tid), CHUNK_ROWS,
-                          tid * 1_000_000 * WINDOW_US, 32)
-        a.push(c)  # warm
-        for i in range(n):
-            a.push(c)
-            if (i + 1) % 64 == 0:
-                a.flush(i)
-                a.poll_all()
-        a.close()
-        out[tid] = n * CHUNK_ROWS
+    del threading  # processes, not threads: the per-chunk ctypes dispatch is
+    # GIL-bound at 4K-row chunks (8 threads measured BELOW one core)
+    import multiprocessing as mp
 
-    done = [0] * ncores
-    threads = [threading.Thread(target=worker, args=(t, done))
-               for t in range(ncores)]
+    ctx = mp.get_context("fork")
     t0 = time.perf_counter()
-    for t in threads:
-        t.start()
-    for t in threads:
-        t.join()
+    with ctx.Pool(ncores) as pool:
+        done = pool.map(_cpu_baseline_worker, [(tid, n) for tid in range(ncores)])
     dt = time.perf_counter() - t0
     rows = sum(done)
     return {
@@ -137,9 +122,29 @@ def cpu_baseline(ffi, rng, target_seconds=10.0):
         "unit": "rows/s",
         "cores": ncores,
         "kind": "port",
-        "sample": f"{rows} q7 rows across {ncores} executor threads "
-                  f"({dt:.1f}s, flush every 64 chunks per thread)",
+        "sample": f"{rows} q7 rows across {ncores} executor processes "
+                  f"({dt:.1f}s, flush every 64 chunks per process; "
+                  f"fork overhead included)",
     }
+
+
+def _cpu_baseline_worker(arg):
+    tid, n = arg
+    from rwtest import ffi
+    from rwtest.ffi import AGG_COUNT_STAR, AGG_MAX, T_I64, oracle
+
+    calls = [(AGG_MAX, 1, T_I64), (AGG_COUNT_STAR, -1, T_I64)]
+    a = ffi.HashAgg(oracle(), [T_I64, T_I64], [0], calls, 1, append_only=True)
+    c = make_q7_chunk(ffi, np.random.default_rng(100 + tid), CHUNK_ROWS,
+                      tid * 1_000_000 * WINDOW_US, 32)
+    a.push(c)  # warm
+    for i in range(n):
+        a.push(c)
+        if (i + 1) % 64 == 0:
+            a.flush(i)
+            a.poll_all()
+    a.close()
+    return n * CHUNK_ROWS
 
 
 def main():

@@ -98,21 +98,47 @@ __global__ void x_count_kernel(XBatch b, int n_keys, uint32_t k0, uint32_t k1,
         uint32_t dest = vn / per_rank;
         if (dest >= (uint32_t)n_ranks) dest = n_ranks - 1;
         dest_of_row[r] = dest;
-        atomicAdd(&counts[dest], 1ull);
+        // wave-aggregated counting: one atomic per (wave, destination) —
+        // a single hot counter serializes at ~88 adds/us otherwise
+        int lane = threadIdx.x & 63;
+        for (int d = 0; d < n_ranks; d++) {
+            uint64_t mask = __ballot(dest == (uint32_t)d);
+            if (mask && lane == (63 - __clzll(mask)))
+                atomicAdd(&counts[d], (unsigned long long)__popcll(mask));
+        }
     }
 }
 
 // pass 2: scatter rows into dense per-destination blocks
-__global__ void x_scatter_kernel(XBatch b, int n_cols, const uint32_t* dest_of_row,
+__global__ void x_scatter_kernel(XBatch b, int n_cols, int n_ranks,
+                                 const uint32_t* dest_of_row,
                                  const unsigned long long* offsets, // [n_ranks]
                                  const unsigned long long* counts,  // [n_ranks]
                                  unsigned long long* cursors,       // [n_ranks]
                                  uint8_t* out /* packed payload */) {
     uint32_t stride = gridDim.x * blockDim.x;
-    for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < b.n_rows;
-         r += stride) {
-        uint32_t dest = dest_of_row[r];
-        unsigned long long idx = atomicAdd(&cursors[dest], 1ull);
+    uint32_t iters = (b.n_rows + stride - 1) / stride;
+    int lane = threadIdx.x & 63;
+    for (uint32_t it = 0; it < iters; it++) {
+        uint32_t r = it * stride + blockIdx.x * blockDim.x + threadIdx.x;
+        bool active = r < b.n_rows;
+        uint32_t dest = active ? dest_of_row[r] : 0xFFFFFFFFu;
+        // wave-aggregated reservation: one atomic per (wave, destination);
+        // same-dest lanes get consecutive slots (coalesced scatter writes)
+        unsigned long long idx = 0;
+        for (int d = 0; d < n_ranks; d++) {
+            uint64_t mask = __ballot(dest == (uint32_t)d);
+            if (!mask) continue;
+            unsigned long long base = 0;
+            int leader = 63 - __clzll(mask);
+            if (lane == leader)
+                base = atomicAdd(&cursors[d], (unsigned long long)__popcll(mask));
+            base = (unsigned long long)__shfl((long long)base, leader);
+            if (dest == (uint32_t)d)
+                idx = base + (unsigned long long)__popcll(mask &
+                                                          ((1ULL << lane) - 1));
+        }
+        if (!active) continue;
         unsigned long long base = offsets[dest]; // byte offset (8-aligned)
         unsigned long long n = counts[dest];
         // block layout (8-aligned base): vals[col][n]*8 ∥ valid[col][n] ∥ ops[n]
@@ -126,6 +152,7 @@ __global__ void x_scatter_kernel(XBatch b, int n_cols, const uint32_t* dest_of_r
         ops[idx] = b.ops[r];
     }
 }
+
 
 struct Exchange {
     ncclComm_t comm = nullptr;
@@ -288,8 +315,9 @@ int rw_exchange_run(void* h, const int64_t* const* col_vals,
     if (off > send_cap) XFAIL(-2, "send buffer too small (%llu)", (unsigned long long)off);
     XHIP(hipMemcpyAsync(d_offsets, offsets, R * 8, hipMemcpyHostToDevice,
                         x->stream));
-    x_scatter_kernel<<<blocks, 256, 0, x->stream>>>(b, n_cols, d_dest, d_offsets,
-                                                    d_counts, d_cursors, send_buf);
+    x_scatter_kernel<<<blocks, 256, 0, x->stream>>>(b, n_cols, R, d_dest,
+                                                    d_offsets, d_counts,
+                                                    d_cursors, send_buf);
     mark("scatter");
 
     // exchange per-peer row counts, then the payload blocks (all-to-all-v)
