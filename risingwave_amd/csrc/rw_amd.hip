@@ -1918,21 +1918,124 @@ int rw_dispatch_compute(const RwDispatchDesc* d, const RwChunk* chunk,
 #define MAX_COLS 8
 #define MAX_OUT 16
 
+// AoS layouts: one 64-B line per key slot, one packed record per row —
+// a random probe touches 1 table line + 1 record line instead of the 8–10
+// lines the SoA arrays cost (DESIGN §8 lever).
+struct alignas(64) JoinSlot {
+    uint32_t state; // SLOT_EMPTY/CLAIMED/READY
+    uint32_t head;  // chain head row index, UINT32_MAX = none
+    uint32_t nulls; // key null mask
+    uint32_t _pad;
+    long long key[4];
+};
+static_assert(sizeof(JoinSlot) == 64, "one cache line per slot");
+
+// row record header; vals[n_cols] i64 follow at offset 16
+struct JoinRowHdr {
+    uint32_t alive; // CAS-claimed tombstone
+    uint32_t next;
+    uint32_t validbits; // bit c = column c non-NULL
+    uint32_t _pad;
+};
+
 struct JoinSideDev {
-    // key table
-    uint32_t* state;
-    int64_t* keys;      // [cap * KW]
-    uint32_t* key_nulls; // [cap]
-    uint32_t* head;     // [cap] chain head row index, UINT32_MAX = none
+    JoinSlot* slots;
     uint32_t cap_mask;
-    // row store (SoA)
-    int64_t* col_vals[MAX_COLS];
-    uint8_t* col_valid[MAX_COLS];
-    uint32_t* next;  // [row_cap]
-    uint32_t* alive; // [row_cap] 1 = alive (u32 for CAS-claimed deletes)
+    uint8_t* rows;       // packed records, row_stride bytes each (16-B aligned)
+    uint32_t row_stride;  // 16 + 8*n_cols
     uint32_t* row_cursor; // single counter
     uint32_t row_cap;
 };
+
+__device__ __forceinline__ JoinRowHdr* jrow(const JoinSideDev& s, uint32_t r) {
+    return (JoinRowHdr*)(s.rows + (size_t)r * s.row_stride);
+}
+__device__ __forceinline__ long long* jvals(JoinRowHdr* h) {
+    return (long long*)((uint8_t*)h + 16);
+}
+
+// own-side find-or-insert with the R1 sc1 protocol on the slot fields
+__device__ __forceinline__ uint32_t jslot_find_or_insert(JoinSlot* slots,
+                                                         uint32_t cap_mask,
+                                                         const int64_t* kw,
+                                                         uint32_t nullmask,
+                                                         int KW) {
+    uint32_t slot = (uint32_t)(hash_key(kw, nullmask, KW) & cap_mask);
+    for (uint32_t probes = 0; probes <= cap_mask; probes++) {
+        JoinSlot* sl = &slots[slot];
+        uint32_t st = ld_u32(&sl->state);
+        if (st == SLOT_EMPTY) {
+            uint32_t prev = atomicCAS(&sl->state, SLOT_EMPTY, SLOT_CLAIMED);
+            if (prev == SLOT_EMPTY) {
+                for (int i = 0; i < KW; i++) st_i64((int64_t*)&sl->key[i], kw[i]);
+                st_u32(&sl->nulls, nullmask);
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); // R1 drain
+                st_u32(&sl->state, SLOT_READY);
+                return slot;
+            }
+            st = prev;
+        }
+        while (st == SLOT_CLAIMED) {
+            __builtin_amdgcn_s_sleep(1);
+            st = ld_u32(&sl->state);
+        }
+        bool eq = ld_u32(&sl->nulls) == nullmask;
+        for (int i = 0; eq && i < KW; i++)
+            eq = ld_i64((const int64_t*)&sl->key[i]) == kw[i];
+        if (eq) return slot;
+        slot = (slot + 1) & cap_mask;
+    }
+    return (uint32_t)-1;
+}
+
+// match-side find with plain cached loads (immutable during the launch)
+__device__ __forceinline__ uint32_t jslot_find_cached(const JoinSlot* slots,
+                                                      uint32_t cap_mask,
+                                                      const int64_t* kw,
+                                                      uint32_t nullmask, int KW) {
+    uint32_t slot = (uint32_t)(hash_key(kw, nullmask, KW) & cap_mask);
+    for (uint32_t probes = 0; probes <= cap_mask; probes++) {
+        const JoinSlot* sl = &slots[slot];
+        uint32_t st = sl->state;
+        if (st == SLOT_EMPTY) return (uint32_t)-1;
+        if (st == SLOT_READY) {
+            bool eq = sl->nulls == nullmask;
+            for (int i = 0; eq && i < KW; i++) eq = sl->key[i] == kw[i];
+            if (eq) return slot;
+        }
+        slot = (slot + 1) & cap_mask;
+    }
+    return (uint32_t)-1;
+}
+
+// sc1 variant of the find (append-only path mutates the match side)
+__device__ __forceinline__ uint32_t jslot_find_sc1(const JoinSlot* slots,
+                                                   uint32_t cap_mask,
+                                                   const int64_t* kw,
+                                                   uint32_t nullmask, int KW) {
+    uint32_t slot = (uint32_t)(hash_key(kw, nullmask, KW) & cap_mask);
+    for (uint32_t probes = 0; probes <= cap_mask; probes++) {
+        const JoinSlot* sl = &slots[slot];
+        uint32_t st = ld_u32(&sl->state);
+        if (st == SLOT_EMPTY) return (uint32_t)-1;
+        if (st == SLOT_READY) {
+            bool eq = ld_u32(&sl->nulls) == nullmask;
+            for (int i = 0; eq && i < KW; i++)
+                eq = ld_i64((const int64_t*)&sl->key[i]) == kw[i];
+            if (eq) return slot;
+        }
+        slot = (slot + 1) & cap_mask;
+    }
+    return (uint32_t)-1;
+}
+
+__global__ void jslot_init_kernel(JoinSlot* slots, size_t cap) {
+    size_t stride = (size_t)gridDim.x * blockDim.x;
+    for (size_t i = blockIdx.x * blockDim.x + threadIdx.x; i < cap; i += stride) {
+        slots[i].state = SLOT_EMPTY;
+        slots[i].head = UINT32_MAX;
+    }
+}
 
 struct JoinMeta {
     int KW;
@@ -1989,16 +2092,16 @@ __device__ __forceinline__ uint32_t table_find_cached(
 
 __device__ __forceinline__ bool join_cond_ok(const JoinMeta& m, int probe_side,
                                              const JoinBatchDev& b, uint32_t r,
-                                             const JoinSideDev& match,
-                                             uint32_t mrow) {
+                                             uint32_t validbits,
+                                             const long long* mvals) {
     if (!m.has_cond) return true;
     auto fetch = [&](uint8_t src, uint8_t col, int64_t* v) -> bool {
         if ((int)src == probe_side) {
             if (!b.col_valid[col][r]) return false;
             *v = b.col_vals[col][r];
         } else {
-            if (!match.col_valid[col][mrow]) return false;
-            *v = match.col_vals[col][mrow];
+            if (!((validbits >> col) & 1)) return false;
+            *v = mvals[col];
         }
         return true;
     };
@@ -2015,12 +2118,10 @@ __device__ __forceinline__ bool join_cond_ok(const JoinMeta& m, int probe_side,
 }
 
 // probe the match side + update own side, one thread per probe row.
-// Emission reserves output rows with ONE wave-level atomicAdd per iteration
-// (per-lane match counts -> wave prefix sum -> lane-0 cursor add), then a
-// second chain walk writes the rows: the match side's state is immutable
-// during a non-append-only launch (probes read the OTHER side; own-side
-// writes never alias it), so the two walks agree. The append-only path
-// mutates the match side (kill-on-match) and keeps per-match atomics.
+// Emission reserves output rows with ONE wave-level atomicAdd per iteration;
+// the match side is immutable during a non-append-only launch, so it is read
+// with plain cached loads (one line per slot, one per record). The
+// append-only path (in-launch kill-on-match) uses the sc1 protocol.
 __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
                                   JoinSideDev match, JoinMeta m, int S,
                                   JoinOutDev out, uint32_t r0, uint32_t r1) {
@@ -2044,8 +2145,7 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
                 kw[i] = valid ? b.col_vals[col][r] : 0;
                 nullmask |= (!valid) << i;
             }
-            // null-safe NeverMatch (hash_join.rs:1004-1016): inner join
-            // forwards nothing and writes no state
+            // null-safe NeverMatch (hash_join.rs:1004-1016)
             if (nullmask & ~(uint32_t)m.null_safe_mask) active = false;
         }
 
@@ -2054,22 +2154,22 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
         uint32_t matched_row = UINT32_MAX;
         if (active) {
             if (m.append_only) {
-                // append-only mutates the match side in-launch: sc1 path
-                mslot = table_find(match.state, match.keys, match.key_nulls,
-                                   match.cap_mask, kw, nullmask, m.KW);
+                mslot = jslot_find_sc1(match.slots, match.cap_mask, kw, nullmask,
+                                       m.KW);
             } else {
-                mslot = table_find_cached(match.state, match.keys,
-                                          match.key_nulls, match.cap_mask, kw,
+                mslot = jslot_find_cached(match.slots, match.cap_mask, kw,
                                           nullmask, m.KW);
             }
             if (mslot != UINT32_MAX && !m.append_only) {
-                uint32_t row = match.head[mslot];
+                uint32_t row = match.slots[mslot].head;
                 while (row != UINT32_MAX) {
-                    if (match.alive[row] && join_cond_ok(m, S, b, r, match, row)) {
+                    JoinRowHdr* h = jrow(match, row);
+                    if (h->alive &&
+                        join_cond_ok(m, S, b, r, h->validbits, jvals(h))) {
                         my_n++;
                         matched_row = row;
                     }
-                    row = match.next[row];
+                    row = h->next;
                 }
             }
         }
@@ -2090,15 +2190,16 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
             if (total && base + total > out.cap) {
                 if (lane == 0) atomicExch(&out.counters[1], 1u); // overflow
             } else if (my_n) {
-                // second walk: emit (JoinStreamChunkBuilder::append_row,
-                // join/builder.rs:87-106)
-                uint32_t row = match.head[mslot];
+                // second walk: emit (JoinStreamChunkBuilder::append_row)
+                uint32_t row = match.slots[mslot].head;
                 uint32_t k = 0;
                 while (row != UINT32_MAX && k < my_n) {
-                    if (match.alive[row] &&
-                        join_cond_ok(m, S, b, r, match, row)) {
+                    JoinRowHdr* h = jrow(match, row);
+                    if (h->alive &&
+                        join_cond_ok(m, S, b, r, h->validbits, jvals(h))) {
                         uint32_t orow = my_base + k;
                         out.ops[orow] = op;
+                        const long long* mv = jvals(h);
                         for (int c = 0; c < m.n_out; c++) {
                             bool from_probe = (int)m.out_src[c] == S;
                             uint8_t col = m.out_col[c];
@@ -2108,30 +2209,30 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
                                 valid = b.col_valid[col][r];
                                 v = b.col_vals[col][r];
                             } else {
-                                valid = match.col_valid[col][row];
-                                v = match.col_vals[col][row];
+                                valid = (h->validbits >> col) & 1;
+                                v = mv[col];
                             }
                             out.vals[(size_t)orow * m.n_out + c] = valid ? v : 0;
                             out.nulls[(size_t)orow * m.n_out + c] = !valid;
                         }
                         k++;
                     }
-                    row = match.next[row];
+                    row = h->next;
                 }
             }
-        }
-        if (m.append_only && active && mslot != UINT32_MAX) {
-            // append-only path: <=1 match (jk superset of pk); per-match
-            // cursor atomics, single walk with kill
-            uint32_t row = ld_u32(&match.head[mslot]);
+        } else if (active && mslot != UINT32_MAX) {
+            // append-only path: <=1 match; sc1 reads, per-match atomics
+            uint32_t row = ld_u32(&match.slots[mslot].head);
             while (row != UINT32_MAX) {
-                if (ld_u32(&match.alive[row]) &&
-                    join_cond_ok(m, S, b, r, match, row)) {
+                JoinRowHdr* h = jrow(match, row);
+                uint32_t vb = ld_u32(&h->validbits);
+                if (ld_u32(&h->alive) && join_cond_ok(m, S, b, r, vb, jvals(h))) {
                     uint32_t orow = atomicAdd(&out.counters[0], 1u);
                     if (orow >= out.cap) {
                         atomicExch(&out.counters[1], 1u);
                     } else {
                         out.ops[orow] = op;
+                        const long long* mv = jvals(h);
                         for (int c = 0; c < m.n_out; c++) {
                             bool from_probe = (int)m.out_src[c] == S;
                             uint8_t col = m.out_col[c];
@@ -2141,8 +2242,8 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
                                 valid = b.col_valid[col][r];
                                 v = b.col_vals[col][r];
                             } else {
-                                valid = match.col_valid[col][row];
-                                v = match.col_vals[col][row];
+                                valid = (vb >> col) & 1;
+                                v = (int64_t)ld_i64((const int64_t*)&mv[col]);
                             }
                             out.vals[(size_t)orow * m.n_out + c] = valid ? v : 0;
                             out.nulls[(size_t)orow * m.n_out + c] = !valid;
@@ -2150,24 +2251,22 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
                     }
                     matched_row = row;
                 }
-                row = match.next[row];
+                row = ld_u32(&h->next);
             }
         }
 
         if (!active) continue;
 
         if (m.append_only && is_insert && matched_row != UINT32_MAX) {
-            // append-only optimize: delete the single matched row and skip
-            // own insert (hash_join.rs:1241-1245)
-            st_u32(&match.alive[matched_row], 0);
+            // append-only optimize (hash_join.rs:1241-1245)
+            st_u32(&jrow(match, matched_row)->alive, 0);
             continue;
         }
 
         // own-side state update (join/hash_join.rs:591-681 without LRU tier)
         if (is_insert) {
-            uint32_t own_slot =
-                table_find_or_insert(own.state, own.keys, own.key_nulls,
-                                     own.cap_mask, kw, nullmask, m.KW);
+            uint32_t own_slot = jslot_find_or_insert(own.slots, own.cap_mask, kw,
+                                                     nullmask, m.KW);
             if (own_slot == UINT32_MAX) {
                 atomicExch(&out.counters[1], 2u); // key table full
                 continue;
@@ -2177,42 +2276,48 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
                 atomicExch(&out.counters[1], 3u); // row store full
                 continue;
             }
+            JoinRowHdr* h = jrow(own, row);
+            uint32_t vb = 0;
+            long long* hv = jvals(h);
             for (int c = 0; c < m.n_cols[S]; c++) {
-                st_i64(&own.col_vals[c][row], b.col_vals[c][r]);
-                own.col_valid[c][row] = b.col_valid[c][r];
+                st_i64((int64_t*)&hv[c], b.col_vals[c][r]);
+                vb |= (uint32_t)(b.col_valid[c][r] != 0) << c;
             }
-            st_u32(&own.alive[row], 1);
-            // lock-free chain push: set next BEFORE publishing the row (a
-            // same-launch delete may walk this chain), R1-draining the row
-            // payload ahead of the CAS publish
-            uint32_t old_head = ld_u32(&own.head[own_slot]);
+            st_u32(&h->validbits, vb);
+            st_u32(&h->alive, 1);
+            // lock-free chain push: next set BEFORE the CAS publish, payload
+            // R1-drained ahead of it (same-launch deletes may walk this)
+            uint32_t* headp = &own.slots[own_slot].head;
+            uint32_t old_head = ld_u32(headp);
             for (;;) {
-                st_u32(&own.next[row], old_head);
+                st_u32(&h->next, old_head);
                 asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); // R1 drain
-                uint32_t prev = atomicCAS(&own.head[own_slot], old_head, row);
+                uint32_t prev = atomicCAS(headp, old_head, row);
                 if (prev == old_head) break;
                 old_head = prev;
             }
         } else {
-            // delete own row (join/hash_join.rs:659-681 deletes by deduped
-            // pk; FULL-row compare keeps same-chunk U-pairs race-free, and
-            // the CAS claim makes two identical deletes kill two distinct
-            // identical rows)
-            uint32_t own_slot = table_find(own.state, own.keys, own.key_nulls,
-                                           own.cap_mask, kw, nullmask, m.KW);
+            // delete own row: FULL-row compare + CAS claim
+            uint32_t own_slot = jslot_find_sc1(own.slots, own.cap_mask, kw,
+                                               nullmask, m.KW);
             if (own_slot == UINT32_MAX) continue;
-            uint32_t row = ld_u32(&own.head[own_slot]);
+            uint32_t row = ld_u32(&own.slots[own_slot].head);
             while (row != UINT32_MAX) {
-                if (ld_u32(&own.alive[row])) {
+                JoinRowHdr* h = jrow(own, row);
+                if (ld_u32(&h->alive)) {
+                    uint32_t vb = ld_u32(&h->validbits);
+                    const long long* hv = jvals(h);
                     bool eq = true;
                     for (int c = 0; eq && c < m.n_cols[S]; c++) {
-                        uint8_t va = b.col_valid[c][r], vb = own.col_valid[c][row];
-                        eq = (va == vb) &&
-                             (!va || b.col_vals[c][r] == ld_i64(&own.col_vals[c][row]));
+                        uint8_t va = b.col_valid[c][r];
+                        uint8_t vbc = (vb >> c) & 1;
+                        eq = (va == vbc) &&
+                             (!va ||
+                              b.col_vals[c][r] == ld_i64((const int64_t*)&hv[c]));
                     }
-                    if (eq && atomicCAS(&own.alive[row], 1u, 0u) == 1u) break;
+                    if (eq && atomicCAS(&h->alive, 1u, 0u) == 1u) break;
                 }
-                row = ld_u32(&own.next[row]);
+                row = ld_u32(&h->next);
             }
         }
     }
@@ -2223,19 +2328,22 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
 // retired in place (slots stay READY so linear probing is undisturbed;
 // memory reclamation is compaction work for a later round)
 __global__ void join_clean_kernel(JoinSideDev sd, int kpos, long long wm, int KW) {
+    (void)KW;
     size_t cap = (size_t)sd.cap_mask + 1;
     size_t stride = (size_t)gridDim.x * blockDim.x;
     for (size_t slot = blockIdx.x * blockDim.x + threadIdx.x; slot < cap;
          slot += stride) {
-        if (ld_u32(&sd.state[(uint32_t)slot]) != SLOT_READY) continue;
-        if ((ld_u32(&sd.key_nulls[(uint32_t)slot]) >> kpos) & 1) continue; // NULLs largest
-        if (ld_i64((const int64_t*)&sd.keys[slot * KW + kpos]) >= wm) continue;
-        uint32_t row = ld_u32(&sd.head[(uint32_t)slot]);
+        JoinSlot* sl = &sd.slots[slot];
+        if (ld_u32(&sl->state) != SLOT_READY) continue;
+        if ((ld_u32(&sl->nulls) >> kpos) & 1) continue; // NULLs largest
+        if (ld_i64((const int64_t*)&sl->key[kpos]) >= wm) continue;
+        uint32_t row = ld_u32(&sl->head);
         while (row != UINT32_MAX) {
-            st_u32(&sd.alive[row], 0);
-            row = ld_u32(&sd.next[row]);
+            JoinRowHdr* h = jrow(sd, row);
+            st_u32(&h->alive, 0);
+            row = ld_u32(&h->next);
         }
-        st_u32(&sd.head[(uint32_t)slot], UINT32_MAX);
+        st_u32(&sl->head, UINT32_MAX);
     }
 }
 
@@ -2363,22 +2471,15 @@ struct HashJoin {
             while (cap < key_cap * 2) cap <<= 1; // ≤50% load factor
             JoinSideDev& js = side[s];
             js.cap_mask = cap - 1;
-            HIP_TRY(hipMalloc(&js.state, (size_t)cap * 4));
-            HIP_TRY(hipMemset(js.state, 0, (size_t)cap * 4));
-            HIP_TRY(hipMalloc(&js.keys, (size_t)cap * m.KW * 8));
-            HIP_TRY(hipMalloc(&js.key_nulls, (size_t)cap * 4));
-            HIP_TRY(hipMalloc(&js.head, (size_t)cap * 4));
-            HIP_TRY(hipMemset(js.head, 0xFF, (size_t)cap * 4));
+            HIP_TRY(hipMalloc(&js.slots, (size_t)cap * sizeof(JoinSlot)));
+            jslot_init_kernel<<<2048, 256, 0, stream>>>(js.slots, cap);
             js.row_cap = (uint32_t)row_cap;
-            for (int c = 0; c < m.n_cols[s]; c++) {
-                HIP_TRY(hipMalloc(&js.col_vals[c], row_cap * 8));
-                HIP_TRY(hipMalloc(&js.col_valid[c], row_cap));
-            }
-            HIP_TRY(hipMalloc(&js.next, row_cap * 4));
-            HIP_TRY(hipMalloc(&js.alive, row_cap * 4));
+            js.row_stride = 16 + 8 * (uint32_t)m.n_cols[s];
+            HIP_TRY(hipMalloc(&js.rows, (size_t)row_cap * js.row_stride));
             HIP_TRY(hipMalloc(&js.row_cursor, 4));
             HIP_TRY(hipMemset(js.row_cursor, 0, 4));
         }
+        HIP_TRY(hipStreamSynchronize(stream));
         out.cap = 1u << 22;
         HIP_TRY(hipMalloc(&out.vals, (size_t)out.cap * m.n_out * 8));
         HIP_TRY(hipMalloc(&out.nulls, (size_t)out.cap * m.n_out));
@@ -2628,17 +2729,9 @@ struct HashJoin {
     ~HashJoin() {
         for (int s = 0; s < 2; s++) {
             JoinSideDev& js = side[s];
-            if (js.state) {
-                hipFree(js.state);
-                hipFree(js.keys);
-                hipFree(js.key_nulls);
-                hipFree(js.head);
-                for (int c = 0; c < m.n_cols[s]; c++) {
-                    hipFree(js.col_vals[c]);
-                    hipFree(js.col_valid[c]);
-                }
-                hipFree(js.next);
-                hipFree(js.alive);
+            if (js.slots) {
+                hipFree(js.slots);
+                hipFree(js.rows);
                 hipFree(js.row_cursor);
             }
             for (int c = 0; c < m.n_cols[s]; c++) {
