@@ -942,9 +942,31 @@ struct HashAgg {
     // staging (host-pinned mirrors reused per push)
     AggBatch stage{};
     uint32_t stage_cap = 0;
-    // kernel timing (HIP events on this executor's own stream)
+    // kernel timing (HIP events on this executor's own stream). Event pairs
+    // live in a ring and are read back lazily — a per-step
+    // hipEventSynchronize costs a full host round-trip (~45 us, 3x the
+    // kernel itself at q7 sizes), so the timed region must stay async.
+    static constexpr int EV_RING = 256;
+    hipEvent_t ev0[EV_RING] = {}, ev1[EV_RING] = {};
+    uint8_t ev_pending[EV_RING] = {};
+    uint64_t ev_head = 0;
     double apply_ms_total = 0;
     uint64_t apply_launches = 0, apply_rows = 0;
+
+    int ev_harvest(int slot) {
+        if (!ev_pending[slot]) return RW_OK;
+        HIP_TRY(hipEventSynchronize(ev1[slot]));
+        float ms = 0;
+        HIP_TRY(hipEventElapsedTime(&ms, ev0[slot], ev1[slot]));
+        apply_ms_total += ms;
+        ev_pending[slot] = 0;
+        return RW_OK;
+    }
+    int ev_harvest_all() {
+        for (int s = 0; s < EV_RING; s++)
+            if (ev_harvest(s) != RW_OK) return RW_E_INTERNAL;
+        return RW_OK;
+    }
     // pending outputs
     std::vector<RwChunk*> outq;
     // checkpoint spill buffer (§8f-2): state-table KV deltas, accumulated at
@@ -1206,11 +1228,17 @@ struct HashAgg {
 
     int apply(const AggBatch& b, bool timed,
               const std::vector<uint32_t>& seg_bounds = {}) {
-        hipEvent_t e0 = nullptr, e1 = nullptr;
+        int slot = -1;
         if (timed) {
-            HIP_TRY(hipEventCreate(&e0));
-            HIP_TRY(hipEventCreate(&e1));
-            HIP_TRY(hipEventRecord(e0, stream));
+            slot = (int)(ev_head++ % EV_RING);
+            if (!ev0[slot]) {
+                HIP_TRY(hipEventCreate(&ev0[slot]));
+                HIP_TRY(hipEventCreate(&ev1[slot]));
+            }
+            // reclaim the slot's previous measurement (long complete by the
+            // time the ring wraps) without stalling the current step
+            if (ev_harvest(slot) != RW_OK) return RW_E_INTERNAL;
+            HIP_TRY(hipEventRecord(ev0[slot], stream));
         }
         uint32_t start = 0;
         for (uint32_t bnd : seg_bounds) {
@@ -1219,15 +1247,10 @@ struct HashAgg {
         }
         if (start < b.n_rows) launch_apply(b, start, b.n_rows);
         if (timed) {
-            HIP_TRY(hipEventRecord(e1, stream));
-            HIP_TRY(hipEventSynchronize(e1));
-            float ms = 0;
-            HIP_TRY(hipEventElapsedTime(&ms, e0, e1));
-            apply_ms_total += ms;
+            HIP_TRY(hipEventRecord(ev1[slot], stream));
+            ev_pending[slot] = 1;
             apply_launches++;
             apply_rows += b.n_rows;
-            hipEventDestroy(e0);
-            hipEventDestroy(e1);
         }
         return RW_OK;
     }
@@ -1424,6 +1447,11 @@ struct HashAgg {
 
     ~HashAgg() {
         free_stage();
+        for (int s = 0; s < EV_RING; s++)
+            if (ev0[s]) {
+                hipEventDestroy(ev0[s]);
+                hipEventDestroy(ev1[s]);
+            }
         if (t.state) {
             hipFree(t.state);
             hipFree(t.keys);
@@ -1607,6 +1635,8 @@ long long rw_agg_flush_device(void* h, uint64_t epoch) {
 
 int rw_agg_kernel_stats(void* h, RwKernelStats* out) {
     auto* agg = (HashAgg*)h;
+    if (agg->ev_harvest_all() != RW_OK)
+        FAIL(RW_E_INTERNAL, "event harvest failed");
     out->launches = agg->apply_launches;
     out->total_ms = agg->apply_ms_total;
     out->rows = agg->apply_rows;
@@ -1695,6 +1725,7 @@ void rw_spill_free(uint8_t* buf) { free(buf); }
 
 int rw_agg_stats_reset(void* h) {
     auto* agg = (HashAgg*)h;
+    agg->ev_harvest_all();
     agg->apply_launches = 0;
     agg->apply_ms_total = 0;
     agg->apply_rows = 0;
@@ -2397,8 +2428,28 @@ struct HashJoin {
     std::vector<uint32_t> wm_pos;
     std::vector<uint8_t> wm_clean;
     std::vector<Wm> wm_side0, wm_side1, wm_out;
+    // lazy event ring (same rationale as HashAgg: no per-step host sync)
+    static constexpr int EV_RING = 256;
+    hipEvent_t ev0[EV_RING] = {}, ev1[EV_RING] = {};
+    uint8_t ev_pending[EV_RING] = {};
+    uint64_t ev_head = 0;
     double probe_ms_total = 0;
     uint64_t probe_launches = 0, probe_rows = 0;
+
+    int ev_harvest(int slot) {
+        if (!ev_pending[slot]) return RW_OK;
+        HIP_TRY(hipEventSynchronize(ev1[slot]));
+        float ms = 0;
+        HIP_TRY(hipEventElapsedTime(&ms, ev0[slot], ev1[slot]));
+        probe_ms_total += ms;
+        ev_pending[slot] = 0;
+        return RW_OK;
+    }
+    int ev_harvest_all() {
+        for (int s = 0; s < EV_RING; s++)
+            if (ev_harvest(s) != RW_OK) return RW_E_INTERNAL;
+        return RW_OK;
+    }
 
     int init(const RwHashJoinDesc* d) {
         if (!gpu_ok()) FAIL(RW_E_NOGPU, "risingwave_amd: no GPU visible (product path has no CPU fallback)");
@@ -2534,24 +2585,23 @@ struct HashJoin {
         uint32_t blocks = (r1 - r0 + 255) / 256;
         if (blocks > 2048) blocks = 2048;
         if (!blocks) blocks = 1;
-        hipEvent_t e0 = nullptr, e1 = nullptr;
+        int slot = -1;
         if (timed) {
-            HIP_TRY(hipEventCreate(&e0));
-            HIP_TRY(hipEventCreate(&e1));
-            HIP_TRY(hipEventRecord(e0, stream));
+            slot = (int)(ev_head++ % EV_RING);
+            if (!ev0[slot]) {
+                HIP_TRY(hipEventCreate(&ev0[slot]));
+                HIP_TRY(hipEventCreate(&ev1[slot]));
+            }
+            if (ev_harvest(slot) != RW_OK) return RW_E_INTERNAL;
+            HIP_TRY(hipEventRecord(ev0[slot], stream));
         }
         join_probe_kernel<<<blocks, 256, 0, stream>>>(b, side[s], side[1 - s], m, s,
                                                       out, r0, r1);
         if (timed) {
-            HIP_TRY(hipEventRecord(e1, stream));
-            HIP_TRY(hipEventSynchronize(e1));
-            float ms = 0;
-            HIP_TRY(hipEventElapsedTime(&ms, e0, e1));
-            probe_ms_total += ms;
+            HIP_TRY(hipEventRecord(ev1[slot], stream));
+            ev_pending[slot] = 1;
             probe_launches++;
             probe_rows += b.n_rows;
-            hipEventDestroy(e0);
-            hipEventDestroy(e1);
         }
         return RW_OK;
     }
@@ -2727,6 +2777,11 @@ struct HashJoin {
     }
 
     ~HashJoin() {
+        for (int s = 0; s < EV_RING; s++)
+            if (ev0[s]) {
+                hipEventDestroy(ev0[s]);
+                hipEventDestroy(ev1[s]);
+            }
         for (int s = 0; s < 2; s++) {
             JoinSideDev& js = side[s];
             if (js.slots) {
@@ -2896,6 +2951,8 @@ int rw_agg_apply_joinout(void* agg_h, void* join_h) {
 
 int rw_join_kernel_stats(void* h, RwKernelStats* out) {
     auto* j = (HashJoin*)h;
+    if (j->ev_harvest_all() != RW_OK)
+        FAIL(RW_E_INTERNAL, "event harvest failed");
     out->launches = j->probe_launches;
     out->total_ms = j->probe_ms_total;
     out->rows = j->probe_rows;
@@ -2904,6 +2961,7 @@ int rw_join_kernel_stats(void* h, RwKernelStats* out) {
 
 int rw_join_stats_reset(void* h) {
     auto* j = (HashJoin*)h;
+    j->ev_harvest_all();
     j->probe_launches = 0;
     j->probe_ms_total = 0;
     j->probe_rows = 0;
