@@ -522,3 +522,53 @@ def test_checkpoint_spill_parity():
         assert rg == ro, f"epoch {epoch+1}: {len(rg)} vs {len(ro)} records"
     g.close()
     o.close()
+
+
+def test_exchange_partition_parity():
+    # vnode partition (slice-by-8 CRC count/scan/scatter kernels) + self-loop
+    # exchange + payload apply vs the oracle on the same chunk. R=1 uses the
+    # device-copy bypass; the RCCL collective itself is exercised by
+    # tests/debug_exchange.py under RW_EXCHANGE_FORCE_NCCL=1.
+    import ctypes
+
+    import bench
+    import risingwave_amd
+
+    L = gpu().lib
+    L.rw_agg_bench_preload.restype = ctypes.c_void_p
+    L.rw_agg_bench_preload.argtypes = [ctypes.c_void_p,
+                                       ctypes.POINTER(ffi.RwChunkC)]
+    L.rw_agg_apply_payload.restype = ctypes.c_int
+    L.rw_agg_apply_payload.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                       ctypes.POINTER(ctypes.c_uint64),
+                                       ctypes.c_int, ctypes.c_int]
+    L.rw_agg_n_batch_slots.restype = ctypes.c_int
+    L.rw_agg_n_batch_slots.argtypes = [ctypes.c_void_p]
+
+    calls = [(ffi.AGG_MAX, 1, T_I64), (ffi.AGG_COUNT_STAR, -1, T_I64)]
+    agg = ffi.HashAgg(gpu(), [T_I64, T_I64], [0], calls, 1, append_only=True)
+    exch = bench.setup_exchange(ffi, 0, 1, None)
+    assert exch is not None, "exchange init failed"
+    rng = np.random.default_rng(3)
+    n = 65536
+    c = bench.make_q7_chunk(ffi, rng, n, 0, 32)
+    cc = c.to_c()
+    batch = L.rw_agg_bench_preload(agg.h, ctypes.byref(cc))
+    assert batch
+    xb = exch.make_buffers(n * 32 * 4)
+    nslots = L.rw_agg_n_batch_slots(agg.h)
+    recv_counts = exch.run(agg.h, batch, xb, n_cols=nslots)
+    assert sum(recv_counts) == n
+    rc = L.rw_agg_apply_payload(agg.h, ctypes.c_void_p(xb.recv), recv_counts,
+                                1, nslots)
+    assert rc == 0
+    agg.flush(1)
+    got = ffi.rows_multiset(agg.poll_all())
+    o = ffi.HashAgg(ffi.oracle(), [T_I64, T_I64], [0], calls, 1,
+                    append_only=True)
+    o.push(c)
+    o.flush(1)
+    want = ffi.rows_multiset(o.poll_all())
+    assert got == want
+    agg.close()
+    o.close()
