@@ -556,7 +556,7 @@ __device__ __forceinline__ long long agg4_comb(uint8_t kind, long long a,
     }
 }
 
-template <int n_calls>
+template <int n_calls, int RPL = 4> // RPL = rows per lane (even)
 __global__ void agg_apply_dense4_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
                                         AggCallDev c1, AggCallDev c2,
                                         AggCallDev c3, uint32_t r0, uint32_t r1) {
@@ -564,7 +564,7 @@ __global__ void agg_apply_dense4_kernel(AggBatch b, AggTableDev t, AggCallDev c0
     int lane = threadIdx.x & 63;
     size_t cap = (size_t)t.cap_mask + 1;
     const uint32_t SLOT_NONE = (uint32_t)-1;
-    uint32_t stride_rows = (gridDim.x * blockDim.x) * 4;
+    uint32_t stride_rows = (gridDim.x * blockDim.x) * RPL;
     uint32_t iters = (r1 - r0 + stride_rows - 1) / stride_rows;
 
     long long memo_key = 0;
@@ -619,35 +619,41 @@ __global__ void agg_apply_dense4_kernel(AggBatch b, AggTableDev t, AggCallDev c0
     };
 
     for (uint32_t it = 0; it < iters; it++) {
-        uint32_t rb =
-            r0 + it * stride_rows + (blockIdx.x * blockDim.x + threadIdx.x) * 4;
-        bool act[4];
-        long long k[4];
-        long long cv[4][4]; // [row][call]
-        bool all4 = rb + 3 < r1;
-        if (all4) {
+        uint32_t rb = r0 + it * stride_rows +
+                      (blockIdx.x * blockDim.x + threadIdx.x) * RPL;
+        bool act[RPL];
+        long long k[RPL];
+        long long cv[RPL][4]; // [row][call]
+        bool all_in = rb + RPL - 1 < r1;
+        if (all_in) {
+            // back-to-back b128 loads: RPL/2 per column, all in flight at once
             const ulonglong2* kp = (const ulonglong2*)(b.col_vals[0] + rb);
-            ulonglong2 k01 = kp[0], k23 = kp[1];
-            k[0] = (long long)k01.x;
-            k[1] = (long long)k01.y;
-            k[2] = (long long)k23.x;
-            k[3] = (long long)k23.y;
-            act[0] = act[1] = act[2] = act[3] = true;
+#pragma unroll
+            for (int h = 0; h < RPL / 2; h++) {
+                ulonglong2 kk = kp[h];
+                k[2 * h] = (long long)kk.x;
+                k[2 * h + 1] = (long long)kk.y;
+            }
+#pragma unroll
+            for (int r = 0; r < RPL; r++) act[r] = true;
             for (int ci = 0; ci < n_calls; ci++) {
                 if (calls[ci].arg < 0) {
-                    for (int r = 0; r < 4; r++) cv[r][ci] = 1;
+#pragma unroll
+                    for (int r = 0; r < RPL; r++) cv[r][ci] = 1;
                     continue;
                 }
                 const ulonglong2* vp =
                     (const ulonglong2*)(b.col_vals[1 + ci] + rb);
-                ulonglong2 v01 = vp[0], v23 = vp[1];
-                cv[0][ci] = agg4_unit(calls[ci].kind, (long long)v01.x);
-                cv[1][ci] = agg4_unit(calls[ci].kind, (long long)v01.y);
-                cv[2][ci] = agg4_unit(calls[ci].kind, (long long)v23.x);
-                cv[3][ci] = agg4_unit(calls[ci].kind, (long long)v23.y);
+#pragma unroll
+                for (int h = 0; h < RPL / 2; h++) {
+                    ulonglong2 vv = vp[h];
+                    cv[2 * h][ci] = agg4_unit(calls[ci].kind, (long long)vv.x);
+                    cv[2 * h + 1][ci] =
+                        agg4_unit(calls[ci].kind, (long long)vv.y);
+                }
             }
         } else {
-            for (int r = 0; r < 4; r++) {
+            for (int r = 0; r < RPL; r++) {
                 act[r] = rb + r < r1;
                 k[r] = act[r] ? b.col_vals[0][rb + r] : 0;
                 for (int ci = 0; ci < n_calls; ci++)
@@ -659,7 +665,7 @@ __global__ void agg_apply_dense4_kernel(AggBatch b, AggTableDev t, AggCallDev c0
                                : 0;
             }
         }
-        // lane-local segments over the 4 rows (inactive rows break runs)
+        // lane-local segments over the RPL rows (inactive rows break runs)
         long long last_key = 0;
         bool any = false, has_bnd = false;
         long long cur_key = 0;
@@ -667,7 +673,7 @@ __global__ void agg_apply_dense4_kernel(AggBatch b, AggTableDev t, AggCallDev c0
         bool have_cur = false, have_pre = false;
         long long preq[4];
         long long pre_key = 0;
-        for (int r = 0; r < 4; r++) {
+        for (int r = 0; r < RPL; r++) {
             if (!act[r]) {
                 if (have_cur) {
                     if (!have_pre) {
@@ -1187,13 +1193,32 @@ struct HashAgg {
         if (b.dense && KW == 1 && b.stride == 1 && n_minput == 0 &&
             debug_mode == 0 && ((uintptr_t)(b.col_vals[0] + r0) & 31) == 0 &&
             (r1 - r0) >= 1024) {
-            int grid = grid_for((r1 - r0 + 3) / 4);
-            switch (n_calls) {
-                case 1: agg_apply_dense4_kernel<1><<<grid, 256, 0, stream>>>(b, t, a0, a1, a2, a3, r0, r1); return;
-                case 2: agg_apply_dense4_kernel<2><<<grid, 256, 0, stream>>>(b, t, a0, a1, a2, a3, r0, r1); return;
-                case 3: agg_apply_dense4_kernel<3><<<grid, 256, 0, stream>>>(b, t, a0, a1, a2, a3, r0, r1); return;
-                case 4: agg_apply_dense4_kernel<4><<<grid, 256, 0, stream>>>(b, t, a0, a1, a2, a3, r0, r1); return;
+            // rows-per-lane: more rows deepen the load pipeline and amortize
+            // the cross-lane scan (A/B-selectable via RW_AGG_RPL)
+            static int rpl = [] {
+                const char* e = getenv("RW_AGG_RPL");
+                int v = e ? atoi(e) : 8;
+                return (v == 4 || v == 8 || v == 16) ? v : 8;
+            }();
+            int grid = grid_for((r1 - r0 + rpl - 1) / rpl);
+            #define RW_DENSE(nc, rp)                                          \
+                agg_apply_dense4_kernel<nc, rp><<<grid, 256, 0, stream>>>(    \
+                    b, t, a0, a1, a2, a3, r0, r1)
+            switch (n_calls * 32 + rpl) {
+                case 1 * 32 + 4: RW_DENSE(1, 4); return;
+                case 1 * 32 + 8: RW_DENSE(1, 8); return;
+                case 1 * 32 + 16: RW_DENSE(1, 16); return;
+                case 2 * 32 + 4: RW_DENSE(2, 4); return;
+                case 2 * 32 + 8: RW_DENSE(2, 8); return;
+                case 2 * 32 + 16: RW_DENSE(2, 16); return;
+                case 3 * 32 + 4: RW_DENSE(3, 4); return;
+                case 3 * 32 + 8: RW_DENSE(3, 8); return;
+                case 3 * 32 + 16: RW_DENSE(3, 16); return;
+                case 4 * 32 + 4: RW_DENSE(4, 4); return;
+                case 4 * 32 + 8: RW_DENSE(4, 8); return;
+                case 4 * 32 + 16: RW_DENSE(4, 16); return;
             }
+            #undef RW_DENSE
         }
         int grid = grid_for(r1 - r0);
         #define RW_LAUNCH(kw, nc)                                             \
