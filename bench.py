@@ -197,8 +197,8 @@ def main():
     L.rw_agg_bench_apply.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
     L.rw_agg_sync.restype = ctypes.c_int
     L.rw_agg_sync.argtypes = [ctypes.c_void_p]
-    L.rw_agg_flush_device.restype = ctypes.c_longlong
-    L.rw_agg_flush_device.argtypes = [ctypes.c_void_p, ctypes.c_uint64]
+    L.rw_agg_flush_launch.restype = ctypes.c_int
+    L.rw_agg_flush_launch.argtypes = [ctypes.c_void_p, ctypes.c_uint64]
     L.rw_agg_kernel_stats.argtypes = [ctypes.c_void_p, ctypes.POINTER(KernelStats)]
     L.rw_agg_stats_reset.argtypes = [ctypes.c_void_p]
     L.rw_agg_apply_payload.restype = ctypes.c_int
@@ -277,12 +277,14 @@ def main():
             rc = L.rw_agg_bench_apply(agg.h, batches[i % n_batches])
             assert rc == 0, gpu_lib.last_error()
         if (i + 1) % args.barrier_every == 0:
-            # checkpoint barrier: change inference + emission into HBM; the
-            # downstream (exchange/sink) consumes device-resident, as in the
-            # q3 pipeline — the host-marshalling flush stays the parity-test
-            # surface (tests/test_gpu_parity.py)
-            n = L.rw_agg_flush_device(agg.h, i)
-            assert n >= 0, gpu_lib.last_error()
+            # checkpoint barrier: change inference + emission into HBM,
+            # stream-ordered with no host round-trip; overflow surfaces at
+            # the end-of-region rw_agg_sync. The downstream (exchange/sink)
+            # consumes device-resident, as in the q3 pipeline — the
+            # host-marshalling flush stays the parity-test surface
+            # (tests/test_gpu_parity.py)
+            rc = L.rw_agg_flush_launch(agg.h, i)
+            assert rc == 0, gpu_lib.last_error()
 
     # ---- warmup ----
     for i in range(args.warmup):
