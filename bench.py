@@ -199,6 +199,11 @@ def main():
     L.rw_agg_sync.argtypes = [ctypes.c_void_p]
     L.rw_agg_flush_launch.restype = ctypes.c_int
     L.rw_agg_flush_launch.argtypes = [ctypes.c_void_p, ctypes.c_uint64]
+    L.rw_agg_bench_run.restype = ctypes.c_int
+    L.rw_agg_bench_run.argtypes = [ctypes.c_void_p,
+                                   ctypes.POINTER(ctypes.c_void_p),
+                                   ctypes.c_int, ctypes.c_int, ctypes.c_int,
+                                   ctypes.c_int]
     L.rw_agg_kernel_stats.argtypes = [ctypes.c_void_p, ctypes.POINTER(KernelStats)]
     L.rw_agg_stats_reset.argtypes = [ctypes.c_void_p]
     L.rw_agg_apply_payload.restype = ctypes.c_int
@@ -286,9 +291,21 @@ def main():
             rc = L.rw_agg_flush_launch(agg.h, i)
             assert rc == 0, gpu_lib.last_error()
 
+    # non-exchange path: the whole step loop runs in C (the Python
+    # interpreter costs more per step than the apply kernel itself)
+    batch_arr = (ctypes.c_void_p * n_batches)(*batches)
+
+    def run_steps(n):
+        if use_exchange:
+            for i in range(n):
+                step(i)
+        else:
+            rc = L.rw_agg_bench_run(agg.h, batch_arr, n_batches, n,
+                                    args.barrier_every, 0)
+            assert rc == 0, gpu_lib.last_error()
+
     # ---- warmup ----
-    for i in range(args.warmup):
-        step(i)
+    run_steps(args.warmup)
     rc = L.rw_agg_sync(agg.h)
     assert rc == 0, gpu_lib.last_error()
     L.rw_agg_stats_reset(agg.h)
@@ -297,8 +314,7 @@ def main():
         dist.barrier()
     _dev_sync()
     t0 = time.perf_counter()
-    for i in range(args.steps):
-        step(i)
+    run_steps(args.steps)
     rc = L.rw_agg_sync(agg.h)
     assert rc == 0, gpu_lib.last_error()
     _dev_sync()

@@ -595,7 +595,7 @@ __global__ void agg_apply_dense4_kernel(AggBatch b, AggTableDev t, AggCallDev c0
             }
             memo_dirtied = true;
         }
-        for (int ci = 0; ci < n_calls; ci++) {
+        _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) {
             long long* acc = t.acc + (size_t)ci * cap;
             uint8_t* has = t.has + (size_t)ci * cap;
             switch (calls[ci].kind) {
@@ -623,7 +623,7 @@ __global__ void agg_apply_dense4_kernel(AggBatch b, AggTableDev t, AggCallDev c0
                       (blockIdx.x * blockDim.x + threadIdx.x) * RPL;
         bool act[RPL];
         long long k[RPL];
-        long long cv[RPL][4]; // [row][call]
+        long long cv[RPL][n_calls]; // [row][call]
         bool all_in = rb + RPL - 1 < r1;
         if (all_in) {
             // back-to-back b128 loads: RPL/2 per column, all in flight at once
@@ -636,7 +636,7 @@ __global__ void agg_apply_dense4_kernel(AggBatch b, AggTableDev t, AggCallDev c0
             }
 #pragma unroll
             for (int r = 0; r < RPL; r++) act[r] = true;
-            for (int ci = 0; ci < n_calls; ci++) {
+            _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) {
                 if (calls[ci].arg < 0) {
 #pragma unroll
                     for (int r = 0; r < RPL; r++) cv[r][ci] = 1;
@@ -656,7 +656,7 @@ __global__ void agg_apply_dense4_kernel(AggBatch b, AggTableDev t, AggCallDev c0
             for (int r = 0; r < RPL; r++) {
                 act[r] = rb + r < r1;
                 k[r] = act[r] ? b.col_vals[0][rb + r] : 0;
-                for (int ci = 0; ci < n_calls; ci++)
+                _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++)
                     cv[r][ci] =
                         act[r] ? agg4_unit(calls[ci].kind,
                                            calls[ci].arg < 0
@@ -669,15 +669,15 @@ __global__ void agg_apply_dense4_kernel(AggBatch b, AggTableDev t, AggCallDev c0
         long long last_key = 0;
         bool any = false, has_bnd = false;
         long long cur_key = 0;
-        long long cur[4];
+        long long cur[n_calls];
         bool have_cur = false, have_pre = false;
-        long long preq[4];
+        long long preq[n_calls];
         long long pre_key = 0;
         for (int r = 0; r < RPL; r++) {
             if (!act[r]) {
                 if (have_cur) {
                     if (!have_pre) {
-                        for (int ci = 0; ci < n_calls; ci++) preq[ci] = cur[ci];
+                        _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) preq[ci] = cur[ci];
                         pre_key = cur_key;
                         have_pre = true;
                     } else {
@@ -690,12 +690,12 @@ __global__ void agg_apply_dense4_kernel(AggBatch b, AggTableDev t, AggCallDev c0
             }
             any = true;
             if (have_cur && k[r] == cur_key) {
-                for (int ci = 0; ci < n_calls; ci++)
+                _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++)
                     cur[ci] = agg4_comb(calls[ci].kind, cur[ci], cv[r][ci]);
             } else {
                 if (have_cur) {
                     if (!have_pre) {
-                        for (int ci = 0; ci < n_calls; ci++) preq[ci] = cur[ci];
+                        _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) preq[ci] = cur[ci];
                         pre_key = cur_key;
                         have_pre = true;
                     } else {
@@ -704,7 +704,7 @@ __global__ void agg_apply_dense4_kernel(AggBatch b, AggTableDev t, AggCallDev c0
                     has_bnd = true;
                 }
                 cur_key = k[r];
-                for (int ci = 0; ci < n_calls; ci++) cur[ci] = cv[r][ci];
+                _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) cur[ci] = cv[r][ci];
                 have_cur = true;
             }
         }
@@ -712,10 +712,10 @@ __global__ void agg_apply_dense4_kernel(AggBatch b, AggTableDev t, AggCallDev c0
         // holds the first closed run (prefix). Gaps between same-key runs
         // are irrelevant: every partition of a key's rows commits correctly
         // under commutative combines.
-        long long sufv[4];
+        long long sufv[n_calls];
         bool have_suf = have_cur;
         long long suf_key = cur_key;
-        for (int ci = 0; ci < n_calls; ci++)
+        _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++)
             sufv[ci] = have_cur ? cur[ci] : agg4_identity(calls[ci].kind);
         last_key = have_suf ? suf_key : (have_pre ? pre_key : 0);
         bool lane_any = any;
@@ -736,17 +736,17 @@ __global__ void agg_apply_dense4_kernel(AggBatch b, AggTableDev t, AggCallDev c0
         uint64_t le_mask = heads_b & (~0ULL >> (63 - lane));
         int run_start = 63 - __clzll(le_mask | 1ULL);
         int run_pos = lane - run_start;
-        long long incl[4];
-        for (int ci = 0; ci < n_calls; ci++) incl[ci] = sufv[ci];
+        long long incl[n_calls];
+        _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) incl[ci] = sufv[ci];
         for (int d = 1; d < 64; d <<= 1) {
-            long long ov[4];
-            for (int ci = 0; ci < n_calls; ci++) ov[ci] = __shfl_up(incl[ci], d);
+            long long ov[n_calls];
+            _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) ov[ci] = __shfl_up(incl[ci], d);
             if (run_pos >= d)
-                for (int ci = 0; ci < n_calls; ci++)
+                _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++)
                     incl[ci] = agg4_comb(calls[ci].kind, incl[ci], ov[ci]);
         }
-        long long incoming[4];
-        for (int ci = 0; ci < n_calls; ci++)
+        long long incoming[n_calls];
+        _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++)
             incoming[ci] = __shfl_up(incl[ci], 1);
 
         uint64_t cont_b = __ballot(cont);
@@ -757,8 +757,8 @@ __global__ void agg_apply_dense4_kernel(AggBatch b, AggTableDev t, AggCallDev c0
         if (lane_any) {
             if (cont && lane_has_bnd) {
                 // closer: commit the incoming run + own prefix
-                long long tot[4];
-                for (int ci = 0; ci < n_calls; ci++)
+                long long tot[n_calls];
+                _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++)
                     tot[ci] = agg4_comb(calls[ci].kind, incoming[ci], preq[ci]);
                 commit(pre_key, tot);
             } else if (!cont && lane_has_bnd && have_pre) {
@@ -1578,6 +1578,26 @@ void* rw_agg_bench_preload(void* h, const RwChunk* c) {
 int rw_agg_bench_apply(void* h, void* batch) {
     auto* agg = (HashAgg*)h;
     return agg->apply(*(AggBatch*)batch, true);
+}
+
+int rw_agg_flush_launch(void* h, uint64_t epoch); // defined below
+
+// C-side step loop: the Python interpreter costs ~15-20 us per step at q7
+// sizes (ctypes + loop bookkeeping), which exceeds the 16 us apply kernel
+// and makes the host the bottleneck. One call runs `steps` applies with the
+// periodic stream-ordered checkpoint flush; everything stays async.
+int rw_agg_bench_run(void* h, void** batches, int n_batches, int steps,
+                     int barrier_every, int step0) {
+    auto* agg = (HashAgg*)h;
+    for (int i = 0; i < steps; i++) {
+        int rc = agg->apply(*(AggBatch*)batches[(step0 + i) % n_batches], true);
+        if (rc != RW_OK) return rc;
+        if (barrier_every > 0 && (step0 + i + 1) % barrier_every == 0) {
+            rc = rw_agg_flush_launch(h, (uint64_t)(step0 + i));
+            if (rc != RW_OK) return rc;
+        }
+    }
+    return RW_OK;
 }
 
 int rw_agg_sync(void* h) {
