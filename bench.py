@@ -204,6 +204,11 @@ def main():
                                    ctypes.POINTER(ctypes.c_void_p),
                                    ctypes.c_int, ctypes.c_int, ctypes.c_int,
                                    ctypes.c_int]
+    L.rw_agg_bench_run_epochs.restype = ctypes.c_int
+    L.rw_agg_bench_run_epochs.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                          ctypes.c_uint64, ctypes.c_int,
+                                          ctypes.c_int, ctypes.c_int,
+                                          ctypes.c_int]
     L.rw_agg_kernel_stats.argtypes = [ctypes.c_void_p, ctypes.POINTER(KernelStats)]
     L.rw_agg_stats_reset.argtypes = [ctypes.c_void_p]
     L.rw_agg_apply_payload.restype = ctypes.c_int
@@ -257,14 +262,38 @@ def main():
     # cdna_hip_programming.md §2)
     n_batches = 16
     batches = []
-    for b in range(n_batches):
-        c = make_q7_chunk(ffi, rng, batch_rows,
-                          window_base + b * args.windows_per_epoch * WINDOW_US,
-                          args.windows_per_epoch)
-        cc = c.to_c()
-        h = L.rw_agg_bench_preload(agg.h, ctypes.byref(cc))
-        assert h, gpu_lib.last_error()
-        batches.append(h)
+    giant = None
+    eff_barrier = (args.barrier_every
+                   if args.barrier_every > 0 and n_batches % args.barrier_every == 0
+                   else n_batches)
+    if use_exchange:
+        for b in range(n_batches):
+            c = make_q7_chunk(ffi, rng, batch_rows,
+                              window_base + b * args.windows_per_epoch * WINDOW_US,
+                              args.windows_per_epoch)
+            cc = c.to_c()
+            h = L.rw_agg_bench_preload(agg.h, ctypes.byref(cc))
+            assert h, gpu_lib.last_error()
+            batches.append(h)
+    else:
+        # one contiguous 16-step region: the engine buffers an epoch's chunks
+        # and applies them as ONE launch before each checkpoint flush
+        w_all, p_all = [], []
+        for b in range(n_batches):
+            base = window_base + b * args.windows_per_epoch * WINDOW_US
+            w_all.append(base +
+                         np.sort(rng.integers(0, args.windows_per_epoch,
+                                              batch_rows)) * WINDOW_US)
+            p_all.append(rng.integers(1, 10**7, batch_rows))
+        total = n_batches * batch_rows
+        giant_c = ffi.Chunk(
+            [T_I64, T_I64], np.zeros(total, np.uint8),
+            [np.concatenate(w_all), np.concatenate(p_all)],
+            [np.ones(total, np.uint8), np.ones(total, np.uint8)])
+        cc = giant_c.to_c()
+        giant = L.rw_agg_bench_preload(agg.h, ctypes.byref(cc))
+        assert giant, gpu_lib.last_error()
+        del giant_c, w_all, p_all
 
     if use_exchange:
         payload_cap = int(batch_rows * (1 + 2 * 9) * 4)  # 4x headroom for skew
@@ -291,17 +320,15 @@ def main():
             rc = L.rw_agg_flush_launch(agg.h, i)
             assert rc == 0, gpu_lib.last_error()
 
-    # non-exchange path: the whole step loop runs in C (the Python
-    # interpreter costs more per step than the apply kernel itself)
-    batch_arr = (ctypes.c_void_p * n_batches)(*batches)
-
+    # non-exchange path: the whole step loop runs in C, epoch-granular (the
+    # Python interpreter costs more per step than the apply kernel itself)
     def run_steps(n):
         if use_exchange:
             for i in range(n):
                 step(i)
         else:
-            rc = L.rw_agg_bench_run(agg.h, batch_arr, n_batches, n,
-                                    args.barrier_every, 0)
+            rc = L.rw_agg_bench_run_epochs(agg.h, giant, batch_rows,
+                                           n_batches, n, eff_barrier, 0)
             assert rc == 0, gpu_lib.last_error()
 
     # ---- warmup ----

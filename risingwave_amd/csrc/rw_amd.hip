@@ -1282,6 +1282,29 @@ struct HashAgg {
         return RW_OK;
     }
 
+    // apply a sub-range of a device batch as one launch (epoch-granular
+    // ingestion: chunks buffered within an epoch are applied together)
+    int apply_range(const AggBatch& b, uint32_t r0, uint32_t r1, bool timed) {
+        int slot = -1;
+        if (timed) {
+            slot = (int)(ev_head++ % EV_RING);
+            if (!ev0[slot]) {
+                HIP_TRY(hipEventCreate(&ev0[slot]));
+                HIP_TRY(hipEventCreate(&ev1[slot]));
+            }
+            if (ev_harvest(slot) != RW_OK) return RW_E_INTERNAL;
+            HIP_TRY(hipEventRecord(ev0[slot], stream));
+        }
+        launch_apply(b, r0, r1);
+        if (timed) {
+            HIP_TRY(hipEventRecord(ev1[slot], stream));
+            ev_pending[slot] = 1;
+            apply_launches++;
+            apply_rows += r1 - r0;
+        }
+        return RW_OK;
+    }
+
     // A minput DELETE targeting a row INSERTed earlier in the same chunk
     // must observe that insert (the reference applies rows in order,
     // hash_agg.rs:332-398); the parallel kernel keeps that only across
@@ -1596,6 +1619,33 @@ int rw_agg_bench_run(void* h, void** batches, int n_batches, int steps,
             rc = rw_agg_flush_launch(h, (uint64_t)(step0 + i));
             if (rc != RW_OK) return rc;
         }
+    }
+    return RW_OK;
+}
+
+// Epoch-granular step loop over one contiguous preloaded region: the
+// engine buffers an epoch's chunks and applies them as ONE launch before
+// the checkpoint flush (order-free value states) — a
+// barrier_every×1M-row launch pipelines HBM loads far better than
+// barrier_every separate 1M-row launches. `giant` holds n_slots logical
+// steps of rows_per_step rows each; step s reads slot s % n_slots.
+// Requires barrier_every | n_slots so every epoch is a contiguous range.
+int rw_agg_bench_run_epochs(void* h, void* giant, uint64_t rows_per_step,
+                            int n_slots, int steps, int barrier_every,
+                            int step0) {
+    auto* agg = (HashAgg*)h;
+    auto* b = (AggBatch*)giant;
+    if (barrier_every <= 0 || n_slots % barrier_every ||
+        step0 % barrier_every)
+        FAIL(RW_E_INVAL, "epoch run needs barrier_every | n_slots, aligned step0");
+    for (int s = 0; s < steps; s += barrier_every) {
+        int width = steps - s < barrier_every ? steps - s : barrier_every;
+        uint32_t r0 = (uint32_t)(((step0 + s) % n_slots) * rows_per_step);
+        uint32_t r1 = r0 + (uint32_t)(width * rows_per_step);
+        int rc = agg->apply_range(*b, r0, r1, true);
+        if (rc != RW_OK) return rc;
+        rc = rw_agg_flush_launch(h, (uint64_t)(step0 + s));
+        if (rc != RW_OK) return rc;
     }
     return RW_OK;
 }
