@@ -363,7 +363,8 @@ def main():
         # roofline: algorithmic bytes per apply launch ÷ measured launch time
         # (HIP events on the executor's stream, inside the C library)
         avg_launch_ms = ks.total_ms / max(ks.launches, 1)
-        achieved_gbs = (BYTES_PER_ROW * batch_rows) / (avg_launch_ms * 1e-3) / 1e9
+        rows_per_launch = ks.rows / max(ks.launches, 1)
+        achieved_gbs = (BYTES_PER_ROW * rows_per_launch) / (avg_launch_ms * 1e-3) / 1e9
         exch_stats = None
         if use_exchange and exch is not None:
             ems, en = exch.stats()
