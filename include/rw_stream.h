@@ -169,6 +169,18 @@ int rw_hash_join_watermark(void* h, int side, uint32_t col_idx, int64_t val,
  * update_watermark); non-EOWC, so nothing is emitted. */
 int rw_hash_agg_watermark(void* h, uint32_t group_key_pos, int64_t val);
 
+/* Rescale re-scoping (ExecutorParams vnode bitmap updates on scale-in/out,
+ * state_table.rs update_vnode_bitmap + executor cache eviction): state
+ * whose distribution-key vnode (dist key = group key for agg, join key for
+ * join) is no longer owned is dropped WITHOUT emitting retractions — the
+ * new owner holds it after recovery. `bitmap` is vnode_count/8 bytes,
+ * LSB-first per byte (common/src/bitmap.rs layout); vnode_count matches
+ * VirtualNode::COUNT (256 default). */
+int rw_hash_agg_update_vnode_bitmap(void* h, const uint8_t* bitmap,
+                                    uint32_t vnode_count);
+int rw_hash_join_update_vnode_bitmap(void* h, const uint8_t* bitmap,
+                                     uint32_t vnode_count);
+
 #ifdef __cplusplus
 }
 #endif

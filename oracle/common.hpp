@@ -187,6 +187,68 @@ struct ChunkBuilder {
 };
 
 // StreamChunk::eliminate_adjacent_noop_update (stream_chunk.rs:331-384).
+// vnode of a key row (Crc32 of the hash_datum byte feed % vnode_count,
+// consistent_hash/vnode.rs:146-181; NULL sentinel 0xfffffff0,
+// array/mod.rs:99) — used by rescale re-scoping (update_vnode_bitmap)
+inline uint32_t vnode_of_key_row(const Row& key,
+                                 const std::vector<uint8_t>& key_types,
+                                 uint32_t vnode_count) {
+    static uint32_t tab[256];
+    static bool init = false;
+    if (!init) {
+        for (uint32_t i = 0; i < 256; i++) {
+            uint32_t c = i;
+            for (int k = 0; k < 8; k++)
+                c = (c & 1) ? 0xEDB88320u ^ (c >> 1) : c >> 1;
+            tab[i] = c;
+        }
+        init = true;
+    }
+    auto feed = [&](uint32_t crc, const void* p, size_t n) {
+        const uint8_t* b = (const uint8_t*)p;
+        for (size_t i = 0; i < n; i++)
+            crc = tab[(crc ^ b[i]) & 0xFF] ^ (crc >> 8);
+        return crc;
+    };
+    uint32_t crc = 0xFFFFFFFFu;
+    for (size_t k = 0; k < key.size(); k++) {
+        const Datum& d = key[k];
+        if (d.null) {
+            uint32_t sentinel = 0xfffffff0u;
+            crc = feed(crc, &sentinel, 4);
+            continue;
+        }
+        switch (key_types[k]) {
+            case RW_T_I32: {
+                int32_t v = (int32_t)d.i;
+                crc = feed(crc, &v, 4);
+                break;
+            }
+            case RW_T_BOOL: {
+                uint8_t v = (uint8_t)d.i;
+                crc = feed(crc, &v, 1);
+                break;
+            }
+            case RW_T_F64: {
+                double v = d.d;
+                crc = feed(crc, &v, 8);
+                break;
+            }
+            case RW_T_F32: {
+                float v = (float)d.d;
+                crc = feed(crc, &v, 4);
+                break;
+            }
+            default: { // I64 / TS (i64 micros)
+                int64_t v = d.i;
+                crc = feed(crc, &v, 8);
+                break;
+            }
+        }
+    }
+    return (crc ^ 0xFFFFFFFFu) % vnode_count;
+}
+
 inline void eliminate_adjacent_noop_update(OwnedChunk& c) {
     size_t len = c.n_rows();
     if (c.vis.empty()) c.vis.assign(len, 1);

@@ -397,6 +397,26 @@ struct HashAggOracle {
         outputs.erase(outputs.begin());
         return chunk_to_c(*c);
     }
+
+    // rescale re-scope (update_vnode_bitmap): drop groups whose dist-key
+    // vnode (dist key = group key) is no longer owned; no retractions —
+    // the new owner holds the state
+    int update_vnode_bitmap(const uint8_t* bm, uint32_t vnode_count) {
+        for (auto it = groups.begin(); it != groups.end();) {
+            uint32_t vn = vnode_of_key_row(it->first, group_key_types,
+                                           vnode_count);
+            if ((bm[vn >> 3] >> (vn & 7)) & 1) {
+                ++it;
+            } else {
+                dirty.erase(it->first);
+                it = groups.erase(it);
+            }
+        }
+        for (auto it = dirty_order.begin(); it != dirty_order.end();)
+            if (!groups.count(*it)) it = dirty_order.erase(it);
+            else ++it;
+        return RW_OK;
+    }
 };
 
 } // namespace orc
@@ -421,6 +441,11 @@ int rw_hash_agg_push_chunk(void* h, const RwChunk* c) {
 int rw_hash_agg_flush(void* h, uint64_t epoch) { return ((HashAggOracle*)h)->flush(epoch); }
 int rw_hash_agg_watermark(void* h, uint32_t pos, int64_t val) {
     return ((HashAggOracle*)h)->watermark(pos, val);
+}
+int rw_hash_agg_update_vnode_bitmap(void* h, const uint8_t* bitmap,
+                                    uint32_t vnode_count) {
+    if (!vnode_count || vnode_count % 8) return RW_E_INVAL;
+    return ((HashAggOracle*)h)->update_vnode_bitmap(bitmap, vnode_count);
 }
 
 // drain the checkpoint spill buffer (caller frees with rw_spill_free)

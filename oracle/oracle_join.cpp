@@ -362,6 +362,22 @@ struct HashJoinOracle {
         }
     }
 
+    // rescale re-scope (update_vnode_bitmap): dist key = join key; both
+    // sides drop unowned keys, no retractions
+    int update_vnode_bitmap(const uint8_t* bm, uint32_t vnode_count) {
+        for (int s = 0; s < 2; s++) {
+            std::vector<uint8_t> kt;
+            for (auto k : side[s].key_idx) kt.push_back(side[s].types[k]);
+            auto& tab = side[s].table;
+            for (auto it = tab.begin(); it != tab.end();) {
+                uint32_t vn = vnode_of_key_row(it->first, kt, vnode_count);
+                if ((bm[vn >> 3] >> (vn & 7)) & 1) ++it;
+                else it = tab.erase(it);
+            }
+        }
+        return RW_OK;
+    }
+
     int watermark(int s, uint32_t col_idx, int64_t val, uint32_t* out_cols,
                   int64_t* out_vals, int max_out) {
         int n_out = 0;
@@ -402,6 +418,12 @@ void* rw_hash_join_create(const RwHashJoinDesc* d) { return new HashJoinOracle(d
 int rw_hash_join_push_chunk(void* h, int side, const RwChunk* c) {
     return ((HashJoinOracle*)h)->push(side, c);
 }
+int rw_hash_join_update_vnode_bitmap(void* h, const uint8_t* bitmap,
+                                     uint32_t vnode_count) {
+    if (!vnode_count || vnode_count % 8) return RW_E_INVAL;
+    return ((HashJoinOracle*)h)->update_vnode_bitmap(bitmap, vnode_count);
+}
+
 int rw_hash_join_watermark(void* h, int side, uint32_t col_idx, int64_t val,
                            uint32_t* out_cols, int64_t* out_vals, int max_out) {
     return ((HashJoinOracle*)h)->watermark(side, col_idx, val, out_cols,

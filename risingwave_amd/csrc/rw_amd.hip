@@ -979,6 +979,7 @@ struct HashAgg {
     // flush, drained by rw_agg_checkpoint_drain
     std::vector<uint8_t> spill;
     int debug_mode = 0; // RW_AGG_DEBUG_MODE: 1 per-lane atomics, 2 no-dedupe
+    uint8_t* d_vnode_bitmap = nullptr; // rescale scope (update_vnode_bitmap)
     int n_minput = 0;   // materialized-input (retractable min/max) calls
     std::vector<uint8_t> call_minput;
     std::vector<uint32_t> stream_key;
@@ -1861,6 +1862,21 @@ int rw_agg_stats_reset(void* h) {
 __constant__ uint32_t g_crc_table[256];
 static bool g_crc_table_init = false;
 
+static int ensure_crc_table() {
+    if (g_crc_table_init) return RW_OK;
+    uint32_t tab[256];
+    for (uint32_t i = 0; i < 256; i++) {
+        uint32_t c = i;
+        for (int k = 0; k < 8; k++) c = (c & 1) ? 0xEDB88320u ^ (c >> 1) : c >> 1;
+        tab[i] = c;
+    }
+    if (hipMemcpyToSymbol(HIP_SYMBOL(g_crc_table), tab, sizeof tab) !=
+        hipSuccess)
+        return RW_E_INTERNAL;
+    g_crc_table_init = true;
+    return RW_OK;
+}
+
 // LDS-staged CRC: divergent indexing of __constant__ memory serializes
 // (each distinct address replays); LDS banks handle it at full rate
 __device__ __forceinline__ uint32_t crc32_bytes(const uint32_t* lut, uint32_t crc,
@@ -1923,17 +1939,7 @@ typedef struct {
 int rw_vnode_compute(const RwVnodeDesc* d, const RwChunk* chunk, uint16_t* out) {
     if (!gpu_ok()) FAIL(RW_E_NOGPU, "no GPU visible");
     if (d->n_keys < 1 || d->n_keys > MAX_KW) FAIL(RW_E_INVAL, "n_keys");
-    if (!g_crc_table_init) {
-        uint32_t tab[256];
-        for (uint32_t i = 0; i < 256; i++) {
-            uint32_t c = i;
-            for (int k = 0; k < 8; k++)
-                c = (c & 1) ? 0xEDB88320u ^ (c >> 1) : c >> 1;
-            tab[i] = c;
-        }
-        HIP_TRY(hipMemcpyToSymbol(HIP_SYMBOL(g_crc_table), tab, sizeof tab));
-        g_crc_table_init = true;
-    }
+    if (ensure_crc_table() != RW_OK) FAIL(RW_E_INTERNAL, "crc table");
     uint32_t n = chunk->n_rows;
     VnodeBatch b{};
     uint8_t types[4] = {0, 0, 0, 0};
@@ -2526,6 +2532,101 @@ __global__ void agg_clean_kernel(AggTableDev t, int kpos, long long wm, int KW,
     }
 }
 
+// Rescale re-scoping (update_vnode_bitmap, state_table.rs + executor cache
+// eviction): state whose distribution-key vnode is no longer owned is
+// dropped WITHOUT emitting retractions — the new owner holds it. Dist key =
+// group key (agg) / join key (join). CRC feed identical to vnode_kernel.
+__device__ __forceinline__ uint32_t crc_key_words(const uint32_t* lut,
+                                                  const int64_t* kw,
+                                                  uint32_t nulls, int KW,
+                                                  const uint8_t* types) {
+    uint32_t crc = 0xFFFFFFFFu;
+    for (int k = 0; k < KW; k++) {
+        if ((nulls >> k) & 1) {
+            uint32_t sentinel = 0xfffffff0u;
+            crc = crc32_bytes(lut, crc, (const uint8_t*)&sentinel, 4);
+        } else if (types[k] == RW_T_I32) {
+            int32_t v = (int32_t)kw[k];
+            crc = crc32_bytes(lut, crc, (const uint8_t*)&v, 4);
+        } else if (types[k] == RW_T_BOOL) {
+            uint8_t v = (uint8_t)kw[k];
+            crc = crc32_bytes(lut, crc, &v, 1);
+        } else {
+            int64_t v = kw[k];
+            crc = crc32_bytes(lut, crc, (const uint8_t*)&v, 8);
+        }
+    }
+    return crc ^ 0xFFFFFFFFu;
+}
+
+__global__ void agg_vnode_scope_kernel(AggTableDev t, int KW, int n_calls,
+                                       AggCallDev c0, AggCallDev c1,
+                                       AggCallDev c2, AggCallDev c3,
+                                       const uint8_t* bitmap,
+                                       uint32_t vnode_count, uint8_t t0,
+                                       uint8_t t1, uint8_t t2, uint8_t t3) {
+    __shared__ uint32_t lut[256];
+    stage_crc_lut(lut);
+    AggCallDev calls[4] = {c0, c1, c2, c3};
+    uint8_t types[4] = {t0, t1, t2, t3};
+    size_t cap = (size_t)t.cap_mask + 1;
+    size_t stride = (size_t)gridDim.x * blockDim.x;
+    for (size_t slot = blockIdx.x * blockDim.x + threadIdx.x; slot < cap;
+         slot += stride) {
+        if (ld_u32(&t.state[(uint32_t)slot]) != SLOT_READY) continue;
+        uint32_t vn = crc_key_words(lut, &t.keys[slot * KW],
+                                    ld_u32(&t.key_nulls[(uint32_t)slot]), KW,
+                                    types) %
+                      vnode_count;
+        if ((bitmap[vn >> 3] >> (vn & 7)) & 1) continue; // still owned
+        for (int ci = 0; ci < n_calls; ci++) {
+            long long init = 0;
+            if (calls[ci].kind == RW_AGG_MIN) init = INT64_MAX;
+            if (calls[ci].kind == RW_AGG_MAX) init = INT64_MIN;
+            t.acc[(size_t)ci * cap + slot] = init;
+            t.has[(size_t)ci * cap + slot] = 0;
+            if (calls[ci].minput) {
+                uint32_t row = t.mheads[(size_t)calls[ci].mord * cap + slot];
+                while (row != UINT32_MAX) {
+                    st_u32(&t.malive[row], 0);
+                    row = ld_u32(&t.mnext[row]);
+                }
+                t.mheads[(size_t)calls[ci].mord * cap + slot] = UINT32_MAX;
+            }
+        }
+        t.has_prev[(uint32_t)slot] = 0;
+        t.dirty_flag[(uint32_t)slot] = 0;
+    }
+}
+
+__global__ void join_vnode_scope_kernel(JoinSideDev sd, int KW,
+                                        const uint8_t* bitmap,
+                                        uint32_t vnode_count, uint8_t t0,
+                                        uint8_t t1, uint8_t t2, uint8_t t3) {
+    __shared__ uint32_t lut[256];
+    stage_crc_lut(lut);
+    uint8_t types[4] = {t0, t1, t2, t3};
+    size_t cap = (size_t)sd.cap_mask + 1;
+    size_t stride = (size_t)gridDim.x * blockDim.x;
+    for (size_t slot = blockIdx.x * blockDim.x + threadIdx.x; slot < cap;
+         slot += stride) {
+        JoinSlot* sl = &sd.slots[slot];
+        if (ld_u32(&sl->state) != SLOT_READY) continue;
+        int64_t kw[4];
+        for (int k = 0; k < KW; k++) kw[k] = ld_i64((const int64_t*)&sl->key[k]);
+        uint32_t vn =
+            crc_key_words(lut, kw, ld_u32(&sl->nulls), KW, types) % vnode_count;
+        if ((bitmap[vn >> 3] >> (vn & 7)) & 1) continue;
+        uint32_t row = ld_u32(&sl->head);
+        while (row != UINT32_MAX) {
+            JoinRowHdr* h = jrow(sd, row);
+            st_u32(&h->alive, 0);
+            row = ld_u32(&h->next);
+        }
+        st_u32(&sl->head, UINT32_MAX);
+    }
+}
+
 struct HashJoin {
     RwHashJoinDesc desc;
     JoinMeta m{};
@@ -2539,6 +2640,7 @@ struct HashJoin {
     uint32_t stage_cap[2] = {0, 0};
     std::vector<RwChunk*> outq;
     uint8_t* zeros = nullptr; // all-zero null flags for pipeline dummy cols
+    uint8_t* d_vnode_bitmap = nullptr; // rescale scope (update_vnode_bitmap)
     // join-key watermark buffering (hash_join.rs:826-867)
     struct Wm { bool has = false; int64_t val = 0; };
     std::vector<uint32_t> wm_pos;
@@ -2954,6 +3056,51 @@ int rw_hash_agg_watermark(void* h, uint32_t group_key_pos, int64_t val) {
         agg->t, (int)group_key_pos, val, agg->KW, agg->n_calls, agg->cd(0),
         agg->cd(1), agg->cd(2), agg->cd(3));
     HIP_TRY(hipStreamSynchronize(agg->stream));
+    return RW_OK;
+}
+
+// rescale re-scope (update_vnode_bitmap): drop state whose dist-key vnode
+// (dist key = group key) is no longer owned; no retractions are emitted.
+// `bitmap` is vnode_count/8 bytes, LSB-first per byte (common Bitmap layout).
+int rw_hash_agg_update_vnode_bitmap(void* h, const uint8_t* bitmap,
+                                    uint32_t vnode_count) {
+    auto* agg = (HashAgg*)h;
+    if (!vnode_count || vnode_count % 8 || vnode_count > 4096)
+        FAIL(RW_E_INVAL, "vnode_count");
+    if (ensure_crc_table() != RW_OK) FAIL(RW_E_INTERNAL, "crc table");
+    if (!agg->d_vnode_bitmap)
+        HIP_TRY(hipMalloc(&agg->d_vnode_bitmap, 512));
+    HIP_TRY(hipMemcpyAsync(agg->d_vnode_bitmap, bitmap, vnode_count / 8,
+                           hipMemcpyHostToDevice, agg->stream));
+    uint8_t ty[4] = {RW_T_I64, RW_T_I64, RW_T_I64, RW_T_I64};
+    for (int k = 0; k < agg->KW && k < 4; k++) ty[k] = agg->out_types[k];
+    agg_vnode_scope_kernel<<<2048, 256, 0, agg->stream>>>(
+        agg->t, agg->KW, agg->n_calls, agg->cd(0), agg->cd(1), agg->cd(2),
+        agg->cd(3), agg->d_vnode_bitmap, vnode_count, ty[0], ty[1], ty[2],
+        ty[3]);
+    HIP_TRY(hipStreamSynchronize(agg->stream));
+    return RW_OK;
+}
+
+// join dist key = join key; both sides are re-scoped together
+int rw_hash_join_update_vnode_bitmap(void* h, const uint8_t* bitmap,
+                                     uint32_t vnode_count) {
+    auto* j = (HashJoin*)h;
+    if (!vnode_count || vnode_count % 8 || vnode_count > 4096)
+        FAIL(RW_E_INVAL, "vnode_count");
+    if (ensure_crc_table() != RW_OK) FAIL(RW_E_INTERNAL, "crc table");
+    if (!j->d_vnode_bitmap) HIP_TRY(hipMalloc(&j->d_vnode_bitmap, 512));
+    HIP_TRY(hipMemcpyAsync(j->d_vnode_bitmap, bitmap, vnode_count / 8,
+                           hipMemcpyHostToDevice, j->stream));
+    for (int s = 0; s < 2; s++) {
+        uint8_t ty[4] = {RW_T_I64, RW_T_I64, RW_T_I64, RW_T_I64};
+        for (int k = 0; k < j->m.KW && k < 4; k++)
+            ty[k] = j->types[s][j->m.key_cols[s][k]];
+        join_vnode_scope_kernel<<<2048, 256, 0, j->stream>>>(
+            j->side[s], j->m.KW, j->d_vnode_bitmap, vnode_count, ty[0], ty[1],
+            ty[2], ty[3]);
+    }
+    HIP_TRY(hipStreamSynchronize(j->stream));
     return RW_OK;
 }
 RwChunk* rw_hash_join_poll(void* h) { return ((HashJoin*)h)->poll(); }

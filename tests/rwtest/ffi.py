@@ -231,6 +231,12 @@ class Lib:
                 C.POINTER(C.c_uint32), C.POINTER(C.c_int64), C.c_int]
             L.rw_hash_agg_watermark.restype = C.c_int
             L.rw_hash_agg_watermark.argtypes = [C.c_void_p, C.c_uint32, C.c_int64]
+            L.rw_hash_agg_update_vnode_bitmap.restype = C.c_int
+            L.rw_hash_agg_update_vnode_bitmap.argtypes = [
+                C.c_void_p, C.POINTER(C.c_uint8), C.c_uint32]
+            L.rw_hash_join_update_vnode_bitmap.restype = C.c_int
+            L.rw_hash_join_update_vnode_bitmap.argtypes = [
+                C.c_void_p, C.POINTER(C.c_uint8), C.c_uint32]
         except AttributeError:
             pass
 
@@ -313,6 +319,13 @@ class HashAgg:
         rc = self.lib.lib.rw_hash_agg_watermark(self.h, group_key_pos, val)
         if rc != 0:
             raise RuntimeError(f"agg watermark failed {rc}")
+
+    def update_vnode_bitmap(self, bitmap_bytes, vnode_count=256):
+        buf = (C.c_uint8 * len(bitmap_bytes))(*bitmap_bytes)
+        rc = self.lib.lib.rw_hash_agg_update_vnode_bitmap(self.h, buf,
+                                                          vnode_count)
+        if rc != 0:
+            raise RuntimeError(f"agg vnode bitmap failed {rc}")
 
     def close(self):
         if self.h:
@@ -404,6 +417,13 @@ class HashJoin:
         if n < 0:
             raise RuntimeError(f"watermark failed {n}: {self.lib.last_error()}")
         return [(int(cols[i]), int(vals[i])) for i in range(n)]
+
+    def update_vnode_bitmap(self, bitmap_bytes, vnode_count=256):
+        buf = (C.c_uint8 * len(bitmap_bytes))(*bitmap_bytes)
+        rc = self.lib.lib.rw_hash_join_update_vnode_bitmap(self.h, buf,
+                                                           vnode_count)
+        if rc != 0:
+            raise RuntimeError(f"join vnode bitmap failed {rc}")
 
     def close(self):
         if self.h:

@@ -572,3 +572,67 @@ def test_exchange_partition_parity():
     assert got == want
     agg.close()
     o.close()
+
+
+def test_update_vnode_bitmap_parity():
+    # rescale re-scope (update_vnode_bitmap): GPU vs oracle — drop half the
+    # vnodes mid-stream on both executors, keep pushing rows (including for
+    # dropped keys, which must restart), compare per-epoch multisets
+    import struct
+    import zlib
+
+    def owned_bitmap(mod):
+        bm = bytearray(32)
+        for v in range(256):
+            if v % 2 == mod:
+                bm[v >> 3] |= 1 << (v & 7)
+        return bytes(bm)
+
+    # --- agg ---
+    calls = [(AGG_COUNT_STAR, -1, T_I64), (AGG_SUM, 1, T_I64)]
+    ga = ffi.HashAgg(gpu(), [T_I64, T_I64], [0], calls, 0)
+    oa = ffi.HashAgg(ffi.oracle(), [T_I64, T_I64], [0], calls, 0)
+    rng = np.random.default_rng(77)
+    keys = rng.integers(0, 4000, 4096)
+    vals = rng.integers(1, 100, 4096)
+    c = mk_chunk([T_I64, T_I64], np.zeros(4096, np.uint8), [keys, vals])
+    for a in (ga, oa):
+        a.push(c)
+        a.flush(1)
+        a.poll_all()
+        a.update_vnode_bitmap(owned_bitmap(0))
+    keys2 = rng.integers(0, 4000, 4096)
+    vals2 = rng.integers(1, 100, 4096)
+    c2 = mk_chunk([T_I64, T_I64], np.zeros(4096, np.uint8), [keys2, vals2])
+    outs = []
+    for a in (ga, oa):
+        a.push(c2)
+        a.flush(2)
+        outs.append(rows_multiset(a.poll_all()))
+    assert outs[0] == outs[1]
+    ga.close()
+    oa.close()
+
+    # --- join ---
+    g, o = join_pair()
+    lk = rng.integers(0, 300, 1024)
+    lv = np.arange(1024)
+    rk = rng.integers(0, 300, 1024)
+    rv = np.arange(1024, 2048)
+    cl = mk_chunk([T_I64, T_I64], np.zeros(1024, np.uint8), [lk, lv])
+    cr = mk_chunk([T_I64, T_I64], np.zeros(1024, np.uint8), [rk, rv])
+    for j in (g, o):
+        j.push(SIDE_LEFT, cl)
+        j.push(SIDE_RIGHT, cr)
+        j.poll_all()
+        j.update_vnode_bitmap(owned_bitmap(1))
+    lk2 = rng.integers(0, 300, 1024)
+    lv2 = np.arange(2048, 3072)
+    cl2 = mk_chunk([T_I64, T_I64], np.zeros(1024, np.uint8), [lk2, lv2])
+    outs = []
+    for j in (g, o):
+        j.push(SIDE_LEFT, cl2)
+        outs.append(rows_multiset(j.poll_all()))
+    assert outs[0] == outs[1]
+    g.close()
+    o.close()

@@ -185,3 +185,47 @@ def test_dispatch_shard_union_equals_single():
         shard_in[1].append(outs[1])
     union = sorted(run(shard_in[0]) + run(shard_in[1]))
     assert union == single
+
+
+def test_update_vnode_bitmap_oracle():
+    # rescale re-scope: groups whose vnode bit is cleared are dropped with
+    # NO retraction; a later row for a dropped key restarts the group
+    # (row_count 0 -> n => Insert, agg_group.rs:131-165)
+    from rwtest.ffi import AGG_COUNT_STAR, AGG_SUM, rows_multiset
+
+    calls = [(AGG_COUNT_STAR, -1, T_I64), (AGG_SUM, 1, T_I64)]
+    a = ffi.HashAgg(oracle(), [T_I64, T_I64], [0], calls, 0)
+    keys = list(range(16))
+    c = from_pretty("I I\n" + "\n".join(f"+ {k} {10*k+1}" for k in keys))
+    a.push(c)
+    a.flush(1)
+    a.poll_all()
+
+    def vnode(k):
+        crc = zlib.crc32(struct.pack("<q", k))
+        return crc % 256
+
+    owned = bytearray(32)
+    keep = set()
+    for k in keys:
+        if vnode(k) % 2 == 0:  # keep even vnodes
+            keep.add(k)
+            owned[vnode(k) >> 3] |= 1 << (vnode(k) & 7)
+    assert 0 < len(keep) < len(keys)
+    a.update_vnode_bitmap(bytes(owned))
+    # push one more row per key: kept keys -> Update pair (count 1->2);
+    # dropped keys restart -> Insert with count 1
+    c2 = from_pretty("I I\n" + "\n".join(f"+ {k} {10*k+2}" for k in keys))
+    a.push(c2)
+    a.flush(2)
+    got = rows_multiset(a.poll_all())
+    want = []
+    for k in keys:
+        if k in keep:
+            want.append(("U-", (k, 1, 10 * k + 1)))
+            want.append(("U+", (k, 2, 20 * k + 3)))
+        else:
+            want.append(("+", (k, 1, 10 * k + 2)))
+    key = lambda r: (r[0], r[1])
+    assert got == sorted(want, key=key)
+    a.close()
