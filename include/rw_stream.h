@@ -181,6 +181,41 @@ int rw_hash_agg_update_vnode_bitmap(void* h, const uint8_t* bitmap,
 int rw_hash_join_update_vnode_bitmap(void* h, const uint8_t* bitmap,
                                      uint32_t vnode_count);
 
+/* ----- GroupTopN (top_n/group_top_n.rs) -----
+ *
+ * Per group (group_by cols), maintain the full ordered row set; the visible
+ * window is rows [offset, offset+limit) in cache-key order (cache key =
+ * order_by cols then the remaining storage-key cols, each asc/desc with the
+ * default NULLS treatment — sort_util.rs defaults: ASC NULLS LAST / DESC
+ * NULLS FIRST). Each push emits the ChangeBuffer-compacted window delta for
+ * the chunk (group_top_n.rs:168-239 + common/change_buffer.rs:76-187):
+ * delete+insert of the same cache key with a different row becomes a U-/U+
+ * pair; equal rows cancel. WITH_TIES is not implemented (the reference's
+ * default instantiation is WITH_TIES = false). */
+typedef struct RwGroupTopNDesc {
+    uint32_t n_cols;
+    const uint8_t* types; /* RwTypeId per input column */
+    uint32_t n_group_by;
+    const uint32_t* group_by;
+    uint32_t n_order_by;
+    const uint32_t* order_cols;
+    const uint8_t* order_desc; /* per order col: 1 = descending */
+    uint32_t n_rest; /* remaining storage-key cols (after group + order) */
+    const uint32_t* rest_cols;
+    const uint8_t* rest_desc;
+    uint64_t offset;
+    uint64_t limit; /* > 0 */
+    uint32_t chunk_size;
+    uint64_t state_capacity_hint; /* expected group count */
+    uint64_t row_capacity_hint;   /* expected resident rows */
+} RwGroupTopNDesc;
+
+void* rw_group_top_n_create(const RwGroupTopNDesc* desc);
+int rw_group_top_n_push_chunk(void* h, const RwChunk* chunk);
+int rw_group_top_n_flush(void* h, uint64_t epoch); /* state commit only */
+RwChunk* rw_group_top_n_poll(void* h);
+void rw_group_top_n_destroy(void* h);
+
 #ifdef __cplusplus
 }
 #endif

@@ -1,0 +1,192 @@
+// oracle/oracle_topn.cpp — CPU restatement of the reference's GroupTopN
+// executor, WITH_TIES = false:
+//
+//  - per-row application and window maintenance:
+//    stream/src/executor/top_n/group_top_n.rs:168-239 (apply_chunk: per row,
+//    group cache lookup → TopNCache insert/delete → TopNStaging)
+//  - window semantics: top_n/top_n_cache.rs:293-520 (TopNCache<false>): the
+//    low/middle/high caches are an LRU optimization over the state table;
+//    the observable behavior is "the window is rows [offset, offset+limit)
+//    of the group's rows in cache-key order", with deltas staged whenever
+//    the middle (= window) content changes
+//  - cache key order: order_by cols then the remaining storage-key cols
+//    (create_cache_key_serde, top_n/utils.rs:130-150), each asc/desc with
+//    sort_util.rs default NULLS (ASC last / DESC first — ordered_cmp)
+//  - per-chunk compaction: TopNStaging over ChangeBuffer
+//    (common/change_buffer.rs:76-187): per cache key, delete+insert with a
+//    different row → Update (U-/U+ pair); insert+delete → dropped; no-op
+//    updates filtered (into_records:182-190)
+//  - output: one StreamChunkBuilder per push (group_top_n.rs:231-239),
+//    chunked at chunk_size with U-pair adjacency
+//
+// This oracle keeps the full ordered set per group (std::map) and computes
+// the window delta per push; identical to the incremental cache by
+// construction (the reference's caches are exact once synced).
+#include <cstring>
+#include <map>
+#include <memory>
+#include <string>
+#include <vector>
+
+#include "../include/rw_chunk.h"
+#include "../include/rw_stream.h"
+#include "common.hpp"
+
+namespace orc {
+
+extern thread_local std::string g_err; // defined in oracle_agg.cpp
+
+struct GroupTopNOracle {
+    std::vector<uint8_t> types;
+    std::vector<uint32_t> group_by;
+    std::vector<uint32_t> ck_cols; // order_by ∥ rest storage-key cols
+    RowOrderLess ck_less;          // over the ck_cols projection
+    RowOrderLess group_less;
+    uint64_t offset, limit;
+    uint32_t chunk_size;
+
+    using Group = std::map<Row, Row, RowOrderLess>; // cache key → full row
+    std::map<Row, Group, RowOrderLess> groups;
+    std::vector<std::unique_ptr<OwnedChunk>> outputs;
+
+    GroupTopNOracle(const RwGroupTopNDesc* d) {
+        types.assign(d->types, d->types + d->n_cols);
+        group_by.assign(d->group_by, d->group_by + d->n_group_by);
+        for (uint32_t i = 0; i < d->n_order_by; i++) {
+            ck_cols.push_back(d->order_cols[i]);
+            ck_less.order.push_back(
+                {types[d->order_cols[i]], d->order_desc[i] != 0});
+        }
+        for (uint32_t i = 0; i < d->n_rest; i++) {
+            ck_cols.push_back(d->rest_cols[i]);
+            ck_less.order.push_back(
+                {types[d->rest_cols[i]], d->rest_desc[i] != 0});
+        }
+        for (auto g : group_by) group_less.order.push_back({types[g], false});
+        offset = d->offset;
+        limit = d->limit;
+        chunk_size = d->chunk_size ? d->chunk_size : 1024;
+        groups = decltype(groups)(group_less);
+    }
+
+    Row project(const Row& row, const std::vector<uint32_t>& idx) const {
+        Row out;
+        out.reserve(idx.size());
+        for (auto i : idx) out.push_back(row[i]);
+        return out;
+    }
+
+    std::vector<Row> window_of(const Group& g) const {
+        std::vector<Row> w; // cache keys only
+        uint64_t i = 0;
+        for (auto& kv : g) {
+            if (i >= offset + limit) break;
+            if (i >= offset) w.push_back(kv.first);
+            i++;
+        }
+        return w;
+    }
+
+    int push_chunk(const RwChunk* c) {
+        ChunkView cv{c};
+        // group-key first-touch order; per group the pre-chunk window is
+        // snapshotted as (cache key, full row) pairs before any mutation
+        std::vector<Row> touched;
+        std::map<Row, std::vector<std::pair<Row, Row>>, RowOrderLess> old_win(
+            group_less);
+        for (size_t r = 0; r < cv.n_rows(); r++) {
+            if (!cv.visible(r)) continue;
+            Row row = cv.row(r);
+            Row gk = project(row, group_by);
+            if (!old_win.count(gk)) {
+                touched.push_back(gk);
+                std::vector<std::pair<Row, Row>> w;
+                auto it = groups.find(gk);
+                if (it != groups.end()) {
+                    uint64_t i = 0;
+                    for (auto& kv : it->second) {
+                        if (i >= offset + limit) break;
+                        if (i >= offset) w.emplace_back(kv.first, kv.second);
+                        i++;
+                    }
+                }
+                old_win.emplace(gk, std::move(w));
+            }
+            Row ck = project(row, ck_cols);
+            uint8_t op = cv.op(r);
+            auto& g = groups.try_emplace(gk, Group(ck_less)).first->second;
+            if (op == RW_OP_INSERT || op == RW_OP_UPDATE_INSERT) {
+                g[ck] = row;
+            } else {
+                auto it = g.find(ck);
+                if (it != g.end()) g.erase(it);
+            }
+        }
+        // per touched group: merge-diff old vs new window with ChangeBuffer
+        // merge rules (delete+insert same key, different row -> U-pair)
+        ChunkBuilder cb(chunk_size, types);
+        std::unique_ptr<OwnedChunk> done;
+        auto emit = [&](uint8_t op, const Row& row) {
+            if (cb.append_row(op, row, &done)) outputs.push_back(std::move(done));
+        };
+        for (auto& gk : touched) {
+            auto git = groups.find(gk);
+            std::vector<Row> neww = window_of(git->second);
+            const auto& oldw = old_win.at(gk);
+            size_t i = 0, j = 0;
+            while (i < oldw.size() || j < neww.size()) {
+                int c2;
+                if (i >= oldw.size()) c2 = 1;
+                else if (j >= neww.size()) c2 = -1;
+                else c2 = ordered_cmp(oldw[i].first, neww[j], ck_less.order);
+                if (c2 == 0) {
+                    const Row& nr = git->second.at(neww[j]);
+                    if (!row_eq(oldw[i].second, nr, types)) {
+                        emit(RW_OP_UPDATE_DELETE, oldw[i].second);
+                        emit(RW_OP_UPDATE_INSERT, nr);
+                    }
+                    i++;
+                    j++;
+                } else if (c2 < 0) {
+                    emit(RW_OP_DELETE, oldw[i].second);
+                    i++;
+                } else {
+                    emit(RW_OP_INSERT, git->second.at(neww[j]));
+                    j++;
+                }
+            }
+        }
+        if (auto rest = cb.take()) outputs.push_back(std::move(rest));
+        return RW_OK;
+    }
+
+    RwChunk* poll() {
+        if (outputs.empty()) return nullptr;
+        auto c = std::move(outputs.front());
+        outputs.erase(outputs.begin());
+        return chunk_to_c(*c);
+    }
+};
+
+} // namespace orc
+
+using namespace orc;
+
+extern "C" {
+
+void* rw_group_top_n_create(const RwGroupTopNDesc* d) {
+    if (!d || !d->limit) return nullptr;
+    return new GroupTopNOracle(d);
+}
+int rw_group_top_n_push_chunk(void* h, const RwChunk* c) {
+    return ((GroupTopNOracle*)h)->push_chunk(c);
+}
+int rw_group_top_n_flush(void* h, uint64_t epoch) {
+    (void)h;
+    (void)epoch;
+    return RW_OK; // emission is per push; state commit is a no-op here
+}
+RwChunk* rw_group_top_n_poll(void* h) { return ((GroupTopNOracle*)h)->poll(); }
+void rw_group_top_n_destroy(void* h) { delete (GroupTopNOracle*)h; }
+
+} // extern "C"

@@ -462,3 +462,102 @@ def oracle():
             subprocess.run(["make", "-C", os.path.join(REPO, "oracle")], check=True)
         _oracle = Lib(path)
     return _oracle
+
+
+class RwGroupTopNDesc(C.Structure):
+    _fields_ = [
+        ("n_cols", C.c_uint32), ("types", C.POINTER(C.c_uint8)),
+        ("n_group_by", C.c_uint32), ("group_by", C.POINTER(C.c_uint32)),
+        ("n_order_by", C.c_uint32), ("order_cols", C.POINTER(C.c_uint32)),
+        ("order_desc", C.POINTER(C.c_uint8)),
+        ("n_rest", C.c_uint32), ("rest_cols", C.POINTER(C.c_uint32)),
+        ("rest_desc", C.POINTER(C.c_uint8)),
+        ("offset", C.c_uint64), ("limit", C.c_uint64),
+        ("chunk_size", C.c_uint32),
+        ("state_capacity_hint", C.c_uint64), ("row_capacity_hint", C.c_uint64),
+    ]
+
+
+class GroupTopN:
+    """GroupTopN executor wrapper (group_top_n.rs; WITH_TIES=false).
+
+    order_by / rest: lists of (col_idx, desc) pairs; rest = the remaining
+    storage-key columns after group_by and order_by."""
+
+    def __init__(self, lib: Lib, input_types, group_by, order_by, rest,
+                 offset=0, limit=1, chunk_size=1024, state_capacity_hint=0,
+                 row_capacity_hint=0):
+        self.lib = lib
+        L = lib.lib
+        L.rw_group_top_n_create.restype = C.c_void_p
+        L.rw_group_top_n_create.argtypes = [C.POINTER(RwGroupTopNDesc)]
+        L.rw_group_top_n_push_chunk.restype = C.c_int
+        L.rw_group_top_n_push_chunk.argtypes = [C.c_void_p,
+                                                C.POINTER(RwChunkC)]
+        L.rw_group_top_n_flush.restype = C.c_int
+        L.rw_group_top_n_flush.argtypes = [C.c_void_p, C.c_uint64]
+        L.rw_group_top_n_poll.restype = C.POINTER(RwChunkC)
+        L.rw_group_top_n_poll.argtypes = [C.c_void_p]
+        L.rw_group_top_n_destroy.restype = None
+        L.rw_group_top_n_destroy.argtypes = [C.c_void_p]
+        d = RwGroupTopNDesc()
+        self._keep = []
+
+        def u32s(v):
+            a = (C.c_uint32 * max(len(v), 1))(*v)
+            self._keep.append(a)
+            return a
+
+        def u8s(v):
+            a = (C.c_uint8 * max(len(v), 1))(*v)
+            self._keep.append(a)
+            return a
+
+        d.n_cols = len(input_types)
+        d.types = u8s(input_types)
+        d.n_group_by = len(group_by)
+        d.group_by = u32s(group_by)
+        d.n_order_by = len(order_by)
+        d.order_cols = u32s([c for c, _ in order_by])
+        d.order_desc = u8s([1 if desc else 0 for _, desc in order_by])
+        d.n_rest = len(rest)
+        d.rest_cols = u32s([c for c, _ in rest])
+        d.rest_desc = u8s([1 if desc else 0 for _, desc in rest])
+        d.offset = offset
+        d.limit = limit
+        d.chunk_size = chunk_size
+        d.state_capacity_hint = state_capacity_hint
+        d.row_capacity_hint = row_capacity_hint
+        self.h = L.rw_group_top_n_create(C.byref(d))
+        if not self.h:
+            raise RuntimeError(f"group_top_n create failed: {lib.last_error()}")
+
+    def push(self, chunk):
+        cc = chunk.to_c()
+        rc = self.lib.lib.rw_group_top_n_push_chunk(self.h, C.byref(cc))
+        if rc != 0:
+            raise RuntimeError(f"push failed {rc}: {self.lib.last_error()}")
+
+    def flush(self, epoch):
+        rc = self.lib.lib.rw_group_top_n_flush(self.h, epoch)
+        if rc != 0:
+            raise RuntimeError(f"flush failed {rc}: {self.lib.last_error()}")
+
+    def poll_all(self):
+        out = []
+        while True:
+            p = self.lib.lib.rw_group_top_n_poll(self.h)
+            if not p:
+                return out
+            out.append(self.lib._read_chunk(p))
+
+    def close(self):
+        if self.h:
+            self.lib.lib.rw_group_top_n_destroy(self.h)
+            self.h = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass

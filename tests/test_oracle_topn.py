@@ -1,0 +1,197 @@
+"""GroupTopN oracle vs the reference's own test fixtures (CPU).
+
+Golden vectors transcribed from
+stream/src/executor/top_n/group_top_n.rs tests (sort_rows comparison =
+per-push multiset):
+  - test_without_offset_and_with_limits (storage_key [1,2,0] asc,
+    group_by [1], order_by [2], offset 0 limit 2)
+  - test_with_offset_and_with_limits (offset 1 limit 2)
+  - test_multi_group_key (group_by [1,2], order_by [0], offset 0 limit 2)
+  - test_compact_changes (ChangeBuffer merge incl. the no-output chunk)
+"""
+import numpy as np
+import pytest
+
+from rwtest import ffi
+from rwtest.ffi import T_I64, from_pretty, oracle, rows_multiset
+
+I3 = [T_I64, T_I64, T_I64]
+
+
+def chunks_0_3():
+    return [
+        from_pretty("""  I I I
+            + 10 9 1
+            +  8 8 2
+            +  7 8 2
+            +  9 1 1
+            + 10 1 1
+            +  8 1 3"""),
+        from_pretty("""  I I I
+            - 10 9 1
+            -  8 8 2
+            - 10 1 1"""),
+        from_pretty(""" I I I
+            - 7 8 2
+            - 8 1 3
+            - 9 1 1"""),
+        from_pretty("""  I I I
+            +  5 1 1
+            +  2 1 1
+            +  3 1 2
+            +  4 1 3"""),
+    ]
+
+
+def expect(t, pretty):
+    got = rows_multiset(t.poll_all())
+    want = rows_multiset([from_pretty(pretty)]) if pretty else []
+    assert got == want, f"got {got}\nwant {want}"
+
+
+def make(lib, offset, limit, group_by, order_by):
+    # storage key = group_by ++ order_by ++ rest of stream key [1,2,0]
+    sk = [1, 2, 0]
+    rest = [(c, False) for c in sk if c not in group_by
+            and c not in [o for o, _ in order_by]]
+    return ffi.GroupTopN(lib, I3, group_by, order_by, rest,
+                         offset=offset, limit=limit)
+
+
+def test_without_offset_and_with_limits():
+    t = make(oracle(), 0, 2, [1], [(2, False)])
+    cs = chunks_0_3()
+    t.push(cs[0])
+    expect(t, """  I I I
+        + 10 9 1
+        +  8 8 2
+        +  7 8 2
+        +  9 1 1
+        + 10 1 1""")
+    t.push(cs[1])
+    expect(t, """  I I I
+        - 10 9 1
+        -  8 8 2
+        - 10 1 1
+        +  8 1 3""")
+    t.push(cs[2])
+    expect(t, """ I I I
+        - 7 8 2
+        - 8 1 3
+        - 9 1 1""")
+    t.push(cs[3])
+    expect(t, """ I I I
+        + 5 1 1
+        + 2 1 1""")
+    t.close()
+
+
+def test_with_offset_and_with_limits():
+    t = make(oracle(), 1, 2, [1], [(2, False)])
+    cs = chunks_0_3()
+    t.push(cs[0])
+    expect(t, """  I I I
+        +  8 8 2
+        + 10 1 1
+        +  8 1 3""")
+    t.push(cs[1])
+    expect(t, """  I I I
+        -  8 8 2
+        - 10 1 1""")
+    t.push(cs[2])
+    expect(t, """ I I I
+        - 8 1 3""")
+    t.push(cs[3])
+    expect(t, """ I I I
+        + 5 1 1
+        + 3 1 2""")
+    t.close()
+
+
+def test_multi_group_key():
+    t = make(oracle(), 0, 2, [1, 2], [(0, False)])
+    cs = chunks_0_3()
+    t.push(cs[0])
+    expect(t, """  I I I
+        + 10 9 1
+        +  8 8 2
+        +  7 8 2
+        +  9 1 1
+        + 10 1 1
+        +  8 1 3""")
+    t.push(cs[1])
+    expect(t, """  I I I
+        - 10 9 1
+        -  8 8 2
+        - 10 1 1""")
+    t.push(cs[2])
+    expect(t, """  I I I
+        - 7 8 2
+        - 8 1 3
+        - 9 1 1""")
+    t.push(cs[3])
+    expect(t, """  I I I
+        +  5 1 1
+        +  2 1 1
+        +  3 1 2
+        +  4 1 3""")
+    t.close()
+
+
+def test_compact_changes():
+    # group_by [0,1], order_by [2]; storage key = [0,1,2]
+    t = ffi.GroupTopN(oracle(), I3, [0, 1], [(2, False)], [],
+                      offset=0, limit=2)
+    t.push(from_pretty("""  I I I
+        +  0 0 9
+        +  0 0 8
+        +  0 0 7
+        +  0 0 6
+        +  0 1 15
+        +  0 1 14"""))
+    expect(t, """  I I I
+        +  0 0 7
+        +  0 0 6
+        +  0 1 15
+        +  0 1 14""")
+    t.push(from_pretty("""  I I I
+        -  0 0 6
+        -  0 0 8
+        +  0 0 4
+        +  0 0 3
+        +  0 1 12
+        +  0 2 26
+        -  0 1 12
+        +  0 1 11"""))
+    expect(t, """  I I I
+        -  0 0 6
+        -  0 0 7
+        +  0 0 4
+        +  0 0 3
+        -  0 1 15
+        +  0 1 11
+        +  0 2 26""")
+    t.push(from_pretty("""  I I I
+        +  0 0 11"""))
+    expect(t, "")  # no chunk output
+    t.close()
+
+
+def test_topn_delete_update_pair():
+    # delete old + insert new with the same cache key in one chunk → U-pair
+    # (ChangeBuffer delete-then-insert merge, change_buffer.rs:76-120); needs
+    # a non-key payload column: schema (k, ord, payload), storage key (ord),
+    # group (k)
+    t = ffi.GroupTopN(oracle(), I3, [0], [(1, False)], [], offset=0, limit=2)
+    t.push(from_pretty(" I I I\n + 1 5 100\n + 1 6 200"))
+    t.poll_all()
+    t.push(from_pretty(" I I I\n - 1 5 100\n + 1 5 999"))
+    outs = t.poll_all()
+    rows = []
+    for c in outs:
+        rows.extend(c.visible_rows())
+    assert ("U-", (1, 5, 100)) in rows and ("U+", (1, 5, 999)) in rows
+    # adjacency: U- immediately followed by U+
+    i = rows.index(("U-", (1, 5, 100)))
+    assert rows[i + 1] == ("U+", (1, 5, 999))
+    t.close()
