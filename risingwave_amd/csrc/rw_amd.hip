@@ -3232,3 +3232,613 @@ int rw_join_stats_reset(void* h) {
 }
 
 } // extern "C"
+
+// ===================== GroupTopN =====================
+//
+// GPU restatement of stream/src/executor/top_n/group_top_n.rs (WITH_TIES =
+// false). Per group the full row set lives in HBM (AoS slot + packed
+// records, shared layout with the join side); the observable semantics are
+// those of TopNCache<false> (top_n_cache.rs:293-520): the visible window is
+// rows [offset, offset+limit) of the group in cache-key order, and each
+// pushed chunk emits the ChangeBuffer-compacted window delta
+// (change_buffer.rs:76-187: replace = U-pair, equal rows cancel).
+//
+// Kernel plan per push (stream-ordered):
+//   1. touch    — find-or-insert group slots, collect the touched list
+//   2. snapshot — per touched group, select the pre-chunk window (row ids)
+//   3. apply    — inserts (upsert = kill old + append) / deletes; same-
+//                 cache-key conflicts within the chunk run in single-row
+//                 segments (host pre-pass, as in the agg/join executors)
+//   4. emit     — per touched group, select the new window, merge-diff
+//                 against the snapshot, reserve + write output rows
+// Parallelism is across groups (2+4) and rows (1+3). Records are never
+// overwritten in place, so snapshot row ids stay readable in step 4.
+
+#define TOPN_MAX_CK 8
+#define TOPN_MAX_WIN 128
+
+struct TopMeta {
+    int KW;     // group key width
+    int n_cols; // input schema width
+    uint8_t gk_cols[MAX_KW];
+    int n_ck; // cache key = order_by then rest storage-key cols
+    uint8_t ck_cols[TOPN_MAX_CK];
+    uint8_t ck_desc[TOPN_MAX_CK];
+    uint8_t ck_float[TOPN_MAX_CK];
+    uint32_t offset, limit;
+};
+
+// cache-key compare on packed records (ordered_cmp semantics: NULLs
+// largest in value order, desc reverses, NaN largest among floats)
+__device__ __forceinline__ int topn_cmp(const TopMeta& m, const long long* va,
+                                        uint32_t nba, const long long* vb,
+                                        uint32_t nbb) {
+    for (int i = 0; i < m.n_ck; i++) {
+        int c = m.ck_cols[i];
+        bool na = !((nba >> c) & 1), nb = !((nbb >> c) & 1);
+        int r;
+        if (na || nb) {
+            r = (na && nb) ? 0 : (na ? 1 : -1);
+        } else if (m.ck_float[i]) {
+            double da = __longlong_as_double(va[c]);
+            double db = __longlong_as_double(vb[c]);
+            bool an = isnan(da), bn = isnan(db);
+            if (an || bn) r = an == bn ? 0 : (an ? 1 : -1);
+            else r = da < db ? -1 : (da > db ? 1 : 0);
+        } else {
+            r = va[c] < vb[c] ? -1 : (va[c] > vb[c] ? 1 : 0);
+        }
+        if (m.ck_desc[i]) r = -r;
+        if (r) return r;
+    }
+    return 0;
+}
+
+__device__ __forceinline__ bool topn_ck_eq_batch(const TopMeta& m,
+                                                 const JoinBatchDev& b,
+                                                 uint32_t r, const long long* v,
+                                                 uint32_t nb) {
+    for (int i = 0; i < m.n_ck; i++) {
+        int c = m.ck_cols[i];
+        bool bn = !b.col_valid[c][r];
+        bool rn = !((nb >> c) & 1);
+        if (bn != rn) return false;
+        if (!bn && b.col_vals[c][r] != v[c]) return false;
+    }
+    return true;
+}
+
+__global__ void topn_touch_kernel(JoinBatchDev b, JoinSideDev sd, TopMeta m,
+                                  uint32_t* touched, uint32_t* touched_list,
+                                  uint32_t* counters /*0=tcur 1=err*/) {
+    uint32_t stride = gridDim.x * blockDim.x;
+    for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < b.n_rows;
+         r += stride) {
+        if (b.vis && !b.vis[r]) continue;
+        int64_t kw[MAX_KW];
+        uint32_t nullmask = 0;
+        for (int i = 0; i < m.KW; i++) {
+            uint8_t col = m.gk_cols[i];
+            bool valid = b.col_valid[col][r];
+            kw[i] = valid ? b.col_vals[col][r] : 0;
+            nullmask |= (!valid) << i;
+        }
+        uint32_t slot =
+            jslot_find_or_insert(sd.slots, sd.cap_mask, kw, nullmask, m.KW);
+        if (slot == UINT32_MAX) {
+            atomicExch(&counters[1], 2u);
+            continue;
+        }
+        if (ld_u32(&touched[slot]) == 0 &&
+            atomicCAS(&touched[slot], 0u, 1u) == 0u)
+            touched_list[atomicAdd(&counters[0], 1u)] = slot;
+    }
+}
+
+// select the first (offset+limit) rows of a chain in cache-key order;
+// returns count. O(chain * K) insertion selection — the window is small.
+__device__ int topn_select(const JoinSideDev& sd, const TopMeta& m,
+                           uint32_t slot, uint32_t* sel, int K) {
+    int n = 0;
+    uint32_t row = sd.slots[slot].head;
+    while (row != UINT32_MAX) {
+        JoinRowHdr* h = jrow(sd, row);
+        if (h->alive) {
+            const long long* v = jvals(h);
+            int pos = n;
+            while (pos > 0) {
+                JoinRowHdr* hp = jrow(sd, sel[pos - 1]);
+                if (topn_cmp(m, jvals(hp), hp->validbits, v, h->validbits) <= 0)
+                    break;
+                pos--;
+            }
+            if (pos < K) {
+                int last = n < K ? n : K - 1;
+                for (int k = last; k > pos; k--) sel[k] = sel[k - 1];
+                sel[pos] = row;
+                if (n < K) n++;
+            }
+        }
+        row = h->next;
+    }
+    return n;
+}
+
+__global__ void topn_snapshot_kernel(JoinSideDev sd, TopMeta m,
+                                     const uint32_t* touched_list,
+                                     const uint32_t* counters,
+                                     uint32_t* old_win, uint32_t* old_n) {
+    int K = (int)(m.offset + m.limit);
+    uint32_t nt = counters[0];
+    uint32_t stride = gridDim.x * blockDim.x;
+    for (uint32_t t = blockIdx.x * blockDim.x + threadIdx.x; t < nt;
+         t += stride) {
+        uint32_t sel[TOPN_MAX_WIN];
+        int n = topn_select(sd, m, touched_list[t], sel, K);
+        old_n[t] = (uint32_t)n;
+        for (int i = 0; i < n; i++) old_win[(size_t)t * K + i] = sel[i];
+    }
+}
+
+__global__ void topn_apply_kernel(JoinBatchDev b, JoinSideDev sd, TopMeta m,
+                                  uint32_t r0, uint32_t r1,
+                                  uint32_t* counters /*1=err*/) {
+    uint32_t stride = gridDim.x * blockDim.x;
+    for (uint32_t r = r0 + blockIdx.x * blockDim.x + threadIdx.x; r < r1;
+         r += stride) {
+        if (b.vis && !b.vis[r]) continue;
+        int64_t kw[MAX_KW];
+        uint32_t nullmask = 0;
+        for (int i = 0; i < m.KW; i++) {
+            uint8_t col = m.gk_cols[i];
+            bool valid = b.col_valid[col][r];
+            kw[i] = valid ? b.col_vals[col][r] : 0;
+            nullmask |= (!valid) << i;
+        }
+        uint32_t slot =
+            jslot_find_cached(sd.slots, sd.cap_mask, kw, nullmask, m.KW);
+        if (slot == UINT32_MAX) continue; // touch pass created it
+        uint8_t op = b.ops[r];
+        bool is_insert = (op == RW_OP_INSERT || op == RW_OP_UPDATE_INSERT);
+        // find the alive row with this cache key (sc1 loads: the chain
+        // mutates within this launch; same-ck rows run in their own
+        // segments, so the kill/append below is race-free)
+        uint32_t row = ld_u32(&sd.slots[slot].head);
+        while (row != UINT32_MAX) {
+            JoinRowHdr* h = jrow(sd, row);
+            uint32_t vb = ld_u32(&h->validbits);
+            if (ld_u32(&h->alive) &&
+                topn_ck_eq_batch(m, b, r, jvals(h), vb)) {
+                st_u32(&h->alive, 0); // upsert kills the old record
+                break;
+            }
+            row = ld_u32(&h->next);
+        }
+        if (!is_insert) continue;
+        uint32_t nrow = atomicAdd(sd.row_cursor, 1u);
+        if (nrow >= sd.row_cap) {
+            atomicExch(&counters[1], 3u);
+            continue;
+        }
+        JoinRowHdr* h = jrow(sd, nrow);
+        uint32_t vb = 0;
+        long long* hv = jvals(h);
+        for (int c = 0; c < m.n_cols; c++) {
+            st_i64((int64_t*)&hv[c], b.col_vals[c][r]);
+            vb |= (uint32_t)(b.col_valid[c][r] != 0) << c;
+        }
+        st_u32(&h->validbits, vb);
+        st_u32(&h->alive, 1);
+        uint32_t* headp = &sd.slots[slot].head;
+        uint32_t old_head = ld_u32(headp);
+        for (;;) {
+            st_u32(&h->next, old_head);
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); // R1 drain
+            uint32_t prev = atomicCAS(headp, old_head, nrow);
+            if (prev == old_head) break;
+            old_head = prev;
+        }
+    }
+}
+
+__global__ void topn_emit_kernel(JoinSideDev sd, TopMeta m,
+                                 const uint32_t* touched_list,
+                                 const uint32_t* tcounters, uint32_t* touched,
+                                 const uint32_t* old_win,
+                                 const uint32_t* old_n, JoinOutDev out,
+                                 int n_cols) {
+    int K = (int)(m.offset + m.limit);
+    uint32_t nt = tcounters[0];
+    uint32_t stride = gridDim.x * blockDim.x;
+    for (uint32_t t = blockIdx.x * blockDim.x + threadIdx.x; t < nt;
+         t += stride) {
+        uint32_t slot = touched_list[t];
+        touched[slot] = 0;
+        uint32_t neww[TOPN_MAX_WIN];
+        int nn = topn_select(sd, m, slot, neww, K);
+        const uint32_t* oldw = old_win + (size_t)t * K;
+        int no = (int)old_n[t];
+        // merge-diff into local delta lists: op + row id (+ paired old id)
+        uint8_t dop[2 * TOPN_MAX_WIN];
+        uint32_t drow[2 * TOPN_MAX_WIN];
+        int nd = 0;
+        int i = 0, j = 0;
+        while (i < no || j < nn) {
+            int c;
+            if (i >= no) c = 1;
+            else if (j >= nn) c = -1;
+            else {
+                JoinRowHdr* ho = jrow(sd, oldw[i]);
+                JoinRowHdr* hn = jrow(sd, neww[j]);
+                c = topn_cmp(m, jvals(ho), ho->validbits, jvals(hn),
+                             hn->validbits);
+            }
+            if (c == 0) {
+                if (oldw[i] != neww[j]) {
+                    // same cache key, different record: replaced. Equal
+                    // rows cancel (ChangeBuffer no-op filter).
+                    JoinRowHdr* ho = jrow(sd, oldw[i]);
+                    JoinRowHdr* hn = jrow(sd, neww[j]);
+                    bool eq = ho->validbits == hn->validbits;
+                    const long long* vo = jvals(ho);
+                    const long long* vn = jvals(hn);
+                    for (int cc = 0; eq && cc < n_cols; cc++)
+                        if (((ho->validbits >> cc) & 1) && vo[cc] != vn[cc])
+                            eq = false;
+                    if (!eq) {
+                        dop[nd] = RW_OP_UPDATE_DELETE;
+                        drow[nd++] = oldw[i];
+                        dop[nd] = RW_OP_UPDATE_INSERT;
+                        drow[nd++] = neww[j];
+                    }
+                }
+                i++;
+                j++;
+            } else if (c < 0) {
+                dop[nd] = RW_OP_DELETE;
+                drow[nd++] = oldw[i];
+                i++;
+            } else {
+                dop[nd] = RW_OP_INSERT;
+                drow[nd++] = neww[j];
+                j++;
+            }
+        }
+        if (!nd) continue;
+        uint32_t base = atomicAdd(&out.counters[0], (uint32_t)nd);
+        if (base + nd > out.cap) {
+            atomicExch(&out.counters[1], 1u);
+            continue;
+        }
+        for (int k = 0; k < nd; k++) {
+            uint32_t orow = base + k;
+            JoinRowHdr* h = jrow(sd, drow[k]);
+            const long long* v = jvals(h);
+            out.ops[orow] = dop[k];
+            for (int c = 0; c < n_cols; c++) {
+                bool valid = (h->validbits >> c) & 1;
+                out.vals[(size_t)orow * n_cols + c] = valid ? v[c] : 0;
+                out.nulls[(size_t)orow * n_cols + c] = !valid;
+            }
+        }
+    }
+}
+
+struct GroupTopN {
+    TopMeta m{};
+    JoinSideDev sd{};
+    JoinBatchDev stage{};
+    uint32_t stage_cap = 0;
+    JoinOutDev out{};
+    uint32_t* touched = nullptr;     // [slot cap]
+    uint32_t* touched_list = nullptr;
+    uint32_t* tcounters = nullptr; // 0=tcursor 1=err
+    uint32_t* old_win = nullptr;
+    uint32_t* old_n = nullptr;
+    uint32_t old_cap = 0; // touched capacity for old_win/old_n
+    std::vector<uint8_t> types;
+    std::vector<uint32_t> ck_col_idx;
+    uint32_t chunk_size;
+    hipStream_t stream = nullptr;
+    std::vector<RwChunk*> outq;
+
+    int init(const RwGroupTopNDesc* d) {
+        if (!gpu_ok())
+            FAIL(RW_E_NOGPU,
+                 "risingwave_amd: no GPU visible (product path has no CPU fallback)");
+        if (!d->limit) FAIL(RW_E_INVAL, "limit must be > 0");
+        if (d->offset + d->limit > TOPN_MAX_WIN)
+            FAIL(RW_E_INVAL, "offset+limit > %d unsupported", TOPN_MAX_WIN);
+        if (d->n_group_by < 1 || d->n_group_by > MAX_KW)
+            FAIL(RW_E_INVAL, "group_by width 1..%d", MAX_KW);
+        if (d->n_order_by + d->n_rest > TOPN_MAX_CK)
+            FAIL(RW_E_INVAL, "cache key width > %d", TOPN_MAX_CK);
+        if (d->n_cols > MAX_COLS) FAIL(RW_E_INVAL, "n_cols > %d", MAX_COLS);
+        types.assign(d->types, d->types + d->n_cols);
+        for (auto t : types)
+            if (t != RW_T_I64 && t != RW_T_TS && t != RW_T_F64)
+                FAIL(RW_E_INVAL, "GroupTopN supports 8-byte datum columns");
+        m.KW = (int)d->n_group_by;
+        m.n_cols = (int)d->n_cols;
+        for (uint32_t i = 0; i < d->n_group_by; i++)
+            m.gk_cols[i] = (uint8_t)d->group_by[i];
+        m.n_ck = (int)(d->n_order_by + d->n_rest);
+        int k = 0;
+        for (uint32_t i = 0; i < d->n_order_by; i++, k++) {
+            m.ck_cols[k] = (uint8_t)d->order_cols[i];
+            m.ck_desc[k] = d->order_desc[i];
+            m.ck_float[k] = types[d->order_cols[i]] == RW_T_F64;
+            ck_col_idx.push_back(d->order_cols[i]);
+        }
+        for (uint32_t i = 0; i < d->n_rest; i++, k++) {
+            m.ck_cols[k] = (uint8_t)d->rest_cols[i];
+            m.ck_desc[k] = d->rest_desc[i];
+            m.ck_float[k] = types[d->rest_cols[i]] == RW_T_F64;
+            ck_col_idx.push_back(d->rest_cols[i]);
+        }
+        m.offset = (uint32_t)d->offset;
+        m.limit = (uint32_t)d->limit;
+        chunk_size = d->chunk_size ? d->chunk_size : 1024;
+
+        uint64_t cap_hint = d->state_capacity_hint ? d->state_capacity_hint
+                                                   : (1ull << 14);
+        size_t cap = 1;
+        while (cap < cap_hint * 2) cap <<= 1;
+        sd.cap_mask = (uint32_t)(cap - 1);
+        uint64_t row_cap = d->row_capacity_hint ? d->row_capacity_hint
+                                                : (1ull << 20);
+        sd.row_stride = 16 + 8u * (uint32_t)m.n_cols;
+        sd.row_cap = (uint32_t)row_cap;
+        HIP_TRY(hipStreamCreate(&stream));
+        HIP_TRY(hipMalloc(&sd.slots, cap * sizeof(JoinSlot)));
+        HIP_TRY(hipMalloc(&sd.rows, (size_t)row_cap * sd.row_stride));
+        HIP_TRY(hipMalloc(&sd.row_cursor, 4));
+        HIP_TRY(hipMemset(sd.row_cursor, 0, 4));
+        HIP_TRY(hipMalloc(&touched, cap * 4));
+        HIP_TRY(hipMemset(touched, 0, cap * 4));
+        HIP_TRY(hipMalloc(&tcounters, 8));
+        jslot_init_kernel<<<2048, 256, 0, stream>>>(sd.slots, cap);
+        uint32_t out_cap = 1 << 18;
+        out.cap = out_cap;
+        HIP_TRY(hipMalloc(&out.vals, (size_t)out_cap * m.n_cols * 8));
+        HIP_TRY(hipMalloc(&out.nulls, (size_t)out_cap * m.n_cols));
+        HIP_TRY(hipMalloc(&out.ops, out_cap));
+        HIP_TRY(hipMalloc(&out.counters, 8));
+        HIP_TRY(hipMemset(out.counters, 0, 8));
+        HIP_TRY(hipStreamSynchronize(stream));
+        return RW_OK;
+    }
+
+    ~GroupTopN() {
+        if (sd.slots) {
+            hipFree(sd.slots);
+            hipFree(sd.rows);
+            hipFree(sd.row_cursor);
+            hipFree(touched);
+            hipFree(tcounters);
+            hipFree(out.vals);
+            hipFree(out.nulls);
+            hipFree(out.ops);
+            hipFree(out.counters);
+        }
+        if (touched_list) hipFree(touched_list);
+        if (old_win) hipFree(old_win);
+        free_stage();
+        if (stream) hipStreamDestroy(stream);
+        for (auto* c : outq) {
+            for (uint32_t i = 0; i < c->n_cols; i++) {
+                delete[] (int64_t*)c->cols[i].data;
+                delete[] (uint8_t*)c->cols[i].valid;
+            }
+            delete[] c->cols;
+            delete[] c->ops;
+            delete c;
+        }
+    }
+
+    void free_stage() {
+        if (!stage_cap) return;
+        for (int i = 0; i < m.n_cols; i++) {
+            hipFree(stage.col_vals[i]);
+            hipFree(stage.col_valid[i]);
+        }
+        hipFree(stage.ops);
+        hipFree(stage.vis);
+        stage_cap = 0;
+    }
+
+    int ensure_caps(uint32_t n) {
+        if (stage_cap < n) {
+            free_stage();
+            for (int i = 0; i < m.n_cols; i++) {
+                HIP_TRY(hipMalloc(&stage.col_vals[i], (size_t)n * 8));
+                HIP_TRY(hipMalloc(&stage.col_valid[i], n));
+            }
+            HIP_TRY(hipMalloc(&stage.ops, n));
+            HIP_TRY(hipMalloc(&stage.vis, n));
+            stage_cap = n;
+        }
+        if (old_cap < n) {
+            if (touched_list) hipFree(touched_list);
+            if (old_win) hipFree(old_win);
+            if (old_n) hipFree(old_n);
+            int K = (int)(m.offset + m.limit);
+            HIP_TRY(hipMalloc(&touched_list, (size_t)n * 4));
+            HIP_TRY(hipMalloc(&old_win, (size_t)n * K * 4));
+            HIP_TRY(hipMalloc(&old_n, (size_t)n * 4));
+            old_cap = n;
+        }
+        return RW_OK;
+    }
+
+    // same-cache-key rows within one chunk run in their own single-row
+    // segments (the reference applies rows in order, group_top_n.rs:168)
+    std::vector<uint32_t> conflict_segments(const RwChunk* c) {
+        std::vector<uint32_t> bounds;
+        std::unordered_map<std::string, int> ck_count;
+        auto ck_of = [&](uint32_t r) {
+            std::string k;
+            for (auto col : ck_col_idx) {
+                uint8_t valid = c->cols[col].valid[r];
+                k.push_back((char)valid);
+                int64_t v = valid ? ((const int64_t*)c->cols[col].data)[r] : 0;
+                k.append((const char*)&v, 8);
+            }
+            return k;
+        };
+        for (uint32_t r = 0; r < c->n_rows; r++) {
+            if (c->vis && !c->vis[r]) continue;
+            ck_count[ck_of(r)]++;
+        }
+        bool any = false;
+        for (auto& kv : ck_count)
+            if (kv.second > 1) {
+                any = true;
+                break;
+            }
+        if (!any) return bounds;
+        for (uint32_t r = 0; r < c->n_rows; r++) {
+            if (c->vis && !c->vis[r]) continue;
+            if (ck_count[ck_of(r)] > 1) {
+                bounds.push_back(r);
+                bounds.push_back(r + 1);
+            }
+        }
+        return bounds;
+    }
+
+    int push_chunk(const RwChunk* c) {
+        uint32_t n = c->n_rows;
+        if (!n) return RW_OK;
+        if (ensure_caps(n) != RW_OK) return RW_E_INTERNAL;
+        JoinBatchDev b = stage;
+        for (int i = 0; i < m.n_cols; i++) {
+            HIP_TRY(hipMemcpyAsync(b.col_vals[i], c->cols[i].data,
+                                   (size_t)n * 8, hipMemcpyHostToDevice,
+                                   stream));
+            HIP_TRY(hipMemcpyAsync(b.col_valid[i], c->cols[i].valid, n,
+                                   hipMemcpyHostToDevice, stream));
+        }
+        HIP_TRY(hipMemcpyAsync(b.ops, c->ops, n, hipMemcpyHostToDevice,
+                               stream));
+        if (c->vis)
+            HIP_TRY(hipMemcpyAsync(b.vis, c->vis, n, hipMemcpyHostToDevice,
+                                   stream));
+        else
+            b.vis = nullptr;
+        b.n_rows = n;
+        HIP_TRY(hipMemsetAsync(tcounters, 0, 8, stream));
+
+        uint32_t blocks = (n + 255) / 256;
+        if (blocks > 2048) blocks = 2048;
+        topn_touch_kernel<<<blocks, 256, 0, stream>>>(b, sd, m, touched,
+                                                      touched_list, tcounters);
+        topn_snapshot_kernel<<<256, 256, 0, stream>>>(
+            sd, m, touched_list, tcounters, old_win, old_n);
+        auto segs = conflict_segments(c);
+        uint32_t start = 0;
+        auto launch_apply = [&](uint32_t a, uint32_t z) {
+            if (z <= a) return;
+            uint32_t bl = (z - a + 255) / 256;
+            if (bl > 2048) bl = 2048;
+            topn_apply_kernel<<<bl, 256, 0, stream>>>(b, sd, m, a, z,
+                                                      tcounters);
+        };
+        for (uint32_t bnd : segs) {
+            launch_apply(start, bnd);
+            start = bnd;
+        }
+        launch_apply(start, n);
+        topn_emit_kernel<<<256, 256, 0, stream>>>(sd, m, touched_list,
+                                                  tcounters, touched, old_win,
+                                                  old_n, out, m.n_cols);
+        HIP_TRY(hipStreamSynchronize(stream));
+        uint32_t tc[2];
+        HIP_TRY(hipMemcpy(tc, tcounters, 8, hipMemcpyDeviceToHost));
+        if (tc[1] == 2) FAIL(RW_E_INTERNAL, "group table full");
+        if (tc[1] == 3) FAIL(RW_E_INTERNAL, "row store full");
+        return drain_output();
+    }
+
+    int drain_output() {
+        uint32_t ctr[2];
+        HIP_TRY(hipMemcpy(ctr, out.counters, 8, hipMemcpyDeviceToHost));
+        if (ctr[1] == 1) FAIL(RW_E_INTERNAL, "topn output overflow");
+        uint32_t n_out = ctr[0];
+        if (n_out) {
+            std::vector<int64_t> vals((size_t)n_out * m.n_cols);
+            std::vector<uint8_t> nulls((size_t)n_out * m.n_cols);
+            std::vector<uint8_t> ops(n_out);
+            HIP_TRY(hipMemcpy(vals.data(), out.vals, vals.size() * 8,
+                              hipMemcpyDeviceToHost));
+            HIP_TRY(hipMemcpy(nulls.data(), out.nulls, nulls.size(),
+                              hipMemcpyDeviceToHost));
+            HIP_TRY(hipMemcpy(ops.data(), out.ops, n_out,
+                              hipMemcpyDeviceToHost));
+            uint32_t max_rows = chunk_size < 2 ? 2 : chunk_size;
+            uint32_t s = 0;
+            while (s < n_out) {
+                uint32_t take = n_out - s;
+                if (take > max_rows) take = max_rows;
+                // U-pair adjacency: never split a U-/U+ pair at the edge
+                if (s + take < n_out &&
+                    ops[s + take - 1] == RW_OP_UPDATE_DELETE)
+                    take++;
+                auto* ch = new RwChunk();
+                auto* cols = new RwColumn[m.n_cols];
+                auto* o = new uint8_t[take];
+                memcpy(o, ops.data() + s, take);
+                for (int ci = 0; ci < m.n_cols; ci++) {
+                    auto* data = new int64_t[take];
+                    auto* valid = new uint8_t[take];
+                    for (uint32_t r = 0; r < take; r++) {
+                        data[r] = vals[(size_t)(s + r) * m.n_cols + ci];
+                        valid[r] = !nulls[(size_t)(s + r) * m.n_cols + ci];
+                    }
+                    cols[ci].type = types[ci];
+                    cols[ci].valid = valid;
+                    cols[ci].data = data;
+                }
+                ch->n_rows = take;
+                ch->n_cols = (uint32_t)m.n_cols;
+                ch->ops = o;
+                ch->vis = nullptr;
+                ch->cols = cols;
+                outq.push_back(ch);
+                s += take;
+            }
+        }
+        HIP_TRY(hipMemset(out.counters, 0, 8));
+        return RW_OK;
+    }
+
+    RwChunk* poll() {
+        if (outq.empty()) return nullptr;
+        RwChunk* c = outq.front();
+        outq.erase(outq.begin());
+        return c;
+    }
+};
+
+extern "C" {
+
+void* rw_group_top_n_create(const RwGroupTopNDesc* d) {
+    auto* t = new GroupTopN();
+    if (t->init(d) != RW_OK) {
+        delete t;
+        return nullptr;
+    }
+    return t;
+}
+int rw_group_top_n_push_chunk(void* h, const RwChunk* c) {
+    return ((GroupTopN*)h)->push_chunk(c);
+}
+int rw_group_top_n_flush(void* h, uint64_t epoch) {
+    (void)h;
+    (void)epoch;
+    return RW_OK; // emission is per push; checkpoint spill is a later row
+}
+RwChunk* rw_group_top_n_poll(void* h) { return ((GroupTopN*)h)->poll(); }
+void rw_group_top_n_destroy(void* h) { delete (GroupTopN*)h; }
+
+} // extern "C"

@@ -636,3 +636,61 @@ def test_update_vnode_bitmap_parity():
     assert outs[0] == outs[1]
     g.close()
     o.close()
+
+
+def _topn_pair(offset, limit, group_by, order_by, rest, types=None):
+    types = types or [T_I64, T_I64, T_I64]
+    g = ffi.GroupTopN(gpu(), types, group_by, order_by, rest,
+                      offset=offset, limit=limit)
+    o = ffi.GroupTopN(ffi.oracle(), types, group_by, order_by, rest,
+                      offset=offset, limit=limit)
+    return g, o
+
+
+def test_topn_golden():
+    # the reference's group_top_n.rs fixtures, GPU vs oracle (the oracle is
+    # pinned to the transcribed outputs in tests/test_oracle_topn.py)
+    from test_oracle_topn import chunks_0_3
+
+    for off, lim, gb, ob in [(0, 2, [1], [(2, False)]),
+                             (1, 2, [1], [(2, False)]),
+                             (0, 2, [1, 2], [(0, False)])]:
+        rest = [(c, False) for c in [1, 2, 0]
+                if c not in gb and c not in [o for o, _ in ob]]
+        g, o = _topn_pair(off, lim, gb, ob, rest)
+        for c in chunks_0_3():
+            g.push(c)
+            o.push(c)
+            mg = rows_multiset(g.poll_all())
+            mo = rows_multiset(o.poll_all())
+            assert mg == mo, f"off={off} lim={lim}: {mg} vs {mo}"
+        g.close()
+        o.close()
+
+
+def test_topn_random():
+    # randomized insert/delete mix incl. same-ck replacements and desc order
+    rng = np.random.default_rng(91)
+    g, o = _topn_pair(1, 3, [0], [(1, True)], [(2, False)])
+    live = []
+    for i in range(10):
+        n = 512
+        gk = rng.integers(0, 40, n)
+        ordv = rng.integers(0, 200, n)
+        pk = rng.integers(0, 10**6, n)
+        ops = np.zeros(n, np.uint8)
+        for r in range(n):
+            if live and rng.random() < 0.35:
+                jx = int(rng.integers(0, len(live)))
+                gk[r], ordv[r], pk[r] = live.pop(jx)
+                ops[r] = ffi.OP_DELETE
+            else:
+                live.append((int(gk[r]), int(ordv[r]), int(pk[r])))
+        c = mk_chunk([T_I64, T_I64, T_I64], ops, [gk, ordv, pk])
+        g.push(c)
+        o.push(c)
+        mg = rows_multiset(g.poll_all())
+        mo = rows_multiset(o.poll_all())
+        assert mg == mo, f"push {i}: {len(mg)} vs {len(mo)} rows"
+    g.close()
+    o.close()
