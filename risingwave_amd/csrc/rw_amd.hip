@@ -3375,8 +3375,11 @@ __global__ void topn_snapshot_kernel(JoinSideDev sd, TopMeta m,
          t += stride) {
         uint32_t sel[TOPN_MAX_WIN];
         int n = topn_select(sd, m, touched_list[t], sel, K);
-        old_n[t] = (uint32_t)n;
-        for (int i = 0; i < n; i++) old_win[(size_t)t * K + i] = sel[i];
+        // visible window = positions [offset, offset+limit)
+        int w0 = n < (int)m.offset ? n : (int)m.offset;
+        old_n[t] = (uint32_t)(n - w0);
+        for (int i = w0; i < n; i++)
+            old_win[(size_t)t * K + (i - w0)] = sel[i];
     }
 }
 
@@ -3454,8 +3457,12 @@ __global__ void topn_emit_kernel(JoinSideDev sd, TopMeta m,
          t += stride) {
         uint32_t slot = touched_list[t];
         touched[slot] = 0;
-        uint32_t neww[TOPN_MAX_WIN];
-        int nn = topn_select(sd, m, slot, neww, K);
+        uint32_t selbuf[TOPN_MAX_WIN];
+        int nsel = topn_select(sd, m, slot, selbuf, K);
+        // visible window = positions [offset, offset+limit)
+        int w0 = nsel < (int)m.offset ? nsel : (int)m.offset;
+        const uint32_t* neww = selbuf + w0;
+        int nn = nsel - w0;
         const uint32_t* oldw = old_win + (size_t)t * K;
         int no = (int)old_n[t];
         // merge-diff into local delta lists: op + row id (+ paired old id)
