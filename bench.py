@@ -150,8 +150,8 @@ def _cpu_baseline_worker(arg):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=30)
-    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--steps", type=int, default=2000)
+    ap.add_argument("--warmup", type=int, default=100)
     ap.add_argument("--barrier-every", type=int, default=16)
     ap.add_argument("--windows-per-epoch", type=int, default=64)
     ap.add_argument("--seed", type=int, default=1)
