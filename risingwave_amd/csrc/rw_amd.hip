@@ -556,7 +556,54 @@ __device__ __forceinline__ long long agg4_comb(uint8_t kind, long long a,
     }
 }
 
-template <int n_calls, int RPL = 4> // RPL = rows per lane (even)
+template <int n_calls, int RPL>
+__device__ __forceinline__ void dense_load(const AggBatch& b,
+                                           const AggCallDev* calls,
+                                           uint32_t rb, uint32_t r1, bool* act,
+                                           long long* k,
+                                           long long (*cv)[n_calls]) {
+    bool all_in = rb + RPL - 1 < r1;
+    if (all_in) {
+        // back-to-back b128 loads: RPL/2 per column, all in flight at once
+        const ulonglong2* kp = (const ulonglong2*)(b.col_vals[0] + rb);
+#pragma unroll
+        for (int h = 0; h < RPL / 2; h++) {
+            ulonglong2 kk = kp[h];
+            k[2 * h] = (long long)kk.x;
+            k[2 * h + 1] = (long long)kk.y;
+        }
+#pragma unroll
+        for (int r = 0; r < RPL; r++) act[r] = true;
+        _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) {
+            if (calls[ci].arg < 0) {
+#pragma unroll
+                for (int r = 0; r < RPL; r++) cv[r][ci] = 1;
+                continue;
+            }
+            const ulonglong2* vp = (const ulonglong2*)(b.col_vals[1 + ci] + rb);
+#pragma unroll
+            for (int h = 0; h < RPL / 2; h++) {
+                ulonglong2 vv = vp[h];
+                cv[2 * h][ci] = agg4_unit(calls[ci].kind, (long long)vv.x);
+                cv[2 * h + 1][ci] = agg4_unit(calls[ci].kind, (long long)vv.y);
+            }
+        }
+    } else {
+        for (int r = 0; r < RPL; r++) {
+            act[r] = rb + r < r1;
+            k[r] = act[r] ? b.col_vals[0][rb + r] : 0;
+            _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++)
+                cv[r][ci] = act[r]
+                                ? agg4_unit(calls[ci].kind,
+                                            calls[ci].arg < 0
+                                                ? 1
+                                                : b.col_vals[1 + ci][rb + r])
+                                : 0;
+        }
+    }
+}
+
+template <int n_calls, int RPL = 4, bool PF = false>
 __global__ void agg_apply_dense4_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
                                         AggCallDev c1, AggCallDev c2,
                                         AggCallDev c3, uint32_t r0, uint32_t r1) {
@@ -618,52 +665,24 @@ __global__ void agg_apply_dense4_kernel(AggBatch b, AggTableDev t, AggCallDev c0
         }
     };
 
+    uint32_t base = r0 + (blockIdx.x * blockDim.x + threadIdx.x) * RPL;
+    bool act[RPL];
+    long long k[RPL];
+    long long cv[RPL][n_calls]; // [row][call]
+    if (PF && iters) dense_load<n_calls, RPL>(b, calls, base, r1, act, k, cv);
     for (uint32_t it = 0; it < iters; it++) {
-        uint32_t rb = r0 + it * stride_rows +
-                      (blockIdx.x * blockDim.x + threadIdx.x) * RPL;
-        bool act[RPL];
-        long long k[RPL];
-        long long cv[RPL][n_calls]; // [row][call]
-        bool all_in = rb + RPL - 1 < r1;
-        if (all_in) {
-            // back-to-back b128 loads: RPL/2 per column, all in flight at once
-            const ulonglong2* kp = (const ulonglong2*)(b.col_vals[0] + rb);
-#pragma unroll
-            for (int h = 0; h < RPL / 2; h++) {
-                ulonglong2 kk = kp[h];
-                k[2 * h] = (long long)kk.x;
-                k[2 * h + 1] = (long long)kk.y;
-            }
-#pragma unroll
-            for (int r = 0; r < RPL; r++) act[r] = true;
-            _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) {
-                if (calls[ci].arg < 0) {
-#pragma unroll
-                    for (int r = 0; r < RPL; r++) cv[r][ci] = 1;
-                    continue;
-                }
-                const ulonglong2* vp =
-                    (const ulonglong2*)(b.col_vals[1 + ci] + rb);
-#pragma unroll
-                for (int h = 0; h < RPL / 2; h++) {
-                    ulonglong2 vv = vp[h];
-                    cv[2 * h][ci] = agg4_unit(calls[ci].kind, (long long)vv.x);
-                    cv[2 * h + 1][ci] =
-                        agg4_unit(calls[ci].kind, (long long)vv.y);
-                }
-            }
+        uint32_t rb = base + it * stride_rows;
+        bool act2[RPL];
+        long long k2[RPL];
+        long long cv2[RPL][n_calls];
+        if (PF) {
+            // double-buffered prefetch: next tile's loads issue before this
+            // tile's scan/commit chain, hiding the HBM round-trip
+            if (it + 1 < iters)
+                dense_load<n_calls, RPL>(b, calls, rb + stride_rows, r1, act2,
+                                         k2, cv2);
         } else {
-            for (int r = 0; r < RPL; r++) {
-                act[r] = rb + r < r1;
-                k[r] = act[r] ? b.col_vals[0][rb + r] : 0;
-                _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++)
-                    cv[r][ci] =
-                        act[r] ? agg4_unit(calls[ci].kind,
-                                           calls[ci].arg < 0
-                                               ? 1
-                                               : b.col_vals[1 + ci][rb + r])
-                               : 0;
-            }
+            dense_load<n_calls, RPL>(b, calls, rb, r1, act, k, cv);
         }
         // lane-local segments over the RPL rows (inactive rows break runs)
         long long last_key = 0;
@@ -769,6 +788,15 @@ __global__ void agg_apply_dense4_kernel(AggBatch b, AggTableDev t, AggCallDev c0
                 commit(suf_key, incl);
             } else if (have_suf && next_closes) {
                 // the next lane commits incoming (== this incl) + its prefix
+            }
+        }
+        if (PF && it + 1 < iters) {
+#pragma unroll
+            for (int r = 0; r < RPL; r++) {
+                act[r] = act2[r];
+                k[r] = k2[r];
+                _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++)
+                    cv[r][ci] = cv2[r][ci];
             }
         }
     }
@@ -1203,7 +1231,20 @@ struct HashAgg {
                 int v = e ? atoi(e) : 4;
                 return (v == 4 || v == 8 || v == 16) ? v : 4;
             }();
+            // PF: double-buffered prefetch variant (A/B via RW_AGG_PF=1)
+            static int pf = [] {
+                const char* e = getenv("RW_AGG_PF");
+                return e && *e == '1';
+            }();
             int grid = grid_for((r1 - r0 + rpl - 1) / rpl);
+            if (pf && rpl == 4) {
+                switch (n_calls) {
+                    case 1: agg_apply_dense4_kernel<1, 4, true><<<grid, 256, 0, stream>>>(b, t, a0, a1, a2, a3, r0, r1); return;
+                    case 2: agg_apply_dense4_kernel<2, 4, true><<<grid, 256, 0, stream>>>(b, t, a0, a1, a2, a3, r0, r1); return;
+                    case 3: agg_apply_dense4_kernel<3, 4, true><<<grid, 256, 0, stream>>>(b, t, a0, a1, a2, a3, r0, r1); return;
+                    case 4: agg_apply_dense4_kernel<4, 4, true><<<grid, 256, 0, stream>>>(b, t, a0, a1, a2, a3, r0, r1); return;
+                }
+            }
             #define RW_DENSE(nc, rp)                                          \
                 agg_apply_dense4_kernel<nc, rp><<<grid, 256, 0, stream>>>(    \
                     b, t, a0, a1, a2, a3, r0, r1)
