@@ -95,26 +95,17 @@ def cpu_baseline(ffi, rng, target_seconds=10.0):
     from rwtest.ffi import AGG_COUNT_STAR, AGG_MAX, T_I64, oracle
 
     ncores = os.cpu_count() or 1
-    calls = [(AGG_MAX, 1, T_I64), (AGG_COUNT_STAR, -1, T_I64)]
-    # calibrate on one instance
-    agg = ffi.HashAgg(oracle(), [T_I64, T_I64], [0], calls, 1, append_only=True)
-    chunk = make_q7_chunk(ffi, rng, CHUNK_ROWS, 0, 32)
-    agg.push(chunk)
-    t0 = time.perf_counter()
-    for _ in range(8):
-        agg.push(chunk)
-    per_chunk = (time.perf_counter() - t0) / 8
-    agg.close()
-    n = max(8, min(int(target_seconds / max(per_chunk, 1e-9)), 400_000))
-
     del threading  # processes, not threads: the per-chunk ctypes dispatch is
     # GIL-bound at 4K-row chunks (8 threads measured BELOW one core)
     import multiprocessing as mp
 
     ctx = mp.get_context("fork")
     t0 = time.perf_counter()
+    # wall-clock-bounded workers: a fixed chunk count misestimates N-way
+    # memory contention by 30x on a 256-core host
     with ctx.Pool(ncores) as pool:
-        done = pool.map(_cpu_baseline_worker, [(tid, n) for tid in range(ncores)])
+        done = pool.map(_cpu_baseline_worker,
+                        [(tid, target_seconds) for tid in range(ncores)])
     dt = time.perf_counter() - t0
     rows = sum(done)
     return {
@@ -129,7 +120,7 @@ def cpu_baseline(ffi, rng, target_seconds=10.0):
 
 
 def _cpu_baseline_worker(arg):
-    tid, n = arg
+    tid, seconds = arg
     from rwtest import ffi
     from rwtest.ffi import AGG_COUNT_STAR, AGG_MAX, T_I64, oracle
 
@@ -138,13 +129,16 @@ def _cpu_baseline_worker(arg):
     c = make_q7_chunk(ffi, np.random.default_rng(100 + tid), CHUNK_ROWS,
                       tid * 1_000_000 * WINDOW_US, 32)
     a.push(c)  # warm
-    for i in range(n):
+    deadline = time.perf_counter() + seconds
+    i = 0
+    while time.perf_counter() < deadline:
         a.push(c)
-        if (i + 1) % 64 == 0:
+        i += 1
+        if i % 64 == 0:
             a.flush(i)
             a.poll_all()
     a.close()
-    return n * CHUNK_ROWS
+    return i * CHUNK_ROWS
 
 
 def main():

@@ -2318,7 +2318,8 @@ __device__ __forceinline__ bool join_cond_ok(const JoinMeta& m, int probe_side,
 // append-only path (in-launch kill-on-match) uses the sc1 protocol.
 __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
                                   JoinSideDev match, JoinMeta m, int S,
-                                  JoinOutDev out, uint32_t r0, uint32_t r1) {
+                                  JoinOutDev out, uint32_t r0, uint32_t r1,
+                                  int dbg_skip = 0) {
     uint32_t stride = gridDim.x * blockDim.x;
     uint32_t n = r1 - r0;
     uint32_t iters = (n + stride - 1) / stride;
@@ -2383,7 +2384,7 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
             uint32_t my_base = base + incl - my_n;
             if (total && base + total > out.cap) {
                 if (lane == 0) atomicExch(&out.counters[1], 1u); // overflow
-            } else if (my_n) {
+            } else if (my_n && !(dbg_skip & 1)) {
                 // second walk: emit (JoinStreamChunkBuilder::append_row)
                 uint32_t row = match.slots[mslot].head;
                 uint32_t k = 0;
@@ -2449,7 +2450,7 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
             }
         }
 
-        if (!active) continue;
+        if (!active || (dbg_skip & 2)) continue;
 
         if (m.append_only && is_insert && matched_row != UINT32_MAX) {
             // append-only optimize (hash_join.rs:1241-1245)
@@ -2854,8 +2855,12 @@ struct HashJoin {
             if (ev_harvest(slot) != RW_OK) return RW_E_INTERNAL;
             HIP_TRY(hipEventRecord(ev0[slot], stream));
         }
+        static int dbg_skip = [] {
+            const char* e = getenv("RW_JOIN_SKIP"); // bench A/B only:
+            return e ? atoi(e) : 0;  // 1=no emit writes, 2=no own insert
+        }();
         join_probe_kernel<<<blocks, 256, 0, stream>>>(b, side[s], side[1 - s], m, s,
-                                                      out, r0, r1);
+                                                      out, r0, r1, dbg_skip);
         if (timed) {
             HIP_TRY(hipEventRecord(ev1[slot], stream));
             ev_pending[slot] = 1;
