@@ -2252,6 +2252,10 @@ struct JoinBatchDev {
     uint8_t* ops;
     uint8_t* vis;
     uint32_t n_rows;
+    // every visible row is an Insert: no same-launch thread walks the own
+    // side, so record payloads may use plain cached stores (no sc1, no
+    // vmcnt drain) — cross-launch visibility comes from the kernel boundary
+    uint8_t all_insert;
 };
 
 struct JoinOutDev {
@@ -2474,22 +2478,44 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
             JoinRowHdr* h = jrow(own, row);
             uint32_t vb = 0;
             long long* hv = jvals(h);
-            for (int c = 0; c < m.n_cols[S]; c++) {
-                st_i64((int64_t*)&hv[c], b.col_vals[c][r]);
-                vb |= (uint32_t)(b.col_valid[c][r] != 0) << c;
-            }
-            st_u32(&h->validbits, vb);
-            st_u32(&h->alive, 1);
-            // lock-free chain push: next set BEFORE the CAS publish, payload
-            // R1-drained ahead of it (same-launch deletes may walk this)
-            uint32_t* headp = &own.slots[own_slot].head;
-            uint32_t old_head = ld_u32(headp);
-            for (;;) {
-                st_u32(&h->next, old_head);
-                asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); // R1 drain
-                uint32_t prev = atomicCAS(headp, old_head, row);
-                if (prev == old_head) break;
-                old_head = prev;
+            if (b.all_insert) {
+                // plain cached stores: nothing walks the own side within
+                // this launch (no deletes), and the kernel boundary flushes
+                // before the next launch reads. The sc1 word-store variant
+                // below costs ~0.5 ms/1M rows (measured via RW_JOIN_SKIP).
+                for (int c = 0; c < m.n_cols[S]; c++) {
+                    hv[c] = b.col_vals[c][r];
+                    vb |= (uint32_t)(b.col_valid[c][r] != 0) << c;
+                }
+                h->validbits = vb;
+                h->alive = 1;
+                uint32_t* headp = &own.slots[own_slot].head;
+                uint32_t old_head = ld_u32(headp);
+                for (;;) {
+                    h->next = old_head;
+                    uint32_t prev = atomicCAS(headp, old_head, row);
+                    if (prev == old_head) break;
+                    old_head = prev;
+                }
+            } else {
+                for (int c = 0; c < m.n_cols[S]; c++) {
+                    st_i64((int64_t*)&hv[c], b.col_vals[c][r]);
+                    vb |= (uint32_t)(b.col_valid[c][r] != 0) << c;
+                }
+                st_u32(&h->validbits, vb);
+                st_u32(&h->alive, 1);
+                // lock-free chain push: next set BEFORE the CAS publish,
+                // payload R1-drained ahead of it (same-launch deletes may
+                // walk this)
+                uint32_t* headp = &own.slots[own_slot].head;
+                uint32_t old_head = ld_u32(headp);
+                for (;;) {
+                    st_u32(&h->next, old_head);
+                    asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); // R1 drain
+                    uint32_t prev = atomicCAS(headp, old_head, row);
+                    if (prev == old_head) break;
+                    old_head = prev;
+                }
             }
         } else {
             // delete own row: FULL-row compare + CAS claim
@@ -2837,6 +2863,11 @@ struct HashJoin {
         else
             b.vis = nullptr;
         b.n_rows = n;
+        b.all_insert = 1;
+        for (uint32_t r = 0; r < n && b.all_insert; r++)
+            if (!(c->vis && !c->vis[r]) && c->ops[r] != RW_OP_INSERT &&
+                c->ops[r] != RW_OP_UPDATE_INSERT)
+                b.all_insert = 0;
         *bout = b;
         return RW_OK;
     }
@@ -3171,6 +3202,10 @@ void* rw_join_bench_preload(void* h, int side, const RwChunk* c) {
     hipMemcpy(b->ops, c->ops, n, hipMemcpyHostToDevice);
     b->vis = nullptr;
     b->n_rows = n;
+    b->all_insert = 1;
+    for (uint32_t r = 0; r < n && b->all_insert; r++)
+        if (c->ops[r] != RW_OP_INSERT && c->ops[r] != RW_OP_UPDATE_INSERT)
+            b->all_insert = 0;
     return b;
 }
 
