@@ -2129,7 +2129,7 @@ struct JoinRowHdr {
     uint32_t alive; // CAS-claimed tombstone
     uint32_t next;
     uint32_t validbits; // bit c = column c non-NULL
-    uint32_t _pad;
+    uint32_t degree;    // matches on the other side (outer/semi/anti types)
 };
 
 struct JoinSideDev {
@@ -2148,7 +2148,14 @@ __device__ __forceinline__ long long* jvals(JoinRowHdr* h) {
     return (long long*)((uint8_t*)h + 16);
 }
 
-// own-side find-or-insert with the R1 sc1 protocol on the slot fields
+// own-side find-or-insert. The probe walk uses PLAIN cached loads: a slot's
+// state and keys share one 64-B line, and the claim protocol drains the sc1
+// key stores to the coherence point BEFORE the sc1 READY store — so any
+// line fill that observes READY also contains the keys (same-line, written
+// earlier at the coherence point), and a stale cached line can only show
+// the older EMPTY/CLAIMED state, which funnels into the coherent CAS/spin
+// path below. Keys never change once READY. This removes the per-visit sc1
+// word loads that dominated the insert path (~0.45 ms/1M inserts measured).
 __device__ __forceinline__ uint32_t jslot_find_or_insert(JoinSlot* slots,
                                                          uint32_t cap_mask,
                                                          const int64_t* kw,
@@ -2157,26 +2164,36 @@ __device__ __forceinline__ uint32_t jslot_find_or_insert(JoinSlot* slots,
     uint32_t slot = (uint32_t)(hash_key(kw, nullmask, KW) & cap_mask);
     for (uint32_t probes = 0; probes <= cap_mask; probes++) {
         JoinSlot* sl = &slots[slot];
-        uint32_t st = ld_u32(&sl->state);
-        if (st == SLOT_EMPTY) {
-            uint32_t prev = atomicCAS(&sl->state, SLOT_EMPTY, SLOT_CLAIMED);
-            if (prev == SLOT_EMPTY) {
-                for (int i = 0; i < KW; i++) st_i64((int64_t*)&sl->key[i], kw[i]);
-                st_u32(&sl->nulls, nullmask);
-                asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); // R1 drain
-                st_u32(&sl->state, SLOT_READY);
-                return slot;
+        uint32_t st = sl->state; // plain (fast path)
+        if (st == SLOT_READY) {
+            bool eq = sl->nulls == nullmask;
+            for (int i = 0; eq && i < KW; i++) eq = sl->key[i] == kw[i];
+            if (eq) return slot;
+        } else {
+            if (st == SLOT_EMPTY) {
+                uint32_t prev = atomicCAS(&sl->state, SLOT_EMPTY, SLOT_CLAIMED);
+                if (prev == SLOT_EMPTY) {
+                    for (int i = 0; i < KW; i++)
+                        st_i64((int64_t*)&sl->key[i], kw[i]);
+                    st_u32(&sl->nulls, nullmask);
+                    asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); // R1 drain
+                    st_u32(&sl->state, SLOT_READY);
+                    return slot;
+                }
+                st = prev;
+            } else {
+                st = ld_u32(&sl->state); // sc1 refresh of a stale CLAIMED
             }
-            st = prev;
+            while (st == SLOT_CLAIMED) {
+                __builtin_amdgcn_s_sleep(1);
+                st = ld_u32(&sl->state);
+            }
+            // coherent compare (the plain line may be stale here)
+            bool eq = ld_u32(&sl->nulls) == nullmask;
+            for (int i = 0; eq && i < KW; i++)
+                eq = ld_i64((const int64_t*)&sl->key[i]) == kw[i];
+            if (eq) return slot;
         }
-        while (st == SLOT_CLAIMED) {
-            __builtin_amdgcn_s_sleep(1);
-            st = ld_u32(&sl->state);
-        }
-        bool eq = ld_u32(&sl->nulls) == nullmask;
-        for (int i = 0; eq && i < KW; i++)
-            eq = ld_i64((const int64_t*)&sl->key[i]) == kw[i];
-        if (eq) return slot;
         slot = (slot + 1) & cap_mask;
     }
     return (uint32_t)-1;
@@ -2244,6 +2261,8 @@ struct JoinMeta {
     uint8_t has_cond, cond_op;
     uint8_t cond_src_l, cond_col_l, cond_src_r, cond_col_r;
     uint8_t append_only;
+    uint8_t join_type;    // RwJoinType
+    uint8_t need_deg[2];  // need_left/right_degree (join/mod.rs:153-165)
 };
 
 struct JoinBatchDev {
@@ -2320,6 +2339,252 @@ __device__ __forceinline__ bool join_cond_ok(const JoinMeta& m, int probe_side,
 // the match side is immutable during a non-append-only launch, so it is read
 // with plain cached loads (one line per slot, one per record). The
 // append-only path (in-launch kill-on-match) uses the sc1 protocol.
+
+// ---- join-type predicates (join/mod.rs:120-165; oracle_join.cpp mirrors) ----
+__host__ __device__ __forceinline__ bool j_is_semi(uint8_t t) {
+    return t == RW_JOIN_LEFT_SEMI || t == RW_JOIN_RIGHT_SEMI;
+}
+__host__ __device__ __forceinline__ bool j_is_anti(uint8_t t) {
+    return t == RW_JOIN_LEFT_ANTI || t == RW_JOIN_RIGHT_ANTI;
+}
+__host__ __device__ __forceinline__ bool j_is_outer_side(uint8_t t, int side) {
+    return t == RW_JOIN_FULL_OUTER ||
+           (t == RW_JOIN_LEFT_OUTER && side == RW_SIDE_LEFT) ||
+           (t == RW_JOIN_RIGHT_OUTER && side == RW_SIDE_RIGHT);
+}
+__host__ __device__ __forceinline__ bool j_outer_side_null(uint8_t t, int side) {
+    return t == RW_JOIN_FULL_OUTER ||
+           (t == RW_JOIN_LEFT_OUTER && side == RW_SIDE_RIGHT) ||
+           (t == RW_JOIN_RIGHT_OUTER && side == RW_SIDE_LEFT);
+}
+__host__ __device__ __forceinline__ bool j_forward_exactly_once(uint8_t t, int side) {
+    return ((t == RW_JOIN_LEFT_SEMI || t == RW_JOIN_LEFT_ANTI) && side == RW_SIDE_LEFT) ||
+           ((t == RW_JOIN_RIGHT_SEMI || t == RW_JOIN_RIGHT_ANTI) && side == RW_SIDE_RIGHT);
+}
+__host__ __device__ __forceinline__ bool j_only_forward_matched_side(uint8_t t, int side) {
+    return ((t == RW_JOIN_LEFT_SEMI || t == RW_JOIN_LEFT_ANTI) && side == RW_SIDE_RIGHT) ||
+           ((t == RW_JOIN_RIGHT_SEMI || t == RW_JOIN_RIGHT_ANTI) && side == RW_SIDE_LEFT);
+}
+__host__ __device__ __forceinline__ bool j_need_degree(uint8_t t, int side) {
+    if (side == RW_SIDE_LEFT)
+        return t == RW_JOIN_FULL_OUTER || t == RW_JOIN_LEFT_OUTER ||
+               t == RW_JOIN_LEFT_ANTI || t == RW_JOIN_LEFT_SEMI;
+    return t == RW_JOIN_FULL_OUTER || t == RW_JOIN_RIGHT_OUTER ||
+           t == RW_JOIN_RIGHT_ANTI || t == RW_JOIN_RIGHT_SEMI;
+}
+
+// emit one output row with per-emission reservation. form bit0 = include
+// the probe side's columns, bit1 = include the matched record's columns;
+// excluded sides are NULL (JoinOutBuilder append_row / _update / _matched,
+// join/builder.rs:153-318).
+__device__ __forceinline__ void jemit_row(JoinOutDev& out, const JoinMeta& m,
+                                          int S, uint8_t op,
+                                          const JoinBatchDev& b, uint32_t r,
+                                          uint32_t match_vb,
+                                          const long long* mv, int form) {
+    uint32_t orow = atomicAdd(&out.counters[0], 1u);
+    if (orow >= out.cap) {
+        atomicExch(&out.counters[1], 1u);
+        return;
+    }
+    out.ops[orow] = op;
+    for (int c = 0; c < m.n_out; c++) {
+        bool from_probe = (int)m.out_src[c] == S;
+        uint8_t col = m.out_col[c];
+        int64_t v = 0;
+        uint8_t valid = 0;
+        if (from_probe && (form & 1)) {
+            valid = b.col_valid[col][r];
+            v = b.col_vals[col][r];
+        } else if (!from_probe && (form & 2)) {
+            valid = (match_vb >> col) & 1;
+            v = mv[col];
+        }
+        out.vals[(size_t)orow * m.n_out + c] = valid ? v : 0;
+        out.nulls[(size_t)orow * m.n_out + c] = !valid;
+    }
+}
+
+// own-side insert/delete (join/hash_join.rs:591-681 without the LRU tier)
+__device__ __forceinline__ void jown_insert(JoinSideDev own, const JoinMeta& m,
+                                            int S, const JoinBatchDev& b,
+                                            uint32_t r, const int64_t* kw,
+                                            uint32_t nullmask, JoinOutDev out,
+                                            uint32_t init_deg) {
+    uint32_t own_slot =
+        jslot_find_or_insert(own.slots, own.cap_mask, kw, nullmask, m.KW);
+    if (own_slot == UINT32_MAX) {
+        atomicExch(&out.counters[1], 2u); // key table full
+        return;
+    }
+    uint32_t row = atomicAdd(own.row_cursor, 1u);
+    if (row >= own.row_cap) {
+        atomicExch(&out.counters[1], 3u); // row store full
+        return;
+    }
+    JoinRowHdr* h = jrow(own, row);
+    uint32_t vb = 0;
+    long long* hv = jvals(h);
+    if (b.all_insert) {
+        // plain cached stores: nothing walks the own side within this
+        // launch (no deletes), and the kernel boundary flushes before the
+        // next launch reads. The sc1 word-store variant below costs
+        // ~0.5 ms/1M rows (measured via RW_JOIN_SKIP).
+        for (int c = 0; c < m.n_cols[S]; c++) {
+            hv[c] = b.col_vals[c][r];
+            vb |= (uint32_t)(b.col_valid[c][r] != 0) << c;
+        }
+        h->validbits = vb;
+        h->degree = init_deg;
+        h->alive = 1;
+        uint32_t* headp = &own.slots[own_slot].head;
+        uint32_t old_head = ld_u32(headp);
+        for (;;) {
+            h->next = old_head;
+            uint32_t prev = atomicCAS(headp, old_head, row);
+            if (prev == old_head) break;
+            old_head = prev;
+        }
+    } else {
+        for (int c = 0; c < m.n_cols[S]; c++) {
+            st_i64((int64_t*)&hv[c], b.col_vals[c][r]);
+            vb |= (uint32_t)(b.col_valid[c][r] != 0) << c;
+        }
+        st_u32(&h->validbits, vb);
+        st_u32(&h->degree, init_deg);
+        st_u32(&h->alive, 1);
+        // lock-free chain push: next set BEFORE the CAS publish, payload
+        // R1-drained ahead of it (same-launch deletes may walk this)
+        uint32_t* headp = &own.slots[own_slot].head;
+        uint32_t old_head = ld_u32(headp);
+        for (;;) {
+            st_u32(&h->next, old_head);
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); // R1 drain
+            uint32_t prev = atomicCAS(headp, old_head, row);
+            if (prev == old_head) break;
+            old_head = prev;
+        }
+    }
+}
+
+__device__ __forceinline__ void jown_delete(JoinSideDev own, const JoinMeta& m,
+                                            int S, const JoinBatchDev& b,
+                                            uint32_t r, const int64_t* kw,
+                                            uint32_t nullmask) {
+    // delete own row: FULL-row compare + CAS claim
+    uint32_t own_slot =
+        jslot_find_sc1(own.slots, own.cap_mask, kw, nullmask, m.KW);
+    if (own_slot == UINT32_MAX) return;
+    uint32_t row = ld_u32(&own.slots[own_slot].head);
+    while (row != UINT32_MAX) {
+        JoinRowHdr* h = jrow(own, row);
+        if (ld_u32(&h->alive)) {
+            uint32_t vb = ld_u32(&h->validbits);
+            const long long* hv = jvals(h);
+            bool eq = true;
+            for (int c = 0; eq && c < m.n_cols[S]; c++) {
+                uint8_t va = b.col_valid[c][r];
+                uint8_t vbc = (vb >> c) & 1;
+                eq = (va == vbc) &&
+                     (!va || b.col_vals[c][r] == ld_i64((const int64_t*)&hv[c]));
+            }
+            if (eq && atomicCAS(&h->alive, 1u, 0u) == 1u) break;
+        }
+        row = ld_u32(&h->next);
+    }
+}
+
+// one probe row of a non-inner join: eq_join_oneside for
+// LeftOuter/RightOuter/FullOuter/Semi/Anti (hash_join.rs:949-1374 +
+// with_match_on_insert/delete, join/builder.rs:173-318). Degree transitions
+// use the matched record's atomic degree; the emitted transition rows are
+// independent of which chunk row wins a concurrent transition, so the
+// per-chunk output multiset matches the reference's sequential apply
+// (mixed insert/delete chunks sharing a join key run in serial segments —
+// conflict_segments).
+__device__ void join_probe_row_noninner(const JoinBatchDev& b, JoinSideDev own,
+                                        JoinSideDev match, const JoinMeta& m,
+                                        int S, JoinOutDev out, uint32_t r,
+                                        bool is_insert, uint8_t op,
+                                        const int64_t* kw, uint32_t nullmask,
+                                        bool never_match) {
+    uint8_t T = m.join_type;
+    bool fwd_once = j_forward_exactly_once(T, S);
+    bool anti = j_is_anti(T), semi = j_is_semi(T);
+    bool outer_self = j_is_outer_side(T, S);
+    bool only_fwd_m = j_only_forward_matched_side(T, S);
+    bool outer_null = j_outer_side_null(T, S);
+    if (never_match) {
+        // null-safe NeverMatch forwards for anti/outer (hash_join.rs:1143-1152)
+        if ((anti && fwd_once) || outer_self)
+            jemit_row(out, m, S, op, b, r, 0, nullptr, 1);
+        return; // no state write
+    }
+    uint32_t mslot =
+        jslot_find_cached(match.slots, match.cap_mask, kw, nullmask, m.KW);
+    uint32_t my_deg = 0;
+    if (mslot != UINT32_MAX) {
+        uint32_t row = match.slots[mslot].head;
+        while (row != UINT32_MAX) {
+            JoinRowHdr* h = jrow(match, row);
+            if (h->alive && join_cond_ok(m, S, b, r, h->validbits, jvals(h))) {
+                my_deg++;
+                bool zero = false;
+                if (m.need_deg[1 - S]) {
+                    // insert: emit BEFORE increment (zero == pre-value 0);
+                    // delete: decrement BEFORE emit (zero == post-value 0)
+                    // (hash_join.rs:1311-1342)
+                    uint32_t old = atomicAdd(&h->degree,
+                                             is_insert ? 1u : (uint32_t)-1);
+                    zero = is_insert ? old == 0 : old == 1;
+                }
+                const long long* mv = jvals(h);
+                uint32_t vb = h->validbits;
+                if (is_insert) {
+                    if (anti) {
+                        if (zero && only_fwd_m)
+                            jemit_row(out, m, S, RW_OP_DELETE, b, r, vb, mv, 2);
+                    } else if (semi) {
+                        if (zero && only_fwd_m)
+                            jemit_row(out, m, S, RW_OP_INSERT, b, r, vb, mv, 2);
+                    } else if (zero && outer_null) {
+                        jemit_row(out, m, S, RW_OP_DELETE, b, r, vb, mv, 2);
+                        jemit_row(out, m, S, RW_OP_INSERT, b, r, vb, mv, 3);
+                    } else if (!fwd_once) {
+                        jemit_row(out, m, S, RW_OP_INSERT, b, r, vb, mv, 3);
+                    }
+                } else {
+                    if (anti) {
+                        if (zero && only_fwd_m)
+                            jemit_row(out, m, S, RW_OP_INSERT, b, r, vb, mv, 2);
+                    } else if (semi) {
+                        if (zero && only_fwd_m)
+                            jemit_row(out, m, S, RW_OP_DELETE, b, r, vb, mv, 2);
+                    } else if (zero && outer_null) {
+                        jemit_row(out, m, S, RW_OP_DELETE, b, r, vb, mv, 3);
+                        jemit_row(out, m, S, RW_OP_INSERT, b, r, vb, mv, 2);
+                    } else if (!fwd_once) {
+                        jemit_row(out, m, S, RW_OP_DELETE, b, r, vb, mv, 3);
+                    }
+                }
+            }
+            row = h->next;
+        }
+    }
+    if (my_deg == 0) {
+        // forward_if_not_matched (join/builder.rs:303-312)
+        if ((anti && fwd_once) || outer_self)
+            jemit_row(out, m, S, op, b, r, 0, nullptr, 1);
+    } else if (semi && fwd_once) {
+        // forward_exactly_once_if_matched (join/builder.rs:287-300)
+        jemit_row(out, m, S, op, b, r, 0, nullptr, 1);
+    }
+    if (is_insert)
+        jown_insert(own, m, S, b, r, kw, nullmask, out, my_deg);
+    else
+        jown_delete(own, m, S, b, r, kw, nullmask);
+}
+
 __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
                                   JoinSideDev match, JoinMeta m, int S,
                                   JoinOutDev out, uint32_t r0, uint32_t r1,
@@ -2337,6 +2602,7 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
         uint8_t op = is_insert ? RW_OP_INSERT : RW_OP_DELETE;
         int64_t kw[MAX_KW];
         uint32_t nullmask = 0;
+        bool never_match = false;
         if (active) {
             for (int i = 0; i < m.KW; i++) {
                 uint8_t col = m.key_cols[S][i];
@@ -2345,8 +2611,16 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
                 nullmask |= (!valid) << i;
             }
             // null-safe NeverMatch (hash_join.rs:1004-1016)
-            if (nullmask & ~(uint32_t)m.null_safe_mask) active = false;
+            if (nullmask & ~(uint32_t)m.null_safe_mask) never_match = true;
         }
+
+        if (m.join_type != RW_JOIN_INNER) {
+            if (active && !(dbg_skip & 2))
+                join_probe_row_noninner(b, own, match, m, S, out, r, is_insert,
+                                        op, kw, nullmask, never_match);
+            continue;
+        }
+        if (never_match) active = false;
 
         uint32_t mslot = UINT32_MAX;
         uint32_t my_n = 0;
@@ -2463,84 +2737,10 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
         }
 
         // own-side state update (join/hash_join.rs:591-681 without LRU tier)
-        if (is_insert) {
-            uint32_t own_slot = jslot_find_or_insert(own.slots, own.cap_mask, kw,
-                                                     nullmask, m.KW);
-            if (own_slot == UINT32_MAX) {
-                atomicExch(&out.counters[1], 2u); // key table full
-                continue;
-            }
-            uint32_t row = atomicAdd(own.row_cursor, 1u);
-            if (row >= own.row_cap) {
-                atomicExch(&out.counters[1], 3u); // row store full
-                continue;
-            }
-            JoinRowHdr* h = jrow(own, row);
-            uint32_t vb = 0;
-            long long* hv = jvals(h);
-            if (b.all_insert) {
-                // plain cached stores: nothing walks the own side within
-                // this launch (no deletes), and the kernel boundary flushes
-                // before the next launch reads. The sc1 word-store variant
-                // below costs ~0.5 ms/1M rows (measured via RW_JOIN_SKIP).
-                for (int c = 0; c < m.n_cols[S]; c++) {
-                    hv[c] = b.col_vals[c][r];
-                    vb |= (uint32_t)(b.col_valid[c][r] != 0) << c;
-                }
-                h->validbits = vb;
-                h->alive = 1;
-                uint32_t* headp = &own.slots[own_slot].head;
-                uint32_t old_head = ld_u32(headp);
-                for (;;) {
-                    h->next = old_head;
-                    uint32_t prev = atomicCAS(headp, old_head, row);
-                    if (prev == old_head) break;
-                    old_head = prev;
-                }
-            } else {
-                for (int c = 0; c < m.n_cols[S]; c++) {
-                    st_i64((int64_t*)&hv[c], b.col_vals[c][r]);
-                    vb |= (uint32_t)(b.col_valid[c][r] != 0) << c;
-                }
-                st_u32(&h->validbits, vb);
-                st_u32(&h->alive, 1);
-                // lock-free chain push: next set BEFORE the CAS publish,
-                // payload R1-drained ahead of it (same-launch deletes may
-                // walk this)
-                uint32_t* headp = &own.slots[own_slot].head;
-                uint32_t old_head = ld_u32(headp);
-                for (;;) {
-                    st_u32(&h->next, old_head);
-                    asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); // R1 drain
-                    uint32_t prev = atomicCAS(headp, old_head, row);
-                    if (prev == old_head) break;
-                    old_head = prev;
-                }
-            }
-        } else {
-            // delete own row: FULL-row compare + CAS claim
-            uint32_t own_slot = jslot_find_sc1(own.slots, own.cap_mask, kw,
-                                               nullmask, m.KW);
-            if (own_slot == UINT32_MAX) continue;
-            uint32_t row = ld_u32(&own.slots[own_slot].head);
-            while (row != UINT32_MAX) {
-                JoinRowHdr* h = jrow(own, row);
-                if (ld_u32(&h->alive)) {
-                    uint32_t vb = ld_u32(&h->validbits);
-                    const long long* hv = jvals(h);
-                    bool eq = true;
-                    for (int c = 0; eq && c < m.n_cols[S]; c++) {
-                        uint8_t va = b.col_valid[c][r];
-                        uint8_t vbc = (vb >> c) & 1;
-                        eq = (va == vbc) &&
-                             (!va ||
-                              b.col_vals[c][r] == ld_i64((const int64_t*)&hv[c]));
-                    }
-                    if (eq && atomicCAS(&h->alive, 1u, 0u) == 1u) break;
-                }
-                row = ld_u32(&h->next);
-            }
-        }
+        if (is_insert)
+            jown_insert(own, m, S, b, r, kw, nullmask, out, 0);
+        else
+            jown_delete(own, m, S, b, r, kw, nullmask);
     }
 }
 
@@ -2740,8 +2940,10 @@ struct HashJoin {
     int init(const RwHashJoinDesc* d) {
         if (!gpu_ok()) FAIL(RW_E_NOGPU, "risingwave_amd: no GPU visible (product path has no CPU fallback)");
         desc = *d;
-        if (d->join_type != RW_JOIN_INNER)
-            FAIL(RW_E_INVAL, "join type %d not in round-1 kernels (Inner only; oracle covers the rest)", d->join_type);
+        if (d->join_type > RW_JOIN_RIGHT_ANTI)
+            FAIL(RW_E_INVAL, "unknown join type %d", d->join_type);
+        if (d->append_only && d->join_type != RW_JOIN_INNER)
+            FAIL(RW_E_INVAL, "append-only optimize kernel is inner-only this round");
         if (d->n_key < 1 || d->n_key > MAX_KW) FAIL(RW_E_INVAL, "n_key %u", d->n_key);
         if (d->n_cols_l > MAX_COLS || d->n_cols_r > MAX_COLS)
             FAIL(RW_E_INVAL, "too many columns for round-1 kernels");
@@ -2792,6 +2994,9 @@ struct HashJoin {
         split(d->cond_l, &m.cond_src_l, &m.cond_col_l);
         split(d->cond_r, &m.cond_src_r, &m.cond_col_r);
         m.append_only = d->append_only;
+        m.join_type = d->join_type;
+        m.need_deg[0] = j_need_degree(d->join_type, RW_SIDE_LEFT);
+        m.need_deg[1] = j_need_degree(d->join_type, RW_SIDE_RIGHT);
         if (d->n_wm_jk) {
             wm_pos.assign(d->wm_jk_pos, d->wm_jk_pos + d->n_wm_jk);
             wm_clean.assign(d->wm_jk_clean, d->wm_jk_clean + d->n_wm_jk);
@@ -2976,21 +3181,40 @@ struct HashJoin {
             }
             return k;
         };
+        auto join_key = [&](uint32_t r) {
+            std::string k;
+            for (int i = 0; i < m.KW; i++) {
+                uint8_t ci = m.key_cols[s][i];
+                uint8_t valid = c->cols[ci].valid[r];
+                k.push_back((char)valid);
+                int64_t v = valid ? ((const int64_t*)c->cols[ci].data)[r] : 0;
+                k.append((const char*)&v, 8);
+            }
+            return k;
+        };
         std::unordered_multiset<std::string> inserts;
+        // degree-carrying types: a mixed insert/delete chunk sharing a JOIN
+        // KEY races the matched rows' degree transitions — isolate all rows
+        // of such keys (sequential semantics, hash_join.rs row order)
+        bool deg = m.need_deg[0] || m.need_deg[1];
+        std::unordered_map<std::string, int> jk_ops; // 1=ins 2=del bits
         for (uint32_t r = 0; r < c->n_rows; r++) {
             if (c->vis && !c->vis[r]) continue;
             uint8_t op = c->ops[r];
-            if (op == RW_OP_INSERT || op == RW_OP_UPDATE_INSERT)
-                inserts.insert(row_key(r));
+            bool ins = (op == RW_OP_INSERT || op == RW_OP_UPDATE_INSERT);
+            if (ins) inserts.insert(row_key(r));
+            if (deg) jk_ops[join_key(r)] |= ins ? 1 : 2;
         }
-        if (inserts.empty()) return bounds;
         for (uint32_t r = 0; r < c->n_rows; r++) {
             if (c->vis && !c->vis[r]) continue;
             uint8_t op = c->ops[r];
-            if ((op == RW_OP_DELETE || op == RW_OP_UPDATE_DELETE) &&
-                inserts.count(row_key(r))) {
-                bounds.push_back(r);     // segment ends before the delete
-                bounds.push_back(r + 1); // the delete runs alone
+            bool is_del = (op == RW_OP_DELETE || op == RW_OP_UPDATE_DELETE);
+            bool conflict =
+                (is_del && !inserts.empty() && inserts.count(row_key(r))) ||
+                (deg && jk_ops[join_key(r)] == 3);
+            if (conflict) {
+                bounds.push_back(r);     // segment ends before this row
+                bounds.push_back(r + 1); // the row runs alone, in order
             }
         }
         return bounds;

@@ -694,3 +694,69 @@ def test_topn_random():
         assert mg == mo, f"push {i}: {len(mg)} vs {len(mo)} rows"
     g.close()
     o.close()
+
+
+@pytest.mark.parametrize("jt", [
+    ffi.JOIN_LEFT_OUTER, ffi.JOIN_RIGHT_OUTER, ffi.JOIN_FULL_OUTER,
+    ffi.JOIN_LEFT_SEMI, ffi.JOIN_LEFT_ANTI, ffi.JOIN_RIGHT_SEMI,
+    ffi.JOIN_RIGHT_ANTI,
+])
+def test_join_noninner_random(jt):
+    # all non-inner types: randomized insert/delete mix on both sides,
+    # GPU vs oracle multiset per push (degree transitions, NULL-side rows,
+    # forward_if_not_matched, semi/anti matched-side emissions)
+    rng = np.random.default_rng(1000 + jt)
+    g = ffi.HashJoin(gpu(), jt, [T_I64, T_I64], [T_I64, T_I64],
+                     key_l=[0], key_r=[0], pk_l=[1], pk_r=[1])
+    o = ffi.HashJoin(oracle(), jt, [T_I64, T_I64], [T_I64, T_I64],
+                     key_l=[0], key_r=[0], pk_l=[1], pk_r=[1])
+    live = {SIDE_LEFT: [], SIDE_RIGHT: []}
+    pk = 0
+    for i in range(10):
+        side = int(rng.integers(0, 2))
+        n = 1024
+        keys = rng.integers(0, 120, n)
+        vals = np.arange(pk, pk + n)
+        pk += n
+        ops = np.zeros(n, np.uint8)
+        for r in range(n):
+            if live[side] and rng.random() < 0.3:
+                jx = int(rng.integers(0, len(live[side])))
+                keys[r], vals[r] = live[side].pop(jx)
+                ops[r] = ffi.OP_DELETE
+            else:
+                live[side].append((int(keys[r]), int(vals[r])))
+        c = mk_chunk([T_I64, T_I64], ops, [keys, vals])
+        g.push(side, c)
+        o.push(side, c)
+        mg = rows_multiset(g.poll_all())
+        mo = rows_multiset(o.poll_all())
+        assert mg == mo, (f"type {jt} push {i} side {side}: "
+                          f"{len(mg)} vs {len(mo)} rows")
+    g.close()
+    o.close()
+
+
+def test_join_left_outer_golden():
+    # transcription anchor: the oracle's left-outer behavior is pinned by
+    # tests/test_oracle_join.py golden vectors; here GPU vs oracle on the
+    # same shaped stream incl. NULL-side transitions
+    g = ffi.HashJoin(gpu(), ffi.JOIN_LEFT_OUTER, [T_I64, T_I64],
+                     [T_I64, T_I64], key_l=[0], key_r=[0], pk_l=[1], pk_r=[1])
+    o = ffi.HashJoin(oracle(), ffi.JOIN_LEFT_OUTER, [T_I64, T_I64],
+                     [T_I64, T_I64], key_l=[0], key_r=[0], pk_l=[1], pk_r=[1])
+    pushes = [
+        (SIDE_LEFT, from_pretty(" I I\n + 1 4\n + 2 5\n + 3 6")),
+        (SIDE_RIGHT, from_pretty(" I I\n + 2 7\n + 4 8")),
+        (SIDE_RIGHT, from_pretty(" I I\n + 2 9\n - 2 7")),
+        (SIDE_LEFT, from_pretty(" I I\n - 2 5")),
+        (SIDE_RIGHT, from_pretty(" I I\n - 2 9")),
+    ]
+    for i, (side, c) in enumerate(pushes):
+        g.push(side, c)
+        o.push(side, c)
+        mg = rows_multiset(g.poll_all())
+        mo = rows_multiset(o.poll_all())
+        assert mg == mo, f"push {i}: {mg} vs {mo}"
+    g.close()
+    o.close()
