@@ -759,7 +759,9 @@ def net_rows(rows):
     out = []
     for key, c in cnt.items():
         out.extend([key] * c)
-    return sorted(out, key=lambda r: (r[0], r[1]))
+    sk = lambda r: (r[0], tuple((v is None, v if v is not None else 0)
+                                for v in r[1]))
+    return sorted(out, key=sk)
 
 
 def test_join_left_outer_golden():
