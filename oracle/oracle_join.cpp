@@ -28,6 +28,7 @@
 
 #include "../include/rw_stream.h"
 #include "common.hpp"
+#include "../include/rw_codec.hpp"
 
 namespace orc {
 
@@ -271,6 +272,7 @@ struct HashJoinOracle {
             if (append_only && have_ao_match && is_insert) {
                 // delete matched row, skip own insert (hash_join.rs:1241-1245)
                 mit->second.erase(ao_match_pk);
+                delta_delete(1 - S, key, ao_match_pk);
                 if (mit->second.empty()) match_side.table.erase(mit);
                 continue;
             }
@@ -281,10 +283,11 @@ struct HashJoinOracle {
             if (is_insert) {
                 auto& m = side_entry(upd_side, key);
                 m[pk] = JoinEntry{row, degree};
+                delta_insert(S, key, pk, row);
             } else {
                 auto it = upd_side.table.find(key);
                 if (it != upd_side.table.end()) {
-                    it->second.erase(pk);
+                    if (it->second.erase(pk)) delta_delete(S, key, pk);
                     if (it->second.empty()) upd_side.table.erase(it);
                 }
             }
@@ -332,6 +335,70 @@ struct HashJoinOracle {
                 b.append_row(RW_OP_DELETE, row, entry.row);
             }
         }
+    }
+
+    // §8f-2 checkpoint spill deltas per side: key = memcmp(jk ∥ pk),
+    // tri-state net per key (DEL / fresh PUT / PUT over a pre-epoch row) —
+    // mirrors the GPU's kill-list netting. Degree tables are not spilled.
+    struct DeltaEnt {
+        int st; // 0 = DEL, 1 = PUT (fresh), 2 = PUT (over pre-epoch row)
+        std::vector<uint8_t> v;
+    };
+    std::map<std::string, DeltaEnt> delta[2];
+
+    std::string enc_key(int s, const Row& key, const Row& pk) {
+        std::vector<uint8_t> kb;
+        for (size_t i = 0; i < key.size(); i++) {
+            uint8_t t = side[s].types[side[s].key_idx[i]];
+            rwcodec::DatumC d{key[i].null, key[i].i, key[i].d};
+            rwcodec::memcmp_encode_datum(kb, t, d, {});
+        }
+        for (size_t i = 0; i < pk.size(); i++) {
+            uint8_t t = side[s].types[side[s].pk_idx[i]];
+            rwcodec::DatumC d{pk[i].null, pk[i].i, pk[i].d};
+            rwcodec::memcmp_encode_datum(kb, t, d, {});
+        }
+        return std::string((const char*)kb.data(), kb.size());
+    }
+
+    void delta_insert(int s, const Row& key, const Row& pk, const Row& row) {
+        std::vector<uint8_t> v;
+        for (size_t c = 0; c < row.size(); c++) {
+            rwcodec::DatumC d{row[c].null, row[c].i, row[c].d};
+            rwcodec::value_encode_datum(v, side[s].types[c], d);
+        }
+        std::string k = enc_key(s, key, pk);
+        auto it = delta[s].find(k);
+        if (it != delta[s].end() && it->second.st == 0)
+            it->second = {2, std::move(v)}; // PUT over a pre-epoch DEL
+        else
+            delta[s][k] = {1, std::move(v)};
+    }
+
+    void delta_delete(int s, const Row& key, const Row& pk) {
+        std::string k = enc_key(s, key, pk);
+        auto it = delta[s].find(k);
+        if (it == delta[s].end()) {
+            delta[s][k] = {0, {}}; // pre-epoch row deleted
+        } else if (it->second.st == 1) {
+            delta[s].erase(it); // fresh insert netted away
+        } else {
+            it->second = {0, {}};
+        }
+    }
+
+    void checkpoint_drain(int s, std::vector<uint8_t>& sp) {
+        auto put32 = [&](uint32_t x) {
+            for (int b = 0; b < 4; b++) sp.push_back((uint8_t)(x >> (8 * b)));
+        };
+        for (auto& [k, e] : delta[s]) {
+            sp.push_back(e.st ? 1 : 0);
+            put32((uint32_t)k.size());
+            sp.insert(sp.end(), k.begin(), k.end());
+            put32(e.st ? (uint32_t)e.v.size() : 0);
+            if (e.st) sp.insert(sp.end(), e.v.begin(), e.v.end());
+        }
+        delta[s].clear();
     }
 
     int cur_side = 0;
@@ -418,6 +485,17 @@ void* rw_hash_join_create(const RwHashJoinDesc* d) { return new HashJoinOracle(d
 int rw_hash_join_push_chunk(void* h, int side, const RwChunk* c) {
     return ((HashJoinOracle*)h)->push(side, c);
 }
+int rw_join_checkpoint_drain(void* h, int side, uint8_t** buf,
+                             uint64_t* len) {
+    if (side != 0 && side != 1) return RW_E_INVAL;
+    std::vector<uint8_t> sp;
+    ((HashJoinOracle*)h)->checkpoint_drain(side, sp);
+    *len = sp.size();
+    *buf = (uint8_t*)malloc(sp.size() ? sp.size() : 1);
+    memcpy(*buf, sp.data(), sp.size());
+    return RW_OK;
+}
+
 int rw_hash_join_update_vnode_bitmap(void* h, const uint8_t* bitmap,
                                      uint32_t vnode_count) {
     if (!vnode_count || vnode_count % 8) return RW_E_INVAL;

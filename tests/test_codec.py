@@ -115,3 +115,37 @@ def test_memcmp_ordering_properties():
     nonnull = [v - 2**64 if v >= 2**63 else v for v in vals if v is not None]
     assert nonnull == sorted(nonnull)
     assert vals[-1] is None  # NULL sorts largest
+
+
+def test_join_spill_bytes():
+    # §8f-2 join-state spill: record framing + memcmp(jk ∥ pk) key +
+    # value-encoded row, with in-epoch netting (insert+delete cancels;
+    # delete of a pre-epoch row emits a DELETE record)
+    import struct
+
+    from rwtest.ffi import JOIN_INNER, SIDE_LEFT, T_I64, from_pretty, oracle
+    from rwtest import ffi
+
+    o = ffi.HashJoin(oracle(), JOIN_INNER, [T_I64, T_I64], [T_I64, T_I64],
+                     key_l=[0], key_r=[0], pk_l=[1], pk_r=[1])
+    o.push(SIDE_LEFT, from_pretty(" I I\n + 5 100\n + 7 200\n - 7 200"))
+    o.poll_all()
+    sp = ffi.join_checkpoint_drain(oracle(), o.h, SIDE_LEFT)
+    # only (5,100) survives the epoch: one PUT record
+    def memcmp_i64(v):
+        return b"\x00" + struct.pack(">q", v ^ -(1 << 63))
+    def val_i64(v):
+        return b"\x01" + struct.pack("<q", v)
+    want_key = memcmp_i64(5) + memcmp_i64(100)
+    want_val = val_i64(5) + val_i64(100)
+    want = (b"\x01" + struct.pack("<I", len(want_key)) + want_key +
+            struct.pack("<I", len(want_val)) + want_val)
+    assert sp == want, f"{sp.hex()} vs {want.hex()}"
+    # next epoch: delete the pre-epoch row -> one DELETE record
+    o.push(SIDE_LEFT, from_pretty(" I I\n - 5 100"))
+    o.poll_all()
+    sp2 = ffi.join_checkpoint_drain(oracle(), o.h, SIDE_LEFT)
+    want2 = (b"\x00" + struct.pack("<I", len(want_key)) + want_key +
+             struct.pack("<I", 0))
+    assert sp2 == want2, f"{sp2.hex()} vs {want2.hex()}"
+    o.close()

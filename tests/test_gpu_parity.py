@@ -787,3 +787,38 @@ def test_join_left_outer_golden():
         assert mg == mo, f"push {i}: {mg} vs {mo}"
     g.close()
     o.close()
+
+
+def test_join_checkpoint_spill_parity():
+    # §8f-2 join-state spill: GPU vs oracle byte-identical per epoch
+    # (memcmp(jk ∥ pk) keys sorted, value-encoded rows, in-epoch netting)
+    rng = np.random.default_rng(123)
+    g, o = join_pair()
+    live = {SIDE_LEFT: [], SIDE_RIGHT: []}
+    pk = 0
+    for epoch in range(4):
+        for side in (SIDE_LEFT, SIDE_RIGHT):
+            n = 1024
+            keys = rng.integers(0, 200, n)
+            vals = np.arange(pk, pk + n)
+            pk += n
+            ops = np.zeros(n, np.uint8)
+            for r in range(n):
+                if live[side] and rng.random() < 0.3:
+                    jx = int(rng.integers(0, len(live[side])))
+                    keys[r], vals[r] = live[side].pop(jx)
+                    ops[r] = ffi.OP_DELETE
+                else:
+                    live[side].append((int(keys[r]), int(vals[r])))
+            c = mk_chunk([T_I64, T_I64], ops, [keys, vals])
+            g.push(side, c)
+            o.push(side, c)
+            g.poll_all()
+            o.poll_all()
+        for side in (SIDE_LEFT, SIDE_RIGHT):
+            sg = ffi.join_checkpoint_drain(gpu(), g.h, side)
+            so = ffi.join_checkpoint_drain(ffi.oracle(), o.h, side)
+            assert sg == so, (f"epoch {epoch} side {side}: "
+                              f"{len(sg)} vs {len(so)} bytes")
+    g.close()
+    o.close()
