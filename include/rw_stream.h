@@ -216,6 +216,22 @@ int rw_group_top_n_flush(void* h, uint64_t epoch); /* state commit only */
 RwChunk* rw_group_top_n_poll(void* h);
 void rw_group_top_n_destroy(void* h);
 
+/* ----- §8f-2 checkpoint spill (state-store boundary) -----
+ *
+ * Per-epoch KV deltas of the executor state tables in the reference's
+ * encodings — key = memcomparable pk (OrderedRowSerde,
+ * memcmp_encoding.rs:35-70), value = value-encoded row
+ * (value_encoding/mod.rs:151-215) — so the drained bytes are what
+ * StateTable::commit would hand the state store. Record framing:
+ * [put u8][klen u32 LE][key][vlen u32 LE][value], emitted in memcmp key
+ * order. Agg: intermediate-state table (group key → outputs). Join: per
+ * side, pk = join key ∥ deduped input pk, value = full row; degree tables
+ * are not spilled (cleanup flows through watermark hints). Caller frees
+ * with rw_spill_free. */
+int rw_agg_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len);
+int rw_join_checkpoint_drain(void* h, int side, uint8_t** buf, uint64_t* len);
+void rw_spill_free(uint8_t* buf);
+
 #ifdef __cplusplus
 }
 #endif

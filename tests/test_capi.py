@@ -29,6 +29,20 @@ SYMBOLS = [
     "rw_agg_bench_apply",
     "rw_agg_sync",
     "rw_agg_kernel_stats",
+    "rw_hash_agg_watermark",
+    "rw_hash_join_watermark",
+    "rw_hash_agg_update_vnode_bitmap",
+    "rw_hash_join_update_vnode_bitmap",
+    "rw_group_top_n_create",
+    "rw_group_top_n_push_chunk",
+    "rw_group_top_n_flush",
+    "rw_group_top_n_poll",
+    "rw_group_top_n_destroy",
+    "rw_agg_checkpoint_drain",
+    "rw_join_checkpoint_drain",
+    "rw_spill_free",
+    "rw_vnode_compute",
+    "rw_dispatch_compute",
 ]
 
 
