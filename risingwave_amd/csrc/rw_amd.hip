@@ -2282,6 +2282,10 @@ struct JoinBatchDev {
     // side, so record payloads may use plain cached stores (no sc1, no
     // vmcnt drain) — cross-launch visibility comes from the kernel boundary
     uint8_t all_insert;
+    // join keys are pairwise distinct within the batch (host-verified; true
+    // by construction for agg-output inputs like Nexmark q8's): at most one
+    // chain push per slot per launch, so the head publish needs no CAS
+    uint8_t unique_keys;
 };
 
 struct JoinOutDev {
@@ -2455,12 +2459,19 @@ __device__ __forceinline__ void jown_insert(JoinSideDev own, const JoinMeta& m,
         h->degree = init_deg;
         h->alive = 1;
         uint32_t* headp = &own.slots[own_slot].head;
-        uint32_t old_head = ld_u32(headp);
-        for (;;) {
-            h->next = old_head;
-            uint32_t prev = atomicCAS(headp, old_head, row);
-            if (prev == old_head) break;
-            old_head = prev;
+        if (b.unique_keys) {
+            // sole writer of this slot's chain in this launch: plain
+            // read-modify-write (1M random-line CASes cost ~0.45 ms/launch)
+            h->next = *headp;
+            *headp = row;
+        } else {
+            uint32_t old_head = ld_u32(headp);
+            for (;;) {
+                h->next = old_head;
+                uint32_t prev = atomicCAS(headp, old_head, row);
+                if (prev == old_head) break;
+                old_head = prev;
+            }
         }
     } else {
         for (int c = 0; c < m.n_cols[S]; c++) {
@@ -3105,6 +3116,26 @@ struct HashJoin {
             if (!(c->vis && !c->vis[r]) && c->ops[r] != RW_OP_INSERT &&
                 c->ops[r] != RW_OP_UPDATE_INSERT)
                 b.all_insert = 0;
+        b.unique_keys = 0;
+        if (b.all_insert) {
+            std::unordered_set<std::string> seen;
+            seen.reserve(n * 2);
+            bool uniq = true;
+            for (uint32_t r = 0; uniq && r < n; r++) {
+                if (c->vis && !c->vis[r]) continue;
+                std::string k;
+                for (int i = 0; i < m.KW; i++) {
+                    uint8_t ci = m.key_cols[s][i];
+                    uint8_t valid = c->cols[ci].valid[r];
+                    k.push_back((char)valid);
+                    int64_t v =
+                        valid ? ((const int64_t*)c->cols[ci].data)[r] : 0;
+                    k.append((const char*)&v, 8);
+                }
+                uniq = seen.insert(std::move(k)).second;
+            }
+            b.unique_keys = uniq;
+        }
         *bout = b;
         return RW_OK;
     }
@@ -3550,6 +3581,24 @@ void* rw_join_bench_preload(void* h, int side, const RwChunk* c) {
     for (uint32_t r = 0; r < n && b->all_insert; r++)
         if (c->ops[r] != RW_OP_INSERT && c->ops[r] != RW_OP_UPDATE_INSERT)
             b->all_insert = 0;
+    b->unique_keys = 0;
+    if (b->all_insert) {
+        std::unordered_set<std::string> seen;
+        seen.reserve(n * 2);
+        bool uniq = true;
+        for (uint32_t r = 0; uniq && r < n; r++) {
+            std::string k;
+            for (int i = 0; i < j->m.KW; i++) {
+                uint8_t ci = j->m.key_cols[side][i];
+                uint8_t valid = c->cols[ci].valid[r];
+                k.push_back((char)valid);
+                int64_t v = valid ? ((const int64_t*)c->cols[ci].data)[r] : 0;
+                k.append((const char*)&v, 8);
+            }
+            uniq = seen.insert(std::move(k)).second;
+        }
+        b->unique_keys = uniq;
+    }
     return b;
 }
 
