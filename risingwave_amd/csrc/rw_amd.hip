@@ -1902,7 +1902,7 @@ int rw_agg_stats_reset(void* h) {
 // array/mod.rs:99) mod vnode_count. Oracle counterpart:
 // oracle/oracle_dispatch.cpp; pinned against zlib.crc32 in tests.
 
-__constant__ uint32_t g_crc_table[256];
+__device__ uint32_t g_crc_table[256];
 static bool g_crc_table_init = false;
 
 static int ensure_crc_table() {

@@ -52,7 +52,7 @@ static thread_local std::string g_xerr;
 // XOR tree of 8 independent LDS lookups instead of an 8-deep dependent
 // chain. Staged to LDS — divergent indexing of __constant__ memory
 // serializes (each distinct address replays); LDS banks handle it at rate.
-__constant__ uint32_t gx_crc8_table[8 * 256];
+__device__ uint32_t gx_crc8_table[8 * 256];
 
 __device__ __forceinline__ void stage_crc_lut(uint32_t* lut) {
     for (int i = threadIdx.x; i < 8 * 256; i += blockDim.x)
