@@ -149,3 +149,38 @@ def test_join_spill_bytes():
              struct.pack("<I", 0))
     assert sp2 == want2, f"{sp2.hex()} vs {want2.hex()}"
     o.close()
+
+
+def test_join_spill_tristate_netting():
+    # across-epoch edge: pre-epoch row X; epoch does del X, ins X', del X'
+    # -> net DELETE (X is gone); and del X, ins X' -> net PUT (overwrite)
+    import struct
+
+    from rwtest.ffi import JOIN_INNER, SIDE_LEFT, T_I64, from_pretty, oracle
+    from rwtest import ffi
+
+    def memcmp_i64(v):
+        return b"\x00" + struct.pack(">q", v ^ -(1 << 63))
+
+    def val_i64(v):
+        return b"\x01" + struct.pack("<q", v)
+
+    o = ffi.HashJoin(oracle(), JOIN_INNER, [T_I64, T_I64], [T_I64, T_I64],
+                     key_l=[0], key_r=[0], pk_l=[1], pk_r=[1])
+    o.push(SIDE_LEFT, from_pretty(" I I\n + 1 10\n + 2 20"))
+    o.poll_all()
+    ffi.join_checkpoint_drain(oracle(), o.h, SIDE_LEFT)  # epoch 1: 2 PUTs
+    # epoch 2: key 1: del, re-ins (same pk), del again -> DELETE;
+    #          key 2: del, re-ins -> PUT
+    o.push(SIDE_LEFT, from_pretty(
+        " I I\n - 1 10\n + 1 10\n - 1 10\n - 2 20\n + 2 20"))
+    o.poll_all()
+    sp = ffi.join_checkpoint_drain(oracle(), o.h, SIDE_LEFT)
+    k1 = memcmp_i64(1) + memcmp_i64(10)
+    k2 = memcmp_i64(2) + memcmp_i64(20)
+    v2 = val_i64(2) + val_i64(20)
+    want = (b"\x00" + struct.pack("<I", len(k1)) + k1 + struct.pack("<I", 0) +
+            b"\x01" + struct.pack("<I", len(k2)) + k2 +
+            struct.pack("<I", len(v2)) + v2)
+    assert sp == want, f"{sp.hex()}\nvs\n{want.hex()}"
+    o.close()

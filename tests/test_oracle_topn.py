@@ -195,3 +195,27 @@ def test_topn_delete_update_pair():
     i = rows.index(("U-", (1, 5, 100)))
     assert rows[i + 1] == ("U+", (1, 5, 999))
     t.close()
+
+
+def test_topn_desc_and_deep_offset():
+    # DESC order key + offset beyond group size + NULL order values
+    # (NULLS FIRST under DESC — sort_util.rs defaults)
+    t = ffi.GroupTopN(oracle(), I3, [0], [(1, True)], [(2, False)],
+                      offset=2, limit=2)
+    t.push(from_pretty(""" I I I
+        + 1 10 1
+        + 1 30 2
+        + 1 20 3
+        + 1 . 4"""))
+    # DESC NULLS FIRST: order = NULL(4), 30(2), 20(3), 10(1);
+    # window [2,4) = 20(3), 10(1)
+    got = rows_multiset(t.poll_all())
+    want = rows_multiset([from_pretty(" I I I\n + 1 20 3\n + 1 10 1")])
+    assert got == want, got
+    # delete the NULL head: order becomes 30, 20, 10; window [2,4) = {10};
+    # old window {20, 10} -> delta = -20 only (10 stays)
+    t.push(from_pretty(" I I I\n - 1 . 4"))
+    got = rows_multiset(t.poll_all())
+    want = rows_multiset([from_pretty(" I I I\n - 1 20 3")])
+    assert got == want, got
+    t.close()
