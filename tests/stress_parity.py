@@ -1,0 +1,140 @@
+"""Soak: randomized GPU-vs-oracle parity across many seeds (not a pytest
+test — a standalone stress loop for gpurun budget). Covers the agg value +
+retractable-min/max states, all 8 join types, GroupTopN, and the join
+checkpoint spill, with per-seed randomized insert/delete mixes.
+
+Usage: python tests/stress_parity.py [n_seeds] [rows_per_push]
+"""
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import numpy as np
+
+import risingwave_amd
+from rwtest import ffi
+from rwtest.ffi import (AGG_COUNT_STAR, AGG_MAX, AGG_MIN, AGG_SUM, SIDE_LEFT,
+                        SIDE_RIGHT, T_I64, oracle, rows_multiset)
+
+risingwave_amd.load_library()
+GPU = ffi.Lib(risingwave_amd.lib_path())
+
+
+def mk(types, ops, cols):
+    return ffi.Chunk(types, ops, cols,
+                     [np.ones(len(ops), np.uint8) for _ in types])
+
+
+def stress_agg(seed, n):
+    rng = np.random.default_rng(seed)
+    calls = [(AGG_COUNT_STAR, -1, T_I64), (AGG_SUM, 1, T_I64),
+             (AGG_MIN, 1, T_I64), (AGG_MAX, 1, T_I64)]
+    g = ffi.HashAgg(GPU, [T_I64, T_I64], [0], calls, 0, stream_key=[1])
+    o = ffi.HashAgg(oracle(), [T_I64, T_I64], [0], calls, 0, stream_key=[1])
+    live = []
+    for ep in range(6):
+        keys = rng.integers(0, 500, n)
+        vals = rng.integers(-1000, 1000, n)
+        ops = np.zeros(n, np.uint8)
+        for r in range(n):
+            if live and rng.random() < 0.35:
+                jx = int(rng.integers(0, len(live)))
+                keys[r], vals[r] = live.pop(jx)
+                ops[r] = ffi.OP_DELETE
+            else:
+                live.append((int(keys[r]), int(vals[r])))
+        c = mk([T_I64, T_I64], ops, [keys, vals])
+        outs = []
+        for a in (g, o):
+            a.push(c)
+            a.flush(ep + 1)
+            outs.append(rows_multiset(a.poll_all()))
+        assert outs[0] == outs[1], f"agg seed {seed} epoch {ep}"
+    g.close()
+    o.close()
+
+
+def stress_join(seed, jt, n):
+    rng = np.random.default_rng(seed * 100 + jt)
+    g = ffi.HashJoin(GPU, jt, [T_I64, T_I64], [T_I64, T_I64],
+                     key_l=[0], key_r=[0], pk_l=[1], pk_r=[1])
+    o = ffi.HashJoin(oracle(), jt, [T_I64, T_I64], [T_I64, T_I64],
+                     key_l=[0], key_r=[0], pk_l=[1], pk_r=[1])
+    live = {SIDE_LEFT: [], SIDE_RIGHT: []}
+    pk = 0
+    for i in range(8):
+        side = int(rng.integers(0, 2))
+        keys = rng.integers(0, 150, n)
+        vals = np.arange(pk, pk + n)
+        pk += n
+        ops = np.zeros(n, np.uint8)
+        for r in range(n):
+            if live[side] and rng.random() < 0.3:
+                jx = int(rng.integers(0, len(live[side])))
+                keys[r], vals[r] = live[side].pop(jx)
+                ops[r] = ffi.OP_DELETE
+            else:
+                live[side].append((int(keys[r]), int(vals[r])))
+        c = mk([T_I64, T_I64], ops, [keys, vals])
+        g.push(side, c)
+        o.push(side, c)
+        from test_gpu_parity import net_rows
+
+        mg = net_rows(rows_multiset(g.poll_all()))
+        mo = net_rows(rows_multiset(o.poll_all()))
+        assert mg == mo, f"join type {jt} seed {seed} push {i}"
+        sg = ffi.join_checkpoint_drain(GPU, g.h, side)
+        so = ffi.join_checkpoint_drain(oracle(), o.h, side)
+        assert sg == so, f"spill type {jt} seed {seed} push {i}"
+    g.close()
+    o.close()
+
+
+def stress_topn(seed, n):
+    rng = np.random.default_rng(seed + 7)
+    t3 = [T_I64, T_I64, T_I64]
+    off, lim = int(rng.integers(0, 3)), int(rng.integers(1, 5))
+    desc = bool(rng.integers(0, 2))
+    g = ffi.GroupTopN(GPU, t3, [0], [(1, desc)], [(2, False)],
+                      offset=off, limit=lim)
+    o = ffi.GroupTopN(oracle(), t3, [0], [(1, desc)], [(2, False)],
+                      offset=off, limit=lim)
+    live = []
+    for i in range(6):
+        gk = rng.integers(0, 30, n)
+        ordv = rng.integers(0, 100, n)
+        pkv = rng.integers(0, 10**7, n)
+        ops = np.zeros(n, np.uint8)
+        for r in range(n):
+            if live and rng.random() < 0.35:
+                jx = int(rng.integers(0, len(live)))
+                gk[r], ordv[r], pkv[r] = live.pop(jx)
+                ops[r] = ffi.OP_DELETE
+            else:
+                live.append((int(gk[r]), int(ordv[r]), int(pkv[r])))
+        c = mk(t3, ops, [gk, ordv, pkv])
+        g.push(c)
+        o.push(c)
+        mg = rows_multiset(g.poll_all())
+        mo = rows_multiset(o.poll_all())
+        assert mg == mo, f"topn seed {seed} push {i} (off={off} lim={lim})"
+    g.close()
+    o.close()
+
+
+def main():
+    n_seeds = int(sys.argv[1]) if len(sys.argv) > 1 else 5
+    n = int(sys.argv[2]) if len(sys.argv) > 2 else 2048
+    for seed in range(1, n_seeds + 1):
+        stress_agg(seed, n)
+        for jt in range(8):
+            stress_join(seed, jt, max(n // 2, 512))
+        stress_topn(seed, max(n // 4, 256))
+        print(f"seed {seed}: agg + 8 join types + topn OK")
+    print(f"STRESS OK: {n_seeds} seeds")
+
+
+if __name__ == "__main__":
+    main()
