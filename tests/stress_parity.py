@@ -28,24 +28,32 @@ def mk(types, ops, cols):
 
 
 def stress_agg(seed, n):
+    # schema (key, val, sk): sk is a UNIQUE stream key — the reference's
+    # upstream guarantees stream-key uniqueness, duplicate stream keys are
+    # out-of-contract input. vals repeat freely (retraction stress).
     rng = np.random.default_rng(seed)
     calls = [(AGG_COUNT_STAR, -1, T_I64), (AGG_SUM, 1, T_I64),
              (AGG_MIN, 1, T_I64), (AGG_MAX, 1, T_I64)]
-    g = ffi.HashAgg(GPU, [T_I64, T_I64], [0], calls, 0, stream_key=[1])
-    o = ffi.HashAgg(oracle(), [T_I64, T_I64], [0], calls, 0, stream_key=[1])
+    t3 = [T_I64, T_I64, T_I64]
+    g = ffi.HashAgg(GPU, t3, [0], calls, 0, stream_key=[2])
+    o = ffi.HashAgg(oracle(), t3, [0], calls, 0, stream_key=[2])
     live = []
+    sk_next = 0
     for ep in range(6):
         keys = rng.integers(0, 500, n)
         vals = rng.integers(-1000, 1000, n)
+        sks = np.zeros(n, np.int64)
         ops = np.zeros(n, np.uint8)
         for r in range(n):
             if live and rng.random() < 0.35:
                 jx = int(rng.integers(0, len(live)))
-                keys[r], vals[r] = live.pop(jx)
+                keys[r], vals[r], sks[r] = live.pop(jx)
                 ops[r] = ffi.OP_DELETE
             else:
-                live.append((int(keys[r]), int(vals[r])))
-        c = mk([T_I64, T_I64], ops, [keys, vals])
+                sks[r] = sk_next
+                sk_next += 1
+                live.append((int(keys[r]), int(vals[r]), int(sks[r])))
+        c = mk(t3, ops, [keys, vals, sks])
         outs = []
         for a in (g, o):
             a.push(c)
