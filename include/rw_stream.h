@@ -66,6 +66,10 @@ typedef struct RwAggCall {
     uint8_t kind;     /* RwAggKind */
     int32_t arg;      /* input column index, -1 for count(*) */
     uint8_t ret_type; /* RwTypeId of the output */
+    uint8_t distinct; /* DISTINCT dedup on `arg` (aggregate/distinct.rs:67-
+                         198): per (group, datum) counts gate the call's row
+                         visibility — an insert is visible iff the count
+                         rises 0→1, a delete iff it falls 1→0 */
 } RwAggCall;
 
 typedef struct RwHashAggDesc {

@@ -111,9 +111,32 @@ struct HashAggOracle {
             bool minput = (c.kind == RW_AGG_MIN || c.kind == RW_AGG_MAX) && !append_only;
             call_is_minput.push_back(minput);
         }
+        // DISTINCT dedup (aggregate/distinct.rs:67-198): one counter map per
+        // distinct column (the calls here carry no per-call filters, so all
+        // calls distinct on a column share one count and one visibility)
+        call_dedup_idx.assign(calls.size(), -1);
+        for (size_t ci = 0; ci < calls.size(); ci++) {
+            if (!calls[ci].distinct || calls[ci].arg < 0) continue;
+            uint32_t col = (uint32_t)calls[ci].arg;
+            size_t di = 0;
+            for (; di < distinct_cols.size(); di++)
+                if (distinct_cols[di] == col) break;
+            if (di == distinct_cols.size()) {
+                distinct_cols.push_back(col);
+                RowOrderLess less;
+                for (auto t : group_key_types) less.order.push_back({t, false});
+                less.order.push_back({input_types[col], false});
+                dedup_counts.emplace_back(less);
+            }
+            call_dedup_idx[ci] = (int)di;
+        }
         groups = decltype(groups)(16, KeyHash{}, KeyEq{&group_key_types});
         dirty = decltype(dirty)(16, KeyHash{}, KeyEq{&group_key_types});
     }
+
+    std::vector<uint32_t> distinct_cols;
+    std::vector<std::map<Row, int64_t, RowOrderLess>> dedup_counts;
+    std::vector<int> call_dedup_idx; // per call: index into dedup_counts or -1
 
     RowOrderLess minput_order(const RwAggCall& c) const {
         // pk of the materialized-input table: value (ASC min / DESC max),
@@ -147,9 +170,11 @@ struct HashAggOracle {
         return it->second;
     }
 
-    int apply_row(AggGroupState& g, const ChunkView& cv, size_t r, bool retract) {
+    int apply_row(AggGroupState& g, const ChunkView& cv, size_t r, bool retract,
+                  const std::vector<bool>& hidden) {
         for (size_t ci = 0; ci < calls.size(); ci++) {
             const auto& c = calls[ci];
+            if (!hidden.empty() && hidden[ci]) continue; // DISTINCT dup row
             if (call_is_minput[ci]) {
                 // minput materializes the row keyed by [value, stream_key...]
                 Row key;
@@ -220,6 +245,8 @@ struct HashAggOracle {
 
     int push_chunk(const RwChunk* chunk) {
         ChunkView cv{chunk};
+        std::vector<bool> hidden;
+        std::vector<bool> col_hidden(distinct_cols.size());
         for (size_t r = 0; r < cv.n_rows(); r++) {
             if (!cv.visible(r)) continue;
             Row key;
@@ -228,7 +255,30 @@ struct HashAggOracle {
             auto& g = touch(key);
             uint8_t op = cv.op(r);
             bool retract = (op == RW_OP_DELETE || op == RW_OP_UPDATE_DELETE);
-            int rc = apply_row(g, cv, r, retract);
+            if (!distinct_cols.empty()) {
+                // DISTINCT dedup (distinct.rs:131-158): count per
+                // (group, datum); insert visible iff count 0→1, delete iff
+                // 1→0; a count that drops to 0 is removed (:185-191)
+                for (size_t di = 0; di < distinct_cols.size(); di++) {
+                    Row dk = key;
+                    dk.push_back(cv.at(r, distinct_cols[di]));
+                    auto& m = dedup_counts[di];
+                    if (!retract) {
+                        int64_t cnt = ++m[dk];
+                        col_hidden[di] = cnt > 1;
+                    } else {
+                        auto it = m.find(dk);
+                        int64_t cnt = it == m.end() ? -1 : --it->second;
+                        col_hidden[di] = cnt > 0;
+                        if (it != m.end() && cnt == 0) m.erase(it);
+                    }
+                }
+                hidden.assign(calls.size(), false);
+                for (size_t ci = 0; ci < calls.size(); ci++)
+                    if (call_dedup_idx[ci] >= 0)
+                        hidden[ci] = col_hidden[call_dedup_idx[ci]];
+            }
+            int rc = apply_row(g, cv, r, retract, hidden);
             if (rc != RW_OK) return rc;
         }
         return RW_OK;

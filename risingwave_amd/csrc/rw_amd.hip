@@ -196,7 +196,197 @@ struct AggBatch {
     uint32_t stride = 1;
     uint8_t valid_inverted = 0;
     uint8_t dense = 0; // all rows visible+Insert+non-null (checked at upload)
+    // DISTINCT dedup visibility (aggregate/distinct.rs): per call, byte per
+    // row (indexed by r*stride) — 1 = hidden duplicate; null = no dedup
+    const uint8_t* call_hidden[MAX_CALLS] = {};
 };
+
+// AoS layouts: one 64-B line per key slot, one packed record per row —
+// a random probe touches 1 table line + 1 record line instead of the 8–10
+// lines the SoA arrays cost (DESIGN §8 lever).
+struct alignas(64) JoinSlot {
+    uint32_t state; // SLOT_EMPTY/CLAIMED/READY
+    uint32_t head;  // chain head row index, UINT32_MAX = none
+    uint32_t nulls; // key null mask
+    uint32_t _pad;
+    long long key[4];
+};
+static_assert(sizeof(JoinSlot) == 64, "one cache line per slot");
+
+// row record header; vals[n_cols] i64 follow at offset 16
+struct JoinRowHdr {
+    uint32_t alive; // CAS-claimed tombstone
+    uint32_t next;
+    uint32_t validbits; // bit c = column c non-NULL
+    uint32_t degree;    // matches on the other side (outer/semi/anti types)
+};
+
+struct JoinSideDev {
+    JoinSlot* slots;
+    uint32_t cap_mask;
+    uint8_t* rows;       // packed records, row_stride bytes each (16-B aligned)
+    uint32_t row_stride;  // 16 + 8*n_cols
+    uint32_t* row_cursor; // single counter
+    uint32_t row_cap;
+    // checkpoint-delta tracking (§8f-2): rows killed by deletes this epoch
+    // (null for executors without spill, e.g. GroupTopN this round)
+    uint32_t* killed;
+    uint32_t* killed_cursor;
+    uint32_t killed_cap;
+};
+
+__device__ __forceinline__ JoinRowHdr* jrow(const JoinSideDev& s, uint32_t r) {
+    return (JoinRowHdr*)(s.rows + (size_t)r * s.row_stride);
+}
+__device__ __forceinline__ long long* jvals(JoinRowHdr* h) {
+    return (long long*)((uint8_t*)h + 16);
+}
+
+// own-side find-or-insert. The probe walk uses PLAIN cached loads: a slot's
+// state and keys share one 64-B line, and the claim protocol drains the sc1
+// key stores to the coherence point BEFORE the sc1 READY store — so any
+// line fill that observes READY also contains the keys (same-line, written
+// earlier at the coherence point), and a stale cached line can only show
+// the older EMPTY/CLAIMED state, which funnels into the coherent CAS/spin
+// path below. Keys never change once READY. This removes the per-visit sc1
+// word loads that dominated the insert path (~0.45 ms/1M inserts measured).
+__device__ __forceinline__ uint32_t jslot_find_or_insert(JoinSlot* slots,
+                                                         uint32_t cap_mask,
+                                                         const int64_t* kw,
+                                                         uint32_t nullmask,
+                                                         int KW) {
+    uint32_t slot = (uint32_t)(hash_key(kw, nullmask, KW) & cap_mask);
+    for (uint32_t probes = 0; probes <= cap_mask; probes++) {
+        JoinSlot* sl = &slots[slot];
+        uint32_t st = sl->state; // plain (fast path)
+        if (st == SLOT_READY) {
+            bool eq = sl->nulls == nullmask;
+            for (int i = 0; eq && i < KW; i++) eq = sl->key[i] == kw[i];
+            if (eq) return slot;
+        } else {
+            if (st == SLOT_EMPTY) {
+                uint32_t prev = atomicCAS(&sl->state, SLOT_EMPTY, SLOT_CLAIMED);
+                if (prev == SLOT_EMPTY) {
+                    for (int i = 0; i < KW; i++)
+                        st_i64((int64_t*)&sl->key[i], kw[i]);
+                    st_u32(&sl->nulls, nullmask);
+                    asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); // R1 drain
+                    st_u32(&sl->state, SLOT_READY);
+                    return slot;
+                }
+                st = prev;
+            } else {
+                st = ld_u32(&sl->state); // sc1 refresh of a stale CLAIMED
+            }
+            while (st == SLOT_CLAIMED) {
+                __builtin_amdgcn_s_sleep(1);
+                st = ld_u32(&sl->state);
+            }
+            // coherent compare (the plain line may be stale here)
+            bool eq = ld_u32(&sl->nulls) == nullmask;
+            for (int i = 0; eq && i < KW; i++)
+                eq = ld_i64((const int64_t*)&sl->key[i]) == kw[i];
+            if (eq) return slot;
+        }
+        slot = (slot + 1) & cap_mask;
+    }
+    return (uint32_t)-1;
+}
+
+// match-side find with plain cached loads (immutable during the launch)
+__device__ __forceinline__ uint32_t jslot_find_cached(const JoinSlot* slots,
+                                                      uint32_t cap_mask,
+                                                      const int64_t* kw,
+                                                      uint32_t nullmask, int KW) {
+    uint32_t slot = (uint32_t)(hash_key(kw, nullmask, KW) & cap_mask);
+    for (uint32_t probes = 0; probes <= cap_mask; probes++) {
+        const JoinSlot* sl = &slots[slot];
+        uint32_t st = sl->state;
+        if (st == SLOT_EMPTY) return (uint32_t)-1;
+        if (st == SLOT_READY) {
+            bool eq = sl->nulls == nullmask;
+            for (int i = 0; eq && i < KW; i++) eq = sl->key[i] == kw[i];
+            if (eq) return slot;
+        }
+        slot = (slot + 1) & cap_mask;
+    }
+    return (uint32_t)-1;
+}
+
+// sc1 variant of the find (append-only path mutates the match side)
+__device__ __forceinline__ uint32_t jslot_find_sc1(const JoinSlot* slots,
+                                                   uint32_t cap_mask,
+                                                   const int64_t* kw,
+                                                   uint32_t nullmask, int KW) {
+    uint32_t slot = (uint32_t)(hash_key(kw, nullmask, KW) & cap_mask);
+    for (uint32_t probes = 0; probes <= cap_mask; probes++) {
+        const JoinSlot* sl = &slots[slot];
+        uint32_t st = ld_u32(&sl->state);
+        if (st == SLOT_EMPTY) return (uint32_t)-1;
+        if (st == SLOT_READY) {
+            bool eq = ld_u32(&sl->nulls) == nullmask;
+            for (int i = 0; eq && i < KW; i++)
+                eq = ld_i64((const int64_t*)&sl->key[i]) == kw[i];
+            if (eq) return slot;
+        }
+        slot = (slot + 1) & cap_mask;
+    }
+    return (uint32_t)-1;
+}
+
+__global__ void jslot_init_kernel(JoinSlot* slots, size_t cap) {
+    size_t stride = (size_t)gridDim.x * blockDim.x;
+    for (size_t i = blockIdx.x * blockDim.x + threadIdx.x; i < cap; i += stride) {
+        slots[i].state = SLOT_EMPTY;
+        slots[i].head = UINT32_MAX;
+    }
+}
+
+// DISTINCT dedup pass (aggregate/distinct.rs:131-158): per (group key,
+// datum) a counter slot (JoinSlot layout; `head` is the count, initialized
+// 0). An insert is visible iff the count rises 0→1, a delete iff it falls
+// 1→0 — transition ownership via the atomic's return value, so concurrent
+// duplicates resolve to the same net visibility multiset as the
+// reference's sequential pass (the accumulators only see the net).
+__global__ void dedup_init_kernel(JoinSlot* slots, size_t cap) {
+    size_t stride = (size_t)gridDim.x * blockDim.x;
+    for (size_t i = blockIdx.x * blockDim.x + threadIdx.x; i < cap; i += stride) {
+        slots[i].state = SLOT_EMPTY;
+        slots[i].head = 0; // count
+    }
+}
+
+__global__ void agg_dedup_kernel(AggBatch b, JoinSlot* slots,
+                                 uint32_t cap_mask, int KW, int dslot,
+                                 uint8_t* hidden, uint32_t* err) {
+    uint32_t stride = gridDim.x * blockDim.x;
+    for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < b.n_rows;
+         r += stride) {
+        size_t rs = (size_t)r * b.stride;
+        if (b.vis && !b.vis[r]) continue;
+        int64_t kw[MAX_KW];
+        uint32_t nullmask = 0;
+        for (int i = 0; i < KW; i++) {
+            bool valid = b.col_valid[i][rs] ^ b.valid_inverted;
+            kw[i] = valid ? b.col_vals[i][rs] : 0;
+            nullmask |= (!valid) << i;
+        }
+        bool dvalid = b.col_valid[dslot][rs] ^ b.valid_inverted;
+        kw[KW] = dvalid ? b.col_vals[dslot][rs] : 0;
+        nullmask |= (!dvalid) << KW;
+        uint32_t slot =
+            jslot_find_or_insert(slots, cap_mask, kw, nullmask, KW + 1);
+        if (slot == UINT32_MAX) {
+            atomicExch(err, 4u); // dedup table full
+            continue;
+        }
+        uint8_t op = b.ops[r];
+        bool retract = (op == RW_OP_DELETE || op == RW_OP_UPDATE_DELETE);
+        uint32_t old = retract ? atomicSub(&slots[slot].head, 1u)
+                               : atomicAdd(&slots[slot].head, 1u);
+        hidden[rs] = retract ? (old != 1) : (old != 0);
+    }
+}
 
 struct AggTableDev {
     uint32_t* state;
@@ -406,8 +596,11 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
                 }
                 continue;
             }
+            bool shown = contributing &&
+                         !(b.call_hidden[ci] && b.call_hidden[ci][rs]);
+            arg_valid = arg_valid && shown;
             switch (c.kind) {
-                case RW_AGG_COUNT_STAR: v[ci] = contributing ? sign : 0; break;
+                case RW_AGG_COUNT_STAR: v[ci] = shown ? sign : 0; break;
                 case RW_AGG_COUNT: v[ci] = arg_valid ? sign : 0; break;
                 case RW_AGG_SUM:
                 case RW_AGG_SUM0:
@@ -1013,6 +1206,13 @@ struct HashAgg {
     int n_minput = 0;   // materialized-input (retractable min/max) calls
     std::vector<uint8_t> call_minput;
     std::vector<uint32_t> stream_key;
+    // DISTINCT dedup state (aggregate/distinct.rs)
+    std::vector<int> distinct_slots;  // batch slot per dedup table
+    std::vector<int> call_dedup_idx;  // per call: dedup table index or -1
+    std::vector<JoinSlot*> dedup_slots;
+    uint32_t dedup_cap_mask = 0;
+    std::vector<uint8_t*> hidden_bufs; // device, per dedup table
+    uint32_t hidden_cap = 0;
 
     int grid_for(uint32_t work) const {
         uint32_t blocks = (work + 255) / 256;
@@ -1057,6 +1257,31 @@ struct HashAgg {
                 if (ty != RW_T_I64 && ty != RW_T_TS)
                     FAIL(RW_E_INVAL, "agg arg type %d unsupported on GPU (i64/ts only)", ty);
             }
+        }
+        // DISTINCT dedup (aggregate/distinct.rs): one counter table per
+        // distinct column. min/max DISTINCT ≡ min/max — the frontend strips
+        // it — so distinct over a materialized-input call is rejected.
+        call_dedup_idx.assign(calls.size(), -1);
+        for (size_t ci = 0; ci < calls.size(); ci++) {
+            if (!calls[ci].distinct || calls[ci].arg < 0) continue;
+            if (call_minput[ci])
+                FAIL(RW_E_INVAL,
+                     "DISTINCT min/max is min/max — plan should strip it");
+            if (KW + 1 > MAX_KW)
+                FAIL(RW_E_INVAL, "group key too wide for DISTINCT dedup");
+            int slot = KW + (int)ci; // arg column's batch slot
+            size_t di = 0;
+            for (; di < distinct_slots.size(); di++)
+                if (distinct_slots[di] == slot) break;
+            // distinct calls on the same arg COLUMN may use different batch
+            // slots (args are duplicated per call) — dedupe by source col
+            for (size_t cj = 0; cj < ci; cj++)
+                if (call_dedup_idx[cj] >= 0 && calls[cj].arg == calls[ci].arg) {
+                    di = (size_t)call_dedup_idx[cj];
+                    break;
+                }
+            if (di == distinct_slots.size()) distinct_slots.push_back(slot);
+            call_dedup_idx[ci] = (int)di;
         }
         if (n_minput) {
             if (stream_key.size() > 2)
@@ -1222,6 +1447,7 @@ struct HashAgg {
     void launch_apply(const AggBatch& b, uint32_t r0, uint32_t r1) {
         auto a0 = cd(0), a1 = cd(1), a2 = cd(2), a3 = cd(3);
         if (b.dense && KW == 1 && b.stride == 1 && n_minput == 0 &&
+            distinct_slots.empty() &&
             debug_mode == 0 && ((uintptr_t)(b.col_vals[0] + r0) & 31) == 0 &&
             (r1 - r0) >= 1024) {
             // rows-per-lane (A/B-selectable via RW_AGG_RPL). Measured on
@@ -1393,10 +1619,47 @@ struct HashAgg {
         return bounds;
     }
 
+    int ensure_dedup(uint32_t n) {
+        if (dedup_slots.empty()) {
+            size_t cap = 1;
+            size_t hint = (size_t)capacity * 4;
+            if (hint < (1u << 16)) hint = 1u << 16;
+            while (cap < hint) cap <<= 1;
+            dedup_cap_mask = (uint32_t)(cap - 1);
+            for (size_t di = 0; di < distinct_slots.size(); di++) {
+                JoinSlot* sl = nullptr;
+                HIP_TRY(hipMalloc(&sl, cap * sizeof(JoinSlot)));
+                dedup_init_kernel<<<2048, 256, 0, stream>>>(sl, cap);
+                dedup_slots.push_back(sl);
+            }
+        }
+        if (hidden_cap < n) {
+            for (auto*& hb : hidden_bufs)
+                if (hb) hipFree(hb);
+            hidden_bufs.assign(distinct_slots.size(), nullptr);
+            for (size_t di = 0; di < distinct_slots.size(); di++)
+                HIP_TRY(hipMalloc(&hidden_bufs[di], n));
+            hidden_cap = n;
+        }
+        return RW_OK;
+    }
+
     int push_chunk(const RwChunk* c) {
         AggBatch b;
         int rc = upload(c, &b, true);
         if (rc != RW_OK) return rc;
+        if (!distinct_slots.empty()) {
+            rc = ensure_dedup(c->n_rows);
+            if (rc != RW_OK) return rc;
+            int grid = grid_for(c->n_rows);
+            for (size_t di = 0; di < distinct_slots.size(); di++)
+                agg_dedup_kernel<<<grid, 256, 0, stream>>>(
+                    b, dedup_slots[di], dedup_cap_mask, KW, distinct_slots[di],
+                    hidden_bufs[di], &t.counters[2]);
+            for (size_t ci = 0; ci < calls.size(); ci++)
+                if (call_dedup_idx[ci] >= 0)
+                    b.call_hidden[ci] = hidden_bufs[call_dedup_idx[ci]];
+        }
         rc = apply(b, true, minput_conflict_segments(c));
         if (rc != RW_OK) return rc;
         HIP_TRY(hipStreamSynchronize(stream)); // staging buffer reuse
@@ -1409,6 +1672,7 @@ struct HashAgg {
         if (ctr[2] == 1) FAIL(RW_E_INTERNAL, "agg state table full (capacity %u)", capacity);
         if (ctr[2] == 2) FAIL(RW_E_INTERNAL, "agg output buffer overflow");
         if (ctr[2] == 3) FAIL(RW_E_INTERNAL, "agg minput row store full");
+        if (ctr[2] == 4) FAIL(RW_E_INTERNAL, "agg distinct dedup table full");
         return RW_OK;
     }
 
@@ -2114,146 +2378,9 @@ int rw_dispatch_compute(const RwDispatchDesc* d, const RwChunk* chunk,
 #define MAX_COLS 8
 #define MAX_OUT 16
 
-// AoS layouts: one 64-B line per key slot, one packed record per row —
-// a random probe touches 1 table line + 1 record line instead of the 8–10
-// lines the SoA arrays cost (DESIGN §8 lever).
-struct alignas(64) JoinSlot {
-    uint32_t state; // SLOT_EMPTY/CLAIMED/READY
-    uint32_t head;  // chain head row index, UINT32_MAX = none
-    uint32_t nulls; // key null mask
-    uint32_t _pad;
-    long long key[4];
-};
-static_assert(sizeof(JoinSlot) == 64, "one cache line per slot");
+// (JoinSlot/jslot helpers and the DISTINCT dedup kernels moved before
+// HashAgg — shared by the agg dedup tables)
 
-// row record header; vals[n_cols] i64 follow at offset 16
-struct JoinRowHdr {
-    uint32_t alive; // CAS-claimed tombstone
-    uint32_t next;
-    uint32_t validbits; // bit c = column c non-NULL
-    uint32_t degree;    // matches on the other side (outer/semi/anti types)
-};
-
-struct JoinSideDev {
-    JoinSlot* slots;
-    uint32_t cap_mask;
-    uint8_t* rows;       // packed records, row_stride bytes each (16-B aligned)
-    uint32_t row_stride;  // 16 + 8*n_cols
-    uint32_t* row_cursor; // single counter
-    uint32_t row_cap;
-    // checkpoint-delta tracking (§8f-2): rows killed by deletes this epoch
-    // (null for executors without spill, e.g. GroupTopN this round)
-    uint32_t* killed;
-    uint32_t* killed_cursor;
-    uint32_t killed_cap;
-};
-
-__device__ __forceinline__ JoinRowHdr* jrow(const JoinSideDev& s, uint32_t r) {
-    return (JoinRowHdr*)(s.rows + (size_t)r * s.row_stride);
-}
-__device__ __forceinline__ long long* jvals(JoinRowHdr* h) {
-    return (long long*)((uint8_t*)h + 16);
-}
-
-// own-side find-or-insert. The probe walk uses PLAIN cached loads: a slot's
-// state and keys share one 64-B line, and the claim protocol drains the sc1
-// key stores to the coherence point BEFORE the sc1 READY store — so any
-// line fill that observes READY also contains the keys (same-line, written
-// earlier at the coherence point), and a stale cached line can only show
-// the older EMPTY/CLAIMED state, which funnels into the coherent CAS/spin
-// path below. Keys never change once READY. This removes the per-visit sc1
-// word loads that dominated the insert path (~0.45 ms/1M inserts measured).
-__device__ __forceinline__ uint32_t jslot_find_or_insert(JoinSlot* slots,
-                                                         uint32_t cap_mask,
-                                                         const int64_t* kw,
-                                                         uint32_t nullmask,
-                                                         int KW) {
-    uint32_t slot = (uint32_t)(hash_key(kw, nullmask, KW) & cap_mask);
-    for (uint32_t probes = 0; probes <= cap_mask; probes++) {
-        JoinSlot* sl = &slots[slot];
-        uint32_t st = sl->state; // plain (fast path)
-        if (st == SLOT_READY) {
-            bool eq = sl->nulls == nullmask;
-            for (int i = 0; eq && i < KW; i++) eq = sl->key[i] == kw[i];
-            if (eq) return slot;
-        } else {
-            if (st == SLOT_EMPTY) {
-                uint32_t prev = atomicCAS(&sl->state, SLOT_EMPTY, SLOT_CLAIMED);
-                if (prev == SLOT_EMPTY) {
-                    for (int i = 0; i < KW; i++)
-                        st_i64((int64_t*)&sl->key[i], kw[i]);
-                    st_u32(&sl->nulls, nullmask);
-                    asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); // R1 drain
-                    st_u32(&sl->state, SLOT_READY);
-                    return slot;
-                }
-                st = prev;
-            } else {
-                st = ld_u32(&sl->state); // sc1 refresh of a stale CLAIMED
-            }
-            while (st == SLOT_CLAIMED) {
-                __builtin_amdgcn_s_sleep(1);
-                st = ld_u32(&sl->state);
-            }
-            // coherent compare (the plain line may be stale here)
-            bool eq = ld_u32(&sl->nulls) == nullmask;
-            for (int i = 0; eq && i < KW; i++)
-                eq = ld_i64((const int64_t*)&sl->key[i]) == kw[i];
-            if (eq) return slot;
-        }
-        slot = (slot + 1) & cap_mask;
-    }
-    return (uint32_t)-1;
-}
-
-// match-side find with plain cached loads (immutable during the launch)
-__device__ __forceinline__ uint32_t jslot_find_cached(const JoinSlot* slots,
-                                                      uint32_t cap_mask,
-                                                      const int64_t* kw,
-                                                      uint32_t nullmask, int KW) {
-    uint32_t slot = (uint32_t)(hash_key(kw, nullmask, KW) & cap_mask);
-    for (uint32_t probes = 0; probes <= cap_mask; probes++) {
-        const JoinSlot* sl = &slots[slot];
-        uint32_t st = sl->state;
-        if (st == SLOT_EMPTY) return (uint32_t)-1;
-        if (st == SLOT_READY) {
-            bool eq = sl->nulls == nullmask;
-            for (int i = 0; eq && i < KW; i++) eq = sl->key[i] == kw[i];
-            if (eq) return slot;
-        }
-        slot = (slot + 1) & cap_mask;
-    }
-    return (uint32_t)-1;
-}
-
-// sc1 variant of the find (append-only path mutates the match side)
-__device__ __forceinline__ uint32_t jslot_find_sc1(const JoinSlot* slots,
-                                                   uint32_t cap_mask,
-                                                   const int64_t* kw,
-                                                   uint32_t nullmask, int KW) {
-    uint32_t slot = (uint32_t)(hash_key(kw, nullmask, KW) & cap_mask);
-    for (uint32_t probes = 0; probes <= cap_mask; probes++) {
-        const JoinSlot* sl = &slots[slot];
-        uint32_t st = ld_u32(&sl->state);
-        if (st == SLOT_EMPTY) return (uint32_t)-1;
-        if (st == SLOT_READY) {
-            bool eq = ld_u32(&sl->nulls) == nullmask;
-            for (int i = 0; eq && i < KW; i++)
-                eq = ld_i64((const int64_t*)&sl->key[i]) == kw[i];
-            if (eq) return slot;
-        }
-        slot = (slot + 1) & cap_mask;
-    }
-    return (uint32_t)-1;
-}
-
-__global__ void jslot_init_kernel(JoinSlot* slots, size_t cap) {
-    size_t stride = (size_t)gridDim.x * blockDim.x;
-    for (size_t i = blockIdx.x * blockDim.x + threadIdx.x; i < cap; i += stride) {
-        slots[i].state = SLOT_EMPTY;
-        slots[i].head = UINT32_MAX;
-    }
-}
 
 struct JoinMeta {
     int KW;

@@ -49,7 +49,8 @@ class RwChunkC(C.Structure):
 
 
 class RwAggCall(C.Structure):
-    _fields_ = [("kind", C.c_uint8), ("arg", C.c_int32), ("ret_type", C.c_uint8)]
+    _fields_ = [("kind", C.c_uint8), ("arg", C.c_int32), ("ret_type", C.c_uint8),
+                ("distinct", C.c_uint8)]
 
 
 class RwHashAggDesc(C.Structure):
@@ -280,10 +281,12 @@ class HashAgg:
         d.group_key_indices = self._gk
         d.n_calls = len(calls)
         self._calls = (RwAggCall * len(calls))()
-        for i, (k, a, rt) in enumerate(calls):
-            self._calls[i].kind = k
-            self._calls[i].arg = a
-            self._calls[i].ret_type = rt
+        for i, call in enumerate(calls):
+            # (kind, arg, ret_type[, distinct])
+            self._calls[i].kind = call[0]
+            self._calls[i].arg = call[1]
+            self._calls[i].ret_type = call[2]
+            self._calls[i].distinct = call[3] if len(call) > 3 else 0
         d.calls = self._calls
         d.row_count_index = row_count_index
         d.n_stream_key = len(stream_key)

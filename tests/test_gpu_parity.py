@@ -822,3 +822,37 @@ def test_join_checkpoint_spill_parity():
                               f"{len(sg)} vs {len(so)} bytes")
     g.close()
     o.close()
+
+
+def test_count_distinct_parity():
+    # DISTINCT dedup (aggregate/distinct.rs): GPU vs oracle — hand case +
+    # randomized duplicate-heavy insert/delete mix
+    from rwtest.ffi import AGG_COUNT
+
+    calls = [(AGG_COUNT_STAR, -1, T_I64), (AGG_COUNT, 1, T_I64, 1),
+             (AGG_SUM, 1, T_I64, 1)]
+    g = ffi.HashAgg(gpu(), [T_I64, T_I64], [0], calls, 0)
+    o = ffi.HashAgg(oracle(), [T_I64, T_I64], [0], calls, 0)
+    rng = np.random.default_rng(321)
+    live = []
+    for ep in range(5):
+        n = 2048
+        keys = rng.integers(0, 50, n)
+        vals = rng.integers(0, 8, n)  # heavy duplication per (group, datum)
+        ops = np.zeros(n, np.uint8)
+        for r in range(n):
+            if live and rng.random() < 0.4:
+                jx = int(rng.integers(0, len(live)))
+                keys[r], vals[r] = live.pop(jx)
+                ops[r] = ffi.OP_DELETE
+            else:
+                live.append((int(keys[r]), int(vals[r])))
+        c = mk_chunk([T_I64, T_I64], ops, [keys, vals])
+        outs = []
+        for a in (g, o):
+            a.push(c)
+            a.flush(ep + 1)
+            outs.append(rows_multiset(a.poll_all()))
+        assert outs[0] == outs[1], f"epoch {ep}"
+    g.close()
+    o.close()

@@ -231,3 +231,37 @@ def test_agg_watermark_group_cleaning():
         [("+", (1, 1, 3)), ("U-", (9, 1, 7)), ("U+", (9, 2, 8))]
     )
     agg.close()
+
+
+def test_count_distinct():
+    # DISTINCT dedup (aggregate/distinct.rs:67-198 semantics at the executor
+    # level): count(*) + count(DISTINCT a) + sum(DISTINCT a), multiset
+    # tracked by hand. Insert visible iff the (group, datum) count goes 0->1,
+    # delete iff 1->0.
+    from rwtest.ffi import AGG_COUNT
+
+    agg = ffi.HashAgg(
+        oracle(), [T_I64, T_I64], [0],
+        calls=[(AGG_COUNT_STAR, -1, T_I64), (AGG_COUNT, 1, T_I64, 1),
+               (AGG_SUM, 1, T_I64, 1)],
+        row_count_index=0,
+    )
+    # group 1 multiset after push: {1x2, 2x2, 3x1} -> distinct {1,2,3}
+    agg.push(from_pretty(" I I\n + 1 1\n + 1 2\n + 1 2\n + 1 1\n + 1 3"))
+    agg.flush(1)
+    assert rows_multiset(agg.poll_all()) == expect([("+", (1, 5, 3, 6))])
+    # -1, -2, +2, -3 -> multiset {1x1, 2x2} -> distinct {1,2}
+    agg.push(from_pretty(" I I\n - 1 1\n - 1 2\n + 1 2\n - 1 3"))
+    agg.flush(2)
+    assert rows_multiset(agg.poll_all()) == expect(
+        [("U-", (1, 5, 3, 6)), ("U+", (1, 3, 2, 3))]
+    )
+    # retract everything -> group deleted (emit prev)
+    agg.push(from_pretty(" I I\n - 1 1\n - 1 2\n - 1 2"))
+    agg.flush(3)
+    assert rows_multiset(agg.poll_all()) == expect([("-", (1, 3, 2, 3))])
+    # reinsert: dedup state must have been cleaned (counts dropped to 0)
+    agg.push(from_pretty(" I I\n + 1 7\n + 1 7"))
+    agg.flush(4)
+    assert rows_multiset(agg.poll_all()) == expect([("+", (1, 2, 1, 7))])
+    agg.close()
