@@ -209,6 +209,10 @@ typedef struct RwGroupTopNDesc {
     const uint8_t* rest_desc;
     uint64_t offset;
     uint64_t limit; /* > 0 */
+    uint8_t with_ties; /* TopNCache<true> (top_n_cache.rs:539-758): the
+                          window extends past `limit` while rows TIE with
+                          the limit-th row on the order-by prefix (offset
+                          must be 0, as in the reference) */
     uint32_t chunk_size;
     uint64_t state_capacity_hint; /* expected group count */
     uint64_t row_capacity_hint;   /* expected resident rows */

@@ -65,8 +65,24 @@ struct GroupTopNOracle {
         for (auto g : group_by) group_less.order.push_back({types[g], false});
         offset = d->offset;
         limit = d->limit;
+        with_ties = d->with_ties != 0;
+        n_order = d->n_order_by;
         chunk_size = d->chunk_size ? d->chunk_size : 1024;
         groups = decltype(groups)(group_less);
+    }
+
+    bool with_ties = false;
+    uint32_t n_order = 0;
+
+    // compare only the order-by prefix of two cache keys (sort key,
+    // top_n_cache.rs:539-560: ties are on (CacheKey).0)
+    int sort_key_cmp(const Row& a, const Row& b) const {
+        for (uint32_t i = 0; i < n_order; i++) {
+            int c = datum_cmp(a[i], b[i], ck_less.order[i].type);
+            if (ck_less.order[i].desc) c = -c;
+            if (c) return c;
+        }
+        return 0;
     }
 
     Row project(const Row& row, const std::vector<uint32_t>& idx) const {
@@ -79,9 +95,18 @@ struct GroupTopNOracle {
     std::vector<Row> window_of(const Group& g) const {
         std::vector<Row> w; // cache keys only
         uint64_t i = 0;
+        const Row* cut = nullptr; // limit-th row's cache key (ties boundary)
         for (auto& kv : g) {
-            if (i >= offset + limit) break;
-            if (i >= offset) w.push_back(kv.first);
+            if (i >= offset + limit) {
+                // WITH TIES: keep rows tying the limit-th row's sort key
+                // (TopNCache<true>, top_n_cache.rs:539-640; offset == 0)
+                if (!with_ties || !cut || sort_key_cmp(kv.first, *cut) != 0)
+                    break;
+                w.push_back(kv.first);
+            } else if (i >= offset) {
+                w.push_back(kv.first);
+                if (i + 1 == offset + limit) cut = &kv.first;
+            }
             i++;
         }
         return w;
@@ -104,9 +129,17 @@ struct GroupTopNOracle {
                 auto it = groups.find(gk);
                 if (it != groups.end()) {
                     uint64_t i = 0;
+                    const Row* cut = nullptr;
                     for (auto& kv : it->second) {
-                        if (i >= offset + limit) break;
-                        if (i >= offset) w.emplace_back(kv.first, kv.second);
+                        if (i >= offset + limit) {
+                            if (!with_ties || !cut ||
+                                sort_key_cmp(kv.first, *cut) != 0)
+                                break;
+                            w.emplace_back(kv.first, kv.second);
+                        } else if (i >= offset) {
+                            w.emplace_back(kv.first, kv.second);
+                            if (i + 1 == offset + limit) cut = &kv.first;
+                        }
                         i++;
                     }
                 }
@@ -176,6 +209,7 @@ extern "C" {
 
 void* rw_group_top_n_create(const RwGroupTopNDesc* d) {
     if (!d || !d->limit) return nullptr;
+    if (d->with_ties && d->offset) return nullptr; // reference asserts too
     return new GroupTopNOracle(d);
 }
 int rw_group_top_n_push_chunk(void* h, const RwChunk* c) {

@@ -3880,14 +3880,17 @@ struct TopMeta {
     uint8_t ck_desc[TOPN_MAX_CK];
     uint8_t ck_float[TOPN_MAX_CK];
     uint32_t offset, limit;
+    uint8_t with_ties; // TopNCache<true>: window extends past limit while
+                       // rows tie the limit-th row's order-by prefix
+    int n_order;       // cache-key prefix length that defines a tie
 };
 
 // cache-key compare on packed records (ordered_cmp semantics: NULLs
 // largest in value order, desc reverses, NaN largest among floats)
 __device__ __forceinline__ int topn_cmp(const TopMeta& m, const long long* va,
                                         uint32_t nba, const long long* vb,
-                                        uint32_t nbb) {
-    for (int i = 0; i < m.n_ck; i++) {
+                                        uint32_t nbb, int ncols) {
+    for (int i = 0; i < ncols; i++) {
         int c = m.ck_cols[i];
         bool na = !((nba >> c) & 1), nb = !((nbb >> c) & 1);
         int r;
@@ -3962,7 +3965,8 @@ __device__ int topn_select(const JoinSideDev& sd, const TopMeta& m,
             int pos = n;
             while (pos > 0) {
                 JoinRowHdr* hp = jrow(sd, sel[pos - 1]);
-                if (topn_cmp(m, jvals(hp), hp->validbits, v, h->validbits) <= 0)
+                if (topn_cmp(m, jvals(hp), hp->validbits, v, h->validbits,
+                             m.n_ck) <= 0)
                     break;
                 pos--;
             }
@@ -3978,18 +3982,45 @@ __device__ int topn_select(const JoinSideDev& sd, const TopMeta& m,
     return n;
 }
 
+// full visible window incl. WITH TIES extension (ties on the order-by
+// prefix of the limit-th row, TopNCache<true>; offset == 0 with ties).
+// Safe-overflow argument: sel is sorted by full cache key and the sort
+// prefix is a key prefix, so prefixes are non-decreasing along sel and any
+// row dropped by the bounded selection is ≥ the last kept row — once the
+// tie trim stops before the buffer end, no dropped row can tie. Only a tie
+// group consuming the ENTIRE buffer is unrepresentable → error 5.
+__device__ int topn_window(const JoinSideDev& sd, const TopMeta& m,
+                           uint32_t slot, uint32_t* sel, uint32_t* err) {
+    if (!m.with_ties)
+        return topn_select(sd, m, slot, sel, (int)(m.offset + m.limit));
+    int n = topn_select(sd, m, slot, sel, TOPN_MAX_WIN);
+    if (n <= (int)m.limit) return n;
+    JoinRowHdr* hc = jrow(sd, sel[m.limit - 1]);
+    int nw = (int)m.limit;
+    while (nw < n) {
+        JoinRowHdr* h = jrow(sd, sel[nw]);
+        if (topn_cmp(m, jvals(h), h->validbits, jvals(hc), hc->validbits,
+                     m.n_order) != 0)
+            break;
+        nw++;
+    }
+    if (nw == TOPN_MAX_WIN) atomicExch(err, 5u); // ties window overflow
+    return nw;
+}
+
 __global__ void topn_snapshot_kernel(JoinSideDev sd, TopMeta m,
                                      const uint32_t* touched_list,
                                      const uint32_t* counters,
-                                     uint32_t* old_win, uint32_t* old_n) {
-    int K = (int)(m.offset + m.limit);
+                                     uint32_t* old_win, uint32_t* old_n,
+                                     uint32_t* err) {
+    int K = m.with_ties ? TOPN_MAX_WIN : (int)(m.offset + m.limit);
     uint32_t nt = counters[0];
     uint32_t stride = gridDim.x * blockDim.x;
     for (uint32_t t = blockIdx.x * blockDim.x + threadIdx.x; t < nt;
          t += stride) {
         uint32_t sel[TOPN_MAX_WIN];
-        int n = topn_select(sd, m, touched_list[t], sel, K);
-        // visible window = positions [offset, offset+limit)
+        int n = topn_window(sd, m, touched_list[t], sel, err);
+        // visible window = positions [offset, offset+limit) (+ ties)
         int w0 = n < (int)m.offset ? n : (int)m.offset;
         old_n[t] = (uint32_t)(n - w0);
         for (int i = w0; i < n; i++)
@@ -4071,8 +4102,8 @@ __global__ void topn_emit_kernel(JoinSideDev sd, TopMeta m,
                                  const uint32_t* tcounters, uint32_t* touched,
                                  const uint32_t* old_win,
                                  const uint32_t* old_n, JoinOutDev out,
-                                 int n_cols) {
-    int K = (int)(m.offset + m.limit);
+                                 int n_cols, uint32_t* err) {
+    int K = m.with_ties ? TOPN_MAX_WIN : (int)(m.offset + m.limit);
     uint32_t nt = tcounters[0];
     uint32_t stride = gridDim.x * blockDim.x;
     for (uint32_t t = blockIdx.x * blockDim.x + threadIdx.x; t < nt;
@@ -4080,7 +4111,7 @@ __global__ void topn_emit_kernel(JoinSideDev sd, TopMeta m,
         uint32_t slot = touched_list[t];
         touched[slot] = 0;
         uint32_t selbuf[TOPN_MAX_WIN];
-        int nsel = topn_select(sd, m, slot, selbuf, K);
+        int nsel = topn_window(sd, m, slot, selbuf, err);
         // visible window = positions [offset, offset+limit)
         int w0 = nsel < (int)m.offset ? nsel : (int)m.offset;
         const uint32_t* neww = selbuf + w0;
@@ -4100,7 +4131,7 @@ __global__ void topn_emit_kernel(JoinSideDev sd, TopMeta m,
                 JoinRowHdr* ho = jrow(sd, oldw[i]);
                 JoinRowHdr* hn = jrow(sd, neww[j]);
                 c = topn_cmp(m, jvals(ho), ho->validbits, jvals(hn),
-                             hn->validbits);
+                             hn->validbits, m.n_ck);
             }
             if (c == 0) {
                 if (oldw[i] != neww[j]) {
@@ -4207,6 +4238,10 @@ struct GroupTopN {
         }
         m.offset = (uint32_t)d->offset;
         m.limit = (uint32_t)d->limit;
+        m.with_ties = d->with_ties;
+        m.n_order = (int)d->n_order_by;
+        if (d->with_ties && d->offset)
+            FAIL(RW_E_INVAL, "offset unsupported WITH TIES (as in the reference)");
         chunk_size = d->chunk_size ? d->chunk_size : 1024;
 
         uint64_t cap_hint = d->state_capacity_hint ? d->state_capacity_hint
@@ -4291,7 +4326,7 @@ struct GroupTopN {
             if (touched_list) hipFree(touched_list);
             if (old_win) hipFree(old_win);
             if (old_n) hipFree(old_n);
-            int K = (int)(m.offset + m.limit);
+            int K = m.with_ties ? TOPN_MAX_WIN : (int)(m.offset + m.limit);
             HIP_TRY(hipMalloc(&touched_list, (size_t)n * 4));
             HIP_TRY(hipMalloc(&old_win, (size_t)n * K * 4));
             HIP_TRY(hipMalloc(&old_n, (size_t)n * 4));
@@ -4363,7 +4398,7 @@ struct GroupTopN {
         topn_touch_kernel<<<blocks, 256, 0, stream>>>(b, sd, m, touched,
                                                       touched_list, tcounters);
         topn_snapshot_kernel<<<256, 256, 0, stream>>>(
-            sd, m, touched_list, tcounters, old_win, old_n);
+            sd, m, touched_list, tcounters, old_win, old_n, tcounters + 1);
         auto segs = conflict_segments(c);
         uint32_t start = 0;
         auto launch_apply = [&](uint32_t a, uint32_t z) {
@@ -4380,12 +4415,15 @@ struct GroupTopN {
         launch_apply(start, n);
         topn_emit_kernel<<<256, 256, 0, stream>>>(sd, m, touched_list,
                                                   tcounters, touched, old_win,
-                                                  old_n, out, m.n_cols);
+                                                  old_n, out, m.n_cols,
+                                                  tcounters + 1);
         HIP_TRY(hipStreamSynchronize(stream));
         uint32_t tc[2];
         HIP_TRY(hipMemcpy(tc, tcounters, 8, hipMemcpyDeviceToHost));
         if (tc[1] == 2) FAIL(RW_E_INTERNAL, "group table full");
         if (tc[1] == 3) FAIL(RW_E_INTERNAL, "row store full");
+        if (tc[1] == 5)
+            FAIL(RW_E_INTERNAL, "WITH TIES window exceeds %d rows", TOPN_MAX_WIN);
         return drain_output();
     }
 

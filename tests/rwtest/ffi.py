@@ -476,6 +476,7 @@ class RwGroupTopNDesc(C.Structure):
         ("n_rest", C.c_uint32), ("rest_cols", C.POINTER(C.c_uint32)),
         ("rest_desc", C.POINTER(C.c_uint8)),
         ("offset", C.c_uint64), ("limit", C.c_uint64),
+        ("with_ties", C.c_uint8),
         ("chunk_size", C.c_uint32),
         ("state_capacity_hint", C.c_uint64), ("row_capacity_hint", C.c_uint64),
     ]
@@ -488,8 +489,8 @@ class GroupTopN:
     storage-key columns after group_by and order_by."""
 
     def __init__(self, lib: Lib, input_types, group_by, order_by, rest,
-                 offset=0, limit=1, chunk_size=1024, state_capacity_hint=0,
-                 row_capacity_hint=0):
+                 offset=0, limit=1, with_ties=False, chunk_size=1024,
+                 state_capacity_hint=0, row_capacity_hint=0):
         self.lib = lib
         L = lib.lib
         L.rw_group_top_n_create.restype = C.c_void_p
@@ -528,6 +529,7 @@ class GroupTopN:
         d.rest_desc = u8s([1 if desc else 0 for _, desc in rest])
         d.offset = offset
         d.limit = limit
+        d.with_ties = 1 if with_ties else 0
         d.chunk_size = chunk_size
         d.state_capacity_hint = state_capacity_hint
         d.row_capacity_hint = row_capacity_hint

@@ -856,3 +856,62 @@ def test_count_distinct_parity():
         assert outs[0] == outs[1], f"epoch {ep}"
     g.close()
     o.close()
+
+
+def test_topn_with_ties_parity():
+    # WITH TIES: the two transcribed top_n_plain.rs fixtures (oracle pinned
+    # in tests/test_oracle_topn.py) replayed GPU vs oracle, plus a
+    # randomized heavy-tie mix
+    from test_oracle_topn import I3, _prepend_group
+
+    for limit, chunks in [
+        (4, [""" I I\n + 1 0\n + 2 1\n + 3 2\n + 10 3\n + 9 4\n + 8 5""",
+             """ I I\n + 7 6\n - 3 2\n - 1 0\n + 5 7\n - 2 1\n + 11 8""",
+             """ I I\n + 6 9\n + 12 10\n + 13 11\n + 14 12""",
+             """ I I\n - 5 7\n - 6 9\n - 11 8"""]),
+        (3, [""" I I\n + 1 0\n + 2 1\n + 3 2\n + 10 3\n + 9 4\n + 8 5""",
+             """ I I\n + 3 6\n + 3 7\n + 1 8\n + 2 9\n + 10 10""",
+             """ I I\n - 1 0""",
+             """ I I\n - 1 8"""]),
+    ]:
+        g = ffi.GroupTopN(gpu(), I3, [0], [(1, False)], [(2, False)],
+                          offset=0, limit=limit, with_ties=True)
+        o = ffi.GroupTopN(oracle(), I3, [0], [(1, False)], [(2, False)],
+                          offset=0, limit=limit, with_ties=True)
+        for i, pretty in enumerate(chunks):
+            c = _prepend_group(pretty)
+            g.push(c)
+            o.push(c)
+            mg = rows_multiset(g.poll_all())
+            mo = rows_multiset(o.poll_all())
+            assert mg == mo, f"limit={limit} push {i}: {mg} vs {mo}"
+        g.close()
+        o.close()
+
+    rng = np.random.default_rng(77)
+    g = ffi.GroupTopN(gpu(), I3, [0], [(1, False)], [(2, False)],
+                      offset=0, limit=3, with_ties=True)
+    o = ffi.GroupTopN(oracle(), I3, [0], [(1, False)], [(2, False)],
+                      offset=0, limit=3, with_ties=True)
+    live = []
+    for i in range(8):
+        n = 512
+        gk = rng.integers(0, 20, n)
+        ordv = rng.integers(0, 6, n)  # heavy ties
+        pk = rng.integers(0, 10**7, n)
+        ops = np.zeros(n, np.uint8)
+        for r in range(n):
+            if live and rng.random() < 0.4:
+                jx = int(rng.integers(0, len(live)))
+                gk[r], ordv[r], pk[r] = live.pop(jx)
+                ops[r] = ffi.OP_DELETE
+            else:
+                live.append((int(gk[r]), int(ordv[r]), int(pk[r])))
+        c = mk_chunk(I3, ops, [gk, ordv, pk])
+        g.push(c)
+        o.push(c)
+        mg = rows_multiset(g.poll_all())
+        mo = rows_multiset(o.poll_all())
+        assert mg == mo, f"random push {i}: {len(mg)} vs {len(mo)}"
+    g.close()
+    o.close()

@@ -219,3 +219,107 @@ def test_topn_desc_and_deep_offset():
     want = rows_multiset([from_pretty(" I I I\n - 1 20 3")])
     assert got == want, got
     t.close()
+
+
+def _prepend_group(pretty):
+    # plain-TopN fixtures → GroupTopN with a constant group column
+    lines = [l.strip() for l in pretty.strip().split("\n") if l.strip()]
+    out = ["I " + lines[0]]
+    for l in lines[1:]:
+        toks = l.split()
+        if toks[0] in ("+", "-", "U-", "U+"):
+            out.append(f"{toks[0]} 0 " + " ".join(toks[1:]))
+        else:
+            out.append("0 " + l)
+    return from_pretty("\n".join(out))
+
+
+def _ties_exec(limit):
+    # schema (g, a, b); storage key (a asc, b asc); order by (a asc)
+    return ffi.GroupTopN(oracle(), I3, [0], [(1, False)], [(2, False)],
+                         offset=0, limit=limit, with_ties=True)
+
+
+def test_with_limit_with_ties():
+    # transcribed: top_n_plain.rs test_top_n_executor_with_limit_with_ties
+    # (limit 4, WITH TIES), per-push sort_rows comparison
+    t = _ties_exec(4)
+    t.push(_prepend_group(""" I I
+        +  1 0
+        +  2 1
+        +  3 2
+        + 10 3
+        +  9 4
+        +  8 5"""))
+    expect(t, """ I I I
+        + 0 1 0
+        + 0 2 1
+        + 0 3 2
+        + 0 8 5""")
+    t.push(_prepend_group(""" I I
+        +  7 6
+        -  3 2
+        -  1 0
+        +  5 7
+        -  2 1
+        + 11 8"""))
+    expect(t, """ I I I
+        + 0 7 6
+        - 0 3 2
+        - 0 1 0
+        + 0 5 7
+        - 0 2 1
+        + 0 9 4""")
+    t.push(_prepend_group("""  I  I
+        +  6  9
+        + 12 10
+        + 13 11
+        + 14 12"""))
+    expect(t, """ I I I
+        - 0 9 4
+        + 0 6 9""")
+    t.push(_prepend_group("""  I  I
+        -  5  7
+        -  6  9
+        - 11  8"""))
+    expect(t, """ I I I
+        - 0 5 7
+        + 0 9 4
+        - 0 6 9
+        + 0 10 3""")
+    t.close()
+
+
+def test_with_ties():
+    # transcribed: top_n_plain.rs test_with_ties (limit 3, real tie groups)
+    t = _ties_exec(3)
+    t.push(_prepend_group("""  I I
+        +  1 0
+        +  2 1
+        +  3 2
+        + 10 3
+        +  9 4
+        +  8 5"""))
+    expect(t, """ I I I
+        + 0 1 0
+        + 0 2 1
+        + 0 3 2""")
+    t.push(_prepend_group("""  I I
+        +  3 6
+        +  3 7
+        +  1 8
+        +  2 9
+        + 10 10"""))
+    expect(t, """ I I I
+        - 0 3 2
+        + 0 1 8
+        + 0 2 9""")
+    t.push(_prepend_group(" I I\n - 1 0"))
+    expect(t, " I I I\n - 0 1 0")
+    t.push(_prepend_group(" I I\n - 1 8"))
+    expect(t, """ I I I
+        - 0 1 8
+        + 0 3 2
+        + 0 3 6
+        + 0 3 7""")
+    t.close()
