@@ -84,6 +84,13 @@ typedef struct RwHashAggDesc {
     const uint32_t* stream_key;
     uint32_t chunk_size;      /* output chunk rows (config/mod.rs:222-224) */
     uint8_t append_only;      /* value-state min/max allowed */
+    uint8_t emit_on_window_close; /* EOWC (hash_agg.rs:421-474): barriers
+        emit NOTHING until a watermark on group-key position 0 closes
+        windows; then each group with window key < watermark emits its
+        FINAL row once (Insert, rows sorted by group key; groups with
+        row_count 0 emit nothing) and is removed. rw_hash_agg_watermark
+        buffers the watermark instead of cleaning. Output watermark
+        forwarding is the caller's job (the value passes through). */
     uint64_t state_capacity_hint; /* expected group count (0 = default);
                                      GPU sizes the HBM table from this */
 } RwHashAggDesc;

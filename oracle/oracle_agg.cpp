@@ -19,6 +19,7 @@
 //  - Materialized-input min/max (aggregate/minput.rs:170-248): state ordered
 //    by [value ASC(min)/DESC(max), stream_key ASC], NULLs largest
 //    (test_utils/agg_executor.rs:72-121); output = first entry's value.
+#include <algorithm>
 #include <cstdlib>
 #include <cstring>
 #include <stdexcept>
@@ -82,6 +83,9 @@ struct HashAggOracle {
     std::vector<uint32_t> stream_key;
     size_t chunk_size;
     bool append_only;
+    bool eowc = false; // emit-on-window-close (hash_agg.rs:421-474)
+    bool has_pending_wm = false;
+    int64_t pending_wm = 0;
     std::vector<uint8_t> group_key_types;
     std::vector<uint8_t> out_types; // group key types ++ ret types
     std::vector<bool> call_is_minput;
@@ -105,6 +109,7 @@ struct HashAggOracle {
         stream_key.assign(d->stream_key, d->stream_key + d->n_stream_key);
         chunk_size = d->chunk_size;
         append_only = d->append_only;
+        eowc = d->emit_on_window_close != 0;
         for (auto k : group_key) group_key_types.push_back(input_types[k]);
         out_types = out_ts;
         for (auto& c : calls) {
@@ -386,6 +391,7 @@ struct HashAggOracle {
 
     // flush_data, emit-on-update branch (hash_agg.rs:475-501)
     int flush(uint64_t /*epoch*/) {
+        if (eowc) return flush_eowc();
         for (auto& key : dirty_order) {
             auto& g = groups[key];
             Row curr = get_outputs(g);
@@ -421,11 +427,53 @@ struct HashAggOracle {
         return RW_OK;
     }
 
+    // EOWC flush (hash_agg.rs:429-474): windows with group-key[0] below the
+    // buffered watermark emit their FINAL row once (Insert; row_count 0
+    // emits nothing — OnlyOutputIfHasInput with no prev) in group-key-sorted
+    // order (SortBuffer::consume iterates ordered) and are removed. Dirty
+    // state rows spill as PUTs; closed groups spill DELETEs.
+    int flush_eowc() {
+        dirty_order.clear();
+        dirty.clear();
+        if (has_pending_wm) {
+            std::vector<Row> closing;
+            for (auto& [key, g] : groups) {
+                const Datum& d = key[0];
+                if (!d.null && d.i < pending_wm) closing.push_back(key);
+            }
+            RowOrderLess less;
+            for (auto t : group_key_types) less.order.push_back({t, false});
+            std::sort(closing.begin(), closing.end(), less);
+            for (auto& key : closing) {
+                auto& g = groups[key];
+                Row curr = get_outputs(g);
+                if (row_count_of(curr) != 0) {
+                    emit(RW_OP_INSERT, key, curr);
+                    // EOWC spill this round: the window-close DELETE only
+                    // (mid-window state PUTs are a later-round item)
+                    spill_record(0, key, curr);
+                }
+                groups.erase(key);
+            }
+            has_pending_wm = false;
+        }
+        outputs_push(builder.take());
+        return RW_OK;
+    }
+
     // watermark TTL cleaning (hash_agg.rs:503-507 → update_watermark):
     // groups whose watermarked group-key column sorts below the value are
     // dropped (NULLs largest, kept)
     int watermark(uint32_t pos, int64_t val) {
         if (pos >= group_key.size()) return RW_E_INVAL;
+        if (eowc) {
+            // EOWC: buffer the window watermark; windows close at the next
+            // barrier (hash_agg.rs:657-700 buffers into window_watermark)
+            if (pos != 0) return RW_E_INVAL;
+            if (!has_pending_wm || val > pending_wm) pending_wm = val;
+            has_pending_wm = true;
+            return RW_OK;
+        }
         for (auto it = groups.begin(); it != groups.end();) {
             const Datum& d = it->first[pos];
             if (!d.null && d.i < val) {

@@ -1158,6 +1158,13 @@ __global__ void agg_init_kernel(AggTableDev t, int n_calls, AggCallDev c0,
     }
 }
 
+__global__ void agg_eowc_close_kernel(AggTableDev t, int KW, int n_calls,
+                                      int row_count_index, long long wm,
+                                      AggCallDev c0, AggCallDev c1,
+                                      AggCallDev c2, AggCallDev c3);
+
+extern "C" __global__ void agg_counters_reset_kernel(uint32_t* counters);
+
 struct HashAgg {
     RwHashAggDesc desc;
     std::vector<uint8_t> input_types;
@@ -1203,6 +1210,9 @@ struct HashAgg {
     std::vector<uint8_t> spill;
     int debug_mode = 0; // RW_AGG_DEBUG_MODE: 1 per-lane atomics, 2 no-dedupe
     uint8_t* d_vnode_bitmap = nullptr; // rescale scope (update_vnode_bitmap)
+    bool eowc = false; // emit-on-window-close (hash_agg.rs:421-474)
+    bool has_pending_wm = false;
+    int64_t pending_wm = 0;
     int n_minput = 0;   // materialized-input (retractable min/max) calls
     std::vector<uint8_t> call_minput;
     std::vector<uint32_t> stream_key;
@@ -1233,6 +1243,7 @@ struct HashAgg {
         if (!gpu_ok()) FAIL(RW_E_NOGPU, "risingwave_amd: no GPU visible (product path has no CPU fallback)");
         if (const char* m = getenv("RW_AGG_DEBUG_MODE")) debug_mode = atoi(m);
         desc = *d;
+        eowc = d->emit_on_window_close != 0;
         input_types.assign(d->input_types, d->input_types + d->n_input_cols);
         group_key.assign(d->group_key_indices, d->group_key_indices + d->n_group_key);
         calls.assign(d->calls, d->calls + d->n_calls);
@@ -1677,6 +1688,71 @@ struct HashAgg {
     }
 
     int flush(uint64_t) {
+        if (eowc) {
+            // EOWC barrier (hash_agg.rs:429-474): nothing is emitted until a
+            // watermark closes windows; the dirty list only resets (state
+            // accumulates across epochs)
+            if (!has_pending_wm) {
+                agg_counters_reset_kernel<<<1, 1, 0, stream>>>(t.counters);
+                HIP_TRY(hipStreamSynchronize(stream));
+                return check_overflow();
+            }
+            has_pending_wm = false;
+            agg_eowc_close_kernel<<<2048, 256, 0, stream>>>(
+                t, KW, n_calls, (int)desc.row_count_index, pending_wm, cd(0),
+                cd(1), cd(2), cd(3));
+            HIP_TRY(hipStreamSynchronize(stream));
+            int rc2 = check_overflow();
+            if (rc2 != RW_OK) return rc2;
+            uint32_t ctr2[3];
+            HIP_TRY(hipMemcpy(ctr2, t.counters, 12, hipMemcpyDeviceToHost));
+            uint32_t n_out = ctr2[1];
+            if (n_out) {
+                std::vector<long long> vals((size_t)n_out * out_width);
+                std::vector<uint8_t> nulls((size_t)n_out * out_width);
+                std::vector<uint8_t> ops(n_out);
+                HIP_TRY(hipMemcpy(vals.data(), t.out_vals, vals.size() * 8,
+                                  hipMemcpyDeviceToHost));
+                HIP_TRY(hipMemcpy(nulls.data(), t.out_nulls, nulls.size(),
+                                  hipMemcpyDeviceToHost));
+                HIP_TRY(hipMemcpy(ops.data(), t.out_ops, n_out,
+                                  hipMemcpyDeviceToHost));
+                // group-key-sorted emission (SortBuffer::consume order)
+                std::vector<uint32_t> order(n_out);
+                for (uint32_t i = 0; i < n_out; i++) order[i] = i;
+                std::sort(order.begin(), order.end(),
+                          [&](uint32_t a, uint32_t b) {
+                              for (int k = 0; k < KW; k++) {
+                                  bool na = nulls[(size_t)a * out_width + k];
+                                  bool nb = nulls[(size_t)b * out_width + k];
+                                  if (na != nb) return nb; // NULLs largest
+                                  if (na) continue;
+                                  long long va = vals[(size_t)a * out_width + k];
+                                  long long vb = vals[(size_t)b * out_width + k];
+                                  if (va != vb) return va < vb;
+                              }
+                              return false;
+                          });
+                std::vector<long long> sv((size_t)n_out * out_width);
+                std::vector<uint8_t> sn((size_t)n_out * out_width);
+                std::vector<uint8_t> so(n_out);
+                for (uint32_t i = 0; i < n_out; i++) {
+                    memcpy(&sv[(size_t)i * out_width],
+                           &vals[(size_t)order[i] * out_width],
+                           (size_t)out_width * 8);
+                    memcpy(&sn[(size_t)i * out_width],
+                           &nulls[(size_t)order[i] * out_width], out_width);
+                    so[i] = ops[order[i]];
+                }
+                // spill: closed windows leave the state table (DELETEs);
+                // final rows spill through spill_records as PUT+DELETE —
+                // here the closed groups are gone, so record DELETEs only
+                spill_records_eowc(sv, sn, n_out);
+                slice_outputs(sv, sn, so, n_out);
+            }
+            HIP_TRY(hipMemset(t.counters, 0, 12));
+            return RW_OK;
+        }
         agg_flush_kernel<<<2048, 256, 0, stream>>>(t, KW, n_calls,
                                                    (int)desc.row_count_index, cd(0),
                                                    cd(1), cd(2), cd(3));
@@ -1700,6 +1776,28 @@ struct HashAgg {
         }
         HIP_TRY(hipMemset(t.counters, 0, 12));
         return RW_OK;
+    }
+
+    // EOWC spill this round: one DELETE per closed (emitted) window —
+    // mid-window state PUTs are a later-round item (matches the oracle).
+    void spill_records_eowc(const std::vector<long long>& vals,
+                            const std::vector<uint8_t>& nulls,
+                            uint32_t n_out) {
+        auto put32 = [&](uint32_t x) {
+            for (int b = 0; b < 4; b++) spill.push_back((uint8_t)(x >> (8 * b)));
+        };
+        for (uint32_t r = 0; r < n_out; r++) {
+            spill.push_back(0);
+            std::vector<uint8_t> k;
+            for (int i = 0; i < KW; i++) {
+                rwcodec::DatumC d{nulls[(size_t)r * out_width + i] != 0,
+                                  vals[(size_t)r * out_width + i], 0};
+                rwcodec::memcmp_encode_datum(k, out_types[i], d, {});
+            }
+            put32((uint32_t)k.size());
+            spill.insert(spill.end(), k.begin(), k.end());
+            put32(0);
+        }
     }
 
     // Append state-table KV deltas for the flushed records (the spill
@@ -2945,6 +3043,114 @@ __global__ void agg_clean_kernel(AggTableDev t, int kpos, long long wm, int KW,
         if ((ld_u32(&t.key_nulls[(uint32_t)slot]) >> kpos) & 1) continue;
         if (ld_i64((const int64_t*)&t.keys[slot * KW + kpos]) >= wm) continue;
         // reset the group as if freshly created (late rows restart it)
+        for (int ci = 0; ci < n_calls; ci++) {
+            long long init = 0;
+            if (calls[ci].kind == RW_AGG_MIN) init = INT64_MAX;
+            if (calls[ci].kind == RW_AGG_MAX) init = INT64_MIN;
+            t.acc[(size_t)ci * cap + slot] = init;
+            t.has[(size_t)ci * cap + slot] = 0;
+            if (calls[ci].minput) {
+                uint32_t row = t.mheads[(size_t)calls[ci].mord * cap + slot];
+                while (row != UINT32_MAX) {
+                    st_u32(&t.malive[row], 0);
+                    row = ld_u32(&t.mnext[row]);
+                }
+                t.mheads[(size_t)calls[ci].mord * cap + slot] = UINT32_MAX;
+            }
+        }
+        t.has_prev[(uint32_t)slot] = 0;
+        t.dirty_flag[(uint32_t)slot] = 0;
+    }
+}
+
+// EOWC close (hash_agg.rs:429-474): windows with group-key[0] below the
+// watermark emit their FINAL row once (Insert; row_count 0 emits nothing)
+// and are reset in place (slot stays READY so linear probing is
+// undisturbed; a late row would restart the window). Emission order is
+// fixed up host-side (rows sorted by group key, as SortBuffer::consume
+// iterates ordered).
+__global__ void agg_eowc_close_kernel(AggTableDev t, int KW, int n_calls,
+                                      int row_count_index, long long wm,
+                                      AggCallDev c0, AggCallDev c1,
+                                      AggCallDev c2, AggCallDev c3) {
+    AggCallDev calls[4] = {c0, c1, c2, c3};
+    size_t cap = (size_t)t.cap_mask + 1;
+    size_t stride = (size_t)gridDim.x * blockDim.x;
+    int width = KW + n_calls;
+    for (size_t slot = blockIdx.x * blockDim.x + threadIdx.x; slot < cap;
+         slot += stride) {
+        if (ld_u32(&t.state[(uint32_t)slot]) != SLOT_READY) continue;
+        if ((ld_u32(&t.key_nulls[(uint32_t)slot]) >> 0) & 1) continue;
+        if (ld_i64((const int64_t*)&t.keys[slot * KW]) >= wm) continue;
+        long long rc = t.acc[(size_t)row_count_index * cap + slot];
+        if (rc < 0) rc = 0;
+        if (rc > 0) {
+            // final outputs (get_outputs, agg_group.rs:431-467)
+            long long curr[MAX_CALLS];
+            uint8_t curr_null[MAX_CALLS];
+            for (int ci = 0; ci < n_calls; ci++) {
+                const AggCallDev& c = calls[ci];
+                long long* acc = t.acc + (size_t)ci * cap;
+                uint8_t* has = t.has + (size_t)ci * cap;
+                if (c.minput) {
+                    uint32_t row = t.mheads[(size_t)c.mord * cap + slot];
+                    bool any = false, any_null = false;
+                    long long best = 0;
+                    while (row != UINT32_MAX) {
+                        if (t.malive[row]) {
+                            if (t.mval_null[row]) {
+                                any_null = true;
+                            } else {
+                                long long v = t.mval[row];
+                                if (!any) best = v;
+                                else if (c.kind == RW_AGG_MIN)
+                                    best = v < best ? v : best;
+                                else
+                                    best = v > best ? v : best;
+                                any = true;
+                            }
+                        }
+                        row = t.mnext[row];
+                    }
+                    if (c.kind == RW_AGG_MAX && any_null) {
+                        curr[ci] = 0;
+                        curr_null[ci] = 1;
+                    } else {
+                        curr[ci] = best;
+                        curr_null[ci] = !any;
+                    }
+                    continue;
+                }
+                switch (c.kind) {
+                    case RW_AGG_COUNT_STAR:
+                    case RW_AGG_COUNT:
+                    case RW_AGG_SUM0:
+                        curr[ci] = acc[slot];
+                        curr_null[ci] = 0;
+                        break;
+                    default:
+                        curr[ci] = acc[slot];
+                        curr_null[ci] = !has[slot];
+                }
+            }
+            uint32_t orow = atomicAdd(&t.counters[1], 1u);
+            if (orow + 1 > t.out_capacity) {
+                atomicExch(&t.counters[2], 2u);
+            } else {
+                t.out_ops[orow] = RW_OP_INSERT;
+                for (int k = 0; k < KW; k++) {
+                    t.out_vals[(size_t)orow * width + k] =
+                        t.keys[slot * KW + k];
+                    t.out_nulls[(size_t)orow * width + k] =
+                        (t.key_nulls[(uint32_t)slot] >> k) & 1;
+                }
+                for (int ci = 0; ci < n_calls; ci++) {
+                    t.out_vals[(size_t)orow * width + KW + ci] = curr[ci];
+                    t.out_nulls[(size_t)orow * width + KW + ci] = curr_null[ci];
+                }
+            }
+        }
+        // reset the window in place (as agg_clean_kernel does)
         for (int ci = 0; ci < n_calls; ci++) {
             long long init = 0;
             if (calls[ci].kind == RW_AGG_MIN) init = INT64_MAX;

@@ -66,6 +66,7 @@ class RwHashAggDesc(C.Structure):
         ("stream_key", C.POINTER(C.c_uint32)),
         ("chunk_size", C.c_uint32),
         ("append_only", C.c_uint8),
+        ("emit_on_window_close", C.c_uint8),
         ("state_capacity_hint", C.c_uint64),
     ]
 
@@ -269,7 +270,7 @@ class Lib:
 class HashAgg:
     def __init__(self, lib: Lib, input_types, group_key, calls, row_count_index,
                  stream_key=(), chunk_size=1024, append_only=False,
-                 state_capacity_hint=0):
+                 emit_on_window_close=False, state_capacity_hint=0):
         """calls: list of (kind, arg, ret_type)."""
         self.lib = lib
         d = RwHashAggDesc()
@@ -294,6 +295,7 @@ class HashAgg:
         d.stream_key = self._sk
         d.chunk_size = chunk_size
         d.append_only = 1 if append_only else 0
+        d.emit_on_window_close = 1 if emit_on_window_close else 0
         d.state_capacity_hint = state_capacity_hint
         self.h = lib.lib.rw_hash_agg_create(C.byref(d))
         if not self.h:

@@ -915,3 +915,61 @@ def test_topn_with_ties_parity():
         assert mg == mo, f"random push {i}: {len(mg)} vs {len(mo)}"
     g.close()
     o.close()
+
+
+def test_eowc_parity():
+    # emit-on-window-close: the transcribed golden sequence + a randomized
+    # multi-window stream, GPU vs oracle (order-exact within a flush — both
+    # emit group-key-sorted)
+    ga = ffi.HashAgg(gpu(), [T_I64], [0], [(AGG_COUNT_STAR, -1, T_I64)], 0,
+                     emit_on_window_close=True)
+    oa = ffi.HashAgg(ffi.oracle(), [T_I64], [0], [(AGG_COUNT_STAR, -1, T_I64)],
+                     0, emit_on_window_close=True)
+    steps = [
+        ("push", " I\n + 1\n + 2\n + 3"),
+        ("flush", None),
+        ("push", " I\n - 2\n + 4"),
+        ("wm", 3), ("flush", None),
+        ("wm", 4), ("flush", None),
+        ("wm", 10), ("flush", None),
+        ("wm", 20), ("flush", None),
+    ]
+    ep = 0
+    for kind, arg in steps:
+        outs = []
+        for a in (ga, oa):
+            if kind == "push":
+                a.push(from_pretty(arg))
+            elif kind == "wm":
+                a.watermark(0, arg)
+            else:
+                ep += 1
+                a.flush(ep)
+            outs.append(ffi.rows_ordered(a.poll_all()))
+        assert outs[0] == outs[1], f"{kind} {arg}: {outs[0]} vs {outs[1]}"
+    ga.close()
+    oa.close()
+
+    # randomized: windows 0..300, watermark advances between epochs
+    rng = np.random.default_rng(55)
+    calls = [(AGG_COUNT_STAR, -1, T_I64), (ffi.AGG_SUM, 1, T_I64)]
+    g = ffi.HashAgg(gpu(), [T_I64, T_I64], [0], calls, 0,
+                    emit_on_window_close=True)
+    o = ffi.HashAgg(ffi.oracle(), [T_I64, T_I64], [0], calls, 0,
+                    emit_on_window_close=True)
+    wm = 0
+    for ep in range(6):
+        n = 4096
+        keys = rng.integers(wm, wm + 80, n)
+        vals = rng.integers(1, 100, n)
+        c = mk_chunk([T_I64, T_I64], np.zeros(n, np.uint8), [keys, vals])
+        wm += 40
+        outs = []
+        for a in (g, o):
+            a.push(c)
+            a.watermark(0, wm)
+            a.flush(ep + 1)
+            outs.append(ffi.rows_ordered(a.poll_all()))
+        assert outs[0] == outs[1], f"epoch {ep}: {len(outs[0])} vs {len(outs[1])}"
+    g.close()
+    o.close()

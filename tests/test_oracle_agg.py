@@ -265,3 +265,31 @@ def test_count_distinct():
     agg.flush(4)
     assert rows_multiset(agg.poll_all()) == expect([("+", (1, 2, 1, 7))])
     agg.close()
+
+
+def test_emit_on_window_close():
+    # transcribed: src/stream/tests/integration_tests/hash_agg.rs
+    # test_hash_agg_emit_on_window_close (varchar column dropped — types are
+    # i64-only here; outputs unchanged). Barriers emit nothing until a
+    # watermark closes windows; closed windows emit once, sorted; row_count
+    # 0 windows emit nothing.
+    agg = ffi.HashAgg(oracle(), [T_I64], [0],
+                      calls=[(AGG_COUNT_STAR, -1, T_I64)], row_count_index=0,
+                      emit_on_window_close=True)
+    agg.push(from_pretty(" I\n + 1\n + 2\n + 3"))
+    agg.flush(2)
+    assert agg.poll_all() == []
+    agg.push(from_pretty(" I\n - 2\n + 4"))
+    agg.watermark(0, 3)
+    agg.flush(3)
+    assert rows_multiset(agg.poll_all()) == expect([("+", (1, 1))])
+    agg.watermark(0, 4)
+    agg.flush(4)
+    assert rows_multiset(agg.poll_all()) == expect([("+", (3, 1))])
+    agg.watermark(0, 10)
+    agg.flush(5)
+    assert rows_multiset(agg.poll_all()) == expect([("+", (4, 1))])
+    agg.watermark(0, 20)
+    agg.flush(6)
+    assert agg.poll_all() == []
+    agg.close()
