@@ -3837,6 +3837,15 @@ int rw_hash_join_watermark(void* h, int side, uint32_t col_idx, int64_t val,
 int rw_hash_agg_watermark(void* h, uint32_t group_key_pos, int64_t val) {
     auto* agg = (HashAgg*)h;
     if (group_key_pos >= (uint32_t)agg->KW) FAIL(RW_E_INVAL, "bad group key pos");
+    if (agg->eowc) {
+        // EOWC: buffer the window watermark; windows close at the next
+        // barrier (hash_agg.rs:657-700)
+        if (group_key_pos != 0) FAIL(RW_E_INVAL, "EOWC watermark must be on group key 0");
+        if (!agg->has_pending_wm || val > agg->pending_wm)
+            agg->pending_wm = val;
+        agg->has_pending_wm = true;
+        return RW_OK;
+    }
     agg_clean_kernel<<<2048, 256, 0, agg->stream>>>(
         agg->t, (int)group_key_pos, val, agg->KW, agg->n_calls, agg->cd(0),
         agg->cd(1), agg->cd(2), agg->cd(3));
