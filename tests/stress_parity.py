@@ -140,9 +140,98 @@ def main():
         for jt in range(8):
             stress_join(seed, jt, max(n // 2, 512))
         stress_topn(seed, max(n // 4, 256))
-        print(f"seed {seed}: agg + 8 join types + topn OK")
+        stress_eowc(seed, max(n // 2, 512))
+        stress_distinct(seed, max(n // 2, 512))
+        stress_topn_ties(seed, max(n // 4, 256))
+        print(f"seed {seed}: agg + 8 joins + topn + eowc + distinct + ties OK")
     print(f"STRESS OK: {n_seeds} seeds")
 
 
 if __name__ == "__main__":
     main()
+
+
+def stress_eowc(seed, n):
+    rng = np.random.default_rng(seed + 99)
+    calls = [(AGG_COUNT_STAR, -1, T_I64), (AGG_SUM, 1, T_I64)]
+    g = ffi.HashAgg(GPU, [T_I64, T_I64], [0], calls, 0,
+                    emit_on_window_close=True)
+    o = ffi.HashAgg(oracle(), [T_I64, T_I64], [0], calls, 0,
+                    emit_on_window_close=True)
+    wm = 0
+    for ep in range(5):
+        keys = rng.integers(wm, wm + 60, n)
+        vals = rng.integers(1, 50, n)
+        c = mk([T_I64, T_I64], np.zeros(n, np.uint8), [keys, vals])
+        wm += 30
+        outs = []
+        for a in (g, o):
+            a.push(c)
+            a.watermark(0, wm)
+            a.flush(ep + 1)
+            outs.append(ffi.rows_ordered(a.poll_all()))
+        assert outs[0] == outs[1], f"eowc seed {seed} epoch {ep}"
+    g.close()
+    o.close()
+
+
+def stress_distinct(seed, n):
+    from rwtest.ffi import AGG_COUNT
+
+    rng = np.random.default_rng(seed + 13)
+    calls = [(AGG_COUNT_STAR, -1, T_I64), (AGG_COUNT, 1, T_I64, 1),
+             (AGG_SUM, 1, T_I64, 1)]
+    g = ffi.HashAgg(GPU, [T_I64, T_I64], [0], calls, 0)
+    o = ffi.HashAgg(oracle(), [T_I64, T_I64], [0], calls, 0)
+    live = []
+    for ep in range(5):
+        keys = rng.integers(0, 40, n)
+        vals = rng.integers(0, 6, n)
+        ops = np.zeros(n, np.uint8)
+        for r in range(n):
+            if live and rng.random() < 0.4:
+                jx = int(rng.integers(0, len(live)))
+                keys[r], vals[r] = live.pop(jx)
+                ops[r] = ffi.OP_DELETE
+            else:
+                live.append((int(keys[r]), int(vals[r])))
+        c = mk([T_I64, T_I64], ops, [keys, vals])
+        outs = []
+        for a in (g, o):
+            a.push(c)
+            a.flush(ep + 1)
+            outs.append(rows_multiset(a.poll_all()))
+        assert outs[0] == outs[1], f"distinct seed {seed} epoch {ep}"
+    g.close()
+    o.close()
+
+
+def stress_topn_ties(seed, n):
+    rng = np.random.default_rng(seed + 31)
+    t3 = [T_I64, T_I64, T_I64]
+    lim = int(rng.integers(1, 5))
+    g = ffi.GroupTopN(GPU, t3, [0], [(1, False)], [(2, False)],
+                      offset=0, limit=lim, with_ties=True)
+    o = ffi.GroupTopN(oracle(), t3, [0], [(1, False)], [(2, False)],
+                      offset=0, limit=lim, with_ties=True)
+    live = []
+    for i in range(5):
+        gk = rng.integers(0, 20, n)
+        ordv = rng.integers(0, 8, n)
+        pkv = rng.integers(0, 10**7, n)
+        ops = np.zeros(n, np.uint8)
+        for r in range(n):
+            if live and rng.random() < 0.35:
+                jx = int(rng.integers(0, len(live)))
+                gk[r], ordv[r], pkv[r] = live.pop(jx)
+                ops[r] = ffi.OP_DELETE
+            else:
+                live.append((int(gk[r]), int(ordv[r]), int(pkv[r])))
+        c = mk(t3, ops, [gk, ordv, pkv])
+        g.push(c)
+        o.push(c)
+        mg = rows_multiset(g.poll_all())
+        mo = rows_multiset(o.poll_all())
+        assert mg == mo, f"ties seed {seed} push {i} (lim={lim})"
+    g.close()
+    o.close()
