@@ -132,23 +132,6 @@ def stress_topn(seed, n):
     o.close()
 
 
-def main():
-    n_seeds = int(sys.argv[1]) if len(sys.argv) > 1 else 5
-    n = int(sys.argv[2]) if len(sys.argv) > 2 else 2048
-    for seed in range(1, n_seeds + 1):
-        stress_agg(seed, n)
-        for jt in range(8):
-            stress_join(seed, jt, max(n // 2, 512))
-        stress_topn(seed, max(n // 4, 256))
-        stress_eowc(seed, max(n // 2, 512))
-        stress_distinct(seed, max(n // 2, 512))
-        stress_topn_ties(seed, max(n // 4, 256))
-        print(f"seed {seed}: agg + 8 joins + topn + eowc + distinct + ties OK")
-    print(f"STRESS OK: {n_seeds} seeds")
-
-
-if __name__ == "__main__":
-    main()
 
 
 def stress_eowc(seed, n):
@@ -235,3 +218,22 @@ def stress_topn_ties(seed, n):
         assert mg == mo, f"ties seed {seed} push {i} (lim={lim})"
     g.close()
     o.close()
+
+
+def main():
+    n_seeds = int(sys.argv[1]) if len(sys.argv) > 1 else 5
+    n = int(sys.argv[2]) if len(sys.argv) > 2 else 2048
+    for seed in range(1, n_seeds + 1):
+        stress_agg(seed, n)
+        for jt in range(8):
+            stress_join(seed, jt, max(n // 2, 512))
+        stress_topn(seed, max(n // 4, 256))
+        stress_eowc(seed, max(n // 2, 512))
+        stress_distinct(seed, max(n // 2, 512))
+        stress_topn_ties(seed, max(n // 4, 256))
+        print(f"seed {seed}: agg + 8 joins + topn + eowc + distinct + ties OK")
+    print(f"STRESS OK: {n_seeds} seeds")
+
+
+if __name__ == "__main__":
+    main()
