@@ -402,3 +402,28 @@ def test_right_outer_join_append_only():
         ("+", (None, None, None, 7, 7, 6)),
     ]
     j.close()
+
+
+def test_null_safe_left_semi_join():
+    # hash_join.rs test_streaming_null_safe_hash_left_semi_join
+    j = classical(JOIN_LEFT_SEMI, null_safe=True)
+    assert push(j, SIDE_LEFT, " I I\n + 1 4\n + 2 5\n + . 6") == []
+    assert push(j, SIDE_LEFT, " I I\n + . 8\n - . 8") == []
+    assert push(j, SIDE_RIGHT, " I I\n + 2 7\n + 4 8\n + 6 9") == rows([("+", 2, 5)])
+    assert push(j, SIDE_RIGHT, " I I\n + . 10\n + 6 11") == rows([("+", None, 6)])
+    assert push(j, SIDE_LEFT, " I I\n + 6 10") == rows([("+", 6, 10)])
+    assert push(j, SIDE_RIGHT, " I I\n - 6 11") == []
+    assert push(j, SIDE_RIGHT, " I I\n - 6 9") == rows([("-", 6, 10)])
+    j.close()
+
+
+def test_right_semi_join_append_only():
+    # hash_join.rs test_streaming_hash_right_semi_join_append_only
+    j = append_only(JOIN_RIGHT_SEMI)
+    assert push(j, SIDE_LEFT, " I I I\n + 1 4 1\n + 2 5 2\n + 3 6 3") == []
+    assert push(j, SIDE_LEFT, " I I I\n + 4 9 4\n + 5 10 5") == []
+    assert push(j, SIDE_RIGHT, " I I I\n + 2 5 1\n + 4 9 2\n + 6 9 3") == rows(
+        [("+", 2, 5, 1), ("+", 4, 9, 2)])
+    assert push(j, SIDE_RIGHT, " I I I\n + 1 4 4\n + 3 6 5") == rows(
+        [("+", 1, 4, 4), ("+", 3, 6, 5)])
+    j.close()
