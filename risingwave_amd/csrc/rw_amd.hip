@@ -4424,8 +4424,10 @@ struct GroupTopN {
         if (!d->limit) FAIL(RW_E_INVAL, "limit must be > 0");
         if (d->offset + d->limit > TOPN_MAX_WIN)
             FAIL(RW_E_INVAL, "offset+limit > %d unsupported", TOPN_MAX_WIN);
-        if (d->n_group_by < 1 || d->n_group_by > MAX_KW)
-            FAIL(RW_E_INVAL, "group_by width 1..%d", MAX_KW);
+        if (d->n_group_by > MAX_KW)
+            FAIL(RW_E_INVAL, "group_by width 0..%d", MAX_KW);
+        // n_group_by == 0 = the plain TopN executor (top_n_plain.rs): every
+        // row lands in the single empty-key group
         if (d->n_order_by + d->n_rest > TOPN_MAX_CK)
             FAIL(RW_E_INVAL, "cache key width > %d", TOPN_MAX_CK);
         if (d->n_cols > MAX_COLS) FAIL(RW_E_INVAL, "n_cols > %d", MAX_COLS);

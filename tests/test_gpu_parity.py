@@ -973,3 +973,34 @@ def test_eowc_parity():
         assert outs[0] == outs[1], f"epoch {ep}: {len(outs[0])} vs {len(outs[1])}"
     g.close()
     o.close()
+
+
+def test_plain_topn_parity():
+    # plain TopN (zero group columns): randomized stream, GPU vs oracle
+    t2 = [T_I64, T_I64]
+    rng = np.random.default_rng(42)
+    g = ffi.GroupTopN(gpu(), t2, [], [(0, False)], [(1, False)],
+                      offset=0, limit=5)
+    o = ffi.GroupTopN(oracle(), t2, [], [(0, False)], [(1, False)],
+                      offset=0, limit=5)
+    live = []
+    for i in range(8):
+        n = 512
+        ordv = rng.integers(0, 500, n)
+        pk = rng.integers(0, 10**7, n)
+        ops = np.zeros(n, np.uint8)
+        for r in range(n):
+            if live and rng.random() < 0.35:
+                jx = int(rng.integers(0, len(live)))
+                ordv[r], pk[r] = live.pop(jx)
+                ops[r] = ffi.OP_DELETE
+            else:
+                live.append((int(ordv[r]), int(pk[r])))
+        c = mk_chunk(t2, ops, [ordv, pk])
+        g.push(c)
+        o.push(c)
+        mg = rows_multiset(g.poll_all())
+        mo = rows_multiset(o.poll_all())
+        assert mg == mo, f"push {i}: {mg} vs {mo}"
+    g.close()
+    o.close()

@@ -323,3 +323,33 @@ def test_with_ties():
         + 0 3 6
         + 0 3 7""")
     t.close()
+
+
+def test_plain_topn_no_group():
+    # plain TopN = GroupTopN with zero group columns (top_n_plain.rs): the
+    # limit-4 WITH TIES fixture replayed on the bare 2-col schema
+    t2 = [T_I64, T_I64]
+    t = ffi.GroupTopN(oracle(), t2, [], [(0, False)], [(1, False)],
+                      offset=0, limit=4, with_ties=True)
+    t.push(from_pretty(""" I I
+        +  1 0
+        +  2 1
+        +  3 2
+        + 10 3
+        +  9 4
+        +  8 5"""))
+    got = rows_multiset(t.poll_all())
+    want = rows_multiset([from_pretty(" I I\n + 1 0\n + 2 1\n + 3 2\n + 8 5")])
+    assert got == want, got
+    t.push(from_pretty(""" I I
+        +  7 6
+        -  3 2
+        -  1 0
+        +  5 7
+        -  2 1
+        + 11 8"""))
+    got = rows_multiset(t.poll_all())
+    want = rows_multiset([from_pretty(
+        " I I\n + 7 6\n - 3 2\n - 1 0\n + 5 7\n - 2 1\n + 9 4")])
+    assert got == want, got
+    t.close()
