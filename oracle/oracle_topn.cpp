@@ -24,6 +24,7 @@
 // construction (the reference's caches are exact once synced).
 #include <cstring>
 #include <map>
+#include <optional>
 #include <memory>
 #include <string>
 #include <vector>
@@ -31,6 +32,7 @@
 #include "../include/rw_chunk.h"
 #include "../include/rw_stream.h"
 #include "common.hpp"
+#include "../include/rw_codec.hpp"
 
 namespace orc {
 
@@ -83,6 +85,66 @@ struct GroupTopNOracle {
             if (c) return c;
         }
         return 0;
+    }
+
+    // §8f-2 checkpoint spill deltas (same tri-state netting and record
+    // framing as the join oracle): key = memcmp storage key (group cols
+    // ASC then cache-key cols with their orders), value = full row.
+    struct DeltaEnt {
+        int st; // 0 = DEL, 1 = PUT (fresh), 2 = PUT (over pre-epoch row)
+        std::vector<uint8_t> v;
+    };
+    std::map<std::string, DeltaEnt> delta;
+
+    std::string enc_key(const Row& gk, const Row& ck) {
+        std::vector<uint8_t> kb;
+        for (size_t i = 0; i < gk.size(); i++) {
+            rwcodec::DatumC d{gk[i].null, gk[i].i, gk[i].d};
+            rwcodec::memcmp_encode_datum(kb, types[group_by[i]], d, {});
+        }
+        for (size_t i = 0; i < ck.size(); i++) {
+            rwcodec::DatumC d{ck[i].null, ck[i].i, ck[i].d};
+            rwcodec::OrderType ot;
+            ot.desc = ck_less.order[i].desc;
+            rwcodec::memcmp_encode_datum(kb, types[ck_cols[i]], d, ot);
+        }
+        return std::string((const char*)kb.data(), kb.size());
+    }
+
+    void delta_insert(const Row& gk, const Row& ck, const Row& row) {
+        std::vector<uint8_t> v;
+        for (size_t c = 0; c < row.size(); c++) {
+            rwcodec::DatumC d{row[c].null, row[c].i, row[c].d};
+            rwcodec::value_encode_datum(v, types[c], d);
+        }
+        std::string k = enc_key(gk, ck);
+        auto it = delta.find(k);
+        if (it != delta.end() && it->second.st == 0)
+            it->second = {2, std::move(v)};
+        else
+            delta[k] = {1, std::move(v)};
+    }
+
+    void delta_delete(const Row& gk, const Row& ck) {
+        std::string k = enc_key(gk, ck);
+        auto it = delta.find(k);
+        if (it == delta.end()) delta[k] = {0, {}};
+        else if (it->second.st == 1) delta.erase(it);
+        else it->second = {0, {}};
+    }
+
+    void checkpoint_drain(std::vector<uint8_t>& sp) {
+        auto put32 = [&](uint32_t x) {
+            for (int b = 0; b < 4; b++) sp.push_back((uint8_t)(x >> (8 * b)));
+        };
+        for (auto& [k, e] : delta) {
+            sp.push_back(e.st ? 1 : 0);
+            put32((uint32_t)k.size());
+            sp.insert(sp.end(), k.begin(), k.end());
+            put32(e.st ? (uint32_t)e.v.size() : 0);
+            if (e.st) sp.insert(sp.end(), e.v.begin(), e.v.end());
+        }
+        delta.clear();
     }
 
     Row project(const Row& row, const std::vector<uint32_t>& idx) const {
@@ -149,10 +211,17 @@ struct GroupTopNOracle {
             uint8_t op = cv.op(r);
             auto& g = groups.try_emplace(gk, Group(ck_less)).first->second;
             if (op == RW_OP_INSERT || op == RW_OP_UPDATE_INSERT) {
+                // upsert over an existing row = delete + insert for the
+                // delta (the pre-epoch row leaves the store)
+                if (g.count(ck)) delta_delete(gk, ck);
                 g[ck] = row;
+                delta_insert(gk, ck, row);
             } else {
                 auto it = g.find(ck);
-                if (it != g.end()) g.erase(it);
+                if (it != g.end()) {
+                    g.erase(it);
+                    delta_delete(gk, ck);
+                }
             }
         }
         // per touched group: merge-diff old vs new window with ChangeBuffer
@@ -221,6 +290,14 @@ int rw_group_top_n_flush(void* h, uint64_t epoch) {
     return RW_OK; // emission is per push; state commit is a no-op here
 }
 RwChunk* rw_group_top_n_poll(void* h) { return ((GroupTopNOracle*)h)->poll(); }
+int rw_topn_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len) {
+    std::vector<uint8_t> sp;
+    ((GroupTopNOracle*)h)->checkpoint_drain(sp);
+    *len = sp.size();
+    *buf = (uint8_t*)malloc(sp.size() ? sp.size() : 1);
+    memcpy(*buf, sp.data(), sp.size());
+    return RW_OK;
+}
 void rw_group_top_n_destroy(void* h) { delete (GroupTopNOracle*)h; }
 
 } // extern "C"

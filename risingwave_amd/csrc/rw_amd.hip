@@ -4273,6 +4273,10 @@ __global__ void topn_apply_kernel(JoinBatchDev b, JoinSideDev sd, TopMeta m,
             if (ld_u32(&h->alive) &&
                 topn_ck_eq_batch(m, b, r, jvals(h), vb)) {
                 st_u32(&h->alive, 0); // upsert kills the old record
+                if (sd.killed) {
+                    uint32_t kidx = atomicAdd(sd.killed_cursor, 1u);
+                    if (kidx < sd.killed_cap) sd.killed[kidx] = row;
+                }
                 break;
             }
             row = ld_u32(&h->next);
@@ -4475,6 +4479,10 @@ struct GroupTopN {
         HIP_TRY(hipMalloc(&sd.rows, (size_t)row_cap * sd.row_stride));
         HIP_TRY(hipMalloc(&sd.row_cursor, 4));
         HIP_TRY(hipMemset(sd.row_cursor, 0, 4));
+        sd.killed_cap = 1u << 22; // checkpoint-delta tracking (§8f-2)
+        HIP_TRY(hipMalloc(&sd.killed, (size_t)sd.killed_cap * 4));
+        HIP_TRY(hipMalloc(&sd.killed_cursor, 4));
+        HIP_TRY(hipMemset(sd.killed_cursor, 0, 4));
         HIP_TRY(hipMalloc(&touched, cap * 4));
         HIP_TRY(hipMemset(touched, 0, cap * 4));
         HIP_TRY(hipMalloc(&tcounters, 8));
@@ -4495,6 +4503,8 @@ struct GroupTopN {
             hipFree(sd.slots);
             hipFree(sd.rows);
             hipFree(sd.row_cursor);
+            hipFree(sd.killed);
+            hipFree(sd.killed_cursor);
             hipFree(touched);
             hipFree(tcounters);
             hipFree(out.vals);
@@ -4696,6 +4706,91 @@ struct GroupTopN {
         return RW_OK;
     }
 
+    // §8f-2 checkpoint spill: the TopN state table's per-epoch KV deltas —
+    // key = memcomparable storage key (group cols ASC, then the cache-key
+    // cols with their declared orders), value = value-encoded full row;
+    // same kill-list netting and record framing as the join drain.
+    uint32_t flush_mark = 0;
+    std::vector<uint8_t> ck_desc_host;
+    int checkpoint_drain(std::vector<uint8_t>& sp) {
+        HIP_TRY(hipStreamSynchronize(stream));
+        uint32_t cur = 0, kcur = 0;
+        HIP_TRY(hipMemcpy(&cur, sd.row_cursor, 4, hipMemcpyDeviceToHost));
+        HIP_TRY(hipMemcpy(&kcur, sd.killed_cursor, 4, hipMemcpyDeviceToHost));
+        if (kcur > sd.killed_cap)
+            FAIL(RW_E_INTERNAL, "topn kill list overflow (lost deltas)");
+        if (cur > sd.row_cap) cur = sd.row_cap;
+        size_t stride = sd.row_stride;
+        std::vector<uint8_t> fresh((size_t)(cur - flush_mark) * stride);
+        if (cur > flush_mark)
+            HIP_TRY(hipMemcpy(fresh.data(),
+                              sd.rows + (size_t)flush_mark * stride,
+                              fresh.size(), hipMemcpyDeviceToHost));
+        std::vector<uint32_t> kills(kcur);
+        if (kcur)
+            HIP_TRY(hipMemcpy(kills.data(), sd.killed, (size_t)kcur * 4,
+                              hipMemcpyDeviceToHost));
+        auto encode_key = [&](const uint8_t* rec, std::string& k) {
+            const uint32_t vb = ((const uint32_t*)rec)[2];
+            const int64_t* vals = (const int64_t*)(rec + 16);
+            std::vector<uint8_t> kb;
+            for (int i = 0; i < m.KW; i++) {
+                uint8_t col = m.gk_cols[i];
+                rwcodec::DatumC d{!((vb >> col) & 1), vals[col], 0};
+                rwcodec::memcmp_encode_datum(kb, types[col], d, {});
+            }
+            for (int i = 0; i < m.n_ck; i++) {
+                uint8_t col = m.ck_cols[i];
+                rwcodec::DatumC d{!((vb >> col) & 1), vals[col], 0};
+                rwcodec::OrderType ot;
+                ot.desc = m.ck_desc[i] != 0;
+                rwcodec::memcmp_encode_datum(kb, types[col], d, ot);
+            }
+            k.assign((const char*)kb.data(), kb.size());
+        };
+        auto encode_val = [&](const uint8_t* rec, std::vector<uint8_t>& v) {
+            const uint32_t vb = ((const uint32_t*)rec)[2];
+            const int64_t* vals = (const int64_t*)(rec + 16);
+            for (int c = 0; c < m.n_cols; c++) {
+                rwcodec::DatumC d{!((vb >> c) & 1), vals[c], 0};
+                rwcodec::value_encode_datum(v, types[c], d);
+            }
+        };
+        std::map<std::string, std::optional<std::vector<uint8_t>>> delta;
+        std::vector<uint8_t> oldrec(stride);
+        for (uint32_t i = 0; i < kcur; i++) {
+            if (kills[i] >= flush_mark) continue;
+            HIP_TRY(hipMemcpy(oldrec.data(),
+                              sd.rows + (size_t)kills[i] * stride, stride,
+                              hipMemcpyDeviceToHost));
+            std::string k;
+            encode_key(oldrec.data(), k);
+            delta[k] = std::nullopt;
+        }
+        for (uint32_t i = flush_mark; i < cur; i++) {
+            const uint8_t* rec = fresh.data() + (size_t)(i - flush_mark) * stride;
+            if (!((const uint32_t*)rec)[0]) continue;
+            std::string k;
+            encode_key(rec, k);
+            std::vector<uint8_t> v;
+            encode_val(rec, v);
+            delta[k] = std::move(v);
+        }
+        auto put32 = [&](uint32_t x) {
+            for (int b2 = 0; b2 < 4; b2++) sp.push_back((uint8_t)(x >> (8 * b2)));
+        };
+        for (auto& [k, v] : delta) {
+            sp.push_back(v.has_value() ? 1 : 0);
+            put32((uint32_t)k.size());
+            sp.insert(sp.end(), k.begin(), k.end());
+            put32(v ? (uint32_t)v->size() : 0);
+            if (v) sp.insert(sp.end(), v->begin(), v->end());
+        }
+        HIP_TRY(hipMemset(sd.killed_cursor, 0, 4));
+        flush_mark = cur;
+        return RW_OK;
+    }
+
     RwChunk* poll() {
         if (outq.empty()) return nullptr;
         RwChunk* c = outq.front();
@@ -4705,6 +4800,16 @@ struct GroupTopN {
 };
 
 extern "C" {
+
+int rw_topn_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len) {
+    std::vector<uint8_t> sp;
+    int rc = ((GroupTopN*)h)->checkpoint_drain(sp);
+    if (rc != RW_OK) return rc;
+    *len = sp.size();
+    *buf = (uint8_t*)malloc(sp.size() ? sp.size() : 1);
+    memcpy(*buf, sp.data(), sp.size());
+    return RW_OK;
+}
 
 void* rw_group_top_n_create(const RwGroupTopNDesc* d) {
     auto* t = new GroupTopN();

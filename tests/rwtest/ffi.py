@@ -586,3 +586,21 @@ def join_checkpoint_drain(lib, h, side):
     L.rw_spill_free.argtypes = [C.c_void_p]
     L.rw_spill_free(C.cast(buf, C.c_void_p))
     return out
+
+
+def topn_checkpoint_drain(lib, h):
+    """Drain the GroupTopN §8f-2 checkpoint spill buffer; returns bytes."""
+    L = lib.lib
+    L.rw_topn_checkpoint_drain.restype = C.c_int
+    L.rw_topn_checkpoint_drain.argtypes = [C.c_void_p,
+                                           C.POINTER(C.POINTER(C.c_uint8)),
+                                           C.POINTER(C.c_uint64)]
+    buf = C.POINTER(C.c_uint8)()
+    ln = C.c_uint64()
+    rc = L.rw_topn_checkpoint_drain(h, C.byref(buf), C.byref(ln))
+    if rc != 0:
+        raise RuntimeError(f"topn drain failed {rc}: {lib.last_error()}")
+    out = bytes(bytearray(buf[i] for i in range(ln.value)))
+    L.rw_spill_free.argtypes = [C.c_void_p]
+    L.rw_spill_free(C.cast(buf, C.c_void_p))
+    return out

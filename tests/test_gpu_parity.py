@@ -1004,3 +1004,38 @@ def test_plain_topn_parity():
         assert mg == mo, f"push {i}: {mg} vs {mo}"
     g.close()
     o.close()
+
+
+def test_topn_checkpoint_spill_parity():
+    # GroupTopN §8f-2 spill: byte-identical GPU vs oracle per epoch,
+    # incl. upserts (same storage key replaced) and delete mixes
+    rng = np.random.default_rng(17)
+    t3 = [T_I64, T_I64, T_I64]
+    g = ffi.GroupTopN(gpu(), t3, [0], [(1, True)], [(2, False)],
+                      offset=0, limit=3)
+    o = ffi.GroupTopN(oracle(), t3, [0], [(1, True)], [(2, False)],
+                      offset=0, limit=3)
+    live = []
+    for ep in range(4):
+        n = 1024
+        gk = rng.integers(0, 30, n)
+        ordv = rng.integers(0, 40, n)
+        pk = rng.integers(0, 50, n)  # small pk space -> frequent upserts
+        ops = np.zeros(n, np.uint8)
+        for r in range(n):
+            if live and rng.random() < 0.3:
+                jx = int(rng.integers(0, len(live)))
+                gk[r], ordv[r], pk[r] = live.pop(jx)
+                ops[r] = ffi.OP_DELETE
+            else:
+                live.append((int(gk[r]), int(ordv[r]), int(pk[r])))
+        c = mk_chunk(t3, ops, [gk, ordv, pk])
+        g.push(c)
+        o.push(c)
+        g.poll_all()
+        o.poll_all()
+        sg = ffi.topn_checkpoint_drain(gpu(), g.h)
+        so = ffi.topn_checkpoint_drain(ffi.oracle(), o.h)
+        assert sg == so, f"epoch {ep}: {len(sg)} vs {len(so)} bytes"
+    g.close()
+    o.close()
