@@ -245,6 +245,7 @@ void rw_group_top_n_destroy(void* h);
  * with rw_spill_free. */
 int rw_agg_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len);
 int rw_join_checkpoint_drain(void* h, int side, uint8_t** buf, uint64_t* len);
+int rw_topn_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len);
 void rw_spill_free(uint8_t* buf);
 
 #ifdef __cplusplus

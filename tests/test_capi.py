@@ -40,6 +40,7 @@ SYMBOLS = [
     "rw_group_top_n_destroy",
     "rw_agg_checkpoint_drain",
     "rw_join_checkpoint_drain",
+    "rw_topn_checkpoint_drain",
     "rw_spill_free",
     "rw_vnode_compute",
     "rw_dispatch_compute",
