@@ -118,7 +118,12 @@ __global__ void x_count_kernel(XBatch b, int n_keys, uint32_t k0, uint32_t k1,
                 else
                     crc = xcrc_u64(lut, crc, (uint64_t)b.col_vals[col][r]);
             }
-            uint32_t vn = (uint32_t)((uint64_t)(crc ^ 0xFFFFFFFFu) % vnode_count);
+            // vnode_count is a power of 2 in practice (256 default):
+            // mask/shift instead of the ~100-instruction u64 div pair
+            uint32_t h = crc ^ 0xFFFFFFFFu;
+            uint32_t vn = (vnode_count & (vnode_count - 1)) == 0
+                              ? (h & (vnode_count - 1))
+                              : (uint32_t)((uint64_t)h % vnode_count);
             dest = vn / per_rank;
             if (dest >= (uint32_t)n_ranks) dest = n_ranks - 1;
             dest_of_row[r] = dest;
