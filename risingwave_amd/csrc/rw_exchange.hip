@@ -373,14 +373,10 @@ int rw_exchange_run(void* h, const int64_t* const* col_vals,
     hipEvent_t e0 = x->e0, e1 = x->e1;
     XHIP(hipEventRecord(e0, x->stream));
 
-    // fewer blocks for the CRC pass: each block stages the 8 KB slice-by-8
-    // LUT from global memory, so 2048 blocks re-read 16 MB of tables for
-    // 9 MB of keys — 512 blocks amortize it 4x (grid-stride covers the rest)
-    uint32_t cblocks = blocks > 512 ? 512 : blocks;
-    x_count_kernel<<<cblocks, 256, 0, x->stream>>>(b, n_keys, k[0], k[1], k[2],
-                                                   k[3], vnode_count, R, d_dest,
-                                                   x->d_block_counts);
-    x_scan_kernel<<<1, 256, 0, x->stream>>>(R, (int)cblocks, x->d_block_counts,
+    x_count_kernel<<<blocks, 256, 0, x->stream>>>(b, n_keys, k[0], k[1], k[2],
+                                                  k[3], vnode_count, R, d_dest,
+                                                  x->d_block_counts);
+    x_scan_kernel<<<1, 256, 0, x->stream>>>(R, (int)blocks, x->d_block_counts,
                                             x->d_block_bases, d_counts);
     mark("count_kernel");
     unsigned long long counts[64];
@@ -401,11 +397,9 @@ int rw_exchange_run(void* h, const int64_t* const* col_vals,
     if (off > send_cap) XFAIL(-2, "send buffer too small (%llu)", (unsigned long long)off);
     XHIP(hipMemcpyAsync(d_offsets, offsets, R * 8, hipMemcpyHostToDevice,
                         x->stream));
-    // same grid as the count pass: block_bases are per-(dest, block) and
-    // the row→block mapping must match exactly
-    x_scatter_kernel<<<cblocks, 256, 0, x->stream>>>(b, n_cols, R, d_dest,
-                                                     d_offsets, d_counts,
-                                                     x->d_block_bases, send_buf);
+    x_scatter_kernel<<<blocks, 256, 0, x->stream>>>(b, n_cols, R, d_dest,
+                                                    d_offsets, d_counts,
+                                                    x->d_block_bases, send_buf);
     mark("scatter");
 
     // exchange per-peer row counts, then the payload blocks (all-to-all-v)
