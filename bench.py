@@ -208,7 +208,8 @@ def main():
     L.rw_agg_apply_payload.restype = ctypes.c_int
     L.rw_agg_apply_payload.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
                                        ctypes.POINTER(ctypes.c_uint64),
-                                       ctypes.c_int, ctypes.c_int]
+                                       ctypes.c_int, ctypes.c_int,
+                                       ctypes.c_int]
     L.rw_agg_n_batch_slots.restype = ctypes.c_int
     L.rw_agg_n_batch_slots.argtypes = [ctypes.c_void_p]
 
@@ -299,7 +300,8 @@ def main():
             recv_blocks = exch.run(agg.h, batches[i % n_batches], xb,
                                    n_cols=nslots)
             rc = L.rw_agg_apply_payload(
-                agg.h, ctypes.c_void_p(xb.recv), recv_blocks, world, nslots)
+                agg.h, ctypes.c_void_p(xb.recv), recv_blocks, world, nslots,
+                1)  # q7 batches are dense (all-Insert, non-null)
             assert rc == 0, gpu_lib.last_error()
         else:
             rc = L.rw_agg_bench_apply(agg.h, batches[i % n_batches])

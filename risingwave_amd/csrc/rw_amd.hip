@@ -2085,30 +2085,39 @@ int rw_agg_batch_ptrs(void* batch, const int64_t** vals, const uint8_t** valids,
 }
 
 // apply an exchange payload (device-resident; per-block layout
-// vals[col][n] ∥ valid[col][n] ∥ ops[n], cols in the executor's batch slot
-// order: group key cols then call args) — one apply launch per block
+// vals[col][npad]*8 ∥ valid[col][n] ∥ ops[n] with npad = n rounded up to 4
+// so every column base is 32-B aligned; cols in the executor's batch slot
+// order: group key cols then call args) — one apply launch per block.
+// `dense` asserts every payload row is a visible non-NULL Insert (the
+// SENDER knows: its source batches carried the dense flag) and routes the
+// receiver through the vectorized dense kernel.
 int rw_agg_apply_payload(void* h, const uint8_t* payload,
-                         const uint64_t* block_rows, int n_blocks, int n_cols) {
+                         const uint64_t* block_rows, int n_blocks, int n_cols,
+                         int dense) {
     auto* agg = (HashAgg*)h;
     if (n_cols != agg->KW + agg->n_calls)
         FAIL(RW_E_INVAL, "payload n_cols %d != %d", n_cols, agg->KW + agg->n_calls);
-    uint64_t row_bytes = 1 + (uint64_t)n_cols * 9;
     uint64_t off = 0;
     for (int bi = 0; bi < n_blocks; bi++) {
         uint64_t n = block_rows[bi];
         if (!n) continue;
+        uint64_t npad = (n + 3) & ~3ull;
         AggBatch b{};
         for (int c = 0; c < n_cols; c++) {
-            b.col_vals[c] = (int64_t*)(payload + off + (uint64_t)c * n * 8);
+            b.col_vals[c] = (int64_t*)(payload + off + (uint64_t)c * npad * 8);
             b.col_valid[c] =
-                (uint8_t*)(payload + off + (uint64_t)n_cols * n * 8 + (uint64_t)c * n);
+                (uint8_t*)(payload + off + (uint64_t)n_cols * npad * 8 +
+                           (uint64_t)c * n);
         }
-        b.ops = (uint8_t*)(payload + off + (uint64_t)n_cols * n * 9);
+        b.ops = (uint8_t*)(payload + off + (uint64_t)n_cols * npad * 8 +
+                           (uint64_t)n_cols * n);
         b.vis = nullptr;
         b.n_rows = (uint32_t)n;
+        b.dense = dense ? 1 : 0;
         int rc = agg->apply(b, true);
         if (rc != RW_OK) return rc;
-        off += (n * row_bytes + 7) & ~7ull;
+        off += ((uint64_t)n_cols * npad * 8 + (uint64_t)n_cols * n + n + 31) &
+               ~31ull;
     }
     return RW_OK;
 }

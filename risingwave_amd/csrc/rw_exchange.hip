@@ -215,14 +215,16 @@ __global__ void x_scatter_kernel(XBatch b, int n_cols, int n_ranks,
                                                           ((1ULL << lane) - 1));
         }
         if (!active) continue;
-        unsigned long long base = offsets[dest]; // byte offset (8-aligned)
+        unsigned long long base = offsets[dest]; // byte offset (32-aligned)
         unsigned long long n = counts[dest];
-        // block layout (8-aligned base): vals[col][n]*8 ∥ valid[col][n] ∥ ops[n]
+        unsigned long long npad = (n + 3) & ~3ull;
+        // block layout: vals[col][npad]*8 ∥ valid[col][n] ∥ ops[n]
         for (int c = 0; c < n_cols; c++) {
-            int64_t* vals = (int64_t*)(out + base + (unsigned long long)c * n * 8);
+            int64_t* vals =
+                (int64_t*)(out + base + (unsigned long long)c * npad * 8);
             vals[idx] = b.col_valid[c][r] ? b.col_vals[c][r] : 0;
         }
-        uint8_t* valids = out + base + (unsigned long long)n_cols * n * 8;
+        uint8_t* valids = out + base + (unsigned long long)n_cols * npad * 8;
         for (int c = 0; c < n_cols; c++) valids[(unsigned long long)c * n + idx] = b.col_valid[c][r];
         uint8_t* ops = valids + (unsigned long long)n_cols * n;
         ops[idx] = b.ops[r];
@@ -386,8 +388,14 @@ int rw_exchange_run(void* h, const int64_t* const* col_vals,
     mark("counts_d2h");
 
     // byte offsets of per-destination blocks in send_buf
-    uint64_t row_bytes = 1 + (uint64_t)n_cols * 9; // op + per col valid+val
-    auto block_bytes = [&](uint64_t nrows) { return (nrows * row_bytes + 7) & ~7ull; };
+    // per-destination block: vals[col][npad]*8 ∥ valid[col][n] ∥ ops[n],
+    // npad = n rounded up to 4 so every vals[col] base is 32-B aligned
+    // (the receiver's dense vectorized apply needs b128-aligned columns)
+    auto block_bytes = [&](uint64_t nrows) {
+        uint64_t npad = (nrows + 3) & ~3ull;
+        return ((uint64_t)n_cols * npad * 8 + (uint64_t)n_cols * nrows + nrows +
+                31) & ~31ull;
+    };
     unsigned long long offsets[64];
     uint64_t off = 0;
     for (int d = 0; d < R; d++) {

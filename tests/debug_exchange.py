@@ -24,7 +24,8 @@ def main():
     L.rw_agg_apply_payload.restype = ctypes.c_int
     L.rw_agg_apply_payload.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
                                        ctypes.POINTER(ctypes.c_uint64),
-                                       ctypes.c_int, ctypes.c_int]
+                                       ctypes.c_int, ctypes.c_int,
+                                       ctypes.c_int]
 
     print("1: create agg + exchange")
     calls = [(ffi.AGG_MAX, 1, ffi.T_I64), (ffi.AGG_COUNT_STAR, -1, ffi.T_I64)]
@@ -49,7 +50,7 @@ def main():
     assert sum(recv_counts) == n
     print("5: apply payload")
     rc = L.rw_agg_apply_payload(agg.h, ctypes.c_void_p(xb.recv), recv_counts, 1,
-                                nslots)
+                                nslots, 1)
     assert rc == 0, gpu.last_error()
     print("6: flush + compare vs oracle")
     agg.flush(1)

@@ -541,7 +541,8 @@ def test_exchange_partition_parity():
     L.rw_agg_apply_payload.restype = ctypes.c_int
     L.rw_agg_apply_payload.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
                                        ctypes.POINTER(ctypes.c_uint64),
-                                       ctypes.c_int, ctypes.c_int]
+                                       ctypes.c_int, ctypes.c_int,
+                                       ctypes.c_int]
     L.rw_agg_n_batch_slots.restype = ctypes.c_int
     L.rw_agg_n_batch_slots.argtypes = [ctypes.c_void_p]
 
@@ -560,7 +561,7 @@ def test_exchange_partition_parity():
     recv_counts = exch.run(agg.h, batch, xb, n_cols=nslots)
     assert sum(recv_counts) == n
     rc = L.rw_agg_apply_payload(agg.h, ctypes.c_void_p(xb.recv), recv_counts,
-                                1, nslots)
+                                1, nslots, 1)
     assert rc == 0
     agg.flush(1)
     got = ffi.rows_multiset(agg.poll_all())
