@@ -293,3 +293,28 @@ def test_emit_on_window_close():
     agg.flush(6)
     assert agg.poll_all() == []
     agg.close()
+
+
+def test_minput_basic_min():
+    # transcribed: aggregate/minput.rs test_extreme_agg_state_basic_min
+    # (values verbatim; the varchar col dropped and int4 widened to int64 —
+    # this build's executors are i64-family, value semantics identical).
+    # Schema (g const, b, c, row_id); min(c), state ordered [c ASC, row_id].
+    agg = ffi.HashAgg(
+        oracle(), [T_I64] * 4, [0],
+        calls=[(AGG_COUNT_STAR, -1, T_I64), (AGG_MIN, 2, T_I64)],
+        row_count_index=0, stream_key=[3])
+    agg.push(from_pretty(""" I I I I
+        + 0 1 8 123
+        + 0 5 2 128
+        - 0 5 2 128
+        + 0 1 3 130"""))
+    agg.flush(1)
+    assert rows_multiset(agg.poll_all()) == expect([("+", (0, 2, 3))])
+    agg.push(from_pretty(""" I I I I
+        + 0 0 8 134
+        + 0 2 2 137"""))
+    agg.flush(2)
+    assert rows_multiset(agg.poll_all()) == expect(
+        [("U-", (0, 2, 3)), ("U+", (0, 4, 2))])
+    agg.close()
