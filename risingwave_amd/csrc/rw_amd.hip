@@ -751,7 +751,22 @@ __device__ __forceinline__ long long agg4_comb(uint8_t kind, long long a,
     }
 }
 
-template <int n_calls, int RPL>
+template <bool NT>
+__device__ __forceinline__ ulonglong2 ld_b128(const ulonglong2* p) {
+    // NT = nontemporal (streaming) load: the input chunk is read exactly
+    // once per epoch, so bypassing L2 retention leaves the cache to the
+    // state table (A/B via RW_AGG_NT=1).
+    if constexpr (NT) {
+        const unsigned long long* q = (const unsigned long long*)p;
+        ulonglong2 r;
+        r.x = __builtin_nontemporal_load(q);
+        r.y = __builtin_nontemporal_load(q + 1);
+        return r;
+    }
+    return *p;
+}
+
+template <int n_calls, int RPL, bool NT = false>
 __device__ __forceinline__ void dense_load(const AggBatch& b,
                                            const AggCallDev* calls,
                                            uint32_t rb, uint32_t r1, bool* act,
@@ -763,7 +778,7 @@ __device__ __forceinline__ void dense_load(const AggBatch& b,
         const ulonglong2* kp = (const ulonglong2*)(b.col_vals[0] + rb);
 #pragma unroll
         for (int h = 0; h < RPL / 2; h++) {
-            ulonglong2 kk = kp[h];
+            ulonglong2 kk = ld_b128<NT>(kp + h);
             k[2 * h] = (long long)kk.x;
             k[2 * h + 1] = (long long)kk.y;
         }
@@ -778,7 +793,7 @@ __device__ __forceinline__ void dense_load(const AggBatch& b,
             const ulonglong2* vp = (const ulonglong2*)(b.col_vals[1 + ci] + rb);
 #pragma unroll
             for (int h = 0; h < RPL / 2; h++) {
-                ulonglong2 vv = vp[h];
+                ulonglong2 vv = ld_b128<NT>(vp + h);
                 cv[2 * h][ci] = agg4_unit(calls[ci].kind, (long long)vv.x);
                 cv[2 * h + 1][ci] = agg4_unit(calls[ci].kind, (long long)vv.y);
             }
@@ -798,10 +813,10 @@ __device__ __forceinline__ void dense_load(const AggBatch& b,
     }
 }
 
-template <int n_calls, int RPL = 4, bool PF = false>
-__global__ void agg_apply_dense4_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
-                                        AggCallDev c1, AggCallDev c2,
-                                        AggCallDev c3, uint32_t r0, uint32_t r1) {
+template <int n_calls, int RPL = 4, bool PF = false, bool NT = false>
+__device__ __forceinline__ void agg_apply_dense4_body(
+        const AggBatch& b, const AggTableDev& t, AggCallDev c0, AggCallDev c1,
+        AggCallDev c2, AggCallDev c3, uint32_t r0, uint32_t r1) {
     AggCallDev calls[4] = {c0, c1, c2, c3};
     int lane = threadIdx.x & 63;
     size_t cap = (size_t)t.cap_mask + 1;
@@ -864,7 +879,7 @@ __global__ void agg_apply_dense4_kernel(AggBatch b, AggTableDev t, AggCallDev c0
     bool act[RPL];
     long long k[RPL];
     long long cv[RPL][n_calls]; // [row][call]
-    if (PF && iters) dense_load<n_calls, RPL>(b, calls, base, r1, act, k, cv);
+    if (PF && iters) dense_load<n_calls, RPL, NT>(b, calls, base, r1, act, k, cv);
     for (uint32_t it = 0; it < iters; it++) {
         uint32_t rb = base + it * stride_rows;
         bool act2[RPL];
@@ -874,10 +889,10 @@ __global__ void agg_apply_dense4_kernel(AggBatch b, AggTableDev t, AggCallDev c0
             // double-buffered prefetch: next tile's loads issue before this
             // tile's scan/commit chain, hiding the HBM round-trip
             if (it + 1 < iters)
-                dense_load<n_calls, RPL>(b, calls, rb + stride_rows, r1, act2,
+                dense_load<n_calls, RPL, NT>(b, calls, rb + stride_rows, r1, act2,
                                          k2, cv2);
         } else {
-            dense_load<n_calls, RPL>(b, calls, rb, r1, act, k, cv);
+            dense_load<n_calls, RPL, NT>(b, calls, rb, r1, act, k, cv);
         }
         // lane-local segments over the RPL rows (inactive rows break runs)
         long long last_key = 0;
@@ -995,6 +1010,23 @@ __global__ void agg_apply_dense4_kernel(AggBatch b, AggTableDev t, AggCallDev c0
             }
         }
     }
+}
+
+template <int n_calls, int RPL = 4, bool PF = false, bool NT = false>
+__global__ void agg_apply_dense4_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
+                                        AggCallDev c1, AggCallDev c2,
+                                        AggCallDev c3, uint32_t r0, uint32_t r1) {
+    agg_apply_dense4_body<n_calls, RPL, PF, NT>(b, t, c0, c1, c2, c3, r0, r1);
+}
+
+// Occupancy experiment (A/B via RW_AGG_W8=1): cap the register budget at
+// 64 VGPRs so 8 waves/SIMD are resident instead of the default build's 7
+// (71 VGPRs) — +14% latency-hiding on a latency-structured kernel.
+template <int n_calls, int RPL = 4, bool NT = false>
+__global__ __launch_bounds__(256, 8) void agg_apply_dense4_kernel_w8(
+        AggBatch b, AggTableDev t, AggCallDev c0, AggCallDev c1, AggCallDev c2,
+        AggCallDev c3, uint32_t r0, uint32_t r1) {
+    agg_apply_dense4_body<n_calls, RPL, false, NT>(b, t, c0, c1, c2, c3, r0, r1);
 }
 
 // agg_flush: flush_data's emit-on-update branch (hash_agg.rs:475-501) +
@@ -1475,7 +1507,45 @@ struct HashAgg {
                 const char* e = getenv("RW_AGG_PF");
                 return e && *e == '1';
             }();
+            // W8: 8-waves/SIMD register-capped variant (A/B via RW_AGG_W8=1)
+            static int w8 = [] {
+                const char* e = getenv("RW_AGG_W8");
+                return e && *e == '1';
+            }();
+            // NT: nontemporal input loads (A/B via RW_AGG_NT=1)
+            static int nt = [] {
+                const char* e = getenv("RW_AGG_NT");
+                return e && *e == '1';
+            }();
             int grid = grid_for((r1 - r0 + rpl - 1) / rpl);
+            if (w8 && rpl == 4) {
+                #define RW_DW8(nc, ntv)                                        \
+                    agg_apply_dense4_kernel_w8<nc, 4, ntv>                     \
+                        <<<grid, 256, 0, stream>>>(b, t, a0, a1, a2, a3, r0, r1)
+                switch (n_calls * 2 + (nt ? 1 : 0)) {
+                    case 1 * 2 + 0: RW_DW8(1, false); return;
+                    case 1 * 2 + 1: RW_DW8(1, true); return;
+                    case 2 * 2 + 0: RW_DW8(2, false); return;
+                    case 2 * 2 + 1: RW_DW8(2, true); return;
+                    case 3 * 2 + 0: RW_DW8(3, false); return;
+                    case 3 * 2 + 1: RW_DW8(3, true); return;
+                    case 4 * 2 + 0: RW_DW8(4, false); return;
+                    case 4 * 2 + 1: RW_DW8(4, true); return;
+                }
+                #undef RW_DW8
+            }
+            if (nt && rpl == 4) {
+                #define RW_DNT(nc)                                             \
+                    agg_apply_dense4_kernel<nc, 4, false, true>                \
+                        <<<grid, 256, 0, stream>>>(b, t, a0, a1, a2, a3, r0, r1)
+                switch (n_calls) {
+                    case 1: RW_DNT(1); return;
+                    case 2: RW_DNT(2); return;
+                    case 3: RW_DNT(3); return;
+                    case 4: RW_DNT(4); return;
+                }
+                #undef RW_DNT
+            }
             if (pf && rpl == 4) {
                 switch (n_calls) {
                     case 1: agg_apply_dense4_kernel<1, 4, true><<<grid, 256, 0, stream>>>(b, t, a0, a1, a2, a3, r0, r1); return;
@@ -4840,3 +4910,48 @@ RwChunk* rw_group_top_n_poll(void* h) { return ((GroupTopN*)h)->poll(); }
 void rw_group_top_n_destroy(void* h) { delete (GroupTopN*)h; }
 
 } // extern "C"
+
+// ---------------------------------------------------------------------------
+// HBM read-bandwidth probe: pins the "achievable" read rate the roofline
+// `frac` is quoted against — the same b128 grid-stride load
+// pattern as dense_load, reduced into a sink so nothing is eliminated.
+// Measurement infrastructure, not a product path.
+__global__ void membw_probe_kernel(const ulonglong2* __restrict__ p, size_t n,
+                                   unsigned long long* sink) {
+    size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    size_t stride = (size_t)gridDim.x * blockDim.x;
+    unsigned long long acc = 0;
+    for (; i + 3 * stride < n; i += 4 * stride) {
+        ulonglong2 a = p[i], b = p[i + stride], c = p[i + 2 * stride],
+                   d = p[i + 3 * stride];
+        acc += a.x + a.y + b.x + b.y + c.x + c.y + d.x + d.y;
+    }
+    for (; i < n; i += stride) acc += p[i].x + p[i].y;
+    if (acc == 0xdeadbeefdeadbeefULL) *sink = acc; // never taken
+}
+
+extern "C" int rw_membw_probe(uint64_t bytes, int iters, double* gbps_out) {
+    size_t n = bytes / sizeof(ulonglong2);
+    ulonglong2* p = nullptr;
+    unsigned long long* sink = nullptr;
+    if (hipMalloc(&p, n * sizeof(ulonglong2)) != hipSuccess) return RW_E_INTERNAL;
+    (void)hipMalloc(&sink, 8);
+    (void)hipMemset(p, 1, n * sizeof(ulonglong2));
+    int grid = 8192; // ≫256 workgroups: fills all 8 XCDs
+    hipEvent_t e0, e1;
+    (void)hipEventCreate(&e0);
+    (void)hipEventCreate(&e1);
+    membw_probe_kernel<<<grid, 256>>>(p, n, sink); // warmup
+    (void)hipEventRecord(e0);
+    for (int it = 0; it < iters; it++) membw_probe_kernel<<<grid, 256>>>(p, n, sink);
+    (void)hipEventRecord(e1);
+    if (hipEventSynchronize(e1) != hipSuccess) return RW_E_INTERNAL;
+    float ms = 0;
+    (void)hipEventElapsedTime(&ms, e0, e1);
+    *gbps_out = (double)bytes * iters / (ms * 1e-3) / 1e9;
+    (void)hipFree(p);
+    (void)hipFree(sink);
+    (void)hipEventDestroy(e0);
+    (void)hipEventDestroy(e1);
+    return RW_OK;
+}
