@@ -766,7 +766,7 @@ __device__ __forceinline__ ulonglong2 ld_b128(const ulonglong2* p) {
     return *p;
 }
 
-template <int n_calls, int RPL, bool NT = false>
+template <int n_calls, int RPL, bool NT = false, int CS = -1>
 __device__ __forceinline__ void dense_load(const AggBatch& b,
                                            const AggCallDev* calls,
                                            uint32_t rb, uint32_t r1, bool* act,
@@ -785,6 +785,7 @@ __device__ __forceinline__ void dense_load(const AggBatch& b,
 #pragma unroll
         for (int r = 0; r < RPL; r++) act[r] = true;
         _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) {
+            if (ci == CS) continue; // count-star carried as run length
             if (calls[ci].arg < 0) {
 #pragma unroll
                 for (int r = 0; r < RPL; r++) cv[r][ci] = 1;
@@ -802,18 +803,26 @@ __device__ __forceinline__ void dense_load(const AggBatch& b,
         for (int r = 0; r < RPL; r++) {
             act[r] = rb + r < r1;
             k[r] = act[r] ? b.col_vals[0][rb + r] : 0;
-            _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++)
+            _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) {
+                if (ci == CS) continue;
                 cv[r][ci] = act[r]
                                 ? agg4_unit(calls[ci].kind,
                                             calls[ci].arg < 0
                                                 ? 1
                                                 : b.col_vals[1 + ci][rb + r])
                                 : 0;
+            }
         }
     }
 }
 
-template <int n_calls, int RPL = 4, bool PF = false, bool NT = false>
+// CS >= 0 names a count-star call (arg < 0): its per-row contribution is
+// the constant 1, so the run's value is the run LENGTH — carried as one
+// u32 beside the scan instead of an i64 inside it (halves the scan's
+// cross-lane traffic for q7's [max, count] shape and frees the VGPRs the
+// count column held). CS == -1 is the generic path, unchanged.
+template <int n_calls, int RPL = 4, bool PF = false, bool NT = false,
+          int CS = -1>
 __device__ __forceinline__ void agg_apply_dense4_body(
         const AggBatch& b, const AggTableDev& t, AggCallDev c0, AggCallDev c1,
         AggCallDev c2, AggCallDev c3, uint32_t r0, uint32_t r1) {
@@ -828,7 +837,8 @@ __device__ __forceinline__ void agg_apply_dense4_body(
     uint32_t memo_slot = SLOT_NONE;
     bool memo_dirtied = false;
 
-    auto commit = [&](long long key, const long long* vals) {
+    auto commit = [&](long long key, const long long* vals, uint32_t cnt) {
+        (void)cnt;
         uint32_t slot;
         if (memo_slot != SLOT_NONE && key == memo_key) {
             slot = memo_slot;
@@ -855,6 +865,10 @@ __device__ __forceinline__ void agg_apply_dense4_body(
         _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) {
             long long* acc = t.acc + (size_t)ci * cap;
             uint8_t* has = t.has + (size_t)ci * cap;
+            if (ci == CS) {
+                atomic_add_i64(&acc[slot], (long long)cnt);
+                continue;
+            }
             switch (calls[ci].kind) {
                 case RW_AGG_MIN:
                     atomic_min_i64(&acc[slot], vals[ci]);
@@ -879,7 +893,7 @@ __device__ __forceinline__ void agg_apply_dense4_body(
     bool act[RPL];
     long long k[RPL];
     long long cv[RPL][n_calls]; // [row][call]
-    if (PF && iters) dense_load<n_calls, RPL, NT>(b, calls, base, r1, act, k, cv);
+    if (PF && iters) dense_load<n_calls, RPL, NT, CS>(b, calls, base, r1, act, k, cv);
     for (uint32_t it = 0; it < iters; it++) {
         uint32_t rb = base + it * stride_rows;
         bool act2[RPL];
@@ -889,16 +903,17 @@ __device__ __forceinline__ void agg_apply_dense4_body(
             // double-buffered prefetch: next tile's loads issue before this
             // tile's scan/commit chain, hiding the HBM round-trip
             if (it + 1 < iters)
-                dense_load<n_calls, RPL, NT>(b, calls, rb + stride_rows, r1, act2,
+                dense_load<n_calls, RPL, NT, CS>(b, calls, rb + stride_rows, r1, act2,
                                          k2, cv2);
         } else {
-            dense_load<n_calls, RPL, NT>(b, calls, rb, r1, act, k, cv);
+            dense_load<n_calls, RPL, NT, CS>(b, calls, rb, r1, act, k, cv);
         }
         // lane-local segments over the RPL rows (inactive rows break runs)
         long long last_key = 0;
         bool any = false, has_bnd = false;
         long long cur_key = 0;
         long long cur[n_calls];
+        uint32_t cur_n = 0, pre_n = 0;
         bool have_cur = false, have_pre = false;
         long long preq[n_calls];
         long long pre_key = 0;
@@ -906,11 +921,15 @@ __device__ __forceinline__ void agg_apply_dense4_body(
             if (!act[r]) {
                 if (have_cur) {
                     if (!have_pre) {
-                        _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) preq[ci] = cur[ci];
+                        _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) {
+                            if (ci == CS) continue;
+                            preq[ci] = cur[ci];
+                        }
                         pre_key = cur_key;
+                        pre_n = cur_n;
                         have_pre = true;
                     } else {
-                        commit(cur_key, cur); // interior complete run
+                        commit(cur_key, cur, cur_n); // interior complete run
                     }
                     has_bnd = true;
                     have_cur = false;
@@ -919,21 +938,32 @@ __device__ __forceinline__ void agg_apply_dense4_body(
             }
             any = true;
             if (have_cur && k[r] == cur_key) {
-                _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++)
+                _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) {
+                    if (ci == CS) continue;
                     cur[ci] = agg4_comb(calls[ci].kind, cur[ci], cv[r][ci]);
+                }
+                cur_n++;
             } else {
                 if (have_cur) {
                     if (!have_pre) {
-                        _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) preq[ci] = cur[ci];
+                        _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) {
+                            if (ci == CS) continue;
+                            preq[ci] = cur[ci];
+                        }
                         pre_key = cur_key;
+                        pre_n = cur_n;
                         have_pre = true;
                     } else {
-                        commit(cur_key, cur);
+                        commit(cur_key, cur, cur_n);
                     }
                     has_bnd = true;
                 }
                 cur_key = k[r];
-                _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) cur[ci] = cv[r][ci];
+                _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) {
+                    if (ci == CS) continue;
+                    cur[ci] = cv[r][ci];
+                }
+                cur_n = 1;
                 have_cur = true;
             }
         }
@@ -944,8 +974,11 @@ __device__ __forceinline__ void agg_apply_dense4_body(
         long long sufv[n_calls];
         bool have_suf = have_cur;
         long long suf_key = cur_key;
-        _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++)
+        _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) {
+            if (ci == CS) continue;
             sufv[ci] = have_cur ? cur[ci] : agg4_identity(calls[ci].kind);
+        }
+        uint32_t suf_n = have_cur ? cur_n : 0;
         last_key = have_suf ? suf_key : (have_pre ? pre_key : 0);
         bool lane_any = any;
         bool lane_has_bnd = has_bnd || (have_pre && have_suf);
@@ -966,17 +999,32 @@ __device__ __forceinline__ void agg_apply_dense4_body(
         int run_start = 63 - __clzll(le_mask | 1ULL);
         int run_pos = lane - run_start;
         long long incl[n_calls];
-        _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) incl[ci] = sufv[ci];
+        _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) {
+            if (ci == CS) continue;
+            incl[ci] = sufv[ci];
+        }
+        uint32_t incl_n = suf_n;
         for (int d = 1; d < 64; d <<= 1) {
             long long ov[n_calls];
-            _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) ov[ci] = __shfl_up(incl[ci], d);
-            if (run_pos >= d)
-                _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++)
+            _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) {
+                if (ci == CS) continue;
+                ov[ci] = __shfl_up(incl[ci], d);
+            }
+            uint32_t ov_n = CS >= 0 ? __shfl_up(incl_n, d) : 0u;
+            if (run_pos >= d) {
+                _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) {
+                    if (ci == CS) continue;
                     incl[ci] = agg4_comb(calls[ci].kind, incl[ci], ov[ci]);
+                }
+                if (CS >= 0) incl_n += ov_n;
+            }
         }
         long long incoming[n_calls];
-        _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++)
+        _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) {
+            if (ci == CS) continue;
             incoming[ci] = __shfl_up(incl[ci], 1);
+        }
+        uint32_t incoming_n = CS >= 0 ? __shfl_up(incl_n, 1) : 0u;
 
         uint64_t cont_b = __ballot(cont);
         uint64_t bnd_b = __ballot(lane_has_bnd);
@@ -987,15 +1035,17 @@ __device__ __forceinline__ void agg_apply_dense4_body(
             if (cont && lane_has_bnd) {
                 // closer: commit the incoming run + own prefix
                 long long tot[n_calls];
-                _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++)
+                _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) {
+                    if (ci == CS) continue;
                     tot[ci] = agg4_comb(calls[ci].kind, incoming[ci], preq[ci]);
-                commit(pre_key, tot);
+                }
+                commit(pre_key, tot, incoming_n + pre_n);
             } else if (!cont && lane_has_bnd && have_pre) {
-                commit(pre_key, preq);
+                commit(pre_key, preq, pre_n);
             }
             bool run_continues = next_cont;
             if (have_suf && !run_continues) {
-                commit(suf_key, incl);
+                commit(suf_key, incl, incl_n);
             } else if (have_suf && next_closes) {
                 // the next lane commits incoming (== this incl) + its prefix
             }
@@ -1005,28 +1055,33 @@ __device__ __forceinline__ void agg_apply_dense4_body(
             for (int r = 0; r < RPL; r++) {
                 act[r] = act2[r];
                 k[r] = k2[r];
-                _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++)
+                _Pragma("unroll") for (int ci = 0; ci < n_calls; ci++) {
+                    if (ci == CS) continue;
                     cv[r][ci] = cv2[r][ci];
+                }
             }
         }
     }
 }
 
-template <int n_calls, int RPL = 4, bool PF = false, bool NT = false>
+template <int n_calls, int RPL = 4, bool PF = false, bool NT = false,
+          int CS = -1>
 __global__ void agg_apply_dense4_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
                                         AggCallDev c1, AggCallDev c2,
                                         AggCallDev c3, uint32_t r0, uint32_t r1) {
-    agg_apply_dense4_body<n_calls, RPL, PF, NT>(b, t, c0, c1, c2, c3, r0, r1);
+    agg_apply_dense4_body<n_calls, RPL, PF, NT, CS>(b, t, c0, c1, c2, c3, r0, r1);
 }
 
-// Occupancy experiment (A/B via RW_AGG_W8=1): cap the register budget at
-// 64 VGPRs so 8 waves/SIMD are resident instead of the default build's 7
-// (71 VGPRs) — +14% latency-hiding on a latency-structured kernel.
-template <int n_calls, int RPL = 4, bool NT = false>
+// Occupancy variant (measured +2.5% at 71→64 VGPRs even with 8 spilled
+// regs, gpurun_out/q7_w8.json): cap the register budget at 64 VGPRs so 8
+// waves/SIMD are resident instead of the default build's 7 — more
+// latency-hiding on a latency-structured kernel. A/B via RW_AGG_W8=0/1.
+template <int n_calls, int RPL = 4, bool NT = false, int CS = -1>
 __global__ __launch_bounds__(256, 8) void agg_apply_dense4_kernel_w8(
         AggBatch b, AggTableDev t, AggCallDev c0, AggCallDev c1, AggCallDev c2,
         AggCallDev c3, uint32_t r0, uint32_t r1) {
-    agg_apply_dense4_body<n_calls, RPL, false, NT>(b, t, c0, c1, c2, c3, r0, r1);
+    agg_apply_dense4_body<n_calls, RPL, false, NT, CS>(b, t, c0, c1, c2, c3, r0,
+                                                       r1);
 }
 
 // agg_flush: flush_data's emit-on-update branch (hash_agg.rs:475-501) +
@@ -1518,6 +1573,33 @@ struct HashAgg {
                 return e && *e == '1';
             }();
             int grid = grid_for((r1 - r0 + rpl - 1) / rpl);
+            // CS: count-star carried as run length beside the scan instead
+            // of an i64 inside it (A/B via RW_AGG_CS=0; default on).
+            static int cs_en = [] {
+                const char* e = getenv("RW_AGG_CS");
+                return !(e && *e == '0');
+            }();
+            int cs = -1;
+            for (int ci = 0; ci < n_calls && cs < 0; ci++)
+                if (cd(ci).arg < 0 && cd(ci).kind == RW_AGG_COUNT_STAR) cs = ci;
+            if (cs_en && cs >= 0 && rpl == 4 && !pf && !nt) {
+                #define RW_DCS(nc, csv)                                        \
+                    agg_apply_dense4_kernel<nc, 4, false, false, csv>          \
+                        <<<grid, 256, 0, stream>>>(b, t, a0, a1, a2, a3, r0, r1)
+                #define RW_DCS8(nc, csv)                                       \
+                    agg_apply_dense4_kernel_w8<nc, 4, false, csv>              \
+                        <<<grid, 256, 0, stream>>>(b, t, a0, a1, a2, a3, r0, r1)
+                switch ((w8 ? 128 : 0) + n_calls * 8 + cs) {
+                    case 1 * 8 + 0: RW_DCS(1, 0); return;
+                    case 2 * 8 + 0: RW_DCS(2, 0); return;
+                    case 2 * 8 + 1: RW_DCS(2, 1); return;
+                    case 128 + 1 * 8 + 0: RW_DCS8(1, 0); return;
+                    case 128 + 2 * 8 + 0: RW_DCS8(2, 0); return;
+                    case 128 + 2 * 8 + 1: RW_DCS8(2, 1); return;
+                }
+                #undef RW_DCS
+                #undef RW_DCS8
+            }
             if (w8 && rpl == 4) {
                 #define RW_DW8(nc, ntv)                                        \
                     agg_apply_dense4_kernel_w8<nc, 4, ntv>                     \

@@ -1,0 +1,32 @@
+#!/usr/bin/env python3
+"""HBM read-bandwidth probe (measurement infrastructure).
+
+Pins the "achievable" read rate that the roofline `frac` is quoted
+against: the same b128 grid-stride load pattern as the dense agg kernel's
+input stream (rw_amd.hip `membw_probe_kernel`), on a buffer far larger
+than L2. Prints one JSON line; not a pytest test and not a product path.
+"""
+import ctypes
+import json
+import os
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
+
+import risingwave_amd
+
+risingwave_amd.load_library()
+L = ctypes.CDLL(risingwave_amd.lib_path())
+L.rw_membw_probe.restype = ctypes.c_int
+L.rw_membw_probe.argtypes = [ctypes.c_uint64, ctypes.c_int,
+                             ctypes.POINTER(ctypes.c_double)]
+
+out = []
+for gib in (1, 4, 16):
+    g = ctypes.c_double(0.0)
+    rc = L.rw_membw_probe(gib << 30, 8, ctypes.byref(g))
+    if rc != 0:
+        print(json.dumps({"error": rc, "GiB": gib}))
+        sys.exit(1)
+    out.append({"GiB": gib, "read_GBps": round(g.value, 1)})
+print(json.dumps({"membw_probe": out}))

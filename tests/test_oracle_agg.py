@@ -318,3 +318,94 @@ def test_minput_basic_min():
     assert rows_multiset(agg.poll_all()) == expect(
         [("U-", (0, 2, 3)), ("U+", (0, 4, 2))])
     agg.close()
+
+
+def test_minput_basic_max():
+    # transcribed: aggregate/minput.rs test_extreme_agg_state_basic_max
+    # (values verbatim; varchar col dropped, int4 widened to int64).
+    # Schema (g const, b, c, row_id); max(c), state ordered [c DESC, row_id].
+    from rwtest.ffi import AGG_MAX
+    agg = ffi.HashAgg(
+        oracle(), [T_I64] * 4, [0],
+        calls=[(AGG_COUNT_STAR, -1, T_I64), (AGG_MAX, 2, T_I64)],
+        row_count_index=0, stream_key=[3])
+    agg.push(from_pretty(""" I I I I
+        + 0 1 8 123
+        + 0 5 2 128
+        - 0 5 2 128
+        + 0 1 3 130"""))
+    agg.flush(1)
+    assert rows_multiset(agg.poll_all()) == expect([("+", (0, 2, 8))])
+    agg.push(from_pretty(""" I I I I
+        + 0 0 9 134
+        + 0 2 2 137"""))
+    agg.flush(2)
+    assert rows_multiset(agg.poll_all()) == expect(
+        [("U-", (0, 2, 8)), ("U+", (0, 4, 9))])
+    agg.close()
+
+
+def test_minput_hidden_input():
+    # transcribed: aggregate/minput.rs test_extreme_agg_state_with_hidden_input
+    # — rows hidden by the visibility bitmap (`D`) must not enter minput
+    # state; NULL agg-column rows materialize as NULL rows. Two states:
+    # min over the varchar col (mapped a=1 b=2 c=3, NULL kept) and max over
+    # the int col; each with its own executor, as the reference keeps two
+    # state tables.
+    from rwtest.ffi import AGG_MAX
+    # min(a): schema (g const, a_code, row_id)
+    agg = ffi.HashAgg(
+        oracle(), [T_I64] * 3, [0],
+        calls=[(AGG_COUNT_STAR, -1, T_I64), (AGG_MIN, 1, T_I64)],
+        row_count_index=0, stream_key=[2])
+    agg.push(from_pretty(""" I I I
+        + 0 1 123
+        + 0 2 128
+        - 0 2 128
+        + 0 3 130
+        + 0 . 131 D
+        + 0 . 132 D
+        + 0 3 133"""))
+    agg.flush(1)
+    assert rows_multiset(agg.poll_all()) == expect([("+", (0, 3, 1))])
+    agg.close()
+    # max(b): schema (g const, b, row_id); the NULL-b row is hidden
+    agg = ffi.HashAgg(
+        oracle(), [T_I64] * 3, [0],
+        calls=[(AGG_COUNT_STAR, -1, T_I64), (AGG_MAX, 1, T_I64)],
+        row_count_index=0, stream_key=[2])
+    agg.push(from_pretty(""" I I I
+        + 0 1 123
+        + 0 5 128
+        - 0 5 128
+        + 0 1 130
+        + 0 9 131
+        + 0 6 132
+        + 0 . 133 D"""))
+    agg.flush(1)
+    assert rows_multiset(agg.poll_all()) == expect([("+", (0, 4, 9))])
+    agg.close()
+
+
+def test_minput_grouped():
+    # transcribed: aggregate/minput.rs test_extreme_agg_state_grouped —
+    # max(b) within group c=8; hidden rows (other groups) never touch it.
+    # Schema (c group, b, row_id).
+    from rwtest.ffi import AGG_MAX
+    agg = ffi.HashAgg(
+        oracle(), [T_I64] * 3, [0],
+        calls=[(AGG_COUNT_STAR, -1, T_I64), (AGG_MAX, 1, T_I64)],
+        row_count_index=0, stream_key=[2])
+    agg.push(from_pretty(""" I I I
+        + 8 1 123
+        + 8 5 128
+        + 3 7 130 D"""))
+    agg.flush(1)
+    assert rows_multiset(agg.poll_all()) == expect([("+", (8, 2, 5))])
+    agg.push(from_pretty(""" I I I
+        + 2 9 134 D
+        + 8 8 137"""))
+    agg.flush(2)
+    assert rows_multiset(agg.poll_all()) == expect(
+        [("U-", (8, 2, 5)), ("U+", (8, 3, 8))])
+    agg.close()
