@@ -154,6 +154,22 @@ def test_agg_min_append_only_golden():
     run_and_compare(g, o, [[e1], [e2]])
 
 
+def test_agg_dense_tail_and_runs():
+    # dense-kernel edge cases: chunk sizes that leave partial tail tiles
+    # (not multiples of the 4-rows/lane x 64-lane wave tile), giant
+    # single-key runs spanning many waves (q7's monotone windows), and the
+    # count-star-as-run-length path (rw_amd.hip agg_apply_dense4_body CS).
+    rng = np.random.default_rng(7)
+    calls = [(AGG_MAX, 1, T_I64), (AGG_COUNT_STAR, -1, T_I64)]
+    for n, n_keys in [(1031, 1), (1537, 2), (4099, 3), (2047, 500)]:
+        g, o = agg_pair(calls, 1, append_only=True)
+        # sorted keys => giant runs; the tail tile ends mid-run
+        keys = np.sort(rng.integers(0, n_keys, n))
+        vals = rng.integers(1, 10**7, n)
+        c = mk_chunk([T_I64, T_I64], np.zeros(n, np.uint8), [keys, vals])
+        run_and_compare(g, o, [[c]])
+
+
 # ---------------- HashJoin ----------------
 
 
