@@ -246,6 +246,15 @@ void rw_group_top_n_destroy(void* h);
 int rw_agg_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len);
 int rw_join_checkpoint_drain(void* h, int side, uint8_t** buf, uint64_t* len);
 int rw_topn_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len);
+/* DISTINCT dedup tables (one StateTable per distinct column in the
+ * reference, from_proto/hash_agg.rs distinct_dedup_tables): pk = group key
+ * ∥ datum, value = full row ++ one i64 count per call distincting on the
+ * column (distinct.rs:89-93,158-185). Same record framing; one record per
+ * (group, datum) touched since the last drain, sorted by pk; DELETE when
+ * the count dropped to 0. `di` indexes tables in first-use order;
+ * rw_agg_n_dedup_tables returns how many exist. */
+int rw_agg_n_dedup_tables(void* h);
+int rw_agg_dedup_drain(void* h, int di, uint8_t** buf, uint64_t* len);
 void rw_spill_free(uint8_t* buf);
 
 #ifdef __cplusplus

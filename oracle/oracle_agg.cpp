@@ -132,6 +132,8 @@ struct HashAggOracle {
                 for (auto t : group_key_types) less.order.push_back({t, false});
                 less.order.push_back({input_types[col], false});
                 dedup_counts.emplace_back(less);
+                dedup_touched.emplace_back(less);
+                dedup_persisted.emplace_back(less);
             }
             call_dedup_idx[ci] = (int)di;
         }
@@ -142,6 +144,10 @@ struct HashAggOracle {
     std::vector<uint32_t> distinct_cols;
     std::vector<std::map<Row, int64_t, RowOrderLess>> dedup_counts;
     std::vector<int> call_dedup_idx; // per call: index into dedup_counts or -1
+    // §8f-2 dedup-table spill: (group, datum) keys touched this epoch and
+    // the set persisted at the last checkpoint (distinct.rs:158-185)
+    std::vector<std::set<Row, RowOrderLess>> dedup_touched;
+    std::vector<std::set<Row, RowOrderLess>> dedup_persisted;
 
     RowOrderLess minput_order(const RwAggCall& c) const {
         // pk of the materialized-input table: value (ASC min / DESC max),
@@ -267,6 +273,7 @@ struct HashAggOracle {
                 for (size_t di = 0; di < distinct_cols.size(); di++) {
                     Row dk = key;
                     dk.push_back(cv.at(r, distinct_cols[di]));
+                    dedup_touched[di].insert(dk);
                     auto& m = dedup_counts[di];
                     if (!retract) {
                         int64_t cnt = ++m[dk];
@@ -375,6 +382,58 @@ struct HashAggOracle {
         spill.insert(spill.end(), k.begin(), k.end());
         put32((uint32_t)v.size());
         spill.insert(spill.end(), v.begin(), v.end());
+    }
+
+    // §8f-2 drain for one dedup table: one record per (group, datum)
+    // touched this epoch — PUT with the current counts, DELETE when the
+    // count dropped to 0 and the row was persisted, nothing when it was
+    // created and died within the epoch (mem-table netting). Sorted by pk
+    // (the canonical form; the reference iterates a HashMap).
+    int dedup_drain(int di, std::vector<uint8_t>& out) {
+        if (di < 0 || (size_t)di >= distinct_cols.size()) return RW_E_INVAL;
+        std::vector<uint8_t> key_types(group_key_types);
+        key_types.push_back(input_types[distinct_cols[di]]);
+        auto put32 = [&](uint32_t x) {
+            for (int b = 0; b < 4; b++) out.push_back((uint8_t)(x >> (8 * b)));
+        };
+        auto& counts = dedup_counts[di];
+        auto& persisted = dedup_persisted[di];
+        for (const Row& dk : dedup_touched[di]) {
+            auto it = counts.find(dk);
+            int64_t cnt = it == counts.end() ? 0 : it->second;
+            uint8_t put;
+            if (cnt > 0) {
+                put = 1;
+                persisted.insert(dk);
+            } else if (persisted.count(dk)) {
+                put = 0;
+                persisted.erase(dk);
+            } else {
+                continue;
+            }
+            std::vector<uint8_t> k, v;
+            for (size_t c = 0; c < dk.size(); c++) {
+                rwcodec::DatumC d{dk[c].null, dk[c].i, dk[c].d};
+                rwcodec::memcmp_encode_datum(k, key_types[c], d, {});
+            }
+            if (put) {
+                for (size_t c = 0; c < dk.size(); c++) {
+                    rwcodec::DatumC d{dk[c].null, dk[c].i, dk[c].d};
+                    rwcodec::value_encode_datum(v, key_types[c], d);
+                }
+                for (size_t ci = 0; ci < calls.size(); ci++)
+                    if (call_dedup_idx[ci] == di)
+                        rwcodec::value_encode_datum(v, RW_T_I64,
+                                                    {false, cnt, 0});
+            }
+            out.push_back(put);
+            put32((uint32_t)k.size());
+            out.insert(out.end(), k.begin(), k.end());
+            put32((uint32_t)v.size());
+            out.insert(out.end(), v.begin(), v.end());
+        }
+        dedup_touched[di].clear();
+        return RW_OK;
     }
 
     void emit(uint8_t op, const Row& key, const Row& outputs) {
@@ -553,6 +612,21 @@ int rw_agg_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len) {
     *buf = (uint8_t*)malloc(a->spill.size() ? a->spill.size() : 1);
     memcpy(*buf, a->spill.data(), a->spill.size());
     a->spill.clear();
+    return RW_OK;
+}
+
+int rw_agg_n_dedup_tables(void* h) {
+    return (int)((HashAggOracle*)h)->distinct_cols.size();
+}
+
+int rw_agg_dedup_drain(void* h, int di, uint8_t** buf, uint64_t* len) {
+    auto* a = (HashAggOracle*)h;
+    std::vector<uint8_t> sp;
+    int rc = a->dedup_drain(di, sp);
+    if (rc != RW_OK) return rc;
+    *len = sp.size();
+    *buf = (uint8_t*)malloc(sp.size() ? sp.size() : 1);
+    memcpy(*buf, sp.data(), sp.size());
     return RW_OK;
 }
 

@@ -358,7 +358,9 @@ __global__ void dedup_init_kernel(JoinSlot* slots, size_t cap) {
 
 __global__ void agg_dedup_kernel(AggBatch b, JoinSlot* slots,
                                  uint32_t cap_mask, int KW, int dslot,
-                                 uint8_t* hidden, uint32_t* err) {
+                                 uint8_t* hidden, uint32_t* err,
+                                 uint32_t* dirty_flag, uint32_t* dirty_list,
+                                 uint32_t* dirty_n) {
     uint32_t stride = gridDim.x * blockDim.x;
     for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < b.n_rows;
          r += stride) {
@@ -385,6 +387,39 @@ __global__ void agg_dedup_kernel(AggBatch b, JoinSlot* slots,
         uint32_t old = retract ? atomicSub(&slots[slot].head, 1u)
                                : atomicAdd(&slots[slot].head, 1u);
         hidden[rs] = retract ? (old != 1) : (old != 0);
+        // §8f-2 spill: record each (group, datum) touched this epoch, once
+        // (the reference writes an update per touched datum at dedup(),
+        // distinct.rs:158-185; the per-epoch drain nets those to one)
+        if (ld_u32(&dirty_flag[slot]) == 0 &&
+            atomicCAS(&dirty_flag[slot], 0u, 1u) == 0u) {
+            uint32_t i = atomicAdd(dirty_n, 1u);
+            dirty_list[i] = slot;
+        }
+    }
+}
+
+// Dedup-table checkpoint gather: pack the dirty slots' (key, count) into a
+// compact buffer for one D2H copy, clearing the dirty flags on the way.
+struct DedupDirtyRec {
+    long long key[MAX_KW]; // group key words ++ datum (KW+1 used)
+    uint32_t nulls;
+    uint32_t count;
+};
+__global__ void dedup_gather_kernel(const JoinSlot* slots,
+                                    uint32_t* dirty_flag,
+                                    const uint32_t* dirty_list,
+                                    const uint32_t* dirty_n,
+                                    DedupDirtyRec* out) {
+    uint32_t n = *dirty_n;
+    uint32_t stride = gridDim.x * blockDim.x;
+    for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        uint32_t s = dirty_list[i];
+        const JoinSlot* sl = &slots[s];
+        for (int w = 0; w < MAX_KW; w++) out[i].key[w] = sl->key[w];
+        out[i].nulls = sl->nulls;
+        out[i].count = sl->head;
+        dirty_flag[s] = 0;
     }
 }
 
@@ -1310,6 +1345,11 @@ struct HashAgg {
     uint32_t dedup_cap_mask = 0;
     std::vector<uint8_t*> hidden_bufs; // device, per dedup table
     uint32_t hidden_cap = 0;
+    // §8f-2 dedup-table spill: per table, device dirty tracking + a host
+    // persisted bitmap (was the row in the table at the last checkpoint?)
+    std::vector<uint8_t> distinct_col_types; // datum col type per table
+    std::vector<uint32_t*> ddirty_flag, ddirty_list, ddirty_n;
+    std::vector<std::vector<uint8_t>> dedup_persisted;
 
     int grid_for(uint32_t work) const {
         uint32_t blocks = (work + 255) / 256;
@@ -1378,7 +1418,10 @@ struct HashAgg {
                     di = (size_t)call_dedup_idx[cj];
                     break;
                 }
-            if (di == distinct_slots.size()) distinct_slots.push_back(slot);
+            if (di == distinct_slots.size()) {
+                distinct_slots.push_back(slot);
+                distinct_col_types.push_back(input_types[calls[ci].arg]);
+            }
             call_dedup_idx[ci] = (int)di;
         }
         if (n_minput) {
@@ -1801,6 +1844,16 @@ struct HashAgg {
                 HIP_TRY(hipMalloc(&sl, cap * sizeof(JoinSlot)));
                 dedup_init_kernel<<<2048, 256, 0, stream>>>(sl, cap);
                 dedup_slots.push_back(sl);
+                uint32_t *df = nullptr, *dl = nullptr, *dn = nullptr;
+                HIP_TRY(hipMalloc(&df, cap * 4));
+                HIP_TRY(hipMemset(df, 0, cap * 4));
+                HIP_TRY(hipMalloc(&dl, cap * 4));
+                HIP_TRY(hipMalloc(&dn, 4));
+                HIP_TRY(hipMemset(dn, 0, 4));
+                ddirty_flag.push_back(df);
+                ddirty_list.push_back(dl);
+                ddirty_n.push_back(dn);
+                dedup_persisted.emplace_back(cap, 0);
             }
         }
         if (hidden_cap < n) {
@@ -1825,7 +1878,8 @@ struct HashAgg {
             for (size_t di = 0; di < distinct_slots.size(); di++)
                 agg_dedup_kernel<<<grid, 256, 0, stream>>>(
                     b, dedup_slots[di], dedup_cap_mask, KW, distinct_slots[di],
-                    hidden_bufs[di], &t.counters[2]);
+                    hidden_bufs[di], &t.counters[2], ddirty_flag[di],
+                    ddirty_list[di], ddirty_n[di]);
             for (size_t ci = 0; ci < calls.size(); ci++)
                 if (call_dedup_idx[ci] >= 0)
                     b.call_hidden[ci] = hidden_bufs[call_dedup_idx[ci]];
@@ -2004,6 +2058,88 @@ struct HashAgg {
             put32((uint32_t)v.size());
             spill.insert(spill.end(), v.begin(), v.end());
         }
+    }
+
+    // §8f-2 drain for one DISTINCT dedup table (its own state table in the
+    // reference, one per distinct column: pk = group key ∥ datum, value =
+    // full row ++ one i64 count per call distincting on the column,
+    // distinct.rs:89-93,158-185; DELETE when the count drops to 0, nothing
+    // when a row created this epoch also dies in it — mem-table netting,
+    // as the join/topn drains). Records sorted by memcomparable pk: the
+    // reference iterates a HashMap (nondeterministic), so sorted order is
+    // the canonical form both this and the oracle emit.
+    int dedup_drain(int di, std::vector<uint8_t>& out) {
+        if (di < 0 || (size_t)di >= distinct_slots.size())
+            FAIL(RW_E_INVAL, "dedup table index %d (have %zu)", di,
+                 distinct_slots.size());
+        if ((size_t)di >= dedup_slots.size()) return RW_OK; // nothing pushed
+        HIP_TRY(hipStreamSynchronize(stream));
+        uint32_t n = 0;
+        HIP_TRY(hipMemcpy(&n, ddirty_n[di], 4, hipMemcpyDeviceToHost));
+        if (!n) return RW_OK;
+        std::vector<uint32_t> slots_h(n);
+        HIP_TRY(hipMemcpy(slots_h.data(), ddirty_list[di], (size_t)n * 4,
+                          hipMemcpyDeviceToHost));
+        DedupDirtyRec* d_out = nullptr;
+        HIP_TRY(hipMalloc(&d_out, (size_t)n * sizeof(DedupDirtyRec)));
+        dedup_gather_kernel<<<grid_for(n), 256, 0, stream>>>(
+            dedup_slots[di], ddirty_flag[di], ddirty_list[di], ddirty_n[di],
+            d_out);
+        std::vector<DedupDirtyRec> recs(n);
+        int rc_cp = hipMemcpy(recs.data(), d_out,
+                              (size_t)n * sizeof(DedupDirtyRec),
+                              hipMemcpyDeviceToHost);
+        hipFree(d_out);
+        if (rc_cp != hipSuccess) FAIL(RW_E_INTERNAL, "dedup gather copy");
+        HIP_TRY(hipMemset(ddirty_n[di], 0, 4));
+        std::vector<uint8_t> key_types(out_types.begin(),
+                                       out_types.begin() + KW);
+        key_types.push_back(distinct_col_types[di]);
+        auto& persisted = dedup_persisted[di];
+        struct Rec { std::vector<uint8_t> k, v; uint8_t put; };
+        std::vector<Rec> rs;
+        rs.reserve(n);
+        for (uint32_t i = 0; i < n; i++) {
+            const DedupDirtyRec& r = recs[i];
+            uint32_t slot = slots_h[i];
+            Rec e;
+            if (r.count > 0) {
+                e.put = 1;
+                persisted[slot] = 1;
+            } else {
+                if (!persisted[slot]) continue; // created+died this epoch
+                e.put = 0;
+                persisted[slot] = 0;
+            }
+            for (int c = 0; c <= KW; c++) {
+                rwcodec::DatumC d{((r.nulls >> c) & 1) != 0, r.key[c], 0};
+                rwcodec::memcmp_encode_datum(e.k, key_types[c], d, {});
+            }
+            if (e.put) {
+                for (int c = 0; c <= KW; c++) {
+                    rwcodec::DatumC d{((r.nulls >> c) & 1) != 0, r.key[c], 0};
+                    rwcodec::value_encode_datum(e.v, key_types[c], d);
+                }
+                for (int ci = 0; ci < n_calls; ci++)
+                    if (call_dedup_idx[ci] == di)
+                        rwcodec::value_encode_datum(
+                            e.v, RW_T_I64, {false, (long long)r.count, 0});
+            }
+            rs.push_back(std::move(e));
+        }
+        std::sort(rs.begin(), rs.end(),
+                  [](const Rec& a, const Rec& b) { return a.k < b.k; });
+        auto put32 = [&](uint32_t x) {
+            for (int b = 0; b < 4; b++) out.push_back((uint8_t)(x >> (8 * b)));
+        };
+        for (auto& e : rs) {
+            out.push_back(e.put);
+            put32((uint32_t)e.k.size());
+            out.insert(out.end(), e.k.begin(), e.k.end());
+            put32((uint32_t)e.v.size());
+            out.insert(out.end(), e.v.begin(), e.v.end());
+        }
+        return RW_OK;
     }
 
     // Host-side chunking with the U-pair no-split rule
@@ -2408,6 +2544,21 @@ int rw_agg_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len) {
     *buf = (uint8_t*)malloc(agg->spill.size() ? agg->spill.size() : 1);
     memcpy(*buf, agg->spill.data(), agg->spill.size());
     agg->spill.clear();
+    return RW_OK;
+}
+
+int rw_agg_n_dedup_tables(void* h) {
+    return (int)((HashAgg*)h)->distinct_slots.size();
+}
+
+int rw_agg_dedup_drain(void* h, int di, uint8_t** buf, uint64_t* len) {
+    auto* agg = (HashAgg*)h;
+    std::vector<uint8_t> sp;
+    int rc = agg->dedup_drain(di, sp);
+    if (rc != RW_OK) return rc;
+    *len = sp.size();
+    *buf = (uint8_t*)malloc(sp.size() ? sp.size() : 1);
+    memcpy(*buf, sp.data(), sp.size());
     return RW_OK;
 }
 

@@ -30,6 +30,29 @@ def drain(lib, agg):
     return parse(data)
 
 
+def dedup_drain(lib, agg, di):
+    L = lib.lib
+    L.rw_agg_dedup_drain.restype = ctypes.c_int
+    L.rw_agg_dedup_drain.argtypes = [ctypes.c_void_p, ctypes.c_int,
+                                     ctypes.POINTER(ctypes.c_void_p),
+                                     ctypes.POINTER(ctypes.c_uint64)]
+    L.rw_spill_free.argtypes = [ctypes.c_void_p]
+    buf = ctypes.c_void_p()
+    ln = ctypes.c_uint64()
+    assert L.rw_agg_dedup_drain(agg.h, di, ctypes.byref(buf),
+                                ctypes.byref(ln)) == 0
+    data = ctypes.string_at(buf, ln.value)
+    L.rw_spill_free(buf)
+    return parse(data)
+
+
+def n_dedup_tables(lib, agg):
+    L = lib.lib
+    L.rw_agg_n_dedup_tables.restype = ctypes.c_int
+    L.rw_agg_n_dedup_tables.argtypes = [ctypes.c_void_p]
+    return L.rw_agg_n_dedup_tables(agg.h)
+
+
 def parse(data):
     recs = []
     off = 0
@@ -183,4 +206,50 @@ def test_join_spill_tristate_netting():
             b"\x01" + struct.pack("<I", len(k2)) + k2 +
             struct.pack("<I", len(v2)) + v2)
     assert sp == want, f"{sp.hex()}\nvs\n{want.hex()}"
+    o.close()
+
+
+def test_dedup_spill_records():
+    # §8f-2 DISTINCT dedup-table drain (distinct.rs:89-93,158-185): pk =
+    # group ∥ datum (memcomparable), value = full row ++ one i64 count per
+    # call distincting on the column; one record per touched (group, datum),
+    # sorted by pk; DELETE at count 0; created+died in the epoch → nothing.
+    from rwtest.ffi import AGG_COUNT
+
+    calls = [(AGG_COUNT_STAR, -1, T_I64), (AGG_COUNT, 1, T_I64, 1)]
+    o = ffi.HashAgg(oracle(), [T_I64, T_I64], [0], calls, 0)
+    from test_codec import dedup_drain, n_dedup_tables  # self-import ok
+    assert n_dedup_tables(oracle(), o) == 1
+    o.push(from_pretty(""" I I
+        + 1 5
+        + 1 5
+        + 1 7
+        + 2 5
+        + 1 9
+        - 1 9"""))
+    o.flush(1)
+    o.poll_all()
+    recs = dedup_drain(oracle(), o, 0)
+    def k(g, d):
+        return memcmp_i64(g) + memcmp_i64(d)
+    def v(g, d, cnt):
+        return value_i64(g) + value_i64(d) + value_i64(cnt)
+    assert recs == [
+        (1, k(1, 5), v(1, 5, 2)),
+        (1, k(1, 7), v(1, 7, 1)),
+        (1, k(2, 5), v(2, 5, 1)),
+    ], recs
+    # epoch 2: (1,5) drops to 1 → PUT; (1,7) drops to 0 → DELETE
+    o.push(from_pretty(""" I I
+        - 1 5
+        - 1 7"""))
+    o.flush(2)
+    o.poll_all()
+    recs = dedup_drain(oracle(), o, 0)
+    assert recs == [
+        (1, k(1, 5), v(1, 5, 1)),
+        (0, k(1, 7), b""),
+    ], recs
+    # epoch 3: untouched → empty drain
+    assert dedup_drain(oracle(), o, 0) == []
     o.close()

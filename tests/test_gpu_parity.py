@@ -871,6 +871,14 @@ def test_count_distinct_parity():
             a.flush(ep + 1)
             outs.append(rows_multiset(a.poll_all()))
         assert outs[0] == outs[1], f"epoch {ep}"
+        # §8f-2 dedup-table spill: byte-identical drains (both sorted by pk)
+        from test_codec import dedup_drain, n_dedup_tables
+        assert n_dedup_tables(gpu(), g) == n_dedup_tables(oracle(), o) == 1
+        dg = dedup_drain(gpu(), g, 0)
+        do = dedup_drain(oracle(), o, 0)
+        assert dg == do, (f"epoch {ep}: dedup spill {len(dg)} vs {len(do)};"
+                          f" first diff: "
+                          f"{next(((a, b) for a, b in zip(dg, do) if a != b), None)}")
     g.close()
     o.close()
 
