@@ -492,6 +492,14 @@ struct HashAggOracle {
     // order (SortBuffer::consume iterates ordered) and are removed. Dirty
     // state rows spill as PUTs; closed groups spill DELETEs.
     int flush_eowc() {
+        // mid-window state PUTs (hash_agg.rs:429-460): every dirty group's
+        // CURRENT states upsert into the intermediate table at the barrier
+        // even though nothing is emitted until the window closes
+        for (auto& key : dirty_order) {
+            auto& g = groups[key];
+            Row curr = get_outputs(g);
+            spill_record(1, key, curr);
+        }
         dirty_order.clear();
         dirty.clear();
         if (has_pending_wm) {
@@ -506,12 +514,11 @@ struct HashAggOracle {
             for (auto& key : closing) {
                 auto& g = groups[key];
                 Row curr = get_outputs(g);
-                if (row_count_of(curr) != 0) {
-                    emit(RW_OP_INSERT, key, curr);
-                    // EOWC spill this round: the window-close DELETE only
-                    // (mid-window state PUTs are a later-round item)
-                    spill_record(0, key, curr);
-                }
+                // every closed window leaves the intermediate table (the
+                // reference deletes all rows < wm); rc==0 windows delete
+                // their mid-window row without emitting
+                spill_record(0, key, curr);
+                if (row_count_of(curr) != 0) emit(RW_OP_INSERT, key, curr);
                 groups.erase(key);
             }
             has_pending_wm = false;

@@ -1285,6 +1285,11 @@ __global__ void agg_eowc_close_kernel(AggTableDev t, int KW, int n_calls,
                                       AggCallDev c0, AggCallDev c1,
                                       AggCallDev c2, AggCallDev c3);
 
+__global__ void agg_eowc_dump_kernel(AggTableDev t, int KW, int n_calls,
+                                     int row_count_index, AggCallDev c0,
+                                     AggCallDev c1, AggCallDev c2,
+                                     AggCallDev c3);
+
 extern "C" __global__ void agg_counters_reset_kernel(uint32_t* counters);
 
 struct HashAgg {
@@ -1902,14 +1907,30 @@ struct HashAgg {
 
     int flush(uint64_t) {
         if (eowc) {
-            // EOWC barrier (hash_agg.rs:429-474): nothing is emitted until a
-            // watermark closes windows; the dirty list only resets (state
-            // accumulates across epochs)
-            if (!has_pending_wm) {
-                agg_counters_reset_kernel<<<1, 1, 0, stream>>>(t.counters);
-                HIP_TRY(hipStreamSynchronize(stream));
-                return check_overflow();
+            // EOWC barrier (hash_agg.rs:429-474): nothing is EMITTED until a
+            // watermark closes windows, but the dirty groups' current states
+            // still upsert into the intermediate table (mid-window PUTs,
+            // :429-460); state accumulates across epochs
+            agg_eowc_dump_kernel<<<2048, 256, 0, stream>>>(
+                t, KW, n_calls, (int)desc.row_count_index, cd(0), cd(1), cd(2),
+                cd(3));
+            HIP_TRY(hipStreamSynchronize(stream));
+            int rcd = check_overflow();
+            if (rcd != RW_OK) return rcd;
+            uint32_t ctrd[3];
+            HIP_TRY(hipMemcpy(ctrd, t.counters, 12, hipMemcpyDeviceToHost));
+            if (uint32_t n_put = ctrd[1]) {
+                std::vector<long long> vals((size_t)n_put * out_width);
+                std::vector<uint8_t> nulls((size_t)n_put * out_width);
+                HIP_TRY(hipMemcpy(vals.data(), t.out_vals, vals.size() * 8,
+                                  hipMemcpyDeviceToHost));
+                HIP_TRY(hipMemcpy(nulls.data(), t.out_nulls, nulls.size(),
+                                  hipMemcpyDeviceToHost));
+                std::vector<uint8_t> put_ops(n_put, RW_OP_INSERT);
+                spill_records(vals, nulls, put_ops, n_put);
             }
+            HIP_TRY(hipMemset(t.counters, 0, 12));
+            if (!has_pending_wm) return RW_OK;
             has_pending_wm = false;
             agg_eowc_close_kernel<<<2048, 256, 0, stream>>>(
                 t, KW, n_calls, (int)desc.row_count_index, pending_wm, cd(0),
@@ -1957,11 +1978,25 @@ struct HashAgg {
                            &nulls[(size_t)order[i] * out_width], out_width);
                     so[i] = ops[order[i]];
                 }
-                // spill: closed windows leave the state table (DELETEs);
-                // final rows spill through spill_records as PUT+DELETE —
-                // here the closed groups are gone, so record DELETEs only
+                // spill: every closed window leaves the state table (the
+                // reference deletes all intermediate rows < wm) — DELETE
+                // records for all rows, incl. the rc==0 marker rows, which
+                // are then filtered out of emission
                 spill_records_eowc(sv, sn, n_out);
-                slice_outputs(sv, sn, so, n_out);
+                uint32_t m = 0;
+                for (uint32_t i = 0; i < n_out; i++) {
+                    if (so[i] != RW_OP_INSERT) continue;
+                    if (m != i) {
+                        memcpy(&sv[(size_t)m * out_width],
+                               &sv[(size_t)i * out_width],
+                               (size_t)out_width * 8);
+                        memcpy(&sn[(size_t)m * out_width],
+                               &sn[(size_t)i * out_width], out_width);
+                    }
+                    so[m] = RW_OP_INSERT;
+                    m++;
+                }
+                slice_outputs(sv, sn, so, m);
             }
             HIP_TRY(hipMemset(t.counters, 0, 12));
             return RW_OK;
@@ -3388,6 +3423,76 @@ __global__ void agg_clean_kernel(AggTableDev t, int kpos, long long wm, int KW,
 // undisturbed; a late row would restart the window). Emission order is
 // fixed up host-side (rows sorted by group key, as SortBuffer::consume
 // iterates ordered).
+// EOWC mid-window state PUTs (hash_agg.rs:429-460): at every barrier the
+// dirty groups' CURRENT states upsert into the intermediate table even
+// though nothing is emitted until a watermark closes the window. One
+// thread per dirty slot; rows carry op=Insert (PUT) and are encoded
+// host-side by spill_records (materialized-input states encode None
+// there, as in the non-EOWC spill).
+__global__ void agg_eowc_dump_kernel(AggTableDev t, int KW, int n_calls,
+                                     int row_count_index, AggCallDev c0,
+                                     AggCallDev c1, AggCallDev c2,
+                                     AggCallDev c3) {
+    AggCallDev calls[4] = {c0, c1, c2, c3};
+    uint32_t n_dirty = t.counters[0];
+    uint32_t stride = gridDim.x * blockDim.x;
+    size_t cap = (size_t)t.cap_mask + 1;
+    int width = KW + n_calls;
+    for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n_dirty;
+         i += stride) {
+        uint32_t slot = t.dirty_list[i];
+        t.dirty_flag[slot] = 0;
+        uint32_t orow = atomicAdd(&t.counters[1], 1u);
+        if (orow + 1 > t.out_capacity) {
+            atomicExch(&t.counters[2], 2u);
+            continue;
+        }
+        long long rc = t.acc[(size_t)row_count_index * cap + slot];
+        if (rc < 0) rc = 0;
+        t.out_ops[orow] = RW_OP_INSERT;
+        for (int k = 0; k < KW; k++) {
+            t.out_vals[(size_t)orow * width + k] = t.keys[(size_t)slot * KW + k];
+            t.out_nulls[(size_t)orow * width + k] =
+                (t.key_nulls[slot] >> k) & 1;
+        }
+        for (int ci = 0; ci < n_calls; ci++) {
+            const AggCallDev& c = calls[ci];
+            long long* acc = t.acc + (size_t)ci * cap;
+            uint8_t* has = t.has + (size_t)ci * cap;
+            if (rc == 0 && !c.minput) {
+                // reset value states (agg_state.rs:149-155), as the flush
+                // kernel and the oracle's get_outputs do
+                switch (c.kind) {
+                    case RW_AGG_MIN: acc[slot] = INT64_MAX; break;
+                    case RW_AGG_MAX: acc[slot] = INT64_MIN; break;
+                    default: acc[slot] = 0;
+                }
+                has[slot] = 0;
+            }
+            long long v;
+            uint8_t nl;
+            if (c.minput) {
+                v = 0;
+                nl = 1; // spill_records encodes None for minput states
+            } else {
+                switch (c.kind) {
+                    case RW_AGG_COUNT_STAR:
+                    case RW_AGG_COUNT:
+                    case RW_AGG_SUM0:
+                        v = acc[slot];
+                        nl = 0;
+                        break;
+                    default:
+                        v = acc[slot];
+                        nl = !has[slot];
+                }
+            }
+            t.out_vals[(size_t)orow * width + KW + ci] = v;
+            t.out_nulls[(size_t)orow * width + KW + ci] = nl;
+        }
+    }
+}
+
 __global__ void agg_eowc_close_kernel(AggTableDev t, int KW, int n_calls,
                                       int row_count_index, long long wm,
                                       AggCallDev c0, AggCallDev c1,
@@ -3466,6 +3571,27 @@ __global__ void agg_eowc_close_kernel(AggTableDev t, int KW, int n_calls,
                 for (int ci = 0; ci < n_calls; ci++) {
                     t.out_vals[(size_t)orow * width + KW + ci] = curr[ci];
                     t.out_nulls[(size_t)orow * width + KW + ci] = curr_null[ci];
+                }
+            }
+        } else {
+            // closed window with row_count 0: nothing is emitted
+            // (OnlyOutputIfHasInput) but the reference still deletes its
+            // intermediate-table row — a DELETE-marker row, spilled but
+            // filtered out of emission host-side
+            uint32_t orow = atomicAdd(&t.counters[1], 1u);
+            if (orow + 1 > t.out_capacity) {
+                atomicExch(&t.counters[2], 2u);
+            } else {
+                t.out_ops[orow] = RW_OP_DELETE;
+                for (int k = 0; k < KW; k++) {
+                    t.out_vals[(size_t)orow * width + k] =
+                        t.keys[slot * KW + k];
+                    t.out_nulls[(size_t)orow * width + k] =
+                        (t.key_nulls[(uint32_t)slot] >> k) & 1;
+                }
+                for (int ci = 0; ci < n_calls; ci++) {
+                    t.out_vals[(size_t)orow * width + KW + ci] = 0;
+                    t.out_nulls[(size_t)orow * width + KW + ci] = 1;
                 }
             }
         }

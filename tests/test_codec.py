@@ -253,3 +253,35 @@ def test_dedup_spill_records():
     # epoch 3: untouched → empty drain
     assert dedup_drain(oracle(), o, 0) == []
     o.close()
+
+
+def test_eowc_spill_records():
+    # §8f-2 EOWC spill (hash_agg.rs:429-474): every barrier PUTs the dirty
+    # groups' current states even though nothing is emitted; a window close
+    # DELETEs every intermediate row below the watermark, including
+    # row_count-0 windows.
+    calls = [(AGG_COUNT_STAR, -1, T_I64)]
+    o = ffi.HashAgg(oracle(), [T_I64], [0], calls, 0,
+                    emit_on_window_close=True)
+    o.push(from_pretty(" I\n + 1\n + 2\n + 2"))
+    o.flush(1)
+    assert o.poll_all() == []  # EOWC: no emission without a watermark
+    recs = drain(oracle(), o)
+    assert recs == [
+        (1, memcmp_i64(1), value_i64(1) + value_i64(1)),
+        (1, memcmp_i64(2), value_i64(2) + value_i64(2)),
+    ], recs
+    # epoch 2: retract window 2 fully, then close windows < 3:
+    # mid-window PUT for the dirty (now rc=0) window 2, then close DELETEs
+    # for both windows — window 1 emits, window 2 does not
+    o.push(from_pretty(" I\n - 2\n - 2"))
+    o.watermark(0, 3)
+    o.flush(2)
+    o.poll_all()
+    recs = drain(oracle(), o)
+    assert recs == [
+        (1, memcmp_i64(2), value_i64(2) + value_i64(0)),
+        (0, memcmp_i64(1), b""),
+        (0, memcmp_i64(2), b""),
+    ], recs
+    o.close()

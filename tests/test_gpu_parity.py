@@ -996,6 +996,14 @@ def test_eowc_parity():
             a.flush(ep + 1)
             outs.append(ffi.rows_ordered(a.poll_all()))
         assert outs[0] == outs[1], f"epoch {ep}: {len(outs[0])} vs {len(outs[1])}"
+        # §8f-2 EOWC spill: mid-window state PUTs + window-close DELETEs —
+        # same record multiset per epoch (sorted; dirty-list order differs)
+        from test_codec import drain
+        sg = sorted(drain(gpu(), g))
+        so = sorted(drain(ffi.oracle(), o))
+        assert sg == so, (f"epoch {ep}: eowc spill {len(sg)} vs {len(so)}; "
+                          f"first diff: "
+                          f"{next(((a, b) for a, b in zip(sg, so) if a != b), None)}")
     g.close()
     o.close()
 
