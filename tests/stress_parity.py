@@ -60,6 +60,11 @@ def stress_agg(seed, n):
             a.flush(ep + 1)
             outs.append(rows_multiset(a.poll_all()))
         assert outs[0] == outs[1], f"agg seed {seed} epoch {ep}"
+        from test_codec import drain
+
+        sg = sorted(drain(GPU, g))
+        so = sorted(drain(oracle(), o))
+        assert sg == so, f"agg spill seed {seed} epoch {ep}"
     g.close()
     o.close()
 
@@ -154,6 +159,11 @@ def stress_eowc(seed, n):
             a.flush(ep + 1)
             outs.append(ffi.rows_ordered(a.poll_all()))
         assert outs[0] == outs[1], f"eowc seed {seed} epoch {ep}"
+        from test_codec import drain
+
+        sg = sorted(drain(GPU, g))
+        so = sorted(drain(oracle(), o))
+        assert sg == so, f"eowc spill seed {seed} epoch {ep}"
     g.close()
     o.close()
 
@@ -185,6 +195,11 @@ def stress_distinct(seed, n):
             a.flush(ep + 1)
             outs.append(rows_multiset(a.poll_all()))
         assert outs[0] == outs[1], f"distinct seed {seed} epoch {ep}"
+        from test_codec import dedup_drain
+
+        dg = dedup_drain(GPU, g, 0)
+        do = dedup_drain(oracle(), o, 0)
+        assert dg == do, f"dedup spill seed {seed} epoch {ep}"
     g.close()
     o.close()
 
