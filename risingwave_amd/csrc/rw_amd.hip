@@ -3450,6 +3450,10 @@ __global__ void agg_eowc_dump_kernel(AggTableDev t, int KW, int n_calls,
         long long rc = t.acc[(size_t)row_count_index * cap + slot];
         if (rc < 0) rc = 0;
         t.out_ops[orow] = RW_OP_INSERT;
+        // the window now has a persisted intermediate-table row; the close
+        // kernel emits its DELETE marker only for persisted windows (and
+        // clears the flag), so re-scanned already-closed slots stay silent
+        t.has_prev[slot] = 1;
         for (int k = 0; k < KW; k++) {
             t.out_vals[(size_t)orow * width + k] = t.keys[(size_t)slot * KW + k];
             t.out_nulls[(size_t)orow * width + k] =
@@ -3573,11 +3577,12 @@ __global__ void agg_eowc_close_kernel(AggTableDev t, int KW, int n_calls,
                     t.out_nulls[(size_t)orow * width + KW + ci] = curr_null[ci];
                 }
             }
-        } else {
-            // closed window with row_count 0: nothing is emitted
-            // (OnlyOutputIfHasInput) but the reference still deletes its
-            // intermediate-table row — a DELETE-marker row, spilled but
-            // filtered out of emission host-side
+        } else if (t.has_prev[(uint32_t)slot]) {
+            // closed window with row_count 0 but a persisted row: nothing
+            // is emitted (OnlyOutputIfHasInput) but the reference still
+            // deletes its intermediate-table row — a DELETE-marker row,
+            // spilled but filtered out of emission host-side. Slots with
+            // no persisted row (already-closed residue) stay silent.
             uint32_t orow = atomicAdd(&t.counters[1], 1u);
             if (orow + 1 > t.out_capacity) {
                 atomicExch(&t.counters[2], 2u);
