@@ -255,6 +255,13 @@ int rw_topn_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len);
  * rw_agg_n_dedup_tables returns how many exist. */
 int rw_agg_n_dedup_tables(void* h);
 int rw_agg_dedup_drain(void* h, int di, uint8_t** buf, uint64_t* len);
+/* Join degree tables (join/row.rs:99-113 build_degree_row): per side
+ * needing degrees (join/mod.rs:153-165), pk = jk ∥ pk as the main table,
+ * value = order key ++ degree i64. The deltas are computed by
+ * rw_join_checkpoint_drain for that side (rows inserted/killed this epoch
+ * plus pre-epoch rows whose degree changed during probes) — call it first,
+ * then this to collect the degree records. */
+int rw_join_degree_drain(void* h, int side, uint8_t** buf, uint64_t* len);
 void rw_spill_free(uint8_t* buf);
 
 #ifdef __cplusplus

@@ -22,6 +22,7 @@
 //    overwrite (non-strict consistency mode, src/stream/src/consistency.rs).
 #include <algorithm>
 #include <map>
+#include <optional>
 #include <memory>
 #include <string>
 #include <vector>
@@ -251,6 +252,9 @@ struct HashJoinOracle {
                             if (match_side.need_deg) entry.degree -= 1;
                             if (!forward_exactly_once(T, S)) with_match(b, op, row, entry);
                         }
+                        if (match_side.need_deg)
+                            deg_touched[1 - S][enc_key(1 - S, key, mpk)] = {key,
+                                                                            mpk};
                     }
                     if (append_only) {
                         // hash_join.rs:1359-1364: jk ⊇ pk ⇒ at most one match
@@ -343,8 +347,13 @@ struct HashJoinOracle {
     struct DeltaEnt {
         int st; // 0 = DEL, 1 = PUT (fresh), 2 = PUT (over pre-epoch row)
         std::vector<uint8_t> v;
+        Row jk, pk; // for the degree-table drain's current-degree lookup
     };
     std::map<std::string, DeltaEnt> delta[2];
+    // §8f-2 degree-table deltas: pre-epoch rows whose degree changed this
+    // epoch via probes (key → (jk, pk) for lookup at drain)
+    std::map<std::string, std::pair<Row, Row>> deg_touched[2];
+    std::vector<uint8_t> deg_spill[2];
 
     std::string enc_key(int s, const Row& key, const Row& pk) {
         std::vector<uint8_t> kb;
@@ -370,9 +379,9 @@ struct HashJoinOracle {
         std::string k = enc_key(s, key, pk);
         auto it = delta[s].find(k);
         if (it != delta[s].end() && it->second.st == 0)
-            it->second = {2, std::move(v)}; // PUT over a pre-epoch DEL
+            it->second = {2, std::move(v), key, pk}; // PUT over a pre-epoch DEL
         else
-            delta[s][k] = {1, std::move(v)};
+            delta[s][k] = {1, std::move(v), key, pk};
     }
 
     void delta_delete(int s, const Row& key, const Row& pk) {
@@ -398,6 +407,69 @@ struct HashJoinOracle {
             put32(e.st ? (uint32_t)e.v.size() : 0);
             if (e.st) sp.insert(sp.end(), e.v.begin(), e.v.end());
         }
+        // §8f-2 degree-table deltas (build_degree_row, join/row.rs:99-113):
+        // keys mirror the main delta (a degree row exists iff its state row
+        // does) plus pre-epoch rows whose degree changed during probes;
+        // PUT value = order key (jk ∥ pk) ++ current degree i64.
+        if (side[s].need_deg) {
+            auto deg_val = [&](const Row& jk, const Row& pk, uint64_t deg,
+                               std::vector<uint8_t>& v) {
+                for (size_t i = 0; i < jk.size(); i++) {
+                    uint8_t t = side[s].types[side[s].key_idx[i]];
+                    rwcodec::DatumC d{jk[i].null, jk[i].i, jk[i].d};
+                    rwcodec::value_encode_datum(v, t, d);
+                }
+                for (size_t i = 0; i < pk.size(); i++) {
+                    uint8_t t = side[s].types[side[s].pk_idx[i]];
+                    rwcodec::DatumC d{pk[i].null, pk[i].i, pk[i].d};
+                    rwcodec::value_encode_datum(v, t, d);
+                }
+                rwcodec::value_encode_datum(v, RW_T_I64,
+                                            {false, (long long)deg, 0});
+            };
+            auto lookup_deg = [&](const Row& jk, const Row& pk,
+                                  uint64_t& deg) {
+                auto ti = side[s].table.find(jk);
+                if (ti == side[s].table.end()) return false;
+                auto ei = ti->second.find(pk);
+                if (ei == ti->second.end()) return false;
+                deg = ei->second.degree;
+                return true;
+            };
+            std::map<std::string, std::optional<std::vector<uint8_t>>> dd;
+            for (auto& [k, e] : delta[s]) {
+                if (e.st == 0) {
+                    dd[k] = std::nullopt;
+                } else {
+                    uint64_t deg = 0;
+                    lookup_deg(e.jk, e.pk, deg); // st>0 ⇒ live
+                    std::vector<uint8_t> v;
+                    deg_val(e.jk, e.pk, deg, v);
+                    dd[k] = std::move(v);
+                }
+            }
+            for (auto& [k, jp] : deg_touched[s]) {
+                if (dd.count(k)) continue; // covered by the main delta
+                uint64_t deg = 0;
+                if (!lookup_deg(jp.first, jp.second, deg)) continue;
+                std::vector<uint8_t> v;
+                deg_val(jp.first, jp.second, deg, v);
+                dd[k] = std::move(v);
+            }
+            auto& dsp = deg_spill[s];
+            auto dput32 = [&](uint32_t x) {
+                for (int b = 0; b < 4; b++)
+                    dsp.push_back((uint8_t)(x >> (8 * b)));
+            };
+            for (auto& [k, v] : dd) {
+                dsp.push_back(v.has_value() ? 1 : 0);
+                dput32((uint32_t)k.size());
+                dsp.insert(dsp.end(), k.begin(), k.end());
+                dput32(v ? (uint32_t)v->size() : 0);
+                if (v) dsp.insert(dsp.end(), v->begin(), v->end());
+            }
+        }
+        deg_touched[s].clear();
         delta[s].clear();
     }
 
@@ -493,6 +565,16 @@ int rw_join_checkpoint_drain(void* h, int side, uint8_t** buf,
     *len = sp.size();
     *buf = (uint8_t*)malloc(sp.size() ? sp.size() : 1);
     memcpy(*buf, sp.data(), sp.size());
+    return RW_OK;
+}
+
+int rw_join_degree_drain(void* h, int side, uint8_t** buf, uint64_t* len) {
+    if (side != 0 && side != 1) return RW_E_INVAL;
+    auto& sp = ((HashJoinOracle*)h)->deg_spill[side];
+    *len = sp.size();
+    *buf = (uint8_t*)malloc(sp.size() ? sp.size() : 1);
+    memcpy(*buf, sp.data(), sp.size());
+    sp.clear();
     return RW_OK;
 }
 

@@ -233,6 +233,11 @@ struct JoinSideDev {
     uint32_t* killed;
     uint32_t* killed_cursor;
     uint32_t killed_cap;
+    // degree-table delta tracking (§8f-2): pre-epoch rows whose degree
+    // changed this epoch (null when the join type needs no degree here)
+    uint32_t* deg_dirty_flag;
+    uint32_t* deg_dirty_list;
+    uint32_t* deg_dirty_n;
 };
 
 __device__ __forceinline__ JoinRowHdr* jrow(const JoinSideDev& s, uint32_t r) {
@@ -3150,6 +3155,13 @@ __device__ void join_probe_row_noninner(const JoinBatchDev& b, JoinSideDev own,
                     uint32_t old = atomicAdd(&h->degree,
                                              is_insert ? 1u : (uint32_t)-1);
                     zero = is_insert ? old == 0 : old == 1;
+                    // §8f-2 degree-table delta: record the touched row once
+                    if (match.deg_dirty_flag &&
+                        ld_u32(&match.deg_dirty_flag[row]) == 0 &&
+                        atomicCAS(&match.deg_dirty_flag[row], 0u, 1u) == 0u) {
+                        uint32_t di = atomicAdd(match.deg_dirty_n, 1u);
+                        match.deg_dirty_list[di] = row;
+                    }
                 }
                 const long long* mv = jvals(h);
                 uint32_t vb = h->validbits;
@@ -3846,6 +3858,16 @@ struct HashJoin {
             HIP_TRY(hipMalloc(&js.killed, (size_t)js.killed_cap * 4));
             HIP_TRY(hipMalloc(&js.killed_cursor, 4));
             HIP_TRY(hipMemset(js.killed_cursor, 0, 4));
+            js.deg_dirty_flag = nullptr;
+            js.deg_dirty_list = nullptr;
+            js.deg_dirty_n = nullptr;
+            if (m.need_deg[s]) {
+                HIP_TRY(hipMalloc(&js.deg_dirty_flag, (size_t)row_cap * 4));
+                HIP_TRY(hipMemset(js.deg_dirty_flag, 0, (size_t)row_cap * 4));
+                HIP_TRY(hipMalloc(&js.deg_dirty_list, (size_t)row_cap * 4));
+                HIP_TRY(hipMalloc(&js.deg_dirty_n, 4));
+                HIP_TRY(hipMemset(js.deg_dirty_n, 0, 4));
+            }
         }
         HIP_TRY(hipStreamSynchronize(stream));
         out.cap = 1u << 22;
@@ -4031,10 +4053,84 @@ struct HashJoin {
             put32(v ? (uint32_t)v->size() : 0);
             if (v) sp.insert(sp.end(), v->begin(), v->end());
         }
+        // §8f-2 degree-table deltas (build_degree_row, join/row.rs:99-113:
+        // pk = jk ∥ pk as the main table, value = order key ++ degree i64).
+        // Keys mirror the main delta (a degree row exists iff its state row
+        // does), plus pre-epoch rows whose degree changed during probes
+        // (deg_dirty list). Serialized into deg_spill[s], returned by
+        // rw_join_degree_drain after this drain.
+        if (m.need_deg[s]) {
+            auto encode_deg_val = [&](const uint8_t* rec,
+                                      std::vector<uint8_t>& v) {
+                const uint32_t vb = ((const uint32_t*)rec)[2];
+                const int64_t* vals = (const int64_t*)(rec + 16);
+                auto put_datum = [&](uint8_t col) {
+                    rwcodec::DatumC d{!((vb >> col) & 1), vals[col], 0};
+                    rwcodec::value_encode_datum(v, types[s][col], d);
+                };
+                for (int i = 0; i < m.KW; i++) put_datum(m.key_cols[s][i]);
+                for (int i = 0; i < m.n_pk[s]; i++) put_datum(m.pk_cols[s][i]);
+                uint32_t deg = ((const uint32_t*)rec)[3];
+                rwcodec::value_encode_datum(v, RW_T_I64,
+                                            {false, (long long)deg, 0});
+            };
+            std::map<std::string, std::optional<std::vector<uint8_t>>> dd;
+            for (uint32_t i = 0; i < kcur; i++) {
+                if (kills[i] >= mark) continue;
+                HIP_TRY(hipMemcpy(oldrec.data(),
+                                  js.rows + (size_t)kills[i] * stride, stride,
+                                  hipMemcpyDeviceToHost));
+                std::string k;
+                encode_key(oldrec.data(), k);
+                dd[k] = std::nullopt;
+            }
+            for (uint32_t i = mark; i < cur; i++) {
+                const uint8_t* rec = fresh.data() + (size_t)(i - mark) * stride;
+                if (!((const uint32_t*)rec)[0]) continue;
+                std::string k;
+                encode_key(rec, k);
+                std::vector<uint8_t> v;
+                encode_deg_val(rec, v);
+                dd[k] = std::move(v);
+            }
+            uint32_t dn = 0;
+            HIP_TRY(hipMemcpy(&dn, js.deg_dirty_n, 4, hipMemcpyDeviceToHost));
+            std::vector<uint32_t> dl(dn);
+            if (dn)
+                HIP_TRY(hipMemcpy(dl.data(), js.deg_dirty_list, (size_t)dn * 4,
+                                  hipMemcpyDeviceToHost));
+            HIP_TRY(hipMemset(js.deg_dirty_flag, 0, (size_t)js.row_cap * 4));
+            HIP_TRY(hipMemset(js.deg_dirty_n, 0, 4));
+            for (uint32_t i = 0; i < dn; i++) {
+                uint32_t r = dl[i];
+                if (r >= mark) continue; // fresh rows already covered
+                HIP_TRY(hipMemcpy(oldrec.data(), js.rows + (size_t)r * stride,
+                                  stride, hipMemcpyDeviceToHost));
+                if (!((const uint32_t*)oldrec.data())[0]) continue; // killed
+                std::string k;
+                encode_key(oldrec.data(), k);
+                std::vector<uint8_t> v;
+                encode_deg_val(oldrec.data(), v);
+                dd[k] = std::move(v);
+            }
+            auto& dsp = deg_spill[s];
+            auto dput32 = [&](uint32_t x) {
+                for (int b = 0; b < 4; b++)
+                    dsp.push_back((uint8_t)(x >> (8 * b)));
+            };
+            for (auto& [k, v] : dd) {
+                dsp.push_back(v.has_value() ? 1 : 0);
+                dput32((uint32_t)k.size());
+                dsp.insert(dsp.end(), k.begin(), k.end());
+                dput32(v ? (uint32_t)v->size() : 0);
+                if (v) dsp.insert(dsp.end(), v->begin(), v->end());
+            }
+        }
         HIP_TRY(hipMemset(js.killed_cursor, 0, 4));
         flush_mark[s] = cur;
         return RW_OK;
     }
+    std::vector<uint8_t> deg_spill[2];
 
     int drain_output() {
         uint32_t ctr[2];
@@ -4498,6 +4594,17 @@ int rw_join_checkpoint_drain(void* h, int side, uint8_t** buf,
     *len = sp.size();
     *buf = (uint8_t*)malloc(sp.size() ? sp.size() : 1);
     memcpy(*buf, sp.data(), sp.size());
+    return RW_OK;
+}
+
+int rw_join_degree_drain(void* h, int side, uint8_t** buf, uint64_t* len) {
+    auto* j = (HashJoin*)h;
+    if (side != 0 && side != 1) FAIL(RW_E_INVAL, "bad side");
+    auto& sp = j->deg_spill[side];
+    *len = sp.size();
+    *buf = (uint8_t*)malloc(sp.size() ? sp.size() : 1);
+    memcpy(*buf, sp.data(), sp.size());
+    sp.clear();
     return RW_OK;
 }
 

@@ -570,6 +570,25 @@ class GroupTopN:
             pass
 
 
+def join_degree_drain(lib, h, side):
+    """Collect one side's degree-table records (computed by the preceding
+    join_checkpoint_drain call for that side); returns bytes."""
+    L = lib.lib
+    L.rw_join_degree_drain.restype = C.c_int
+    L.rw_join_degree_drain.argtypes = [C.c_void_p, C.c_int,
+                                       C.POINTER(C.POINTER(C.c_uint8)),
+                                       C.POINTER(C.c_uint64)]
+    buf = C.POINTER(C.c_uint8)()
+    ln = C.c_uint64()
+    rc = L.rw_join_degree_drain(h, side, C.byref(buf), C.byref(ln))
+    if rc != 0:
+        raise RuntimeError(f"degree drain failed {rc}: {lib.last_error()}")
+    out = bytes(bytearray(buf[i] for i in range(ln.value)))
+    L.rw_spill_free.argtypes = [C.c_void_p]
+    L.rw_spill_free(C.cast(buf, C.c_void_p))
+    return out
+
+
 def join_checkpoint_drain(lib, h, side):
     """Drain one side's §8f-2 checkpoint spill buffer; returns bytes."""
     L = lib.lib

@@ -101,6 +101,9 @@ def stress_join(seed, jt, n):
         sg = ffi.join_checkpoint_drain(GPU, g.h, side)
         so = ffi.join_checkpoint_drain(oracle(), o.h, side)
         assert sg == so, f"spill type {jt} seed {seed} push {i}"
+        dg = ffi.join_degree_drain(GPU, g.h, side)
+        do = ffi.join_degree_drain(oracle(), o.h, side)
+        assert dg == do, f"degree spill type {jt} seed {seed} push {i}"
     g.close()
     o.close()
 

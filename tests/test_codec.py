@@ -285,3 +285,52 @@ def test_eowc_spill_records():
         (0, memcmp_i64(2), b""),
     ], recs
     o.close()
+
+
+def test_join_degree_spill_bytes():
+    # §8f-2 degree-table drain (join/row.rs:99-113 build_degree_row): pk =
+    # jk ∥ pk as the main table, value = order key ++ degree i64. Deltas =
+    # main-table keys (insert/delete) plus pre-epoch rows whose degree
+    # changed during probes of the other side.
+    from rwtest.ffi import JOIN_LEFT_SEMI, SIDE_LEFT, SIDE_RIGHT
+
+    o = ffi.HashJoin(oracle(), JOIN_LEFT_SEMI, [T_I64, T_I64],
+                     [T_I64, T_I64], key_l=[0], key_r=[0], pk_l=[1],
+                     pk_r=[1])
+    k = memcmp_i64(1) + memcmp_i64(10)
+
+    def deg_rec(put, deg=None):
+        import struct
+        if put:
+            v = value_i64(1) + value_i64(10) + value_i64(deg)
+            return (b"\x01" + struct.pack("<I", len(k)) + k +
+                    struct.pack("<I", len(v)) + v)
+        return b"\x00" + struct.pack("<I", len(k)) + k + struct.pack("<I", 0)
+
+    # epoch 1: left row (1, 10), no right matches -> degree 0 PUT
+    o.push(SIDE_LEFT, from_pretty(" I I\n + 1 10"))
+    o.poll_all()
+    ffi.join_checkpoint_drain(oracle(), o.h, SIDE_LEFT)
+    assert ffi.join_degree_drain(oracle(), o.h, SIDE_LEFT) == deg_rec(1, 0)
+    # epoch 2: right insert matches -> pre-epoch left row degree 1
+    o.push(SIDE_RIGHT, from_pretty(" I I\n + 1 99"))
+    o.poll_all()
+    ffi.join_checkpoint_drain(oracle(), o.h, SIDE_LEFT)
+    assert ffi.join_degree_drain(oracle(), o.h, SIDE_LEFT) == deg_rec(1, 1)
+    # right side of a LEFT SEMI needs no degree table -> empty
+    ffi.join_checkpoint_drain(oracle(), o.h, SIDE_RIGHT)
+    assert ffi.join_degree_drain(oracle(), o.h, SIDE_RIGHT) == b""
+    # epoch 3: right delete -> degree back to 0
+    o.push(SIDE_RIGHT, from_pretty(" I I\n - 1 99"))
+    o.poll_all()
+    ffi.join_checkpoint_drain(oracle(), o.h, SIDE_LEFT)
+    assert ffi.join_degree_drain(oracle(), o.h, SIDE_LEFT) == deg_rec(1, 0)
+    # epoch 4: left delete -> degree-row DELETE
+    o.push(SIDE_LEFT, from_pretty(" I I\n - 1 10"))
+    o.poll_all()
+    ffi.join_checkpoint_drain(oracle(), o.h, SIDE_LEFT)
+    assert ffi.join_degree_drain(oracle(), o.h, SIDE_LEFT) == deg_rec(0)
+    # epoch 5: untouched -> empty
+    ffi.join_checkpoint_drain(oracle(), o.h, SIDE_LEFT)
+    assert ffi.join_degree_drain(oracle(), o.h, SIDE_LEFT) == b""
+    o.close()

@@ -841,6 +841,50 @@ def test_join_checkpoint_spill_parity():
     o.close()
 
 
+def test_join_degree_spill_parity():
+    # §8f-2 degree-table spill: GPU vs oracle byte-identical per epoch for
+    # degree-carrying join types (main-delta keys + pre-epoch rows whose
+    # degree changed during probes)
+    from rwtest.ffi import JOIN_FULL_OUTER, JOIN_LEFT_OUTER, JOIN_LEFT_SEMI
+
+    for jt in (JOIN_LEFT_OUTER, JOIN_FULL_OUTER, JOIN_LEFT_SEMI):
+        rng = np.random.default_rng(77 + jt)
+        kw = dict(key_l=[0], key_r=[0], pk_l=[1], pk_r=[1])
+        g = ffi.HashJoin(gpu(), jt, [T_I64, T_I64], [T_I64, T_I64], **kw)
+        o = ffi.HashJoin(oracle(), jt, [T_I64, T_I64], [T_I64, T_I64], **kw)
+        live = {SIDE_LEFT: [], SIDE_RIGHT: []}
+        pk = 0
+        for epoch in range(4):
+            for side in (SIDE_LEFT, SIDE_RIGHT):
+                n = 768
+                keys = rng.integers(0, 60, n)  # dense keys: many matches
+                vals = np.arange(pk, pk + n)
+                pk += n
+                ops = np.zeros(n, np.uint8)
+                for r in range(n):
+                    if live[side] and rng.random() < 0.3:
+                        jx = int(rng.integers(0, len(live[side])))
+                        keys[r], vals[r] = live[side].pop(jx)
+                        ops[r] = ffi.OP_DELETE
+                    else:
+                        live[side].append((int(keys[r]), int(vals[r])))
+                c = mk_chunk([T_I64, T_I64], ops, [keys, vals])
+                g.push(side, c)
+                o.push(side, c)
+                g.poll_all()
+                o.poll_all()
+            for side in (SIDE_LEFT, SIDE_RIGHT):
+                mg = ffi.join_checkpoint_drain(gpu(), g.h, side)
+                mo = ffi.join_checkpoint_drain(ffi.oracle(), o.h, side)
+                assert mg == mo, f"jt {jt} epoch {epoch} side {side}: main"
+                dg = ffi.join_degree_drain(gpu(), g.h, side)
+                do = ffi.join_degree_drain(ffi.oracle(), o.h, side)
+                assert dg == do, (f"jt {jt} epoch {epoch} side {side}: "
+                                  f"degree {len(dg)} vs {len(do)} bytes")
+        g.close()
+        o.close()
+
+
 def test_count_distinct_parity():
     # DISTINCT dedup (aggregate/distinct.rs): GPU vs oracle — hand case +
     # randomized duplicate-heavy insert/delete mix
