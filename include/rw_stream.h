@@ -240,9 +240,9 @@ void rw_group_top_n_destroy(void* h);
  * StateTable::commit would hand the state store. Record framing:
  * [put u8][klen u32 LE][key][vlen u32 LE][value], emitted in memcmp key
  * order. Agg: intermediate-state table (group key → outputs). Join: per
- * side, pk = join key ∥ deduped input pk, value = full row; degree tables
- * are not spilled (cleanup flows through watermark hints). Caller frees
- * with rw_spill_free. */
+ * side, pk = join key ∥ deduped input pk, value = full row (degree
+ * tables via rw_join_degree_drain below). Caller frees with
+ * rw_spill_free. */
 int rw_agg_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len);
 int rw_join_checkpoint_drain(void* h, int side, uint8_t** buf, uint64_t* len);
 int rw_topn_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len);

@@ -343,7 +343,8 @@ struct HashJoinOracle {
 
     // §8f-2 checkpoint spill deltas per side: key = memcmp(jk ∥ pk),
     // tri-state net per key (DEL / fresh PUT / PUT over a pre-epoch row) —
-    // mirrors the GPU's kill-list netting. Degree tables are not spilled.
+    // mirrors the GPU's kill-list netting. Degree tables spill via
+    // rw_join_degree_drain (computed in checkpoint_drain below).
     struct DeltaEnt {
         int st; // 0 = DEL, 1 = PUT (fresh), 2 = PUT (over pre-epoch row)
         std::vector<uint8_t> v;

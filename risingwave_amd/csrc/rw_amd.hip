@@ -3982,8 +3982,9 @@ struct HashJoin {
     // memcmp key order). Deltas: rows appended since the last drain that
     // are still alive → PUT; rows killed this epoch that predate the last
     // drain → DELETE; kill-after-insert within the epoch nets away (the
-    // record is dead, its PUT is skipped). Degree tables and watermark-TTL
-    // cleanup are not spilled (the reference cleans via watermark hints).
+    // record is dead, its PUT is skipped). Degree tables spill too (the
+    // block below, drained via rw_join_degree_drain); watermark-TTL
+    // cleanup is not spilled (the reference cleans via watermark hints).
     uint32_t flush_mark[2] = {0, 0};
     int checkpoint_drain(int s, std::vector<uint8_t>& sp) {
         HIP_TRY(hipStreamSynchronize(stream));
