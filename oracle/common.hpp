@@ -13,6 +13,7 @@
 //  - StreamChunk::eliminate_adjacent_noop_update
 //    (common/src/array/stream_chunk.rs:331-384).
 #pragma once
+#include <algorithm>
 #include <cassert>
 #include <cmath>
 #include <cstdint>
@@ -383,6 +384,48 @@ inline void chunk_free_c(RwChunk* ch) {
     delete[] ch->ops;
     delete[] ch->vis;
     delete ch;
+}
+
+} // namespace orc
+
+namespace orc {
+
+// Re-emit a spill stream ([put u8][klen u32 LE][key][vlen u32 LE][val]
+// frames) in memcomparable-key order (stable: same-key PUT/DELETE within an
+// epoch keep their relative order) — the canonical drain order both the GPU
+// library and this oracle emit, enabling byte-compare parity.
+inline void sort_spill_frames(std::vector<uint8_t>& sp) {
+    struct Frame { const uint8_t* p; size_t n; };
+    std::vector<Frame> frames;
+    size_t off = 0;
+    auto rd32 = [&](size_t o) {
+        return (uint32_t)sp[o] | ((uint32_t)sp[o + 1] << 8) |
+               ((uint32_t)sp[o + 2] << 16) | ((uint32_t)sp[o + 3] << 24);
+    };
+    while (off + 5 <= sp.size()) {
+        size_t start = off;
+        uint32_t klen = rd32(off + 1);
+        off += 5 + klen;
+        if (off + 4 > sp.size()) return; // malformed: leave unsorted
+        uint32_t vlen = rd32(off);
+        off += 4 + vlen;
+        if (off > sp.size()) return;
+        frames.push_back({sp.data() + start, off - start});
+    }
+    std::stable_sort(frames.begin(), frames.end(),
+                     [](const Frame& a, const Frame& b) {
+        uint32_t ka = (uint32_t)a.p[1] | ((uint32_t)a.p[2] << 8) |
+                      ((uint32_t)a.p[3] << 16) | ((uint32_t)a.p[4] << 24);
+        uint32_t kb = (uint32_t)b.p[1] | ((uint32_t)b.p[2] << 8) |
+                      ((uint32_t)b.p[3] << 16) | ((uint32_t)b.p[4] << 24);
+        int c = memcmp(a.p + 5, b.p + 5, ka < kb ? ka : kb);
+        if (c) return c < 0;
+        return ka < kb;
+    });
+    std::vector<uint8_t> out;
+    out.reserve(sp.size());
+    for (auto& f : frames) out.insert(out.end(), f.p, f.p + f.n);
+    sp.swap(out);
 }
 
 } // namespace orc

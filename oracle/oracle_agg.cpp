@@ -615,6 +615,8 @@ int rw_hash_agg_update_vnode_bitmap(void* h, const uint8_t* bitmap,
 // drain the checkpoint spill buffer (caller frees with rw_spill_free)
 int rw_agg_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len) {
     auto* a = (HashAggOracle*)h;
+    // memcmp-key order (canonical drain form; matches the GPU library)
+    orc::sort_spill_frames(a->spill);
     *len = a->spill.size();
     *buf = (uint8_t*)malloc(a->spill.size() ? a->spill.size() : 1);
     memcpy(*buf, a->spill.data(), a->spill.size());

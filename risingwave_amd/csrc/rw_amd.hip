@@ -1910,7 +1910,39 @@ struct HashAgg {
         return RW_OK;
     }
 
+    // Grow the flush output buffers to hold `need` rows. Called on the sync
+    // flush paths BEFORE launching the flush kernel (the kernel clears dirty
+    // flags and updates prev as it emits, so an overflowed launch is not
+    // retryable): large dirty sets — e.g. q3's 512k-order windows × 16-step
+    // epochs — emit up to 2 rows per dirty group.
+    int ensure_out_capacity(uint64_t need) {
+        if (need <= t.out_capacity) return RW_OK;
+        uint64_t cap = t.out_capacity;
+        while (cap < need) cap <<= 1;
+        if (cap > UINT32_MAX) FAIL(RW_E_INVAL, "flush output > 4G rows");
+        hipFree(t.out_vals);
+        hipFree(t.out_nulls);
+        hipFree(t.out_ops);
+        t.out_vals = nullptr; t.out_nulls = nullptr; t.out_ops = nullptr;
+        HIP_TRY(hipMalloc(&t.out_vals, (size_t)cap * out_width * 8));
+        HIP_TRY(hipMalloc(&t.out_nulls, (size_t)cap * out_width));
+        HIP_TRY(hipMalloc(&t.out_ops, cap));
+        t.out_capacity = (uint32_t)cap;
+        return RW_OK;
+    }
+
+    // sync the stream, read the dirty count, and size the output buffers for
+    // the worst case (2 rows per dirty group) ahead of the flush launch
+    int presize_flush_out() {
+        HIP_TRY(hipStreamSynchronize(stream));
+        uint32_t nd = 0;
+        HIP_TRY(hipMemcpy(&nd, t.counters, 4, hipMemcpyDeviceToHost));
+        return ensure_out_capacity((uint64_t)nd * 2);
+    }
+
     int flush(uint64_t) {
+        int rcp = presize_flush_out();
+        if (rcp != RW_OK) return rcp;
         if (eowc) {
             // EOWC barrier (hash_agg.rs:429-474): nothing is EMITTED until a
             // watermark closes windows, but the dirty groups' current states
@@ -2143,7 +2175,10 @@ struct HashAgg {
             const DedupDirtyRec& r = recs[i];
             uint32_t slot = slots_h[i];
             Rec e;
-            if (r.count > 0) {
+            // signed view: a retract of a never-inserted datum wraps the u32
+            // below 0 — match the oracle's signed-count semantics (the
+            // reference's count is i64; inconsistent input tolerance)
+            if ((int32_t)r.count > 0) {
                 e.put = 1;
                 persisted[slot] = 1;
             } else {
@@ -2277,6 +2312,56 @@ struct HashAgg {
 // ---------------------------------------------------------------------------
 // C ABI
 // ---------------------------------------------------------------------------
+
+// Shared checked export for the drain functions: copies a host spill buffer
+// into a malloc'd block the caller frees with rw_spill_free. Fails loudly on
+// allocation failure instead of crashing in memcpy.
+static int spill_export(const std::vector<uint8_t>& sp, uint8_t** buf,
+                        uint64_t* len) {
+    *len = sp.size();
+    *buf = (uint8_t*)malloc(sp.size() ? sp.size() : 1);
+    if (!*buf) FAIL(RW_E_INTERNAL, "spill export: host alloc of %zu bytes failed",
+                    sp.size());
+    memcpy(*buf, sp.data(), sp.size());
+    return RW_OK;
+}
+
+// Re-emit a spill stream ([put u8][klen u32 LE][key][vlen u32 LE][val] frames)
+// in memcomparable-key order. stable: a PUT and DELETE of the same key within
+// one epoch (EOWC mid-window PUT then window-close DELETE) keep their order.
+static void sort_spill_frames(std::vector<uint8_t>& sp) {
+    struct Frame { const uint8_t* p; size_t n; };
+    std::vector<Frame> frames;
+    size_t off = 0;
+    auto rd32 = [&](size_t o) {
+        return (uint32_t)sp[o] | ((uint32_t)sp[o + 1] << 8) |
+               ((uint32_t)sp[o + 2] << 16) | ((uint32_t)sp[o + 3] << 24);
+    };
+    while (off + 5 <= sp.size()) {
+        size_t start = off;
+        uint32_t klen = rd32(off + 1);
+        off += 5 + klen;
+        if (off + 4 > sp.size()) return; // malformed: leave unsorted
+        uint32_t vlen = rd32(off);
+        off += 4 + vlen;
+        if (off > sp.size()) return;
+        frames.push_back({sp.data() + start, off - start});
+    }
+    std::stable_sort(frames.begin(), frames.end(), [](const Frame& a,
+                                                      const Frame& b) {
+        uint32_t ka = (uint32_t)a.p[1] | ((uint32_t)a.p[2] << 8) |
+                      ((uint32_t)a.p[3] << 16) | ((uint32_t)a.p[4] << 24);
+        uint32_t kb = (uint32_t)b.p[1] | ((uint32_t)b.p[2] << 8) |
+                      ((uint32_t)b.p[3] << 16) | ((uint32_t)b.p[4] << 24);
+        int c = memcmp(a.p + 5, b.p + 5, ka < kb ? ka : kb);
+        if (c) return c < 0;
+        return ka < kb;
+    });
+    std::vector<uint8_t> out;
+    out.reserve(sp.size());
+    for (auto& f : frames) out.insert(out.end(), f.p, f.p + f.n);
+    sp.swap(out);
+}
 
 extern "C" {
 
@@ -2464,6 +2549,7 @@ int rw_agg_apply_payload(void* h, const uint8_t* payload,
 long long rw_agg_flush_device(void* h, uint64_t epoch) {
     auto* agg = (HashAgg*)h;
     (void)epoch;
+    if (agg->presize_flush_out() != RW_OK) return -1;
     agg_flush_kernel<<<2048, 256, 0, agg->stream>>>(
         agg->t, agg->KW, agg->n_calls, (int)agg->desc.row_count_index,
         agg->cd(0), agg->cd(1), agg->cd(2), agg->cd(3));
@@ -2580,9 +2666,13 @@ int rw_agg_debug_dirty(void* h, uint32_t max_n, uint32_t* slots, uint32_t* state
 
 int rw_agg_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len) {
     auto* agg = (HashAgg*)h;
-    *len = agg->spill.size();
-    *buf = (uint8_t*)malloc(agg->spill.size() ? agg->spill.size() : 1);
-    memcpy(*buf, agg->spill.data(), agg->spill.size());
+    // memcmp-key order, as the join/topn/dedup drains (rw_stream.h's ordering
+    // guarantee): the flush kernel's atomic-cursor emission order is
+    // nondeterministic, so the canonical sorted form is what both this and
+    // the oracle emit — enabling byte-compare parity.
+    sort_spill_frames(agg->spill);
+    int rc = spill_export(agg->spill, buf, len);
+    if (rc != RW_OK) return rc;
     agg->spill.clear();
     return RW_OK;
 }
@@ -2596,10 +2686,7 @@ int rw_agg_dedup_drain(void* h, int di, uint8_t** buf, uint64_t* len) {
     std::vector<uint8_t> sp;
     int rc = agg->dedup_drain(di, sp);
     if (rc != RW_OK) return rc;
-    *len = sp.size();
-    *buf = (uint8_t*)malloc(sp.size() ? sp.size() : 1);
-    memcpy(*buf, sp.data(), sp.size());
-    return RW_OK;
+    return spill_export(sp, buf, len);
 }
 
 void rw_spill_free(uint8_t* buf) { free(buf); }
@@ -4592,19 +4679,15 @@ int rw_join_checkpoint_drain(void* h, int side, uint8_t** buf,
     std::vector<uint8_t> sp;
     int rc = j->checkpoint_drain(side, sp);
     if (rc != RW_OK) return rc;
-    *len = sp.size();
-    *buf = (uint8_t*)malloc(sp.size() ? sp.size() : 1);
-    memcpy(*buf, sp.data(), sp.size());
-    return RW_OK;
+    return spill_export(sp, buf, len);
 }
 
 int rw_join_degree_drain(void* h, int side, uint8_t** buf, uint64_t* len) {
     auto* j = (HashJoin*)h;
     if (side != 0 && side != 1) FAIL(RW_E_INVAL, "bad side");
     auto& sp = j->deg_spill[side];
-    *len = sp.size();
-    *buf = (uint8_t*)malloc(sp.size() ? sp.size() : 1);
-    memcpy(*buf, sp.data(), sp.size());
+    int rc = spill_export(sp, buf, len);
+    if (rc != RW_OK) return rc;
     sp.clear();
     return RW_OK;
 }
@@ -5363,10 +5446,7 @@ int rw_topn_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len) {
     std::vector<uint8_t> sp;
     int rc = ((GroupTopN*)h)->checkpoint_drain(sp);
     if (rc != RW_OK) return rc;
-    *len = sp.size();
-    *buf = (uint8_t*)malloc(sp.size() ? sp.size() : 1);
-    memcpy(*buf, sp.data(), sp.size());
-    return RW_OK;
+    return spill_export(sp, buf, len);
 }
 
 void* rw_group_top_n_create(const RwGroupTopNDesc* d) {

@@ -273,15 +273,17 @@ def test_eowc_spill_records():
     ], recs
     # epoch 2: retract window 2 fully, then close windows < 3:
     # mid-window PUT for the dirty (now rc=0) window 2, then close DELETEs
-    # for both windows — window 1 emits, window 2 does not
+    # for both windows — window 1 emits, window 2 does not. Drains emit in
+    # memcmp-key order (stable within a key: window 2's PUT precedes its
+    # close DELETE).
     o.push(from_pretty(" I\n - 2\n - 2"))
     o.watermark(0, 3)
     o.flush(2)
     o.poll_all()
     recs = drain(oracle(), o)
     assert recs == [
-        (1, memcmp_i64(2), value_i64(2) + value_i64(0)),
         (0, memcmp_i64(1), b""),
+        (1, memcmp_i64(2), value_i64(2) + value_i64(0)),
         (0, memcmp_i64(2), b""),
     ], recs
     o.close()
