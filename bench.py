@@ -1,15 +1,18 @@
 #!/usr/bin/env python3
 """bench.py — measures the hot path per the driver contract.
 
-Workload (BASELINE.json configs[1], the single-GPU config the metric is
-quoted on): Nexmark q7 windowed hash-agg — append-only HashAgg
-group_key=[window $expr1], aggs=[max(price), count] (reference plan
-`nexmark.yaml` q7 block) — on synthetic bid-shaped chunks: price ~
-uniform[1,1e7) i64, date_time monotone in 10s windows, all-Insert ops,
-seeded. A step = one pass of the apply kernel over one 1M-row batch
-(= 256 ingested 4K-row chunks staged within the epoch)
-already resident in HBM; a checkpoint barrier (flush) fires every
---barrier-every steps inside the timed region.
+BASELINE.json's metric is "input rows/sec/GPU on Nexmark q7+q8"; the default
+run emits BOTH lines:
+- q7 (configs[1]): append-only HashAgg group_key=[window $expr1],
+  aggs=[max(price), count] (reference plan `nexmark.yaml` q7 block) on
+  synthetic bid-shaped chunks: price ~ uniform[1,1e7) i64, date_time
+  monotone in 10s windows, all-Insert ops, seeded. A step = one checkpoint
+  EPOCH: 16 x 1M-row chunk-batches (= 4096 ingested 4K-row chunks,
+  DESIGN.md §3.1) already resident in HBM, applied as one launch, plus the
+  stream-ordered flush.
+- q8 (configs[2]): stream-stream hash-join, 10M-key build side in HBM;
+  a step = one 1M-row probe batch.
+--workload q3 selects the TPC-H-q3-stream pipeline (configs[4]).
 
 Parity gate: before timing, a small q7 run is compared row-for-row
 (multiset per epoch) against the CPU oracle; a mismatch aborts the bench.
@@ -41,13 +44,18 @@ class KernelStats(ctypes.Structure):
 
 
 CHUNK_ROWS = 4096
-CHUNKS_PER_BATCH = 256  # 1M rows per step
-WINDOW_US = 10_000_000  # 10s tumble (q7)
-# algorithmic bytes per input row for the q7 agg_apply kernel: key 8 + price 8
-# + op 1 + 2 validity bytes = 19 B of mandatory input-stream traffic (the few
-# window slots stay cache-resident; SURVEY §8d's 48 B/row includes an
-# HBM-resident probe, reported separately via rocprof traffic)
-BYTES_PER_ROW = 19
+CHUNKS_PER_BATCH = 256   # 1M rows per logical ingest batch
+EPOCH_BATCHES = 16       # batches buffered per checkpoint epoch
+WINDOW_US = 10_000_000   # 10s tumble (q7)
+# Algorithmic bytes per input row for the q7 dense agg_apply kernel: the two
+# 8-B input streams (window key, price). The dense path reads no ops/validity
+# streams and the few window slots stay cache-resident. PMC counters agree:
+# 2x134,747 KB FETCH + 562 KB WRITE per 16,777,216-row epoch launch
+# (profiles/r01_q7_pmc_{fetch,write}_v7.txt, x2 = the gfx950 FETCH_SIZE
+# correction, MI355X_MICROARCH.md §HBM) = 16.5 B/row measured vs 16.0
+# algorithmic.
+Q7_BYTES_PER_ROW = 16.0
+Q7_TRAFFIC_B_PER_ROW = (2 * 134_747 + 562) * 1024 / (EPOCH_BATCHES * CHUNK_ROWS * CHUNKS_PER_BATCH)
 HBM_PEAK_GBS = 8000.0  # spec peak (MI355X_MICROARCH.md)
 
 
@@ -144,18 +152,21 @@ def _cpu_baseline_worker(arg):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=2000)
-    ap.add_argument("--warmup", type=int, default=100)
+    ap.add_argument("--steps", type=int, default=500)
+    ap.add_argument("--warmup", type=int, default=25)
     ap.add_argument("--barrier-every", type=int, default=16)
     ap.add_argument("--windows-per-epoch", type=int, default=64)
     ap.add_argument("--seed", type=int, default=1)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
-    ap.add_argument("--workload", choices=["q7", "q8", "q3"], default="q7",
-                    help="q7 = windowed hash-agg (BASELINE configs[1], the "
-                         "default the driver measures); q8 = stream-stream "
-                         "hash-join, 10M-key build side (configs[2]); q3 = "
-                         "TPC-H-stream join+agg pipeline with insert/delete "
-                         "mix (configs[4], minus the Hummock spill)")
+    ap.add_argument("--workload", choices=["all", "q7", "q8", "q3"],
+                    default="all",
+                    help="all (default) = the BASELINE metric 'q7+q8': one "
+                         "run emits the q7 line then the q8 line. q7 = "
+                         "windowed hash-agg only (configs[1]); q8 = "
+                         "stream-stream hash-join, 10M-key build side "
+                         "(configs[2]); q3 = TPC-H-stream join+agg pipeline "
+                         "with insert/delete mix (configs[4], minus the "
+                         "Hummock spill)")
     ap.add_argument("--q3-orders", type=int, default=100_000_000)
     ap.add_argument("--exchange", choices=["auto", "on", "off"], default="auto",
                     help="vnode partition + RCCL all-to-all-v before the agg "
@@ -219,18 +230,28 @@ def main():
     if rank == 0:
         parity_gate(ffi, gpu_lib, np.random.default_rng(99))
 
-    if args.workload == "q8":
-        bench_q8(args, ffi, gpu_lib, rng, rank, world, dist)
+    try:
+        if args.workload in ("all", "q7"):
+            bench_q7(args, ffi, gpu_lib, rng, rank, world, dist)
+        if args.workload in ("all", "q8"):
+            bench_q8(args, ffi, gpu_lib, np.random.default_rng(args.seed + rank),
+                     rank, world, dist)
+        if args.workload == "q3":
+            bench_q3(args, ffi, gpu_lib, rng, rank, world, dist)
+    finally:
         if dist:
             dist.destroy_process_group()
-        return
-    if args.workload == "q3":
-        bench_q3(args, ffi, gpu_lib, rng, rank, world, dist)
-        if dist:
-            dist.destroy_process_group()
-        return
 
-    # ---- build the timed executor + preloaded batches ----
+
+def bench_q7(args, ffi, gpu_lib, rng, rank, world, dist):
+    """Nexmark q7 windowed hash-agg (BASELINE configs[1]). A step = one
+    checkpoint epoch: EPOCH_BATCHES x 1M-row chunk-batches applied as ONE
+    kernel launch (order-free value states) followed by the
+    stream-ordered flush — so every timed step is a real launch and small
+    --steps counts still average over `steps` launches."""
+    import ctypes
+
+    L = gpu_lib.lib
     from rwtest.ffi import AGG_COUNT_STAR, AGG_MAX, T_I64
 
     calls = [(AGG_MAX, 1, T_I64), (AGG_COUNT_STAR, -1, T_I64)]
@@ -252,15 +273,12 @@ def main():
     # exchange already routed keys — DESIGN.md §7). With exchange: a GLOBAL
     # window space; the RCCL all-to-all routes each window to its vnode owner.
     window_base = 0 if use_exchange else rank * 1_000_000 * WINDOW_US
-    # 16 × 19 MB ≈ 300 MB of resident input > the 256 MB Infinity Cache, so
+    # 16 × 16 MB ≈ 270 MB of resident input > the 256 MB Infinity Cache, so
     # the timed region streams from HBM (L3-masking gotcha,
     # cdna_hip_programming.md §2)
-    n_batches = 16
+    n_batches = EPOCH_BATCHES
     batches = []
     giant = None
-    eff_barrier = (args.barrier_every
-                   if args.barrier_every > 0 and n_batches % args.barrier_every == 0
-                   else n_batches)
     if use_exchange:
         for b in range(n_batches):
             c = make_q7_chunk(ffi, rng, batch_rows,
@@ -294,37 +312,31 @@ def main():
         payload_cap = int(batch_rows * (1 + 2 * 9) * 4)  # 4x headroom for skew
         xb = exch.make_buffers(payload_cap)
 
-    def step(i):
-        if use_exchange:
-            nslots = L.rw_agg_n_batch_slots(agg.h)
-            recv_blocks = exch.run(agg.h, batches[i % n_batches], xb,
-                                   n_cols=nslots)
-            rc = L.rw_agg_apply_payload(
-                agg.h, ctypes.c_void_p(xb.recv), recv_blocks, world, nslots,
-                1)  # q7 batches are dense (all-Insert, non-null)
-            assert rc == 0, gpu_lib.last_error()
-        else:
-            rc = L.rw_agg_bench_apply(agg.h, batches[i % n_batches])
-            assert rc == 0, gpu_lib.last_error()
-        if (i + 1) % args.barrier_every == 0:
-            # checkpoint barrier: change inference + emission into HBM,
-            # stream-ordered with no host round-trip; overflow surfaces at
-            # the end-of-region rw_agg_sync. The downstream (exchange/sink)
-            # consumes device-resident, as in the q3 pipeline — the
-            # host-marshalling flush stays the parity-test surface
-            # (tests/test_gpu_parity.py)
-            rc = L.rw_agg_flush_launch(agg.h, i)
-            assert rc == 0, gpu_lib.last_error()
-
-    # non-exchange path: the whole step loop runs in C, epoch-granular (the
-    # Python interpreter costs more per step than the apply kernel itself)
+    # A STEP = ONE EPOCH in both modes: EPOCH_BATCHES 1M-row batches applied
+    # (one fused launch without exchange; per-batch route+apply with it),
+    # then the stream-ordered checkpoint flush. The downstream
+    # (exchange/sink) consumes device-resident, as in the q3 pipeline — the
+    # host-marshalling flush stays the parity-test surface
+    # (tests/test_gpu_parity.py).
     def run_steps(n):
         if use_exchange:
-            for i in range(n):
-                step(i)
+            nslots = L.rw_agg_n_batch_slots(agg.h)
+            for e in range(n):
+                for b in range(n_batches):
+                    recv_blocks = exch.run(agg.h, batches[b], xb,
+                                           n_cols=nslots)
+                    rc = L.rw_agg_apply_payload(
+                        agg.h, ctypes.c_void_p(xb.recv), recv_blocks, world,
+                        nslots, 1)  # q7 batches: dense all-Insert, non-null
+                    assert rc == 0, gpu_lib.last_error()
+                rc = L.rw_agg_flush_launch(agg.h, e)
+                assert rc == 0, gpu_lib.last_error()
         else:
+            # the whole epoch loop runs in C (the Python interpreter costs
+            # more per step than the apply kernel itself)
             rc = L.rw_agg_bench_run_epochs(agg.h, giant, batch_rows,
-                                           n_batches, n, eff_barrier, 0)
+                                           n_batches, n * n_batches,
+                                           n_batches, 0)
             assert rc == 0, gpu_lib.last_error()
 
     # ---- warmup ----
@@ -352,15 +364,60 @@ def main():
     ks = KernelStats()
     L.rw_agg_kernel_stats(agg.h, ctypes.byref(ks))
 
+    # ---- untimed extra segment (rank 0, N=1): the checkpoint-INCLUSIVE
+    # epoch cost — apply + host-marshalling flush + device→host spill drain
+    # (rw_agg_checkpoint_drain), the preserved-checkpoint path the north
+    # star requires as a measured number ----
+    ckpt = None
+    if rank == 0 and world == 1 and not use_exchange:
+        L.rw_agg_checkpoint_drain.restype = ctypes.c_int
+        L.rw_agg_checkpoint_drain.argtypes = [
+            ctypes.c_void_p, ctypes.POINTER(ctypes.POINTER(ctypes.c_uint8)),
+            ctypes.POINTER(ctypes.c_uint64)]
+        L.rw_spill_free.argtypes = [ctypes.POINTER(ctypes.c_uint8)]
+        n_ck = 4
+        drained = 0
+        t0 = time.perf_counter()
+        for e in range(n_ck):
+            # the giant region IS one epoch: one fused apply launch
+            rc = L.rw_agg_bench_apply(agg.h, giant)
+            assert rc == 0, gpu_lib.last_error()
+            rc = L.rw_hash_agg_flush(agg.h, 1000 + e)
+            assert rc == 0, gpu_lib.last_error()
+            while True:
+                c = L.rw_hash_agg_poll(agg.h)
+                if not c:
+                    break
+                L.rw_chunk_free(c)
+            buf = ctypes.POINTER(ctypes.c_uint8)()
+            ln = ctypes.c_uint64()
+            rc = L.rw_agg_checkpoint_drain(agg.h, ctypes.byref(buf),
+                                           ctypes.byref(ln))
+            assert rc == 0, gpu_lib.last_error()
+            drained += ln.value
+            L.rw_spill_free(buf)
+        ck_elapsed = time.perf_counter() - t0
+        ckpt = {
+            "ms_per_epoch": ck_elapsed * 1000.0 / n_ck,
+            "rows_per_s": n_ck * n_batches * batch_rows / ck_elapsed,
+            "drained_bytes_per_epoch": drained / n_ck,
+            "epochs": n_ck,
+            "note": "epoch incl. host-marshalled emission + spill drain "
+                    "(checkpoint-inclusive rate; untimed in the headline)",
+        }
+
     if rank == 0:
-        total_rows = args.steps * batch_rows * world
+        epoch_rows = n_batches * batch_rows
+        total_rows = args.steps * epoch_rows * world
         value = total_rows / elapsed
         ms_per_step = elapsed * 1000.0 / args.steps
         # roofline: algorithmic bytes per apply launch ÷ measured launch time
-        # (HIP events on the executor's stream, inside the C library)
+        # (HIP events on the executor's stream, inside the C library);
+        # traffic = the PMC-measured per-row bytes (see Q7_TRAFFIC_B_PER_ROW)
+        # scaled to this run's rows-per-launch
         avg_launch_ms = ks.total_ms / max(ks.launches, 1)
         rows_per_launch = ks.rows / max(ks.launches, 1)
-        achieved_gbs = (BYTES_PER_ROW * rows_per_launch) / (avg_launch_ms * 1e-3) / 1e9
+        achieved_gbs = (Q7_BYTES_PER_ROW * rows_per_launch) / (avg_launch_ms * 1e-3) / 1e9
         exch_stats = None
         if use_exchange and exch is not None:
             ems, en = exch.stats()
@@ -382,9 +439,10 @@ def main():
             "config": {
                 "workload": "nexmark_q7",
                 "chunk_rows": CHUNK_ROWS,
-                "chunks_per_step": CHUNKS_PER_BATCH,
+                "chunks_per_epoch_step": n_batches * CHUNKS_PER_BATCH,
+                "rows_per_step": epoch_rows,
                 "windows_per_epoch": args.windows_per_epoch,
-                "barrier_every_steps": args.barrier_every,
+                "barriers": "one checkpoint flush per step (a step = one epoch)",
                 "agg": "max(price), count group by 10s window (append-only)",
                 "parallelism": f"dp{world}",
             },
@@ -394,17 +452,16 @@ def main():
                 "peak": HBM_PEAK_GBS,
                 "unit": "GB/s",
                 "frac": achieved_gbs / HBM_PEAK_GBS,
-                "traffic": None,  # filled from rocprofv3 PMC runs (profiles/)
+                "traffic": Q7_TRAFFIC_B_PER_ROW * rows_per_launch,
             },
+            "checkpoint": ckpt,
             "cpu_baseline": None,
         }
         if not args.skip_cpu_baseline and world == 1:
             result["cpu_baseline"] = cpu_baseline(ffi, np.random.default_rng(5))
-        print(json.dumps(result))
+        print(json.dumps(result), flush=True)
 
     agg.close()
-    if dist:
-        dist.destroy_process_group()
 
 
 class ExchangeCtx:
