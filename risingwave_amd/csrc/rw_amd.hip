@@ -247,6 +247,14 @@ __device__ __forceinline__ long long* jvals(JoinRowHdr* h) {
     return (long long*)((uint8_t*)h + 16);
 }
 
+// First slot for a join-table probe: TOP bits of the hash, so the slot index
+// is monotone in h and a hash-prefix partition maps to a CONTIGUOUS slot
+// region — the locality lever of the partitioned probe/insert pipeline
+// (jpart_* kernels below). Distribution-equivalent to h & cap_mask.
+__device__ __forceinline__ uint32_t jslot_start(uint64_t h, uint32_t cap_mask) {
+    return (uint32_t)(h >> (64 - __popc(cap_mask)));
+}
+
 // own-side find-or-insert. The probe walk uses PLAIN cached loads: a slot's
 // state and keys share one 64-B line, and the claim protocol drains the sc1
 // key stores to the coherence point BEFORE the sc1 READY store — so any
@@ -260,7 +268,7 @@ __device__ __forceinline__ uint32_t jslot_find_or_insert(JoinSlot* slots,
                                                          const int64_t* kw,
                                                          uint32_t nullmask,
                                                          int KW) {
-    uint32_t slot = (uint32_t)(hash_key(kw, nullmask, KW) & cap_mask);
+    uint32_t slot = jslot_start(hash_key(kw, nullmask, KW), cap_mask);
     for (uint32_t probes = 0; probes <= cap_mask; probes++) {
         JoinSlot* sl = &slots[slot];
         uint32_t st = sl->state; // plain (fast path)
@@ -303,7 +311,7 @@ __device__ __forceinline__ uint32_t jslot_find_cached(const JoinSlot* slots,
                                                       uint32_t cap_mask,
                                                       const int64_t* kw,
                                                       uint32_t nullmask, int KW) {
-    uint32_t slot = (uint32_t)(hash_key(kw, nullmask, KW) & cap_mask);
+    uint32_t slot = jslot_start(hash_key(kw, nullmask, KW), cap_mask);
     for (uint32_t probes = 0; probes <= cap_mask; probes++) {
         const JoinSlot* sl = &slots[slot];
         uint32_t st = sl->state;
@@ -323,7 +331,7 @@ __device__ __forceinline__ uint32_t jslot_find_sc1(const JoinSlot* slots,
                                                    uint32_t cap_mask,
                                                    const int64_t* kw,
                                                    uint32_t nullmask, int KW) {
-    uint32_t slot = (uint32_t)(hash_key(kw, nullmask, KW) & cap_mask);
+    uint32_t slot = jslot_start(hash_key(kw, nullmask, KW), cap_mask);
     for (uint32_t probes = 0; probes <= cap_mask; probes++) {
         const JoinSlot* sl = &slots[slot];
         uint32_t st = ld_u32(&sl->state);
@@ -3460,6 +3468,306 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
     }
 }
 
+// ============ Partitioned probe/insert pipeline (q8 hot path) ============
+//
+// The round-1 single-kernel path was random-access-bound: each probe row
+// touched ~3 random 64-B lines spread over GBs of slot table + record
+// store (PMC: 500–1000 B fetched per row vs the 128-B algorithmic model,
+// profiles/r01_q8_pmc_fetch.txt), sustaining ~2 G own-side inserts/s.
+// This pipeline partitions an all-Insert unique-key batch by the TOP
+// JPART_LOG2 bits of the join-key hash. jslot_start() derives slot
+// indices from the same top bits, so partition p's slots are the
+// CONTIGUOUS window [p*cap/P, (p+1)*cap/P) of BOTH sides' tables, and its
+// records land in one contiguous row-store run per batch (the scatter
+// writes each record into its FINAL row-store position). The fused
+// probe+insert phase runs one workgroup per partition: its slot window
+// (~cap*64/P bytes) stays in the XCD's L2 and its match-record reads fall
+// in per-batch partition runs — DRAM-row-local instead of uniformly
+// random. eq_join_oneside semantics are unchanged (inner join, all
+// visible rows Insert, keys pairwise distinct — the host gates on exactly
+// the conditions under which batch-parallel insert+probe commutes with
+// the reference's sequential per-row loop; hash_join.rs:949-1075).
+//
+// Phases (stream-ordered, no host round-trip):
+//   1. jpart_count          per-block LDS histograms of partition ids
+//   2. jpart_scan           scatter bases + partition ranges + row-store
+//                           reservation (single workgroup)
+//   3. jpart_scatter        write final row records, partition-clustered
+//   4. jpart_probe_insert   per partition: probe match side + emit, then
+//                           link own-side chains (plain RMW — unique keys
+//                           give every touched slot exactly one writer)
+
+#define JPART_LOG2 11
+#define JPART_P (1u << JPART_LOG2)
+#define JPART_NBLK 256
+#define JPART_MIN_ROWS 131072u
+
+__device__ __forceinline__ uint32_t jpart_of(uint64_t h) {
+    return (uint32_t)(h >> (64 - JPART_LOG2));
+}
+
+// per-row key extraction from the input batch; returns false for rows that
+// carry no state (invisible is excluded by the host gate; never-match NULL
+// keys are not inserted and emit nothing under inner join,
+// hash_join.rs:1004-1016)
+__device__ __forceinline__ bool jpart_row_key(const JoinBatchDev& b,
+                                              const JoinMeta& m, int S,
+                                              uint32_t r, int64_t* kw,
+                                              uint32_t* nullmask) {
+    uint32_t nm = 0;
+    for (int i = 0; i < m.KW; i++) {
+        uint8_t col = m.key_cols[S][i];
+        bool valid = b.col_valid[col][r];
+        kw[i] = valid ? b.col_vals[col][r] : 0;
+        nm |= (uint32_t)(!valid) << i;
+    }
+    *nullmask = nm;
+    return !(nm & ~(uint32_t)m.null_safe_mask);
+}
+
+__global__ void jpart_count_kernel(JoinBatchDev b, JoinMeta m, int S,
+                                   uint32_t* pcount) {
+    __shared__ uint32_t hist[JPART_P];
+    for (uint32_t i = threadIdx.x; i < JPART_P; i += blockDim.x) hist[i] = 0;
+    __syncthreads();
+    uint32_t n = b.n_rows;
+    uint32_t per = (n + gridDim.x - 1) / gridDim.x;
+    uint32_t r0 = blockIdx.x * per;
+    uint32_t r1 = r0 + per < n ? r0 + per : n;
+    int64_t kw[MAX_KW];
+    uint32_t nm;
+    for (uint32_t r = r0 + threadIdx.x; r < r1; r += blockDim.x) {
+        if (!jpart_row_key(b, m, S, r, kw, &nm)) continue;
+        atomicAdd(&hist[jpart_of(hash_key(kw, nm, m.KW))], 1u);
+    }
+    __syncthreads();
+    for (uint32_t p = threadIdx.x; p < JPART_P; p += blockDim.x)
+        pcount[(size_t)blockIdx.x * JPART_P + p] = hist[p];
+}
+
+// single-workgroup scan: per-partition totals → exclusive bases; rewrites
+// pcount[blk][p] into per-block scatter bases; reserves the batch's row
+// range on the device cursor (err code 3 = row store full, as the
+// wave-aggregated reservation in jown_insert)
+__global__ void jpart_scan_kernel(uint32_t* pcount, int nblk,
+                                  uint32_t* part_base, uint32_t* row_cursor,
+                                  uint32_t row_cap, uint32_t* row_base,
+                                  uint32_t* err) {
+    __shared__ uint32_t tot[JPART_P];
+    __shared__ uint32_t excl[JPART_P];
+    for (uint32_t p = threadIdx.x; p < JPART_P; p += blockDim.x) {
+        uint32_t s = 0;
+        for (int bk = 0; bk < nblk; bk++)
+            s += pcount[(size_t)bk * JPART_P + p];
+        tot[p] = s;
+        excl[p] = s;
+    }
+    __syncthreads();
+    // Hillis-Steele inclusive scan over excl[]
+    for (uint32_t off = 1; off < JPART_P; off <<= 1) {
+        uint32_t v[JPART_P / 1024];
+        for (uint32_t i = threadIdx.x, k = 0; i < JPART_P;
+             i += blockDim.x, k++)
+            v[k] = i >= off ? excl[i - off] : 0;
+        __syncthreads();
+        for (uint32_t i = threadIdx.x, k = 0; i < JPART_P;
+             i += blockDim.x, k++)
+            excl[i] += v[k];
+        __syncthreads();
+    }
+    uint32_t total = excl[JPART_P - 1];
+    if (threadIdx.x == 0) {
+        uint32_t base = atomicAdd(row_cursor, total);
+        if ((uint64_t)base + total > row_cap) atomicExch(err, 3u);
+        *row_base = base;
+        part_base[JPART_P] = total;
+    }
+    __syncthreads();
+    for (uint32_t p = threadIdx.x; p < JPART_P; p += blockDim.x) {
+        uint32_t run = excl[p] - tot[p]; // exclusive base of partition p
+        part_base[p] = run;
+        for (int bk = 0; bk < nblk; bk++) {
+            uint32_t c = pcount[(size_t)bk * JPART_P + p];
+            pcount[(size_t)bk * JPART_P + p] = run;
+            run += c;
+        }
+    }
+}
+
+__global__ void jpart_scatter_kernel(JoinBatchDev b, JoinMeta m, int S,
+                                     JoinSideDev own, uint32_t* pcount,
+                                     const uint32_t* row_base,
+                                     const uint32_t* err) {
+    if (*err) return;
+    __shared__ uint32_t cur[JPART_P];
+    for (uint32_t p = threadIdx.x; p < JPART_P; p += blockDim.x)
+        cur[p] = pcount[(size_t)blockIdx.x * JPART_P + p];
+    __syncthreads();
+    uint32_t rb = *row_base;
+    uint32_t n = b.n_rows;
+    uint32_t per = (n + gridDim.x - 1) / gridDim.x;
+    uint32_t r0 = blockIdx.x * per;
+    uint32_t r1 = r0 + per < n ? r0 + per : n;
+    int64_t kw[MAX_KW];
+    uint32_t nm;
+    for (uint32_t r = r0 + threadIdx.x; r < r1; r += blockDim.x) {
+        if (!jpart_row_key(b, m, S, r, kw, &nm)) continue;
+        uint32_t p = jpart_of(hash_key(kw, nm, m.KW));
+        uint32_t row = rb + atomicAdd(&cur[p], 1u);
+        // final row record (plain cached stores: the next kernel on the
+        // stream observes them through the inter-dispatch cache flush)
+        JoinRowHdr* hd = jrow(own, row);
+        long long* hv = jvals(hd);
+        uint32_t vb = 0;
+        for (int c = 0; c < m.n_cols[S]; c++) {
+            hv[c] = b.col_vals[c][r];
+            vb |= (uint32_t)(b.col_valid[c][r] != 0) << c;
+        }
+        hd->validbits = vb;
+        hd->degree = 0;
+        hd->alive = 1;
+        hd->next = UINT32_MAX;
+    }
+}
+
+// inner-join condition over a probe ROW RECORD (vs join_cond_ok's batch row)
+__device__ __forceinline__ bool jpart_cond_ok(const JoinMeta& m, int S,
+                                              const long long* pv,
+                                              uint32_t pvb, uint32_t mvb,
+                                              const long long* mv) {
+    if (!m.has_cond) return true;
+    auto fetch = [&](uint8_t src, uint8_t col, int64_t* v) -> bool {
+        if ((int)src == S) {
+            if (!((pvb >> col) & 1)) return false;
+            *v = pv[col];
+        } else {
+            if (!((mvb >> col) & 1)) return false;
+            *v = mv[col];
+        }
+        return true;
+    };
+    int64_t a, c;
+    if (!fetch(m.cond_src_l, m.cond_col_l, &a)) return false;
+    if (!fetch(m.cond_src_r, m.cond_col_r, &c)) return false;
+    switch (m.cond_op) {
+        case RW_CMP_LT: return a < c;
+        case RW_CMP_LE: return a <= c;
+        case RW_CMP_GT: return a > c;
+        case RW_CMP_GE: return a >= c;
+    }
+    return false;
+}
+
+__global__ __launch_bounds__(256) void jpart_probe_insert_kernel(
+    JoinSideDev own, JoinSideDev match, JoinMeta m, int S, JoinOutDev out,
+    const uint32_t* part_base, const uint32_t* row_base,
+    const uint32_t* err) {
+    if (*err) return;
+    uint32_t rb = *row_base;
+    uint32_t lo = part_base[blockIdx.x];
+    uint32_t hi = part_base[blockIdx.x + 1]; // [JPART_P] holds the total
+    int lane = threadIdx.x & 63;
+    uint32_t n = hi - lo;
+    uint32_t iters = (n + blockDim.x - 1) / blockDim.x;
+    for (uint32_t it = 0; it < iters; it++) {
+        uint32_t i = it * blockDim.x + threadIdx.x;
+        bool active = i < n;
+        uint32_t row = rb + lo + i;
+        int64_t kw[MAX_KW];
+        uint32_t nm = 0;
+        JoinRowHdr* hd = nullptr;
+        long long* hv = nullptr;
+        uint32_t pvb = 0;
+        if (active) {
+            hd = jrow(own, row);
+            hv = jvals(hd);
+            pvb = hd->validbits;
+            for (int k = 0; k < m.KW; k++) {
+                uint8_t col = m.key_cols[S][k];
+                bool valid = (pvb >> col) & 1;
+                kw[k] = valid ? hv[col] : 0;
+                nm |= (uint32_t)(!valid) << k;
+            }
+        }
+        // probe the match side (immutable during this launch: the batch
+        // mutates only `own`)
+        uint32_t mslot = UINT32_MAX;
+        uint32_t my_n = 0;
+        if (active) {
+            mslot = jslot_find_cached(match.slots, match.cap_mask, kw, nm,
+                                      m.KW);
+            if (mslot != UINT32_MAX) {
+                uint32_t mr = match.slots[mslot].head;
+                while (mr != UINT32_MAX) {
+                    JoinRowHdr* mh = jrow(match, mr);
+                    if (mh->alive &&
+                        jpart_cond_ok(m, S, hv, pvb, mh->validbits, jvals(mh)))
+                        my_n++;
+                    mr = mh->next;
+                }
+            }
+        }
+        // wave-aggregated output reservation + emit (multiset parity; the
+        // reference's intra-epoch order is nondeterministic)
+        uint32_t incl = my_n;
+        for (int d = 1; d < 64; d <<= 1) {
+            uint32_t o = __shfl_up(incl, d);
+            if (lane >= d) incl += o;
+        }
+        uint32_t total = (uint32_t)__shfl((int)incl, 63);
+        uint32_t base = 0;
+        if (lane == 0 && total) base = atomicAdd(&out.counters[0], total);
+        base = (uint32_t)__shfl((int)base, 0);
+        uint32_t my_base = base + incl - my_n;
+        if (total && base + total > out.cap) {
+            if (lane == 0) atomicExch(&out.counters[1], 1u);
+        } else if (my_n) {
+            uint32_t mr = match.slots[mslot].head;
+            uint32_t k = 0;
+            while (mr != UINT32_MAX && k < my_n) {
+                JoinRowHdr* mh = jrow(match, mr);
+                if (mh->alive &&
+                    jpart_cond_ok(m, S, hv, pvb, mh->validbits, jvals(mh))) {
+                    uint32_t orow = my_base + k;
+                    out.ops[orow] = RW_OP_INSERT;
+                    const long long* mv = jvals(mh);
+                    for (int c = 0; c < m.n_out; c++) {
+                        bool from_probe = (int)m.out_src[c] == S;
+                        uint8_t col = m.out_col[c];
+                        int64_t v;
+                        uint8_t valid;
+                        if (from_probe) {
+                            valid = (pvb >> col) & 1;
+                            v = hv[col];
+                        } else {
+                            valid = (mh->validbits >> col) & 1;
+                            v = mv[col];
+                        }
+                        out.vals[(size_t)orow * m.n_out + c] = valid ? v : 0;
+                        out.nulls[(size_t)orow * m.n_out + c] = !valid;
+                    }
+                    k++;
+                }
+                mr = mh->next;
+            }
+        }
+        // own-side insert: the record is already in place; link the chain.
+        // Keys are pairwise distinct in the batch, so each touched slot has
+        // exactly one linking thread — plain read-modify-write.
+        if (active) {
+            uint32_t os = jslot_find_or_insert(own.slots, own.cap_mask, kw,
+                                               nm, m.KW);
+            if (os == UINT32_MAX) {
+                atomicExch(&out.counters[1], 2u);
+            } else {
+                uint32_t* headp = &own.slots[os].head;
+                hd->next = *headp;
+                *headp = row;
+            }
+        }
+    }
+}
+
 // watermark TTL sweeps (state_table watermark cleaning):
 // rows/groups whose watermarked key column sorts below the value are
 // retired in place (slots stay READY so linear probing is undisturbed;
@@ -4032,6 +4340,40 @@ struct HashJoin {
         return RW_OK;
     }
 
+    // partitioned-pipeline device buffers (lazily allocated; ~2 MB)
+    uint32_t* d_pcount = nullptr;
+    uint32_t* d_part_base = nullptr;
+    uint32_t* d_row_base = nullptr;
+
+    // The partitioned pipeline handles exactly the shape under which the
+    // batch-parallel insert commutes with the reference's sequential loop
+    // without CAS chain publishes: inner join, every visible row an Insert,
+    // keys pairwise distinct (host-verified at upload), whole-batch range.
+    bool can_partition(const JoinBatchDev& b, uint32_t r0, uint32_t r1) {
+        static int en = [] {
+            const char* e = getenv("RW_JOIN_PART"); // A/B: 0 = old path
+            return e ? atoi(e) : 1;
+        }();
+        if (!en) return false;
+        if (m.join_type != RW_JOIN_INNER || m.append_only) return false;
+        if (!b.all_insert || !b.unique_keys || b.vis) return false;
+        if (r0 != 0 || r1 != b.n_rows) return false;
+        if (b.n_rows < JPART_MIN_ROWS) return false;
+        // partitions must map to multi-slot windows on both sides
+        if (side[0].cap_mask < 4 * JPART_P - 1 ||
+            side[1].cap_mask < 4 * JPART_P - 1)
+            return false;
+        return true;
+    }
+
+    int ensure_part_bufs() {
+        if (d_pcount) return RW_OK;
+        HIP_TRY(hipMalloc(&d_pcount, (size_t)JPART_NBLK * JPART_P * 4));
+        HIP_TRY(hipMalloc(&d_part_base, (size_t)(JPART_P + 1) * 4));
+        HIP_TRY(hipMalloc(&d_row_base, 4));
+        return RW_OK;
+    }
+
     int probe(int s, const JoinBatchDev& b, bool timed, uint32_t r0, uint32_t r1) {
         uint32_t blocks = (r1 - r0 + 255) / 256;
         if (blocks > 2048) blocks = 2048;
@@ -4050,8 +4392,25 @@ struct HashJoin {
             const char* e = getenv("RW_JOIN_SKIP"); // bench A/B only:
             return e ? atoi(e) : 0;  // 1=no emit writes, 2=no own insert
         }();
-        join_probe_kernel<<<blocks, 256, 0, stream>>>(b, side[s], side[1 - s], m, s,
-                                                      out, r0, r1, dbg_skip);
+        if (can_partition(b, r0, r1)) {
+            int rc = ensure_part_bufs();
+            if (rc != RW_OK) return rc;
+            jpart_count_kernel<<<JPART_NBLK, 256, 0, stream>>>(b, m, s,
+                                                               d_pcount);
+            jpart_scan_kernel<<<1, 1024, 0, stream>>>(
+                d_pcount, JPART_NBLK, d_part_base, side[s].row_cursor,
+                side[s].row_cap, d_row_base, out.counters + 1);
+            jpart_scatter_kernel<<<JPART_NBLK, 256, 0, stream>>>(
+                b, m, s, side[s], d_pcount, d_row_base, out.counters + 1);
+            jpart_probe_insert_kernel<<<JPART_P, 256, 0, stream>>>(
+                side[s], side[1 - s], m, s, out, d_part_base, d_row_base,
+                out.counters + 1);
+        } else {
+            join_probe_kernel<<<blocks, 256, 0, stream>>>(b, side[s],
+                                                          side[1 - s], m, s,
+                                                          out, r0, r1,
+                                                          dbg_skip);
+        }
         if (timed) {
             HIP_TRY(hipEventRecord(ev1[slot], stream));
             ev_pending[slot] = 1;
@@ -4415,6 +4774,11 @@ struct HashJoin {
                 hipEventDestroy(ev0[s]);
                 hipEventDestroy(ev1[s]);
             }
+        if (d_pcount) {
+            hipFree(d_pcount);
+            hipFree(d_part_base);
+            hipFree(d_row_base);
+        }
         for (int s = 0; s < 2; s++) {
             JoinSideDev& js = side[s];
             if (js.slots) {

@@ -1116,3 +1116,64 @@ def test_topn_checkpoint_spill_parity():
         assert sg == so, f"epoch {ep}: {len(sg)} vs {len(so)} bytes"
     g.close()
     o.close()
+
+
+def test_join_partitioned_pipeline_parity():
+    # The hash-prefix-partitioned probe/insert pipeline (jpart_* kernels)
+    # engages only for all-Insert unique-key inner batches of >= 131072
+    # rows (rw_amd.hip can_partition). Drive it with q8-shaped batches on
+    # BOTH sides — so partitioned inserts build the chains that later
+    # partitioned probes walk — and compare output multisets AND the
+    # checkpoint spill bytes (which read back the scattered row records)
+    # against the oracle.
+    rng = np.random.default_rng(77)
+    t4 = [T_I64, T_TS, T_TS, T_I64]
+    kw = dict(key_l=[0, 1, 2], key_r=[0, 1, 2], pk_l=[3], pk_r=[3])
+    g = ffi.HashJoin(gpu(), JOIN_INNER, t4, t4, **kw)
+    o = ffi.HashJoin(oracle(), JOIN_INNER, t4, t4, **kw)
+    N = 140_000  # > JPART_MIN_ROWS
+    rowid = 0
+
+    def batch(ids, null_every=0):
+        nonlocal rowid
+        n = len(ids)
+        ws = (ids % 7) * 10_000_000
+        we = ws + 10_000_000
+        rid = np.arange(rowid, rowid + n)
+        rowid += n
+        valid = np.ones(n, np.uint8)
+        if null_every:
+            # null join keys (never-match under non-null-safe): distinct
+            # (ws, we) keeps the batch's keys pairwise distinct so the
+            # partitioned path still engages
+            valid[::null_every] = 0
+        return ffi.Chunk(t4, np.zeros(n, np.uint8), [ids, ws, we, rid],
+                         [valid, np.ones(n, np.uint8), np.ones(n, np.uint8),
+                          np.ones(n, np.uint8)])
+
+    # build side: two partitioned insert batches (distinct ids, no matches)
+    perm = rng.permutation(400_000).astype(np.int64)
+    pushes = [
+        (SIDE_RIGHT, batch(perm[:N])),
+        (SIDE_RIGHT, batch(perm[N:2 * N])),
+        # probe batches: ~half the keys hit the build side; some null keys
+        (SIDE_LEFT, batch(np.concatenate([perm[:N // 2],
+                                          perm[2 * N:2 * N + N // 2]]),
+                          null_every=997)),
+        # second probe batch repeats earlier LEFT keys -> left chains of
+        # length 2, then a RIGHT batch probes those chains
+        (SIDE_LEFT, batch(perm[:N // 2 + N // 2])),
+        (SIDE_RIGHT, batch(perm[2 * N + N: 2 * N + N + N // 4])),
+    ]
+    for i, (side, c) in enumerate(pushes):
+        g.push(side, c)
+        o.push(side, c)
+        mg = rows_multiset(g.poll_all())
+        mo = rows_multiset(o.poll_all())
+        assert mg == mo, f"push {i}: GPU {len(mg)} rows vs oracle {len(mo)}"
+    for side in (SIDE_LEFT, SIDE_RIGHT):
+        sg = ffi.join_checkpoint_drain(gpu(), g.h, side)
+        so = ffi.join_checkpoint_drain(ffi.oracle(), o.h, side)
+        assert sg == so, (f"side {side}: spill {len(sg)} vs {len(so)} bytes")
+    g.close()
+    o.close()
