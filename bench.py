@@ -585,9 +585,11 @@ def bench_q8(args, ffi, gpu_lib, rng, rank, world, dist):
     BUILD_KEYS = 10_000_000
     batch_rows = CHUNK_ROWS * CHUNKS_PER_BATCH
     t4 = [T_I64, T_TS, T_TS, T_I64]
+    # hint 2^23 -> cap 2^24 slots x 8 B = 134 MB per side: BOTH slot tables
+    # sit in the 256 MB Infinity Cache (load factor 0.6 at 10M keys)
     j = ffi.HashJoin(gpu_lib, JOIN_INNER, t4, t4, key_l=[0, 1, 2],
                      key_r=[0, 1, 2], pk_l=[3], pk_r=[3],
-                     state_capacity_hint=1 << 24,
+                     state_capacity_hint=1 << 23,
                      row_capacity_hint=BUILD_KEYS + (args.steps + args.warmup + 4)
                      * batch_rows + 1_000_000)
 

@@ -222,7 +222,8 @@ struct JoinRowHdr {
 };
 
 struct JoinSideDev {
-    JoinSlot* slots;
+    JoinSlot* slots;     // 64-B AoS slots (GroupTopN tables)
+    uint64_t* slots8;    // compact 8-B slots (join executor tables; see below)
     uint32_t cap_mask;
     uint8_t* rows;       // packed records, row_stride bytes each (16-B aligned)
     uint32_t row_stride;  // 16 + 8*n_cols
@@ -253,6 +254,133 @@ __device__ __forceinline__ long long* jvals(JoinRowHdr* h) {
 // (jpart_* kernels below). Distribution-equivalent to h & cap_mask.
 __device__ __forceinline__ uint32_t jslot_start(uint64_t h, uint32_t cap_mask) {
     return (uint32_t)(h >> (64 - __popc(cap_mask)));
+}
+
+// ---- compact 8-B join slots -------------------------------------------
+// The join executor's slot table packs {tag u32, head u32} into one u64:
+// tag = a low hash slice (forced nonzero), head = chain-head row index;
+// the KEY WORDS live only in the row records. Rationale (round-2 roofline
+// work): the round-1 64-B JoinSlot table was 2 GB at q8 capacity, so every
+// probe/insert slot touch was an HBM row miss (PMC: 500–1000 B/row vs the
+// 128-B model). At 8 B/slot the whole q8 table is 128–256 MB — it sits in
+// the 256 MB Infinity Cache and the random slot touch becomes an LIC hit.
+// A tag match is verified against the chain-head record's key columns
+// (which a real match reads anyway); a tag mismatch probes on with no
+// record read. packed==0 means empty; claim+publish is ONE 64-bit CAS (no
+// CLAIMED spin state). A cleaned chain keeps its {tag, head} with dead
+// records — probes walk past it by key verification, never by slot state.
+
+#define JTAG(h) ((uint32_t)(h) | 1u)
+
+__device__ __forceinline__ uint64_t jpack(uint32_t tag, uint32_t head) {
+    return ((uint64_t)head << 32) | tag;
+}
+__device__ __forceinline__ uint32_t jhead_of(uint64_t packed) {
+    return (uint32_t)(packed >> 32);
+}
+
+// does record `row`'s key (columns key_cols[0..KW)) equal (kw, nullmask)?
+// sc1: coherent loads for records that may be written concurrently in this
+// launch (every publish drains the payload to the coherence point first,
+// so a visible head implies fetchable keys).
+__device__ __forceinline__ bool jrec_key_eq(const JoinSideDev& sd,
+                                            const uint8_t* key_cols, int KW,
+                                            uint32_t row, const int64_t* kw,
+                                            uint32_t nullmask, bool sc1) {
+    JoinRowHdr* hd = jrow(sd, row);
+    uint32_t vb = sc1 ? ld_u32(&hd->validbits) : hd->validbits;
+    const long long* hv = (const long long*)((const uint8_t*)hd + 16);
+    for (int i = 0; i < KW; i++) {
+        uint8_t col = key_cols[i];
+        bool valid = (vb >> col) & 1;
+        if (valid == (bool)((nullmask >> i) & 1)) return false;
+        if (valid) {
+            long long v = sc1 ? ld_i64((const int64_t*)&hv[col]) : hv[col];
+            if (v != kw[i]) return false;
+        }
+    }
+    return true;
+}
+
+// find the slot of key (kw, nullmask); UINT32_MAX if absent. sc1 = coherent
+// slot loads (the side is mutated during this launch).
+__device__ __forceinline__ uint32_t jslot8_find(const JoinSideDev& sd,
+                                                const uint8_t* key_cols,
+                                                int KW, const int64_t* kw,
+                                                uint32_t nullmask, uint64_t h,
+                                                bool sc1) {
+    const uint64_t* slots = sd.slots8;
+    uint32_t cap_mask = sd.cap_mask;
+    uint32_t tag = JTAG(h);
+    uint32_t slot = jslot_start(h, cap_mask);
+    for (uint32_t probes = 0; probes <= cap_mask; probes++) {
+        uint64_t packed = sc1 ? __hip_atomic_load((uint64_t*)&slots[slot], RLX)
+                              : slots[slot];
+        if (packed == 0) return UINT32_MAX;
+        if ((uint32_t)packed == tag &&
+            jrec_key_eq(sd, key_cols, KW, jhead_of(packed), kw, nullmask, sc1))
+            return slot;
+        slot = (slot + 1) & cap_mask;
+    }
+    return UINT32_MAX;
+}
+
+// find-or-insert + chain push of `row` (whose record is fully written
+// except `next`, which is linked here). plain_push: unique-key all-Insert
+// batches — claims still CAS (distinct keys can race one empty slot) but
+// the push of an existing chain is a plain RMW (sole writer per key).
+// drain_payload: the record was written with sc1 stores IN THIS LAUNCH and
+// same-launch readers exist — drain before the publishing CAS (R1).
+// Returns 0, or -1 on table full.
+__device__ __forceinline__ int jslot8_insert(const JoinSideDev& sd,
+                                             const uint8_t* key_cols, int KW,
+                                             const int64_t* kw,
+                                             uint32_t nullmask, uint64_t h,
+                                             uint32_t row, bool plain_push,
+                                             bool drain_payload) {
+    uint64_t* slots = sd.slots8;
+    uint32_t cap_mask = sd.cap_mask;
+    uint32_t tag = JTAG(h);
+    uint32_t slot = jslot_start(h, cap_mask);
+    JoinRowHdr* hd = jrow(sd, row);
+    for (uint32_t probes = 0; probes <= cap_mask; probes++) {
+        uint64_t packed = __hip_atomic_load(&slots[slot], RLX);
+        if (packed == 0) {
+            // publish order: payload (and next) at the coherence point
+            // BEFORE the head becomes visible, so a concurrent
+            // tag-collision verify that sees this head fetches real keys.
+            // The drain covers this thread's earlier payload stores too
+            // (cost ~nil, round-1 A/B).
+            if (drain_payload) st_u32(&hd->next, UINT32_MAX);
+            else hd->next = UINT32_MAX;
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); // R1 drain
+            uint64_t prev = atomicCAS((unsigned long long*)&slots[slot], 0ull,
+                                      jpack(tag, row));
+            if (prev == 0) return 0;
+            packed = prev;
+        }
+        if ((uint32_t)packed == tag &&
+            jrec_key_eq(sd, key_cols, KW, jhead_of(packed), kw, nullmask,
+                        drain_payload)) {
+            if (plain_push) {
+                hd->next = jhead_of(packed);
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+                slots[slot] = jpack(tag, row);
+                return 0;
+            }
+            for (;;) {
+                if (drain_payload) st_u32(&hd->next, jhead_of(packed));
+                else hd->next = jhead_of(packed);
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+                uint64_t prev = atomicCAS((unsigned long long*)&slots[slot],
+                                          packed, jpack(tag, row));
+                if (prev == packed) return 0;
+                packed = prev; // only our key's chain pushes here
+            }
+        }
+        slot = (slot + 1) & cap_mask;
+    }
+    return -1;
 }
 
 // own-side find-or-insert. The probe walk uses PLAIN cached loads: a slot's
@@ -3101,12 +3229,6 @@ __device__ __forceinline__ void jown_insert(JoinSideDev own, const JoinMeta& m,
                                             uint32_t r, const int64_t* kw,
                                             uint32_t nullmask, JoinOutDev out,
                                             uint32_t init_deg) {
-    uint32_t own_slot =
-        jslot_find_or_insert(own.slots, own.cap_mask, kw, nullmask, m.KW);
-    if (own_slot == UINT32_MAX) {
-        atomicExch(&out.counters[1], 2u); // key table full
-        return;
-    }
     // wave-aggregated row reservation: the single cursor sustains only
     // ~2.1 G same-line atomicAdds/s, so 1M per-lane adds cost ~0.47 ms per
     // launch (measured via RW_JOIN_SKIP) — one leader add per wave instead
@@ -3122,14 +3244,15 @@ __device__ __forceinline__ void jown_insert(JoinSideDev own, const JoinMeta& m,
         atomicExch(&out.counters[1], 3u); // row store full
         return;
     }
+    uint64_t h64 = hash_key(kw, nullmask, m.KW);
     JoinRowHdr* h = jrow(own, row);
     uint32_t vb = 0;
     long long* hv = jvals(h);
+    int rc;
     if (b.all_insert) {
         // plain cached stores: nothing walks the own side within this
-        // launch (no deletes), and the kernel boundary flushes before the
-        // next launch reads. The sc1 word-store variant below costs
-        // ~0.5 ms/1M rows (measured via RW_JOIN_SKIP).
+        // launch (no deletes); jslot8_insert drains them to the coherence
+        // point before publishing the head.
         for (int c = 0; c < m.n_cols[S]; c++) {
             hv[c] = b.col_vals[c][r];
             vb |= (uint32_t)(b.col_valid[c][r] != 0) << c;
@@ -3137,21 +3260,9 @@ __device__ __forceinline__ void jown_insert(JoinSideDev own, const JoinMeta& m,
         h->validbits = vb;
         h->degree = init_deg;
         h->alive = 1;
-        uint32_t* headp = &own.slots[own_slot].head;
-        if (b.unique_keys) {
-            // sole writer of this slot's chain in this launch: plain
-            // read-modify-write (1M random-line CASes cost ~0.45 ms/launch)
-            h->next = *headp;
-            *headp = row;
-        } else {
-            uint32_t old_head = ld_u32(headp);
-            for (;;) {
-                h->next = old_head;
-                uint32_t prev = atomicCAS(headp, old_head, row);
-                if (prev == old_head) break;
-                old_head = prev;
-            }
-        }
+        rc = jslot8_insert(own, m.key_cols[S], m.KW, kw, nullmask, h64, row,
+                           /*plain_push=*/b.unique_keys != 0,
+                           /*drain_payload=*/false);
     } else {
         for (int c = 0; c < m.n_cols[S]; c++) {
             st_i64((int64_t*)&hv[c], b.col_vals[c][r]);
@@ -3160,18 +3271,10 @@ __device__ __forceinline__ void jown_insert(JoinSideDev own, const JoinMeta& m,
         st_u32(&h->validbits, vb);
         st_u32(&h->degree, init_deg);
         st_u32(&h->alive, 1);
-        // lock-free chain push: next set BEFORE the CAS publish, payload
-        // R1-drained ahead of it (same-launch deletes may walk this)
-        uint32_t* headp = &own.slots[own_slot].head;
-        uint32_t old_head = ld_u32(headp);
-        for (;;) {
-            st_u32(&h->next, old_head);
-            asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); // R1 drain
-            uint32_t prev = atomicCAS(headp, old_head, row);
-            if (prev == old_head) break;
-            old_head = prev;
-        }
+        rc = jslot8_insert(own, m.key_cols[S], m.KW, kw, nullmask, h64, row,
+                           false, /*drain_payload=*/true);
     }
+    if (rc != 0) atomicExch(&out.counters[1], 2u); // key table full
 }
 
 __device__ __forceinline__ void jown_delete(JoinSideDev own, const JoinMeta& m,
@@ -3179,10 +3282,11 @@ __device__ __forceinline__ void jown_delete(JoinSideDev own, const JoinMeta& m,
                                             uint32_t r, const int64_t* kw,
                                             uint32_t nullmask) {
     // delete own row: FULL-row compare + CAS claim
-    uint32_t own_slot =
-        jslot_find_sc1(own.slots, own.cap_mask, kw, nullmask, m.KW);
+    uint32_t own_slot = jslot8_find(own, m.key_cols[S], m.KW, kw, nullmask,
+                                    hash_key(kw, nullmask, m.KW), true);
     if (own_slot == UINT32_MAX) return;
-    uint32_t row = ld_u32(&own.slots[own_slot].head);
+    uint32_t row =
+        jhead_of(__hip_atomic_load(&own.slots8[own_slot], RLX));
     while (row != UINT32_MAX) {
         JoinRowHdr* h = jrow(own, row);
         if (ld_u32(&h->alive)) {
@@ -3233,11 +3337,12 @@ __device__ void join_probe_row_noninner(const JoinBatchDev& b, JoinSideDev own,
             jemit_row(out, m, S, op, b, r, 0, nullptr, 1);
         return; // no state write
     }
-    uint32_t mslot =
-        jslot_find_cached(match.slots, match.cap_mask, kw, nullmask, m.KW);
+    uint32_t mslot = jslot8_find(match, m.key_cols[1 - S], m.KW, kw,
+                                 nullmask, hash_key(kw, nullmask, m.KW),
+                                 false);
     uint32_t my_deg = 0;
     if (mslot != UINT32_MAX) {
-        uint32_t row = match.slots[mslot].head;
+        uint32_t row = jhead_of(match.slots8[mslot]);
         while (row != UINT32_MAX) {
             JoinRowHdr* h = jrow(match, row);
             if (h->alive && join_cond_ok(m, S, b, r, h->validbits, jvals(h))) {
@@ -3346,15 +3451,11 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
         uint32_t my_n = 0;
         uint32_t matched_row = UINT32_MAX;
         if (active) {
-            if (m.append_only) {
-                mslot = jslot_find_sc1(match.slots, match.cap_mask, kw, nullmask,
-                                       m.KW);
-            } else {
-                mslot = jslot_find_cached(match.slots, match.cap_mask, kw,
-                                          nullmask, m.KW);
-            }
+            mslot = jslot8_find(match, m.key_cols[1 - S], m.KW, kw,
+                                nullmask, hash_key(kw, nullmask, m.KW),
+                                m.append_only != 0);
             if (mslot != UINT32_MAX && !m.append_only) {
-                uint32_t row = match.slots[mslot].head;
+                uint32_t row = jhead_of(match.slots8[mslot]);
                 while (row != UINT32_MAX) {
                     JoinRowHdr* h = jrow(match, row);
                     if (h->alive &&
@@ -3384,7 +3485,7 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
                 if (lane == 0) atomicExch(&out.counters[1], 1u); // overflow
             } else if (my_n && !(dbg_skip & 1)) {
                 // second walk: emit (JoinStreamChunkBuilder::append_row)
-                uint32_t row = match.slots[mslot].head;
+                uint32_t row = jhead_of(match.slots8[mslot]);
                 uint32_t k = 0;
                 while (row != UINT32_MAX && k < my_n) {
                     JoinRowHdr* h = jrow(match, row);
@@ -3415,7 +3516,8 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
             }
         } else if (active && mslot != UINT32_MAX) {
             // append-only path: <=1 match; sc1 reads, per-match atomics
-            uint32_t row = ld_u32(&match.slots[mslot].head);
+            uint32_t row =
+                jhead_of(__hip_atomic_load(&match.slots8[mslot], RLX));
             while (row != UINT32_MAX) {
                 JoinRowHdr* h = jrow(match, row);
                 uint32_t vb = ld_u32(&h->validbits);
@@ -3525,8 +3627,12 @@ __device__ __forceinline__ bool jpart_row_key(const JoinBatchDev& b,
     return !(nm & ~(uint32_t)m.null_safe_mask);
 }
 
+// one counter per 64-B line: same-line atomics from different CUs
+// serialize at the line's home (~12.6 ns each, round-1 measurement)
+#define JPART_PAD 16
+
 __global__ void jpart_count_kernel(JoinBatchDev b, JoinMeta m, int S,
-                                   uint32_t* pcount) {
+                                   uint32_t* ptot) {
     __shared__ uint32_t hist[JPART_P];
     for (uint32_t i = threadIdx.x; i < JPART_P; i += blockDim.x) hist[i] = 0;
     __syncthreads();
@@ -3542,23 +3648,22 @@ __global__ void jpart_count_kernel(JoinBatchDev b, JoinMeta m, int S,
     }
     __syncthreads();
     for (uint32_t p = threadIdx.x; p < JPART_P; p += blockDim.x)
-        pcount[(size_t)blockIdx.x * JPART_P + p] = hist[p];
+        if (hist[p]) atomicAdd(&ptot[(size_t)p * JPART_PAD], hist[p]);
 }
 
 // single-workgroup scan: per-partition totals → exclusive bases; rewrites
 // pcount[blk][p] into per-block scatter bases; reserves the batch's row
 // range on the device cursor (err code 3 = row store full, as the
 // wave-aggregated reservation in jown_insert)
-__global__ void jpart_scan_kernel(uint32_t* pcount, int nblk,
+__global__ void jpart_scan_kernel(uint32_t* ptot, uint32_t* pcur,
                                   uint32_t* part_base, uint32_t* row_cursor,
                                   uint32_t row_cap, uint32_t* row_base,
                                   uint32_t* err) {
     __shared__ uint32_t tot[JPART_P];
     __shared__ uint32_t excl[JPART_P];
     for (uint32_t p = threadIdx.x; p < JPART_P; p += blockDim.x) {
-        uint32_t s = 0;
-        for (int bk = 0; bk < nblk; bk++)
-            s += pcount[(size_t)bk * JPART_P + p];
+        uint32_t s = ptot[(size_t)p * JPART_PAD];
+        ptot[(size_t)p * JPART_PAD] = 0; // ready for the next batch
         tot[p] = s;
         excl[p] = s;
     }
@@ -3584,25 +3689,17 @@ __global__ void jpart_scan_kernel(uint32_t* pcount, int nblk,
     }
     __syncthreads();
     for (uint32_t p = threadIdx.x; p < JPART_P; p += blockDim.x) {
-        uint32_t run = excl[p] - tot[p]; // exclusive base of partition p
-        part_base[p] = run;
-        for (int bk = 0; bk < nblk; bk++) {
-            uint32_t c = pcount[(size_t)bk * JPART_P + p];
-            pcount[(size_t)bk * JPART_P + p] = run;
-            run += c;
-        }
+        uint32_t base = excl[p] - tot[p]; // exclusive base of partition p
+        part_base[p] = base;
+        pcur[(size_t)p * JPART_PAD] = base; // scatter cursors
     }
 }
 
 __global__ void jpart_scatter_kernel(JoinBatchDev b, JoinMeta m, int S,
-                                     JoinSideDev own, uint32_t* pcount,
+                                     JoinSideDev own, uint32_t* pcur,
                                      const uint32_t* row_base,
                                      const uint32_t* err) {
     if (*err) return;
-    __shared__ uint32_t cur[JPART_P];
-    for (uint32_t p = threadIdx.x; p < JPART_P; p += blockDim.x)
-        cur[p] = pcount[(size_t)blockIdx.x * JPART_P + p];
-    __syncthreads();
     uint32_t rb = *row_base;
     uint32_t n = b.n_rows;
     uint32_t per = (n + gridDim.x - 1) / gridDim.x;
@@ -3613,7 +3710,7 @@ __global__ void jpart_scatter_kernel(JoinBatchDev b, JoinMeta m, int S,
     for (uint32_t r = r0 + threadIdx.x; r < r1; r += blockDim.x) {
         if (!jpart_row_key(b, m, S, r, kw, &nm)) continue;
         uint32_t p = jpart_of(hash_key(kw, nm, m.KW));
-        uint32_t row = rb + atomicAdd(&cur[p], 1u);
+        uint32_t row = rb + atomicAdd(&pcur[(size_t)p * JPART_PAD], 1u);
         // final row record (plain cached stores: the next kernel on the
         // stream observes them through the inter-dispatch cache flush)
         JoinRowHdr* hd = jrow(own, row);
@@ -3694,10 +3791,10 @@ __global__ __launch_bounds__(256) void jpart_probe_insert_kernel(
         uint32_t mslot = UINT32_MAX;
         uint32_t my_n = 0;
         if (active) {
-            mslot = jslot_find_cached(match.slots, match.cap_mask, kw, nm,
-                                      m.KW);
+            mslot = jslot8_find(match, m.key_cols[1 - S], m.KW, kw, nm,
+                                hash_key(kw, nm, m.KW), false);
             if (mslot != UINT32_MAX) {
-                uint32_t mr = match.slots[mslot].head;
+                uint32_t mr = jhead_of(match.slots8[mslot]);
                 while (mr != UINT32_MAX) {
                     JoinRowHdr* mh = jrow(match, mr);
                     if (mh->alive &&
@@ -3722,7 +3819,7 @@ __global__ __launch_bounds__(256) void jpart_probe_insert_kernel(
         if (total && base + total > out.cap) {
             if (lane == 0) atomicExch(&out.counters[1], 1u);
         } else if (my_n) {
-            uint32_t mr = match.slots[mslot].head;
+            uint32_t mr = jhead_of(match.slots8[mslot]);
             uint32_t k = 0;
             while (mr != UINT32_MAX && k < my_n) {
                 JoinRowHdr* mh = jrow(match, mr);
@@ -3755,15 +3852,12 @@ __global__ __launch_bounds__(256) void jpart_probe_insert_kernel(
         // Keys are pairwise distinct in the batch, so each touched slot has
         // exactly one linking thread — plain read-modify-write.
         if (active) {
-            uint32_t os = jslot_find_or_insert(own.slots, own.cap_mask, kw,
-                                               nm, m.KW);
-            if (os == UINT32_MAX) {
+            // record already in place from the scatter kernel (globally
+            // visible across the dispatch boundary); link the chain. Keys
+            // are pairwise distinct in the batch -> plain push.
+            if (jslot8_insert(own, m.key_cols[S], m.KW, kw, nm,
+                              hash_key(kw, nm, m.KW), row, true, false) != 0)
                 atomicExch(&out.counters[1], 2u);
-            } else {
-                uint32_t* headp = &own.slots[os].head;
-                hd->next = *headp;
-                *headp = row;
-            }
         }
     }
 }
@@ -3772,23 +3866,28 @@ __global__ __launch_bounds__(256) void jpart_probe_insert_kernel(
 // rows/groups whose watermarked key column sorts below the value are
 // retired in place (slots stay READY so linear probing is undisturbed;
 // memory reclamation is compaction work for a later round)
-__global__ void join_clean_kernel(JoinSideDev sd, int kpos, long long wm, int KW) {
+__global__ void join_clean_kernel(JoinSideDev sd, int kcol, long long wm,
+                                  int KW) {
+    // kcol = RECORD column of the watermarked key position (8-B slots keep
+    // no key words; the chain head identifies the key). The chain keeps its
+    // slot entry — probes skip dead records by the alive flag.
     (void)KW;
     size_t cap = (size_t)sd.cap_mask + 1;
     size_t stride = (size_t)gridDim.x * blockDim.x;
     for (size_t slot = blockIdx.x * blockDim.x + threadIdx.x; slot < cap;
          slot += stride) {
-        JoinSlot* sl = &sd.slots[slot];
-        if (ld_u32(&sl->state) != SLOT_READY) continue;
-        if ((ld_u32(&sl->nulls) >> kpos) & 1) continue; // NULLs largest
-        if (ld_i64((const int64_t*)&sl->key[kpos]) >= wm) continue;
-        uint32_t row = ld_u32(&sl->head);
+        uint64_t packed = __hip_atomic_load(&sd.slots8[slot], RLX);
+        if (packed == 0) continue;
+        uint32_t head = jhead_of(packed);
+        JoinRowHdr* hh = jrow(sd, head);
+        if (!((ld_u32(&hh->validbits) >> kcol) & 1)) continue; // NULLs largest
+        if (ld_i64((const int64_t*)&jvals(hh)[kcol]) >= wm) continue;
+        uint32_t row = head;
         while (row != UINT32_MAX) {
             JoinRowHdr* h = jrow(sd, row);
             st_u32(&h->alive, 0);
             row = ld_u32(&h->next);
         }
-        st_u32(&sl->head, UINT32_MAX);
     }
 }
 
@@ -4098,28 +4197,37 @@ __global__ void agg_vnode_scope_kernel(AggTableDev t, int KW, int n_calls,
 __global__ void join_vnode_scope_kernel(JoinSideDev sd, int KW,
                                         const uint8_t* bitmap,
                                         uint32_t vnode_count, uint8_t t0,
-                                        uint8_t t1, uint8_t t2, uint8_t t3) {
+                                        uint8_t t1, uint8_t t2, uint8_t t3,
+                                        uint8_t k0, uint8_t k1, uint8_t k2,
+                                        uint8_t k3) {
     __shared__ uint32_t lut[256];
     stage_crc_lut(lut);
     uint8_t types[4] = {t0, t1, t2, t3};
+    uint8_t kcols[4] = {k0, k1, k2, k3};
     size_t cap = (size_t)sd.cap_mask + 1;
     size_t stride = (size_t)gridDim.x * blockDim.x;
     for (size_t slot = blockIdx.x * blockDim.x + threadIdx.x; slot < cap;
          slot += stride) {
-        JoinSlot* sl = &sd.slots[slot];
-        if (ld_u32(&sl->state) != SLOT_READY) continue;
+        uint64_t packed = __hip_atomic_load(&sd.slots8[slot], RLX);
+        if (packed == 0) continue;
+        JoinRowHdr* hh = jrow(sd, jhead_of(packed));
+        uint32_t vb = ld_u32(&hh->validbits);
+        const long long* hv = jvals(hh);
         int64_t kw[4];
-        for (int k = 0; k < KW; k++) kw[k] = ld_i64((const int64_t*)&sl->key[k]);
-        uint32_t vn =
-            crc_key_words(lut, kw, ld_u32(&sl->nulls), KW, types) % vnode_count;
+        uint32_t nulls = 0;
+        for (int k = 0; k < KW; k++) {
+            bool valid = (vb >> kcols[k]) & 1;
+            kw[k] = valid ? ld_i64((const int64_t*)&hv[kcols[k]]) : 0;
+            nulls |= (uint32_t)(!valid) << k;
+        }
+        uint32_t vn = crc_key_words(lut, kw, nulls, KW, types) % vnode_count;
         if ((bitmap[vn >> 3] >> (vn & 7)) & 1) continue;
-        uint32_t row = ld_u32(&sl->head);
+        uint32_t row = jhead_of(packed);
         while (row != UINT32_MAX) {
             JoinRowHdr* h = jrow(sd, row);
             st_u32(&h->alive, 0);
             row = ld_u32(&h->next);
         }
-        st_u32(&sl->head, UINT32_MAX);
     }
 }
 
@@ -4241,8 +4349,9 @@ struct HashJoin {
             while (cap < key_cap * 2) cap <<= 1; // ≤50% load factor
             JoinSideDev& js = side[s];
             js.cap_mask = cap - 1;
-            HIP_TRY(hipMalloc(&js.slots, (size_t)cap * sizeof(JoinSlot)));
-            jslot_init_kernel<<<2048, 256, 0, stream>>>(js.slots, cap);
+            js.slots = nullptr; // join tables use compact 8-B slots
+            HIP_TRY(hipMalloc(&js.slots8, (size_t)cap * 8));
+            HIP_TRY(hipMemsetAsync(js.slots8, 0, (size_t)cap * 8, stream));
             js.row_cap = (uint32_t)row_cap;
             js.row_stride = 16 + 8 * (uint32_t)m.n_cols[s];
             HIP_TRY(hipMalloc(&js.rows, (size_t)row_cap * js.row_stride));
@@ -4340,8 +4449,9 @@ struct HashJoin {
         return RW_OK;
     }
 
-    // partitioned-pipeline device buffers (lazily allocated; ~2 MB)
-    uint32_t* d_pcount = nullptr;
+    // partitioned-pipeline device buffers (lazily allocated; ~260 KB)
+    uint32_t* d_ptot = nullptr;  // padded per-partition totals (count)
+    uint32_t* d_pcur = nullptr;  // padded per-partition scatter cursors
     uint32_t* d_part_base = nullptr;
     uint32_t* d_row_base = nullptr;
 
@@ -4367,8 +4477,11 @@ struct HashJoin {
     }
 
     int ensure_part_bufs() {
-        if (d_pcount) return RW_OK;
-        HIP_TRY(hipMalloc(&d_pcount, (size_t)JPART_NBLK * JPART_P * 4));
+        if (d_ptot) return RW_OK;
+        HIP_TRY(hipMalloc(&d_ptot, (size_t)JPART_P * JPART_PAD * 4));
+        HIP_TRY(hipMemsetAsync(d_ptot, 0, (size_t)JPART_P * JPART_PAD * 4,
+                               stream));
+        HIP_TRY(hipMalloc(&d_pcur, (size_t)JPART_P * JPART_PAD * 4));
         HIP_TRY(hipMalloc(&d_part_base, (size_t)(JPART_P + 1) * 4));
         HIP_TRY(hipMalloc(&d_row_base, 4));
         return RW_OK;
@@ -4396,12 +4509,12 @@ struct HashJoin {
             int rc = ensure_part_bufs();
             if (rc != RW_OK) return rc;
             jpart_count_kernel<<<JPART_NBLK, 256, 0, stream>>>(b, m, s,
-                                                               d_pcount);
+                                                               d_ptot);
             jpart_scan_kernel<<<1, 1024, 0, stream>>>(
-                d_pcount, JPART_NBLK, d_part_base, side[s].row_cursor,
+                d_ptot, d_pcur, d_part_base, side[s].row_cursor,
                 side[s].row_cap, d_row_base, out.counters + 1);
             jpart_scatter_kernel<<<JPART_NBLK, 256, 0, stream>>>(
-                b, m, s, side[s], d_pcount, d_row_base, out.counters + 1);
+                b, m, s, side[s], d_pcur, d_row_base, out.counters + 1);
             jpart_probe_insert_kernel<<<JPART_P, 256, 0, stream>>>(
                 side[s], side[1 - s], m, s, out, d_part_base, d_row_base,
                 out.counters + 1);
@@ -4736,9 +4849,8 @@ struct HashJoin {
                 if (wm_pos[w] == idx && wm_clean[w]) clean = true;
             if (clean) {
                 for (int sd = 0; sd < 2; sd++)
-                    join_clean_kernel<<<2048, 256, 0, stream>>>(side[sd],
-                                                                (int)idx, sel,
-                                                                m.KW);
+                    join_clean_kernel<<<2048, 256, 0, stream>>>(
+                        side[sd], (int)m.key_cols[sd][idx], sel, m.KW);
                 HIP_TRY(hipStreamSynchronize(stream));
             }
             // update side's output columns first, then the match side's
@@ -4774,15 +4886,16 @@ struct HashJoin {
                 hipEventDestroy(ev0[s]);
                 hipEventDestroy(ev1[s]);
             }
-        if (d_pcount) {
-            hipFree(d_pcount);
+        if (d_ptot) {
+            hipFree(d_ptot);
+            hipFree(d_pcur);
             hipFree(d_part_base);
             hipFree(d_row_base);
         }
         for (int s = 0; s < 2; s++) {
             JoinSideDev& js = side[s];
-            if (js.slots) {
-                hipFree(js.slots);
+            if (js.slots8) {
+                hipFree(js.slots8);
                 hipFree(js.rows);
                 hipFree(js.row_cursor);
                 if (js.killed) {
@@ -4886,11 +4999,14 @@ int rw_hash_join_update_vnode_bitmap(void* h, const uint8_t* bitmap,
                            hipMemcpyHostToDevice, j->stream));
     for (int s = 0; s < 2; s++) {
         uint8_t ty[4] = {RW_T_I64, RW_T_I64, RW_T_I64, RW_T_I64};
-        for (int k = 0; k < j->m.KW && k < 4; k++)
+        uint8_t kc[4] = {0, 0, 0, 0};
+        for (int k = 0; k < j->m.KW && k < 4; k++) {
             ty[k] = j->types[s][j->m.key_cols[s][k]];
+            kc[k] = j->m.key_cols[s][k];
+        }
         join_vnode_scope_kernel<<<2048, 256, 0, j->stream>>>(
             j->side[s], j->m.KW, j->d_vnode_bitmap, vnode_count, ty[0], ty[1],
-            ty[2], ty[3]);
+            ty[2], ty[3], kc[0], kc[1], kc[2], kc[3]);
     }
     HIP_TRY(hipStreamSynchronize(j->stream));
     return RW_OK;
