@@ -361,7 +361,7 @@ __device__ __forceinline__ int jslot8_insert(const JoinSideDev& sd,
         }
         if ((uint32_t)packed == tag &&
             jrec_key_eq(sd, key_cols, KW, jhead_of(packed), kw, nullmask,
-                        drain_payload)) {
+                        /*sc1=*/!plain_push)) {
             if (plain_push) {
                 hd->next = jhead_of(packed);
                 asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -3249,10 +3249,13 @@ __device__ __forceinline__ void jown_insert(JoinSideDev own, const JoinMeta& m,
     uint32_t vb = 0;
     long long* hv = jvals(h);
     int rc;
-    if (b.all_insert) {
-        // plain cached stores: nothing walks the own side within this
-        // launch (no deletes); jslot8_insert drains them to the coherence
-        // point before publishing the head.
+    if (b.all_insert && b.unique_keys) {
+        // plain cached stores are only safe with UNIQUE keys: a duplicate
+        // key's second inserter must verify the first one's record, and
+        // plain stores reach only the writer's L2 (not the coherence
+        // point), so a cross-XCD verify would read stale bytes. Unique
+        // batches have no same-key verify; tag-collision verifies use sc1
+        // loads (see jslot8_insert).
         for (int c = 0; c < m.n_cols[S]; c++) {
             hv[c] = b.col_vals[c][r];
             vb |= (uint32_t)(b.col_valid[c][r] != 0) << c;
@@ -3261,8 +3264,7 @@ __device__ __forceinline__ void jown_insert(JoinSideDev own, const JoinMeta& m,
         h->degree = init_deg;
         h->alive = 1;
         rc = jslot8_insert(own, m.key_cols[S], m.KW, kw, nullmask, h64, row,
-                           /*plain_push=*/b.unique_keys != 0,
-                           /*drain_payload=*/false);
+                           /*plain_push=*/true, /*drain_payload=*/false);
     } else {
         for (int c = 0; c < m.n_cols[S]; c++) {
             st_i64((int64_t*)&hv[c], b.col_vals[c][r]);
@@ -3633,22 +3635,20 @@ __device__ __forceinline__ bool jpart_row_key(const JoinBatchDev& b,
 
 __global__ void jpart_count_kernel(JoinBatchDev b, JoinMeta m, int S,
                                    uint32_t* ptot) {
-    __shared__ uint32_t hist[JPART_P];
-    for (uint32_t i = threadIdx.x; i < JPART_P; i += blockDim.x) hist[i] = 0;
-    __syncthreads();
+    // grid-stride, no LDS: 1M adds spread over 2048 line-padded counters
+    // serialize only ~512-deep per line (~6.5 us total) and the full grid
+    // keeps every CU at occupancy (the 256-block LDS-histogram variant ran
+    // one block per CU and was latency-bound at 68 us).
     uint32_t n = b.n_rows;
-    uint32_t per = (n + gridDim.x - 1) / gridDim.x;
-    uint32_t r0 = blockIdx.x * per;
-    uint32_t r1 = r0 + per < n ? r0 + per : n;
+    uint32_t stride = gridDim.x * blockDim.x;
     int64_t kw[MAX_KW];
     uint32_t nm;
-    for (uint32_t r = r0 + threadIdx.x; r < r1; r += blockDim.x) {
+    for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < n;
+         r += stride) {
         if (!jpart_row_key(b, m, S, r, kw, &nm)) continue;
-        atomicAdd(&hist[jpart_of(hash_key(kw, nm, m.KW))], 1u);
+        atomicAdd(&ptot[(size_t)jpart_of(hash_key(kw, nm, m.KW)) * JPART_PAD],
+                  1u);
     }
-    __syncthreads();
-    for (uint32_t p = threadIdx.x; p < JPART_P; p += blockDim.x)
-        if (hist[p]) atomicAdd(&ptot[(size_t)p * JPART_PAD], hist[p]);
 }
 
 // single-workgroup scan: per-partition totals → exclusive bases; rewrites
@@ -3702,12 +3702,11 @@ __global__ void jpart_scatter_kernel(JoinBatchDev b, JoinMeta m, int S,
     if (*err) return;
     uint32_t rb = *row_base;
     uint32_t n = b.n_rows;
-    uint32_t per = (n + gridDim.x - 1) / gridDim.x;
-    uint32_t r0 = blockIdx.x * per;
-    uint32_t r1 = r0 + per < n ? r0 + per : n;
+    uint32_t stride = gridDim.x * blockDim.x;
     int64_t kw[MAX_KW];
     uint32_t nm;
-    for (uint32_t r = r0 + threadIdx.x; r < r1; r += blockDim.x) {
+    for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < n;
+         r += stride) {
         if (!jpart_row_key(b, m, S, r, kw, &nm)) continue;
         uint32_t p = jpart_of(hash_key(kw, nm, m.KW));
         uint32_t row = rb + atomicAdd(&pcur[(size_t)p * JPART_PAD], 1u);
@@ -3755,7 +3754,7 @@ __device__ __forceinline__ bool jpart_cond_ok(const JoinMeta& m, int S,
     return false;
 }
 
-__global__ __launch_bounds__(256) void jpart_probe_insert_kernel(
+__global__ __launch_bounds__(256, 8) void jpart_probe_insert_kernel(
     JoinSideDev own, JoinSideDev match, JoinMeta m, int S, JoinOutDev out,
     const uint32_t* part_base, const uint32_t* row_base,
     const uint32_t* err) {
@@ -4508,12 +4507,11 @@ struct HashJoin {
         if (can_partition(b, r0, r1)) {
             int rc = ensure_part_bufs();
             if (rc != RW_OK) return rc;
-            jpart_count_kernel<<<JPART_NBLK, 256, 0, stream>>>(b, m, s,
-                                                               d_ptot);
+            jpart_count_kernel<<<2048, 256, 0, stream>>>(b, m, s, d_ptot);
             jpart_scan_kernel<<<1, 1024, 0, stream>>>(
                 d_ptot, d_pcur, d_part_base, side[s].row_cursor,
                 side[s].row_cap, d_row_base, out.counters + 1);
-            jpart_scatter_kernel<<<JPART_NBLK, 256, 0, stream>>>(
+            jpart_scatter_kernel<<<2048, 256, 0, stream>>>(
                 b, m, s, side[s], d_pcur, d_row_base, out.counters + 1);
             jpart_probe_insert_kernel<<<JPART_P, 256, 0, stream>>>(
                 side[s], side[1 - s], m, s, out, d_part_base, d_row_base,
