@@ -3633,12 +3633,14 @@ __device__ __forceinline__ bool jpart_row_key(const JoinBatchDev& b,
 // serialize at the line's home (~12.6 ns each, round-1 measurement)
 #define JPART_PAD 16
 
-__global__ void jpart_count_kernel(JoinBatchDev b, JoinMeta m, int S,
-                                   uint32_t* ptot) {
-    // grid-stride, no LDS: 1M adds spread over 2048 line-padded counters
-    // serialize only ~512-deep per line (~6.5 us total) and the full grid
-    // keeps every CU at occupancy (the 256-block LDS-histogram variant ran
-    // one block per CU and was latency-bound at 68 us).
+__global__ __launch_bounds__(1024) void jpart_count_kernel(
+    JoinBatchDev b, JoinMeta m, int S, uint32_t* ptot) {
+    // LDS histogram per block at 1024 threads (4 waves/SIMD — the
+    // 256-thread variant ran 1 wave/SIMD and was latency-bound), then one
+    // global add per nonzero partition per block (~450K adds total).
+    __shared__ uint32_t hist[JPART_P];
+    for (uint32_t i = threadIdx.x; i < JPART_P; i += blockDim.x) hist[i] = 0;
+    __syncthreads();
     uint32_t n = b.n_rows;
     uint32_t stride = gridDim.x * blockDim.x;
     int64_t kw[MAX_KW];
@@ -3646,9 +3648,11 @@ __global__ void jpart_count_kernel(JoinBatchDev b, JoinMeta m, int S,
     for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < n;
          r += stride) {
         if (!jpart_row_key(b, m, S, r, kw, &nm)) continue;
-        atomicAdd(&ptot[(size_t)jpart_of(hash_key(kw, nm, m.KW)) * JPART_PAD],
-                  1u);
+        atomicAdd(&hist[jpart_of(hash_key(kw, nm, m.KW))], 1u);
     }
+    __syncthreads();
+    for (uint32_t p = threadIdx.x; p < JPART_P; p += blockDim.x)
+        if (hist[p]) atomicAdd(&ptot[(size_t)p * JPART_PAD], hist[p]);
 }
 
 // single-workgroup scan: per-partition totals → exclusive bases; rewrites
@@ -3710,19 +3714,30 @@ __global__ void jpart_scatter_kernel(JoinBatchDev b, JoinMeta m, int S,
         if (!jpart_row_key(b, m, S, r, kw, &nm)) continue;
         uint32_t p = jpart_of(hash_key(kw, nm, m.KW));
         uint32_t row = rb + atomicAdd(&pcur[(size_t)p * JPART_PAD], 1u);
-        // final row record (plain cached stores: the next kernel on the
-        // stream observes them through the inter-dispatch cache flush)
+        // final row record, built in registers and stored with 16-B vector
+        // writes (plain cached stores: the next kernel on the stream
+        // observes them through the inter-dispatch cache flush)
         JoinRowHdr* hd = jrow(own, row);
-        long long* hv = jvals(hd);
+        long long vals[MAX_COLS];
         uint32_t vb = 0;
-        for (int c = 0; c < m.n_cols[S]; c++) {
-            hv[c] = b.col_vals[c][r];
+        int nc = m.n_cols[S];
+        for (int c = 0; c < nc; c++) {
+            vals[c] = b.col_vals[c][r];
             vb |= (uint32_t)(b.col_valid[c][r] != 0) << c;
         }
-        hd->validbits = vb;
-        hd->degree = 0;
-        hd->alive = 1;
-        hd->next = UINT32_MAX;
+        ulonglong2* dst = (ulonglong2*)hd;
+        ulonglong2 h0;
+        h0.x = ((uint64_t)UINT32_MAX << 32) | 1u;        // alive=1, next=~0
+        h0.y = (uint64_t)vb;                              // validbits, degree=0
+        dst[0] = h0;
+        int c = 0;
+        for (; c + 1 < nc; c += 2) {
+            ulonglong2 v;
+            v.x = (uint64_t)vals[c];
+            v.y = (uint64_t)vals[c + 1];
+            dst[1 + c / 2] = v;
+        }
+        if (c < nc) jvals(hd)[c] = vals[c];
     }
 }
 
@@ -4513,7 +4528,7 @@ struct HashJoin {
         if (can_partition(b, r0, r1)) {
             int rc = ensure_part_bufs();
             if (rc != RW_OK) return rc;
-            jpart_count_kernel<<<2048, 256, 0, stream>>>(b, m, s, d_ptot);
+            jpart_count_kernel<<<256, 1024, 0, stream>>>(b, m, s, d_ptot);
             jpart_scan_kernel<<<1, 1024, 0, stream>>>(
                 d_ptot, d_pcur, d_part_base, side[s].row_cursor,
                 side[s].row_cap, d_row_base, out.counters + 1);
