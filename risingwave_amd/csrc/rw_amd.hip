@@ -256,38 +256,72 @@ __device__ __forceinline__ uint32_t jslot_start(uint64_t h, uint32_t cap_mask) {
     return (uint32_t)(h >> (64 - __popc(cap_mask)));
 }
 
-// ---- compact 8-B join slots -------------------------------------------
-// The join executor's slot table packs {tag u32, head u32} into one u64:
-// tag = a low hash slice (forced nonzero), head = chain-head row index;
-// the KEY WORDS live only in the row records. Rationale (round-2 roofline
-// work): the round-1 64-B JoinSlot table was 2 GB at q8 capacity, so every
-// probe/insert slot touch was an HBM row miss (PMC: 500–1000 B/row vs the
-// 128-B model). At 8 B/slot the whole q8 table is 128–256 MB — it sits in
-// the 256 MB Infinity Cache and the random slot touch becomes an LIC hit.
-// A tag match is verified against the chain-head record's key columns
-// (which a real match reads anyway); a tag mismatch probes on with no
-// record read. packed==0 means empty; claim+publish is ONE 64-bit CAS (no
-// CLAIMED spin state). A cleaned chain keeps its {tag, head} with dead
-// records — probes walk past it by key verification, never by slot state.
+// ---- compact 8-B join slots: CHAINED buckets with bloom tags ----------
+// The join executor's slot table packs {head u32 (high), bloom u32 (low)}
+// into one u64; bloom==0 means empty. The KEY WORDS live only in the row
+// records, and a bucket CHAINS every row whose hash lands on the slot —
+// chained hashing, not open addressing. Rationale (round-2 roofline work):
+// the round-1 64-B open-addressed slots made every probe/insert touch
+// 1–3 random HBM lines (PMC: 500–1000 B/row vs the 128-B model), and even
+// compact open addressing needs a verify read of the head record per
+// insert. With buckets:
+//  - INSERT = one 64-bit CAS on the slot (8-B table, 128–256 MB at q8
+//    capacity -> Infinity-Cache-resident) — no probing, no verify read.
+//  - MISS   = one slot load: the bloom bit (bit (h>>32)&31, set by every
+//    insert, never cleared) rejects absent keys with ~k/32 false-positive
+//    walks for a k-key bucket.
+//  - MATCH  = walk the chain and filter by FULL key from the records the
+//    match reads anyway (no tag-identity assumption anywhere).
+// A bucket may mix keys (same-slot hash collisions), so every chain walk
+// carries the key filter; deletes/cleans kill records individually.
 
-#define JTAG(h) ((uint32_t)(h) | 1u)
-
-__device__ __forceinline__ uint64_t jpack(uint32_t tag, uint32_t head) {
-    return ((uint64_t)head << 32) | tag;
+__device__ __forceinline__ uint32_t jbloom_bit(uint64_t h) {
+    return 1u << ((uint32_t)(h >> 32) & 31);
 }
 __device__ __forceinline__ uint32_t jhead_of(uint64_t packed) {
     return (uint32_t)(packed >> 32);
 }
-
-// does record `row`'s key (columns key_cols[0..KW)) equal (kw, nullmask)?
-// sc1: coherent loads for records that may be written concurrently in this
-// launch (every publish drains the payload to the coherence point first,
-// so a visible head implies fetchable keys).
-__device__ __forceinline__ bool jrec_key_eq(const JoinSideDev& sd,
-                                            const uint8_t* key_cols, int KW,
-                                            uint32_t row, const int64_t* kw,
-                                            uint32_t nullmask, bool sc1) {
+// head row of the bucket for hash h, or UINT32_MAX if the bloom rejects.
+// sc1: coherent slot load (the side is mutated during this launch).
+__device__ __forceinline__ uint32_t jbucket_head(const JoinSideDev& sd,
+                                                 uint64_t h, bool sc1) {
+    uint32_t slot = jslot_start(h, sd.cap_mask);
+    uint64_t packed = sc1 ? __hip_atomic_load(&sd.slots8[slot], RLX)
+                          : sd.slots8[slot];
+    if (!((uint32_t)packed & jbloom_bit(h))) return UINT32_MAX;
+    return jhead_of(packed);
+}
+// push row onto its bucket: record payload fully written by the caller
+// (drained to the coherence point before the CAS lands, so a walker that
+// sees the new head fetches real bytes). sc1_next: the launch has
+// same-side walkers (mixed-op chunks) — store `next` coherently.
+__device__ __forceinline__ void jbucket_insert(const JoinSideDev& sd,
+                                               uint64_t h, uint32_t row,
+                                               bool sc1_next) {
+    uint32_t slot = jslot_start(h, sd.cap_mask);
     JoinRowHdr* hd = jrow(sd, row);
+    uint64_t old = __hip_atomic_load(&sd.slots8[slot], RLX);
+    for (;;) {
+        uint32_t old_head =
+            (uint32_t)old ? jhead_of(old) : UINT32_MAX; // bloom 0 = empty
+        if (sc1_next) st_u32(&hd->next, old_head);
+        else hd->next = old_head;
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); // R1 drain
+        uint64_t want =
+            ((uint64_t)row << 32) | ((uint32_t)old | jbloom_bit(h));
+        uint64_t prev = atomicCAS((unsigned long long*)&sd.slots8[slot], old,
+                                  want);
+        if (prev == old) return;
+        old = prev;
+    }
+}
+
+// key filter for chain walks given the record header pointer (buckets mix
+// keys; every walk filters by full key)
+__device__ __forceinline__ bool jhdr_key_eq(JoinRowHdr* hd,
+                                            const uint8_t* key_cols, int KW,
+                                            const int64_t* kw,
+                                            uint32_t nullmask, bool sc1) {
     uint32_t vb = sc1 ? ld_u32(&hd->validbits) : hd->validbits;
     const long long* hv = (const long long*)((const uint8_t*)hd + 16);
     for (int i = 0; i < KW; i++) {
@@ -300,87 +334,6 @@ __device__ __forceinline__ bool jrec_key_eq(const JoinSideDev& sd,
         }
     }
     return true;
-}
-
-// find the slot of key (kw, nullmask); UINT32_MAX if absent. sc1 = coherent
-// slot loads (the side is mutated during this launch).
-__device__ __forceinline__ uint32_t jslot8_find(const JoinSideDev& sd,
-                                                const uint8_t* key_cols,
-                                                int KW, const int64_t* kw,
-                                                uint32_t nullmask, uint64_t h,
-                                                bool sc1) {
-    const uint64_t* slots = sd.slots8;
-    uint32_t cap_mask = sd.cap_mask;
-    uint32_t tag = JTAG(h);
-    uint32_t slot = jslot_start(h, cap_mask);
-    for (uint32_t probes = 0; probes <= cap_mask; probes++) {
-        uint64_t packed = sc1 ? __hip_atomic_load((uint64_t*)&slots[slot], RLX)
-                              : slots[slot];
-        if (packed == 0) return UINT32_MAX;
-        if ((uint32_t)packed == tag &&
-            jrec_key_eq(sd, key_cols, KW, jhead_of(packed), kw, nullmask, sc1))
-            return slot;
-        slot = (slot + 1) & cap_mask;
-    }
-    return UINT32_MAX;
-}
-
-// find-or-insert + chain push of `row` (whose record is fully written
-// except `next`, which is linked here). plain_push: unique-key all-Insert
-// batches — claims still CAS (distinct keys can race one empty slot) but
-// the push of an existing chain is a plain RMW (sole writer per key).
-// drain_payload: the record was written with sc1 stores IN THIS LAUNCH and
-// same-launch readers exist — drain before the publishing CAS (R1).
-// Returns 0, or -1 on table full.
-__device__ __forceinline__ int jslot8_insert(const JoinSideDev& sd,
-                                             const uint8_t* key_cols, int KW,
-                                             const int64_t* kw,
-                                             uint32_t nullmask, uint64_t h,
-                                             uint32_t row, bool plain_push,
-                                             bool drain_payload) {
-    uint64_t* slots = sd.slots8;
-    uint32_t cap_mask = sd.cap_mask;
-    uint32_t tag = JTAG(h);
-    uint32_t slot = jslot_start(h, cap_mask);
-    JoinRowHdr* hd = jrow(sd, row);
-    for (uint32_t probes = 0; probes <= cap_mask; probes++) {
-        uint64_t packed = __hip_atomic_load(&slots[slot], RLX);
-        if (packed == 0) {
-            // publish order: payload (and next) at the coherence point
-            // BEFORE the head becomes visible, so a concurrent
-            // tag-collision verify that sees this head fetches real keys.
-            // The drain covers this thread's earlier payload stores too
-            // (cost ~nil, round-1 A/B).
-            if (drain_payload) st_u32(&hd->next, UINT32_MAX);
-            else hd->next = UINT32_MAX;
-            asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); // R1 drain
-            uint64_t prev = atomicCAS((unsigned long long*)&slots[slot], 0ull,
-                                      jpack(tag, row));
-            if (prev == 0) return 0;
-            packed = prev;
-        }
-        if ((uint32_t)packed == tag &&
-            jrec_key_eq(sd, key_cols, KW, jhead_of(packed), kw, nullmask,
-                        /*sc1=*/!plain_push)) {
-            if (plain_push) {
-                hd->next = jhead_of(packed);
-                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-                slots[slot] = jpack(tag, row);
-                return 0;
-            }
-            for (;;) {
-                if (drain_payload) st_u32(&hd->next, jhead_of(packed));
-                else hd->next = jhead_of(packed);
-                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-                uint64_t prev = atomicCAS((unsigned long long*)&slots[slot],
-                                          packed, jpack(tag, row));
-                if (prev == packed) return 0;
-                packed = prev; // only our key's chain pushes here
-            }
-        }
-        slot = (slot + 1) & cap_mask;
-    }
-    return -1;
 }
 
 // own-side find-or-insert. The probe walk uses PLAIN cached loads: a slot's
@@ -3248,14 +3201,11 @@ __device__ __forceinline__ void jown_insert(JoinSideDev own, const JoinMeta& m,
     JoinRowHdr* h = jrow(own, row);
     uint32_t vb = 0;
     long long* hv = jvals(h);
-    int rc;
-    if (b.all_insert && b.unique_keys) {
-        // plain cached stores are only safe with UNIQUE keys: a duplicate
-        // key's second inserter must verify the first one's record, and
-        // plain stores reach only the writer's L2 (not the coherence
-        // point), so a cross-XCD verify would read stale bytes. Unique
-        // batches have no same-key verify; tag-collision verifies use sc1
-        // loads (see jslot8_insert).
+    if (b.all_insert) {
+        // plain cached stores: inserts never read records (chained
+        // buckets), and all-Insert launches have no same-side walkers;
+        // jbucket_insert drains to the coherence point before the CAS
+        // publish so later foreign-key walks fetch real bytes.
         for (int c = 0; c < m.n_cols[S]; c++) {
             hv[c] = b.col_vals[c][r];
             vb |= (uint32_t)(b.col_valid[c][r] != 0) << c;
@@ -3263,8 +3213,7 @@ __device__ __forceinline__ void jown_insert(JoinSideDev own, const JoinMeta& m,
         h->validbits = vb;
         h->degree = init_deg;
         h->alive = 1;
-        rc = jslot8_insert(own, m.key_cols[S], m.KW, kw, nullmask, h64, row,
-                           /*plain_push=*/true, /*drain_payload=*/false);
+        jbucket_insert(own, h64, row, /*sc1_next=*/false);
     } else {
         for (int c = 0; c < m.n_cols[S]; c++) {
             st_i64((int64_t*)&hv[c], b.col_vals[c][r]);
@@ -3273,22 +3222,17 @@ __device__ __forceinline__ void jown_insert(JoinSideDev own, const JoinMeta& m,
         st_u32(&h->validbits, vb);
         st_u32(&h->degree, init_deg);
         st_u32(&h->alive, 1);
-        rc = jslot8_insert(own, m.key_cols[S], m.KW, kw, nullmask, h64, row,
-                           false, /*drain_payload=*/true);
+        jbucket_insert(own, h64, row, /*sc1_next=*/true);
     }
-    if (rc != 0) atomicExch(&out.counters[1], 2u); // key table full
 }
 
 __device__ __forceinline__ void jown_delete(JoinSideDev own, const JoinMeta& m,
                                             int S, const JoinBatchDev& b,
                                             uint32_t r, const int64_t* kw,
                                             uint32_t nullmask) {
-    // delete own row: FULL-row compare + CAS claim
-    uint32_t own_slot = jslot8_find(own, m.key_cols[S], m.KW, kw, nullmask,
-                                    hash_key(kw, nullmask, m.KW), true);
-    if (own_slot == UINT32_MAX) return;
-    uint32_t row =
-        jhead_of(__hip_atomic_load(&own.slots8[own_slot], RLX));
+    // delete own row: FULL-row compare + CAS claim; the
+    // full-row compare subsumes the key filter (key cols are row cols)
+    uint32_t row = jbucket_head(own, hash_key(kw, nullmask, m.KW), true);
     while (row != UINT32_MAX) {
         JoinRowHdr* h = jrow(own, row);
         if (ld_u32(&h->alive)) {
@@ -3339,15 +3283,16 @@ __device__ void join_probe_row_noninner(const JoinBatchDev& b, JoinSideDev own,
             jemit_row(out, m, S, op, b, r, 0, nullptr, 1);
         return; // no state write
     }
-    uint32_t mslot = jslot8_find(match, m.key_cols[1 - S], m.KW, kw,
-                                 nullmask, hash_key(kw, nullmask, m.KW),
-                                 false);
+    uint32_t mhead =
+        jbucket_head(match, hash_key(kw, nullmask, m.KW), false);
     uint32_t my_deg = 0;
-    if (mslot != UINT32_MAX) {
-        uint32_t row = jhead_of(match.slots8[mslot]);
+    {
+        uint32_t row = mhead;
         while (row != UINT32_MAX) {
             JoinRowHdr* h = jrow(match, row);
-            if (h->alive && join_cond_ok(m, S, b, r, h->validbits, jvals(h))) {
+            if (h->alive &&
+                jhdr_key_eq(h, m.key_cols[1 - S], m.KW, kw, nullmask, false) &&
+                join_cond_ok(m, S, b, r, h->validbits, jvals(h))) {
                 my_deg++;
                 bool zero = false;
                 if (m.need_deg[1 - S]) {
@@ -3449,18 +3394,19 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
         }
         if (never_match) active = false;
 
-        uint32_t mslot = UINT32_MAX;
+        uint32_t mhead = UINT32_MAX;
         uint32_t my_n = 0;
         uint32_t matched_row = UINT32_MAX;
         if (active) {
-            mslot = jslot8_find(match, m.key_cols[1 - S], m.KW, kw,
-                                nullmask, hash_key(kw, nullmask, m.KW),
-                                m.append_only != 0);
-            if (mslot != UINT32_MAX && !m.append_only) {
-                uint32_t row = jhead_of(match.slots8[mslot]);
+            mhead = jbucket_head(match, hash_key(kw, nullmask, m.KW),
+                                 m.append_only != 0);
+            if (mhead != UINT32_MAX && !m.append_only) {
+                uint32_t row = mhead;
                 while (row != UINT32_MAX) {
                     JoinRowHdr* h = jrow(match, row);
                     if (h->alive &&
+                        jhdr_key_eq(h, m.key_cols[1 - S], m.KW, kw, nullmask,
+                                    false) &&
                         join_cond_ok(m, S, b, r, h->validbits, jvals(h))) {
                         my_n++;
                         matched_row = row;
@@ -3487,11 +3433,13 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
                 if (lane == 0) atomicExch(&out.counters[1], 1u); // overflow
             } else if (my_n && !(dbg_skip & 1)) {
                 // second walk: emit (JoinStreamChunkBuilder::append_row)
-                uint32_t row = jhead_of(match.slots8[mslot]);
+                uint32_t row = mhead;
                 uint32_t k = 0;
                 while (row != UINT32_MAX && k < my_n) {
                     JoinRowHdr* h = jrow(match, row);
                     if (h->alive &&
+                        jhdr_key_eq(h, m.key_cols[1 - S], m.KW, kw, nullmask,
+                                    false) &&
                         join_cond_ok(m, S, b, r, h->validbits, jvals(h))) {
                         uint32_t orow = my_base + k;
                         out.ops[orow] = op;
@@ -3516,14 +3464,16 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
                     row = h->next;
                 }
             }
-        } else if (active && mslot != UINT32_MAX) {
+        } else if (active && mhead != UINT32_MAX) {
             // append-only path: <=1 match; sc1 reads, per-match atomics
-            uint32_t row =
-                jhead_of(__hip_atomic_load(&match.slots8[mslot], RLX));
+            uint32_t row = mhead;
             while (row != UINT32_MAX) {
                 JoinRowHdr* h = jrow(match, row);
                 uint32_t vb = ld_u32(&h->validbits);
-                if (ld_u32(&h->alive) && join_cond_ok(m, S, b, r, vb, jvals(h))) {
+                if (ld_u32(&h->alive) &&
+                    jhdr_key_eq(h, m.key_cols[1 - S], m.KW, kw, nullmask,
+                                true) &&
+                    join_cond_ok(m, S, b, r, vb, jvals(h))) {
                     uint32_t orow = atomicAdd(&out.counters[0], 1u);
                     if (orow >= out.cap) {
                         atomicExch(&out.counters[1], 1u);
@@ -3802,20 +3752,18 @@ __global__ __launch_bounds__(256, 8) void jpart_probe_insert_kernel(
         }
         // probe the match side (immutable during this launch: the batch
         // mutates only `own`)
-        uint32_t mslot = UINT32_MAX;
+        uint32_t mhead = UINT32_MAX;
         uint32_t my_n = 0;
         if (active) {
-            mslot = jslot8_find(match, m.key_cols[1 - S], m.KW, kw, nm,
-                                hash_key(kw, nm, m.KW), false);
-            if (mslot != UINT32_MAX) {
-                uint32_t mr = jhead_of(match.slots8[mslot]);
-                while (mr != UINT32_MAX) {
-                    JoinRowHdr* mh = jrow(match, mr);
-                    if (mh->alive &&
-                        jpart_cond_ok(m, S, hv, pvb, mh->validbits, jvals(mh)))
-                        my_n++;
-                    mr = mh->next;
-                }
+            mhead = jbucket_head(match, hash_key(kw, nm, m.KW), false);
+            uint32_t mr = mhead;
+            while (mr != UINT32_MAX) {
+                JoinRowHdr* mh = jrow(match, mr);
+                if (mh->alive &&
+                    jhdr_key_eq(mh, m.key_cols[1 - S], m.KW, kw, nm, false) &&
+                    jpart_cond_ok(m, S, hv, pvb, mh->validbits, jvals(mh)))
+                    my_n++;
+                mr = mh->next;
             }
         }
         // wave-aggregated output reservation + emit (multiset parity; the
@@ -3833,11 +3781,12 @@ __global__ __launch_bounds__(256, 8) void jpart_probe_insert_kernel(
         if (total && base + total > out.cap) {
             if (lane == 0) atomicExch(&out.counters[1], 1u);
         } else if (my_n) {
-            uint32_t mr = jhead_of(match.slots8[mslot]);
+            uint32_t mr = mhead;
             uint32_t k = 0;
             while (mr != UINT32_MAX && k < my_n) {
                 JoinRowHdr* mh = jrow(match, mr);
                 if (mh->alive &&
+                    jhdr_key_eq(mh, m.key_cols[1 - S], m.KW, kw, nm, false) &&
                     jpart_cond_ok(m, S, hv, pvb, mh->validbits, jvals(mh))) {
                     uint32_t orow = my_base + k;
                     out.ops[orow] = RW_OP_INSERT;
@@ -3867,11 +3816,8 @@ __global__ __launch_bounds__(256, 8) void jpart_probe_insert_kernel(
         // exactly one linking thread — plain read-modify-write.
         if (active) {
             // record already in place from the scatter kernel (globally
-            // visible across the dispatch boundary); link the chain. Keys
-            // are pairwise distinct in the batch -> plain push.
-            if (jslot8_insert(own, m.key_cols[S], m.KW, kw, nm,
-                              hash_key(kw, nm, m.KW), row, true, false) != 0)
-                atomicExch(&out.counters[1], 2u);
+            // visible across the dispatch boundary); one CAS links it
+            jbucket_insert(own, hash_key(kw, nm, m.KW), row, false);
         }
     }
 }
@@ -3882,24 +3828,24 @@ __global__ __launch_bounds__(256, 8) void jpart_probe_insert_kernel(
 // memory reclamation is compaction work for a later round)
 __global__ void join_clean_kernel(JoinSideDev sd, int kcol, long long wm,
                                   int KW) {
-    // kcol = RECORD column of the watermarked key position (8-B slots keep
-    // no key words; the chain head identifies the key). The chain keeps its
-    // slot entry — probes skip dead records by the alive flag.
+    // kcol = RECORD column of the watermarked key position. Buckets mix
+    // keys, so the watermark predicate runs PER RECORD; rows below the
+    // watermark are retired in place (slot entries stay — probes skip dead
+    // records by the alive flag).
     (void)KW;
     size_t cap = (size_t)sd.cap_mask + 1;
     size_t stride = (size_t)gridDim.x * blockDim.x;
     for (size_t slot = blockIdx.x * blockDim.x + threadIdx.x; slot < cap;
          slot += stride) {
         uint64_t packed = __hip_atomic_load(&sd.slots8[slot], RLX);
-        if (packed == 0) continue;
-        uint32_t head = jhead_of(packed);
-        JoinRowHdr* hh = jrow(sd, head);
-        if (!((ld_u32(&hh->validbits) >> kcol) & 1)) continue; // NULLs largest
-        if (ld_i64((const int64_t*)&jvals(hh)[kcol]) >= wm) continue;
-        uint32_t row = head;
+        if ((uint32_t)packed == 0) continue; // bloom 0 = empty
+        uint32_t row = jhead_of(packed);
         while (row != UINT32_MAX) {
             JoinRowHdr* h = jrow(sd, row);
-            st_u32(&h->alive, 0);
+            if (ld_u32(&h->alive) &&
+                ((ld_u32(&h->validbits) >> kcol) & 1) && // NULLs largest
+                ld_i64((const int64_t*)&jvals(h)[kcol]) < wm)
+                st_u32(&h->alive, 0);
             row = ld_u32(&h->next);
         }
     }
@@ -4223,23 +4169,25 @@ __global__ void join_vnode_scope_kernel(JoinSideDev sd, int KW,
     for (size_t slot = blockIdx.x * blockDim.x + threadIdx.x; slot < cap;
          slot += stride) {
         uint64_t packed = __hip_atomic_load(&sd.slots8[slot], RLX);
-        if (packed == 0) continue;
-        JoinRowHdr* hh = jrow(sd, jhead_of(packed));
-        uint32_t vb = ld_u32(&hh->validbits);
-        const long long* hv = jvals(hh);
-        int64_t kw[4];
-        uint32_t nulls = 0;
-        for (int k = 0; k < KW; k++) {
-            bool valid = (vb >> kcols[k]) & 1;
-            kw[k] = valid ? ld_i64((const int64_t*)&hv[kcols[k]]) : 0;
-            nulls |= (uint32_t)(!valid) << k;
-        }
-        uint32_t vn = crc_key_words(lut, kw, nulls, KW, types) % vnode_count;
-        if ((bitmap[vn >> 3] >> (vn & 7)) & 1) continue;
+        if ((uint32_t)packed == 0) continue; // bloom 0 = empty
         uint32_t row = jhead_of(packed);
-        while (row != UINT32_MAX) {
+        while (row != UINT32_MAX) { // buckets mix keys: vnode PER RECORD
             JoinRowHdr* h = jrow(sd, row);
-            st_u32(&h->alive, 0);
+            if (ld_u32(&h->alive)) {
+                uint32_t vb = ld_u32(&h->validbits);
+                const long long* hv = jvals(h);
+                int64_t kw[4];
+                uint32_t nulls = 0;
+                for (int k = 0; k < KW; k++) {
+                    bool valid = (vb >> kcols[k]) & 1;
+                    kw[k] = valid ? ld_i64((const int64_t*)&hv[kcols[k]]) : 0;
+                    nulls |= (uint32_t)(!valid) << k;
+                }
+                uint32_t vn =
+                    crc_key_words(lut, kw, nulls, KW, types) % vnode_count;
+                if (!((bitmap[vn >> 3] >> (vn & 7)) & 1))
+                    st_u32(&h->alive, 0);
+            }
             row = ld_u32(&h->next);
         }
     }
@@ -4445,26 +4393,9 @@ struct HashJoin {
             if (!(c->vis && !c->vis[r]) && c->ops[r] != RW_OP_INSERT &&
                 c->ops[r] != RW_OP_UPDATE_INSERT)
                 b.all_insert = 0;
+        // chained buckets need no uniqueness pre-pass (round-1's plain
+        // head-push lever is gone: inserts are single-CAS regardless)
         b.unique_keys = 0;
-        if (b.all_insert) {
-            std::unordered_set<std::string> seen;
-            seen.reserve(n * 2);
-            bool uniq = true;
-            for (uint32_t r = 0; uniq && r < n; r++) {
-                if (c->vis && !c->vis[r]) continue;
-                std::string k;
-                for (int i = 0; i < m.KW; i++) {
-                    uint8_t ci = m.key_cols[s][i];
-                    uint8_t valid = c->cols[ci].valid[r];
-                    k.push_back((char)valid);
-                    int64_t v =
-                        valid ? ((const int64_t*)c->cols[ci].data)[r] : 0;
-                    k.append((const char*)&v, 8);
-                }
-                uniq = seen.insert(std::move(k)).second;
-            }
-            b.unique_keys = uniq;
-        }
         *bout = b;
         return RW_OK;
     }
@@ -4475,10 +4406,11 @@ struct HashJoin {
     uint32_t* d_part_base = nullptr;
     uint32_t* d_row_base = nullptr;
 
-    // The partitioned pipeline handles exactly the shape under which the
-    // batch-parallel insert commutes with the reference's sequential loop
-    // without CAS chain publishes: inner join, every visible row an Insert,
-    // keys pairwise distinct (host-verified at upload), whole-batch range.
+    // The partitioned pipeline handles the shape under which the
+    // batch-parallel insert commutes with the reference's sequential loop:
+    // inner join, every visible row an Insert (same-side rows never match
+    // each other, so equal keys within the batch are independent),
+    // whole-batch range.
     bool can_partition(const JoinBatchDev& b, uint32_t r0, uint32_t r1) {
         static int en = [] {
             const char* e = getenv("RW_JOIN_PART"); // A/B: 0 = old path
@@ -4486,7 +4418,7 @@ struct HashJoin {
         }();
         if (!en) return false;
         if (m.join_type != RW_JOIN_INNER || m.append_only) return false;
-        if (!b.all_insert || !b.unique_keys || b.vis) return false;
+        if (!b.all_insert || b.vis) return false;
         if (r0 != 0 || r1 != b.n_rows) return false;
         if (b.n_rows < JPART_MIN_ROWS) return false;
         // partitions must map to multi-slot windows on both sides
@@ -5057,23 +4989,6 @@ void* rw_join_bench_preload(void* h, int side, const RwChunk* c) {
         if (c->ops[r] != RW_OP_INSERT && c->ops[r] != RW_OP_UPDATE_INSERT)
             b->all_insert = 0;
     b->unique_keys = 0;
-    if (b->all_insert) {
-        std::unordered_set<std::string> seen;
-        seen.reserve(n * 2);
-        bool uniq = true;
-        for (uint32_t r = 0; uniq && r < n; r++) {
-            std::string k;
-            for (int i = 0; i < j->m.KW; i++) {
-                uint8_t ci = j->m.key_cols[side][i];
-                uint8_t valid = c->cols[ci].valid[r];
-                k.push_back((char)valid);
-                int64_t v = valid ? ((const int64_t*)c->cols[ci].data)[r] : 0;
-                k.append((const char*)&v, 8);
-            }
-            uniq = seen.insert(std::move(k)).second;
-        }
-        b->unique_keys = uniq;
-    }
     return b;
 }
 
