@@ -3551,7 +3551,7 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
 //                           link own-side chains (plain RMW — unique keys
 //                           give every touched slot exactly one writer)
 
-#define JPART_LOG2 11
+#define JPART_LOG2 12
 #define JPART_P (1u << JPART_LOG2)
 #define JPART_NBLK 256
 #define JPART_MIN_ROWS 131072u
@@ -3820,6 +3820,134 @@ __global__ __launch_bounds__(256, 8) void jpart_probe_insert_kernel(
             jbucket_insert(own, hash_key(kw, nm, m.KW), row, false);
         }
     }
+}
+
+// LDS-window variant of the fused probe+insert phase: with chained
+// buckets and top-bit slot indexing, partition p's OWN slot window
+// [p*cap/P, (p+1)*cap/P) is touched by NO other block during the launch —
+// so the block stages the whole window in LDS, performs the bucket pushes
+// as LDS CASes, and writes the window back sequentially. This converts the
+// insert path's random global CAS traffic (the dominant q8 cost: 290 us of
+// the 540 us monolith step by RW_JOIN_SKIP A/B) into two sequential window
+// copies + LDS atomics. Dynamic LDS = (cap/P)*8 bytes.
+__global__ __launch_bounds__(256) void jpart_probe_insert_lds_kernel(
+    JoinSideDev own, JoinSideDev match, JoinMeta m, int S, JoinOutDev out,
+    const uint32_t* part_base, const uint32_t* row_base,
+    const uint32_t* err) {
+    if (*err) return;
+    extern __shared__ uint64_t win[];
+    uint32_t rb = *row_base;
+    uint32_t lo = part_base[blockIdx.x];
+    uint32_t hi = part_base[blockIdx.x + 1]; // [JPART_P] holds the total
+    uint32_t n = hi - lo;
+    if (n == 0) return; // window untouched
+    uint32_t wslots = (own.cap_mask + 1) >> JPART_LOG2;
+    uint32_t wbase = blockIdx.x * wslots;
+    for (uint32_t i = threadIdx.x; i < wslots; i += blockDim.x)
+        win[i] = own.slots8[wbase + i];
+    __syncthreads();
+    int lane = threadIdx.x & 63;
+    uint32_t iters = (n + blockDim.x - 1) / blockDim.x;
+    for (uint32_t it = 0; it < iters; it++) {
+        uint32_t i = it * blockDim.x + threadIdx.x;
+        bool active = i < n;
+        uint32_t row = rb + lo + i;
+        int64_t kw[MAX_KW];
+        uint32_t nm = 0;
+        JoinRowHdr* hd = nullptr;
+        long long* hv = nullptr;
+        uint32_t pvb = 0;
+        uint64_t h64 = 0;
+        if (active) {
+            hd = jrow(own, row);
+            hv = jvals(hd);
+            pvb = hd->validbits;
+            for (int k = 0; k < m.KW; k++) {
+                uint8_t col = m.key_cols[S][k];
+                bool valid = (pvb >> col) & 1;
+                kw[k] = valid ? hv[col] : 0;
+                nm |= (uint32_t)(!valid) << k;
+            }
+            h64 = hash_key(kw, nm, m.KW);
+        }
+        // probe the match side (immutable during this launch)
+        uint32_t mhead = UINT32_MAX;
+        uint32_t my_n = 0;
+        if (active) {
+            mhead = jbucket_head(match, h64, false);
+            uint32_t mr = mhead;
+            while (mr != UINT32_MAX) {
+                JoinRowHdr* mh = jrow(match, mr);
+                if (mh->alive &&
+                    jhdr_key_eq(mh, m.key_cols[1 - S], m.KW, kw, nm, false) &&
+                    jpart_cond_ok(m, S, hv, pvb, mh->validbits, jvals(mh)))
+                    my_n++;
+                mr = mh->next;
+            }
+        }
+        uint32_t incl = my_n;
+        for (int d = 1; d < 64; d <<= 1) {
+            uint32_t o = __shfl_up(incl, d);
+            if (lane >= d) incl += o;
+        }
+        uint32_t total = (uint32_t)__shfl((int)incl, 63);
+        uint32_t base = 0;
+        if (lane == 0 && total) base = atomicAdd(&out.counters[0], total);
+        base = (uint32_t)__shfl((int)base, 0);
+        uint32_t my_base = base + incl - my_n;
+        if (total && base + total > out.cap) {
+            if (lane == 0) atomicExch(&out.counters[1], 1u);
+        } else if (my_n) {
+            uint32_t mr = mhead;
+            uint32_t k = 0;
+            while (mr != UINT32_MAX && k < my_n) {
+                JoinRowHdr* mh = jrow(match, mr);
+                if (mh->alive &&
+                    jhdr_key_eq(mh, m.key_cols[1 - S], m.KW, kw, nm, false) &&
+                    jpart_cond_ok(m, S, hv, pvb, mh->validbits, jvals(mh))) {
+                    uint32_t orow = my_base + k;
+                    out.ops[orow] = RW_OP_INSERT;
+                    const long long* mv = jvals(mh);
+                    for (int c = 0; c < m.n_out; c++) {
+                        bool from_probe = (int)m.out_src[c] == S;
+                        uint8_t col = m.out_col[c];
+                        int64_t v;
+                        uint8_t valid;
+                        if (from_probe) {
+                            valid = (pvb >> col) & 1;
+                            v = hv[col];
+                        } else {
+                            valid = (mh->validbits >> col) & 1;
+                            v = mv[col];
+                        }
+                        out.vals[(size_t)orow * m.n_out + c] = valid ? v : 0;
+                        out.nulls[(size_t)orow * m.n_out + c] = !valid;
+                    }
+                    k++;
+                }
+                mr = mh->next;
+            }
+        }
+        // own-side bucket push in LDS (visibility via the window writeback
+        // + the inter-dispatch flush; no drains needed — nothing walks the
+        // own side within an all-Insert launch)
+        if (active) {
+            uint32_t local = jslot_start(h64, own.cap_mask) - wbase;
+            uint64_t old = win[local];
+            for (;;) {
+                hd->next = (uint32_t)old ? jhead_of(old) : UINT32_MAX;
+                uint64_t want =
+                    ((uint64_t)row << 32) | ((uint32_t)old | jbloom_bit(h64));
+                uint64_t prev = atomicCAS((unsigned long long*)&win[local],
+                                          old, want);
+                if (prev == old) break;
+                old = prev;
+            }
+        }
+    }
+    __syncthreads();
+    for (uint32_t i = threadIdx.x; i < wslots; i += blockDim.x)
+        own.slots8[wbase + i] = win[i];
 }
 
 // watermark TTL sweeps (state_table watermark cleaning):
@@ -4466,9 +4594,17 @@ struct HashJoin {
                 side[s].row_cap, d_row_base, out.counters + 1);
             jpart_scatter_kernel<<<2048, 256, 0, stream>>>(
                 b, m, s, side[s], d_pcur, d_row_base, out.counters + 1);
-            jpart_probe_insert_kernel<<<JPART_P, 256, 0, stream>>>(
-                side[s], side[1 - s], m, s, out, d_part_base, d_row_base,
-                out.counters + 1);
+            uint32_t wslots = (side[s].cap_mask + 1) >> JPART_LOG2;
+            if ((size_t)wslots * 8 <= 64 * 1024) {
+                jpart_probe_insert_lds_kernel<<<JPART_P, 256, wslots * 8,
+                                                stream>>>(
+                    side[s], side[1 - s], m, s, out, d_part_base, d_row_base,
+                    out.counters + 1);
+            } else {
+                jpart_probe_insert_kernel<<<JPART_P, 256, 0, stream>>>(
+                    side[s], side[1 - s], m, s, out, d_part_base, d_row_base,
+                    out.counters + 1);
+            }
         } else {
             join_probe_kernel<<<blocks, 256, 0, stream>>>(b, side[s],
                                                           side[1 - s], m, s,
