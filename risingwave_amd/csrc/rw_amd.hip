@@ -3835,6 +3835,10 @@ __global__ __launch_bounds__(256) void jpart_probe_insert_lds_kernel(
     const uint32_t* part_base, const uint32_t* row_base,
     const uint32_t* err) {
     if (*err) return;
+    // dynamic LDS: [0, wslots) = the block's own slot window;
+    // [wslots, ...) = a per-iteration staging area for blockDim.x records
+    // (a lane-per-record read of 64-B records issues 64 line transactions
+    // per wave per field — staging streams them COALESCED instead)
     extern __shared__ uint64_t win[];
     uint32_t rb = *row_base;
     uint32_t lo = part_base[blockIdx.x];
@@ -3843,25 +3847,41 @@ __global__ __launch_bounds__(256) void jpart_probe_insert_lds_kernel(
     if (n == 0) return; // window untouched
     uint32_t wslots = (own.cap_mask + 1) >> JPART_LOG2;
     uint32_t wbase = blockIdx.x * wslots;
+    uint32_t rwords = own.row_stride >> 3;
+    uint64_t* stage = win + wslots;
     for (uint32_t i = threadIdx.x; i < wslots; i += blockDim.x)
         win[i] = own.slots8[wbase + i];
-    __syncthreads();
     int lane = threadIdx.x & 63;
     uint32_t iters = (n + blockDim.x - 1) / blockDim.x;
     for (uint32_t it = 0; it < iters; it++) {
         uint32_t i = it * blockDim.x + threadIdx.x;
         bool active = i < n;
         uint32_t row = rb + lo + i;
+        // cooperative coalesced staging of this iteration's records
+        {
+            uint32_t first = lo + it * blockDim.x;
+            uint32_t count = n - it * blockDim.x;
+            if (count > blockDim.x) count = blockDim.x;
+            const uint64_t* src =
+                (const uint64_t*)(own.rows +
+                                  (size_t)(rb + first) * own.row_stride);
+            uint32_t words = count * rwords;
+            __syncthreads();
+            for (uint32_t w = threadIdx.x; w < words; w += blockDim.x)
+                stage[w] = src[w];
+            __syncthreads();
+        }
         int64_t kw[MAX_KW];
         uint32_t nm = 0;
-        JoinRowHdr* hd = nullptr;
-        long long* hv = nullptr;
+        JoinRowHdr* hd = nullptr;   // GLOBAL record (next-link target)
+        long long* hv = nullptr;    // LDS copy (reads)
         uint32_t pvb = 0;
         uint64_t h64 = 0;
         if (active) {
             hd = jrow(own, row);
-            hv = jvals(hd);
-            pvb = hd->validbits;
+            JoinRowHdr* lh = (JoinRowHdr*)(stage + (size_t)threadIdx.x * rwords);
+            hv = (long long*)((uint8_t*)lh + 16);
+            pvb = lh->validbits;
             for (int k = 0; k < m.KW; k++) {
                 uint8_t col = m.key_cols[S][k];
                 bool valid = (pvb >> col) & 1;
@@ -4595,8 +4615,9 @@ struct HashJoin {
             jpart_scatter_kernel<<<2048, 256, 0, stream>>>(
                 b, m, s, side[s], d_pcur, d_row_base, out.counters + 1);
             uint32_t wslots = (side[s].cap_mask + 1) >> JPART_LOG2;
-            if ((size_t)wslots * 8 <= 64 * 1024) {
-                jpart_probe_insert_lds_kernel<<<JPART_P, 256, wslots * 8,
+            size_t shmem = (size_t)wslots * 8 + 256 * side[s].row_stride;
+            if (shmem <= 64 * 1024) {
+                jpart_probe_insert_lds_kernel<<<JPART_P, 256, shmem,
                                                 stream>>>(
                     side[s], side[1 - s], m, s, out, d_part_base, d_row_base,
                     out.counters + 1);
