@@ -3049,8 +3049,11 @@ struct JoinBatchDev {
 };
 
 struct JoinOutDev {
-    int64_t* vals;   // [cap * n_out]
-    uint8_t* nulls;  // [cap * n_out]
+    // COLUMNAR output block: column c occupies vals[c*cap ..), so emit
+    // stores are lane-coalesced and the q3 agg hop reads it as plain
+    // columnar input (stride 1)
+    int64_t* vals;   // [n_out][cap]
+    uint8_t* nulls;  // [n_out][cap]
     uint8_t* ops;    // [cap]
     uint32_t* counters; // [0]=cursor [1]=overflow
     uint32_t cap;
@@ -3171,8 +3174,8 @@ __device__ __forceinline__ void jemit_row(JoinOutDev& out, const JoinMeta& m,
             valid = (match_vb >> col) & 1;
             v = mv[col];
         }
-        out.vals[(size_t)orow * m.n_out + c] = valid ? v : 0;
-        out.nulls[(size_t)orow * m.n_out + c] = !valid;
+        out.vals[(size_t)c * out.cap + orow] = valid ? v : 0;
+        out.nulls[(size_t)c * out.cap + orow] = !valid;
     }
 }
 
@@ -3456,8 +3459,8 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
                                 valid = (h->validbits >> col) & 1;
                                 v = mv[col];
                             }
-                            out.vals[(size_t)orow * m.n_out + c] = valid ? v : 0;
-                            out.nulls[(size_t)orow * m.n_out + c] = !valid;
+                            out.vals[(size_t)c * out.cap + orow] = valid ? v : 0;
+                            out.nulls[(size_t)c * out.cap + orow] = !valid;
                         }
                         k++;
                     }
@@ -3492,8 +3495,8 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
                                 valid = (vb >> col) & 1;
                                 v = (int64_t)ld_i64((const int64_t*)&mv[col]);
                             }
-                            out.vals[(size_t)orow * m.n_out + c] = valid ? v : 0;
-                            out.nulls[(size_t)orow * m.n_out + c] = !valid;
+                            out.vals[(size_t)c * out.cap + orow] = valid ? v : 0;
+                            out.nulls[(size_t)c * out.cap + orow] = !valid;
                         }
                     }
                     matched_row = row;
@@ -3803,8 +3806,8 @@ __global__ __launch_bounds__(256, 8) void jpart_probe_insert_kernel(
                             valid = (mh->validbits >> col) & 1;
                             v = mv[col];
                         }
-                        out.vals[(size_t)orow * m.n_out + c] = valid ? v : 0;
-                        out.nulls[(size_t)orow * m.n_out + c] = !valid;
+                        out.vals[(size_t)c * out.cap + orow] = valid ? v : 0;
+                        out.nulls[(size_t)c * out.cap + orow] = !valid;
                     }
                     k++;
                 }
@@ -3940,8 +3943,8 @@ __global__ __launch_bounds__(256) void jpart_probe_insert_lds_kernel(
                             valid = (mh->validbits >> col) & 1;
                             v = mv[col];
                         }
-                        out.vals[(size_t)orow * m.n_out + c] = valid ? v : 0;
-                        out.nulls[(size_t)orow * m.n_out + c] = !valid;
+                        out.vals[(size_t)c * out.cap + orow] = valid ? v : 0;
+                        out.nulls[(size_t)c * out.cap + orow] = !valid;
                     }
                     k++;
                 }
@@ -4808,13 +4811,18 @@ struct HashJoin {
         if (ctr[1] == 3) FAIL(RW_E_INTERNAL, "join row store full");
         uint32_t n_out = ctr[0];
         if (n_out) {
-            std::vector<int64_t> vals((size_t)n_out * m.n_out);
+            // columnar device block: copy each column's first n_out entries
+            std::vector<int64_t> vals((size_t)n_out * m.n_out); // [c][n]
             std::vector<uint8_t> nulls((size_t)n_out * m.n_out);
             std::vector<uint8_t> ops(n_out);
-            HIP_TRY(hipMemcpy(vals.data(), out.vals, vals.size() * 8,
-                              hipMemcpyDeviceToHost));
-            HIP_TRY(hipMemcpy(nulls.data(), out.nulls, nulls.size(),
-                              hipMemcpyDeviceToHost));
+            for (int ci = 0; ci < m.n_out; ci++) {
+                HIP_TRY(hipMemcpy(vals.data() + (size_t)ci * n_out,
+                                  out.vals + (size_t)ci * out.cap,
+                                  (size_t)n_out * 8, hipMemcpyDeviceToHost));
+                HIP_TRY(hipMemcpy(nulls.data() + (size_t)ci * n_out,
+                                  out.nulls + (size_t)ci * out.cap, n_out,
+                                  hipMemcpyDeviceToHost));
+            }
             HIP_TRY(hipMemcpy(ops.data(), out.ops, n_out, hipMemcpyDeviceToHost));
             uint32_t max_rows = desc.chunk_size ? desc.chunk_size : 1024;
             if (max_rows < 2) max_rows = 2;
@@ -4829,8 +4837,8 @@ struct HashJoin {
                     auto* data = new int64_t[take];
                     auto* valid = new uint8_t[take];
                     for (uint32_t r = 0; r < take; r++) {
-                        data[r] = vals[(size_t)(start + r) * m.n_out + ci];
-                        valid[r] = !nulls[(size_t)(start + r) * m.n_out + ci];
+                        data[r] = vals[(size_t)ci * n_out + start + r];
+                        valid[r] = !nulls[(size_t)ci * n_out + start + r];
                     }
                     cols[ci].type = out_types[ci];
                     cols[ci].valid = valid;
@@ -5214,11 +5222,11 @@ int rw_agg_apply_joinout(void* agg_h, void* join_h) {
         HIP_TRY(hipMemset(j->zeros, 0, (size_t)j->out.cap * j->m.n_out));
     }
     AggBatch b{};
-    b.stride = (uint32_t)j->m.n_out;
+    b.stride = 1; // columnar output block: plain columnar agg input
     b.valid_inverted = 1;
     auto bind = [&](int slot, int src_col) {
-        b.col_vals[slot] = j->out.vals + src_col;
-        b.col_valid[slot] = j->out.nulls + src_col;
+        b.col_vals[slot] = j->out.vals + (size_t)src_col * j->out.cap;
+        b.col_valid[slot] = j->out.nulls + (size_t)src_col * j->out.cap;
     };
     for (int i = 0; i < agg->KW; i++) {
         if ((int)agg->group_key[i] >= j->m.n_out)
