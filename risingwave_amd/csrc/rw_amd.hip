@@ -4563,11 +4563,13 @@ struct HashJoin {
     // each other, so equal keys within the batch are independent),
     // whole-batch range.
     bool can_partition(const JoinBatchDev& b, uint32_t r0, uint32_t r1) {
-        static int en = [] {
-            const char* e = getenv("RW_JOIN_PART"); // A/B: 0 = old path
-            return e ? atoi(e) : 1;
-        }();
-        if (!en) return false;
+        // Default OFF: with LIC-resident bloom-bucket slots the fused
+        // monolithic kernel beat the 4-phase pipeline on q8 (0.50 vs
+        // 0.61 ms/1M rows — the prelude's ~125 us exceeds the probe
+        // phase's locality win; profiles/r02 series). Kept selectable
+        // (RW_JOIN_PART=1) and parity-covered.
+        const char* e = getenv("RW_JOIN_PART");
+        if (!e || atoi(e) == 0) return false;
         if (m.join_type != RW_JOIN_INNER || m.append_only) return false;
         if (!b.all_insert || b.vis) return false;
         if (r0 != 0 || r1 != b.n_rows) return false;

@@ -1126,6 +1126,8 @@ def test_join_partitioned_pipeline_parity():
     # partitioned probes walk — and compare output multisets AND the
     # checkpoint spill bytes (which read back the scattered row records)
     # against the oracle.
+    import os
+    os.environ["RW_JOIN_PART"] = "1"  # opt into the partitioned pipeline
     rng = np.random.default_rng(77)
     t4 = [T_I64, T_TS, T_TS, T_I64]
     kw = dict(key_l=[0, 1, 2], key_r=[0, 1, 2], pk_l=[3], pk_r=[3])
@@ -1177,3 +1179,4 @@ def test_join_partitioned_pipeline_parity():
         assert sg == so, (f"side {side}: spill {len(sg)} vs {len(so)} bytes")
     g.close()
     o.close()
+    del os.environ["RW_JOIN_PART"]
