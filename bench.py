@@ -158,7 +158,8 @@ def main():
     ap.add_argument("--windows-per-epoch", type=int, default=64)
     ap.add_argument("--seed", type=int, default=1)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
-    ap.add_argument("--workload", choices=["all", "q7", "q8", "q3"],
+    ap.add_argument("--workload",
+                    choices=["all", "q7", "q7pipe", "q8", "q3"],
                     default="all",
                     help="all (default) = the BASELINE metric 'q7+q8': one "
                          "run emits the q7 line then the q8 line. q7 = "
@@ -233,6 +234,10 @@ def main():
     try:
         if args.workload in ("all", "q7"):
             bench_q7(args, ffi, gpu_lib, rng, rank, world, dist)
+        if args.workload in ("all", "q7pipe"):
+            bench_q7pipe(args, ffi, gpu_lib,
+                         np.random.default_rng(args.seed + rank + 7),
+                         rank, world, dist)
         if args.workload in ("all", "q8"):
             bench_q8(args, ffi, gpu_lib, np.random.default_rng(args.seed + rank),
                      rank, world, dist)
@@ -560,6 +565,194 @@ def setup_exchange(ffi, rank, world, dist):
     if not h:
         return None
     return ExchangeCtx(lib, h, world)
+
+
+def bench_q7pipe(args, ffi, gpu_lib, rng, rank, world, dist):
+    """The FULL Nexmark q7 stream plan, device-resident (reference
+    stream_plan, nexmark.yaml q7 block): bid -> project(window,price) ->
+    [vnode hop] -> StreamHashAgg[append_only] max(price) by 10s window ->
+    flush change stream -> StreamHashJoin Inner on bid.price = max(price)
+    (right input = the agg's U-/U+ stream) with the StreamFilter
+    `date_time BETWEEN $expr1-10s AND $expr1` fused into inner emission,
+    plus the bid -> [vnode hop] -> join-left input. A step = one 1M-row
+    bid batch through BOTH branches; agg flush + join-right apply at every
+    --barrier-every steps. value = bid input rows/s."""
+    import ctypes
+
+    from rwtest.ffi import AGG_COUNT_STAR, AGG_MAX, CMP_GE, CMP_LE, \
+        SIDE_LEFT, SIDE_RIGHT, JOIN_INNER, T_I64
+
+    L = gpu_lib.lib
+    L.rw_agg_bench_preload.restype = ctypes.c_void_p
+    L.rw_agg_bench_preload.argtypes = [ctypes.c_void_p, ctypes.POINTER(ffi.RwChunkC)]
+    L.rw_join_bench_preload.restype = ctypes.c_void_p
+    L.rw_join_bench_preload.argtypes = [ctypes.c_void_p, ctypes.c_int,
+                                        ctypes.POINTER(ffi.RwChunkC)]
+    L.rw_join_bench_apply.restype = ctypes.c_int
+    L.rw_join_bench_apply.argtypes = [ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p]
+    L.rw_join_bench_drain.restype = ctypes.c_longlong
+    L.rw_join_bench_drain.argtypes = [ctypes.c_void_p]
+    L.rw_agg_bench_apply.restype = ctypes.c_int
+    L.rw_agg_bench_apply.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
+    L.rw_agg_flush_device.restype = ctypes.c_longlong
+    L.rw_agg_flush_device.argtypes = [ctypes.c_void_p, ctypes.c_uint64]
+    L.rw_join_apply_aggout.restype = ctypes.c_int
+    L.rw_join_apply_aggout.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                       ctypes.c_int,
+                                       ctypes.POINTER(ctypes.c_uint32),
+                                       ctypes.c_int, ctypes.c_uint64]
+    L.rw_join_vnode_hop.restype = ctypes.c_int
+    L.rw_join_vnode_hop.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                    ctypes.c_uint32, ctypes.c_uint8,
+                                    ctypes.c_uint32]
+    L.rw_join_kernel_stats.argtypes = [ctypes.c_void_p, ctypes.POINTER(KernelStats)]
+    L.rw_join_stats_reset.argtypes = [ctypes.c_void_p]
+    L.rw_agg_kernel_stats.argtypes = [ctypes.c_void_p, ctypes.POINTER(KernelStats)]
+
+    batch_rows = CHUNK_ROWS * CHUNKS_PER_BATCH
+    calls = [(AGG_MAX, 1, T_I64), (AGG_COUNT_STAR, -1, T_I64)]
+    agg = ffi.HashAgg(gpu_lib, [T_I64, T_I64], [0], calls, 1,
+                      append_only=True, state_capacity_hint=1 << 20)
+    # join: left = bid [auction,bidder,price,date_time,rowid], right = the
+    # agg change stream projected to [window_end, maxprice]. Key: price =
+    # max(price). Fused BETWEEN: dt >= w - 10s AND dt <= w.
+    t5 = [T_I64] * 5
+    j = ffi.HashJoin(gpu_lib, JOIN_INNER, t5, [T_I64, T_I64],
+                     key_l=[2], key_r=[1], pk_l=[4], pk_r=[0],
+                     cond=(CMP_GE, 3, 5, -WINDOW_US), cond2=(CMP_LE, 3, 5, 0),
+                     state_capacity_hint=1 << 23,
+                     row_capacity_hint=(args.steps + args.warmup + 4)
+                     * batch_rows + 1_000_000)
+
+    n_batches = EPOCH_BATCHES
+    ones = lambda n: np.ones(n, np.uint8)
+    agg_batches, join_batches = [], []
+    rowid = 0
+    for b in range(n_batches):
+        # monotone date_time: windows_per_epoch windows per epoch
+        dt = np.sort(rng.integers(b * args.windows_per_epoch * WINDOW_US,
+                                  (b + 1) * args.windows_per_epoch * WINDOW_US,
+                                  batch_rows))
+        w = (dt // WINDOW_US + 1) * WINDOW_US  # TumbleStart + 10s = $expr1
+        price = rng.integers(1, 10**7, batch_rows)
+        auction = rng.integers(0, 1_000_000, batch_rows)
+        bidder = rng.integers(0, 1_000_000, batch_rows)
+        rid = np.arange(rowid, rowid + batch_rows)
+        rowid += batch_rows
+        ca = ffi.Chunk([T_I64, T_I64], np.zeros(batch_rows, np.uint8),
+                       [w, price], [ones(batch_rows)] * 2)
+        cc = ca.to_c()
+        h = L.rw_agg_bench_preload(agg.h, ctypes.byref(cc))
+        assert h, gpu_lib.last_error()
+        agg_batches.append(h)
+        cj = ffi.Chunk(t5, np.zeros(batch_rows, np.uint8),
+                       [auction, bidder, price, dt, rid],
+                       [ones(batch_rows)] * 5)
+        cc = cj.to_c()
+        h = L.rw_join_bench_preload(j.h, SIDE_LEFT, ctypes.byref(cc))
+        assert h, gpu_lib.last_error()
+        join_batches.append(h)
+
+    cmap = (ctypes.c_uint32 * 2)(0, 1)  # agg record [w, max, count] -> [w, max]
+    agg_ms = [0.0]
+
+    def step(i):
+        # exchange hop 1: vnode of the window key feeding the agg fragment
+        rc = L.rw_join_vnode_hop(j.h, join_batches[i % n_batches], 3,
+                                 ffi.T_I64, 256)
+        assert rc == 0, gpu_lib.last_error()
+        rc = L.rw_agg_bench_apply(agg.h, agg_batches[i % n_batches])
+        assert rc == 0, gpu_lib.last_error()
+        # exchange hop 2: vnode of price feeding the join-left fragment
+        rc = L.rw_join_vnode_hop(j.h, join_batches[i % n_batches], 2,
+                                 ffi.T_I64, 256)
+        assert rc == 0, gpu_lib.last_error()
+        rc = L.rw_join_bench_apply(j.h, SIDE_LEFT, join_batches[i % n_batches])
+        assert rc == 0, gpu_lib.last_error()
+        if (i + 1) % args.barrier_every == 0:
+            t0 = time.perf_counter()
+            n = L.rw_agg_flush_device(agg.h, i)
+            assert n >= 0, gpu_lib.last_error()
+            rc = L.rw_join_apply_aggout(j.h, agg.h, SIDE_RIGHT, cmap, 2,
+                                        ctypes.c_uint64(n))
+            assert rc == 0, gpu_lib.last_error()
+            agg_ms[0] += time.perf_counter() - t0
+            assert L.rw_join_bench_drain(j.h) >= 0, gpu_lib.last_error()
+
+    for i in range(args.warmup):
+        step(i)
+    L.rw_join_stats_reset(j.h)
+    L.rw_agg_stats_reset(agg.h)
+    agg_ms[0] = 0.0
+    if dist:
+        dist.barrier()
+    _dev_sync()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(i)
+    assert L.rw_join_bench_drain(j.h) >= 0, gpu_lib.last_error()
+    _dev_sync()
+    elapsed = time.perf_counter() - t0
+    if dist:
+        import torch
+
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    jks = KernelStats()
+    aks = KernelStats()
+    L.rw_join_kernel_stats(j.h, ctypes.byref(jks))
+    L.rw_agg_kernel_stats(agg.h, ctypes.byref(aks))
+    if rank == 0:
+        total_rows = args.steps * batch_rows * world
+        avg_probe_ms = jks.total_ms / max(jks.launches, 1)
+        # dominant kernel = the join-left probe+insert: ~120 B/row
+        # algorithmic (5x8 cols + 5 valid + op reads, 64-B record write,
+        # 8-B slot CAS; the right table is LIC-resident)
+        achieved = (120 * batch_rows) / (avg_probe_ms * 1e-3) / 1e9
+        result = {
+            "metric": "input rows/sec/GPU on Nexmark q7 full stream plan",
+            "value": total_rows / elapsed,
+            "unit": "rows/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed * 1000.0 / args.steps,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "int64",
+            "data": "synthetic",
+            "config": {
+                "workload": "nexmark_q7_pipeline",
+                "plan": "project -> vnode hop -> hash-agg max(price) by 10s "
+                        "window -> change stream -> inner join price = "
+                        "maxprice (BETWEEN fused) <- vnode hop <- bid",
+                "rows_per_step": batch_rows,
+                "windows_per_epoch": args.windows_per_epoch,
+                "barrier_every_steps": args.barrier_every,
+                "parallelism": f"dp{world}",
+            },
+            "kernels": {
+                "join_probe_avg_ms": avg_probe_ms,
+                "join_launches": jks.launches,
+                "agg_apply_avg_ms": aks.total_ms / max(aks.launches, 1),
+                "barrier_flush_plus_aggout_ms_total": agg_ms[0] * 1000.0,
+            },
+            "roofline": {
+                "bound": "hbm",
+                "achieved": achieved,
+                "peak": HBM_PEAK_GBS,
+                "unit": "GB/s",
+                "frac": achieved / HBM_PEAK_GBS,
+                "traffic": None,
+            },
+            "cpu_baseline": None,
+        }
+        print(json.dumps(result), flush=True)
+    j.close()
+    agg.close()
 
 
 def bench_q8(args, ffi, gpu_lib, rng, rank, world, dist):

@@ -146,6 +146,18 @@ typedef struct RwHashJoinDesc {
     uint8_t cond_op;  /* RwCmpOp */
     uint32_t cond_l;  /* column indices into the concatenated row */
     uint32_t cond_r;
+    /* Constant added to the RIGHT operand, and an optional second
+     * conjunct — enough for the reference q7 plan's fused
+     * `date_time BETWEEN $expr1 - 10s AND $expr1` predicate
+     * (nexmark.yaml q7: StreamFilter over StreamHashJoin; fusing a
+     * post-join filter into inner-join emission is semantics-preserving
+     * because inner joins carry no degrees). */
+    int64_t cond_rconst;
+    uint8_t has_cond2;
+    uint8_t cond2_op;
+    uint32_t cond2_l;
+    uint32_t cond2_r;
+    int64_t cond2_rconst;
     uint32_t chunk_size;
     uint64_t state_capacity_hint; /* expected distinct keys per side (0 = default) */
     uint64_t row_capacity_hint;   /* expected resident rows per side (0 = default) */

@@ -193,18 +193,30 @@ struct HashJoinOracle {
         auto at = [&](uint32_t i) -> const Datum& {
             return i < types_l.size() ? lrow[i] : rrow[i - types_l.size()];
         };
-        const Datum& a = at(d_store.cond_l);
-        const Datum& b = at(d_store.cond_r);
-        if (a.null || b.null) return false; // NULL comparison ⇒ false
-        uint8_t t = concat_types[d_store.cond_l];
-        int c = datum_cmp(a, b, t);
-        switch (d_store.cond_op) {
-            case RW_CMP_LT: return c < 0;
-            case RW_CMP_LE: return c <= 0;
-            case RW_CMP_GT: return c > 0;
-            case RW_CMP_GE: return c >= 0;
-        }
-        return false;
+        auto one = [&](uint8_t op, uint32_t li, uint32_t ri,
+                       long long rconst) -> bool {
+            const Datum& a = at(li);
+            Datum b = at(ri);
+            if (a.null || b.null) return false; // NULL comparison ⇒ false
+            b.i += rconst; // constant offset on the right operand
+            uint8_t t = concat_types[li];
+            int c = datum_cmp(a, b, t);
+            switch (op) {
+                case RW_CMP_LT: return c < 0;
+                case RW_CMP_LE: return c <= 0;
+                case RW_CMP_GT: return c > 0;
+                case RW_CMP_GE: return c >= 0;
+            }
+            return false;
+        };
+        if (!one(d_store.cond_op, d_store.cond_l, d_store.cond_r,
+                 d_store.cond_rconst))
+            return false;
+        if (d_store.has_cond2 &&
+            !one(d_store.cond2_op, d_store.cond2_l, d_store.cond2_r,
+                 d_store.cond2_rconst))
+            return false;
+        return true;
     }
 
     int push_chunk(int S, const RwChunk* chunk) {

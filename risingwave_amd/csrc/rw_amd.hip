@@ -1431,6 +1431,8 @@ struct HashAgg {
     std::vector<uint8_t> spill;
     int debug_mode = 0; // RW_AGG_DEBUG_MODE: 1 per-lane atomics, 2 no-dedupe
     uint8_t* d_vnode_bitmap = nullptr; // rescale scope (update_vnode_bitmap)
+    uint16_t* d_vnode_hop = nullptr;   // q7-pipeline exchange-hop vnodes
+    uint32_t d_vnode_hop_cap = 0;
     bool eowc = false; // emit-on-window-close (hash_agg.rs:421-474)
     bool has_pending_wm = false;
     int64_t pending_wm = 0;
@@ -3025,8 +3027,11 @@ struct JoinMeta {
     int n_out;
     uint8_t out_src[MAX_OUT]; // 0 = left, 1 = right
     uint8_t out_col[MAX_OUT];
-    uint8_t has_cond, cond_op;
-    uint8_t cond_src_l, cond_col_l, cond_src_r, cond_col_r;
+    uint8_t n_cond; // 0..2 conjuncts (AND)
+    struct {
+        uint8_t op, src_l, col_l, src_r, col_r;
+        long long rconst; // added to the right operand
+    } cond[2];
     uint8_t append_only;
     uint8_t join_type;    // RwJoinType
     uint8_t need_deg[2];  // need_left/right_degree (join/mod.rs:153-165)
@@ -3085,27 +3090,33 @@ __device__ __forceinline__ bool join_cond_ok(const JoinMeta& m, int probe_side,
                                              const JoinBatchDev& b, uint32_t r,
                                              uint32_t validbits,
                                              const long long* mvals) {
-    if (!m.has_cond) return true;
-    auto fetch = [&](uint8_t src, uint8_t col, int64_t* v) -> bool {
-        if ((int)src == probe_side) {
-            if (!b.col_valid[col][r]) return false;
-            *v = b.col_vals[col][r];
-        } else {
-            if (!((validbits >> col) & 1)) return false;
-            *v = mvals[col];
+    for (int ci = 0; ci < m.n_cond; ci++) {
+        auto& cd = m.cond[ci];
+        auto fetch = [&](uint8_t src, uint8_t col, int64_t* v) -> bool {
+            if ((int)src == probe_side) {
+                if (!b.col_valid[col][r]) return false;
+                *v = b.col_vals[col][r];
+            } else {
+                if (!((validbits >> col) & 1)) return false;
+                *v = mvals[col];
+            }
+            return true;
+        };
+        int64_t a, c;
+        if (!fetch(cd.src_l, cd.col_l, &a)) return false; // NULL ⇒ false
+        if (!fetch(cd.src_r, cd.col_r, &c)) return false;
+        c += cd.rconst;
+        bool ok;
+        switch (cd.op) {
+            case RW_CMP_LT: ok = a < c; break;
+            case RW_CMP_LE: ok = a <= c; break;
+            case RW_CMP_GT: ok = a > c; break;
+            case RW_CMP_GE: ok = a >= c; break;
+            default: ok = false;
         }
-        return true;
-    };
-    int64_t a, c;
-    if (!fetch(m.cond_src_l, m.cond_col_l, &a)) return false; // NULL ⇒ false
-    if (!fetch(m.cond_src_r, m.cond_col_r, &c)) return false;
-    switch (m.cond_op) {
-        case RW_CMP_LT: return a < c;
-        case RW_CMP_LE: return a <= c;
-        case RW_CMP_GT: return a > c;
-        case RW_CMP_GE: return a >= c;
+        if (!ok) return false;
     }
-    return false;
+    return true;
 }
 
 // probe the match side + update own side, one thread per probe row.
@@ -3699,27 +3710,33 @@ __device__ __forceinline__ bool jpart_cond_ok(const JoinMeta& m, int S,
                                               const long long* pv,
                                               uint32_t pvb, uint32_t mvb,
                                               const long long* mv) {
-    if (!m.has_cond) return true;
-    auto fetch = [&](uint8_t src, uint8_t col, int64_t* v) -> bool {
-        if ((int)src == S) {
-            if (!((pvb >> col) & 1)) return false;
-            *v = pv[col];
-        } else {
-            if (!((mvb >> col) & 1)) return false;
-            *v = mv[col];
+    for (int ci = 0; ci < m.n_cond; ci++) {
+        auto& cd = m.cond[ci];
+        auto fetch = [&](uint8_t src, uint8_t col, int64_t* v) -> bool {
+            if ((int)src == S) {
+                if (!((pvb >> col) & 1)) return false;
+                *v = pv[col];
+            } else {
+                if (!((mvb >> col) & 1)) return false;
+                *v = mv[col];
+            }
+            return true;
+        };
+        int64_t a, c;
+        if (!fetch(cd.src_l, cd.col_l, &a)) return false;
+        if (!fetch(cd.src_r, cd.col_r, &c)) return false;
+        c += cd.rconst;
+        bool ok;
+        switch (cd.op) {
+            case RW_CMP_LT: ok = a < c; break;
+            case RW_CMP_LE: ok = a <= c; break;
+            case RW_CMP_GT: ok = a > c; break;
+            case RW_CMP_GE: ok = a >= c; break;
+            default: ok = false;
         }
-        return true;
-    };
-    int64_t a, c;
-    if (!fetch(m.cond_src_l, m.cond_col_l, &a)) return false;
-    if (!fetch(m.cond_src_r, m.cond_col_r, &c)) return false;
-    switch (m.cond_op) {
-        case RW_CMP_LT: return a < c;
-        case RW_CMP_LE: return a <= c;
-        case RW_CMP_GT: return a > c;
-        case RW_CMP_GE: return a >= c;
+        if (!ok) return false;
     }
-    return false;
+    return true;
 }
 
 __global__ __launch_bounds__(256, 8) void jpart_probe_insert_kernel(
@@ -4358,6 +4375,8 @@ struct HashJoin {
     std::vector<RwChunk*> outq;
     uint8_t* zeros = nullptr; // all-zero null flags for pipeline dummy cols
     uint8_t* d_vnode_bitmap = nullptr; // rescale scope (update_vnode_bitmap)
+    uint16_t* d_vnode_hop = nullptr;   // q7-pipeline exchange-hop vnodes
+    uint32_t d_vnode_hop_cap = 0;
     // join-key watermark buffering (hash_join.rs:826-867)
     struct Wm { bool has = false; int64_t val = 0; };
     std::vector<uint32_t> wm_pos;
@@ -4429,8 +4448,6 @@ struct HashJoin {
                 out_types.push_back(types[1][idx - d->n_cols_l]);
             }
         }
-        m.has_cond = d->has_cond;
-        m.cond_op = d->cond_op;
         auto split = [&](uint32_t idx, uint8_t* src, uint8_t* col) {
             if (idx < d->n_cols_l) {
                 *src = 0;
@@ -4440,8 +4457,21 @@ struct HashJoin {
                 *col = (uint8_t)(idx - d->n_cols_l);
             }
         };
-        split(d->cond_l, &m.cond_src_l, &m.cond_col_l);
-        split(d->cond_r, &m.cond_src_r, &m.cond_col_r);
+        m.n_cond = 0;
+        if (d->has_cond) {
+            m.cond[0].op = d->cond_op;
+            split(d->cond_l, &m.cond[0].src_l, &m.cond[0].col_l);
+            split(d->cond_r, &m.cond[0].src_r, &m.cond[0].col_r);
+            m.cond[0].rconst = d->cond_rconst;
+            m.n_cond = 1;
+            if (d->has_cond2) {
+                m.cond[1].op = d->cond2_op;
+                split(d->cond2_l, &m.cond[1].src_l, &m.cond[1].col_l);
+                split(d->cond2_r, &m.cond[1].src_r, &m.cond[1].col_r);
+                m.cond[1].rconst = d->cond2_rconst;
+                m.n_cond = 2;
+            }
+        }
         m.append_only = d->append_only;
         m.join_type = d->join_type;
         m.need_deg[0] = j_need_degree(d->join_type, RW_SIDE_LEFT);
@@ -5209,6 +5239,107 @@ long long rw_join_bench_drain(void* h) {
 // view — the fragment edge agg(join(...)) of the TPC-H q3 MV without
 // leaving HBM. The agg must have been created with the join's output
 // schema as its input schema. Resets the join's output cursor.
+// agg-output -> join-input transpose: the agg's flush output block is
+// row-major [row][gk ++ calls]; the join consumes columnar batches. The
+// map picks which record columns become join input columns.
+struct AggOutColMap {
+    uint8_t n;
+    uint8_t map[MAX_COLS];
+};
+__global__ void aggout_to_join_kernel(const long long* avals,
+                                      const uint8_t* anulls,
+                                      const uint8_t* aops, uint32_t n,
+                                      int width, AggOutColMap cm,
+                                      JoinBatchDev b) {
+    uint32_t stride = gridDim.x * blockDim.x;
+    for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < n;
+         r += stride) {
+        for (int i = 0; i < cm.n; i++) {
+            b.col_vals[i][r] = avals[(size_t)r * width + cm.map[i]];
+            b.col_valid[i][r] = !anulls[(size_t)r * width + cm.map[i]];
+        }
+        b.ops[r] = aops[r];
+    }
+}
+
+// Device-resident fragment edge agg -> join (the q7 plan's
+// StreamHashJoin right input = the windowed max's change stream,
+// nexmark.yaml q7): applies the agg's device-resident flush output
+// (rw_agg_flush_device's emitted rows, still in HBM) as one join input
+// batch on `side`. Safe without the host conflict pre-pass: a U-/Delete
+// record's target row is always a PREVIOUS barrier's U+/Insert (the agg
+// emits at most one change pair per group per barrier), so no delete
+// targets a same-batch insert (DESIGN §3.2's only inner-join conflict).
+int rw_join_apply_aggout(void* join_h, void* agg_h, int side,
+                         const uint32_t* col_map, int n_map,
+                         uint64_t n_rows) {
+    auto* j = (HashJoin*)join_h;
+    auto* agg = (HashAgg*)agg_h;
+    if (side != 0 && side != 1) FAIL(RW_E_INVAL, "bad side");
+    if (n_map != j->m.n_cols[side])
+        FAIL(RW_E_INVAL, "col_map size %d != side cols %d", n_map,
+             j->m.n_cols[side]);
+    if (!n_rows) return RW_OK;
+    if (n_rows > UINT32_MAX) FAIL(RW_E_INVAL, "batch too large");
+    HIP_TRY(hipStreamSynchronize(agg->stream));
+    int rc = j->ensure_stage(side, (uint32_t)n_rows);
+    if (rc != RW_OK) return rc;
+    AggOutColMap cm{};
+    cm.n = (uint8_t)n_map;
+    for (int i = 0; i < n_map; i++) {
+        if (col_map[i] >= (uint32_t)agg->out_width)
+            FAIL(RW_E_INVAL, "col_map[%d]=%u outside agg record", i,
+                 col_map[i]);
+        cm.map[i] = (uint8_t)col_map[i];
+    }
+    JoinBatchDev b = j->stage[side];
+    b.vis = nullptr;
+    b.n_rows = (uint32_t)n_rows;
+    b.all_insert = 0; // change stream carries U-/U+ pairs
+    b.unique_keys = 0;
+    uint32_t blocks = ((uint32_t)n_rows + 255) / 256;
+    if (blocks > 2048) blocks = 2048;
+    aggout_to_join_kernel<<<blocks, 256, 0, j->stream>>>(
+        agg->t.out_vals, agg->t.out_nulls, agg->t.out_ops, (uint32_t)n_rows,
+        agg->out_width, cm, b);
+    return j->probe(side, b, true, 0, (uint32_t)n_rows);
+}
+
+// marshal the join's device-resident output buffer into host chunks on
+// the poll queue (what push_chunk does after its probe) — for callers of
+// rw_join_apply_aggout that want the emissions as chunks (parity tests)
+int rw_join_marshal_output(void* join_h) {
+    auto* j = (HashJoin*)join_h;
+    HIP_TRY(hipStreamSynchronize(j->stream));
+    return j->drain_output();
+}
+
+// device-resident VirtualNode::compute_chunk (vnode.rs:146-181) over a
+// preloaded join batch's columns, launched on the join's stream — the q7
+// pipeline's exchange-hop vnode computation at R=1 (the reference computes
+// vnodes in HashDataDispatcher even with one downstream; the N>1 payload
+// exchange is librw_exchange's path)
+int rw_join_vnode_hop(void* join_h, void* batch, uint32_t col, uint8_t type,
+                      uint32_t vnode_count) {
+    auto* j = (HashJoin*)join_h;
+    auto* b = (JoinBatchDev*)batch;
+    if (j->d_vnode_hop_cap < b->n_rows) {
+        if (j->d_vnode_hop) hipFree(j->d_vnode_hop);
+        HIP_TRY(hipMalloc(&j->d_vnode_hop, (size_t)b->n_rows * 2 + 2));
+        j->d_vnode_hop_cap = b->n_rows;
+    }
+    VnodeBatch vb{};
+    vb.col_vals[0] = b->col_vals[col];
+    vb.col_valid[0] = b->col_valid[col];
+    vb.n_rows = b->n_rows;
+    uint32_t blocks = (b->n_rows + 255) / 256;
+    if (blocks > 2048) blocks = 2048;
+    vnode_kernel<<<blocks, 256, 0, j->stream>>>(vb, 1, vnode_count,
+                                                j->d_vnode_hop, type, 0, 0,
+                                                0);
+    return RW_OK;
+}
+
 int rw_agg_apply_joinout(void* agg_h, void* join_h) {
     auto* agg = (HashAgg*)agg_h;
     auto* j = (HashJoin*)join_h;

@@ -93,6 +93,12 @@ class RwHashJoinDesc(C.Structure):
         ("cond_op", C.c_uint8),
         ("cond_l", C.c_uint32),
         ("cond_r", C.c_uint32),
+        ("cond_rconst", C.c_int64),
+        ("has_cond2", C.c_uint8),
+        ("cond2_op", C.c_uint8),
+        ("cond2_l", C.c_uint32),
+        ("cond2_r", C.c_uint32),
+        ("cond2_rconst", C.c_int64),
         ("chunk_size", C.c_uint32),
         ("state_capacity_hint", C.c_uint64),
         ("row_capacity_hint", C.c_uint64),
@@ -341,6 +347,7 @@ class HashAgg:
 class HashJoin:
     def __init__(self, lib: Lib, join_type, types_l, types_r, key_l, key_r,
                  pk_l, pk_r, output_indices=None, null_safe=None, cond=None,
+                 cond2=None,
                  chunk_size=1024, append_only=False, state_capacity_hint=0,
                  row_capacity_hint=0, wm_jk=()):
         """cond: (op, cond_l, cond_r) into the concatenated row, or None."""
@@ -379,10 +386,19 @@ class HashJoin:
         d.output_indices = self._oi
         if cond is not None:
             d.has_cond = 1
-            d.cond_op, d.cond_l, d.cond_r = cond
+            if len(cond) == 3:
+                d.cond_op, d.cond_l, d.cond_r = cond
+            else:
+                d.cond_op, d.cond_l, d.cond_r, d.cond_rconst = cond
         else:
             d.has_cond = 0
             d.cond_op = d.cond_l = d.cond_r = 0
+        if cond2 is not None:
+            d.has_cond2 = 1
+            if len(cond2) == 3:
+                d.cond2_op, d.cond2_l, d.cond2_r = cond2
+            else:
+                d.cond2_op, d.cond2_l, d.cond2_r, d.cond2_rconst = cond2
         d.chunk_size = chunk_size
         d.state_capacity_hint = state_capacity_hint
         d.row_capacity_hint = row_capacity_hint

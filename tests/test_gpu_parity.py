@@ -1180,3 +1180,89 @@ def test_join_partitioned_pipeline_parity():
     g.close()
     o.close()
     del os.environ["RW_JOIN_PART"]
+
+
+def test_q7_pipeline_parity():
+    # Full q7 plan chain: HashAgg max(price) by window
+    # -> change stream -> inner join on price = max(price) with the BETWEEN
+    # filter fused into emission (reference stream_plan nexmark.yaml q7).
+    # GPU chain: device-resident hop (rw_agg_flush_device +
+    # rw_join_apply_aggout); oracle chain: agg poll -> project -> join push.
+    # Join outputs compared as multisets per barrier.
+    import ctypes
+
+    from rwtest.ffi import AGG_COUNT_STAR, AGG_MAX, CMP_GE, CMP_LE, \
+        JOIN_INNER, SIDE_LEFT, SIDE_RIGHT
+
+    rng = np.random.default_rng(41)
+    W = 10_000_000
+    calls = [(AGG_MAX, 1, T_I64), (AGG_COUNT_STAR, -1, T_I64)]
+    ga = ffi.HashAgg(gpu(), [T_I64, T_I64], [0], calls, 1, append_only=True)
+    oa = ffi.HashAgg(oracle(), [T_I64, T_I64], [0], calls, 1,
+                     append_only=True)
+    t5 = [T_I64] * 5
+    jkw = dict(key_l=[2], key_r=[1], pk_l=[4], pk_r=[0],
+               cond=(CMP_GE, 3, 5, -W), cond2=(CMP_LE, 3, 5, 0))
+    gj = ffi.HashJoin(gpu(), JOIN_INNER, t5, [T_I64, T_I64], **jkw)
+    oj = ffi.HashJoin(oracle(), JOIN_INNER, t5, [T_I64, T_I64], **jkw)
+
+    L = gpu().lib
+    L.rw_agg_flush_device.restype = ctypes.c_longlong
+    L.rw_agg_flush_device.argtypes = [ctypes.c_void_p, ctypes.c_uint64]
+    L.rw_join_apply_aggout.restype = ctypes.c_int
+    L.rw_join_apply_aggout.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                       ctypes.c_int,
+                                       ctypes.POINTER(ctypes.c_uint32),
+                                       ctypes.c_int, ctypes.c_uint64]
+    L.rw_join_marshal_output.restype = ctypes.c_int
+    L.rw_join_marshal_output.argtypes = [ctypes.c_void_p]
+    cmap = (ctypes.c_uint32 * 2)(0, 1)
+
+    rowid = 0
+    for epoch in range(5):
+        for push in range(3):
+            n = 1024
+            dt = np.sort(rng.integers(epoch * 4 * W, (epoch * 4 + 4) * W, n))
+            w = (dt // W + 1) * W
+            # prices in a SMALL space so join matches and max-updates happen
+            price = rng.integers(1, 50, n)
+            auction = rng.integers(0, 100, n)
+            bidder = rng.integers(0, 100, n)
+            rid = np.arange(rowid, rowid + n)
+            rowid += n
+            ca = mk_chunk([T_I64, T_I64], np.zeros(n, np.uint8), [w, price])
+            cj = mk_chunk(t5, np.zeros(n, np.uint8),
+                          [auction, bidder, price, dt, rid])
+            ga.push(ca)
+            oa.push(ca)
+            gj.push(SIDE_LEFT, cj)
+            oj.push(SIDE_LEFT, cj)
+        # barrier: agg flush; change stream feeds the join's right side
+        n_changes = L.rw_agg_flush_device(ga.h, epoch + 1)
+        assert n_changes >= 0, gpu().last_error()
+        rc = L.rw_join_apply_aggout(gj.h, ga.h, SIDE_RIGHT, cmap, 2,
+                                    ctypes.c_uint64(n_changes))
+        assert rc == 0, gpu().last_error()
+        assert L.rw_join_marshal_output(gj.h) == 0, gpu().last_error()
+
+        oa.flush(epoch + 1)
+        ochunks = oa.poll_all()
+        ops, rows = [], []
+        for c in ochunks:
+            for op_tok, vals in c.visible_rows():
+                ops.append(ffi.OP_BY_TOKEN[op_tok])
+                rows.append(vals)
+        if rows:
+            # project the change stream to [window_end, maxprice]
+            wcol = np.array([r[0] for r in rows], dtype=np.int64)
+            mcol = np.array([r[1] for r in rows], dtype=np.int64)
+            assert not any(r[0] is None or r[1] is None for r in rows)
+            oj.push(SIDE_RIGHT,
+                    mk_chunk([T_I64, T_I64], np.array(ops, np.uint8),
+                             [wcol, mcol]))
+        mg = rows_multiset(gj.poll_all())
+        mo = rows_multiset(oj.poll_all())
+        assert mg == mo, (f"epoch {epoch}: GPU {len(mg)} join rows vs "
+                          f"oracle {len(mo)}")
+    for x in (ga, oa, gj, oj):
+        x.close()
