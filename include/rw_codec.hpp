@@ -124,4 +124,75 @@ inline void value_encode_datum(std::vector<uint8_t>& buf, uint8_t type,
     }
 }
 
+// inverse of value_encode_datum: reads one datum, returns bytes consumed
+// (0 on underflow/unknown type)
+inline size_t value_decode_datum(const uint8_t* p, size_t avail, uint8_t type,
+                                 DatumC* out) {
+    if (avail < 1) return 0;
+    if (p[0] == 0) {
+        *out = {true, 0, 0};
+        return 1;
+    }
+    auto le = [&](int n) {
+        uint64_t v = 0;
+        for (int k = 0; k < n; k++) v |= (uint64_t)p[1 + k] << (8 * k);
+        return v;
+    };
+    switch (type) {
+        case RW_T_I64:
+        case RW_T_TS:
+            if (avail < 9) return 0;
+            *out = {false, (int64_t)le(8), 0};
+            return 9;
+        case RW_T_I32:
+            if (avail < 5) return 0;
+            *out = {false, (int32_t)(uint32_t)le(4), 0};
+            return 5;
+        case RW_T_BOOL:
+            if (avail < 2) return 0;
+            *out = {false, (int64_t)p[1], 0};
+            return 2;
+        case RW_T_F64: {
+            if (avail < 9) return 0;
+            uint64_t bits = le(8);
+            double d;
+            std::memcpy(&d, &bits, 8);
+            *out = {false, 0, d};
+            return 9;
+        }
+        case RW_T_F32: {
+            if (avail < 5) return 0;
+            uint32_t bits = (uint32_t)le(4);
+            float f;
+            std::memcpy(&f, &bits, 4);
+            *out = {false, 0, (double)f};
+            return 5;
+        }
+    }
+    return 0;
+}
+
+// iterate spill frames ([put u8][klen u32 LE][key][vlen u32 LE][value]);
+// calls fn(put, key_ptr, klen, val_ptr, vlen); returns false on malformed
+template <typename F>
+inline bool for_each_frame(const uint8_t* buf, uint64_t len, F&& fn) {
+    uint64_t off = 0;
+    auto rd32 = [&](uint64_t o) {
+        return (uint32_t)buf[o] | ((uint32_t)buf[o + 1] << 8) |
+               ((uint32_t)buf[o + 2] << 16) | ((uint32_t)buf[o + 3] << 24);
+    };
+    while (off + 5 <= len) {
+        uint8_t put = buf[off];
+        uint32_t klen = rd32(off + 1);
+        if (off + 5 + klen + 4 > len) return false;
+        const uint8_t* k = buf + off + 5;
+        uint32_t vlen = rd32(off + 5 + klen);
+        if (off + 5 + klen + 4 + vlen > len) return false;
+        const uint8_t* v = buf + off + 5 + klen + 4;
+        fn(put, k, klen, v, vlen);
+        off += 5 + (uint64_t)klen + 4 + vlen;
+    }
+    return off == len;
+}
+
 } // namespace rwcodec

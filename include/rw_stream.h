@@ -256,6 +256,23 @@ void rw_group_top_n_destroy(void* h);
  * tables via rw_join_degree_drain below). Caller frees with
  * rw_spill_free. */
 int rw_agg_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len);
+/* ----- state restore (crash recovery / cache rehydration) -----
+ *
+ * Rebuild device state from spill records previously produced by the
+ * drains (the reference restores executor state from its state tables on
+ * recovery — src/meta/src/barrier/worker.rs:1074 — and on cache miss,
+ * join/hash_join.rs:232-260). `buf` = concatenated drain outputs in epoch
+ * order (PUTs and DELETEs; later frames win — the state store's merged
+ * view). Must be called on a freshly created executor, before any input.
+ * Restored rows are NOT re-emitted by the next drain (they predate the
+ * epoch). Aggregates with materialized-input state (retractable min/max)
+ * reject restore loudly: their row sets live in minput state tables that
+ * are not yet spilled. Join: `deg_buf` = the degree-table
+ * drain stream (same key encoding), required for join types that keep
+ * degrees; pass NULL/0 otherwise. */
+int rw_hash_agg_restore(void* h, const uint8_t* buf, uint64_t len);
+int rw_hash_join_restore(void* h, int side, const uint8_t* buf, uint64_t len,
+                         const uint8_t* deg_buf, uint64_t deg_len);
 int rw_join_checkpoint_drain(void* h, int side, uint8_t** buf, uint64_t* len);
 int rw_topn_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len);
 /* DISTINCT dedup tables (one StateTable per distinct column in the

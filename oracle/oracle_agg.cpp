@@ -33,6 +33,14 @@ namespace orc {
 
 thread_local std::string g_err;
 
+#define FAIL(code, ...)                         \
+    do {                                        \
+        char _b[256];                           \
+        snprintf(_b, sizeof(_b), __VA_ARGS__);  \
+        g_err = _b;                             \
+        return code;                            \
+    } while (0)
+
 struct ValueState {
     // count / sum: has=false ⇔ Rust state None (sum) — count init 0 has=true
     bool has = false;
@@ -304,6 +312,86 @@ struct HashAggOracle {
     }
 
     // get_outputs (agg_group.rs:431-467)
+    // Rebuild state from drained spill records (rw_stream.h restore
+    // contract): net PUT/DELETE frames by key, decode the value-encoded
+    // row (group key ++ outputs), and seed the value states + prev
+    // outputs so the next flush emits only real changes.
+    int restore(const uint8_t* buf, uint64_t len) {
+        for (auto m : call_is_minput)
+            if (m)
+                FAIL(RW_E_INVAL,
+                     "restore with materialized-input aggregates requires "
+                     "minput-table spill (not yet drained)");
+        if (!groups.empty() || !dirty.empty())
+            FAIL(RW_E_INVAL, "restore requires a fresh executor");
+        std::map<std::string, std::vector<uint8_t>> merged;
+        bool ok = rwcodec::for_each_frame(
+            buf, len,
+            [&](uint8_t put, const uint8_t* k, uint32_t klen,
+                const uint8_t* v, uint32_t vlen) {
+                std::string key((const char*)k, klen);
+                if (put)
+                    merged[key].assign(v, v + vlen);
+                else
+                    merged.erase(key);
+            });
+        if (!ok) FAIL(RW_E_INVAL, "malformed spill stream");
+        for (auto& [kbytes, val] : merged) {
+            (void)kbytes;
+            // decode group key ++ per-call outputs
+            size_t off = 0;
+            Row key(group_key_types.size());
+            for (size_t i = 0; i < group_key_types.size(); i++) {
+                rwcodec::DatumC d;
+                size_t n = rwcodec::value_decode_datum(
+                    val.data() + off, val.size() - off, group_key_types[i],
+                    &d);
+                if (!n) FAIL(RW_E_INVAL, "restore: bad group datum");
+                off += n;
+                key[i] = d.null ? Datum()
+                                : (type_is_float(group_key_types[i])
+                                       ? Datum::of_d(d.d)
+                                       : Datum::of_i(d.i));
+            }
+            Row outs(calls.size());
+            for (size_t ci = 0; ci < calls.size(); ci++) {
+                rwcodec::DatumC d;
+                size_t n = rwcodec::value_decode_datum(
+                    val.data() + off, val.size() - off, calls[ci].ret_type,
+                    &d);
+                if (!n) FAIL(RW_E_INVAL, "restore: bad state datum");
+                off += n;
+                outs[ci] = d.null ? Datum()
+                                  : (type_is_float(calls[ci].ret_type)
+                                         ? Datum::of_d(d.d)
+                                         : Datum::of_i(d.i));
+            }
+            AggGroupState& g = touch(key);
+            for (size_t ci = 0; ci < calls.size(); ci++) {
+                ValueState& v2 = g.vstates[ci];
+                const Datum& o = outs[ci];
+                v2.has = !o.null;
+                if (!o.null) {
+                    if (v2.is_float) v2.d = o.d;
+                    else v2.i = o.i;
+                }
+                // count/sum0 state is always Some
+                if (calls[ci].kind == RW_AGG_COUNT_STAR ||
+                    calls[ci].kind == RW_AGG_COUNT ||
+                    calls[ci].kind == RW_AGG_SUM0)
+                    v2.has = true;
+            }
+            if (!eowc) {
+                g.has_prev = true;
+                g.prev_outputs = outs;
+            }
+        }
+        // restored groups are clean (they predate the epoch)
+        dirty_order.clear();
+        dirty.clear();
+        return RW_OK;
+    }
+
     Row get_outputs(AggGroupState& g) {
         // current row count from the count(*) value state
         int64_t rc = g.vstates[row_count_index].i;
@@ -622,6 +710,10 @@ int rw_agg_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len) {
     memcpy(*buf, a->spill.data(), a->spill.size());
     a->spill.clear();
     return RW_OK;
+}
+
+int rw_hash_agg_restore(void* h, const uint8_t* buf, uint64_t len) {
+    return ((HashAggOracle*)h)->restore(buf, len);
 }
 
 int rw_agg_n_dedup_tables(void* h) {

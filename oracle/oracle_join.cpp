@@ -33,7 +33,15 @@
 
 namespace orc {
 
-extern thread_local std::string g_err; // defined in oracle_agg.cpp
+extern thread_local std::string g_err;
+
+#define FAIL(code, ...)                         \
+    do {                                        \
+        char _b[256];                           \
+        snprintf(_b, sizeof(_b), __VA_ARGS__);  \
+        g_err = _b;                             \
+        return code;                            \
+    } while (0) // defined in oracle_agg.cpp
 
 // join type predicates (executor/join/mod.rs:103-165)
 static bool is_outer_side(uint8_t t, int side) {
@@ -217,6 +225,74 @@ struct HashJoinOracle {
                  d_store.cond2_rconst))
             return false;
         return true;
+    }
+
+    // Rebuild one side's state from drained spill records (rw_stream.h
+    // restore contract): net PUT/DELETE frames by key, decode the
+    // value-encoded full row, re-derive jk/pk, and attach degrees from the
+    // degree-table drain stream (matched by the shared memcmp(jk ∥ pk)
+    // key encoding, join/row.rs:99-113).
+    int restore(int S, const uint8_t* buf, uint64_t len, const uint8_t* dbuf,
+                uint64_t dlen) {
+        JoinSideState& sd = side[S];
+        if (!sd.table.empty())
+            FAIL(RW_E_INVAL, "restore requires a fresh executor side");
+        std::map<std::string, std::vector<uint8_t>> merged;
+        bool ok = rwcodec::for_each_frame(
+            buf, len,
+            [&](uint8_t put, const uint8_t* k, uint32_t klen,
+                const uint8_t* v, uint32_t vlen) {
+                std::string key((const char*)k, klen);
+                if (put)
+                    merged[key].assign(v, v + vlen);
+                else
+                    merged.erase(key);
+            });
+        if (!ok) FAIL(RW_E_INVAL, "malformed spill stream");
+        std::map<std::string, uint64_t> degs;
+        if (dbuf && dlen) {
+            ok = rwcodec::for_each_frame(
+                dbuf, dlen,
+                [&](uint8_t put, const uint8_t* k, uint32_t klen,
+                    const uint8_t* v, uint32_t vlen) {
+                    std::string key((const char*)k, klen);
+                    if (put && vlen >= 9 && v[vlen - 9] == 1) {
+                        uint64_t d = 0;
+                        for (int b = 0; b < 8; b++)
+                            d |= (uint64_t)v[vlen - 8 + b] << (8 * b);
+                        degs[key] = d;
+                    } else if (!put) {
+                        degs.erase(key);
+                    }
+                });
+            if (!ok) FAIL(RW_E_INVAL, "malformed degree spill stream");
+        }
+        for (auto& [kbytes, val] : merged) {
+            Row row(sd.types.size());
+            size_t off = 0;
+            for (size_t c = 0; c < sd.types.size(); c++) {
+                rwcodec::DatumC d;
+                size_t n = rwcodec::value_decode_datum(
+                    val.data() + off, val.size() - off, sd.types[c], &d);
+                if (!n) FAIL(RW_E_INVAL, "restore: bad row datum");
+                off += n;
+                row[c] = d.null ? Datum()
+                                : (type_is_float(sd.types[c])
+                                       ? Datum::of_d(d.d)
+                                       : Datum::of_i(d.i));
+            }
+            Row jk(sd.key_idx.size()), pk(sd.pk_idx.size());
+            for (size_t i = 0; i < sd.key_idx.size(); i++)
+                jk[i] = row[sd.key_idx[i]];
+            for (size_t i = 0; i < sd.pk_idx.size(); i++)
+                pk[i] = row[sd.pk_idx[i]];
+            JoinEntry e;
+            e.row = std::move(row);
+            auto di = degs.find(kbytes);
+            if (di != degs.end()) e.degree = di->second;
+            side_entry(sd, jk).emplace(std::move(pk), std::move(e));
+        }
+        return RW_OK;
     }
 
     int push_chunk(int S, const RwChunk* chunk) {
@@ -565,6 +641,12 @@ struct HashJoinOracle {
 using namespace orc;
 
 extern "C" {
+
+int rw_hash_join_restore(void* h, int side, const uint8_t* buf, uint64_t len,
+                         const uint8_t* deg_buf, uint64_t deg_len) {
+    if (side != 0 && side != 1) return RW_E_INVAL;
+    return ((HashJoinOracle*)h)->restore(side, buf, len, deg_buf, deg_len);
+}
 
 void* rw_hash_join_create(const RwHashJoinDesc* d) { return new HashJoinOracle(d); }
 int rw_hash_join_push_chunk(void* h, int side, const RwChunk* c) {

@@ -1386,6 +1386,45 @@ __global__ void agg_eowc_dump_kernel(AggTableDev t, int KW, int n_calls,
 
 extern "C" __global__ void agg_counters_reset_kernel(uint32_t* counters);
 
+// state restore (rw_stream.h contract): seed value states + prev outputs
+// from decoded state rows. err code 1 = table full.
+__global__ void agg_restore_kernel(AggTableDev t, int KW, int n_calls,
+                                   const int64_t* keys,
+                                   const uint32_t* knulls,
+                                   const long long* vals,
+                                   const uint8_t* vnulls, uint32_t n,
+                                   int set_prev, AggCallDev c0, AggCallDev c1,
+                                   AggCallDev c2, AggCallDev c3) {
+    AggCallDev calls[4] = {c0, c1, c2, c3};
+    size_t cap = (size_t)t.cap_mask + 1;
+    uint32_t stride = gridDim.x * blockDim.x;
+    for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        int64_t kw[MAX_KW];
+        for (int k = 0; k < KW; k++) kw[k] = keys[(size_t)i * KW + k];
+        uint32_t slot = table_find_or_insert(t.state, t.keys, t.key_nulls,
+                                             t.cap_mask, kw, knulls[i], KW);
+        if (slot == UINT32_MAX) {
+            atomicExch(&t.counters[2], 1u);
+            continue;
+        }
+        for (int ci = 0; ci < n_calls; ci++) {
+            long long v = vals[(size_t)ci * n + i];
+            uint8_t nu = vnulls[(size_t)ci * n + i];
+            long long init = 0;
+            if (calls[ci].kind == RW_AGG_MIN) init = INT64_MAX;
+            if (calls[ci].kind == RW_AGG_MAX) init = INT64_MIN;
+            t.acc[(size_t)ci * cap + slot] = nu ? init : v;
+            t.has[(size_t)ci * cap + slot] = !nu;
+            if (set_prev) {
+                t.prev[(size_t)ci * cap + slot] = v;
+                t.prev_null[(size_t)ci * cap + slot] = nu;
+            }
+        }
+        if (set_prev) t.has_prev[slot] = 1;
+    }
+}
+
 struct HashAgg {
     RwHashAggDesc desc;
     std::vector<uint8_t> input_types;
@@ -2768,6 +2807,81 @@ int rw_agg_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len) {
     return RW_OK;
 }
 
+int rw_hash_agg_restore(void* h, const uint8_t* buf, uint64_t len) {
+    auto* agg = (HashAgg*)h;
+    if (agg->n_minput > 0)
+        FAIL(RW_E_INVAL,
+             "restore with materialized-input aggregates requires "
+             "minput-table spill (not yet drained)");
+    std::map<std::string, std::vector<uint8_t>> merged;
+    bool ok = rwcodec::for_each_frame(
+        buf, len,
+        [&](uint8_t put, const uint8_t* k, uint32_t klen, const uint8_t* v,
+            uint32_t vlen) {
+            std::string key((const char*)k, klen);
+            if (put)
+                merged[key].assign(v, v + vlen);
+            else
+                merged.erase(key);
+        });
+    if (!ok) FAIL(RW_E_INVAL, "malformed spill stream");
+    uint32_t n = (uint32_t)merged.size();
+    if (!n) return RW_OK;
+    int KW = agg->KW, nc = agg->n_calls;
+    std::vector<int64_t> keys((size_t)n * KW);
+    std::vector<uint32_t> knulls(n, 0);
+    std::vector<long long> vals((size_t)nc * n);
+    std::vector<uint8_t> vnulls((size_t)nc * n);
+    uint32_t i = 0;
+    for (auto& [kbytes, val] : merged) {
+        (void)kbytes;
+        size_t off = 0;
+        for (int c = 0; c < KW + nc; c++) {
+            rwcodec::DatumC d;
+            uint8_t ty = agg->out_types[c];
+            size_t got = rwcodec::value_decode_datum(val.data() + off,
+                                                     val.size() - off, ty, &d);
+            if (!got) FAIL(RW_E_INVAL, "restore: bad state datum");
+            off += got;
+            if (c < KW) {
+                keys[(size_t)i * KW + c] = d.null ? 0 : d.i;
+                knulls[i] |= (uint32_t)(d.null != 0) << c;
+            } else {
+                vals[(size_t)(c - KW) * n + i] = d.i;
+                vnulls[(size_t)(c - KW) * n + i] = d.null;
+            }
+        }
+        i++;
+    }
+    int64_t* dkeys = nullptr;
+    uint32_t* dknulls = nullptr;
+    long long* dvals = nullptr;
+    uint8_t* dvnulls = nullptr;
+    HIP_TRY(hipMalloc(&dkeys, keys.size() * 8));
+    HIP_TRY(hipMalloc(&dknulls, knulls.size() * 4));
+    HIP_TRY(hipMalloc(&dvals, vals.size() * 8));
+    HIP_TRY(hipMalloc(&dvnulls, vnulls.size()));
+    HIP_TRY(hipMemcpy(dkeys, keys.data(), keys.size() * 8,
+                      hipMemcpyHostToDevice));
+    HIP_TRY(hipMemcpy(dknulls, knulls.data(), knulls.size() * 4,
+                      hipMemcpyHostToDevice));
+    HIP_TRY(hipMemcpy(dvals, vals.data(), vals.size() * 8,
+                      hipMemcpyHostToDevice));
+    HIP_TRY(hipMemcpy(dvnulls, vnulls.data(), vnulls.size(),
+                      hipMemcpyHostToDevice));
+    agg_restore_kernel<<<agg->grid_for(n), 256, 0, agg->stream>>>(
+        agg->t, KW, nc, dkeys, dknulls, dvals, dvnulls, n,
+        agg->eowc ? 0 : 1, agg->cd(0), agg->cd(1), agg->cd(2), agg->cd(3));
+    int rc = hipStreamSynchronize(agg->stream) == hipSuccess ? RW_OK
+                                                             : RW_E_INTERNAL;
+    hipFree(dkeys);
+    hipFree(dknulls);
+    hipFree(dvals);
+    hipFree(dvnulls);
+    if (rc != RW_OK) FAIL(RW_E_INTERNAL, "restore sync failed");
+    return agg->check_overflow();
+}
+
 int rw_agg_n_dedup_tables(void* h) {
     return (int)((HashAgg*)h)->distinct_slots.size();
 }
@@ -3988,6 +4102,52 @@ __global__ __launch_bounds__(256) void jpart_probe_insert_lds_kernel(
     __syncthreads();
     for (uint32_t i = threadIdx.x; i < wslots; i += blockDim.x)
         own.slots8[wbase + i] = win[i];
+}
+
+// join state restore: append decoded state rows + bucket links + degrees
+// (rw_stream.h contract). err: 3 = row store full.
+__global__ void join_restore_kernel(JoinBatchDev b, JoinSideDev sd,
+                                    JoinMeta m, int S,
+                                    const uint32_t* degrees, uint32_t* err) {
+    uint32_t stride = gridDim.x * blockDim.x;
+    int lane = threadIdx.x & 63;
+    uint32_t n = b.n_rows;
+    uint32_t iters = (n + stride - 1) / stride;
+    for (uint32_t it = 0; it < iters; it++) {
+        uint32_t r = it * stride + blockIdx.x * blockDim.x + threadIdx.x;
+        bool active = r < n;
+        uint64_t wmask = __ballot(active);
+        if (!active) continue;
+        int leader = 63 - __clzll(wmask);
+        uint32_t base = 0;
+        if (lane == leader)
+            base = atomicAdd(sd.row_cursor, (uint32_t)__popcll(wmask));
+        base = (uint32_t)__shfl((int)base, leader);
+        uint32_t row = base + (uint32_t)__popcll(wmask & ((1ULL << lane) - 1));
+        if (row >= sd.row_cap) {
+            atomicExch(err, 3u);
+            continue;
+        }
+        JoinRowHdr* hd = jrow(sd, row);
+        long long* hv = jvals(hd);
+        uint32_t vb = 0;
+        for (int c = 0; c < m.n_cols[S]; c++) {
+            hv[c] = b.col_vals[c][r];
+            vb |= (uint32_t)(b.col_valid[c][r] != 0) << c;
+        }
+        hd->validbits = vb;
+        hd->degree = degrees ? degrees[r] : 0;
+        hd->alive = 1;
+        int64_t kw[MAX_KW];
+        uint32_t nm = 0;
+        for (int i = 0; i < m.KW; i++) {
+            uint8_t col = m.key_cols[S][i];
+            bool valid = (vb >> col) & 1;
+            kw[i] = valid ? hv[col] : 0;
+            nm |= (uint32_t)(!valid) << i;
+        }
+        jbucket_insert(sd, hash_key(kw, nm, m.KW), row, false);
+    }
 }
 
 // watermark TTL sweeps (state_table watermark cleaning):
@@ -5414,6 +5574,99 @@ int rw_join_checkpoint_drain(void* h, int side, uint8_t** buf,
     int rc = j->checkpoint_drain(side, sp);
     if (rc != RW_OK) return rc;
     return spill_export(sp, buf, len);
+}
+
+int rw_hash_join_restore(void* h, int side, const uint8_t* buf,
+                         uint64_t len, const uint8_t* deg_buf,
+                         uint64_t deg_len) {
+    auto* j = (HashJoin*)h;
+    if (side != 0 && side != 1) FAIL(RW_E_INVAL, "bad side");
+    std::map<std::string, std::vector<uint8_t>> merged;
+    bool ok = rwcodec::for_each_frame(
+        buf, len,
+        [&](uint8_t put, const uint8_t* k, uint32_t klen, const uint8_t* v,
+            uint32_t vlen) {
+            std::string key((const char*)k, klen);
+            if (put)
+                merged[key].assign(v, v + vlen);
+            else
+                merged.erase(key);
+        });
+    if (!ok) FAIL(RW_E_INVAL, "malformed spill stream");
+    std::map<std::string, uint32_t> degs;
+    if (deg_buf && deg_len) {
+        ok = rwcodec::for_each_frame(
+            deg_buf, deg_len,
+            [&](uint8_t put, const uint8_t* k, uint32_t klen,
+                const uint8_t* v, uint32_t vlen) {
+                std::string key((const char*)k, klen);
+                if (put && vlen >= 9 && v[vlen - 9] == 1) {
+                    uint64_t d = 0;
+                    for (int b = 0; b < 8; b++)
+                        d |= (uint64_t)v[vlen - 8 + b] << (8 * b);
+                    degs[key] = (uint32_t)d;
+                } else if (!put) {
+                    degs.erase(key);
+                }
+            });
+        if (!ok) FAIL(RW_E_INVAL, "malformed degree spill stream");
+    }
+    uint32_t n = (uint32_t)merged.size();
+    if (!n) return RW_OK;
+    int ncols = j->m.n_cols[side];
+    std::vector<std::vector<int64_t>> cols(ncols,
+                                           std::vector<int64_t>(n));
+    std::vector<std::vector<uint8_t>> valid(ncols,
+                                            std::vector<uint8_t>(n));
+    std::vector<uint32_t> degrees(n, 0);
+    uint32_t i = 0;
+    for (auto& [kbytes, val] : merged) {
+        size_t off = 0;
+        for (int c = 0; c < ncols; c++) {
+            rwcodec::DatumC d;
+            size_t got = rwcodec::value_decode_datum(
+                val.data() + off, val.size() - off, j->types[side][c], &d);
+            if (!got) FAIL(RW_E_INVAL, "restore: bad row datum");
+            off += got;
+            cols[c][i] = d.null ? 0 : d.i;
+            valid[c][i] = !d.null;
+        }
+        auto di = degs.find(kbytes);
+        if (di != degs.end()) degrees[i] = di->second;
+        i++;
+    }
+    int rc = j->ensure_stage(side, n);
+    if (rc != RW_OK) return rc;
+    JoinBatchDev b = j->stage[side];
+    for (int c = 0; c < ncols; c++) {
+        HIP_TRY(hipMemcpy(b.col_vals[c], cols[c].data(), (size_t)n * 8,
+                          hipMemcpyHostToDevice));
+        HIP_TRY(hipMemcpy(b.col_valid[c], valid[c].data(), n,
+                          hipMemcpyHostToDevice));
+    }
+    b.vis = nullptr;
+    b.n_rows = n;
+    uint32_t* ddeg = nullptr;
+    HIP_TRY(hipMalloc(&ddeg, (size_t)n * 4));
+    HIP_TRY(hipMemcpy(ddeg, degrees.data(), (size_t)n * 4,
+                      hipMemcpyHostToDevice));
+    uint32_t blocks = (n + 255) / 256;
+    if (blocks > 2048) blocks = 2048;
+    join_restore_kernel<<<blocks, 256, 0, j->stream>>>(
+        b, j->side[side], j->m, side, ddeg, j->out.counters + 1);
+    int rcs = hipStreamSynchronize(j->stream) == hipSuccess ? RW_OK
+                                                            : RW_E_INTERNAL;
+    hipFree(ddeg);
+    if (rcs != RW_OK) FAIL(RW_E_INTERNAL, "restore sync failed");
+    uint32_t ctr[2];
+    HIP_TRY(hipMemcpy(ctr, j->out.counters, 8, hipMemcpyDeviceToHost));
+    if (ctr[1]) FAIL(RW_E_INTERNAL, "restore overflow (code %u)", ctr[1]);
+    // restored rows predate the epoch: the next drain must not re-PUT them
+    uint32_t cur = 0;
+    HIP_TRY(hipMemcpy(&cur, j->side[side].row_cursor, 4,
+                      hipMemcpyDeviceToHost));
+    j->flush_mark[side] = cur;
+    return RW_OK;
 }
 
 int rw_join_degree_drain(void* h, int side, uint8_t** buf, uint64_t* len) {

@@ -639,3 +639,44 @@ def topn_checkpoint_drain(lib, h):
     L.rw_spill_free.argtypes = [C.c_void_p]
     L.rw_spill_free(C.cast(buf, C.c_void_p))
     return out
+
+
+def agg_checkpoint_drain_bytes(lib, h):
+    """Drain the agg's §8f-2 checkpoint spill buffer; returns raw bytes."""
+    L = lib.lib
+    L.rw_agg_checkpoint_drain.restype = C.c_int
+    L.rw_agg_checkpoint_drain.argtypes = [C.c_void_p,
+                                          C.POINTER(C.POINTER(C.c_uint8)),
+                                          C.POINTER(C.c_uint64)]
+    buf = C.POINTER(C.c_uint8)()
+    ln = C.c_uint64()
+    rc = L.rw_agg_checkpoint_drain(h, C.byref(buf), C.byref(ln))
+    if rc != 0:
+        raise RuntimeError(f"agg drain failed {rc}: {lib.last_error()}")
+    out = bytes(bytearray(buf[i] for i in range(ln.value)))
+    L.rw_spill_free.argtypes = [C.c_void_p]
+    L.rw_spill_free(C.cast(buf, C.c_void_p))
+    return out
+
+
+def agg_restore(lib, h, buf):
+    """Rebuild agg state from concatenated drain bytes (rw_stream.h)."""
+    L = lib.lib
+    L.rw_hash_agg_restore.restype = C.c_int
+    L.rw_hash_agg_restore.argtypes = [C.c_void_p, C.c_char_p, C.c_uint64]
+    rc = L.rw_hash_agg_restore(h, buf, len(buf))
+    if rc != 0:
+        raise RuntimeError(f"agg restore failed {rc}: {lib.last_error()}")
+
+
+def join_restore(lib, h, side, buf, deg_buf=b""):
+    """Rebuild one join side from concatenated drain bytes (rw_stream.h)."""
+    L = lib.lib
+    L.rw_hash_join_restore.restype = C.c_int
+    L.rw_hash_join_restore.argtypes = [C.c_void_p, C.c_int, C.c_char_p,
+                                       C.c_uint64, C.c_char_p, C.c_uint64]
+    rc = L.rw_hash_join_restore(h, side, buf, len(buf),
+                                deg_buf if deg_buf else None,
+                                len(deg_buf) if deg_buf else 0)
+    if rc != 0:
+        raise RuntimeError(f"join restore failed {rc}: {lib.last_error()}")

@@ -1,0 +1,213 @@
+"""State restore / crash-recovery parity (rw_stream.h
+restore contract). The reference rebuilds executor state from its state
+tables on recovery (src/meta/src/barrier/worker.rs:1074) and on cache miss
+(join/hash_join.rs:232-260); here the drained §8f-2 spill records are the
+state store's view, and rw_hash_{agg,join}_restore must rebuild a fresh
+executor so that continuing produces EXACTLY what an uninterrupted run
+produces — same emissions, same subsequent drain bytes."""
+import numpy as np
+import pytest
+
+from rwtest import ffi
+from rwtest.ffi import (AGG_COUNT_STAR, AGG_MAX, AGG_SUM, JOIN_INNER,
+                        JOIN_LEFT_SEMI, SIDE_LEFT, SIDE_RIGHT, T_I64,
+                        agg_checkpoint_drain_bytes, agg_restore,
+                        join_checkpoint_drain, join_degree_drain,
+                        join_restore, oracle, rows_multiset)
+
+
+def mk_chunk(types, ops, cols, valids=None):
+    cols = [np.asarray(c, np.int64) for c in cols]
+    n = len(cols[0])
+    if valids is None:
+        valids = [np.ones(n, np.uint8) for _ in cols]
+    return ffi.Chunk(types, np.asarray(ops, np.uint8), cols, valids)
+
+
+def _agg_epoch_chunks(rng, epoch, live):
+    """Insert/retract mix over a bounded group space; `live` tracks rows
+    (group, val) inserted so retracts target real state."""
+    chunks = []
+    for _ in range(3):
+        n = 512
+        g = rng.integers(0, 40, n)
+        v = rng.integers(1, 1000, n)
+        ops = np.zeros(n, np.uint8)
+        for r in range(n):
+            if live and rng.random() < 0.25:
+                j = int(rng.integers(0, len(live)))
+                g[r], v[r] = live.pop(j)
+                ops[r] = ffi.OP_DELETE
+            else:
+                live.append((int(g[r]), int(v[r])))
+        chunks.append(mk_chunk([T_I64, T_I64], ops, [g, v]))
+    return chunks
+
+
+def _drive_agg(a, chunks, epoch):
+    for c in chunks:
+        a.push(c)
+    a.flush(epoch)
+    return rows_multiset(a.poll_all())
+
+
+def _agg_restore_flow(lib):
+    calls = [(AGG_SUM, 1, T_I64), (AGG_COUNT_STAR, -1, T_I64)]
+    mk = lambda: ffi.HashAgg(lib, [T_I64, T_I64], [0], calls, 1)
+    a = mk()
+    rng = np.random.default_rng(1234)
+    live = []
+    drains = []
+    # 3 pre-crash epochs
+    pre_inputs = [_agg_epoch_chunks(rng, e, live) for e in range(3)]
+    for e, chunks in enumerate(pre_inputs):
+        _drive_agg(a, chunks, e + 1)
+        drains.append(agg_checkpoint_drain_bytes(lib, a.h))
+    # post-crash epochs, same for both executors
+    post_inputs = [_agg_epoch_chunks(rng, 3 + e, live) for e in range(3)]
+    b = mk()
+    agg_restore(lib, b.h, b"".join(drains))
+    outs_a, outs_b, dr_a, dr_b = [], [], [], []
+    for e, chunks in enumerate(post_inputs):
+        outs_a.append(_drive_agg(a, chunks, 4 + e))
+        outs_b.append(_drive_agg(b, chunks, 4 + e))
+        dr_a.append(agg_checkpoint_drain_bytes(lib, a.h))
+        dr_b.append(agg_checkpoint_drain_bytes(lib, b.h))
+    a.close()
+    b.close()
+    assert outs_a == outs_b, "restored agg diverged from uninterrupted run"
+    assert dr_a == dr_b, "restored agg spill stream diverged"
+
+
+def test_agg_restore_oracle():
+    _agg_restore_flow(oracle())
+
+
+@pytest.mark.gpu
+def test_agg_restore_gpu():
+    import risingwave_amd
+
+    risingwave_amd.load_library()
+    _agg_restore_flow(ffi.Lib(risingwave_amd.lib_path()))
+
+
+def _join_epoch_pushes(rng, live, pk_counter):
+    pushes = []
+    for _ in range(2):
+        for side in (SIDE_LEFT, SIDE_RIGHT):
+            n = 384
+            k = rng.integers(0, 60, n)
+            v = np.arange(pk_counter[0], pk_counter[0] + n)
+            pk_counter[0] += n
+            ops = np.zeros(n, np.uint8)
+            for r in range(n):
+                if live[side] and rng.random() < 0.25:
+                    j = int(rng.integers(0, len(live[side])))
+                    k[r], v[r] = live[side].pop(j)
+                    ops[r] = ffi.OP_DELETE
+                else:
+                    live[side].append((int(k[r]), int(v[r])))
+            pushes.append((side, mk_chunk([T_I64, T_I64], ops, [k, v])))
+    return pushes
+
+
+def _join_restore_flow(lib, join_type):
+    mk = lambda: ffi.HashJoin(lib, join_type, [T_I64, T_I64],
+                              [T_I64, T_I64], key_l=[0], key_r=[0],
+                              pk_l=[1], pk_r=[1])
+    a = mk()
+    rng = np.random.default_rng(99)
+    live = {SIDE_LEFT: [], SIDE_RIGHT: []}
+    pk_counter = [0]
+    state = {s: b"" for s in (SIDE_LEFT, SIDE_RIGHT)}
+    degs = {s: b"" for s in (SIDE_LEFT, SIDE_RIGHT)}
+    pre = [_join_epoch_pushes(rng, live, pk_counter) for _ in range(3)]
+    for e, pushes in enumerate(pre):
+        for side, c in pushes:
+            a.push(side, c)
+            a.poll_all()
+        for s in (SIDE_LEFT, SIDE_RIGHT):
+            state[s] += join_checkpoint_drain(lib, a.h, s)
+            degs[s] += join_degree_drain(lib, a.h, s)
+    post = [_join_epoch_pushes(rng, live, pk_counter) for _ in range(3)]
+    b = mk()
+    for s in (SIDE_LEFT, SIDE_RIGHT):
+        join_restore(lib, b.h, s, state[s], degs[s])
+    for e, pushes in enumerate(post):
+        for side, c in pushes:
+            a.push(side, c)
+            b.push(side, c)
+            ma = rows_multiset(a.poll_all())
+            mb = rows_multiset(b.poll_all())
+            assert ma == mb, (f"epoch {e}: restored join diverged "
+                              f"({len(ma)} vs {len(mb)} rows)")
+        for s in (SIDE_LEFT, SIDE_RIGHT):
+            da = join_checkpoint_drain(lib, a.h, s)
+            db = join_checkpoint_drain(lib, b.h, s)
+            assert da == db, f"epoch {e} side {s}: state drain diverged"
+            ga = join_degree_drain(lib, a.h, s)
+            gb = join_degree_drain(lib, b.h, s)
+            assert ga == gb, f"epoch {e} side {s}: degree drain diverged"
+    a.close()
+    b.close()
+
+
+def test_join_restore_oracle_inner():
+    _join_restore_flow(oracle(), JOIN_INNER)
+
+
+def test_join_restore_oracle_semi_degrees():
+    _join_restore_flow(oracle(), JOIN_LEFT_SEMI)
+
+
+@pytest.mark.gpu
+def test_join_restore_gpu_inner():
+    import risingwave_amd
+
+    risingwave_amd.load_library()
+    _join_restore_flow(ffi.Lib(risingwave_amd.lib_path()), JOIN_INNER)
+
+
+@pytest.mark.gpu
+def test_join_restore_gpu_semi_degrees():
+    import risingwave_amd
+
+    risingwave_amd.load_library()
+    _join_restore_flow(ffi.Lib(risingwave_amd.lib_path()), JOIN_LEFT_SEMI)
+
+
+@pytest.mark.gpu
+def test_restore_gpu_matches_oracle():
+    # cross-check: a GPU executor restored from GPU drains continues in
+    # lockstep with an ORACLE restored from ORACLE drains
+    import risingwave_amd
+
+    risingwave_amd.load_library()
+    glib = ffi.Lib(risingwave_amd.lib_path())
+    calls = [(AGG_MAX, 1, T_I64), (AGG_COUNT_STAR, -1, T_I64)]
+    rng = np.random.default_rng(7)
+    execs = {}
+    drains = {}
+    for name, lib in (("gpu", glib), ("orc", oracle())):
+        a = ffi.HashAgg(lib, [T_I64, T_I64], [0], calls, 1)
+        r = np.random.default_rng(7)
+        live = []
+        d = b""
+        for e in range(3):
+            _drive_agg(a, _agg_epoch_chunks(r, e, live), e + 1)
+            d += agg_checkpoint_drain_bytes(lib, a.h)
+        a.close()
+        b = ffi.HashAgg(lib, [T_I64, T_I64], [0], calls, 1)
+        agg_restore(lib, b.h, d)
+        execs[name] = (lib, b)
+        drains[name] = d
+    assert drains["gpu"] == drains["orc"], "pre-crash drains diverged"
+    r1 = np.random.default_rng(8)
+    r2 = np.random.default_rng(8)
+    live1, live2 = [], []
+    for e in range(3):
+        og = _drive_agg(execs["gpu"][1], _agg_epoch_chunks(r1, e, live1), e)
+        oo = _drive_agg(execs["orc"][1], _agg_epoch_chunks(r2, e, live2), e)
+        assert og == oo, f"epoch {e}: restored GPU != restored oracle"
+    for lib, b in execs.values():
+        b.close()
