@@ -7,6 +7,8 @@ test_hash_agg_min_append_only :177-256). The reference snapshots are taken
 with sort_chunk(true) (snapshot.rs:29-40), so comparison is the per-epoch
 row multiset — same bar as upstream.
 """
+import numpy as np
+
 from rwtest import ffi
 from rwtest.ffi import (
     AGG_COUNT_STAR, AGG_MIN, AGG_SUM, T_I64, from_pretty, oracle, rows_multiset,
@@ -409,3 +411,42 @@ def test_minput_grouped():
     assert rows_multiset(agg.poll_all()) == expect(
         [("U-", (8, 2, 5)), ("U+", (8, 3, 8))])
     agg.close()
+
+
+def test_eowc_reference_fixture():
+    # VERBATIM transcription of the reference's emit-on-window-close
+    # integration test (src/stream/tests/integration_tests/hash_agg.rs:
+    # 258-400, test_hash_agg_emit_on_window_close): count() grouped by the
+    # window column (input col 1), EOWC on. The reference's col 0 is an
+    # all-NULL Varchar ("to ensure correct group key column mapping");
+    # varchar is out of the GPU type set, so the stand-in is an all-NULL
+    # i64 column — it is never read.
+    o = ffi.HashAgg(oracle(), [T_I64, T_I64], [1],
+                    [(AGG_COUNT_STAR, -1, T_I64)], 0,
+                    emit_on_window_close=True)
+
+    def push(tokens):
+        n = len(tokens)
+        ops = np.array([ffi.OP_BY_TOKEN[t[0]] for t in tokens], np.uint8)
+        w = np.array([t[1] for t in tokens], np.int64)
+        o.push(ffi.Chunk([T_I64, T_I64], ops,
+                         [np.zeros(n, np.int64), w],
+                         [np.zeros(n, np.uint8), np.ones(n, np.uint8)]))
+
+    def flush_rows(epoch):
+        o.flush(epoch)
+        return rows_multiset(o.poll_all())
+
+    assert flush_rows(1) == []                      # barrier 1
+    push([("+", 1), ("+", 2), ("+", 3)])
+    assert flush_rows(2) == []                      # barrier 2: no watermark
+    push([("-", 2), ("+", 4)])
+    o.watermark(0, 3)
+    assert flush_rows(3) == [("+", (1, 1))]         # closes 1 (and empty 2)
+    o.watermark(0, 4)
+    assert flush_rows(4) == [("+", (3, 1))]
+    o.watermark(0, 10)
+    assert flush_rows(5) == [("+", (4, 1))]
+    o.watermark(0, 20)
+    assert flush_rows(6) == []
+    o.close()

@@ -427,3 +427,22 @@ def test_right_semi_join_append_only():
     assert push(j, SIDE_RIGHT, " I I I\n + 1 4 4\n + 3 6 5") == rows(
         [("+", 1, 4, 4), ("+", 3, 6, 5)])
     j.close()
+
+
+def test_watermark_reference_fixture():
+    # VERBATIM transcription of test_streaming_hash_join_watermark
+    # (src/stream/src/executor/hash_join.rs:3667-3737): watermarks on the
+    # join-key column of both sides; the executor emits the advanced MIN
+    # for the update side's concat column first, then the match side's
+    # (hash_join.rs:852-866). Inner Key64 join, state cleaning enabled
+    # (`vec![(0, true)]` in the fixture).
+    from rwtest.ffi import JOIN_INNER, SIDE_LEFT, SIDE_RIGHT
+
+    o = ffi.HashJoin(oracle(), JOIN_INNER, [T_I64, T_I64], [T_I64, T_I64],
+                     key_l=[0], key_r=[0], pk_l=[1], pk_r=[1],
+                     wm_jk=[(0, True)])
+    assert o.watermark(SIDE_LEFT, 0, 100) == []
+    assert o.watermark(SIDE_LEFT, 0, 200) == []
+    assert o.watermark(SIDE_RIGHT, 0, 50) == [(2, 50), (0, 50)]
+    assert o.watermark(SIDE_RIGHT, 0, 100) == [(2, 100), (0, 100)]
+    o.close()
