@@ -98,6 +98,14 @@ typedef struct RwHashAggDesc {
 void* rw_hash_agg_create(const RwHashAggDesc* desc);
 int rw_hash_agg_push_chunk(void* h, const RwChunk* chunk);
 int rw_hash_agg_flush(void* h, uint64_t epoch); /* barrier */
+/* Epoch-batched ingestion: with mode 1, push_chunk STAGES chunks into a
+ * device-resident epoch buffer (no apply launch) and the barrier flush
+ * applies the whole epoch as ONE kernel launch — the engine's
+ * one-launch-per-epoch execution shape, reachable from
+ * the reference-side binding. Legal for order-free value
+ * states only; rejects materialized-input / DISTINCT aggregates and
+ * chunks with visibility bitmaps. */
+int rw_hash_agg_ingest_mode(void* h, int epoch_batched);
 RwChunk* rw_hash_agg_poll(void* h);
 void rw_hash_agg_destroy(void* h);
 

@@ -1266,3 +1266,47 @@ def test_q7_pipeline_parity():
                           f"oracle {len(mo)}")
     for x in (ga, oa, gj, oj):
         x.close()
+
+
+def test_epoch_ingest_mode_parity():
+    # product-reachable epoch-batched ingestion (rw_hash_agg_ingest_mode):
+    # staged pushes + one apply launch at the barrier must equal the
+    # per-push apply path AND the oracle on the same inputs
+    import ctypes
+
+    from rwtest.ffi import AGG_COUNT_STAR, AGG_MAX, AGG_SUM
+
+    rng = np.random.default_rng(31)
+    calls = [(AGG_MAX, 1, T_I64), (AGG_SUM, 1, T_I64),
+             (AGG_COUNT_STAR, -1, T_I64)]
+    a_epoch = ffi.HashAgg(gpu(), [T_I64, T_I64], [0], calls, 2)
+    a_plain = ffi.HashAgg(gpu(), [T_I64, T_I64], [0], calls, 2)
+    a_orc = ffi.HashAgg(ffi.oracle(), [T_I64, T_I64], [0], calls, 2)
+    L = gpu().lib
+    L.rw_hash_agg_ingest_mode.restype = ctypes.c_int
+    L.rw_hash_agg_ingest_mode.argtypes = [ctypes.c_void_p, ctypes.c_int]
+    assert L.rw_hash_agg_ingest_mode(a_epoch.h, 1) == 0, gpu().last_error()
+    live = []
+    for epoch in range(4):
+        for _ in range(5):
+            n = 2048
+            g = rng.integers(0, 100, n)
+            v = rng.integers(1, 10_000, n)
+            ops = np.zeros(n, np.uint8)
+            for r in range(n):
+                if live and rng.random() < 0.2:
+                    j = int(rng.integers(0, len(live)))
+                    g[r], v[r] = live.pop(j)
+                    ops[r] = ffi.OP_DELETE
+                else:
+                    live.append((int(g[r]), int(v[r])))
+            c = mk_chunk([T_I64, T_I64], ops, [g, v])
+            for a in (a_epoch, a_plain, a_orc):
+                a.push(c)
+        outs = []
+        for a in (a_epoch, a_plain, a_orc):
+            a.flush(epoch + 1)
+            outs.append(rows_multiset(a.poll_all()))
+        assert outs[0] == outs[1] == outs[2], f"epoch {epoch} diverged"
+    for a in (a_epoch, a_plain, a_orc):
+        a.close()
