@@ -35,6 +35,11 @@ enum RwTypeId {
     RW_T_F32 = 3,
     RW_T_BOOL = 4,
     RW_T_TS = 5,
+    /* 16-byte decimal in rust_decimal 1.40.0's serialize layout (the
+     * reference's Decimal::unordered_serialize, types/decimal.rs:583-592):
+     * u32 flags (scale in bits 16-23, sign in bit 31; byte 0 = 1/2/3 for
+     * NaN/+Inf/-Inf specials) ++ u32 lo ++ u32 mid ++ u32 hi, all LE. */
+    RW_T_DECIMAL = 6,
 };
 
 typedef struct RwColumn {
@@ -59,6 +64,7 @@ static inline uint32_t rw_type_size(uint8_t t) {
         case RW_T_F32: return 4;
         case RW_T_BOOL: return 1;
         case RW_T_TS: return 8;
+        case RW_T_DECIMAL: return 16;
         default: return 0;
     }
 }

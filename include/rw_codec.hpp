@@ -35,6 +35,7 @@ struct DatumC {
     bool null;
     int64_t i;
     double d;
+    int64_t i2 = 0; // decimal datums: high half of the 16-byte image
 };
 
 inline void put_be(std::vector<uint8_t>& buf, uint64_t v, int n) {
@@ -92,6 +93,14 @@ inline void value_encode_datum(std::vector<uint8_t>& buf, uint8_t type,
     }
     buf.push_back(1);
     switch (type) {
+        case RW_T_DECIMAL: {
+            // serialize_decimal (value_encoding/mod.rs:344-350): the raw
+            // 16-byte unordered_serialize image
+            uint64_t a = (uint64_t)dat.i, b = (uint64_t)dat.i2;
+            for (int k = 0; k < 8; k++) buf.push_back((uint8_t)(a >> (8 * k)));
+            for (int k = 0; k < 8; k++) buf.push_back((uint8_t)(b >> (8 * k)));
+            break;
+        }
         case RW_T_I64:
         case RW_T_TS: {
             uint64_t v = (uint64_t)dat.i;
@@ -139,6 +148,14 @@ inline size_t value_decode_datum(const uint8_t* p, size_t avail, uint8_t type,
         return v;
     };
     switch (type) {
+        case RW_T_DECIMAL: {
+            if (avail < 17) return 0;
+            uint64_t a = 0, b = 0;
+            for (int k = 0; k < 8; k++) a |= (uint64_t)p[1 + k] << (8 * k);
+            for (int k = 0; k < 8; k++) b |= (uint64_t)p[9 + k] << (8 * k);
+            *out = {false, (int64_t)a, 0, (int64_t)b};
+            return 17;
+        }
         case RW_T_I64:
         case RW_T_TS:
             if (avail < 9) return 0;

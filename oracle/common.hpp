@@ -47,12 +47,183 @@ struct Datum {
         x.d = v;
         return x;
     }
+    // decimal datums carry the 16-byte rust_decimal serialize image as two
+    // LE halves: i = bytes 0..8 (flags ++ lo), i2 = bytes 8..16 (mid ++ hi)
+    int64_t i2 = 0;
+    static Datum of_dec(int64_t lohalf, int64_t hihalf) {
+        Datum x;
+        x.null = false;
+        x.i = lohalf;
+        x.i2 = hihalf;
+        return x;
+    }
 };
+
+// ---- decimal arithmetic restatement (rust_decimal 1.40.0, a Cargo.lock
+// dependency not vendored under /root/reference; the reference wraps it as
+// Decimal{NegativeInf, Normalized, PositiveInf, NaN}, types/decimal.rs:36-44,
+// with sum via checked_add, expr general.rs:23). The EXACT domain is
+// restated: a decimal is sign x mantissa96 x 10^-scale (scale 0..28);
+// addition aligns scales and adds exactly. Sums are accumulated as exact
+// signed 256-bit integers at scale 28 (each addend = mantissa x
+// 10^(28-scale), <= 10^56 << 2^191; 2^64 such addends cannot overflow
+// i256), so the result is the exact rational sum, converted back at the
+// group's max input scale. If the exact result's mantissa exceeds 96 bits
+// the reference's sequential rust_decimal add would have entered its
+// order-dependent precision-loss rescale path — that domain raises loudly
+// instead of silently diverging. Specials follow the
+// reference's Add table (decimal.rs:259-276): any NaN -> NaN, +Inf + -Inf
+// -> NaN, else Inf dominates; they are counted exactly so retraction
+// works. ----
+
+struct DecVal {
+    int special = 0; // 0 normal, 1 NaN, 2 +Inf, 3 -Inf (serialize byte 0)
+    bool neg = false;
+    uint32_t scale = 0;
+    uint64_t lo = 0;  // mantissa low 64
+    uint32_t hi = 0;  // mantissa high 32
+};
+
+inline DecVal dec_parse(int64_t lohalf, int64_t hihalf) {
+    uint8_t b[16];
+    std::memcpy(b, &lohalf, 8);
+    std::memcpy(b + 8, &hihalf, 8);
+    DecVal v;
+    if (b[0] == 1 || b[0] == 2 || b[0] == 3) {
+        v.special = b[0];
+        return v;
+    }
+    uint32_t flags, lo32, mid, hi;
+    std::memcpy(&flags, b, 4);
+    std::memcpy(&lo32, b + 4, 4);
+    std::memcpy(&mid, b + 8, 4);
+    std::memcpy(&hi, b + 12, 4);
+    v.neg = (flags >> 31) & 1;
+    v.scale = (flags >> 16) & 0xFF;
+    v.lo = (uint64_t)mid << 32 | lo32;
+    v.hi = hi;
+    return v;
+}
+
+inline void dec_serialize(const DecVal& v, int64_t* lohalf, int64_t* hihalf) {
+    uint8_t b[16] = {};
+    if (v.special) {
+        b[0] = (uint8_t)v.special;
+    } else {
+        uint32_t flags = (v.scale << 16) | ((uint32_t)v.neg << 31);
+        uint32_t lo32 = (uint32_t)v.lo, mid = (uint32_t)(v.lo >> 32);
+        std::memcpy(b, &flags, 4);
+        std::memcpy(b + 4, &lo32, 4);
+        std::memcpy(b + 8, &mid, 4);
+        std::memcpy(b + 12, &v.hi, 4);
+    }
+    std::memcpy(lohalf, b, 8);
+    std::memcpy(hihalf, b + 8, 8);
+}
+
+// exact signed-256-bit accumulator (two's complement over 4 u64 words)
+struct I256 {
+    uint64_t w[4] = {0, 0, 0, 0};
+    void add(const I256& o) {
+        unsigned __int128 c = 0;
+        for (int k = 0; k < 4; k++) {
+            unsigned __int128 t = (unsigned __int128)w[k] + o.w[k] + c;
+            w[k] = (uint64_t)t;
+            c = t >> 64;
+        }
+    }
+    I256 negated() const {
+        I256 r;
+        unsigned __int128 c = 1;
+        for (int k = 0; k < 4; k++) {
+            unsigned __int128 t = (unsigned __int128)(~w[k]) + c;
+            r.w[k] = (uint64_t)t;
+            c = t >> 64;
+        }
+        return r;
+    }
+    bool is_neg() const { return w[3] >> 63; }
+    bool is_zero() const { return !(w[0] | w[1] | w[2] | w[3]); }
+};
+
+inline uint64_t dec_pow10_u64(int k) { // k <= 19
+    uint64_t p = 1;
+    while (k-- > 0) p *= 10;
+    return p;
+}
+
+// addend = sign * mantissa96 * 10^(28 - scale) as I256
+inline I256 dec_addend(const DecVal& v) {
+    // mantissa96 * 10^k with k = 28-scale, split as 10^a * 10^b (a,b <= 14)
+    int k = 28 - (int)v.scale;
+    uint64_t pa = dec_pow10_u64(k / 2), pb = dec_pow10_u64(k - k / 2);
+    // m = hi*2^64 + lo; multiply by pa then pb using 128-bit partials
+    uint64_t m[4] = {v.lo, v.hi, 0, 0};
+    for (uint64_t p : {pa, pb}) {
+        unsigned __int128 c = 0;
+        for (int i = 0; i < 4; i++) {
+            unsigned __int128 t = (unsigned __int128)m[i] * p + c;
+            m[i] = (uint64_t)t;
+            c = t >> 64;
+        }
+    }
+    I256 r;
+    for (int i = 0; i < 4; i++) r.w[i] = m[i];
+    if (v.neg) r = r.negated();
+    return r;
+}
+
+// exact divide of |S| by 10^(28 - out_scale); returns false if the result
+// mantissa exceeds 96 bits (the reference's precision-loss domain)
+inline bool dec_from_sum(const I256& S, uint32_t out_scale, DecVal* out) {
+    I256 a = S;
+    bool neg = a.is_neg();
+    if (neg) a = a.negated();
+    int k = 28 - (int)out_scale;
+    // long division by 10^k in two <=10^14 chunks (the sum is a multiple of
+    // 10^k by construction, so remainders are exactly 0)
+    for (uint64_t p : {dec_pow10_u64(k / 2), dec_pow10_u64(k - k / 2)}) {
+        if (p == 1) continue;
+        unsigned __int128 rem = 0;
+        for (int i = 3; i >= 0; i--) {
+            unsigned __int128 cur = (rem << 64) | a.w[i];
+            a.w[i] = (uint64_t)(cur / p);
+            rem = cur % p;
+        }
+        if (rem != 0) return false; // not exact: internal invariant breach
+    }
+    if (a.w[3] || a.w[2] || (a.w[1] >> 32)) return false; // > 96 bits
+    out->special = 0;
+    out->neg = neg && !a.is_zero();
+    out->scale = out_scale;
+    out->lo = a.w[0];
+    out->hi = (uint32_t)a.w[1];
+    return true;
+}
+
+// value equality (rust_decimal PartialEq compares VALUES: 1.2 == 1.20)
+inline bool dec_value_eq(const DecVal& a, const DecVal& b) {
+    if (a.special || b.special) return a.special == b.special;
+    I256 x = dec_addend(a), y = dec_addend(b);
+    return x.w[0] == y.w[0] && x.w[1] == y.w[1] && x.w[2] == y.w[2] &&
+           x.w[3] == y.w[3];
+}
 
 inline bool type_is_float(uint8_t t) { return t == RW_T_F64 || t == RW_T_F32; }
 
 // ScalarImpl PartialEq: floats are OrderedFloat (NaN == NaN).
+inline bool datum_eq(const Datum& a, const Datum& b, uint8_t t);
+
+inline bool datum_eq_impl_decimal(const Datum& a, const Datum& b) {
+    return dec_value_eq(dec_parse(a.i, a.i2), dec_parse(b.i, b.i2));
+}
+
 inline bool datum_eq(const Datum& a, const Datum& b, uint8_t t) {
+    if (t == RW_T_DECIMAL) {
+        if (a.null != b.null) return false;
+        if (a.null) return true;
+        return datum_eq_impl_decimal(a, b);
+    }
     if (a.null || b.null) return a.null == b.null;
     if (type_is_float(t)) {
         return a.d == b.d || (std::isnan(a.d) && std::isnan(b.d));
@@ -138,6 +309,10 @@ struct ChunkView {
         switch (cc.type) {
             case RW_T_I64:
             case RW_T_TS: return Datum::of_i(((const int64_t*)cc.data)[r]);
+            case RW_T_DECIMAL: {
+                const int64_t* p = (const int64_t*)cc.data;
+                return Datum::of_dec(p[2 * r], p[2 * r + 1]);
+            }
             case RW_T_I32: return Datum::of_i(((const int32_t*)cc.data)[r]);
             case RW_T_BOOL: return Datum::of_i(((const uint8_t*)cc.data)[r]);
             case RW_T_F64: return Datum::of_d(((const double*)cc.data)[r]);
@@ -336,6 +511,16 @@ inline RwChunk* chunk_to_c(const OwnedChunk& oc) {
                 data = p;
                 break;
             }
+            case RW_T_DECIMAL: {
+                auto* p = new int64_t[2 * n];
+                for (size_t r = 0; r < n; r++) {
+                    valid[r] = !oc.cols[c][r].null;
+                    p[2 * r] = valid[r] ? oc.cols[c][r].i : 0;
+                    p[2 * r + 1] = valid[r] ? oc.cols[c][r].i2 : 0;
+                }
+                data = p;
+                break;
+            }
             case RW_T_F64: {
                 auto* p = new double[n];
                 for (size_t r = 0; r < n; r++) {
@@ -374,6 +559,7 @@ inline void chunk_free_c(RwChunk* ch) {
         switch (ch->cols[c].type) {
             case RW_T_I64:
             case RW_T_TS: delete[] (int64_t*)ch->cols[c].data; break;
+            case RW_T_DECIMAL: delete[] (int64_t*)ch->cols[c].data; break;
             case RW_T_I32: delete[] (int32_t*)ch->cols[c].data; break;
             case RW_T_BOOL: delete[] (uint8_t*)ch->cols[c].data; break;
             case RW_T_F64: delete[] (double*)ch->cols[c].data; break;

@@ -56,8 +56,16 @@ struct MInputState {
     explicit MInputState(RowOrderLess less) : entries(std::move(less)) {}
 };
 
+// exact decimal-sum accumulator (see common.hpp's decimal restatement)
+struct DecSumState {
+    I256 sum;
+    uint32_t maxscale = 0;
+    int64_t nonnull = 0, nan = 0, pinf = 0, ninf = 0;
+};
+
 struct AggGroupState {
     std::vector<ValueState> vstates;
+    std::vector<DecSumState> dstates; // per call; used for decimal sums
     std::vector<MInputState> mstates; // indexed per call; unused for value calls
     bool has_prev = false;
     Row prev_outputs; // agg outputs only (no group key)
@@ -118,6 +126,21 @@ struct HashAggOracle {
         chunk_size = d->chunk_size;
         append_only = d->append_only;
         eowc = d->emit_on_window_close != 0;
+        // decimal scope this round: sum/count ARGUMENTS only
+        for (auto k : group_key)
+            if (input_types[k] == RW_T_DECIMAL)
+                throw std::invalid_argument("decimal group keys unsupported");
+        for (auto k : stream_key)
+            if (input_types[k] == RW_T_DECIMAL)
+                throw std::invalid_argument("decimal stream keys unsupported");
+        for (auto& c : calls)
+            if (c.arg >= 0 && input_types[c.arg] == RW_T_DECIMAL &&
+                (c.distinct ||
+                 !(c.kind == RW_AGG_SUM || c.kind == RW_AGG_SUM0 ||
+                   c.kind == RW_AGG_COUNT)))
+                throw std::invalid_argument(
+                    "decimal supported as non-DISTINCT sum/count "
+                    "arguments only");
         for (auto k : group_key) group_key_types.push_back(input_types[k]);
         out_types = out_ts;
         for (auto& c : calls) {
@@ -166,6 +189,12 @@ struct HashAggOracle {
         return less;
     }
 
+    bool call_is_decimal(size_t ci) const {
+        return !call_is_minput[ci] && calls[ci].arg >= 0 &&
+               (size_t)calls[ci].arg < input_types.size() &&
+               input_types[calls[ci].arg] == RW_T_DECIMAL;
+    }
+
     AggGroupState& touch(const Row& key) {
         auto it = groups.find(key);
         if (it == groups.end()) {
@@ -180,6 +209,7 @@ struct HashAggOracle {
                 g.vstates.push_back(v);
                 g.mstates.emplace_back(minput_order(c));
             }
+            g.dstates.resize(calls.size());
             it = groups.emplace(key, std::move(g)).first;
         }
         if (!dirty.count(key)) {
@@ -222,6 +252,25 @@ struct HashAggOracle {
                 case RW_AGG_SUM0: {
                     Datum a = cv.at(r, c.arg);
                     if (a.null) break;
+                    if (input_types[c.arg] == RW_T_DECIMAL) {
+                        // exact-domain decimal sum (common.hpp restatement)
+                        DecSumState& ds = g.dstates[ci];
+                        DecVal dv = dec_parse(a.i, a.i2);
+                        int64_t delta = retract ? -1 : 1;
+                        ds.nonnull += delta;
+                        if (dv.special == 1) ds.nan += delta;
+                        else if (dv.special == 2) ds.pinf += delta;
+                        else if (dv.special == 3) ds.ninf += delta;
+                        else {
+                            I256 add = dec_addend(dv);
+                            if (retract) add = add.negated();
+                            ds.sum.add(add);
+                            if (dv.scale > ds.maxscale)
+                                ds.maxscale = dv.scale;
+                        }
+                        v.has = true;
+                        break;
+                    }
                     if (v.is_float) {
                         double x = type_is_float(input_types[c.arg]) ? a.d : (double)a.i;
                         v.d = retract ? v.d - x : v.d + x;
@@ -362,15 +411,30 @@ struct HashAggOracle {
                 if (!n) FAIL(RW_E_INVAL, "restore: bad state datum");
                 off += n;
                 outs[ci] = d.null ? Datum()
-                                  : (type_is_float(calls[ci].ret_type)
-                                         ? Datum::of_d(d.d)
-                                         : Datum::of_i(d.i));
+                                  : (calls[ci].ret_type == RW_T_DECIMAL
+                                         ? Datum::of_dec(d.i, d.i2)
+                                         : (type_is_float(calls[ci].ret_type)
+                                                ? Datum::of_d(d.d)
+                                                : Datum::of_i(d.i)));
             }
             AggGroupState& g = touch(key);
             for (size_t ci = 0; ci < calls.size(); ci++) {
                 ValueState& v2 = g.vstates[ci];
                 const Datum& o = outs[ci];
                 v2.has = !o.null;
+                if (!o.null && call_is_decimal(ci)) {
+                    DecSumState& ds = g.dstates[ci];
+                    ds = DecSumState{};
+                    DecVal dv = dec_parse(o.i, o.i2);
+                    if (dv.special == 1) ds.nan = 1;
+                    else if (dv.special == 2) ds.pinf = 1;
+                    else if (dv.special == 3) ds.ninf = 1;
+                    else {
+                        ds.sum = dec_addend(dv);
+                        ds.maxscale = dv.scale;
+                    }
+                    continue;
+                }
                 if (!o.null) {
                     if (v2.is_float) v2.d = o.d;
                     else v2.i = o.i;
@@ -407,6 +471,7 @@ struct HashAggOracle {
                     v.has = true;
                 v.is_float = g.vstates[ci].is_float;
                 g.vstates[ci] = v;
+                g.dstates[ci] = DecSumState{};
             }
         }
         Row out(calls.size());
@@ -423,6 +488,30 @@ struct HashAggOracle {
                 case RW_AGG_COUNT:
                 case RW_AGG_SUM0: out[ci] = Datum::of_i(v.i); break;
                 case RW_AGG_SUM:
+                    if (call_is_decimal(ci)) {
+                        if (!v.has) {
+                            out[ci] = Datum();
+                            break;
+                        }
+                        const DecSumState& ds = g.dstates[ci];
+                        DecVal r2;
+                        if (ds.nan > 0 || (ds.pinf > 0 && ds.ninf > 0)) {
+                            r2.special = 1; // NaN (decimal.rs:259-276)
+                        } else if (ds.pinf > 0) {
+                            r2.special = 2;
+                        } else if (ds.ninf > 0) {
+                            r2.special = 3;
+                        } else if (!dec_from_sum(ds.sum, ds.maxscale, &r2)) {
+                            throw std::runtime_error(
+                                "decimal sum outside the exact 96-bit "
+                                "domain (the reference's order-dependent "
+                                "rescale path; unsupported)");
+                        }
+                        int64_t a2, b2;
+                        dec_serialize(r2, &a2, &b2);
+                        out[ci] = Datum::of_dec(a2, b2);
+                        break;
+                    }
                     out[ci] = v.has ? (v.is_float ? Datum::of_d(v.d) : Datum::of_i(v.i))
                                     : Datum();
                     break;
@@ -459,7 +548,7 @@ struct HashAggOracle {
                 } else {
                     const Datum& o = outputs[ci];
                     rwcodec::value_encode_datum(v, calls[ci].ret_type,
-                                                {o.null, o.i, o.d});
+                                                {o.null, o.i, o.d, o.i2});
                 }
             }
         }
@@ -537,7 +626,16 @@ struct HashAggOracle {
     }
 
     // flush_data, emit-on-update branch (hash_agg.rs:475-501)
-    int flush(uint64_t /*epoch*/) {
+    int flush(uint64_t epoch) {
+        try {
+            return flush_inner(epoch);
+        } catch (const std::runtime_error& e) {
+            g_err = e.what();
+            return RW_E_OVERFLOW;
+        }
+    }
+
+    int flush_inner(uint64_t /*epoch*/) {
         if (eowc) return flush_eowc();
         for (auto& key : dirty_order) {
             auto& g = groups[key];
@@ -685,7 +783,12 @@ void* rw_hash_agg_create(const RwHashAggDesc* d) {
     for (uint32_t i = 0; i < d->n_group_key; i++)
         out_types.push_back(d->input_types[d->group_key_indices[i]]);
     for (uint32_t i = 0; i < d->n_calls; i++) out_types.push_back(d->calls[i].ret_type);
-    return new HashAggOracle(d, out_types);
+    try {
+        return new HashAggOracle(d, out_types);
+    } catch (const std::invalid_argument& e) {
+        g_err = e.what();
+        return nullptr;
+    }
 }
 int rw_hash_agg_push_chunk(void* h, const RwChunk* c) {
     return ((HashAggOracle*)h)->push_chunk(c);

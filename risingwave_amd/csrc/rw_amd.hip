@@ -89,6 +89,146 @@ __device__ __forceinline__ void atomic_max_i64(long long* p, long long v) {
     atomicMax(p, v);
 }
 
+// ---- decimal (device mirror of the exact-domain restatement in
+// oracle/common.hpp; rust_decimal 1.40.0 layout — see include/rw_chunk.h
+// RW_T_DECIMAL and DESIGN.md §9). Sums accumulate as exact signed 256-bit
+// integers at scale 28 via per-word atomics with cascading carries. ----
+struct DecValD {
+    int special; // 0 normal, 1 NaN, 2 +Inf, 3 -Inf
+    bool neg;
+    uint32_t scale;
+    uint64_t lo;
+    uint32_t hi;
+};
+
+__device__ __forceinline__ DecValD dec_parse_d(int64_t a, int64_t b) {
+    DecValD v{};
+    uint32_t flags = (uint32_t)a;
+    uint8_t b0 = (uint8_t)flags;
+    if (b0 == 1 || b0 == 2 || b0 == 3) {
+        v.special = b0;
+        return v;
+    }
+    v.neg = (flags >> 31) & 1;
+    v.scale = (flags >> 16) & 0xFF;
+    uint32_t lo32 = (uint32_t)((uint64_t)a >> 32);
+    uint32_t mid = (uint32_t)b;
+    v.hi = (uint32_t)((uint64_t)b >> 32);
+    v.lo = ((uint64_t)mid << 32) | lo32;
+    return v;
+}
+
+__device__ __forceinline__ void dec_serialize_d(const DecValD& v, long long* a,
+                                                long long* b) {
+    if (v.special) {
+        *a = (long long)(uint32_t)v.special;
+        *b = 0;
+        return;
+    }
+    uint32_t flags = (v.scale << 16) | ((uint32_t)v.neg << 31);
+    *a = (long long)(((uint64_t)(uint32_t)v.lo << 32) | flags);
+    *b = (long long)(((uint64_t)v.hi << 32) | (uint32_t)(v.lo >> 32));
+}
+
+__device__ __forceinline__ uint64_t dec_pow10_d(int k) {
+    uint64_t p = 1;
+    while (k-- > 0) p *= 10;
+    return p;
+}
+
+// addend words = sign * mantissa96 * 10^(28-scale) (two's complement i256)
+__device__ __forceinline__ void dec_addend_d(const DecValD& v, uint64_t w[4]) {
+    int k = 28 - (int)v.scale;
+    uint64_t pa = dec_pow10_d(k / 2), pb = dec_pow10_d(k - k / 2);
+    w[0] = v.lo;
+    w[1] = v.hi;
+    w[2] = w[3] = 0;
+    for (int t = 0; t < 2; t++) {
+        uint64_t p = t ? pb : pa;
+        unsigned __int128 c = 0;
+        for (int i = 0; i < 4; i++) {
+            unsigned __int128 x = (unsigned __int128)w[i] * p + c;
+            w[i] = (uint64_t)x;
+            c = x >> 64;
+        }
+    }
+    if (v.neg) {
+        unsigned __int128 c = 1;
+        for (int i = 0; i < 4; i++) {
+            unsigned __int128 x = (unsigned __int128)(~w[i]) + c;
+            w[i] = (uint64_t)x;
+            c = x >> 64;
+        }
+    }
+}
+
+// per-word atomic accumulation with cascading carries (mod 2^256 two's
+// complement — negative addends just work)
+__device__ __forceinline__ void dec_i256_atomic_add(
+    unsigned long long* base, size_t cap, uint32_t slot, const uint64_t w[4]) {
+    unsigned long long carry = 0;
+    for (int k = 0; k < 4; k++) {
+        unsigned long long add = (unsigned long long)w[k] + carry;
+        carry = add < w[k] ? 1 : 0;
+        if (add) {
+            unsigned long long old = atomicAdd(&base[(size_t)k * cap + slot],
+                                               add);
+            if (old + add < add) carry += 1;
+        }
+    }
+}
+
+// exact conversion of the slot's i256 sum back to a decimal at out_scale;
+// false when the mantissa exceeds 96 bits (the reference's order-dependent
+// precision-loss rescale domain)
+__device__ __forceinline__ bool dec_from_sum_d(const unsigned long long* base,
+                                               size_t cap, uint32_t slot,
+                                               uint32_t out_scale,
+                                               DecValD* out) {
+    uint64_t a[4];
+    for (int k = 0; k < 4; k++) a[k] = base[(size_t)k * cap + slot];
+    bool neg = a[3] >> 63;
+    if (neg) {
+        unsigned __int128 c = 1;
+        for (int k = 0; k < 4; k++) {
+            unsigned __int128 x = (unsigned __int128)(~a[k]) + c;
+            a[k] = (uint64_t)x;
+            c = x >> 64;
+        }
+    }
+    int k = 28 - (int)out_scale;
+    for (int t = 0; t < 2; t++) {
+        uint64_t p = t ? dec_pow10_d(k - k / 2) : dec_pow10_d(k / 2);
+        if (p == 1) continue;
+        unsigned __int128 rem = 0;
+        for (int i = 3; i >= 0; i--) {
+            unsigned __int128 cur = (rem << 64) | a[i];
+            a[i] = (uint64_t)(cur / p);
+            rem = cur % p;
+        }
+        if (rem != 0) return false;
+    }
+    if (a[3] || a[2] || (a[1] >> 32)) return false;
+    out->special = 0;
+    out->neg = neg && (a[0] | a[1]);
+    out->scale = out_scale;
+    out->lo = a[0];
+    out->hi = (uint32_t)a[1];
+    return true;
+}
+
+// rust_decimal PartialEq compares VALUES (1.2 == 1.20)
+__device__ __forceinline__ bool dec_value_eq_d(int64_t a0, int64_t a1,
+                                               int64_t b0, int64_t b1) {
+    DecValD x = dec_parse_d(a0, a1), y = dec_parse_d(b0, b1);
+    if (x.special || y.special) return x.special == y.special;
+    uint64_t wx[4], wy[4];
+    dec_addend_d(x, wx);
+    dec_addend_d(y, wy);
+    return wx[0] == wy[0] && wx[1] == wy[1] && wx[2] == wy[2] &&
+           wx[3] == wy[3];
+}
+
 // slot states
 #define SLOT_EMPTY 0u
 #define SLOT_CLAIMED 1u
@@ -181,6 +321,8 @@ struct AggCallDev {
     int32_t arg;  // column index in the input BATCH (gk cols ∥ arg cols ∥ sk cols)
     uint8_t minput; // materialized-input state (retractable min/max)
     int8_t mord;    // ordinal among minput calls
+    uint8_t decimal; // RW_T_DECIMAL sum/count argument (16-B datums)
+    int8_t dord;     // ordinal among decimal calls
 };
 
 // Device-side batch of input rows (SoA), i64-widened values. `stride` (in
@@ -549,6 +691,13 @@ struct AggTableDev {
     uint32_t* mcursor; // single counter
     uint32_t mrow_cap;
     int n_sk;
+    // decimal-sum state (exact i256 at scale 28):
+    unsigned long long* dsum; // [n_dec][4][cap]
+    uint32_t* dscl;           // [n_dec][cap] running max input scale
+    long long* dspec;         // [n_dec][3][cap] NaN/+Inf/-Inf counts
+    long long* out_vals2;     // [out_capacity][width] decimal high halves
+    long long* prev2;         // [n_calls][cap] decimal high halves
+    int n_dec;
 };
 
 // agg_apply: HashAggExecutor::apply_chunk (hash_agg.rs:332-409) as one
@@ -725,6 +874,42 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
                 }
                 continue;
             }
+            if (c.decimal) {
+                // exact decimal sum (device restatement above): per-lane
+                // atomics, no wave pre-reduction
+                v[ci] = 0;
+                if (contributing &&
+                    (DENSE || (b.col_valid[col][rs] ^ b.valid_inverted))) {
+                    const int64_t* dp = b.col_vals[col]; // 2 words/row
+                    DecValD dv = dec_parse_d(dp[2 * rs], dp[2 * rs + 1]);
+                    size_t dcap = cap;
+                    if (dv.special) {
+                        atomic_add_i64(
+                            &t.dspec[((size_t)c.dord * 3 + (dv.special - 1)) *
+                                         dcap +
+                                     slot],
+                            sign);
+                    } else {
+                        uint64_t w[4];
+                        dec_addend_d(dv, w);
+                        if (sign < 0) {
+                            unsigned __int128 cc = 1;
+                            for (int k = 0; k < 4; k++) {
+                                unsigned __int128 x =
+                                    (unsigned __int128)(~w[k]) + cc;
+                                w[k] = (uint64_t)x;
+                                cc = x >> 64;
+                            }
+                        }
+                        dec_i256_atomic_add(
+                            t.dsum + (size_t)c.dord * 4 * dcap, dcap, slot, w);
+                        atomicMax(&t.dscl[(size_t)c.dord * dcap + slot],
+                                  dv.scale);
+                    }
+                    t.has[(size_t)ci * cap + slot] = 1;
+                }
+                continue;
+            }
             bool shown = contributing &&
                          !(b.call_hidden[ci] && b.call_hidden[ci][rs]);
             arg_valid = arg_valid && shown;
@@ -750,7 +935,7 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
             // debug fallback: per-lane atomics, no wave aggregation
             if (contributing) {
                 for (int ci = 0; ci < n_calls; ci++) {
-                    if (calls[ci].minput) continue;
+                    if (calls[ci].minput || calls[ci].decimal) continue;
                     long long* acc = t.acc + (size_t)ci * cap;
                     uint8_t* has = t.has + (size_t)ci * cap;
                     switch (calls[ci].kind) {
@@ -815,7 +1000,7 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
         }
         if (tail) {
             for (int ci = 0; ci < n_calls; ci++) {
-                if (calls[ci].minput) continue;
+                if (calls[ci].minput || calls[ci].decimal) continue;
                 long long* acc = t.acc + (size_t)ci * cap;
                 uint8_t* has = t.has + (size_t)ci * cap;
                 switch (calls[ci].kind) {
@@ -1233,11 +1418,13 @@ __global__ void agg_flush_kernel(AggTableDev t, int KW, int n_calls,
         long long rc = t.acc[(size_t)row_count_index * cap + slot];
         if (rc < 0) rc = 0; // row_count_of clamp (agg_group.rs:63-75)
         long long curr[MAX_CALLS];
+        long long curr2[MAX_CALLS];
         uint8_t curr_null[MAX_CALLS];
         for (int ci = 0; ci < n_calls; ci++) {
             const AggCallDev& c = calls[ci];
             long long* acc = t.acc + (size_t)ci * cap;
             uint8_t* has = t.has + (size_t)ci * cap;
+            curr2[ci] = 0;
             if (rc == 0) {
                 // reset value states (agg_state.rs:149-155)
                 switch (c.kind) {
@@ -1246,6 +1433,42 @@ __global__ void agg_flush_kernel(AggTableDev t, int KW, int n_calls,
                     default: acc[slot] = 0;
                 }
                 has[slot] = 0;
+                if (c.decimal) {
+                    for (int k = 0; k < 4; k++)
+                        t.dsum[((size_t)c.dord * 4 + k) * cap + slot] = 0;
+                    t.dscl[(size_t)c.dord * cap + slot] = 0;
+                    for (int k = 0; k < 3; k++)
+                        t.dspec[((size_t)c.dord * 3 + k) * cap + slot] = 0;
+                }
+            }
+            if (c.decimal) {
+                // exact decimal sum output (specials per decimal.rs:259-276)
+                if (!has[slot]) {
+                    curr[ci] = 0;
+                    curr_null[ci] = 1;
+                    continue;
+                }
+                long long nan = t.dspec[((size_t)c.dord * 3 + 0) * cap + slot];
+                long long pinf = t.dspec[((size_t)c.dord * 3 + 1) * cap + slot];
+                long long ninf = t.dspec[((size_t)c.dord * 3 + 2) * cap + slot];
+                DecValD r2{};
+                if (nan > 0 || (pinf > 0 && ninf > 0)) {
+                    r2.special = 1;
+                } else if (pinf > 0) {
+                    r2.special = 2;
+                } else if (ninf > 0) {
+                    r2.special = 3;
+                } else if (!dec_from_sum_d(
+                               t.dsum + (size_t)c.dord * 4 * cap, cap, slot,
+                               t.dscl[(size_t)c.dord * cap + slot], &r2)) {
+                    atomicExch(&t.counters[2], 5u); // outside exact domain
+                    curr[ci] = 0;
+                    curr_null[ci] = 1;
+                    continue;
+                }
+                dec_serialize_d(r2, &curr[ci], &curr2[ci]);
+                curr_null[ci] = 0;
+                continue;
             }
             if (c.minput) {
                 // output_first over the materialized rows (minput.rs:236-241):
@@ -1306,8 +1529,17 @@ __global__ void agg_flush_kernel(AggTableDev t, int KW, int n_calls,
             bool eq = true;
             for (int ci = 0; eq && ci < n_calls; ci++) {
                 uint8_t pn = t.prev_null[(size_t)ci * cap + slot];
-                eq = (pn == curr_null[ci]) &&
-                     (pn || t.prev[(size_t)ci * cap + slot] == curr[ci]);
+                if (calls[ci].decimal) {
+                    // rust_decimal Eq compares VALUES (1.2 == 1.20)
+                    eq = (pn == curr_null[ci]) &&
+                         (pn ||
+                          dec_value_eq_d(t.prev[(size_t)ci * cap + slot],
+                                         t.prev2[(size_t)ci * cap + slot],
+                                         curr[ci], curr2[ci]));
+                } else {
+                    eq = (pn == curr_null[ci]) &&
+                         (pn || t.prev[(size_t)ci * cap + slot] == curr[ci]);
+                }
             }
             change = eq ? 0 : 3;
         }
@@ -1319,7 +1551,7 @@ __global__ void agg_flush_kernel(AggTableDev t, int KW, int n_calls,
             continue;
         }
         auto write_row = [&](uint32_t orow, uint8_t op, const long long* vals,
-                             const uint8_t* nulls) {
+                             const uint8_t* nulls, const long long* vals2) {
             t.out_ops[orow] = op;
             for (int k = 0; k < KW; k++) {
                 t.out_vals[(size_t)orow * width + k] = t.keys[(size_t)slot * KW + k];
@@ -1329,21 +1561,25 @@ __global__ void agg_flush_kernel(AggTableDev t, int KW, int n_calls,
             for (int ci = 0; ci < n_calls; ci++) {
                 t.out_vals[(size_t)orow * width + KW + ci] = vals[ci];
                 t.out_nulls[(size_t)orow * width + KW + ci] = nulls[ci];
+                if (t.out_vals2)
+                    t.out_vals2[(size_t)orow * width + KW + ci] = vals2[ci];
             }
         };
         long long prevv[MAX_CALLS];
+        long long prevv2[MAX_CALLS];
         uint8_t prevn[MAX_CALLS];
         for (int ci = 0; ci < n_calls; ci++) {
             prevv[ci] = t.prev[(size_t)ci * cap + slot];
+            prevv2[ci] = t.prev2 ? t.prev2[(size_t)ci * cap + slot] : 0;
             prevn[ci] = t.prev_null[(size_t)ci * cap + slot];
         }
         if (change == 1) {
-            write_row(base, RW_OP_INSERT, curr, curr_null);
+            write_row(base, RW_OP_INSERT, curr, curr_null, curr2);
         } else if (change == 2) {
-            write_row(base, RW_OP_DELETE, prevv, prevn);
+            write_row(base, RW_OP_DELETE, prevv, prevn, prevv2);
         } else {
-            write_row(base, RW_OP_UPDATE_DELETE, prevv, prevn);
-            write_row(base + 1, RW_OP_UPDATE_INSERT, curr, curr_null);
+            write_row(base, RW_OP_UPDATE_DELETE, prevv, prevn, prevv2);
+            write_row(base + 1, RW_OP_UPDATE_INSERT, curr, curr_null, curr2);
         }
         // prev := curr (or cleared on delete) — agg_group.rs:571-607
         if (change == 2) {
@@ -1352,6 +1588,7 @@ __global__ void agg_flush_kernel(AggTableDev t, int KW, int n_calls,
             t.has_prev[slot] = 1;
             for (int ci = 0; ci < n_calls; ci++) {
                 t.prev[(size_t)ci * cap + slot] = curr[ci];
+                if (t.prev2) t.prev2[(size_t)ci * cap + slot] = curr2[ci];
                 t.prev_null[(size_t)ci * cap + slot] = curr_null[ci];
             }
         }
@@ -1476,6 +1713,8 @@ struct HashAgg {
     bool has_pending_wm = false;
     int64_t pending_wm = 0;
     int n_minput = 0;   // materialized-input (retractable min/max) calls
+    std::vector<uint8_t> call_decimal;
+    int n_dec = 0;      // decimal-sum calls (exact i256 state)
     std::vector<uint8_t> call_minput;
     std::vector<uint32_t> stream_key;
     // DISTINCT dedup state (aggregate/distinct.rs)
@@ -1499,11 +1738,15 @@ struct HashAgg {
 
     AggCallDev cd(int i) const {
         if (i < n_calls) {
-            int8_t mord = 0;
-            for (int j = 0; j < i; j++) mord += call_minput[j];
-            return AggCallDev{calls[i].kind, calls[i].arg, call_minput[i], mord};
+            int8_t mord = 0, dord = 0;
+            for (int j = 0; j < i; j++) {
+                mord += call_minput[j];
+                dord += call_decimal[j];
+            }
+            return AggCallDev{calls[i].kind, calls[i].arg, call_minput[i],
+                              mord, call_decimal[i], dord};
         }
-        return AggCallDev{0, -1, 0, 0};
+        return AggCallDev{0, -1, 0, 0, 0, 0};
     }
 
     int init(const RwHashAggDesc* d) {
@@ -1530,12 +1773,31 @@ struct HashAgg {
                           !d->append_only;
             call_minput.push_back(minput);
             n_minput += minput;
+            bool dec = false;
             if (c.arg >= 0) {
                 uint8_t ty = input_types[c.arg];
-                if (ty != RW_T_I64 && ty != RW_T_TS)
+                if (ty == RW_T_DECIMAL) {
+                    // decimal scope: non-DISTINCT sum/count
+                    // arguments, non-EOWC executors
+                    if (c.distinct ||
+                        !(c.kind == RW_AGG_SUM || c.kind == RW_AGG_SUM0 ||
+                          c.kind == RW_AGG_COUNT))
+                        FAIL(RW_E_INVAL,
+                             "decimal supported as non-DISTINCT sum/count "
+                             "arguments only");
+                    if (eowc)
+                        FAIL(RW_E_INVAL, "decimal + EOWC unsupported");
+                    dec = c.kind != RW_AGG_COUNT; // count needs only validity
+                } else if (ty != RW_T_I64 && ty != RW_T_TS) {
                     FAIL(RW_E_INVAL, "agg arg type %d unsupported on GPU (i64/ts only)", ty);
+                }
             }
+            call_decimal.push_back(dec);
+            n_dec += dec;
         }
+        for (auto k : stream_key)
+            if (input_types[k] == RW_T_DECIMAL)
+                FAIL(RW_E_INVAL, "decimal stream keys unsupported");
         // DISTINCT dedup (aggregate/distinct.rs): one counter table per
         // distinct column. min/max DISTINCT ≡ min/max — the frontend strips
         // it — so distinct over a materialized-input call is rejected.
@@ -1604,6 +1866,19 @@ struct HashAgg {
         HIP_TRY(hipMalloc(&t.out_vals, (size_t)t.out_capacity * out_width * 8));
         HIP_TRY(hipMalloc(&t.out_nulls, (size_t)t.out_capacity * out_width));
         HIP_TRY(hipMalloc(&t.out_ops, t.out_capacity));
+        t.n_dec = n_dec;
+        if (n_dec) {
+            HIP_TRY(hipMalloc(&t.dsum, (size_t)n_dec * 4 * cap * 8));
+            HIP_TRY(hipMemset(t.dsum, 0, (size_t)n_dec * 4 * cap * 8));
+            HIP_TRY(hipMalloc(&t.dscl, (size_t)n_dec * cap * 4));
+            HIP_TRY(hipMemset(t.dscl, 0, (size_t)n_dec * cap * 4));
+            HIP_TRY(hipMalloc(&t.dspec, (size_t)n_dec * 3 * cap * 8));
+            HIP_TRY(hipMemset(t.dspec, 0, (size_t)n_dec * 3 * cap * 8));
+            HIP_TRY(hipMalloc(&t.out_vals2,
+                              (size_t)t.out_capacity * out_width * 8));
+            HIP_TRY(hipMalloc(&t.prev2, (size_t)n_calls * cap * 8));
+            HIP_TRY(hipMemset(t.prev2, 0, (size_t)n_calls * cap * 8));
+        }
         t.n_sk = (int)stream_key.size();
         if (n_minput) {
             t.mrow_cap = 1u << 22;
@@ -1638,7 +1913,8 @@ struct HashAgg {
         uint32_t cap = 4096;
         while (cap < n_rows) cap <<= 1;
         for (int i = 0; i < n_batch_slots(); i++) {
-            HIP_TRY(hipMalloc(&stage.col_vals[i], (size_t)cap * 8));
+            // 16 B/row so decimal slots (2 words/row) fit
+            HIP_TRY(hipMalloc(&stage.col_vals[i], (size_t)cap * 16));
             HIP_TRY(hipMalloc(&stage.col_valid[i], cap));
         }
         HIP_TRY(hipMalloc(&stage.ops, cap));
@@ -1671,9 +1947,11 @@ struct HashAgg {
         }
         auto upcol = [&](int bi, uint32_t col_idx) -> int {
             const RwColumn& col = c->cols[col_idx];
-            if (col.type != RW_T_I64 && col.type != RW_T_TS)
+            size_t w = 8;
+            if (col.type == RW_T_DECIMAL) w = 16; // 2 words/row
+            else if (col.type != RW_T_I64 && col.type != RW_T_TS)
                 FAIL(RW_E_INVAL, "column type %d unsupported on GPU", col.type);
-            HIP_TRY(hipMemcpyAsync(b.col_vals[bi], col.data, (size_t)n * 8,
+            HIP_TRY(hipMemcpyAsync(b.col_vals[bi], col.data, (size_t)n * w,
                                    hipMemcpyHostToDevice, stream));
             HIP_TRY(hipMemcpyAsync(b.col_valid[bi], col.valid, n,
                                    hipMemcpyHostToDevice, stream));
@@ -1728,7 +2006,7 @@ struct HashAgg {
     void launch_apply(const AggBatch& b, uint32_t r0, uint32_t r1) {
         auto a0 = cd(0), a1 = cd(1), a2 = cd(2), a3 = cd(3);
         if (b.dense && KW == 1 && b.stride == 1 && n_minput == 0 &&
-            distinct_slots.empty() &&
+            n_dec == 0 && distinct_slots.empty() &&
             debug_mode == 0 && ((uintptr_t)(b.col_vals[0] + r0) & 31) == 0 &&
             (r1 - r0) >= 1024) {
             // rows-per-lane (A/B-selectable via RW_AGG_RPL). Measured on
@@ -2125,6 +2403,10 @@ struct HashAgg {
         if (ctr[2] == 2) FAIL(RW_E_INTERNAL, "agg output buffer overflow");
         if (ctr[2] == 3) FAIL(RW_E_INTERNAL, "agg minput row store full");
         if (ctr[2] == 4) FAIL(RW_E_INTERNAL, "agg distinct dedup table full");
+        if (ctr[2] == 5)
+            FAIL(RW_E_OVERFLOW,
+                 "decimal sum outside the exact 96-bit domain (the "
+                 "reference's order-dependent rescale path; unsupported)");
         return RW_OK;
     }
 
@@ -2145,6 +2427,11 @@ struct HashAgg {
         HIP_TRY(hipMalloc(&t.out_vals, (size_t)cap * out_width * 8));
         HIP_TRY(hipMalloc(&t.out_nulls, (size_t)cap * out_width));
         HIP_TRY(hipMalloc(&t.out_ops, cap));
+        if (n_dec) {
+            hipFree(t.out_vals2);
+            t.out_vals2 = nullptr;
+            HIP_TRY(hipMalloc(&t.out_vals2, (size_t)cap * out_width * 8));
+        }
         t.out_capacity = (uint32_t)cap;
         return RW_OK;
     }
@@ -2273,13 +2560,19 @@ struct HashAgg {
             std::vector<long long> vals((size_t)n_out * out_width);
             std::vector<uint8_t> nulls((size_t)n_out * out_width);
             std::vector<uint8_t> ops(n_out);
+            std::vector<long long> vals2;
             HIP_TRY(hipMemcpy(vals.data(), t.out_vals, vals.size() * 8,
                               hipMemcpyDeviceToHost));
             HIP_TRY(hipMemcpy(nulls.data(), t.out_nulls, nulls.size(),
                               hipMemcpyDeviceToHost));
             HIP_TRY(hipMemcpy(ops.data(), t.out_ops, n_out, hipMemcpyDeviceToHost));
-            spill_records(vals, nulls, ops, n_out);
-            slice_outputs(vals, nulls, ops, n_out);
+            if (n_dec) {
+                vals2.resize(vals.size());
+                HIP_TRY(hipMemcpy(vals2.data(), t.out_vals2,
+                                  vals2.size() * 8, hipMemcpyDeviceToHost));
+            }
+            spill_records(vals, nulls, ops, n_out, vals2);
+            slice_outputs(vals, nulls, ops, n_out, vals2);
         }
         HIP_TRY(hipMemset(t.counters, 0, 12));
         return RW_OK;
@@ -2315,7 +2608,8 @@ struct HashAgg {
     // input states encode None, agg_group.rs:417-421).
     void spill_records(const std::vector<long long>& vals,
                        const std::vector<uint8_t>& nulls,
-                       const std::vector<uint8_t>& ops, uint32_t n_out) {
+                       const std::vector<uint8_t>& ops, uint32_t n_out,
+                       const std::vector<long long>& vals2 = {}) {
         auto put32 = [&](uint32_t x) {
             for (int b = 0; b < 4; b++) spill.push_back((uint8_t)(x >> (8 * b)));
         };
@@ -2342,7 +2636,8 @@ struct HashAgg {
                                                     {true, 0, 0});
                     } else {
                         size_t ix = (size_t)r * out_width + KW + ci;
-                        rwcodec::DatumC d{nulls[ix] != 0, vals[ix], 0};
+                        rwcodec::DatumC d{nulls[ix] != 0, vals[ix], 0,
+                                          vals2.empty() ? 0 : vals2[ix]};
                         rwcodec::value_encode_datum(v, calls[ci].ret_type, d);
                     }
                 }
@@ -2443,7 +2738,8 @@ struct HashAgg {
     // (stream_chunk_builder.rs:188-218)
     void slice_outputs(const std::vector<long long>& vals,
                        const std::vector<uint8_t>& nulls,
-                       const std::vector<uint8_t>& ops, uint32_t n_out) {
+                       const std::vector<uint8_t>& ops, uint32_t n_out,
+                       const std::vector<long long>& vals2 = {}) {
         uint32_t start = 0;
         uint32_t max_rows = desc.chunk_size ? desc.chunk_size : 1024;
         while (start < n_out) {
@@ -2454,7 +2750,7 @@ struct HashAgg {
                 // extend by one (the builder's size max+1 case)
                 if (ops[start + take - 1] == RW_OP_UPDATE_DELETE) take += 1;
             }
-            outq.push_back(make_chunk(vals, nulls, ops, start, take));
+            outq.push_back(make_chunk(vals, nulls, ops, start, take, vals2));
             start += take;
         }
     }
@@ -2462,12 +2758,28 @@ struct HashAgg {
     RwChunk* make_chunk(const std::vector<long long>& vals,
                         const std::vector<uint8_t>& nulls,
                         const std::vector<uint8_t>& ops, uint32_t start,
-                        uint32_t n) {
+                        uint32_t n,
+                        const std::vector<long long>& vals2 = {}) {
         auto* ch = new RwChunk();
         auto* cols = new RwColumn[out_width];
         auto* o = new uint8_t[n];
         memcpy(o, ops.data() + start, n);
         for (int ci = 0; ci < out_width; ci++) {
+            uint8_t ty = out_types[ci];
+            if (ty == RW_T_DECIMAL && !vals2.empty()) {
+                auto* data = new int64_t[(size_t)2 * n];
+                auto* valid = new uint8_t[n];
+                for (uint32_t r = 0; r < n; r++) {
+                    size_t ix = (size_t)(start + r) * out_width + ci;
+                    valid[r] = !nulls[ix];
+                    data[2 * r] = vals[ix];
+                    data[2 * r + 1] = vals2[ix];
+                }
+                cols[ci].type = ty;
+                cols[ci].valid = valid;
+                cols[ci].data = data;
+                continue;
+            }
             auto* data = new int64_t[n];
             auto* valid = new uint8_t[n];
             for (uint32_t r = 0; r < n; r++) {
@@ -2522,6 +2834,13 @@ struct HashAgg {
             hipFree(t.out_vals);
             hipFree(t.out_nulls);
             hipFree(t.out_ops);
+            if (t.dsum) {
+                hipFree(t.dsum);
+                hipFree(t.dscl);
+                hipFree(t.dspec);
+                hipFree(t.out_vals2);
+                hipFree(t.prev2);
+            }
             if (t.mheads) {
                 hipFree(t.mheads);
                 hipFree(t.mval);
@@ -2624,6 +2943,8 @@ int rw_hash_agg_flush(void* h, uint64_t epoch) { return ((HashAgg*)h)->flush(epo
 // epoch-batched ingest mode (rw_stream.h): push_chunk stages, flush applies
 int rw_hash_agg_ingest_mode(void* h, int epoch_batched) {
     auto* agg = (HashAgg*)h;
+    if (epoch_batched && agg->n_dec)
+        FAIL(RW_E_INVAL, "epoch-batched ingest + decimal not yet supported");
     if (epoch_batched && (agg->n_minput > 0 || !agg->distinct_slots.empty()))
         FAIL(RW_E_INVAL,
              "epoch-batched ingest needs order-free value states "
@@ -2643,7 +2964,8 @@ void* rw_agg_bench_preload(void* h, const RwChunk* c) {
     auto* b = new AggBatch{};
     uint32_t n = c->n_rows;
     for (int i = 0; i < agg->n_batch_slots(); i++) {
-        if (hipMalloc(&b->col_vals[i], (size_t)n * 8) != hipSuccess) return nullptr;
+        // 16 B/row so decimal slots (2 words/row) fit
+        if (hipMalloc(&b->col_vals[i], (size_t)n * 16) != hipSuccess) return nullptr;
         if (hipMalloc(&b->col_valid[i], n) != hipSuccess) return nullptr;
     }
     if (hipMalloc(&b->ops, n) != hipSuccess) return nullptr;
@@ -2757,6 +3079,8 @@ int rw_agg_apply_payload(void* h, const uint8_t* payload,
                          const uint64_t* block_rows, int n_blocks, int n_cols,
                          int dense) {
     auto* agg = (HashAgg*)h;
+    if (agg->n_dec)
+        FAIL(RW_E_INVAL, "decimal columns not yet carried by the exchange");
     if (n_cols != agg->KW + agg->n_calls)
         FAIL(RW_E_INVAL, "payload n_cols %d != %d", n_cols, agg->KW + agg->n_calls);
     uint64_t off = 0;
@@ -2925,6 +3249,8 @@ int rw_hash_agg_restore(void* h, const uint8_t* buf, uint64_t len) {
         FAIL(RW_E_INVAL,
              "restore with materialized-input aggregates requires "
              "minput-table spill (not yet drained)");
+    if (agg->n_dec > 0)
+        FAIL(RW_E_INVAL, "decimal restore not yet implemented on GPU");
     std::map<std::string, std::vector<uint8_t>> merged;
     bool ok = rwcodec::for_each_frame(
         buf, len,
@@ -5615,6 +5941,8 @@ int rw_join_vnode_hop(void* join_h, void* batch, uint32_t col, uint8_t type,
 int rw_agg_apply_joinout(void* agg_h, void* join_h) {
     auto* agg = (HashAgg*)agg_h;
     auto* j = (HashJoin*)join_h;
+    if (agg->n_dec)
+        FAIL(RW_E_INVAL, "decimal columns not yet in the join output block");
     if (hipStreamSynchronize(j->stream) != hipSuccess)
         FAIL(RW_E_INTERNAL, "join sync failed");
     uint32_t ctr[2];

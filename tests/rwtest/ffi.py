@@ -13,6 +13,7 @@ REPO = os.path.dirname(os.path.dirname(os.path.dirname(os.path.abspath(__file__)
 
 # type ids (rw_chunk.h)
 T_I64, T_I32, T_F64, T_F32, T_BOOL, T_TS = 0, 1, 2, 3, 4, 5
+T_DECIMAL = 6
 OP_INSERT, OP_DELETE, OP_UPDATE_DELETE, OP_UPDATE_INSERT = 0, 1, 2, 3
 OP_BY_TOKEN = {"+": OP_INSERT, "-": OP_DELETE, "U-": OP_UPDATE_DELETE, "U+": OP_UPDATE_INSERT}
 TOKEN_BY_OP = {v: k for k, v in OP_BY_TOKEN.items()}
@@ -24,6 +25,7 @@ NP_BY_TYPE = {
     T_F32: np.float32,
     T_BOOL: np.uint8,
     T_TS: np.int64,
+    T_DECIMAL: np.dtype("V16"),  # raw rust_decimal serialize image
 }
 
 AGG_COUNT_STAR, AGG_COUNT, AGG_SUM, AGG_SUM0, AGG_MIN, AGG_MAX = 0, 1, 2, 3, 4, 5
@@ -161,6 +163,8 @@ class Chunk:
             for t, d, v in zip(self.types, self.cols, self.valids):
                 if not v[r]:
                     vals.append(None)
+                elif t == T_DECIMAL:
+                    vals.append(bytes(d[r]))
                 elif t in (T_F64, T_F32):
                     vals.append(float(d[r]))
                 else:
@@ -680,3 +684,43 @@ def join_restore(lib, h, side, buf, deg_buf=b""):
                                 len(deg_buf) if deg_buf else 0)
     if rc != 0:
         raise RuntimeError(f"join restore failed {rc}: {lib.last_error()}")
+
+
+# ---- decimal helpers (rust_decimal 1.40.0 serialize layout; see
+# include/rw_chunk.h RW_T_DECIMAL) ----
+import decimal as _pydec
+import struct as _struct
+
+
+def dec16(v):
+    """16-byte rust_decimal image from a python Decimal / str / int.
+    "NaN" / "Inf" / "-Inf" map to the reference's special encodings
+    (types/decimal.rs:583-592)."""
+    if isinstance(v, str) and v in ("NaN", "Inf", "-Inf"):
+        b0 = {"NaN": 1, "Inf": 2, "-Inf": 3}[v]
+        return bytes([b0]) + b"\x00" * 15
+    d = _pydec.Decimal(v)
+    sign, digits, exp = d.as_tuple()
+    assert exp <= 0 and -exp <= 28, f"scale {-exp} out of range"
+    m = int("".join(map(str, digits)))
+    assert m < (1 << 96), "mantissa exceeds 96 bits"
+    flags = ((-exp) << 16) | (0x80000000 if sign else 0)
+    return _struct.pack("<IIII", flags, m & 0xFFFFFFFF,
+                        (m >> 32) & 0xFFFFFFFF, (m >> 64) & 0xFFFFFFFF)
+
+
+def dec16_value(b):
+    """Python Decimal (or 'NaN'/'Inf'/'-Inf') from a 16-byte image."""
+    if b[0] in (1, 2, 3):
+        return {1: "NaN", 2: "Inf", 3: "-Inf"}[b[0]]
+    flags, lo, mid, hi = _struct.unpack("<IIII", b)
+    m = (hi << 64) | (mid << 32) | lo
+    scale = (flags >> 16) & 0xFF
+    sign = -1 if (flags >> 31) & 1 else 1
+    return _pydec.Decimal(sign * m).scaleb(-scale)
+
+
+def dec_col(values):
+    """numpy V16 column from an iterable of dec16-able values."""
+    raw = b"".join(dec16(v) for v in values)
+    return np.frombuffer(raw, dtype="V16").copy()
