@@ -6916,6 +6916,58 @@ __global__ void membw_probe_kernel(const ulonglong2* __restrict__ p, size_t n,
     if (acc == 0xdeadbeefdeadbeefULL) *sink = acc; // never taken
 }
 
+// RANDOM-access ceiling probe: each lane chases splitmix-scattered 64-B
+// lines over `bytes` of HBM — the achievable rate for the hash-join's
+// probe/insert access pattern (one random line per touch), the honest
+// denominator for q8's roofline discussion. Reads the first
+// u64 of a random line per step; reported as GB/s of 64-B lines touched.
+__global__ void membw_rand_probe_kernel(const unsigned long long* __restrict__ p,
+                                        size_t n_lines, int steps_per_thread,
+                                        unsigned long long* sink) {
+    uint64_t x = (uint64_t)(blockIdx.x * blockDim.x + threadIdx.x) *
+                     0x9e3779b97f4a7c15ULL +
+                 0x243f6a8885a308d3ULL;
+    unsigned long long acc = 0;
+    for (int i = 0; i < steps_per_thread; i++) {
+        x += 0x9e3779b97f4a7c15ULL;
+        uint64_t h = x;
+        h = (h ^ (h >> 30)) * 0xbf58476d1ce4e5b9ULL;
+        h = (h ^ (h >> 27)) * 0x94d049bb133111ebULL;
+        h ^= h >> 31;
+        acc += p[(h % n_lines) * 8]; // 64-B line stride (8 u64)
+    }
+    if (acc == 0xdeadbeefdeadbeefULL) *sink = acc;
+}
+
+extern "C" int rw_membw_rand_probe(uint64_t bytes, int steps_per_thread,
+                                   double* glines_out, double* gbps_out) {
+    size_t n_lines = bytes / 64;
+    unsigned long long* p = nullptr;
+    unsigned long long* sink = nullptr;
+    if (hipMalloc(&p, n_lines * 64) != hipSuccess) return RW_E_INTERNAL;
+    (void)hipMalloc(&sink, 8);
+    (void)hipMemset(p, 1, n_lines * 64);
+    int grid = 2048, blk = 256;
+    hipEvent_t e0, e1;
+    (void)hipEventCreate(&e0);
+    (void)hipEventCreate(&e1);
+    membw_rand_probe_kernel<<<grid, blk>>>(p, n_lines, steps_per_thread, sink);
+    (void)hipEventRecord(e0);
+    membw_rand_probe_kernel<<<grid, blk>>>(p, n_lines, steps_per_thread, sink);
+    (void)hipEventRecord(e1);
+    if (hipEventSynchronize(e1) != hipSuccess) return RW_E_INTERNAL;
+    float ms = 0;
+    (void)hipEventElapsedTime(&ms, e0, e1);
+    double touches = (double)grid * blk * steps_per_thread;
+    *glines_out = touches / (ms * 1e-3) / 1e9;
+    *gbps_out = touches * 64 / (ms * 1e-3) / 1e9;
+    (void)hipFree(p);
+    (void)hipFree(sink);
+    (void)hipEventDestroy(e0);
+    (void)hipEventDestroy(e1);
+    return RW_OK;
+}
+
 extern "C" int rw_membw_probe(uint64_t bytes, int iters, double* gbps_out) {
     size_t n = bytes / sizeof(ulonglong2);
     ulonglong2* p = nullptr;

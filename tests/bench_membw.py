@@ -29,4 +29,23 @@ for gib in (1, 4, 16):
         print(json.dumps({"error": rc, "GiB": gib}))
         sys.exit(1)
     out.append({"GiB": gib, "read_GBps": round(g.value, 1)})
-print(json.dumps({"membw_probe": out}))
+
+# RANDOM 64-B line touches (the hash probe/insert pattern): the honest
+# ceiling for q8's random accesses (rw_membw_rand_probe)
+L.rw_membw_rand_probe.restype = ctypes.c_int
+L.rw_membw_rand_probe.argtypes = [ctypes.c_uint64, ctypes.c_int,
+                                  ctypes.POINTER(ctypes.c_double),
+                                  ctypes.POINTER(ctypes.c_double)]
+rand = []
+for mib, name in ((128, "LIC-resident (128 MiB)"), (1024, "1 GiB"),
+                  (8192, "8 GiB")):
+    gl = ctypes.c_double(0.0)
+    gb = ctypes.c_double(0.0)
+    rc = L.rw_membw_rand_probe(mib << 20, 4096, ctypes.byref(gl),
+                               ctypes.byref(gb))
+    if rc != 0:
+        print(json.dumps({"error": rc, "MiB": mib}))
+        sys.exit(1)
+    rand.append({"region": name, "Glines_per_s": round(gl.value, 2),
+                 "line_GBps": round(gb.value, 1)})
+print(json.dumps({"membw_probe": out, "membw_random_64B": rand}))
