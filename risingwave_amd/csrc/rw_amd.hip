@@ -5234,10 +5234,8 @@ struct HashJoin {
             if (ev_harvest(slot) != RW_OK) return RW_E_INTERNAL;
             HIP_TRY(hipEventRecord(ev0[slot], stream));
         }
-        static int dbg_skip = [] {
-            const char* e = getenv("RW_JOIN_SKIP"); // bench A/B only:
-            return e ? atoi(e) : 0;  // 1=no emit writes, 2=no own insert
-        }();
+        const char* dbg_e = getenv("RW_JOIN_SKIP"); // bench A/B only:
+        int dbg_skip = dbg_e ? atoi(dbg_e) : 0; // 1=no emits, 2=no own insert
         if (can_partition(b, r0, r1)) {
             int rc = ensure_part_bufs();
             if (rc != RW_OK) return rc;
@@ -6937,6 +6935,100 @@ __global__ void membw_rand_probe_kernel(const unsigned long long* __restrict__ p
         acc += p[(h % n_lines) * 8]; // 64-B line stride (8 u64)
     }
     if (acc == 0xdeadbeefdeadbeefULL) *sink = acc;
+}
+
+// random atomicAdd throughput (the insert path's slot-CAS analogue)
+__global__ void membw_rand_atomic_kernel(unsigned long long* p, size_t n_lines,
+                                         int steps_per_thread) {
+    uint64_t x = (uint64_t)(blockIdx.x * blockDim.x + threadIdx.x) *
+                     0x9e3779b97f4a7c15ULL +
+                 0x243f6a8885a308d3ULL;
+    for (int i = 0; i < steps_per_thread; i++) {
+        x += 0x9e3779b97f4a7c15ULL;
+        uint64_t h = x;
+        h = (h ^ (h >> 30)) * 0xbf58476d1ce4e5b9ULL;
+        h = (h ^ (h >> 27)) * 0x94d049bb133111ebULL;
+        h ^= h >> 31;
+        atomicAdd(&p[(h % n_lines) * 8], 1ull);
+    }
+}
+
+// DEPENDENT random chase: each step's address depends on the previous
+// load's value (the probe kernel's hash->slot->record chain shape);
+// `mlp` independent chains per thread expose memory-level parallelism
+__global__ void membw_rand_chase_kernel(const unsigned long long* __restrict__ p,
+                                        size_t n_lines, int steps, int mlp,
+                                        unsigned long long* sink) {
+    uint64_t x[8];
+    for (int j = 0; j < mlp && j < 8; j++)
+        x[j] = (uint64_t)(blockIdx.x * blockDim.x + threadIdx.x) *
+                   0x9e3779b97f4a7c15ULL +
+               j * 0x94d049bb133111ebULL + 1;
+    unsigned long long acc = 0;
+    for (int i = 0; i < steps; i++) {
+        for (int j = 0; j < mlp && j < 8; j++) {
+            uint64_t h = x[j];
+            h = (h ^ (h >> 30)) * 0xbf58476d1ce4e5b9ULL;
+            h = (h ^ (h >> 27)) * 0x94d049bb133111ebULL;
+            h ^= h >> 31;
+            unsigned long long v = p[(h % n_lines) * 8];
+            x[j] += v + 0x9e3779b97f4a7c15ULL; // DEPENDS on the load
+            acc += v;
+        }
+    }
+    if (acc == 0xdeadbeefdeadbeefULL) *sink = acc;
+}
+
+extern "C" int rw_membw_rand_atomic(uint64_t bytes, int steps_per_thread,
+                                    double* gops_out) {
+    size_t n_lines = bytes / 64;
+    unsigned long long* p = nullptr;
+    if (hipMalloc(&p, n_lines * 64) != hipSuccess) return RW_E_INTERNAL;
+    (void)hipMemset(p, 0, n_lines * 64);
+    int grid = 2048, blk = 256;
+    hipEvent_t e0, e1;
+    (void)hipEventCreate(&e0);
+    (void)hipEventCreate(&e1);
+    membw_rand_atomic_kernel<<<grid, blk>>>(p, n_lines, steps_per_thread);
+    (void)hipEventRecord(e0);
+    membw_rand_atomic_kernel<<<grid, blk>>>(p, n_lines, steps_per_thread);
+    (void)hipEventRecord(e1);
+    if (hipEventSynchronize(e1) != hipSuccess) return RW_E_INTERNAL;
+    float ms = 0;
+    (void)hipEventElapsedTime(&ms, e0, e1);
+    *gops_out = (double)grid * blk * steps_per_thread / (ms * 1e-3) / 1e9;
+    (void)hipFree(p);
+    (void)hipEventDestroy(e0);
+    (void)hipEventDestroy(e1);
+    return RW_OK;
+}
+
+extern "C" int rw_membw_rand_chase(uint64_t bytes, int steps, int mlp,
+                                   double* glines_out) {
+    size_t n_lines = bytes / 64;
+    unsigned long long* p = nullptr;
+    unsigned long long* sink = nullptr;
+    if (hipMalloc(&p, n_lines * 64) != hipSuccess) return RW_E_INTERNAL;
+    (void)hipMalloc(&sink, 8);
+    (void)hipMemset(p, 1, n_lines * 64);
+    int grid = 2048, blk = 256;
+    hipEvent_t e0, e1;
+    (void)hipEventCreate(&e0);
+    (void)hipEventCreate(&e1);
+    membw_rand_chase_kernel<<<grid, blk>>>(p, n_lines, steps, mlp, sink);
+    (void)hipEventRecord(e0);
+    membw_rand_chase_kernel<<<grid, blk>>>(p, n_lines, steps, mlp, sink);
+    (void)hipEventRecord(e1);
+    if (hipEventSynchronize(e1) != hipSuccess) return RW_E_INTERNAL;
+    float ms = 0;
+    (void)hipEventElapsedTime(&ms, e0, e1);
+    *glines_out =
+        (double)grid * blk * steps * mlp / (ms * 1e-3) / 1e9;
+    (void)hipFree(p);
+    (void)hipFree(sink);
+    (void)hipEventDestroy(e0);
+    (void)hipEventDestroy(e1);
+    return RW_OK;
 }
 
 extern "C" int rw_membw_rand_probe(uint64_t bytes, int steps_per_thread,

@@ -48,4 +48,25 @@ for mib, name in ((128, "LIC-resident (128 MiB)"), (1024, "1 GiB"),
         sys.exit(1)
     rand.append({"region": name, "Glines_per_s": round(gl.value, 2),
                  "line_GBps": round(gb.value, 1)})
-print(json.dumps({"membw_probe": out, "membw_random_64B": rand}))
+L.rw_membw_rand_atomic.restype = ctypes.c_int
+L.rw_membw_rand_atomic.argtypes = [ctypes.c_uint64, ctypes.c_int,
+                                   ctypes.POINTER(ctypes.c_double)]
+L.rw_membw_rand_chase.restype = ctypes.c_int
+L.rw_membw_rand_chase.argtypes = [ctypes.c_uint64, ctypes.c_int,
+                                  ctypes.c_int,
+                                  ctypes.POINTER(ctypes.c_double)]
+at = []
+for mib in (128, 1024):
+    g = ctypes.c_double(0.0)
+    rc = L.rw_membw_rand_atomic(mib << 20, 2048, ctypes.byref(g))
+    assert rc == 0
+    at.append({"MiB": mib, "Gatomics_per_s": round(g.value, 2)})
+ch = []
+for mlp in (1, 2, 4, 8):
+    g = ctypes.c_double(0.0)
+    rc = L.rw_membw_rand_chase(1 << 30, 512, mlp, ctypes.byref(g))
+    assert rc == 0
+    ch.append({"mlp": mlp, "Glines_per_s": round(g.value, 2)})
+print(json.dumps({"membw_probe": out, "membw_random_64B": rand,
+                  "membw_random_atomic": at,
+                  "membw_dependent_chase_1GiB": ch}))
