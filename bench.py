@@ -832,6 +832,12 @@ def bench_q8(args, ffi, gpu_lib, rng, rank, world, dist):
         batches.append(preload(SIDE_LEFT, ids, rowid))
         rowid += batch_rows
 
+    # A/B hook: RW_JOIN_SKIP_PROBE applies RW_JOIN_SKIP only to the PROBE
+    # phase (the build above ran unskipped, so the tables are populated —
+    # setting RW_JOIN_SKIP outside would also skip the build inserts and
+    # measure an empty-table probe)
+    if os.environ.get("RW_JOIN_SKIP_PROBE"):
+        os.environ["RW_JOIN_SKIP"] = os.environ["RW_JOIN_SKIP_PROBE"]
     # the step loop runs in C with a stream-ordered cursor reset per step —
     # the Python-side per-step drain cost ~0.15 ms of pure sync overhead
     L.rw_join_bench_run.restype = ctypes.c_int
