@@ -3998,7 +3998,11 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
             if (total && base + total > out.cap) {
                 if (lane == 0) atomicExch(&out.counters[1], 1u); // overflow
             } else if (my_n && !(dbg_skip & 1)) {
-                // second walk: emit (JoinStreamChunkBuilder::append_row)
+                // second walk: emit (JoinStreamChunkBuilder::append_row).
+                // Emit stores are NONTEMPORAL: the rows are written once
+                // and only read by later launches, and regular stores pay
+                // a read-for-ownership line fetch per partial write
+                // (RW_JOIN_NT_OFF=1 re-measures the cached variant).
                 uint32_t row = mhead;
                 uint32_t k = 0;
                 while (row != UINT32_MAX && k < my_n) {
@@ -4008,22 +4012,45 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
                                     false) &&
                         join_cond_ok(m, S, b, r, h->validbits, jvals(h))) {
                         uint32_t orow = my_base + k;
-                        out.ops[orow] = op;
                         const long long* mv = jvals(h);
-                        for (int c = 0; c < m.n_out; c++) {
-                            bool from_probe = (int)m.out_src[c] == S;
-                            uint8_t col = m.out_col[c];
-                            int64_t v;
-                            uint8_t valid;
-                            if (from_probe) {
-                                valid = b.col_valid[col][r];
-                                v = b.col_vals[col][r];
-                            } else {
-                                valid = (h->validbits >> col) & 1;
-                                v = mv[col];
+                        if (dbg_skip & 8) {
+                            out.ops[orow] = op;
+                            for (int c = 0; c < m.n_out; c++) {
+                                bool from_probe = (int)m.out_src[c] == S;
+                                uint8_t col = m.out_col[c];
+                                int64_t v;
+                                uint8_t valid;
+                                if (from_probe) {
+                                    valid = b.col_valid[col][r];
+                                    v = b.col_vals[col][r];
+                                } else {
+                                    valid = (h->validbits >> col) & 1;
+                                    v = mv[col];
+                                }
+                                out.vals[(size_t)c * out.cap + orow] = valid ? v : 0;
+                                out.nulls[(size_t)c * out.cap + orow] = !valid;
                             }
-                            out.vals[(size_t)c * out.cap + orow] = valid ? v : 0;
-                            out.nulls[(size_t)c * out.cap + orow] = !valid;
+                        } else {
+                            __builtin_nontemporal_store(op, &out.ops[orow]);
+                            for (int c = 0; c < m.n_out; c++) {
+                                bool from_probe = (int)m.out_src[c] == S;
+                                uint8_t col = m.out_col[c];
+                                int64_t v;
+                                uint8_t valid;
+                                if (from_probe) {
+                                    valid = b.col_valid[col][r];
+                                    v = b.col_vals[col][r];
+                                } else {
+                                    valid = (h->validbits >> col) & 1;
+                                    v = mv[col];
+                                }
+                                __builtin_nontemporal_store(
+                                    valid ? v : 0,
+                                    &out.vals[(size_t)c * out.cap + orow]);
+                                __builtin_nontemporal_store(
+                                    (uint8_t)!valid,
+                                    &out.nulls[(size_t)c * out.cap + orow]);
+                            }
                         }
                         k++;
                     }
