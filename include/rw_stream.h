@@ -279,6 +279,16 @@ int rw_agg_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len);
  * drain stream (same key encoding), required for join types that keep
  * degrees; pass NULL/0 otherwise. */
 int rw_hash_agg_restore(void* h, const uint8_t* buf, uint64_t len);
+/* Materialized-input (retractable min/max) state TABLES — one per minput
+ * call in the reference (AggStateStorage::MaterializedInput,
+ * test_utils/agg_executor.rs:63-121): pk = group key ∥ value (ASC min /
+ * DESC max) ∥ stream key, value = the full row. Same record framing and
+ * per-epoch delta semantics as the other drains. On recovery, restore
+ * EVERY minput table BEFORE rw_hash_agg_restore (prev outputs are
+ * recomputed from the hydrated chains, agg_group.rs:219-221). */
+int rw_agg_n_minput_tables(void* h);
+int rw_agg_minput_drain(void* h, int mi, uint8_t** buf, uint64_t* len);
+int rw_agg_minput_restore(void* h, int mi, const uint8_t* buf, uint64_t len);
 int rw_hash_join_restore(void* h, int side, const uint8_t* buf, uint64_t len,
                          const uint8_t* deg_buf, uint64_t deg_len);
 int rw_join_checkpoint_drain(void* h, int side, uint8_t** buf, uint64_t* len);

@@ -180,6 +180,139 @@ struct HashAggOracle {
     std::vector<std::set<Row, RowOrderLess>> dedup_touched;
     std::vector<std::set<Row, RowOrderLess>> dedup_persisted;
 
+    // §8f-2 round-2: materialized-input state-TABLE spill. One table per
+    // retractable min/max call in the reference (AggStateStorage::
+    // MaterializedInput, test_utils/agg_executor.rs:63-121): columns =
+    // group key ++ arg value ++ stream key, pk ordered [group ASC, value
+    // ASC(min)/DESC(max), stream key ASC], NULLs largest. Per-epoch
+    // deltas keyed by the memcomparable pk with mem-table netting.
+    std::vector<std::map<std::string, std::pair<uint8_t, std::vector<uint8_t>>>>
+        minput_delta;
+
+    int minput_ordinal(size_t ci) const {
+        int o = 0;
+        for (size_t j = 0; j < ci; j++) o += call_is_minput[j];
+        return o;
+    }
+
+    void minput_encode(size_t ci, const Row& gkey, const Row& entry_key,
+                       std::string* kout, std::vector<uint8_t>* vout) {
+        const auto& c = calls[ci];
+        std::vector<uint8_t> k;
+        for (size_t i = 0; i < gkey.size(); i++) {
+            rwcodec::DatumC d{gkey[i].null, gkey[i].i, gkey[i].d, gkey[i].i2};
+            rwcodec::memcmp_encode_datum(k, group_key_types[i], d, {});
+        }
+        rwcodec::DatumC dv{entry_key[0].null, entry_key[0].i, entry_key[0].d,
+                           entry_key[0].i2};
+        rwcodec::memcmp_encode_datum(k, input_types[c.arg], dv,
+                                     {c.kind == RW_AGG_MAX, true});
+        for (size_t j = 0; j < stream_key.size(); j++) {
+            const Datum& d0 = entry_key[1 + j];
+            rwcodec::DatumC d{d0.null, d0.i, d0.d, d0.i2};
+            rwcodec::memcmp_encode_datum(k, input_types[stream_key[j]], d, {});
+        }
+        kout->assign((const char*)k.data(), k.size());
+        if (vout) {
+            for (size_t i = 0; i < gkey.size(); i++) {
+                rwcodec::DatumC d{gkey[i].null, gkey[i].i, gkey[i].d,
+                                  gkey[i].i2};
+                rwcodec::value_encode_datum(*vout, group_key_types[i], d);
+            }
+            rwcodec::DatumC dv2{entry_key[0].null, entry_key[0].i,
+                                entry_key[0].d, entry_key[0].i2};
+            rwcodec::value_encode_datum(*vout, input_types[c.arg], dv2);
+            for (size_t j = 0; j < stream_key.size(); j++) {
+                const Datum& d0 = entry_key[1 + j];
+                rwcodec::DatumC d{d0.null, d0.i, d0.d, d0.i2};
+                rwcodec::value_encode_datum(*vout, input_types[stream_key[j]],
+                                            d);
+            }
+        }
+    }
+
+    int minput_drain(int mi, std::vector<uint8_t>& out) {
+        if (mi < 0 || mi >= n_minput_total())
+            FAIL(RW_E_INVAL, "minput table index %d", mi);
+        if ((size_t)mi >= minput_delta.size()) return RW_OK;
+        auto put32 = [&](uint32_t x) {
+            for (int b = 0; b < 4; b++) out.push_back((uint8_t)(x >> (8 * b)));
+        };
+        for (auto& [k, pv] : minput_delta[mi]) {
+            out.push_back(pv.first);
+            put32((uint32_t)k.size());
+            out.insert(out.end(), k.begin(), k.end());
+            put32((uint32_t)pv.second.size());
+            out.insert(out.end(), pv.second.begin(), pv.second.end());
+        }
+        minput_delta[mi].clear();
+        return RW_OK;
+    }
+
+    int n_minput_total() const {
+        int n = 0;
+        for (auto m : call_is_minput) n += m;
+        return n;
+    }
+
+    int minput_restore(int mi, const uint8_t* buf, uint64_t len) {
+        if (mi < 0 || mi >= n_minput_total())
+            FAIL(RW_E_INVAL, "minput table index %d", mi);
+        // locate the call with this minput ordinal
+        size_t ci = 0;
+        int o = -1;
+        for (; ci < calls.size(); ci++) {
+            if (call_is_minput[ci]) o++;
+            if (o == mi) break;
+        }
+        std::map<std::string, std::vector<uint8_t>> merged;
+        bool ok = rwcodec::for_each_frame(
+            buf, len,
+            [&](uint8_t put, const uint8_t* k, uint32_t klen,
+                const uint8_t* v, uint32_t vlen) {
+                std::string key((const char*)k, klen);
+                if (put)
+                    merged[key].assign(v, v + vlen);
+                else
+                    merged.erase(key);
+            });
+        if (!ok) FAIL(RW_E_INVAL, "malformed minput spill stream");
+        for (auto& [kb, val] : merged) {
+            (void)kb;
+            size_t off = 0;
+            auto rd = [&](uint8_t ty, Datum* out2) -> bool {
+                rwcodec::DatumC d;
+                size_t n2 = rwcodec::value_decode_datum(val.data() + off,
+                                                        val.size() - off, ty,
+                                                        &d);
+                if (!n2) return false;
+                off += n2;
+                *out2 = d.null ? Datum()
+                               : (ty == RW_T_DECIMAL
+                                      ? Datum::of_dec(d.i, d.i2)
+                                      : (type_is_float(ty) ? Datum::of_d(d.d)
+                                                           : Datum::of_i(d.i)));
+                return true;
+            };
+            Row gkey(group_key_types.size());
+            for (size_t i = 0; i < gkey.size(); i++)
+                if (!rd(group_key_types[i], &gkey[i]))
+                    FAIL(RW_E_INVAL, "minput restore: bad group datum");
+            Row ek(1 + stream_key.size());
+            if (!rd(input_types[calls[ci].arg], &ek[0]))
+                FAIL(RW_E_INVAL, "minput restore: bad value datum");
+            for (size_t j = 0; j < stream_key.size(); j++)
+                if (!rd(input_types[stream_key[j]], &ek[1 + j]))
+                    FAIL(RW_E_INVAL, "minput restore: bad stream-key datum");
+            AggGroupState& g = touch(gkey);
+            g.mstates[ci].entries.emplace(ek, ek[0]);
+        }
+        // restored rows predate the epoch: clean, and no deltas
+        dirty_order.clear();
+        dirty.clear();
+        return RW_OK;
+    }
+
     RowOrderLess minput_order(const RwAggCall& c) const {
         // pk of the materialized-input table: value (ASC min / DESC max),
         // then stream key ASC (agg_executor.rs:90-105)
@@ -219,8 +352,8 @@ struct HashAggOracle {
         return it->second;
     }
 
-    int apply_row(AggGroupState& g, const ChunkView& cv, size_t r, bool retract,
-                  const std::vector<bool>& hidden) {
+    int apply_row(AggGroupState& g, const Row& gkey, const ChunkView& cv,
+                  size_t r, bool retract, const std::vector<bool>& hidden) {
         for (size_t ci = 0; ci < calls.size(); ci++) {
             const auto& c = calls[ci];
             if (!hidden.empty() && hidden[ci]) continue; // DISTINCT dup row
@@ -230,11 +363,28 @@ struct HashAggOracle {
                 key.push_back(cv.at(r, c.arg));
                 for (auto sk : stream_key) key.push_back(cv.at(r, sk));
                 auto& m = g.mstates[ci].entries;
+                int mi = minput_ordinal(ci);
+                if ((size_t)mi >= minput_delta.size())
+                    minput_delta.resize(mi + 1);
+                auto& delta = minput_delta[mi];
                 if (!retract) {
+                    std::string kb;
+                    std::vector<uint8_t> vb;
+                    minput_encode(ci, gkey, key, &kb, &vb);
+                    delta[kb] = {1, std::move(vb)};
                     m.emplace(std::move(key), cv.at(r, c.arg));
                 } else {
                     auto it = m.find(key);
-                    if (it != m.end()) m.erase(it);
+                    if (it != m.end()) {
+                        std::string kb;
+                        minput_encode(ci, gkey, key, &kb, nullptr);
+                        auto di = delta.find(kb);
+                        if (di != delta.end() && di->second.first == 1)
+                            delta.erase(di); // created+died this epoch
+                        else
+                            delta[kb] = {0, {}};
+                        m.erase(it);
+                    }
                 }
                 continue;
             }
@@ -347,7 +497,7 @@ struct HashAggOracle {
                     if (call_dedup_idx[ci] >= 0)
                         hidden[ci] = col_hidden[call_dedup_idx[ci]];
             }
-            int rc = apply_row(g, cv, r, retract, hidden);
+            int rc = apply_row(g, key, cv, r, retract, hidden);
             if (rc != RW_OK) return rc;
         }
         return RW_OK;
@@ -365,13 +515,12 @@ struct HashAggOracle {
     // contract): net PUT/DELETE frames by key, decode the value-encoded
     // row (group key ++ outputs), and seed the value states + prev
     // outputs so the next flush emits only real changes.
+    // For executors with materialized-input aggregates, call
+    // rw_agg_minput_restore for EVERY minput table BEFORE this (the
+    // reference restores prev outputs from the hydrated minput tables,
+    // agg_group.rs:219-221 + minput.rs first-entry output).
     int restore(const uint8_t* buf, uint64_t len) {
-        for (auto m : call_is_minput)
-            if (m)
-                FAIL(RW_E_INVAL,
-                     "restore with materialized-input aggregates requires "
-                     "minput-table spill (not yet drained)");
-        if (!groups.empty() || !dirty.empty())
+        if (!dirty.empty())
             FAIL(RW_E_INVAL, "restore requires a fresh executor");
         std::map<std::string, std::vector<uint8_t>> merged;
         bool ok = rwcodec::for_each_frame(
@@ -419,6 +568,13 @@ struct HashAggOracle {
             }
             AggGroupState& g = touch(key);
             for (size_t ci = 0; ci < calls.size(); ci++) {
+                if (call_is_minput[ci]) {
+                    // prev output = first entry of the (already restored)
+                    // minput table; the intermediate record stores None
+                    const auto& m = g.mstates[ci].entries;
+                    outs[ci] = m.empty() ? Datum() : m.begin()->second;
+                    continue;
+                }
                 ValueState& v2 = g.vstates[ci];
                 const Datum& o = outs[ci];
                 v2.has = !o.null;
@@ -817,6 +973,24 @@ int rw_agg_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len) {
 
 int rw_hash_agg_restore(void* h, const uint8_t* buf, uint64_t len) {
     return ((HashAggOracle*)h)->restore(buf, len);
+}
+
+int rw_agg_n_minput_tables(void* h) {
+    return ((HashAggOracle*)h)->n_minput_total();
+}
+int rw_agg_minput_drain(void* h, int mi, uint8_t** buf, uint64_t* len) {
+    auto* a = (HashAggOracle*)h;
+    std::vector<uint8_t> sp;
+    int rc = a->minput_drain(mi, sp);
+    if (rc != RW_OK) return rc;
+    *len = sp.size();
+    *buf = (uint8_t*)malloc(sp.size() ? sp.size() : 1);
+    if (!*buf) return RW_E_INTERNAL;
+    memcpy(*buf, sp.data(), sp.size());
+    return RW_OK;
+}
+int rw_agg_minput_restore(void* h, int mi, const uint8_t* buf, uint64_t len) {
+    return ((HashAggOracle*)h)->minput_restore(mi, buf, len);
 }
 
 int rw_agg_n_dedup_tables(void* h) {

@@ -691,6 +691,13 @@ struct AggTableDev {
     uint32_t* mcursor; // single counter
     uint32_t mrow_cap;
     int n_sk;
+    // minput-table spill bookkeeping (§8f-2 round 2): owner of each mrow
+    // + per-epoch kill list (deletes of pre-drain rows)
+    uint32_t* mslot;          // [mrow_cap] owning group slot
+    uint8_t* mcall;           // [mrow_cap] minput ordinal
+    uint32_t* mkilled;        // [mkilled_cap]
+    uint32_t* mkilled_cursor; // single counter
+    uint32_t mkilled_cap;
     // decimal-sum state (exact i256 at scale 28):
     unsigned long long* dsum; // [n_dec][4][cap]
     uint32_t* dscl;           // [n_dec][cap] running max input scale
@@ -831,6 +838,8 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
                         if (row >= t.mrow_cap) {
                             atomicExch(&t.counters[2], 3u); // minput store full
                         } else {
+                            st_u32(&t.mslot[row], slot);
+                            t.mcall[row] = (uint8_t)c.mord;
                             st_i64((int64_t*)&t.mval[row], val);
                             t.mval_null[row] = !valid;
                             for (int k = 0; k < t.n_sk; k++) {
@@ -865,8 +874,15 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
                                          (!va || b.col_vals[skc][rs] ==
                                                      ld_i64((const int64_t*)&t.msk[(size_t)k * t.mrow_cap + row]));
                                 }
-                                if (eq && atomicCAS(&t.malive[row], 1u, 0u) == 1u)
+                                if (eq && atomicCAS(&t.malive[row], 1u, 0u) == 1u) {
+                                    uint32_t ki =
+                                        atomicAdd(t.mkilled_cursor, 1u);
+                                    if (ki < t.mkilled_cap)
+                                        t.mkilled[ki] = row;
+                                    else
+                                        atomicExch(&t.counters[2], 3u);
                                     break;
+                                }
                             }
                             row = ld_u32(&t.mnext[row]);
                         }
@@ -1646,6 +1662,45 @@ __global__ void agg_restore_kernel(AggTableDev t, int KW, int n_calls,
             continue;
         }
         for (int ci = 0; ci < n_calls; ci++) {
+            if (calls[ci].minput) {
+                // prev output = first entry of the (already restored)
+                // minput chain (output_first, minput.rs:236-241; see
+                // agg_flush_kernel's identical walk)
+                const AggCallDev& c = calls[ci];
+                uint32_t row = t.mheads[(size_t)c.mord * cap + slot];
+                bool any = false, any_null = false;
+                long long best = 0;
+                while (row != UINT32_MAX) {
+                    if (t.malive[row]) {
+                        if (t.mval_null[row]) {
+                            any_null = true;
+                        } else {
+                            long long v2 = t.mval[row];
+                            if (!any) best = v2;
+                            else if (c.kind == RW_AGG_MIN)
+                                best = v2 < best ? v2 : best;
+                            else
+                                best = v2 > best ? v2 : best;
+                            any = true;
+                        }
+                    }
+                    row = t.mnext[row];
+                }
+                long long pv;
+                uint8_t pn;
+                if (c.kind == RW_AGG_MAX && any_null) {
+                    pv = 0;
+                    pn = 1;
+                } else {
+                    pv = best;
+                    pn = !any;
+                }
+                if (set_prev) {
+                    t.prev[(size_t)ci * cap + slot] = pv;
+                    t.prev_null[(size_t)ci * cap + slot] = pn;
+                }
+                continue;
+            }
             long long v = vals[(size_t)ci * n + i];
             uint8_t nu = vnulls[(size_t)ci * n + i];
             long long init = 0;
@@ -1659,6 +1714,55 @@ __global__ void agg_restore_kernel(AggTableDev t, int KW, int n_calls,
             }
         }
         if (set_prev) t.has_prev[slot] = 1;
+    }
+}
+
+// minput-table restore: rebuild the materialized-input chains from decoded
+// rows (rw_stream.h restore contract)
+__global__ void agg_minput_restore_kernel(AggTableDev t, int KW, int mord,
+                                          const int64_t* gkeys,
+                                          const uint32_t* gnulls,
+                                          const long long* vals,
+                                          const uint8_t* vnulls,
+                                          const long long* sks,
+                                          const uint8_t* sknulls, uint32_t n) {
+    size_t cap = (size_t)t.cap_mask + 1;
+    uint32_t stride = gridDim.x * blockDim.x;
+    for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        int64_t kw[MAX_KW];
+        for (int k = 0; k < KW; k++) kw[k] = gkeys[(size_t)i * KW + k];
+        uint32_t slot = table_find_or_insert(t.state, t.keys, t.key_nulls,
+                                             t.cap_mask, kw, gnulls[i], KW);
+        if (slot == UINT32_MAX) {
+            atomicExch(&t.counters[2], 1u);
+            continue;
+        }
+        uint32_t row = atomicAdd(t.mcursor, 1u);
+        if (row >= t.mrow_cap) {
+            atomicExch(&t.counters[2], 3u);
+            continue;
+        }
+        st_u32(&t.mslot[row], slot);
+        t.mcall[row] = (uint8_t)mord;
+        st_i64((int64_t*)&t.mval[row], vals[i]);
+        t.mval_null[row] = vnulls[i];
+        for (int k = 0; k < t.n_sk; k++) {
+            st_i64((int64_t*)&t.msk[(size_t)k * t.mrow_cap + row],
+                   sks[(size_t)k * n + i]);
+            t.msk_null[(size_t)k * t.mrow_cap + row] =
+                sknulls[(size_t)k * n + i];
+        }
+        st_u32(&t.malive[row], 1);
+        uint32_t* headp = t.mheads + (size_t)mord * cap + slot;
+        uint32_t old_head = ld_u32(headp);
+        for (;;) {
+            st_u32(&t.mnext[row], old_head);
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            uint32_t prev = atomicCAS(headp, old_head, row);
+            if (prev == old_head) break;
+            old_head = prev;
+        }
     }
 }
 
@@ -1898,6 +2002,12 @@ struct HashAgg {
             HIP_TRY(hipMalloc(&t.malive, (size_t)t.mrow_cap * 4));
             HIP_TRY(hipMalloc(&t.mcursor, 4));
             HIP_TRY(hipMemset(t.mcursor, 0, 4));
+            HIP_TRY(hipMalloc(&t.mslot, (size_t)t.mrow_cap * 4));
+            HIP_TRY(hipMalloc(&t.mcall, t.mrow_cap));
+            t.mkilled_cap = 1u << 22;
+            HIP_TRY(hipMalloc(&t.mkilled, (size_t)t.mkilled_cap * 4));
+            HIP_TRY(hipMalloc(&t.mkilled_cursor, 4));
+            HIP_TRY(hipMemset(t.mkilled_cursor, 0, 4));
         }
         agg_init_kernel<<<2048, 256, 0, stream>>>(t, n_calls, cd(0), cd(1), cd(2),
                                                   cd(3));
@@ -2734,6 +2844,262 @@ struct HashAgg {
         return RW_OK;
     }
 
+    // ----- §8f-2 round 2: materialized-input state-TABLE spill. One
+    // table per retractable min/max call (AggStateStorage::
+    // MaterializedInput, test_utils/agg_executor.rs:63-121): pk =
+    // [group ASC ∥ value ASC(min)/DESC(max) ∥ stream key ASC], NULLs
+    // largest; value = the full row. Per-epoch deltas with mem-table
+    // netting: fresh alive rows → PUT, pre-drain rows killed this epoch →
+    // DELETE, created+died nets away. The drain copies the row arrays up
+    // to the cursor (host-marshal path; cost noted). -----
+    std::vector<uint32_t> mflush_mark, mkill_mark;
+
+    int minput_drain(int mi, std::vector<uint8_t>& out) {
+        if (mi < 0 || mi >= n_minput) FAIL(RW_E_INVAL, "minput table %d", mi);
+        HIP_TRY(hipStreamSynchronize(stream));
+        if ((int)mflush_mark.size() < n_minput) {
+            mflush_mark.assign(n_minput, 0);
+            mkill_mark.assign(n_minput, 0);
+        }
+        uint32_t cur = 0, kcur = 0;
+        HIP_TRY(hipMemcpy(&cur, t.mcursor, 4, hipMemcpyDeviceToHost));
+        HIP_TRY(hipMemcpy(&kcur, t.mkilled_cursor, 4, hipMemcpyDeviceToHost));
+        if (kcur > t.mkilled_cap)
+            FAIL(RW_E_INTERNAL, "minput kill list overflow (lost deltas)");
+        if (cur > t.mrow_cap) cur = t.mrow_cap;
+        size_t cap = (size_t)t.cap_mask + 1;
+        std::vector<long long> mval(cur);
+        std::vector<uint8_t> mnull(cur), mc(cur), alive4(cur * 4);
+        std::vector<uint32_t> mslot_h(cur), alive(cur);
+        std::vector<long long> sk((size_t)t.n_sk * cur);
+        std::vector<uint8_t> skn((size_t)t.n_sk * cur);
+        if (cur) {
+            HIP_TRY(hipMemcpy(mval.data(), t.mval, (size_t)cur * 8,
+                              hipMemcpyDeviceToHost));
+            HIP_TRY(hipMemcpy(mnull.data(), t.mval_null, cur,
+                              hipMemcpyDeviceToHost));
+            HIP_TRY(hipMemcpy(mc.data(), t.mcall, cur, hipMemcpyDeviceToHost));
+            HIP_TRY(hipMemcpy(mslot_h.data(), t.mslot, (size_t)cur * 4,
+                              hipMemcpyDeviceToHost));
+            HIP_TRY(hipMemcpy(alive.data(), t.malive, (size_t)cur * 4,
+                              hipMemcpyDeviceToHost));
+            for (int k = 0; k < t.n_sk; k++) {
+                HIP_TRY(hipMemcpy(sk.data() + (size_t)k * cur,
+                                  t.msk + (size_t)k * t.mrow_cap,
+                                  (size_t)cur * 8, hipMemcpyDeviceToHost));
+                HIP_TRY(hipMemcpy(skn.data() + (size_t)k * cur,
+                                  t.msk_null + (size_t)k * t.mrow_cap, cur,
+                                  hipMemcpyDeviceToHost));
+            }
+        }
+        std::vector<uint32_t> kills(kcur);
+        if (kcur)
+            HIP_TRY(hipMemcpy(kills.data(), t.mkilled, (size_t)kcur * 4,
+                              hipMemcpyDeviceToHost));
+        std::vector<int64_t> gkeys(cap * KW);
+        std::vector<uint32_t> gnulls(cap);
+        HIP_TRY(hipMemcpy(gkeys.data(), t.keys, cap * KW * 8,
+                          hipMemcpyDeviceToHost));
+        HIP_TRY(hipMemcpy(gnulls.data(), t.key_nulls, cap * 4,
+                          hipMemcpyDeviceToHost));
+        // ordinal -> (call index)
+        int ci = -1, o = -1;
+        for (size_t j = 0; j < calls.size(); j++) {
+            if (call_minput[j]) o++;
+            if (o == mi) {
+                ci = (int)j;
+                break;
+            }
+        }
+        auto encode = [&](uint32_t row, std::string* k,
+                          std::vector<uint8_t>* v) {
+            std::vector<uint8_t> kb;
+            uint32_t slot = mslot_h[row];
+            for (int i2 = 0; i2 < KW; i2++) {
+                rwcodec::DatumC d{((gnulls[slot] >> i2) & 1) != 0,
+                                  gkeys[(size_t)slot * KW + i2], 0};
+                rwcodec::memcmp_encode_datum(kb, out_types[i2], d, {});
+            }
+            rwcodec::DatumC dv{mnull[row] != 0, mval[row], 0};
+            rwcodec::memcmp_encode_datum(
+                kb, input_types[calls[ci].arg], dv,
+                {calls[ci].kind == RW_AGG_MAX, true});
+            for (int k2 = 0; k2 < t.n_sk; k2++) {
+                rwcodec::DatumC d{skn[(size_t)k2 * cur + row] != 0,
+                                  sk[(size_t)k2 * cur + row], 0};
+                rwcodec::memcmp_encode_datum(
+                    kb, input_types[stream_key[k2]], d, {});
+            }
+            k->assign((const char*)kb.data(), kb.size());
+            if (v) {
+                for (int i2 = 0; i2 < KW; i2++) {
+                    rwcodec::DatumC d{((gnulls[slot] >> i2) & 1) != 0,
+                                      gkeys[(size_t)slot * KW + i2], 0};
+                    rwcodec::value_encode_datum(*v, out_types[i2], d);
+                }
+                rwcodec::DatumC dv2{mnull[row] != 0, mval[row], 0};
+                rwcodec::value_encode_datum(*v, input_types[calls[ci].arg],
+                                            dv2);
+                for (int k2 = 0; k2 < t.n_sk; k2++) {
+                    rwcodec::DatumC d{skn[(size_t)k2 * cur + row] != 0,
+                                      sk[(size_t)k2 * cur + row], 0};
+                    rwcodec::value_encode_datum(
+                        *v, input_types[stream_key[k2]], d);
+                }
+            }
+        };
+        struct Rec {
+            std::string k;
+            std::vector<uint8_t> v;
+            uint8_t put;
+        };
+        std::vector<Rec> rs;
+        for (uint32_t row = mflush_mark[mi]; row < cur; row++) {
+            if (mc[row] != (uint8_t)mi || !alive[row]) continue;
+            Rec e;
+            e.put = 1;
+            encode(row, &e.k, &e.v);
+            rs.push_back(std::move(e));
+        }
+        for (uint32_t i2 = mkill_mark[mi]; i2 < kcur; i2++) {
+            uint32_t row = kills[i2];
+            if (row >= cur || mc[row] != (uint8_t)mi) continue;
+            if (row >= mflush_mark[mi]) continue; // created+died this epoch
+            Rec e;
+            e.put = 0;
+            encode(row, &e.k, nullptr);
+            rs.push_back(std::move(e));
+        }
+        std::stable_sort(rs.begin(), rs.end(),
+                         [](const Rec& a, const Rec& b) { return a.k < b.k; });
+        auto put32 = [&](uint32_t x) {
+            for (int b = 0; b < 4; b++) out.push_back((uint8_t)(x >> (8 * b)));
+        };
+        for (auto& e : rs) {
+            out.push_back(e.put);
+            put32((uint32_t)e.k.size());
+            out.insert(out.end(), e.k.begin(), e.k.end());
+            put32((uint32_t)e.v.size());
+            out.insert(out.end(), e.v.begin(), e.v.end());
+        }
+        mflush_mark[mi] = cur;
+        mkill_mark[mi] = kcur;
+        return RW_OK;
+    }
+
+    int minput_restore(int mi, const uint8_t* buf, uint64_t len) {
+        if (mi < 0 || mi >= n_minput) FAIL(RW_E_INVAL, "minput table %d", mi);
+        int ci = -1, o = -1;
+        for (size_t j = 0; j < calls.size(); j++) {
+            if (call_minput[j]) o++;
+            if (o == mi) {
+                ci = (int)j;
+                break;
+            }
+        }
+        std::map<std::string, std::vector<uint8_t>> merged;
+        bool ok = rwcodec::for_each_frame(
+            buf, len,
+            [&](uint8_t put, const uint8_t* k, uint32_t klen,
+                const uint8_t* v, uint32_t vlen) {
+                std::string key((const char*)k, klen);
+                if (put)
+                    merged[key].assign(v, v + vlen);
+                else
+                    merged.erase(key);
+            });
+        if (!ok) FAIL(RW_E_INVAL, "malformed minput spill stream");
+        uint32_t n = (uint32_t)merged.size();
+        if (!n) return RW_OK;
+        std::vector<int64_t> gk((size_t)n * KW);
+        std::vector<uint32_t> gn(n, 0);
+        std::vector<long long> mv(n);
+        std::vector<uint8_t> mn(n);
+        std::vector<long long> sks((size_t)t.n_sk * n, 0);
+        std::vector<uint8_t> skns((size_t)t.n_sk * n, 0);
+        uint32_t i2 = 0;
+        for (auto& [kb, val] : merged) {
+            (void)kb;
+            size_t off = 0;
+            auto rd = [&](uint8_t ty, long long* vv, uint8_t* nn) -> bool {
+                rwcodec::DatumC d;
+                size_t got = rwcodec::value_decode_datum(
+                    val.data() + off, val.size() - off, ty, &d);
+                if (!got) return false;
+                off += got;
+                *vv = d.null ? 0 : d.i;
+                *nn = d.null;
+                return true;
+            };
+            for (int k2 = 0; k2 < KW; k2++) {
+                long long vv;
+                uint8_t nn;
+                if (!rd(out_types[k2], &vv, &nn))
+                    FAIL(RW_E_INVAL, "minput restore: bad group datum");
+                gk[(size_t)i2 * KW + k2] = vv;
+                gn[i2] |= (uint32_t)(nn != 0) << k2;
+            }
+            if (!rd(input_types[calls[ci].arg], &mv[i2], &mn[i2]))
+                FAIL(RW_E_INVAL, "minput restore: bad value datum");
+            for (int k2 = 0; k2 < t.n_sk; k2++) {
+                long long vv;
+                uint8_t nn;
+                if (!rd(input_types[stream_key[k2]], &vv, &nn))
+                    FAIL(RW_E_INVAL, "minput restore: bad stream-key datum");
+                sks[(size_t)k2 * n + i2] = vv;
+                skns[(size_t)k2 * n + i2] = nn;
+            }
+            i2++;
+        }
+        int64_t* dgk = nullptr;
+        uint32_t* dgn = nullptr;
+        long long* dmv = nullptr;
+        uint8_t* dmn = nullptr;
+        long long* dsk = nullptr;
+        uint8_t* dskn = nullptr;
+        HIP_TRY(hipMalloc(&dgk, gk.size() * 8));
+        HIP_TRY(hipMalloc(&dgn, gn.size() * 4));
+        HIP_TRY(hipMalloc(&dmv, mv.size() * 8));
+        HIP_TRY(hipMalloc(&dmn, mn.size()));
+        HIP_TRY(hipMalloc(&dsk, sks.size() * 8 + 8));
+        HIP_TRY(hipMalloc(&dskn, skns.size() + 1));
+        HIP_TRY(hipMemcpy(dgk, gk.data(), gk.size() * 8,
+                          hipMemcpyHostToDevice));
+        HIP_TRY(hipMemcpy(dgn, gn.data(), gn.size() * 4,
+                          hipMemcpyHostToDevice));
+        HIP_TRY(hipMemcpy(dmv, mv.data(), mv.size() * 8,
+                          hipMemcpyHostToDevice));
+        HIP_TRY(hipMemcpy(dmn, mn.data(), mn.size(), hipMemcpyHostToDevice));
+        if (!sks.empty()) {
+            HIP_TRY(hipMemcpy(dsk, sks.data(), sks.size() * 8,
+                              hipMemcpyHostToDevice));
+            HIP_TRY(hipMemcpy(dskn, skns.data(), skns.size(),
+                              hipMemcpyHostToDevice));
+        }
+        agg_minput_restore_kernel<<<grid_for(n), 256, 0, stream>>>(
+            t, KW, mi, dgk, dgn, dmv, dmn, dsk, dskn, n);
+        int rc2 = hipStreamSynchronize(stream) == hipSuccess ? RW_OK
+                                                             : RW_E_INTERNAL;
+        hipFree(dgk);
+        hipFree(dgn);
+        hipFree(dmv);
+        hipFree(dmn);
+        hipFree(dsk);
+        hipFree(dskn);
+        if (rc2 != RW_OK) FAIL(RW_E_INTERNAL, "minput restore sync failed");
+        int rc3 = check_overflow();
+        if (rc3 != RW_OK) return rc3;
+        // restored rows predate the epoch
+        if ((int)mflush_mark.size() < n_minput) {
+            mflush_mark.assign(n_minput, 0);
+            mkill_mark.assign(n_minput, 0);
+        }
+        uint32_t cur = 0;
+        HIP_TRY(hipMemcpy(&cur, t.mcursor, 4, hipMemcpyDeviceToHost));
+        for (int j = 0; j < n_minput; j++) mflush_mark[j] = cur;
+        return RW_OK;
+    }
+
     // Host-side chunking with the U-pair no-split rule
     // (stream_chunk_builder.rs:188-218)
     void slice_outputs(const std::vector<long long>& vals,
@@ -2850,6 +3216,10 @@ struct HashAgg {
                 hipFree(t.mnext);
                 hipFree(t.malive);
                 hipFree(t.mcursor);
+                hipFree(t.mslot);
+                hipFree(t.mcall);
+                hipFree(t.mkilled);
+                hipFree(t.mkilled_cursor);
             }
             hipStreamDestroy(stream);
         }
@@ -3245,10 +3615,9 @@ int rw_agg_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len) {
 
 int rw_hash_agg_restore(void* h, const uint8_t* buf, uint64_t len) {
     auto* agg = (HashAgg*)h;
-    if (agg->n_minput > 0)
-        FAIL(RW_E_INVAL,
-             "restore with materialized-input aggregates requires "
-             "minput-table spill (not yet drained)");
+    // minput-bearing executors: rw_agg_minput_restore must run for every
+    // minput table BEFORE this call (prev outputs are recomputed from the
+    // hydrated chains, mirroring agg_group.rs:219-221)
     if (agg->n_dec > 0)
         FAIL(RW_E_INVAL, "decimal restore not yet implemented on GPU");
     std::map<std::string, std::vector<uint8_t>> merged;
@@ -3318,6 +3687,18 @@ int rw_hash_agg_restore(void* h, const uint8_t* buf, uint64_t len) {
     hipFree(dvnulls);
     if (rc != RW_OK) FAIL(RW_E_INTERNAL, "restore sync failed");
     return agg->check_overflow();
+}
+
+int rw_agg_n_minput_tables(void* h) { return ((HashAgg*)h)->n_minput; }
+int rw_agg_minput_drain(void* h, int mi, uint8_t** buf, uint64_t* len) {
+    auto* agg = (HashAgg*)h;
+    std::vector<uint8_t> sp;
+    int rc = agg->minput_drain(mi, sp);
+    if (rc != RW_OK) return rc;
+    return spill_export(sp, buf, len);
+}
+int rw_agg_minput_restore(void* h, int mi, const uint8_t* buf, uint64_t len) {
+    return ((HashAgg*)h)->minput_restore(mi, buf, len);
 }
 
 int rw_agg_n_dedup_tables(void* h) {

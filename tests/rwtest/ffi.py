@@ -724,3 +724,31 @@ def dec_col(values):
     """numpy V16 column from an iterable of dec16-able values."""
     raw = b"".join(dec16(v) for v in values)
     return np.frombuffer(raw, dtype="V16").copy()
+
+
+def agg_minput_drain_bytes(lib, h, mi):
+    """Drain one materialized-input state table; returns raw bytes."""
+    L = lib.lib
+    L.rw_agg_minput_drain.restype = C.c_int
+    L.rw_agg_minput_drain.argtypes = [C.c_void_p, C.c_int,
+                                      C.POINTER(C.POINTER(C.c_uint8)),
+                                      C.POINTER(C.c_uint64)]
+    buf = C.POINTER(C.c_uint8)()
+    ln = C.c_uint64()
+    rc = L.rw_agg_minput_drain(h, mi, C.byref(buf), C.byref(ln))
+    if rc != 0:
+        raise RuntimeError(f"minput drain failed {rc}: {lib.last_error()}")
+    out = bytes(bytearray(buf[i] for i in range(ln.value)))
+    L.rw_spill_free.argtypes = [C.c_void_p]
+    L.rw_spill_free(C.cast(buf, C.c_void_p))
+    return out
+
+
+def agg_minput_restore(lib, h, mi, buf):
+    L = lib.lib
+    L.rw_agg_minput_restore.restype = C.c_int
+    L.rw_agg_minput_restore.argtypes = [C.c_void_p, C.c_int, C.c_char_p,
+                                        C.c_uint64]
+    rc = L.rw_agg_minput_restore(h, mi, buf, len(buf))
+    if rc != 0:
+        raise RuntimeError(f"minput restore failed {rc}: {lib.last_error()}")

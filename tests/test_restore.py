@@ -212,3 +212,93 @@ def test_restore_gpu_matches_oracle():
         assert og == oo, f"epoch {e}: restored GPU != restored oracle"
     for lib, b in execs.values():
         b.close()
+
+
+def _agg_minput_restore_flow(lib):
+    # retractable min/max (materialized-input state TABLES, spilled per
+    # call): restore = minput tables FIRST, then the intermediate table
+    from rwtest.ffi import (AGG_MIN, agg_minput_drain_bytes,
+                            agg_minput_restore)
+
+    calls = [(ffi.AGG_MIN, 1, T_I64), (AGG_MAX, 1, T_I64),
+             (AGG_COUNT_STAR, -1, T_I64)]
+    # stream key = col 2 (unique row id) so minput rows have distinct pks
+    mk = lambda: ffi.HashAgg(lib, [T_I64, T_I64, T_I64], [0], calls, 2,
+                             stream_key=(2,))
+    a = mk()
+    rng = np.random.default_rng(55)
+    live = []
+    rid = [0]
+
+    def chunks():
+        out = []
+        for _ in range(3):
+            n = 384
+            g = rng.integers(0, 25, n)
+            v = rng.integers(-500, 500, n)
+            r2 = np.arange(rid[0], rid[0] + n)
+            rid[0] += n
+            ops = np.zeros(n, np.uint8)
+            for r in range(n):
+                if live and rng.random() < 0.3:
+                    j = int(rng.integers(0, len(live)))
+                    g[r], v[r], r2[r] = live.pop(j)
+                    ops[r] = ffi.OP_DELETE
+                else:
+                    live.append((int(g[r]), int(v[r]), int(r2[r])))
+            out.append(mk_chunk([T_I64, T_I64, T_I64], ops, [g, v, r2]))
+        return out
+
+    inter = b""
+    minp = [b"", b""]
+    for e in range(3):
+        _drive_agg(a, chunks(), e + 1)
+        inter += agg_checkpoint_drain_bytes(lib, a.h)
+        for mi in range(2):
+            minp[mi] += agg_minput_drain_bytes(lib, a.h, mi)
+    b = mk()
+    for mi in range(2):
+        agg_minput_restore(lib, b.h, mi, minp[mi])
+    agg_restore(lib, b.h, inter)
+    for e in range(3):
+        cs = chunks()
+        oa = _drive_agg(a, cs, 5 + e)
+        ob = _drive_agg(b, cs, 5 + e)
+        assert oa == ob, f"epoch {e}: restored minput agg diverged"
+        da = agg_checkpoint_drain_bytes(lib, a.h)
+        db = agg_checkpoint_drain_bytes(lib, b.h)
+        assert da == db, f"epoch {e}: intermediate drain diverged"
+        for mi in range(2):
+            ma = agg_minput_drain_bytes(lib, a.h, mi)
+            mb = agg_minput_drain_bytes(lib, b.h, mi)
+            assert ma == mb, f"epoch {e} minput {mi}: drain diverged"
+    a.close()
+    b.close()
+
+
+def test_agg_minput_restore_oracle():
+    _agg_minput_restore_flow(oracle())
+
+
+@pytest.mark.gpu
+def test_agg_minput_restore_gpu():
+    import risingwave_amd
+
+    risingwave_amd.load_library()
+    _agg_minput_restore_flow(ffi.Lib(risingwave_amd.lib_path()))
+
+
+def test_agg_minput_drain_parity_cpu_noop():
+    # sanity: a fresh executor drains empty minput tables
+    from rwtest.ffi import agg_minput_drain_bytes
+
+    a = ffi.HashAgg(oracle(), [T_I64, T_I64, T_I64], [0],
+                    [(ffi.AGG_MIN, 1, T_I64), (AGG_COUNT_STAR, -1, T_I64)],
+                    1, stream_key=(2,))
+    import ctypes
+
+    a.lib.lib.rw_agg_n_minput_tables.restype = ctypes.c_int
+    a.lib.lib.rw_agg_n_minput_tables.argtypes = [ctypes.c_void_p]
+    assert a.lib.lib.rw_agg_n_minput_tables(a.h) == 1
+    assert agg_minput_drain_bytes(oracle(), a.h, 0) == b""
+    a.close()
