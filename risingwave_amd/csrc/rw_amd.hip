@@ -7345,6 +7345,171 @@ __global__ void membw_rand_probe_kernel(const unsigned long long* __restrict__ p
     if (acc == 0xdeadbeefdeadbeefULL) *sink = acc;
 }
 
+// q8-SHAPE ladder: a stripped reproduction of the inner probe kernel's
+// per-row structure with stages toggled by `mode` bits, against synthetic
+// tables of the bench's sizes — pinpoints which stage carries the gap
+// between the ~30 us/1M-row random-access ceiling and the ~490 us kernel.
+// mode bits: 1 = wave-scan output reservation, 2 = emit writes (8 cols,
+// columnar), 4 = own-side insert (slot CAS + 64-B record write + link).
+__global__ void q8shape_kernel(const int64_t* __restrict__ k0,
+                               const int64_t* __restrict__ k1,
+                               const int64_t* __restrict__ k2,
+                               const uint64_t* __restrict__ mslots,
+                               uint32_t mcap_mask,
+                               const uint64_t* __restrict__ mrows,
+                               uint32_t mrow_mask, uint64_t* oslots,
+                               uint32_t ocap_mask, uint64_t* orows,
+                               uint32_t* ocursor, int64_t* out_vals,
+                               uint8_t* out_ops, uint32_t out_cap,
+                               uint32_t n, int mode, uint32_t* sink) {
+    uint32_t stride = gridDim.x * blockDim.x;
+    uint32_t iters = (n + stride - 1) / stride;
+    int lane = threadIdx.x & 63;
+    for (uint32_t it = 0; it < iters; it++) {
+        uint32_t r = it * stride + blockIdx.x * blockDim.x + threadIdx.x;
+        bool active = r < n;
+        int64_t kw[3] = {0, 0, 0};
+        uint64_t h = 0x20210401u;
+        if (active) {
+            kw[0] = k0[r];
+            kw[1] = k1[r];
+            kw[2] = k2[r];
+            for (int i = 0; i < 3; i++) h = mix64(h ^ (uint64_t)kw[i]);
+        }
+        uint32_t my_n = 0;
+        uint64_t mv0 = 0;
+        if (active) {
+            uint32_t slot = (uint32_t)(h >> (64 - __popc(mcap_mask)));
+            uint64_t packed = mslots[slot];
+            if ((uint32_t)packed & (1u << ((uint32_t)(h >> 32) & 31))) {
+                uint32_t head = (uint32_t)(packed >> 32) & mrow_mask;
+                // one 64-B record line: read all 8 words (key compare +
+                // payload, as the real walk does)
+                const uint64_t* rec = mrows + (size_t)head * 8;
+                uint64_t acc = 0;
+                for (int w = 0; w < 8; w++) acc += rec[w];
+                mv0 = acc;
+                my_n = 1;
+            }
+        }
+        uint32_t my_base = 0;
+        if (mode & 1) {
+            uint32_t incl = my_n;
+            for (int d = 1; d < 64; d <<= 1) {
+                uint32_t o = __shfl_up(incl, d);
+                if (lane >= d) incl += o;
+            }
+            uint32_t total = (uint32_t)__shfl((int)incl, 63);
+            uint32_t base = 0;
+            if (lane == 0 && total) base = atomicAdd(ocursor, total);
+            base = (uint32_t)__shfl((int)base, 0);
+            my_base = base + incl - my_n;
+        }
+        if ((mode & 2) && my_n) {
+            uint32_t orow = my_base & (out_cap - 1);
+            out_ops[orow] = 0;
+            for (int c = 0; c < 8; c++)
+                out_vals[(size_t)c * out_cap + orow] =
+                    (int64_t)(mv0 + kw[c % 3]);
+        }
+        if ((mode & 4) && active) {
+            uint32_t oslot = (uint32_t)(h >> (64 - __popc(ocap_mask)));
+            // record append at a wave-aggregated cursor position
+            uint64_t wmask = __ballot(true);
+            int leader = 63 - __clzll(wmask);
+            uint32_t base2 = 0;
+            if (lane == leader)
+                base2 = atomicAdd(ocursor + 16, (uint32_t)__popcll(wmask));
+            base2 = (uint32_t)__shfl((int)base2, leader);
+            uint32_t row =
+                (base2 + (uint32_t)__popcll(wmask & ((1ULL << lane) - 1))) &
+                mrow_mask;
+            uint64_t* rec = orows + (size_t)row * 8;
+            for (int w = 0; w < 8; w++) rec[w] = h + w;
+            // bucket push: CAS on the slot
+            uint64_t old = __hip_atomic_load(&oslots[oslot], RLX);
+            for (;;) {
+                rec[1] = (uint32_t)old ? (old >> 32) : 0xFFFFFFFFull;
+                asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+                uint64_t want = ((uint64_t)row << 32) |
+                                ((uint32_t)old | (1u << ((uint32_t)(h >> 32) & 31)));
+                uint64_t prev = atomicCAS((unsigned long long*)&oslots[oslot],
+                                          old, want);
+                if (prev == old) break;
+                old = prev;
+            }
+        }
+        if (active && mv0 == 0xdeadbeefdeadbeefULL) *sink = 1;
+    }
+}
+
+__global__ void q8shape_fill_kernel(int64_t* k0, int64_t* k1, int64_t* k2,
+                                    uint32_t n) {
+    uint32_t stride = gridDim.x * blockDim.x;
+    for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < n;
+         r += stride) {
+        k0[r] = (int64_t)mix64(r * 2654435761u + 1);
+        k1[r] = (int64_t)mix64(r * 40503u + 2);
+        k2[r] = (int64_t)mix64(r * 2246822519u + 3);
+    }
+}
+
+extern "C" int rw_q8shape_probe(uint32_t n_rows, int mode, double* us_out) {
+    // bench-sized synthetic state: 2^24-slot tables (134 MB), 16M-row
+    // record stores (1 GiB), q8-like 50%-hit blooms
+    uint32_t mcap = 1u << 24, rmask = (1u << 24) - 1;
+    size_t rbytes = (size_t)(rmask + 1) * 64;
+    int64_t *k0, *k1, *k2;
+    uint64_t *mslots, *mrows, *oslots, *orows;
+    uint32_t* ocursor;
+    int64_t* out_vals;
+    uint8_t* out_ops;
+    uint32_t out_cap = 1u << 22;
+    if (hipMalloc(&k0, (size_t)n_rows * 8) != hipSuccess) return RW_E_INTERNAL;
+    (void)hipMalloc(&k1, (size_t)n_rows * 8);
+    (void)hipMalloc(&k2, (size_t)n_rows * 8);
+    (void)hipMalloc(&mslots, (size_t)mcap * 8);
+    (void)hipMalloc(&mrows, rbytes);
+    (void)hipMalloc(&oslots, (size_t)mcap * 8);
+    (void)hipMalloc(&orows, rbytes);
+    (void)hipMalloc(&ocursor, 256);
+    (void)hipMalloc(&out_vals, (size_t)8 * out_cap * 8);
+    (void)hipMalloc(&out_ops, out_cap);
+    q8shape_fill_kernel<<<2048, 256>>>(k0, k1, k2, n_rows);
+    (void)hipMemset(mslots, 0xAB, (size_t)mcap * 8); // ~every bloom bit set
+    (void)hipMemset(mrows, 1, rbytes);
+    (void)hipMemset(oslots, 0, (size_t)mcap * 8);
+    (void)hipMemset(ocursor, 0, 256);
+    uint32_t* sink;
+    (void)hipMalloc(&sink, 4);
+    uint32_t blocks = (n_rows + 255) / 256;
+    if (blocks > 2048) blocks = 2048;
+    hipEvent_t e0, e1;
+    (void)hipEventCreate(&e0);
+    (void)hipEventCreate(&e1);
+    q8shape_kernel<<<blocks, 256>>>(k0, k1, k2, mslots, mcap - 1, mrows,
+                                    rmask, oslots, mcap - 1, orows, ocursor,
+                                    out_vals, out_ops, out_cap, n_rows, mode,
+                                    sink);
+    (void)hipEventRecord(e0);
+    for (int i = 0; i < 8; i++)
+        q8shape_kernel<<<blocks, 256>>>(k0, k1, k2, mslots, mcap - 1, mrows,
+                                        rmask, oslots, mcap - 1, orows,
+                                        ocursor, out_vals, out_ops, out_cap,
+                                        n_rows, mode, sink);
+    (void)hipEventRecord(e1);
+    if (hipEventSynchronize(e1) != hipSuccess) return RW_E_INTERNAL;
+    float ms = 0;
+    (void)hipEventElapsedTime(&ms, e0, e1);
+    *us_out = ms * 1000.0 / 8;
+    hipFree(k0); hipFree(k1); hipFree(k2); hipFree(mslots); hipFree(mrows);
+    hipFree(oslots); hipFree(orows); hipFree(ocursor); hipFree(out_vals);
+    hipFree(out_ops); hipFree(sink);
+    (void)hipEventDestroy(e0);
+    (void)hipEventDestroy(e1);
+    return RW_OK;
+}
+
 // random atomicAdd throughput (the insert path's slot-CAS analogue)
 __global__ void membw_rand_atomic_kernel(unsigned long long* p, size_t n_lines,
                                          int steps_per_thread) {

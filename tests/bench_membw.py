@@ -70,3 +70,15 @@ for mlp in (1, 2, 4, 8):
 print(json.dumps({"membw_probe": out, "membw_random_64B": rand,
                   "membw_random_atomic": at,
                   "membw_dependent_chase_1GiB": ch}))
+
+# q8-shape ladder (see rw_q8shape_probe)
+L.rw_q8shape_probe.restype = ctypes.c_int
+L.rw_q8shape_probe.argtypes = [ctypes.c_uint32, ctypes.c_int,
+                               ctypes.POINTER(ctypes.c_double)]
+shape = []
+for mode in (0, 1, 3, 5, 7):
+    us = ctypes.c_double(0.0)
+    rc = L.rw_q8shape_probe(1 << 20, mode, ctypes.byref(us))
+    assert rc == 0
+    shape.append({"mode": mode, "us_per_1M_rows": round(us.value, 1)})
+print(json.dumps({"q8_shape_ladder": shape}))
