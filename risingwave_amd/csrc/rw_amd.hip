@@ -4364,18 +4364,34 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
         }
 
         if (!m.append_only) {
-            // wave-aggregated output reservation
-            uint32_t incl = my_n;
-            for (int d = 1; d < 64; d <<= 1) {
-                uint32_t o = __shfl_up(incl, d);
-                if (lane >= d) incl += o;
-            }
-            uint32_t total = (uint32_t)__shfl((int)incl, 63);
+            // wave-aggregated output reservation. FAST PATH: when every
+            // lane matched <=1 row (the dominant hash-join case), one
+            // ballot + popcount replaces the 6-round shuffle prefix scan —
+            // the synthetic q8-shape ladder measured the scan chain at
+            // +173 us per 1M rows vs +~0 for the ballot form
+            // (profiles/r02_membw_probes.json q8_shape_ladder).
+            uint64_t multi = __ballot(my_n > 1);
+            uint32_t total, my_base;
             uint32_t base = 0;
-            if (lane == 0 && total)
-                base = atomicAdd(&out.counters[0], total);
-            base = (uint32_t)__shfl((int)base, 0);
-            uint32_t my_base = base + incl - my_n;
+            if (!multi) {
+                uint64_t got = __ballot(my_n != 0);
+                total = (uint32_t)__popcll(got);
+                if (lane == 0 && total)
+                    base = atomicAdd(&out.counters[0], total);
+                base = (uint32_t)__shfl((int)base, 0);
+                my_base = base + (uint32_t)__popcll(got & ((1ULL << lane) - 1));
+            } else {
+                uint32_t incl = my_n;
+                for (int d = 1; d < 64; d <<= 1) {
+                    uint32_t o = __shfl_up(incl, d);
+                    if (lane >= d) incl += o;
+                }
+                total = (uint32_t)__shfl((int)incl, 63);
+                if (lane == 0 && total)
+                    base = atomicAdd(&out.counters[0], total);
+                base = (uint32_t)__shfl((int)base, 0);
+                my_base = base + incl - my_n;
+            }
             if (total && base + total > out.cap) {
                 if (lane == 0) atomicExch(&out.counters[1], 1u); // overflow
             } else if (my_n && !(dbg_skip & 1)) {
@@ -4748,16 +4764,27 @@ __global__ __launch_bounds__(256, 8) void jpart_probe_insert_kernel(
         }
         // wave-aggregated output reservation + emit (multiset parity; the
         // reference's intra-epoch order is nondeterministic)
-        uint32_t incl = my_n;
-        for (int d = 1; d < 64; d <<= 1) {
-            uint32_t o = __shfl_up(incl, d);
-            if (lane >= d) incl += o;
-        }
-        uint32_t total = (uint32_t)__shfl((int)incl, 63);
+        // ballot fast path for my_n<=1 (see join_probe_kernel)
+        uint64_t multi = __ballot(my_n > 1);
+        uint32_t total, my_base;
         uint32_t base = 0;
-        if (lane == 0 && total) base = atomicAdd(&out.counters[0], total);
-        base = (uint32_t)__shfl((int)base, 0);
-        uint32_t my_base = base + incl - my_n;
+        if (!multi) {
+            uint64_t got = __ballot(my_n != 0);
+            total = (uint32_t)__popcll(got);
+            if (lane == 0 && total) base = atomicAdd(&out.counters[0], total);
+            base = (uint32_t)__shfl((int)base, 0);
+            my_base = base + (uint32_t)__popcll(got & ((1ULL << lane) - 1));
+        } else {
+            uint32_t incl = my_n;
+            for (int d = 1; d < 64; d <<= 1) {
+                uint32_t o = __shfl_up(incl, d);
+                if (lane >= d) incl += o;
+            }
+            total = (uint32_t)__shfl((int)incl, 63);
+            if (lane == 0 && total) base = atomicAdd(&out.counters[0], total);
+            base = (uint32_t)__shfl((int)base, 0);
+            my_base = base + incl - my_n;
+        }
         if (total && base + total > out.cap) {
             if (lane == 0) atomicExch(&out.counters[1], 1u);
         } else if (my_n) {
@@ -4885,16 +4912,27 @@ __global__ __launch_bounds__(256) void jpart_probe_insert_lds_kernel(
                 mr = mh->next;
             }
         }
-        uint32_t incl = my_n;
-        for (int d = 1; d < 64; d <<= 1) {
-            uint32_t o = __shfl_up(incl, d);
-            if (lane >= d) incl += o;
-        }
-        uint32_t total = (uint32_t)__shfl((int)incl, 63);
+        // ballot fast path for my_n<=1 (see join_probe_kernel)
+        uint64_t multi = __ballot(my_n > 1);
+        uint32_t total, my_base;
         uint32_t base = 0;
-        if (lane == 0 && total) base = atomicAdd(&out.counters[0], total);
-        base = (uint32_t)__shfl((int)base, 0);
-        uint32_t my_base = base + incl - my_n;
+        if (!multi) {
+            uint64_t got = __ballot(my_n != 0);
+            total = (uint32_t)__popcll(got);
+            if (lane == 0 && total) base = atomicAdd(&out.counters[0], total);
+            base = (uint32_t)__shfl((int)base, 0);
+            my_base = base + (uint32_t)__popcll(got & ((1ULL << lane) - 1));
+        } else {
+            uint32_t incl = my_n;
+            for (int d = 1; d < 64; d <<= 1) {
+                uint32_t o = __shfl_up(incl, d);
+                if (lane >= d) incl += o;
+            }
+            total = (uint32_t)__shfl((int)incl, 63);
+            if (lane == 0 && total) base = atomicAdd(&out.counters[0], total);
+            base = (uint32_t)__shfl((int)base, 0);
+            my_base = base + incl - my_n;
+        }
         if (total && base + total > out.cap) {
             if (lane == 0) atomicExch(&out.counters[1], 1u);
         } else if (my_n) {
