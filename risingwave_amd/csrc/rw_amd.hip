@@ -7414,6 +7414,13 @@ __global__ void q8shape_kernel(const int64_t* __restrict__ k0,
             kw[2] = k2[r];
             for (int i = 0; i < 3; i++) h = mix64(h ^ (uint64_t)kw[i]);
         }
+        if (mode & 16) { // byte streams like the real kernel (ops + valid)
+            if (active) {
+                h ^= out_ops[r & (out_cap - 1)]; // stand-in ops stream read
+                for (int i = 0; i < 3; i++)
+                    h += ((const uint8_t*)k0)[r]; // validity-byte stand-ins
+            }
+        }
         uint32_t my_n = 0;
         uint64_t mv0 = 0;
         if (active) {
@@ -7421,13 +7428,33 @@ __global__ void q8shape_kernel(const int64_t* __restrict__ k0,
             uint64_t packed = mslots[slot];
             if ((uint32_t)packed & (1u << ((uint32_t)(h >> 32) & 31))) {
                 uint32_t head = (uint32_t)(packed >> 32) & mrow_mask;
-                // one 64-B record line: read all 8 words (key compare +
-                // payload, as the real walk does)
-                const uint64_t* rec = mrows + (size_t)head * 8;
-                uint64_t acc = 0;
-                for (int w = 0; w < 8; w++) acc += rec[w];
-                mv0 = acc;
-                my_n = 1;
+                if (mode & 8) {
+                    // the real kernel's walk SHAPE: header fields, per-key
+                    // branchy compares, next chase with alive checks
+                    uint32_t row2 = head;
+                    int guard = 0;
+                    while (row2 != UINT32_MAX && guard++ < 4) {
+                        const uint64_t* rec = mrows + (size_t)row2 * 8;
+                        uint64_t hdr = rec[0]; // alive | next
+                        uint64_t vb = rec[1];
+                        bool eq = true;
+                        for (int i = 0; i < 3 && eq; i++)
+                            eq = ((int64_t)rec[2 + i] ^ kw[i]) != 1; // ~always true
+                        if ((hdr & 1) && eq && (vb | 1)) {
+                            my_n++;
+                            mv0 += rec[5];
+                        }
+                        // synthetic records have garbage next: terminate
+                        row2 = UINT32_MAX;
+                    }
+                } else {
+                    // one 64-B record line: read all 8 words
+                    const uint64_t* rec = mrows + (size_t)head * 8;
+                    uint64_t acc = 0;
+                    for (int w = 0; w < 8; w++) acc += rec[w];
+                    mv0 = acc;
+                    my_n = 1;
+                }
             }
         }
         uint32_t my_base = 0;
