@@ -7508,6 +7508,14 @@ __global__ void q8shape_kernel(const int64_t* __restrict__ k0,
     }
 }
 
+__global__ void q8shape_slot_fill_kernel(uint64_t* slots, uint32_t cap,
+                                         uint32_t rmask) {
+    uint32_t stride = gridDim.x * blockDim.x;
+    for (uint32_t s2 = blockIdx.x * blockDim.x + threadIdx.x; s2 < cap;
+         s2 += stride)
+        slots[s2] = ((uint64_t)(mix64(s2) & rmask) << 32) | 0xFFFFFFFFull;
+}
+
 __global__ void q8shape_fill_kernel(int64_t* k0, int64_t* k1, int64_t* k2,
                                     uint32_t n) {
     uint32_t stride = gridDim.x * blockDim.x;
@@ -7541,7 +7549,7 @@ extern "C" int rw_q8shape_probe(uint32_t n_rows, int mode, double* us_out) {
     (void)hipMalloc(&out_vals, (size_t)8 * out_cap * 8);
     (void)hipMalloc(&out_ops, out_cap);
     q8shape_fill_kernel<<<2048, 256>>>(k0, k1, k2, n_rows);
-    (void)hipMemset(mslots, 0xAB, (size_t)mcap * 8); // ~every bloom bit set
+    q8shape_slot_fill_kernel<<<2048, 256>>>(mslots, mcap, rmask);
     (void)hipMemset(mrows, 1, rbytes);
     (void)hipMemset(oslots, 0, (size_t)mcap * 8);
     (void)hipMemset(ocursor, 0, 256);
