@@ -7470,6 +7470,17 @@ __global__ void q8shape_kernel(const int64_t* __restrict__ k0,
             base = (uint32_t)__shfl((int)base, 0);
             my_base = base + incl - my_n;
         }
+        if (mode & 64) { // ballot-form reservation (1 ballot + popcounts)
+            uint64_t got = __ballot(my_n != 0);
+            uint32_t total = (uint32_t)__popcll(got);
+            uint32_t base = 0;
+            if (lane == 0 && total) base = atomicAdd(ocursor, total);
+            base = (uint32_t)__shfl((int)base, 0);
+            my_base = base + (uint32_t)__popcll(got & ((1ULL << lane) - 1));
+        }
+        if (mode & 128) { // pre-assigned output slot: NO cross-lane ops
+            my_base = r;
+        }
         if ((mode & 2) && my_n) {
             uint32_t orow = my_base & (out_cap - 1);
             out_ops[orow] = 0;

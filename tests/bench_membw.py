@@ -76,7 +76,7 @@ L.rw_q8shape_probe.restype = ctypes.c_int
 L.rw_q8shape_probe.argtypes = [ctypes.c_uint32, ctypes.c_int,
                                ctypes.POINTER(ctypes.c_double)]
 shape = []
-for mode in (0, 1, 3, 5, 7, 8, 9, 24, 25, 31):
+for mode in (0, 8, 9, 25, 24 + 64, 24 + 64 + 2, 24 + 128, 24 + 128 + 2, 31, 24 + 128 + 2 + 4):
     us = ctypes.c_double(0.0)
     rc = L.rw_q8shape_probe(1 << 20, mode, ctypes.byref(us))
     assert rc == 0
