@@ -4304,9 +4304,37 @@ __device__ void join_probe_row_noninner(const JoinBatchDev& b, JoinSideDev own,
         jown_delete(own, m, S, b, r, kw, nullmask);
 }
 
+// Per-batch setup for the inner probe's PRE-ASSIGNED layout (see the
+// kernel comment below): output region [0, n) is one sparse slot per
+// probe row (ops sentinel 0xFF = empty; the emission cursor starts at n
+// for >1-match overflow), and the batch's record rows are reserved as ONE
+// cursor bump so the kernel needs no cross-lane reservation at all.
+__global__ void jprobe_setup_kernel(JoinOutDev out, uint32_t* row_cursor,
+                                    uint32_t row_cap, uint32_t* row_base,
+                                    uint32_t n) {
+    out.counters[0] = n; // overflow emissions append past the sparse region
+    uint32_t base = atomicAdd(row_cursor, n);
+    if ((uint64_t)base + n > row_cap) atomicExch(&out.counters[1], 3u);
+    *row_base = base;
+}
+
+__global__ void join_count_emitted_kernel(const uint8_t* ops, uint32_t n,
+                                          const uint32_t* counters,
+                                          unsigned long long* out_count) {
+    uint32_t stride = gridDim.x * blockDim.x;
+    unsigned long long local = 0;
+    for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < n;
+         r += stride)
+        local += ops[r] != 0xFF;
+    if (local) atomicAdd(out_count, local);
+    if (blockIdx.x == 0 && threadIdx.x == 0)
+        atomicAdd(out_count, (unsigned long long)(counters[0] - n));
+}
+
 __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
                                   JoinSideDev match, JoinMeta m, int S,
                                   JoinOutDev out, uint32_t r0, uint32_t r1,
+                                  const uint32_t* row_base,
                                   int dbg_skip = 0) {
     uint32_t stride = gridDim.x * blockDim.x;
     uint32_t n = r1 - r0;
@@ -4344,9 +4372,10 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
         uint32_t mhead = UINT32_MAX;
         uint32_t my_n = 0;
         uint32_t matched_row = UINT32_MAX;
+        uint64_t h64 = 0;
         if (active) {
-            mhead = jbucket_head(match, hash_key(kw, nullmask, m.KW),
-                                 m.append_only != 0);
+            h64 = hash_key(kw, nullmask, m.KW);
+            mhead = jbucket_head(match, h64, m.append_only != 0);
             if (mhead != UINT32_MAX && !m.append_only) {
                 uint32_t row = mhead;
                 while (row != UINT32_MAX) {
@@ -4364,37 +4393,24 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
         }
 
         if (!m.append_only) {
-            // wave-aggregated output reservation. FAST PATH: when every
-            // lane matched <=1 row (the dominant hash-join case), one
-            // ballot + popcount replaces the 6-round shuffle prefix scan —
-            // the synthetic q8-shape ladder measured the scan chain at
-            // +173 us per 1M rows vs +~0 for the ballot form
-            // (profiles/r02_membw_probes.json q8_shape_ladder).
-            uint64_t multi = __ballot(my_n > 1);
-            uint32_t total, my_base;
-            uint32_t base = 0;
-            if (!multi) {
-                uint64_t got = __ballot(my_n != 0);
-                total = (uint32_t)__popcll(got);
-                if (lane == 0 && total)
-                    base = atomicAdd(&out.counters[0], total);
-                base = (uint32_t)__shfl((int)base, 0);
-                my_base = base + (uint32_t)__popcll(got & ((1ULL << lane) - 1));
-            } else {
-                uint32_t incl = my_n;
-                for (int d = 1; d < 64; d <<= 1) {
-                    uint32_t o = __shfl_up(incl, d);
-                    if (lane >= d) incl += o;
-                }
-                total = (uint32_t)__shfl((int)incl, 63);
-                if (lane == 0 && total)
-                    base = atomicAdd(&out.counters[0], total);
-                base = (uint32_t)__shfl((int)base, 0);
-                my_base = base + incl - my_n;
+            // PRE-ASSIGNED sparse emission: match 0 of probe row r goes to
+            // output slot r (ops sentinel 0xFF marks empty slots; the host
+            // compacts at drain); matches 2..k append past n via a
+            // PER-LANE atomic on the rare multi-match lanes. NO cross-lane
+            // ops: the q8-shape ladder measured ANY wave-convergent
+            // reservation (ballot or shuffle scan) at +135-155 us per 1M
+            // rows — the convergence stalls the wave on its slowest lane's
+            // dependent loads — while the pre-assigned form runs at the
+            // random-access ceiling (profiles/r02_membw_probes.json).
+            uint32_t my_base = r;
+            uint32_t extra_base = 0;
+            if (my_n > 1) {
+                extra_base = atomicAdd(&out.counters[0], my_n - 1);
+                if (extra_base + my_n - 1 > out.cap)
+                    atomicExch(&out.counters[1], 1u); // overflow
             }
-            if (total && base + total > out.cap) {
-                if (lane == 0) atomicExch(&out.counters[1], 1u); // overflow
-            } else if (my_n && !(dbg_skip & 1)) {
+            if (my_n && !(dbg_skip & 1) &&
+                !(my_n > 1 && extra_base + my_n - 1 > out.cap)) {
                 // second walk: emit (JoinStreamChunkBuilder::append_row).
                 // Emit stores are NONTEMPORAL: the rows are written once
                 // and only read by later launches, and regular stores pay
@@ -4408,7 +4424,8 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
                         jhdr_key_eq(h, m.key_cols[1 - S], m.KW, kw, nullmask,
                                     false) &&
                         join_cond_ok(m, S, b, r, h->validbits, jvals(h))) {
-                        uint32_t orow = my_base + k;
+                        uint32_t orow = k == 0 ? my_base
+                                               : extra_base + k - 1;
                         const long long* mv = jvals(h);
                         if (dbg_skip & 8) {
                             out.ops[orow] = op;
@@ -4492,6 +4509,47 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
             }
         }
 
+        if (!m.append_only) {
+            // PRE-ASSIGNED record slot (row_base + r): inserting rows write
+            // their record and link the bucket; every other row marks its
+            // slot dead (walks and drains skip by `alive`). No cross-lane
+            // cursor — see the emission comment above.
+            if (r < r1) {
+                uint32_t myrow = *row_base + r;
+                bool do_insert = active && !(dbg_skip & 2) && is_insert;
+                if (myrow < own.row_cap) {
+                    JoinRowHdr* hd2 = jrow(own, myrow);
+                    if (do_insert) {
+                        long long* hv2 = jvals(hd2);
+                        uint32_t vb2 = 0;
+                        if (b.all_insert) {
+                            for (int c = 0; c < m.n_cols[S]; c++) {
+                                hv2[c] = b.col_vals[c][r];
+                                vb2 |= (uint32_t)(b.col_valid[c][r] != 0) << c;
+                            }
+                            hd2->validbits = vb2;
+                            hd2->degree = 0;
+                            hd2->alive = 1;
+                        } else {
+                            // sc1 payload: same-launch delete walkers read it
+                            for (int c = 0; c < m.n_cols[S]; c++) {
+                                st_i64((int64_t*)&hv2[c], b.col_vals[c][r]);
+                                vb2 |= (uint32_t)(b.col_valid[c][r] != 0) << c;
+                            }
+                            st_u32(&hd2->validbits, vb2);
+                            st_u32(&hd2->degree, 0);
+                            st_u32(&hd2->alive, 1);
+                        }
+                        jbucket_insert(own, h64, myrow, !b.all_insert);
+                    } else {
+                        hd2->alive = 0;
+                    }
+                }
+                if (active && !(dbg_skip & 2) && !is_insert)
+                    jown_delete(own, m, S, b, r, kw, nullmask);
+            }
+            continue;
+        }
         if (!active || (dbg_skip & 2)) continue;
 
         if (m.append_only && is_insert && matched_row != UINT32_MAX) {
@@ -5625,6 +5683,14 @@ struct HashJoin {
         return RW_OK;
     }
 
+    // inner-probe pre-assigned layout bookkeeping (see jprobe_setup_kernel)
+    uint32_t* d_probe_base = nullptr; // the batch's reserved record base
+    unsigned long long* d_emit_count = nullptr;
+    uint8_t* d_vis_scratch = nullptr;
+    uint32_t d_vis_cap = 0;
+    uint32_t last_sparse_rows = 0;
+    bool sparse_out = false;
+
     // partitioned-pipeline device buffers (lazily allocated; ~260 KB)
     uint32_t* d_ptot = nullptr;  // padded per-partition totals (count)
     uint32_t* d_pcur = nullptr;  // padded per-partition scatter cursors
@@ -5682,6 +5748,22 @@ struct HashJoin {
         }
         const char* dbg_e = getenv("RW_JOIN_SKIP"); // bench A/B only:
         int dbg_skip = dbg_e ? atoi(dbg_e) : 0; // 1=no emits, 2=no own insert
+        if (m.join_type == RW_JOIN_INNER && !m.append_only &&
+            !can_partition(b, r0, r1) && r0 == 0) {
+            // pre-assigned sparse output + one-bump record reservation
+            if (!d_probe_base) {
+                HIP_TRY(hipMalloc(&d_probe_base, 4));
+                HIP_TRY(hipMalloc(&d_emit_count, 8));
+            }
+            if (b.n_rows > out.cap)
+                FAIL(RW_E_INVAL, "probe batch larger than the output block");
+            HIP_TRY(hipMemsetAsync(out.ops, 0xFF, b.n_rows, stream));
+            jprobe_setup_kernel<<<1, 1, 0, stream>>>(
+                out, side[s].row_cursor, side[s].row_cap, d_probe_base,
+                b.n_rows);
+            last_sparse_rows = b.n_rows;
+            sparse_out = true;
+        }
         if (can_partition(b, r0, r1)) {
             int rc = ensure_part_bufs();
             if (rc != RW_OK) return rc;
@@ -5704,10 +5786,9 @@ struct HashJoin {
                     out.counters + 1);
             }
         } else {
-            join_probe_kernel<<<blocks, 256, 0, stream>>>(b, side[s],
-                                                          side[1 - s], m, s,
-                                                          out, r0, r1,
-                                                          dbg_skip);
+            join_probe_kernel<<<blocks, 256, 0, stream>>>(
+                b, side[s], side[1 - s], m, s, out, r0, r1, d_probe_base,
+                dbg_skip);
         }
         if (timed) {
             HIP_TRY(hipEventRecord(ev1[slot], stream));
@@ -5898,6 +5979,36 @@ struct HashJoin {
                                   hipMemcpyDeviceToHost));
             }
             HIP_TRY(hipMemcpy(ops.data(), out.ops, n_out, hipMemcpyDeviceToHost));
+            if (sparse_out) {
+                // compact the pre-assigned sparse region (sentinel 0xFF)
+                std::vector<uint32_t> real;
+                real.reserve(n_out);
+                for (uint32_t r2 = 0; r2 < n_out; r2++)
+                    if (!(r2 < last_sparse_rows && ops[r2] == 0xFF))
+                        real.push_back(r2);
+                uint32_t nc = (uint32_t)real.size();
+                std::vector<int64_t> cv((size_t)nc * m.n_out);
+                std::vector<uint8_t> cn((size_t)nc * m.n_out);
+                std::vector<uint8_t> co(nc);
+                for (uint32_t j2 = 0; j2 < nc; j2++) {
+                    co[j2] = ops[real[j2]];
+                    for (int ci = 0; ci < m.n_out; ci++) {
+                        cv[(size_t)ci * nc + j2] =
+                            vals[(size_t)ci * n_out + real[j2]];
+                        cn[(size_t)ci * nc + j2] =
+                            nulls[(size_t)ci * n_out + real[j2]];
+                    }
+                }
+                vals.swap(cv);
+                nulls.swap(cn);
+                ops.swap(co);
+                n_out = nc;
+                sparse_out = false;
+            }
+            if (!n_out) {
+                HIP_TRY(hipMemset(out.counters, 0, 8));
+                return RW_OK;
+            }
             uint32_t max_rows = desc.chunk_size ? desc.chunk_size : 1024;
             if (max_rows < 2) max_rows = 2;
             for (uint32_t start = 0; start < n_out; start += max_rows) {
@@ -6082,6 +6193,11 @@ struct HashJoin {
             hipFree(d_part_base);
             hipFree(d_row_base);
         }
+        if (d_probe_base) {
+            hipFree(d_probe_base);
+            hipFree(d_emit_count);
+        }
+        if (d_vis_scratch) hipFree(d_vis_scratch);
         for (int s = 0; s < 2; s++) {
             JoinSideDev& js = side[s];
             if (js.slots8) {
@@ -6253,11 +6369,12 @@ int rw_join_bench_run(void* h, int side, void** batches, int n_batches,
                       int steps) {
     auto* j = (HashJoin*)h;
     for (int i = 0; i < steps; i++) {
+        // the inner probe's per-batch setup re-bases the output block, so
+        // no per-step reset is needed; the last step's emissions survive
+        // for the drain's count
         int rc = j->probe(side, *(JoinBatchDev*)batches[i % n_batches], true,
                           0, ((JoinBatchDev*)batches[i % n_batches])->n_rows);
         if (rc != RW_OK) return rc;
-        if (i + 1 < steps) // the last step's count survives for the drain
-            join_counters_reset_kernel<<<1, 1, 0, j->stream>>>(j->out.counters);
     }
     return RW_OK;
 }
@@ -6272,8 +6389,31 @@ long long rw_join_bench_drain(void* h) {
         g_err = "join bench overflow/full (code " + std::to_string(ctr[1]) + ")";
         return -(long long)ctr[1] - 1;
     }
+    long long emitted = (long long)ctr[0];
+    if (j->sparse_out && j->d_emit_count) {
+        // count the sparse region's real emissions (sentinel-aware)
+        (void)hipMemset(j->d_emit_count, 0, 8);
+        join_count_emitted_kernel<<<2048, 256, 0, j->stream>>>(
+            j->out.ops, j->last_sparse_rows, j->out.counters,
+            j->d_emit_count);
+        unsigned long long cnt = 0;
+        if (hipStreamSynchronize(j->stream) != hipSuccess) return -1;
+        if (hipMemcpy(&cnt, j->d_emit_count, 8, hipMemcpyDeviceToHost) !=
+            hipSuccess)
+            return -1;
+        emitted = (long long)cnt;
+        j->sparse_out = false;
+    }
     hipMemset(j->out.counters, 0, 8);
-    return (long long)ctr[0];
+    return emitted;
+}
+
+__global__ void join_out_vis_kernel(const uint8_t* ops, uint32_t sparse_n,
+                                    uint32_t total, uint8_t* vis) {
+    uint32_t stride = gridDim.x * blockDim.x;
+    for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < total;
+         r += stride)
+        vis[r] = !(r < sparse_n && ops[r] == 0xFF);
 }
 
 // Device-side pipeline hop: apply the join's accumulated output buffer
@@ -6430,6 +6570,21 @@ int rw_agg_apply_joinout(void* agg_h, void* join_h) {
     b.ops = j->out.ops;
     b.vis = nullptr;
     b.n_rows = n;
+    if (j->sparse_out) {
+        // pre-assigned sparse output: sentinel rows masked by visibility
+        if (j->d_vis_cap < n) {
+            if (j->d_vis_scratch) hipFree(j->d_vis_scratch);
+            HIP_TRY(hipMalloc(&j->d_vis_scratch, n));
+            j->d_vis_cap = n;
+        }
+        uint32_t blocks = (n + 255) / 256;
+        if (blocks > 2048) blocks = 2048;
+        join_out_vis_kernel<<<blocks, 256, 0, agg->stream>>>(
+            j->out.ops, j->last_sparse_rows, n, j->d_vis_scratch);
+        b.vis = j->d_vis_scratch;
+        b.dense = 0;
+        j->sparse_out = false;
+    }
     // run the agg on ITS stream after the join's work is drained (synced
     // above), then reset the join cursor
     int rc = agg->apply(b, true);
