@@ -4331,11 +4331,10 @@ __global__ void join_count_emitted_kernel(const uint8_t* ops, uint32_t n,
         atomicAdd(out_count, (unsigned long long)(counters[0] - n));
 }
 
-__global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
-                                  JoinSideDev match, JoinMeta m, int S,
-                                  JoinOutDev out, uint32_t r0, uint32_t r1,
-                                  const uint32_t* row_base,
-                                  int dbg_skip = 0) {
+__global__ __launch_bounds__(256, 8) void join_probe_kernel(
+    JoinBatchDev b, JoinSideDev own, JoinSideDev match, JoinMeta m, int S,
+    JoinOutDev out, uint32_t r0, uint32_t r1, const uint32_t* row_base,
+    int dbg_skip = 0) {
     uint32_t stride = gridDim.x * blockDim.x;
     uint32_t n = r1 - r0;
     uint32_t iters = (n + stride - 1) / stride;
@@ -4409,7 +4408,32 @@ __global__ void join_probe_kernel(JoinBatchDev b, JoinSideDev own,
                 if (extra_base + my_n - 1 > out.cap)
                     atomicExch(&out.counters[1], 1u); // overflow
             }
-            if (my_n && !(dbg_skip & 1) &&
+            if (my_n == 1 && !(dbg_skip & 1)) {
+                // SINGLE-PASS emission for the dominant <=1-match case:
+                // the count walk already identified the row — no re-walk
+                JoinRowHdr* h = jrow(match, matched_row);
+                uint32_t orow = my_base;
+                const long long* mv = jvals(h);
+                __builtin_nontemporal_store(op, &out.ops[orow]);
+                for (int c = 0; c < m.n_out; c++) {
+                    bool from_probe = (int)m.out_src[c] == S;
+                    uint8_t col = m.out_col[c];
+                    int64_t v;
+                    uint8_t valid;
+                    if (from_probe) {
+                        valid = b.col_valid[col][r];
+                        v = b.col_vals[col][r];
+                    } else {
+                        valid = (h->validbits >> col) & 1;
+                        v = mv[col];
+                    }
+                    __builtin_nontemporal_store(
+                        valid ? v : 0, &out.vals[(size_t)c * out.cap + orow]);
+                    __builtin_nontemporal_store(
+                        (uint8_t)!valid,
+                        &out.nulls[(size_t)c * out.cap + orow]);
+                }
+            } else if (my_n && !(dbg_skip & 1) &&
                 !(my_n > 1 && extra_base + my_n - 1 > out.cap)) {
                 // second walk: emit (JoinStreamChunkBuilder::append_row).
                 // Emit stores are NONTEMPORAL: the rows are written once
