@@ -446,9 +446,17 @@ __device__ __forceinline__ void jbucket_insert(const JoinSideDev& sd,
     for (;;) {
         uint32_t old_head =
             (uint32_t)old ? jhead_of(old) : UINT32_MAX; // bloom 0 = empty
-        if (sc1_next) st_u32(&hd->next, old_head);
-        else hd->next = old_head;
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); // R1 drain
+        if (sc1_next) {
+            st_u32(&hd->next, old_head);
+            // R1 drain: same-launch walkers (mixed-op chunks) must fetch
+            // real payload bytes once they see the new head. All-Insert
+            // launches have NO same-side walkers — their visibility comes
+            // from the inter-dispatch flush, so the pipeline-stalling
+            // drain is skipped (sc1_next == false).
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        } else {
+            hd->next = old_head;
+        }
         uint64_t want =
             ((uint64_t)row << 32) | ((uint32_t)old | jbloom_bit(h));
         uint64_t prev = atomicCAS((unsigned long long*)&sd.slots8[slot], old,
