@@ -1653,7 +1653,8 @@ __global__ void agg_restore_kernel(AggTableDev t, int KW, int n_calls,
                                    const int64_t* keys,
                                    const uint32_t* knulls,
                                    const long long* vals,
-                                   const uint8_t* vnulls, uint32_t n,
+                                   const uint8_t* vnulls,
+                                   const long long* vals2, uint32_t n,
                                    int set_prev, AggCallDev c0, AggCallDev c1,
                                    AggCallDev c2, AggCallDev c3) {
     AggCallDev calls[4] = {c0, c1, c2, c3};
@@ -1670,6 +1671,36 @@ __global__ void agg_restore_kernel(AggTableDev t, int KW, int n_calls,
             continue;
         }
         for (int ci = 0; ci < n_calls; ci++) {
+            if (calls[ci].decimal) {
+                // seed the exact-sum state from the stored decimal output
+                // (sum == output in the exact domain; oracle restore does
+                // the same)
+                long long w0 = vals[(size_t)ci * n + i];
+                long long w1 = vals2 ? vals2[(size_t)ci * n + i] : 0;
+                uint8_t nu = vnulls[(size_t)ci * n + i];
+                int dord = calls[ci].dord;
+                t.has[(size_t)ci * cap + slot] = !nu;
+                if (!nu) {
+                    DecValD dv = dec_parse_d(w0, w1);
+                    if (dv.special) {
+                        t.dspec[((size_t)dord * 3 + (dv.special - 1)) * cap +
+                                slot] = 1;
+                    } else {
+                        uint64_t w[4];
+                        dec_addend_d(dv, w);
+                        for (int k = 0; k < 4; k++)
+                            t.dsum[((size_t)dord * 4 + k) * cap + slot] =
+                                w[k];
+                        t.dscl[(size_t)dord * cap + slot] = dv.scale;
+                    }
+                }
+                if (set_prev) {
+                    t.prev[(size_t)ci * cap + slot] = w0;
+                    if (t.prev2) t.prev2[(size_t)ci * cap + slot] = w1;
+                    t.prev_null[(size_t)ci * cap + slot] = nu;
+                }
+                continue;
+            }
             if (calls[ci].minput) {
                 // prev output = first entry of the (already restored)
                 // minput chain (output_first, minput.rs:236-241; see
@@ -3626,8 +3657,6 @@ int rw_hash_agg_restore(void* h, const uint8_t* buf, uint64_t len) {
     // minput-bearing executors: rw_agg_minput_restore must run for every
     // minput table BEFORE this call (prev outputs are recomputed from the
     // hydrated chains, mirroring agg_group.rs:219-221)
-    if (agg->n_dec > 0)
-        FAIL(RW_E_INVAL, "decimal restore not yet implemented on GPU");
     std::map<std::string, std::vector<uint8_t>> merged;
     bool ok = rwcodec::for_each_frame(
         buf, len,
@@ -3647,6 +3676,8 @@ int rw_hash_agg_restore(void* h, const uint8_t* buf, uint64_t len) {
     std::vector<uint32_t> knulls(n, 0);
     std::vector<long long> vals((size_t)nc * n);
     std::vector<uint8_t> vnulls((size_t)nc * n);
+    std::vector<long long> vals2;
+    if (agg->n_dec) vals2.assign((size_t)nc * n, 0);
     uint32_t i = 0;
     for (auto& [kbytes, val] : merged) {
         (void)kbytes;
@@ -3664,6 +3695,7 @@ int rw_hash_agg_restore(void* h, const uint8_t* buf, uint64_t len) {
             } else {
                 vals[(size_t)(c - KW) * n + i] = d.i;
                 vnulls[(size_t)(c - KW) * n + i] = d.null;
+                if (agg->n_dec) vals2[(size_t)(c - KW) * n + i] = d.i2;
             }
         }
         i++;
@@ -3672,6 +3704,7 @@ int rw_hash_agg_restore(void* h, const uint8_t* buf, uint64_t len) {
     uint32_t* dknulls = nullptr;
     long long* dvals = nullptr;
     uint8_t* dvnulls = nullptr;
+    long long* dvals2 = nullptr;
     HIP_TRY(hipMalloc(&dkeys, keys.size() * 8));
     HIP_TRY(hipMalloc(&dknulls, knulls.size() * 4));
     HIP_TRY(hipMalloc(&dvals, vals.size() * 8));
@@ -3684,8 +3717,13 @@ int rw_hash_agg_restore(void* h, const uint8_t* buf, uint64_t len) {
                       hipMemcpyHostToDevice));
     HIP_TRY(hipMemcpy(dvnulls, vnulls.data(), vnulls.size(),
                       hipMemcpyHostToDevice));
+    if (agg->n_dec) {
+        HIP_TRY(hipMalloc(&dvals2, vals2.size() * 8));
+        HIP_TRY(hipMemcpy(dvals2, vals2.data(), vals2.size() * 8,
+                          hipMemcpyHostToDevice));
+    }
     agg_restore_kernel<<<agg->grid_for(n), 256, 0, agg->stream>>>(
-        agg->t, KW, nc, dkeys, dknulls, dvals, dvnulls, n,
+        agg->t, KW, nc, dkeys, dknulls, dvals, dvnulls, dvals2, n,
         agg->eowc ? 0 : 1, agg->cd(0), agg->cd(1), agg->cd(2), agg->cd(3));
     int rc = hipStreamSynchronize(agg->stream) == hipSuccess ? RW_OK
                                                              : RW_E_INTERNAL;
@@ -3693,6 +3731,7 @@ int rw_hash_agg_restore(void* h, const uint8_t* buf, uint64_t len) {
     hipFree(dknulls);
     hipFree(dvals);
     hipFree(dvnulls);
+    if (dvals2) hipFree(dvals2);
     if (rc != RW_OK) FAIL(RW_E_INTERNAL, "restore sync failed");
     return agg->check_overflow();
 }

@@ -194,3 +194,67 @@ def test_decimal_gpu_parity():
         assert dg == do, f"epoch {epoch}: spill bytes diverged"
     g.close()
     o.close()
+
+
+def _decimal_restore_flow(lib):
+    from rwtest.ffi import agg_checkpoint_drain_bytes, agg_restore
+
+    rng = np.random.default_rng(77)
+    a = mk_agg(lib)
+    live = []
+    drains = b""
+
+    def chunks():
+        out = []
+        for _ in range(2):
+            n = 256
+            gs, ds, ops = [], [], []
+            for r in range(n):
+                if live and rng.random() < 0.2:
+                    j = int(rng.integers(0, len(live)))
+                    g, d = live.pop(j)
+                    ops.append(ffi.OP_DELETE)
+                else:
+                    g = int(rng.integers(0, 15))
+                    d = _rand_dec(rng)
+                    live.append((g, d))
+                    ops.append(ffi.OP_INSERT)
+                gs.append(g)
+                ds.append(d)
+            out.append(mk_dec_chunk(gs, ds, ops))
+        return out
+
+    for e in range(3):
+        for c in chunks():
+            a.push(c)
+        a.flush(e + 1)
+        a.poll_all()
+        drains += agg_checkpoint_drain_bytes(lib, a.h)
+    b = mk_agg(lib)
+    agg_restore(lib, b.h, drains)
+    for e in range(3):
+        cs = chunks()
+        outs = []
+        for x in (a, b):
+            for c in cs:
+                x.push(c)
+            x.flush(5 + e)
+            outs.append(rows_multiset(x.poll_all()))
+        assert outs[0] == outs[1], f"epoch {e}: restored decimal agg diverged"
+        da = agg_checkpoint_drain_bytes(lib, a.h)
+        db = agg_checkpoint_drain_bytes(lib, b.h)
+        assert da == db, f"epoch {e}: decimal drain diverged"
+    a.close()
+    b.close()
+
+
+def test_decimal_restore_oracle():
+    _decimal_restore_flow(oracle())
+
+
+@pytest.mark.gpu
+def test_decimal_restore_gpu():
+    import risingwave_amd
+
+    risingwave_amd.load_library()
+    _decimal_restore_flow(ffi.Lib(risingwave_amd.lib_path()))
