@@ -1311,3 +1311,31 @@ def test_epoch_ingest_mode_parity():
         assert outs[0] == outs[1] == outs[2], f"epoch {epoch} diverged"
     for a in (a_epoch, a_plain, a_orc):
         a.close()
+
+
+def test_join_inner_multimatch_sparse_overflow():
+    # the pre-assigned sparse emission's EXTRAS path: probe rows matching
+    # MANY build rows (my_n up to 16) append past the sparse region via
+    # per-lane atomics; drain compaction must reassemble the multiset
+    rng = np.random.default_rng(91)
+    g, o = join_pair()
+    # build: 16 rows per key for 64 keys
+    keys = np.repeat(np.arange(64), 16)
+    vals = np.arange(len(keys))
+    c = mk_chunk([T_I64, T_I64], np.zeros(len(keys), np.uint8), [keys, vals])
+    g.push(SIDE_RIGHT, c)
+    o.push(SIDE_RIGHT, c)
+    g.poll_all()
+    o.poll_all()
+    # probe: hits interleaved with misses
+    n = 4096
+    pk = rng.integers(0, 128, n)  # half miss
+    pv = np.arange(10_000, 10_000 + n)
+    c = mk_chunk([T_I64, T_I64], np.zeros(n, np.uint8), [pk, pv])
+    g.push(SIDE_LEFT, c)
+    o.push(SIDE_LEFT, c)
+    mg = rows_multiset(g.poll_all())
+    mo = rows_multiset(o.poll_all())
+    assert mg == mo and len(mg) > 30_000, f"{len(mg)} vs {len(mo)}"
+    g.close()
+    o.close()
