@@ -182,6 +182,12 @@ enum RwJoinSide { RW_SIDE_LEFT = 0, RW_SIDE_RIGHT = 1 };
 void* rw_hash_join_create(const RwHashJoinDesc* desc);
 int rw_hash_join_push_chunk(void* h, int side, const RwChunk* chunk);
 int rw_hash_join_flush(void* h, uint64_t epoch); /* aligned barrier */
+/* Epoch-batched ingestion (the agg mode's analogue): consecutive
+ * SAME-SIDE chunks merge into one staged batch applied as one launch at
+ * the next side switch, watermark or barrier. Exactly order-equivalent:
+ * probes read only the match side, which a same-side run never mutates;
+ * own-side interleavings are the conflict-segment pre-pass's job. */
+int rw_hash_join_ingest_mode(void* h, int epoch_batched);
 RwChunk* rw_hash_join_poll(void* h);
 void rw_hash_join_destroy(void* h);
 

@@ -6195,8 +6195,76 @@ struct HashJoin {
         return bounds;
     }
 
+    // ----- epoch-batched ingestion (the agg's rw_hash_agg_ingest_mode
+    // analogue): consecutive SAME-SIDE chunks merge into one staged host
+    // batch applied as one launch at the next side switch / barrier /
+    // watermark. Exactly order-equivalent: probes read only the MATCH
+    // side (which a same-side run never mutates), own-side insert/delete
+    // interleavings within the merged batch are the conflict-segment
+    // pre-pass's existing job, and cross-SIDE boundaries always flush. -----
+    bool epoch_ingest = false;
+    int pend_side = -1;
+    std::vector<uint8_t> pend_ops;
+    std::vector<std::vector<int64_t>> pend_vals;  // per column (16-B decimal n/a)
+    std::vector<std::vector<uint8_t>> pend_valid;
+
+    int pending_flush() {
+        if (pend_side < 0 || pend_ops.empty()) {
+            pend_side = -1;
+            return RW_OK;
+        }
+        int s2 = pend_side;
+        uint32_t n = (uint32_t)pend_ops.size();
+        std::vector<RwColumn> cols(m.n_cols[s2]);
+        for (int c2 = 0; c2 < m.n_cols[s2]; c2++) {
+            cols[c2].type = types[s2][c2];
+            cols[c2].valid = pend_valid[c2].data();
+            cols[c2].data = pend_vals[c2].data();
+        }
+        RwChunk ch{};
+        ch.n_rows = n;
+        ch.n_cols = (uint32_t)cols.size();
+        ch.ops = pend_ops.data();
+        ch.vis = nullptr;
+        ch.cols = cols.data();
+        pend_side = -1;
+        int rc = push_chunk_now(s2, &ch);
+        pend_ops.clear();
+        for (auto& v : pend_vals) v.clear();
+        for (auto& v : pend_valid) v.clear();
+        return rc;
+    }
+
     int push_chunk(int s, const RwChunk* c) {
         if (s != 0 && s != 1) FAIL(RW_E_INVAL, "bad side");
+        if (epoch_ingest) {
+            if (c->vis)
+                FAIL(RW_E_INVAL,
+                     "epoch-batched ingest requires visibility-compacted "
+                     "chunks");
+            if (pend_side >= 0 && pend_side != s) {
+                int rc = pending_flush();
+                if (rc != RW_OK) return rc;
+            }
+            if (pend_side < 0) {
+                pend_side = s;
+                pend_vals.assign(m.n_cols[s], {});
+                pend_valid.assign(m.n_cols[s], {});
+            }
+            pend_ops.insert(pend_ops.end(), c->ops, c->ops + c->n_rows);
+            for (int c2 = 0; c2 < m.n_cols[s]; c2++) {
+                const int64_t* d = (const int64_t*)c->cols[c2].data;
+                pend_vals[c2].insert(pend_vals[c2].end(), d, d + c->n_rows);
+                pend_valid[c2].insert(pend_valid[c2].end(),
+                                      c->cols[c2].valid,
+                                      c->cols[c2].valid + c->n_rows);
+            }
+            return RW_OK;
+        }
+        return push_chunk_now(s, c);
+    }
+
+    int push_chunk_now(int s, const RwChunk* c) {
         JoinBatchDev b;
         int rc = upload(s, c, &b);
         if (rc != RW_OK) return rc;
@@ -6218,12 +6286,20 @@ struct HashJoin {
     }
 
     int flush(uint64_t) {
+        if (epoch_ingest) {
+            int rc = pending_flush();
+            if (rc != RW_OK) return rc;
+        }
         HIP_TRY(hipStreamSynchronize(stream));
         return RW_OK;
     }
 
     int watermark(int s, uint32_t col_idx, int64_t val, uint32_t* out_cols,
                   int64_t* out_vals, int max_out, int* n_out_p) {
+        if (epoch_ingest) {
+            int rcp = pending_flush();
+            if (rcp != RW_OK) return rcp;
+        }
         int n_out = 0;
         auto& mine = s == 0 ? wm_side0 : wm_side1;
         for (uint32_t idx = 0; idx < (uint32_t)m.KW; idx++) {
@@ -6330,6 +6406,16 @@ int rw_hash_join_push_chunk(void* h, int side, const RwChunk* c) {
     return ((HashJoin*)h)->push_chunk(side, c);
 }
 int rw_hash_join_flush(void* h, uint64_t epoch) { return ((HashJoin*)h)->flush(epoch); }
+
+// epoch-batched ingest mode (rw_stream.h): same-side chunk runs merge into
+// one launch; barriers/watermarks/side switches flush the staged run
+int rw_hash_join_ingest_mode(void* h, int epoch_batched) {
+    auto* j = (HashJoin*)h;
+    if (!epoch_batched && j->pend_side >= 0)
+        FAIL(RW_E_INVAL, "staged chunks pending; flush before disabling");
+    j->epoch_ingest = epoch_batched != 0;
+    return RW_OK;
+}
 
 int rw_hash_join_watermark(void* h, int side, uint32_t col_idx, int64_t val,
                            uint32_t* out_cols, int64_t* out_vals, int max_out) {

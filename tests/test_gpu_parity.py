@@ -1339,3 +1339,50 @@ def test_join_inner_multimatch_sparse_overflow():
     assert mg == mo and len(mg) > 30_000, f"{len(mg)} vs {len(mo)}"
     g.close()
     o.close()
+
+
+def test_join_epoch_ingest_mode_parity():
+    # rw_hash_join_ingest_mode: same-side chunk runs merged into one launch
+    # must equal the per-push path AND the oracle
+    import ctypes
+
+    rng = np.random.default_rng(61)
+    g_epoch, o = join_pair()
+    g_plain = ffi.HashJoin(gpu(), JOIN_INNER, [T_I64, T_I64],
+                           [T_I64, T_I64], key_l=[0], key_r=[0], pk_l=[1],
+                           pk_r=[1])
+    L = gpu().lib
+    L.rw_hash_join_ingest_mode.restype = ctypes.c_int
+    L.rw_hash_join_ingest_mode.argtypes = [ctypes.c_void_p, ctypes.c_int]
+    assert L.rw_hash_join_ingest_mode(g_epoch.h, 1) == 0, gpu().last_error()
+    live = {SIDE_LEFT: [], SIDE_RIGHT: []}
+    pk = 0
+    for epoch in range(3):
+        # runs of same-side chunks with inserts + deletes
+        for side in (SIDE_LEFT, SIDE_LEFT, SIDE_RIGHT, SIDE_RIGHT, SIDE_LEFT):
+            n = 768
+            keys = rng.integers(0, 120, n)
+            vals = np.arange(pk, pk + n)
+            pk += n
+            ops = np.zeros(n, np.uint8)
+            for r in range(n):
+                if live[side] and rng.random() < 0.25:
+                    j = int(rng.integers(0, len(live[side])))
+                    keys[r], vals[r] = live[side].pop(j)
+                    ops[r] = ffi.OP_DELETE
+                else:
+                    live[side].append((int(keys[r]), int(vals[r])))
+            c = mk_chunk([T_I64, T_I64], ops, [keys, vals])
+            for x in (g_epoch, g_plain, o):
+                x.push(side, c)
+        outs = []
+        for x in (g_epoch, g_plain, o):
+            x.flush(epoch + 1)
+            outs.append(rows_multiset(x.poll_all()))
+        assert outs[0] == outs[1] == outs[2], f"epoch {epoch} diverged"
+        d0 = ffi.join_checkpoint_drain(gpu(), g_epoch.h, SIDE_LEFT)
+        d1 = ffi.join_checkpoint_drain(gpu(), g_plain.h, SIDE_LEFT)
+        d2 = ffi.join_checkpoint_drain(ffi.oracle(), o.h, SIDE_LEFT)
+        assert d0 == d1 == d2, f"epoch {epoch}: left drains diverged"
+    for x in (g_epoch, g_plain, o):
+        x.close()
