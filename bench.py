@@ -655,41 +655,30 @@ def bench_q7pipe(args, ffi, gpu_lib, rng, rank, world, dist):
 
     cmap = (ctypes.c_uint32 * 2)(0, 1)  # agg record [w, max, count] -> [w, max]
     agg_ms = [0.0]
+    # the whole step loop runs in C (rw_q7pipe_bench_run): python-side
+    # ctypes dispatch costs ~30-40 us/step, ~20% at this step size
+    L.rw_q7pipe_bench_run.restype = ctypes.c_int
+    L.rw_q7pipe_bench_run.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.POINTER(ctypes.c_void_p),
+        ctypes.POINTER(ctypes.c_void_p), ctypes.c_int, ctypes.c_int,
+        ctypes.c_int, ctypes.POINTER(ctypes.c_uint32), ctypes.c_int,
+        ctypes.c_int]
+    a_arr = (ctypes.c_void_p * n_batches)(*agg_batches)
+    j_arr = (ctypes.c_void_p * n_batches)(*join_batches)
 
-    def step(i):
-        # exchange hop 1: vnode of the window key feeding the agg fragment
-        rc = L.rw_join_vnode_hop(j.h, join_batches[i % n_batches], 3,
-                                 ffi.T_I64, 256)
+    def run_steps(n, step0):
+        rc = L.rw_q7pipe_bench_run(agg.h, j.h, a_arr, j_arr, n_batches, n,
+                                   args.barrier_every, cmap, 2, step0)
         assert rc == 0, gpu_lib.last_error()
-        rc = L.rw_agg_bench_apply(agg.h, agg_batches[i % n_batches])
-        assert rc == 0, gpu_lib.last_error()
-        # exchange hop 2: vnode of price feeding the join-left fragment
-        rc = L.rw_join_vnode_hop(j.h, join_batches[i % n_batches], 2,
-                                 ffi.T_I64, 256)
-        assert rc == 0, gpu_lib.last_error()
-        rc = L.rw_join_bench_apply(j.h, SIDE_LEFT, join_batches[i % n_batches])
-        assert rc == 0, gpu_lib.last_error()
-        if (i + 1) % args.barrier_every == 0:
-            t0 = time.perf_counter()
-            n = L.rw_agg_flush_device(agg.h, i)
-            assert n >= 0, gpu_lib.last_error()
-            rc = L.rw_join_apply_aggout(j.h, agg.h, SIDE_RIGHT, cmap, 2,
-                                        ctypes.c_uint64(n))
-            assert rc == 0, gpu_lib.last_error()
-            agg_ms[0] += time.perf_counter() - t0
-            assert L.rw_join_bench_drain(j.h) >= 0, gpu_lib.last_error()
 
-    for i in range(args.warmup):
-        step(i)
+    run_steps(args.warmup, 0)
     L.rw_join_stats_reset(j.h)
     L.rw_agg_stats_reset(agg.h)
-    agg_ms[0] = 0.0
     if dist:
         dist.barrier()
     _dev_sync()
     t0 = time.perf_counter()
-    for i in range(args.steps):
-        step(i)
+    run_steps(args.steps, args.warmup)
     assert L.rw_join_bench_drain(j.h) >= 0, gpu_lib.last_error()
     _dev_sync()
     elapsed = time.perf_counter() - t0

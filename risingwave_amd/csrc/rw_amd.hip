@@ -6773,6 +6773,42 @@ int rw_agg_apply_joinout(void* agg_h, void* join_h) {
     return agg->check_overflow();
 }
 
+// C-side q7-pipeline step loop (the full q7 stream plan, bench.py
+// bench_q7pipe): per step, two vnode exchange hops + the agg apply + the
+// join-left probe; every barrier_every steps the agg flush's change
+// stream feeds the join's right side device-resident. The Python
+// interpreter costs ~30-40 us/step in ctypes dispatch — at a ~0.2 ms
+// step that is 20% overhead, so the loop lives here.
+int rw_q7pipe_bench_run(void* agg_h, void* join_h, void** agg_batches,
+                        void** join_batches, int n_batches, int steps,
+                        int barrier_every, const uint32_t* col_map,
+                        int n_map, int step0) {
+    auto* agg = (HashAgg*)agg_h;
+    auto* j = (HashJoin*)join_h;
+    for (int i = 0; i < steps; i++) {
+        int bi = (step0 + i) % n_batches;
+        auto* jb = (JoinBatchDev*)join_batches[bi];
+        int rc = rw_join_vnode_hop(join_h, jb, 3, RW_T_I64, 256);
+        if (rc != RW_OK) return rc;
+        rc = agg->apply(*(AggBatch*)agg_batches[bi], true);
+        if (rc != RW_OK) return rc;
+        rc = rw_join_vnode_hop(join_h, jb, 2, RW_T_I64, 256);
+        if (rc != RW_OK) return rc;
+        rc = j->probe(RW_SIDE_LEFT, *jb, true, 0, jb->n_rows);
+        if (rc != RW_OK) return rc;
+        if (barrier_every > 0 && (step0 + i + 1) % barrier_every == 0) {
+            long long n = rw_agg_flush_device(agg_h, (uint64_t)(step0 + i));
+            if (n < 0) return RW_E_INTERNAL;
+            rc = rw_join_apply_aggout(join_h, agg_h, RW_SIDE_RIGHT, col_map,
+                                      n_map, (uint64_t)n);
+            if (rc != RW_OK) return rc;
+            long long e = rw_join_bench_drain(join_h);
+            if (e < 0) return RW_E_INTERNAL;
+        }
+    }
+    return RW_OK;
+}
+
 int rw_join_kernel_stats(void* h, RwKernelStats* out) {
     auto* j = (HashJoin*)h;
     if (j->ev_harvest_all() != RW_OK)
