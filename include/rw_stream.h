@@ -299,6 +299,11 @@ int rw_hash_join_restore(void* h, int side, const uint8_t* buf, uint64_t len,
                          const uint8_t* deg_buf, uint64_t deg_len);
 int rw_join_checkpoint_drain(void* h, int side, uint8_t** buf, uint64_t* len);
 int rw_topn_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len);
+/* Replay concatenated rw_topn_checkpoint_drain streams into a freshly
+ * created GroupTopN executor (PUT/DELETE frames net host-side; the
+ * surviving full rows rebuild the per-group state, marked persisted so the
+ * next drain does not re-PUT them). Must run before any input. */
+int rw_topn_restore(void* h, const uint8_t* buf, uint64_t len);
 /* DISTINCT dedup tables (one StateTable per distinct column in the
  * reference, from_proto/hash_agg.rs distinct_dedup_tables): pk = group key
  * ∥ datum, value = full row ++ one i64 count per call distincting on the
@@ -308,6 +313,11 @@ int rw_topn_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len);
  * rw_agg_n_dedup_tables returns how many exist. */
 int rw_agg_n_dedup_tables(void* h);
 int rw_agg_dedup_drain(void* h, int di, uint8_t** buf, uint64_t* len);
+/* Replay concatenated rw_agg_dedup_drain streams for table `di` into a
+ * freshly created executor (before any input; independent of the
+ * intermediate-table restore order). Restored counts are persisted: a
+ * later drop to 0 drains as DELETE, exactly as uninterrupted. */
+int rw_agg_dedup_restore(void* h, int di, const uint8_t* buf, uint64_t len);
 /* Join degree tables (join/row.rs:99-113 build_degree_row): per side
  * needing degrees (join/mod.rs:153-165), pk = jk ∥ pk as the main table,
  * value = order key ++ degree i64. The deltas are computed by

@@ -769,6 +769,60 @@ struct HashAggOracle {
         return RW_OK;
     }
 
+    // §8f-5 recovery for one dedup table: net concatenated drain frames
+    // and seed the counter map; restored keys are persisted (a later drop
+    // to 0 drains as DELETE, exactly as uninterrupted).
+    int dedup_restore(int di, const uint8_t* buf, uint64_t len) {
+        if (di < 0 || (size_t)di >= distinct_cols.size()) return RW_E_INVAL;
+        std::map<std::string, std::vector<uint8_t>> merged;
+        bool ok = rwcodec::for_each_frame(
+            buf, len,
+            [&](uint8_t put, const uint8_t* k, uint32_t klen,
+                const uint8_t* v, uint32_t vlen) {
+                std::string key((const char*)k, klen);
+                if (put)
+                    merged[key].assign(v, v + vlen);
+                else
+                    merged.erase(key);
+            });
+        if (!ok) {
+            g_err = "malformed spill stream";
+            return RW_E_INVAL;
+        }
+        std::vector<uint8_t> key_types(group_key_types);
+        key_types.push_back(input_types[distinct_cols[di]]);
+        for (auto& [kbytes, val] : merged) {
+            (void)kbytes;
+            Row dk;
+            size_t off = 0;
+            for (size_t c = 0; c < key_types.size(); c++) {
+                rwcodec::DatumC d;
+                size_t got = rwcodec::value_decode_datum(
+                    val.data() + off, val.size() - off, key_types[c], &d);
+                if (!got) {
+                    g_err = "dedup restore: bad key datum";
+                    return RW_E_INVAL;
+                }
+                off += got;
+                Datum dm;
+                dm.null = d.null;
+                if (key_types[c] == RW_T_F64) dm.d = d.d;
+                else dm.i = d.i;
+                dk.push_back(dm);
+            }
+            rwcodec::DatumC d;
+            size_t got = rwcodec::value_decode_datum(
+                val.data() + off, val.size() - off, RW_T_I64, &d);
+            if (!got || d.null) {
+                g_err = "dedup restore: bad count datum";
+                return RW_E_INVAL;
+            }
+            dedup_counts[di][dk] = d.i;
+            dedup_persisted[di].insert(dk);
+        }
+        return RW_OK;
+    }
+
     void emit(uint8_t op, const Row& key, const Row& outputs) {
         Row row;
         row.reserve(key.size() + outputs.size());
@@ -1006,6 +1060,10 @@ int rw_agg_dedup_drain(void* h, int di, uint8_t** buf, uint64_t* len) {
     *buf = (uint8_t*)malloc(sp.size() ? sp.size() : 1);
     memcpy(*buf, sp.data(), sp.size());
     return RW_OK;
+}
+
+int rw_agg_dedup_restore(void* h, int di, const uint8_t* buf, uint64_t len) {
+    return ((HashAggOracle*)h)->dedup_restore(di, buf, len);
 }
 
 void rw_spill_free(uint8_t* buf) { free(buf); }

@@ -268,6 +268,53 @@ struct GroupTopNOracle {
         outputs.erase(outputs.begin());
         return chunk_to_c(*c);
     }
+
+    // §8f-5 recovery: net concatenated drain frames (PUT last-write-wins,
+    // DELETE removes — KV compaction semantics) and rebuild `groups` from
+    // the surviving value-encoded full rows. Restored rows predate the
+    // epoch, so they leave no delta entries.
+    int restore(const uint8_t* buf, uint64_t len) {
+        std::map<std::string, std::vector<uint8_t>> merged;
+        bool ok = rwcodec::for_each_frame(
+            buf, len,
+            [&](uint8_t put, const uint8_t* k, uint32_t klen,
+                const uint8_t* v, uint32_t vlen) {
+                std::string key((const char*)k, klen);
+                if (put)
+                    merged[key].assign(v, v + vlen);
+                else
+                    merged.erase(key);
+            });
+        if (!ok) {
+            g_err = "malformed spill stream";
+            return RW_E_INVAL;
+        }
+        for (auto& [kbytes, val] : merged) {
+            (void)kbytes;
+            Row row;
+            size_t off = 0;
+            for (size_t c = 0; c < types.size(); c++) {
+                rwcodec::DatumC d;
+                size_t got = rwcodec::value_decode_datum(
+                    val.data() + off, val.size() - off, types[c], &d);
+                if (!got) {
+                    g_err = "restore: bad row datum";
+                    return RW_E_INVAL;
+                }
+                off += got;
+                Datum dm;
+                dm.null = d.null;
+                if (types[c] == RW_T_F64) dm.d = d.d;
+                else dm.i = d.i;
+                row.push_back(dm);
+            }
+            Row gk = project(row, group_by);
+            Row ck = project(row, ck_cols);
+            auto& g = groups.try_emplace(gk, Group(ck_less)).first->second;
+            g[ck] = row;
+        }
+        return RW_OK;
+    }
 };
 
 } // namespace orc
@@ -290,6 +337,9 @@ int rw_group_top_n_flush(void* h, uint64_t epoch) {
     return RW_OK; // emission is per push; state commit is a no-op here
 }
 RwChunk* rw_group_top_n_poll(void* h) { return ((GroupTopNOracle*)h)->poll(); }
+int rw_topn_restore(void* h, const uint8_t* buf, uint64_t len) {
+    return ((GroupTopNOracle*)h)->restore(buf, len);
+}
 int rw_topn_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len) {
     std::vector<uint8_t> sp;
     ((GroupTopNOracle*)h)->checkpoint_drain(sp);

@@ -667,6 +667,32 @@ __global__ void dedup_gather_kernel(const JoinSlot* slots,
     }
 }
 
+// §8f-5 recovery: seed dedup counter slots from a netted spill replay.
+// Keys in the netted map are unique, so each lane owns its slot; the slot
+// index is reported back so the host can mark it persisted (the next drain
+// must emit DELETE, not silence, when a restored count dies).
+__global__ void dedup_restore_kernel(const long long* keys,
+                                     const uint32_t* nulls,
+                                     const uint32_t* counts, uint32_t n,
+                                     int KW1, JoinSlot* slots,
+                                     uint32_t cap_mask, uint32_t* out_slots,
+                                     uint32_t* err) {
+    uint32_t stride = gridDim.x * blockDim.x;
+    for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < n;
+         r += stride) {
+        int64_t kw[MAX_KW];
+        for (int i = 0; i < KW1; i++) kw[i] = keys[(size_t)r * KW1 + i];
+        uint32_t slot =
+            jslot_find_or_insert(slots, cap_mask, kw, nulls[r], KW1);
+        if (slot == UINT32_MAX) {
+            atomicExch(err, 4u);
+            continue;
+        }
+        slots[slot].head = counts[r];
+        out_slots[r] = slot;
+    }
+}
+
 struct AggTableDev {
     uint32_t* state;
     int64_t* keys;      // [cap * KW]
@@ -2883,6 +2909,97 @@ struct HashAgg {
         return RW_OK;
     }
 
+    // §8f-5 recovery for one DISTINCT dedup table: replay concatenated
+    // dedup_drain streams (PUT last-write-wins / DELETE removes), decode
+    // the surviving (group ∥ datum, count) records and seed the counter
+    // slots; restored slots are marked persisted so a later 1→0 drop
+    // drains as DELETE exactly as in an uninterrupted run.
+    int dedup_restore(int di, const uint8_t* buf, uint64_t len) {
+        if (di < 0 || (size_t)di >= distinct_slots.size())
+            FAIL(RW_E_INVAL, "dedup table index %d (have %zu)", di,
+                 distinct_slots.size());
+        int rc = ensure_dedup(1);
+        if (rc != RW_OK) return rc;
+        std::map<std::string, std::vector<uint8_t>> merged;
+        bool ok = rwcodec::for_each_frame(
+            buf, len,
+            [&](uint8_t put, const uint8_t* k, uint32_t klen,
+                const uint8_t* v, uint32_t vlen) {
+                std::string key((const char*)k, klen);
+                if (put)
+                    merged[key].assign(v, v + vlen);
+                else
+                    merged.erase(key);
+            });
+        if (!ok) FAIL(RW_E_INVAL, "malformed spill stream");
+        uint32_t n = (uint32_t)merged.size();
+        if (!n) return RW_OK;
+        int KW1 = KW + 1;
+        std::vector<uint8_t> key_types(out_types.begin(),
+                                       out_types.begin() + KW);
+        key_types.push_back(distinct_col_types[di]);
+        std::vector<long long> keys((size_t)n * KW1);
+        std::vector<uint32_t> nulls(n, 0), counts(n);
+        uint32_t i = 0;
+        for (auto& [kbytes, val] : merged) {
+            (void)kbytes;
+            size_t off = 0;
+            for (int c = 0; c < KW1; c++) {
+                rwcodec::DatumC d;
+                size_t got = rwcodec::value_decode_datum(
+                    val.data() + off, val.size() - off, key_types[c], &d);
+                if (!got) FAIL(RW_E_INVAL, "dedup restore: bad key datum");
+                off += got;
+                keys[(size_t)i * KW1 + c] = d.null ? 0 : d.i;
+                nulls[i] |= (uint32_t)(d.null != 0) << c;
+            }
+            rwcodec::DatumC d;
+            size_t got = rwcodec::value_decode_datum(
+                val.data() + off, val.size() - off, RW_T_I64, &d);
+            if (!got || d.null)
+                FAIL(RW_E_INVAL, "dedup restore: bad count datum");
+            counts[i] = (uint32_t)d.i;
+            i++;
+        }
+        long long* dkeys = nullptr;
+        uint32_t *dnulls = nullptr, *dcounts = nullptr, *dslots_out = nullptr,
+                 *derr = nullptr;
+        HIP_TRY(hipMalloc(&dkeys, keys.size() * 8));
+        HIP_TRY(hipMalloc(&dnulls, (size_t)n * 4));
+        HIP_TRY(hipMalloc(&dcounts, (size_t)n * 4));
+        HIP_TRY(hipMalloc(&dslots_out, (size_t)n * 4));
+        HIP_TRY(hipMalloc(&derr, 4));
+        HIP_TRY(hipMemcpy(dkeys, keys.data(), keys.size() * 8,
+                          hipMemcpyHostToDevice));
+        HIP_TRY(hipMemcpy(dnulls, nulls.data(), (size_t)n * 4,
+                          hipMemcpyHostToDevice));
+        HIP_TRY(hipMemcpy(dcounts, counts.data(), (size_t)n * 4,
+                          hipMemcpyHostToDevice));
+        HIP_TRY(hipMemset(derr, 0, 4));
+        dedup_restore_kernel<<<grid_for(n), 256, 0, stream>>>(
+            dkeys, dnulls, dcounts, n, KW1, dedup_slots[di], dedup_cap_mask,
+            dslots_out, derr);
+        std::vector<uint32_t> slots_h(n);
+        uint32_t errv = 0;
+        int rc_s = hipStreamSynchronize(stream) == hipSuccess ? RW_OK
+                                                              : RW_E_INTERNAL;
+        if (rc_s == RW_OK) {
+            hipMemcpy(slots_h.data(), dslots_out, (size_t)n * 4,
+                      hipMemcpyDeviceToHost);
+            hipMemcpy(&errv, derr, 4, hipMemcpyDeviceToHost);
+        }
+        hipFree(dkeys);
+        hipFree(dnulls);
+        hipFree(dcounts);
+        hipFree(dslots_out);
+        hipFree(derr);
+        if (rc_s != RW_OK) FAIL(RW_E_INTERNAL, "dedup restore sync failed");
+        if (errv) FAIL(RW_E_INTERNAL, "dedup restore overflow (code %u)", errv);
+        auto& persisted = dedup_persisted[di];
+        for (uint32_t r = 0; r < n; r++) persisted[slots_h[r]] = 1;
+        return RW_OK;
+    }
+
     // ----- §8f-2 round 2: materialized-input state-TABLE spill. One
     // table per retractable min/max call (AggStateStorage::
     // MaterializedInput, test_utils/agg_executor.rs:63-121): pk =
@@ -3758,6 +3875,10 @@ int rw_agg_dedup_drain(void* h, int di, uint8_t** buf, uint64_t* len) {
     int rc = agg->dedup_drain(di, sp);
     if (rc != RW_OK) return rc;
     return spill_export(sp, buf, len);
+}
+
+int rw_agg_dedup_restore(void* h, int di, const uint8_t* buf, uint64_t len) {
+    return ((HashAgg*)h)->dedup_restore(di, buf, len);
 }
 
 void rw_spill_free(uint8_t* buf) { free(buf); }
@@ -7284,6 +7405,67 @@ __global__ void topn_emit_kernel(JoinSideDev sd, TopMeta m,
     }
 }
 
+// §8f-5 recovery: rebuild the TopN state table from a netted spill replay.
+// Each restored row is appended to the record store and chained onto its
+// group's slot (cache keys in the netted map are unique, so no same-key
+// conflicts; concurrent head pushes on one group resolve by CAS).
+__global__ void topn_restore_kernel(JoinBatchDev b, JoinSideDev sd, TopMeta m,
+                                    uint32_t* err) {
+    uint32_t stride = gridDim.x * blockDim.x;
+    int lane = threadIdx.x & 63;
+    uint32_t n = b.n_rows;
+    uint32_t iters = (n + stride - 1) / stride;
+    for (uint32_t it = 0; it < iters; it++) {
+        uint32_t r = it * stride + blockIdx.x * blockDim.x + threadIdx.x;
+        bool active = r < n;
+        uint64_t wmask = __ballot(active);
+        if (!active) continue;
+        int leader = 63 - __clzll(wmask);
+        uint32_t base = 0;
+        if (lane == leader)
+            base = atomicAdd(sd.row_cursor, (uint32_t)__popcll(wmask));
+        base = (uint32_t)__shfl((int)base, leader);
+        uint32_t row = base + (uint32_t)__popcll(wmask & ((1ULL << lane) - 1));
+        if (row >= sd.row_cap) {
+            atomicExch(err, 3u);
+            continue;
+        }
+        JoinRowHdr* h = jrow(sd, row);
+        long long* hv = jvals(h);
+        uint32_t vb = 0;
+        for (int c = 0; c < m.n_cols; c++) {
+            hv[c] = b.col_vals[c][r];
+            vb |= (uint32_t)(b.col_valid[c][r] != 0) << c;
+        }
+        h->validbits = vb;
+        h->degree = 0;
+        h->alive = 1;
+        int64_t kw[MAX_KW];
+        uint32_t nm = 0;
+        for (int i = 0; i < m.KW; i++) {
+            uint8_t col = m.gk_cols[i];
+            bool valid = (vb >> col) & 1;
+            kw[i] = valid ? hv[col] : 0;
+            nm |= (uint32_t)(!valid) << i;
+        }
+        uint32_t slot =
+            jslot_find_or_insert(sd.slots, sd.cap_mask, kw, nm, m.KW);
+        if (slot == UINT32_MAX) {
+            atomicExch(err, 2u);
+            continue;
+        }
+        uint32_t* headp = &sd.slots[slot].head;
+        uint32_t old_head = ld_u32(headp);
+        for (;;) {
+            st_u32(&h->next, old_head);
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); // R1 drain
+            uint32_t prev = atomicCAS(headp, old_head, row);
+            if (prev == old_head) break;
+            old_head = prev;
+        }
+    }
+}
+
 struct GroupTopN {
     TopMeta m{};
     JoinSideDev sd{};
@@ -7687,6 +7869,71 @@ int rw_topn_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len) {
     int rc = ((GroupTopN*)h)->checkpoint_drain(sp);
     if (rc != RW_OK) return rc;
     return spill_export(sp, buf, len);
+}
+
+// §8f-5 recovery: replay one or more concatenated epoch drains into a fresh
+// GroupTopN executor. PUT/DELETE frames net host-side (last write wins, as a
+// KV store compaction would); the surviving rows rebuild the device state
+// and are marked persisted so the next drain does not re-PUT them.
+int rw_topn_restore(void* h, const uint8_t* buf, uint64_t len) {
+    auto* t = (GroupTopN*)h;
+    std::map<std::string, std::vector<uint8_t>> merged;
+    bool ok = rwcodec::for_each_frame(
+        buf, len,
+        [&](uint8_t put, const uint8_t* k, uint32_t klen, const uint8_t* v,
+            uint32_t vlen) {
+            std::string key((const char*)k, klen);
+            if (put)
+                merged[key].assign(v, v + vlen);
+            else
+                merged.erase(key);
+        });
+    if (!ok) FAIL(RW_E_INVAL, "malformed spill stream");
+    uint32_t n = (uint32_t)merged.size();
+    if (!n) return RW_OK;
+    int ncols = t->m.n_cols;
+    std::vector<std::vector<int64_t>> cols(ncols, std::vector<int64_t>(n));
+    std::vector<std::vector<uint8_t>> valid(ncols, std::vector<uint8_t>(n));
+    uint32_t i = 0;
+    for (auto& [kbytes, val] : merged) {
+        (void)kbytes;
+        size_t off = 0;
+        for (int c = 0; c < ncols; c++) {
+            rwcodec::DatumC d;
+            size_t got = rwcodec::value_decode_datum(
+                val.data() + off, val.size() - off, t->types[c], &d);
+            if (!got) FAIL(RW_E_INVAL, "restore: bad row datum");
+            off += got;
+            cols[c][i] = d.null ? 0 : d.i;
+            valid[c][i] = !d.null;
+        }
+        i++;
+    }
+    int rc = t->ensure_caps(n);
+    if (rc != RW_OK) return rc;
+    JoinBatchDev b = t->stage;
+    for (int c = 0; c < ncols; c++) {
+        HIP_TRY(hipMemcpy(b.col_vals[c], cols[c].data(), (size_t)n * 8,
+                          hipMemcpyHostToDevice));
+        HIP_TRY(hipMemcpy(b.col_valid[c], valid[c].data(), n,
+                          hipMemcpyHostToDevice));
+    }
+    b.vis = nullptr;
+    b.n_rows = n;
+    HIP_TRY(hipMemset(t->tcounters, 0, 8));
+    uint32_t blocks = (n + 255) / 256;
+    if (blocks > 2048) blocks = 2048;
+    topn_restore_kernel<<<blocks, 256, 0, t->stream>>>(b, t->sd, t->m,
+                                                       t->tcounters + 1);
+    if (hipStreamSynchronize(t->stream) != hipSuccess)
+        FAIL(RW_E_INTERNAL, "restore sync failed");
+    uint32_t tc[2];
+    HIP_TRY(hipMemcpy(tc, t->tcounters, 8, hipMemcpyDeviceToHost));
+    if (tc[1]) FAIL(RW_E_INTERNAL, "restore overflow (code %u)", tc[1]);
+    uint32_t cur = 0;
+    HIP_TRY(hipMemcpy(&cur, t->sd.row_cursor, 4, hipMemcpyDeviceToHost));
+    t->flush_mark = cur;
+    return RW_OK;
 }
 
 void* rw_group_top_n_create(const RwGroupTopNDesc* d) {

@@ -645,6 +645,16 @@ def topn_checkpoint_drain(lib, h):
     return out
 
 
+def topn_restore(lib, h, buf):
+    """Rebuild GroupTopN state from concatenated drain bytes (rw_stream.h)."""
+    L = lib.lib
+    L.rw_topn_restore.restype = C.c_int
+    L.rw_topn_restore.argtypes = [C.c_void_p, C.c_char_p, C.c_uint64]
+    rc = L.rw_topn_restore(h, buf, len(buf))
+    if rc != 0:
+        raise RuntimeError(f"topn restore failed {rc}: {lib.last_error()}")
+
+
 def agg_checkpoint_drain_bytes(lib, h):
     """Drain the agg's §8f-2 checkpoint spill buffer; returns raw bytes."""
     L = lib.lib
@@ -742,6 +752,35 @@ def agg_minput_drain_bytes(lib, h, mi):
     L.rw_spill_free.argtypes = [C.c_void_p]
     L.rw_spill_free(C.cast(buf, C.c_void_p))
     return out
+
+
+def agg_dedup_drain_bytes(lib, h, di):
+    """Drain one DISTINCT dedup table's spill buffer; returns raw bytes."""
+    L = lib.lib
+    L.rw_agg_dedup_drain.restype = C.c_int
+    L.rw_agg_dedup_drain.argtypes = [C.c_void_p, C.c_int,
+                                     C.POINTER(C.POINTER(C.c_uint8)),
+                                     C.POINTER(C.c_uint64)]
+    buf = C.POINTER(C.c_uint8)()
+    ln = C.c_uint64()
+    rc = L.rw_agg_dedup_drain(h, di, C.byref(buf), C.byref(ln))
+    if rc != 0:
+        raise RuntimeError(f"dedup drain failed {rc}: {lib.last_error()}")
+    out = bytes(bytearray(buf[i] for i in range(ln.value)))
+    L.rw_spill_free.argtypes = [C.c_void_p]
+    L.rw_spill_free(C.cast(buf, C.c_void_p))
+    return out
+
+
+def agg_dedup_restore(lib, h, di, buf):
+    """Rebuild one dedup table from concatenated drain bytes (rw_stream.h)."""
+    L = lib.lib
+    L.rw_agg_dedup_restore.restype = C.c_int
+    L.rw_agg_dedup_restore.argtypes = [C.c_void_p, C.c_int, C.c_char_p,
+                                       C.c_uint64]
+    rc = L.rw_agg_dedup_restore(h, di, buf, len(buf))
+    if rc != 0:
+        raise RuntimeError(f"dedup restore failed {rc}: {lib.last_error()}")
 
 
 def agg_minput_restore(lib, h, mi, buf):

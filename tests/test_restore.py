@@ -288,6 +288,173 @@ def test_agg_minput_restore_gpu():
     _agg_minput_restore_flow(ffi.Lib(risingwave_amd.lib_path()))
 
 
+def _topn_epoch_chunks(rng, live, rid):
+    """Insert/delete mix over bounded groups; rows keyed by a unique id so
+    deletes target real state rows by their full cache key."""
+    chunks = []
+    for _ in range(3):
+        n = 256
+        g = rng.integers(0, 16, n)
+        v = rng.integers(0, 200, n)
+        r2 = np.arange(rid[0], rid[0] + n)
+        rid[0] += n
+        ops = np.zeros(n, np.uint8)
+        for r in range(n):
+            if live and rng.random() < 0.3:
+                j = int(rng.integers(0, len(live)))
+                g[r], v[r], r2[r] = live.pop(j)
+                ops[r] = ffi.OP_DELETE
+            else:
+                live.append((int(g[r]), int(v[r]), int(r2[r])))
+        chunks.append(mk_chunk([T_I64, T_I64, T_I64], ops, [g, v, r2]))
+    return chunks
+
+
+def _topn_restore_flow(lib):
+    from rwtest.ffi import topn_checkpoint_drain, topn_restore
+
+    mk = lambda: ffi.GroupTopN(lib, [T_I64, T_I64, T_I64], [0],
+                               [(1, False)], [(2, False)], offset=1, limit=3)
+    a = mk()
+    rng = np.random.default_rng(4242)
+    live = []
+    rid = [0]
+    drains = []
+    for e in range(3):
+        for c in _topn_epoch_chunks(rng, live, rid):
+            a.push(c)
+        a.poll_all()
+        drains.append(topn_checkpoint_drain(lib, a.h))
+    b = mk()
+    topn_restore(lib, b.h, b"".join(drains))
+    for e in range(3):
+        cs = _topn_epoch_chunks(rng, live, rid)
+        for c in cs:
+            a.push(c)
+            b.push(c)
+            ma = rows_multiset(a.poll_all())
+            mb = rows_multiset(b.poll_all())
+            assert ma == mb, f"epoch {e}: restored topn diverged"
+        da = topn_checkpoint_drain(lib, a.h)
+        db = topn_checkpoint_drain(lib, b.h)
+        assert da == db, f"epoch {e}: topn state drain diverged"
+    a.close()
+    b.close()
+
+
+def test_topn_restore_oracle():
+    _topn_restore_flow(oracle())
+
+
+@pytest.mark.gpu
+def test_topn_restore_gpu():
+    import risingwave_amd
+
+    risingwave_amd.load_library()
+    _topn_restore_flow(ffi.Lib(risingwave_amd.lib_path()))
+
+
+@pytest.mark.gpu
+def test_topn_restore_gpu_matches_oracle_drains():
+    # a GPU executor restored from GPU drains and an oracle restored from
+    # oracle drains must produce byte-identical subsequent drains
+    import risingwave_amd
+    from rwtest.ffi import topn_checkpoint_drain, topn_restore
+
+    risingwave_amd.load_library()
+    glib = ffi.Lib(risingwave_amd.lib_path())
+    streams = {}
+    for name, lib in (("gpu", glib), ("orc", oracle())):
+        t = ffi.GroupTopN(lib, [T_I64, T_I64, T_I64], [0],
+                          [(1, True)], [(2, False)], limit=2)
+        rng = np.random.default_rng(31)
+        live, rid = [], [0]
+        d = b""
+        for e in range(2):
+            for c in _topn_epoch_chunks(rng, live, rid):
+                t.push(c)
+            t.poll_all()
+            d += topn_checkpoint_drain(lib, t.h)
+        t.close()
+        t2 = ffi.GroupTopN(lib, [T_I64, T_I64, T_I64], [0],
+                           [(1, True)], [(2, False)], limit=2)
+        topn_restore(lib, t2.h, d)
+        for c in _topn_epoch_chunks(rng, live, rid):
+            t2.push(c)
+        t2.poll_all()
+        streams[name] = (d, topn_checkpoint_drain(lib, t2.h))
+        t2.close()
+    assert streams["gpu"] == streams["orc"]
+
+
+def _agg_dedup_restore_flow(lib):
+    # COUNT(DISTINCT col) + plain SUM: the dedup counter table spills per
+    # epoch and must restore so duplicate-visibility transitions (0→1, 1→0)
+    # continue exactly as uninterrupted
+    from rwtest.ffi import (AGG_COUNT, agg_dedup_drain_bytes,
+                            agg_dedup_restore)
+
+    calls = [(AGG_COUNT, 1, T_I64, 1), (AGG_SUM, 1, T_I64)]
+    mk = lambda: ffi.HashAgg(lib, [T_I64, T_I64], [0], calls, 1)
+    a = mk()
+    rng = np.random.default_rng(777)
+    live = []
+
+    def chunks():
+        # small value domain (0..12) so distinct counts really exercise
+        # duplicate suppression both ways
+        out = []
+        for _ in range(3):
+            n = 256
+            g = rng.integers(0, 10, n)
+            v = rng.integers(0, 12, n)
+            ops = np.zeros(n, np.uint8)
+            for r in range(n):
+                if live and rng.random() < 0.35:
+                    j = int(rng.integers(0, len(live)))
+                    g[r], v[r] = live.pop(j)
+                    ops[r] = ffi.OP_DELETE
+                else:
+                    live.append((int(g[r]), int(v[r])))
+            out.append(mk_chunk([T_I64, T_I64], ops, [g, v]))
+        return out
+
+    inter = b""
+    ded = b""
+    for e in range(3):
+        _drive_agg(a, chunks(), e + 1)
+        inter += agg_checkpoint_drain_bytes(lib, a.h)
+        ded += agg_dedup_drain_bytes(lib, a.h, 0)
+    b = mk()
+    agg_dedup_restore(lib, b.h, 0, ded)
+    agg_restore(lib, b.h, inter)
+    for e in range(3):
+        cs = chunks()
+        oa = _drive_agg(a, cs, 5 + e)
+        ob = _drive_agg(b, cs, 5 + e)
+        assert oa == ob, f"epoch {e}: restored distinct agg diverged"
+        da = agg_checkpoint_drain_bytes(lib, a.h)
+        db = agg_checkpoint_drain_bytes(lib, b.h)
+        assert da == db, f"epoch {e}: intermediate drain diverged"
+        xa = agg_dedup_drain_bytes(lib, a.h, 0)
+        xb = agg_dedup_drain_bytes(lib, b.h, 0)
+        assert xa == xb, f"epoch {e}: dedup drain diverged"
+    a.close()
+    b.close()
+
+
+def test_agg_dedup_restore_oracle():
+    _agg_dedup_restore_flow(oracle())
+
+
+@pytest.mark.gpu
+def test_agg_dedup_restore_gpu():
+    import risingwave_amd
+
+    risingwave_amd.load_library()
+    _agg_dedup_restore_flow(ffi.Lib(risingwave_amd.lib_path()))
+
+
 def test_agg_minput_drain_parity_cpu_noop():
     # sanity: a fresh executor drains empty minput tables
     from rwtest.ffi import agg_minput_drain_bytes
