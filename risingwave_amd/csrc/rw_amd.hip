@@ -4190,6 +4190,53 @@ __device__ __forceinline__ uint32_t table_find_cached(
     return (uint32_t)-1;
 }
 
+// Select word `idx` of a register-resident record copy. Unrolled selects:
+// a dynamic index into a private array would demote it to scratch, which
+// defeats the point of the register walk.
+__device__ __forceinline__ long long rec_sel(const long long* recv, int idx) {
+    long long v = 0;
+#pragma unroll
+    for (int w = 0; w < 6; w++)
+        if (w == idx) v = recv[w];
+    return v;
+}
+
+// join_cond_ok over a register-resident match record (no memory re-reads)
+__device__ __forceinline__ bool join_cond_ok_rec(const JoinMeta& m,
+                                                 int probe_side,
+                                                 const JoinBatchDev& b,
+                                                 uint32_t r,
+                                                 uint32_t validbits,
+                                                 const long long* recv) {
+    for (int ci = 0; ci < m.n_cond; ci++) {
+        auto& cd = m.cond[ci];
+        auto fetch = [&](uint8_t src, uint8_t col, int64_t* v) -> bool {
+            if ((int)src == probe_side) {
+                if (!b.col_valid[col][r]) return false;
+                *v = b.col_vals[col][r];
+            } else {
+                if (!((validbits >> col) & 1)) return false;
+                *v = rec_sel(recv, col);
+            }
+            return true;
+        };
+        int64_t a, c;
+        if (!fetch(cd.src_l, cd.col_l, &a)) return false; // NULL ⇒ false
+        if (!fetch(cd.src_r, cd.col_r, &c)) return false;
+        c += cd.rconst;
+        bool ok;
+        switch (cd.op) {
+            case RW_CMP_LT: ok = a < c; break;
+            case RW_CMP_LE: ok = a <= c; break;
+            case RW_CMP_GT: ok = a > c; break;
+            case RW_CMP_GE: ok = a >= c; break;
+            default: ok = false;
+        }
+        if (!ok) return false;
+    }
+    return true;
+}
+
 __device__ __forceinline__ bool join_cond_ok(const JoinMeta& m, int probe_side,
                                              const JoinBatchDev& b, uint32_t r,
                                              uint32_t validbits,
@@ -4502,14 +4549,13 @@ __global__ void join_count_emitted_kernel(const uint8_t* ops, uint32_t n,
         atomicAdd(out_count, (unsigned long long)(counters[0] - n));
 }
 
-__global__ __launch_bounds__(256, 8) void join_probe_kernel(
+__global__ __launch_bounds__(256, 6) void join_probe_kernel(
     JoinBatchDev b, JoinSideDev own, JoinSideDev match, JoinMeta m, int S,
     JoinOutDev out, uint32_t r0, uint32_t r1, const uint32_t* row_base,
     int dbg_skip = 0) {
     uint32_t stride = gridDim.x * blockDim.x;
     uint32_t n = r1 - r0;
     uint32_t iters = (n + stride - 1) / stride;
-    int lane = threadIdx.x & 63;
 
     const bool dense = b.all_insert && b.all_valid && !b.vis;
     for (uint32_t it = 0; it < iters; it++) {
@@ -4546,21 +4592,72 @@ __global__ __launch_bounds__(256, 8) void join_probe_kernel(
         uint32_t my_n = 0;
         uint32_t matched_row = UINT32_MAX;
         uint64_t h64 = 0;
+        // register-resident copy of the last record the walk loaded, and
+        // whether it is the (single) match — lets the emit below run with
+        // zero record re-reads in the dominant unique-key case
+        long long rec[8];
+        bool rec_match = false;
         if (active) {
             h64 = hash_key(kw, nullmask, m.KW);
             mhead = jbucket_head(match, h64, m.append_only != 0);
-            if (mhead != UINT32_MAX && !m.append_only) {
+            if (mhead != UINT32_MAX && !m.append_only && !(dbg_skip & 16)) {
+                // dbg 16 = hash+slot only (stage isolation: no record walk)
                 uint32_t row = mhead;
-                while (row != UINT32_MAX) {
-                    JoinRowHdr* h = jrow(match, row);
-                    if (h->alive &&
-                        jhdr_key_eq(h, m.key_cols[1 - S], m.KW, kw, nullmask,
-                                    false) &&
-                        join_cond_ok(m, S, b, r, h->validbits, jvals(h))) {
-                        my_n++;
-                        matched_row = row;
+                // REGISTER-RESIDENT walk (narrow records, 16-B-aligned
+                // stride): each record is fetched ONCE as wide b128 loads
+                // into registers and evaluated from there. The serialized
+                // field-by-field walk refetches its line from HBM 2-3x —
+                // with ~2048 lanes in flight per CU the 32-KB vL0 (and the
+                // 4-MB XCD L2) evict a line between a record's `alive`
+                // check and its key loads; PMC measured ~323 B fetched per
+                // 64-B record (gpurun_out/pmcW67.txt) vs the one-line 128-B
+                // model. dbg 32 forces the legacy field walk for A/B.
+                if (match.row_stride <= 64 && !(match.row_stride & 15) &&
+                    !(dbg_skip & 32)) {
+                    int nw16 = (int)(match.row_stride >> 4);
+                    while (row != UINT32_MAX) {
+                        const long long* rp =
+                            (const long long*)jrow(match, row);
+#pragma unroll
+                        for (int w = 0; w < 4; w++)
+                            if (w < nw16)
+                                *(longlong2*)&rec[2 * w] =
+                                    *(const longlong2*)(rp + 2 * w);
+                        rec_match = false; // rec just overwritten
+                        uint32_t alive = (uint32_t)(uint64_t)rec[0];
+                        uint32_t nxt = (uint32_t)((uint64_t)rec[0] >> 32);
+                        uint32_t vb = (uint32_t)(uint64_t)rec[1];
+                        bool eq = alive != 0;
+                        for (int i = 0; eq && i < m.KW; i++) {
+                            uint8_t col = m.key_cols[1 - S][i];
+                            bool valid = (vb >> col) & 1;
+                            if (valid == (bool)((nullmask >> i) & 1))
+                                eq = false;
+                            else if (valid &&
+                                     rec_sel(rec + 2, col) != kw[i])
+                                eq = false;
+                        }
+                        if (eq &&
+                            join_cond_ok_rec(m, S, b, r, vb, rec + 2)) {
+                            my_n++;
+                            matched_row = row;
+                            rec_match = true; // match is register-resident
+                        }
+                        row = nxt;
                     }
-                    row = h->next;
+                } else {
+                    while (row != UINT32_MAX) {
+                        JoinRowHdr* h = jrow(match, row);
+                        if (h->alive &&
+                            jhdr_key_eq(h, m.key_cols[1 - S], m.KW, kw,
+                                        nullmask, false) &&
+                            join_cond_ok(m, S, b, r, h->validbits,
+                                         jvals(h))) {
+                            my_n++;
+                            matched_row = row;
+                        }
+                        row = h->next;
+                    }
                 }
             }
         }
@@ -4582,7 +4679,31 @@ __global__ __launch_bounds__(256, 8) void join_probe_kernel(
                 if (extra_base + my_n - 1 > out.cap)
                     atomicExch(&out.counters[1], 1u); // overflow
             }
-            if (my_n == 1 && !(dbg_skip & 1)) {
+            if (my_n == 1 && !(dbg_skip & 1) && rec_match) {
+                // SINGLE-PASS emission from the REGISTER-RESIDENT match
+                // (the walk's last-loaded record): zero record re-reads
+                uint32_t orow = my_base;
+                uint32_t mvb = (uint32_t)(uint64_t)rec[1];
+                __builtin_nontemporal_store(op, &out.ops[orow]);
+                for (int c = 0; c < m.n_out; c++) {
+                    bool from_probe = (int)m.out_src[c] == S;
+                    uint8_t col = m.out_col[c];
+                    int64_t v;
+                    uint8_t valid;
+                    if (from_probe) {
+                        valid = dense || b.col_valid[col][r];
+                        v = b.col_vals[col][r];
+                    } else {
+                        valid = (mvb >> col) & 1;
+                        v = rec_sel(rec + 2, col);
+                    }
+                    __builtin_nontemporal_store(
+                        valid ? v : 0, &out.vals[(size_t)c * out.cap + orow]);
+                    __builtin_nontemporal_store(
+                        (uint8_t)!valid,
+                        &out.nulls[(size_t)c * out.cap + orow]);
+                }
+            } else if (my_n == 1 && !(dbg_skip & 1)) {
                 // SINGLE-PASS emission for the dominant <=1-match case:
                 // the count walk already identified the row — no re-walk
                 JoinRowHdr* h = jrow(match, matched_row);
@@ -4711,8 +4832,11 @@ __global__ __launch_bounds__(256, 8) void join_probe_kernel(
             // PRE-ASSIGNED record slot (row_base + r): inserting rows write
             // their record and link the bucket; every other row marks its
             // slot dead (walks and drains skip by `alive`). No cross-lane
-            // cursor — see the emission comment above.
-            if (r < r1) {
+            // cursor — see the emission comment above. dbg 64 = skip this
+            // whole section (stage isolation: clean walk timing without
+            // the pre-kill store pass the production path replaces with
+            // full record writes).
+            if (r < r1 && !(dbg_skip & 64)) {
                 uint32_t myrow = *row_base + r;
                 bool do_insert = active && !(dbg_skip & 2) && is_insert;
                 if (myrow < own.row_cap) {
