@@ -4844,6 +4844,13 @@ __global__ __launch_bounds__(256, 6) void join_probe_kernel(
                     if (do_insert) {
                         long long* hv2 = jvals(hd2);
                         uint32_t vb2 = 0;
+                        // (Measured dead end: nontemporal record-insert
+                        // stores — both 8-B fields and register-assembled
+                        // b128s — ran 0.320 vs 0.248 ms/step on q8. The
+                        // 64-B lane stride means no single instruction
+                        // covers a full line, so NT partial-line writes
+                        // reach HBM unmerged, while the cached path's RFO
+                        // is amortized by L2 byte-granular write merge.)
                         if (dense) {
                             for (int c = 0; c < m.n_cols[S]; c++)
                                 hv2[c] = b.col_vals[c][r];
