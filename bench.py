@@ -779,9 +779,12 @@ def bench_q8(args, ffi, gpu_lib, rng, rank, world, dist):
     t4 = [T_I64, T_TS, T_TS, T_I64]
     # hint 2^23 -> cap 2^24 slots x 8 B = 134 MB per side: BOTH slot tables
     # sit in the 256 MB Infinity Cache (load factor 0.6 at 10M keys)
+    # RW_Q8_HINT_LOG2 A/B hook: 23 (default) -> cap 2^24 slots/side; 22
+    # halves both tables so slots + records contend less for the LIC
+    hint_log2 = int(os.environ.get("RW_Q8_HINT_LOG2", "23"))
     j = ffi.HashJoin(gpu_lib, JOIN_INNER, t4, t4, key_l=[0, 1, 2],
                      key_r=[0, 1, 2], pk_l=[3], pk_r=[3],
-                     state_capacity_hint=1 << 23,
+                     state_capacity_hint=1 << hint_log2,
                      row_capacity_hint=BUILD_KEYS + (args.steps + args.warmup + 4)
                      * batch_rows + 1_000_000)
 

@@ -4681,10 +4681,19 @@ __global__ __launch_bounds__(256, 6) void join_probe_kernel(
             }
             if (my_n == 1 && !(dbg_skip & 1) && rec_match) {
                 // SINGLE-PASS emission from the REGISTER-RESIDENT match
-                // (the walk's last-loaded record): zero record re-reads
+                // (the walk's last-loaded record): zero record re-reads.
+                // Store policy A/B (dbg 8 = fully cached, dbg 256 = vals
+                // NT + 1-byte ops/nulls cached): measured 0.251 / 0.255 /
+                // 0.255 ms/step — full NT is the (narrow) winner, the
+                // dense orow=r layout coalesces well under every policy.
                 uint32_t orow = my_base;
                 uint32_t mvb = (uint32_t)(uint64_t)rec[1];
-                __builtin_nontemporal_store(op, &out.ops[orow]);
+                bool nt_vals = !(dbg_skip & 8);
+                bool nt_meta = !(dbg_skip & (8 | 256));
+                if (nt_meta)
+                    __builtin_nontemporal_store(op, &out.ops[orow]);
+                else
+                    out.ops[orow] = op;
                 for (int c = 0; c < m.n_out; c++) {
                     bool from_probe = (int)m.out_src[c] == S;
                     uint8_t col = m.out_col[c];
@@ -4697,11 +4706,19 @@ __global__ __launch_bounds__(256, 6) void join_probe_kernel(
                         valid = (mvb >> col) & 1;
                         v = rec_sel(rec + 2, col);
                     }
-                    __builtin_nontemporal_store(
-                        valid ? v : 0, &out.vals[(size_t)c * out.cap + orow]);
-                    __builtin_nontemporal_store(
-                        (uint8_t)!valid,
-                        &out.nulls[(size_t)c * out.cap + orow]);
+                    if (nt_vals)
+                        __builtin_nontemporal_store(
+                            valid ? v : 0,
+                            &out.vals[(size_t)c * out.cap + orow]);
+                    else
+                        out.vals[(size_t)c * out.cap + orow] = valid ? v : 0;
+                    if (nt_meta)
+                        __builtin_nontemporal_store(
+                            (uint8_t)!valid,
+                            &out.nulls[(size_t)c * out.cap + orow]);
+                    else
+                        out.nulls[(size_t)c * out.cap + orow] =
+                            (uint8_t)!valid;
                 }
             } else if (my_n == 1 && !(dbg_skip & 1)) {
                 // SINGLE-PASS emission for the dominant <=1-match case:
