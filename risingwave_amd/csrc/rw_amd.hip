@@ -6091,7 +6091,11 @@ struct HashJoin {
 
     int probe(int s, const JoinBatchDev& b, bool timed, uint32_t r0, uint32_t r1) {
         uint32_t blocks = (r1 - r0 + 255) / 256;
-        if (blocks > 2048) blocks = 2048;
+        // one row per lane up to 16K blocks: a grid-stride iteration
+        // serializes its dependent walk behind the previous row's, while
+        // fresh blocks give the scheduler independent work as waves
+        // retire (measured on the q8 shape; 2048-cap was the G11 default)
+        if (blocks > 16384) blocks = 16384;
         if (!blocks) blocks = 1;
         int slot = -1;
         if (timed) {
