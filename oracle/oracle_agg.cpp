@@ -939,6 +939,32 @@ struct HashAggOracle {
         for (auto it = groups.begin(); it != groups.end();) {
             const Datum& d = it->first[pos];
             if (!d.null && d.i < val) {
+                // state-cleaning spill deltas (the reference's commit
+                // applies the watermark as a range delete on the store,
+                // state_table.rs:1707): persisted groups net to DELETE
+                // so a restore replay cannot resurrect them
+                if (it->second.has_prev)
+                    spill_record(0, it->first, it->second.prev_outputs);
+                // minput state-table rows of the cleaned group
+                for (size_t ci = 0; ci < calls.size(); ci++) {
+                    if (!call_is_minput[ci] ||
+                        ci >= it->second.mstates.size())
+                        continue;
+                    int mi = minput_ordinal(ci);
+                    if ((size_t)mi >= minput_delta.size())
+                        minput_delta.resize(mi + 1);
+                    auto& delta = minput_delta[mi];
+                    for (auto& [ekey, dat] : it->second.mstates[ci].entries) {
+                        (void)dat;
+                        std::string kb;
+                        minput_encode(ci, it->first, ekey, &kb, nullptr);
+                        auto di = delta.find(kb);
+                        if (di != delta.end() && di->second.first == 1)
+                            delta.erase(di); // created+died this epoch
+                        else
+                            delta[kb] = {0, {}};
+                    }
+                }
                 dirty.erase(it->first);
                 it = groups.erase(it);
             } else {
@@ -948,6 +974,21 @@ struct HashAggOracle {
         for (auto it = dirty_order.begin(); it != dirty_order.end();)
             if (!groups.count(*it)) it = dirty_order.erase(it);
             else ++it;
+        // the same watermark cleans the DISTINCT dedup tables' group
+        // prefix: counts reset (late rows restart visibility from 0) and
+        // the next dedup drain nets persisted rows to DELETE
+        for (size_t di = 0; di < dedup_counts.size(); di++) {
+            auto& counts = dedup_counts[di];
+            for (auto it2 = counts.begin(); it2 != counts.end();) {
+                const Datum& d2 = it2->first[pos];
+                if (!d2.null && d2.i < val) {
+                    dedup_touched[di].insert(it2->first);
+                    it2 = counts.erase(it2);
+                } else {
+                    ++it2;
+                }
+            }
+        }
         return RW_OK;
     }
 

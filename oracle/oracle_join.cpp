@@ -584,8 +584,18 @@ struct HashJoinOracle {
             auto& tab = side[s].table;
             for (auto it = tab.begin(); it != tab.end();) {
                 const Datum& kd = it->first[jk_idx];
-                if (!kd.null && kd.i < sel) it = tab.erase(it);
-                else ++it;
+                if (!kd.null && kd.i < sel) {
+                    // state-cleaning spill deltas: the reference's commit
+                    // applies the watermark as a range delete on the
+                    // store (state_table.rs:1707), so the spill stream
+                    // must carry DELETEs for the cleaned rows — a netted
+                    // restore replay would otherwise resurrect them
+                    for (auto& [pk, entry] : it->second)
+                        delta_delete(s, it->first, pk);
+                    it = tab.erase(it);
+                } else {
+                    ++it;
+                }
             }
         }
     }

@@ -5467,16 +5467,36 @@ __global__ void join_clean_kernel(JoinSideDev sd, int kcol, long long wm,
             JoinRowHdr* h = jrow(sd, row);
             if (ld_u32(&h->alive) &&
                 ((ld_u32(&h->validbits) >> kcol) & 1) && // NULLs largest
-                ld_i64((const int64_t*)&jvals(h)[kcol]) < wm)
+                ld_i64((const int64_t*)&jvals(h)[kcol]) < wm) {
                 st_u32(&h->alive, 0);
+                // §8f-2 state-cleaning spill deltas (the reference's
+                // commit applies the watermark as a range delete on the
+                // store, state_table.rs:1707): route retired rows
+                // through the kill list so the next drain nets them to
+                // DELETE frames — a restore replay must not resurrect
+                // them
+                if (sd.killed) {
+                    uint32_t kidx = atomicAdd(sd.killed_cursor, 1u);
+                    if (kidx < sd.killed_cap) sd.killed[kidx] = row;
+                }
+            }
             row = ld_u32(&h->next);
         }
     }
 }
 
+// crecs/crec_n/crec_cap/cover: capture the cleaned PERSISTED groups'
+// keys so the host can append DELETE spill frames — the reference's
+// commit applies the watermark as a range delete on the store
+// (state_table.rs:1707), so the spill stream must net cleaned rows out
+// or a restore replay resurrects them. On capture overflow the slot is
+// LEFT UNCLEANED (cover set); the host grows the buffer and reruns —
+// already-cleaned slots skip via the has_prev they cleared.
 __global__ void agg_clean_kernel(AggTableDev t, int kpos, long long wm, int KW,
                                  int n_calls, AggCallDev c0, AggCallDev c1,
-                                 AggCallDev c2, AggCallDev c3) {
+                                 AggCallDev c2, AggCallDev c3,
+                                 DedupDirtyRec* crecs, uint32_t* crec_n,
+                                 uint32_t crec_cap, uint32_t* cover) {
     AggCallDev calls[4] = {c0, c1, c2, c3};
     size_t cap = (size_t)t.cap_mask + 1;
     size_t stride = (size_t)gridDim.x * blockDim.x;
@@ -5485,6 +5505,17 @@ __global__ void agg_clean_kernel(AggTableDev t, int kpos, long long wm, int KW,
         if (ld_u32(&t.state[(uint32_t)slot]) != SLOT_READY) continue;
         if ((ld_u32(&t.key_nulls[(uint32_t)slot]) >> kpos) & 1) continue;
         if (ld_i64((const int64_t*)&t.keys[slot * KW + kpos]) >= wm) continue;
+        if (t.has_prev[(uint32_t)slot]) {
+            uint32_t ci2 = atomicAdd(crec_n, 1u);
+            if (ci2 >= crec_cap) {
+                atomicExch(cover, 1u);
+                continue; // retried after the host grows the buffer
+            }
+            for (int w = 0; w < KW; w++)
+                crecs[ci2].key[w] = t.keys[slot * KW + w];
+            crecs[ci2].nulls = t.key_nulls[(uint32_t)slot];
+            crecs[ci2].count = 0;
+        }
         // reset the group as if freshly created (late rows restart it)
         for (int ci = 0; ci < n_calls; ci++) {
             long long init = 0;
@@ -5496,6 +5527,13 @@ __global__ void agg_clean_kernel(AggTableDev t, int kpos, long long wm, int KW,
                 uint32_t row = t.mheads[(size_t)calls[ci].mord * cap + slot];
                 while (row != UINT32_MAX) {
                     st_u32(&t.malive[row], 0);
+                    // minput state-table deltas for the cleaned rows
+                    // (netted to DELETE by minput_drain, as apply-path
+                    // retractions are)
+                    if (t.mkilled) {
+                        uint32_t ki = atomicAdd(t.mkilled_cursor, 1u);
+                        if (ki < t.mkilled_cap) t.mkilled[ki] = row;
+                    }
                     row = ld_u32(&t.mnext[row]);
                 }
                 t.mheads[(size_t)calls[ci].mord * cap + slot] = UINT32_MAX;
@@ -5503,6 +5541,28 @@ __global__ void agg_clean_kernel(AggTableDev t, int kpos, long long wm, int KW,
         }
         t.has_prev[(uint32_t)slot] = 0;
         t.dirty_flag[(uint32_t)slot] = 0;
+    }
+}
+
+// DISTINCT dedup-table state cleaning (the same watermark applies to the
+// dedup tables' group-key prefix in the reference): reset counts below
+// the watermark and mark the slot dirty so the next dedup drain nets a
+// DELETE for persisted rows — and so LATE rows restart visibility
+// transitions from 0, as the reference's freshly-deleted state row would.
+__global__ void dedup_clean_kernel(JoinSlot* slots, size_t cap, int kpos,
+                                   long long wm, uint32_t* dirty_flag,
+                                   uint32_t* dirty_list, uint32_t* dirty_n) {
+    size_t stride = (size_t)gridDim.x * blockDim.x;
+    for (size_t slot = blockIdx.x * blockDim.x + threadIdx.x; slot < cap;
+         slot += stride) {
+        if (slots[slot].state != SLOT_READY) continue;
+        if ((slots[slot].nulls >> kpos) & 1) continue;
+        if (slots[slot].key[kpos] >= wm) continue;
+        if (slots[slot].head == 0) continue; // nothing live to clean
+        slots[slot].head = 0;
+        if (ld_u32(&dirty_flag[slot]) == 0 &&
+            atomicCAS(&dirty_flag[slot], 0u, 1u) == 0u)
+            dirty_list[atomicAdd(dirty_n, 1u)] = (uint32_t)slot;
     }
 }
 
@@ -6710,9 +6770,58 @@ int rw_hash_agg_watermark(void* h, uint32_t group_key_pos, int64_t val) {
         agg->has_pending_wm = true;
         return RW_OK;
     }
-    agg_clean_kernel<<<2048, 256, 0, agg->stream>>>(
-        agg->t, (int)group_key_pos, val, agg->KW, agg->n_calls, agg->cd(0),
-        agg->cd(1), agg->cd(2), agg->cd(3));
+    // clean + capture persisted cleaned groups for DELETE spill frames
+    // (state-table watermark range delete, state_table.rs:1707)
+    uint32_t crec_cap = 1u << 18;
+    for (;;) {
+        DedupDirtyRec* crecs = nullptr;
+        uint32_t* cmeta = nullptr; // [0]=n [1]=overflow
+        HIP_TRY(hipMalloc(&crecs, (size_t)crec_cap * sizeof(DedupDirtyRec)));
+        HIP_TRY(hipMalloc(&cmeta, 8));
+        HIP_TRY(hipMemset(cmeta, 0, 8));
+        agg_clean_kernel<<<2048, 256, 0, agg->stream>>>(
+            agg->t, (int)group_key_pos, val, agg->KW, agg->n_calls,
+            agg->cd(0), agg->cd(1), agg->cd(2), agg->cd(3), crecs, cmeta,
+            crec_cap, cmeta + 1);
+        int rcs = hipStreamSynchronize(agg->stream) == hipSuccess
+                      ? RW_OK : RW_E_INTERNAL;
+        uint32_t meta[2] = {0, 0};
+        if (rcs == RW_OK)
+            hipMemcpy(meta, cmeta, 8, hipMemcpyDeviceToHost);
+        uint32_t n = meta[0] < crec_cap ? meta[0] : crec_cap;
+        std::vector<DedupDirtyRec> recs(n);
+        if (rcs == RW_OK && n)
+            hipMemcpy(recs.data(), crecs, (size_t)n * sizeof(DedupDirtyRec),
+                      hipMemcpyDeviceToHost);
+        hipFree(crecs);
+        hipFree(cmeta);
+        if (rcs != RW_OK) FAIL(RW_E_INTERNAL, "agg clean sync failed");
+        auto put32 = [&](uint32_t x) {
+            for (int b = 0; b < 4; b++)
+                agg->spill.push_back((uint8_t)(x >> (8 * b)));
+        };
+        for (uint32_t i = 0; i < n; i++) {
+            agg->spill.push_back(0); // DELETE
+            std::vector<uint8_t> k;
+            for (int w = 0; w < agg->KW; w++) {
+                rwcodec::DatumC d{((recs[i].nulls >> w) & 1) != 0,
+                                  recs[i].key[w], 0};
+                rwcodec::memcmp_encode_datum(k, agg->out_types[w], d, {});
+            }
+            put32((uint32_t)k.size());
+            agg->spill.insert(agg->spill.end(), k.begin(), k.end());
+            put32(0);
+        }
+        if (!meta[1]) break; // no overflow: every slot cleaned
+        crec_cap *= 4;
+    }
+    // the same watermark cleans the DISTINCT dedup tables' group prefix
+    for (size_t di = 0; di < agg->dedup_slots.size(); di++) {
+        size_t dcap = (size_t)agg->dedup_cap_mask + 1;
+        dedup_clean_kernel<<<2048, 256, 0, agg->stream>>>(
+            agg->dedup_slots[di], dcap, (int)group_key_pos, val,
+            agg->ddirty_flag[di], agg->ddirty_list[di], agg->ddirty_n[di]);
+    }
     HIP_TRY(hipStreamSynchronize(agg->stream));
     return RW_OK;
 }

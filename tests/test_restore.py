@@ -455,6 +455,181 @@ def test_agg_dedup_restore_gpu():
     _agg_dedup_restore_flow(ffi.Lib(risingwave_amd.lib_path()))
 
 
+def _join_clean_restore_flow(lib, join_type):
+    """Watermark state-cleaning must surface in the spill stream: cleaned
+    rows net to DELETE frames so a restore replay cannot resurrect them
+    (the reference's commit applies the watermark as a range delete,
+    state_table.rs:1707). Returns all drain streams for cross-impl
+    byte-comparison."""
+    mk = lambda: ffi.HashJoin(lib, join_type, [T_I64, T_I64],
+                              [T_I64, T_I64], key_l=[0], key_r=[0],
+                              pk_l=[1], pk_r=[1], wm_jk=((0, True),))
+    rng = np.random.default_rng(616)
+    pk = [0]
+
+    def epoch_pushes(e):
+        pushes = []
+        for side in (SIDE_LEFT, SIDE_RIGHT):
+            n = 200
+            k = rng.integers(e * 10, e * 10 + 25, n)
+            v = np.arange(pk[0], pk[0] + n)
+            pk[0] += n
+            pushes.append((side, mk_chunk([T_I64, T_I64],
+                                          np.zeros(n, np.uint8), [k, v])))
+        return pushes
+
+    a = mk()
+    state = {s: b"" for s in (SIDE_LEFT, SIDE_RIGHT)}
+    degs = {s: b"" for s in (SIDE_LEFT, SIDE_RIGHT)}
+    streams = []
+    for e in range(3):
+        for side, c in epoch_pushes(e):
+            a.push(side, c)
+            a.poll_all()
+        # watermark advances on both sides -> clean below e*10
+        a.watermark(SIDE_LEFT, 0, e * 10)
+        a.watermark(SIDE_RIGHT, 0, e * 10)
+        for s in (SIDE_LEFT, SIDE_RIGHT):
+            d = join_checkpoint_drain(lib, a.h, s)
+            state[s] += d
+            streams.append(d)
+            g = join_degree_drain(lib, a.h, s)
+            degs[s] += g
+            streams.append(g)
+    # crash + restore
+    b = mk()
+    for s in (SIDE_LEFT, SIDE_RIGHT):
+        join_restore(lib, b.h, s, state[s], degs[s])
+    # replay the SAME watermark state on the restored executor (the
+    # embedder re-establishes watermarks on recovery, as the reference's
+    # barrier/actor context does)
+    b.watermark(SIDE_LEFT, 0, 2 * 10)
+    b.watermark(SIDE_RIGHT, 0, 2 * 10)
+    for e in range(3, 6):
+        for side, c in epoch_pushes(e):
+            a.push(side, c)
+            b.push(side, c)
+            ma = rows_multiset(a.poll_all())
+            mb = rows_multiset(b.poll_all())
+            assert ma == mb, f"epoch {e}: restored join diverged after clean"
+        a.watermark(SIDE_LEFT, 0, e * 10)
+        a.watermark(SIDE_RIGHT, 0, e * 10)
+        b.watermark(SIDE_LEFT, 0, e * 10)
+        b.watermark(SIDE_RIGHT, 0, e * 10)
+        for s in (SIDE_LEFT, SIDE_RIGHT):
+            da = join_checkpoint_drain(lib, a.h, s)
+            db = join_checkpoint_drain(lib, b.h, s)
+            assert da == db, f"epoch {e} side {s}: drain diverged after clean"
+            streams.append(da)
+            ga = join_degree_drain(lib, a.h, s)
+            gb = join_degree_drain(lib, b.h, s)
+            assert ga == gb, f"epoch {e} side {s}: degree drain diverged"
+            streams.append(ga)
+    a.close()
+    b.close()
+    return streams
+
+
+def test_join_clean_restore_oracle():
+    _join_clean_restore_flow(oracle(), JOIN_INNER)
+
+
+def test_join_clean_restore_oracle_semi():
+    _join_clean_restore_flow(oracle(), JOIN_LEFT_SEMI)
+
+
+@pytest.mark.gpu
+def test_join_clean_restore_gpu_and_parity():
+    import risingwave_amd
+
+    risingwave_amd.load_library()
+    glib = ffi.Lib(risingwave_amd.lib_path())
+    for jt in (JOIN_INNER, JOIN_LEFT_SEMI):
+        sg = _join_clean_restore_flow(glib, jt)
+        so = _join_clean_restore_flow(oracle(), jt)
+        assert sg == so, f"join type {jt}: clean drain streams diverged"
+
+
+def _agg_clean_restore_flow(lib):
+    """Agg watermark cleaning: intermediate DELETE frames, minput row
+    deltas, and dedup count resets must all reach the spill streams."""
+    from rwtest.ffi import (AGG_COUNT, AGG_MIN, agg_dedup_drain_bytes,
+                            agg_dedup_restore, agg_minput_drain_bytes,
+                            agg_minput_restore)
+
+    # group col 0 is watermarked; MIN is materialized-input; COUNT
+    # DISTINCT on col 2 exercises the dedup table
+    calls = [(AGG_SUM, 1, T_I64), (AGG_MIN, 1, T_I64),
+             (AGG_COUNT, 2, T_I64, 1), (AGG_COUNT_STAR, -1, T_I64)]
+    mk = lambda: ffi.HashAgg(lib, [T_I64, T_I64, T_I64, T_I64], [0], calls,
+                             3, stream_key=(3,))
+    rng = np.random.default_rng(909)
+    rid = [0]
+
+    def epoch_chunks(e):
+        out = []
+        for _ in range(2):
+            n = 200
+            g = rng.integers(e * 10, e * 10 + 25, n)
+            v = rng.integers(-100, 100, n)
+            d = rng.integers(0, 6, n)
+            r2 = np.arange(rid[0], rid[0] + n)
+            rid[0] += n
+            out.append(mk_chunk([T_I64] * 4, np.zeros(n, np.uint8),
+                                [g, v, d, r2]))
+        return out
+
+    a = mk()
+    inter, minp, ded = b"", b"", b""
+    streams = []
+    for e in range(3):
+        _drive_agg(a, epoch_chunks(e), e + 1)
+        a.watermark(0, e * 10)
+        d1 = agg_checkpoint_drain_bytes(lib, a.h)
+        d2 = agg_minput_drain_bytes(lib, a.h, 0)
+        d3 = agg_dedup_drain_bytes(lib, a.h, 0)
+        inter += d1
+        minp += d2
+        ded += d3
+        streams += [d1, d2, d3]
+    b = mk()
+    agg_minput_restore(lib, b.h, 0, minp)
+    agg_dedup_restore(lib, b.h, 0, ded)
+    agg_restore(lib, b.h, inter)
+    for e in range(3, 6):
+        cs = epoch_chunks(e)
+        oa = _drive_agg(a, cs, e + 1)
+        ob = _drive_agg(b, cs, e + 1)
+        assert oa == ob, f"epoch {e}: restored agg diverged after clean"
+        a.watermark(0, e * 10)
+        b.watermark(0, e * 10)
+        for fn in (agg_checkpoint_drain_bytes,
+                   lambda l, h: agg_minput_drain_bytes(l, h, 0),
+                   lambda l, h: agg_dedup_drain_bytes(l, h, 0)):
+            da = fn(lib, a.h)
+            db = fn(lib, b.h)
+            assert da == db, f"epoch {e}: drain diverged after clean"
+            streams.append(da)
+    a.close()
+    b.close()
+    return streams
+
+
+def test_agg_clean_restore_oracle():
+    _agg_clean_restore_flow(oracle())
+
+
+@pytest.mark.gpu
+def test_agg_clean_restore_gpu_and_parity():
+    import risingwave_amd
+
+    risingwave_amd.load_library()
+    glib = ffi.Lib(risingwave_amd.lib_path())
+    sg = _agg_clean_restore_flow(glib)
+    so = _agg_clean_restore_flow(oracle())
+    assert sg == so, "agg clean drain streams diverged from oracle"
+
+
 def test_agg_minput_drain_parity_cpu_noop():
     # sanity: a fresh executor drains empty minput tables
     from rwtest.ffi import agg_minput_drain_bytes
