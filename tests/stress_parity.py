@@ -238,6 +238,67 @@ def stress_topn_ties(seed, n):
     o.close()
 
 
+def stress_clean_restore(seed, n):
+    """Randomized watermark cleaning + crash/restore: rising key ranges,
+    random watermark advances (state cleaning w/ spill deltas), a restore
+    from the accumulated drains mid-run, byte-compared GPU vs oracle."""
+    rng = np.random.default_rng(seed + 71)
+    jt = int(rng.integers(0, 2)) * 4  # INNER or LEFT_SEMI (degrees)
+    mkj = lambda lib: ffi.HashJoin(lib, jt, [T_I64, T_I64], [T_I64, T_I64],
+                                   key_l=[0], key_r=[0], pk_l=[1], pk_r=[1],
+                                   wm_jk=((0, True),))
+    g, o = mkj(GPU), mkj(oracle())
+    state = {0: [b"", b""], 1: [b"", b""]}  # side -> [gpu, orc]
+    degs = {0: [b"", b""], 1: [b"", b""]}
+    pk = 0
+    wm = 0
+    for i in range(6):
+        base = i * 12
+        for side in (SIDE_LEFT, SIDE_RIGHT):
+            keys = rng.integers(base, base + 30, n)
+            vals = np.arange(pk, pk + n)
+            pk += n
+            c = mk([T_I64, T_I64], np.zeros(n, np.uint8), [keys, vals])
+            g.push(side, c)
+            o.push(side, c)
+            from test_gpu_parity import net_rows
+
+            assert net_rows(rows_multiset(g.poll_all())) == net_rows(
+                rows_multiset(o.poll_all())), \
+                f"clean seed {seed} push {i} side {side} (jt={jt})"
+        if rng.random() < 0.7:
+            wm = base + int(rng.integers(0, 10))
+            for a in (g, o):
+                a.watermark(SIDE_LEFT, 0, wm)
+                a.watermark(SIDE_RIGHT, 0, wm)
+        for side in (SIDE_LEFT, SIDE_RIGHT):
+            sg = ffi.join_checkpoint_drain(GPU, g.h, side)
+            so = ffi.join_checkpoint_drain(oracle(), o.h, side)
+            assert sg == so, f"clean spill seed {seed} push {i} (jt={jt})"
+            dg = ffi.join_degree_drain(GPU, g.h, side)
+            do = ffi.join_degree_drain(oracle(), o.h, side)
+            assert dg == do, f"clean degree seed {seed} push {i} (jt={jt})"
+            state[side][0] += sg
+            state[side][1] += so
+            degs[side][0] += dg
+            degs[side][1] += do
+        if i == 3:
+            # crash: swap in executors restored from the drain streams
+            g.close()
+            o.close()
+            g, o = mkj(GPU), mkj(oracle())
+            for side in (SIDE_LEFT, SIDE_RIGHT):
+                ffi.join_restore(GPU, g.h, side, state[side][0],
+                                 degs[side][0])
+                ffi.join_restore(oracle(), o.h, side, state[side][1],
+                                 degs[side][1])
+            for a in (g, o):
+                a.watermark(SIDE_LEFT, 0, wm)
+                a.watermark(SIDE_RIGHT, 0, wm)
+    g.close()
+    o.close()
+
+
 def main():
     n_seeds = int(sys.argv[1]) if len(sys.argv) > 1 else 5
     n = int(sys.argv[2]) if len(sys.argv) > 2 else 2048
@@ -249,7 +310,9 @@ def main():
         stress_eowc(seed, max(n // 2, 512))
         stress_distinct(seed, max(n // 2, 512))
         stress_topn_ties(seed, max(n // 4, 256))
-        print(f"seed {seed}: agg + 8 joins + topn + eowc + distinct + ties OK")
+        stress_clean_restore(seed, max(n // 2, 512))
+        print(f"seed {seed}: agg + 8 joins + topn + eowc + distinct + ties"
+              " + clean/restore OK")
     print(f"STRESS OK: {n_seeds} seeds")
 
 
