@@ -304,6 +304,11 @@ int rw_topn_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len);
  * surviving full rows rebuild the per-group state, marked persisted so the
  * next drain does not re-PUT them). Must run before any input. */
 int rw_topn_restore(void* h, const uint8_t* buf, uint64_t len);
+/* handle_watermark (group_top_n.rs:266-273): a watermark on input column
+ * `col_idx` cleans state below `val` and is forwarded iff col_idx is the
+ * FIRST group-by column. Returns 1 = forwarded (state cleaned; retired
+ * rows net to DELETE in the next drain), 0 = absorbed, <0 = error. */
+int rw_group_top_n_watermark(void* h, uint32_t col_idx, int64_t val);
 /* DISTINCT dedup tables (one StateTable per distinct column in the
  * reference, from_proto/hash_agg.rs distinct_dedup_tables): pk = group key
  * ∥ datum, value = full row ++ one i64 count per call distincting on the

@@ -269,6 +269,28 @@ struct GroupTopNOracle {
         return chunk_to_c(*c);
     }
 
+    // handle_watermark (group_top_n.rs:266-273): a watermark on
+    // group_by[0] cleans the state table below it (range delete on the
+    // store) and is forwarded; other columns' watermarks are absorbed.
+    // Cleaned rows net to DELETE spill frames — a restore replay must
+    // not resurrect them.
+    int watermark(uint32_t col_idx, int64_t val) {
+        if (group_by.empty() || group_by[0] != col_idx) return 0;
+        for (auto it = groups.begin(); it != groups.end();) {
+            const Datum& d = it->first[0];
+            if (!d.null && d.i < val) {
+                for (auto& [ck, row] : it->second) {
+                    (void)row;
+                    delta_delete(it->first, ck);
+                }
+                it = groups.erase(it);
+            } else {
+                ++it;
+            }
+        }
+        return 1;
+    }
+
     // §8f-5 recovery: net concatenated drain frames (PUT last-write-wins,
     // DELETE removes — KV compaction semantics) and rebuild `groups` from
     // the surviving value-encoded full rows. Restored rows predate the
@@ -339,6 +361,9 @@ int rw_group_top_n_flush(void* h, uint64_t epoch) {
 RwChunk* rw_group_top_n_poll(void* h) { return ((GroupTopNOracle*)h)->poll(); }
 int rw_topn_restore(void* h, const uint8_t* buf, uint64_t len) {
     return ((GroupTopNOracle*)h)->restore(buf, len);
+}
+int rw_group_top_n_watermark(void* h, uint32_t col_idx, int64_t val) {
+    return ((GroupTopNOracle*)h)->watermark(col_idx, val);
 }
 int rw_topn_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len) {
     std::vector<uint8_t> sp;

@@ -7666,6 +7666,34 @@ __global__ void topn_emit_kernel(JoinSideDev sd, TopMeta m,
     }
 }
 
+// GroupTopN watermark state cleaning (group_top_n.rs:266-273: a watermark
+// on group_by[0] becomes a state-table watermark = range delete on the
+// store). Rows of groups below the watermark are retired in place and
+// routed through the kill list so the next drain nets them to DELETE
+// frames (slots stay READY; a late row restarts the group).
+__global__ void topn_clean_kernel(JoinSideDev sd, int kpos, long long wm) {
+    size_t cap = (size_t)sd.cap_mask + 1;
+    size_t stride = (size_t)gridDim.x * blockDim.x;
+    for (size_t slot = blockIdx.x * blockDim.x + threadIdx.x; slot < cap;
+         slot += stride) {
+        if (sd.slots[slot].state != SLOT_READY) continue;
+        if ((sd.slots[slot].nulls >> kpos) & 1) continue; // NULLs largest
+        if (sd.slots[slot].key[kpos] >= wm) continue;
+        uint32_t row = sd.slots[slot].head;
+        while (row != UINT32_MAX) {
+            JoinRowHdr* h = jrow(sd, row);
+            if (h->alive) {
+                h->alive = 0;
+                if (sd.killed) {
+                    uint32_t kidx = atomicAdd(sd.killed_cursor, 1u);
+                    if (kidx < sd.killed_cap) sd.killed[kidx] = row;
+                }
+            }
+            row = h->next;
+        }
+    }
+}
+
 // §8f-5 recovery: rebuild the TopN state table from a netted spill replay.
 // Each restored row is appended to the record store and chained onto its
 // group's slot (cache keys in the netted map are unique, so no same-key
@@ -8204,6 +8232,18 @@ void* rw_group_top_n_create(const RwGroupTopNDesc* d) {
         return nullptr;
     }
     return t;
+}
+
+// handle_watermark (group_top_n.rs:266-273): a watermark on the FIRST
+// group-by column cleans the state table below it and is forwarded
+// (returns 1); any other column's watermark is absorbed (returns 0).
+int rw_group_top_n_watermark(void* h, uint32_t col_idx, int64_t val) {
+    auto* t = (GroupTopN*)h;
+    if (t->m.KW == 0 || t->m.gk_cols[0] != (uint8_t)col_idx) return 0;
+    topn_clean_kernel<<<2048, 256, 0, t->stream>>>(t->sd, 0, val);
+    if (hipStreamSynchronize(t->stream) != hipSuccess)
+        FAIL(RW_E_INTERNAL, "topn clean sync failed");
+    return 1;
 }
 int rw_group_top_n_push_chunk(void* h, const RwChunk* c) {
     return ((GroupTopN*)h)->push_chunk(c);

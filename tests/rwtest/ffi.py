@@ -565,6 +565,17 @@ class GroupTopN:
         if rc != 0:
             raise RuntimeError(f"push failed {rc}: {self.lib.last_error()}")
 
+    def watermark(self, col_idx, val):
+        L = self.lib.lib
+        L.rw_group_top_n_watermark.restype = C.c_int
+        L.rw_group_top_n_watermark.argtypes = [C.c_void_p, C.c_uint32,
+                                               C.c_int64]
+        rc = L.rw_group_top_n_watermark(self.h, col_idx, val)
+        if rc < 0:
+            raise RuntimeError(
+                f"topn watermark failed {rc}: {self.lib.last_error()}")
+        return rc  # 1 = forwarded, 0 = absorbed
+
     def flush(self, epoch):
         rc = self.lib.lib.rw_group_top_n_flush(self.h, epoch)
         if rc != 0:

@@ -615,6 +615,77 @@ def _agg_clean_restore_flow(lib):
     return streams
 
 
+def _topn_clean_restore_flow(lib):
+    """GroupTopN watermark cleaning (group_top_n.rs:266-273): rows of
+    groups below the watermark net to DELETE spill frames; restore +
+    continue matches an uninterrupted run."""
+    from rwtest.ffi import topn_checkpoint_drain, topn_restore
+
+    mk = lambda: ffi.GroupTopN(lib, [T_I64, T_I64, T_I64], [0],
+                               [(1, False)], [(2, False)], limit=3)
+    rng = np.random.default_rng(515)
+    rid = [0]
+
+    def epoch_chunks(e):
+        n = 150
+        g = rng.integers(e * 8, e * 8 + 20, n)
+        v = rng.integers(0, 100, n)
+        r2 = np.arange(rid[0], rid[0] + n)
+        rid[0] += n
+        return [mk_chunk([T_I64] * 3, np.zeros(n, np.uint8), [g, v, r2])]
+
+    a = mk()
+    state = b""
+    streams = []
+    wm = 0
+    for e in range(3):
+        for c in epoch_chunks(e):
+            a.push(c)
+        a.poll_all()
+        wm = e * 8
+        assert a.watermark(0, wm) == 1  # group col -> forwarded
+        assert a.watermark(1, 10**9) == 0  # non-group col -> absorbed
+        d = topn_checkpoint_drain(lib, a.h)
+        state += d
+        streams.append(d)
+    b = mk()
+    topn_restore(lib, b.h, state)
+    b.watermark(0, wm)
+    for e in range(3, 6):
+        cs = epoch_chunks(e)
+        for c in cs:
+            a.push(c)
+            b.push(c)
+            ma = rows_multiset(a.poll_all())
+            mb = rows_multiset(b.poll_all())
+            assert ma == mb, f"epoch {e}: restored topn diverged after clean"
+        wm = e * 8
+        a.watermark(0, wm)
+        b.watermark(0, wm)
+        da = topn_checkpoint_drain(lib, a.h)
+        db = topn_checkpoint_drain(lib, b.h)
+        assert da == db, f"epoch {e}: topn drain diverged after clean"
+        streams.append(da)
+    a.close()
+    b.close()
+    return streams
+
+
+def test_topn_clean_restore_oracle():
+    _topn_clean_restore_flow(oracle())
+
+
+@pytest.mark.gpu
+def test_topn_clean_restore_gpu_and_parity():
+    import risingwave_amd
+
+    risingwave_amd.load_library()
+    glib = ffi.Lib(risingwave_amd.lib_path())
+    sg = _topn_clean_restore_flow(glib)
+    so = _topn_clean_restore_flow(oracle())
+    assert sg == so, "topn clean drain streams diverged from oracle"
+
+
 def test_agg_clean_restore_oracle():
     _agg_clean_restore_flow(oracle())
 
