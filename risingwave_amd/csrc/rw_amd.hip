@@ -2753,8 +2753,11 @@ struct HashAgg {
         return RW_OK;
     }
 
-    // EOWC spill this round: one DELETE per closed (emitted) window —
-    // mid-window state PUTs are a later-round item (matches the oracle).
+    // EOWC close spill: one DELETE per closed (emitted) window. Paired
+    // with agg_eowc_dump_kernel's per-barrier mid-window PUTs of the
+    // dirty groups' current states (hash_agg.rs:429-460), the stream is
+    // the full state-table view — an EOWC executor restores from it
+    // (tests/test_restore.py::test_agg_eowc_restore_*).
     void spill_records_eowc(const std::vector<long long>& vals,
                             const std::vector<uint8_t>& nulls,
                             uint32_t n_out) {

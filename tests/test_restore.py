@@ -671,6 +671,66 @@ def _topn_clean_restore_flow(lib):
     return streams
 
 
+def _agg_eowc_restore_flow(lib):
+    """EOWC executors restore from their drain stream (mid-window PUTs +
+    close DELETEs are the full state-table view, hash_agg.rs:429-474)."""
+    calls = [(AGG_COUNT_STAR, -1, T_I64), (AGG_SUM, 1, T_I64)]
+    mk = lambda: ffi.HashAgg(lib, [T_I64, T_I64], [0], calls, 0,
+                             emit_on_window_close=True)
+    rng = np.random.default_rng(42)
+    a = mk()
+    d = b""
+    wm = 0
+    outs = []
+    for ep in range(3):
+        n = 200
+        k = rng.integers(wm, wm + 40, n)
+        v = rng.integers(1, 9, n)
+        a.push(mk_chunk([T_I64] * 2, np.zeros(n, np.uint8), [k, v]))
+        wm += 20
+        a.watermark(0, wm)
+        a.flush(ep + 1)
+        a.poll_all()
+        d += agg_checkpoint_drain_bytes(lib, a.h)
+    b = mk()
+    agg_restore(lib, b.h, d)
+    for ep in range(3, 6):
+        n = 200
+        k = rng.integers(wm, wm + 40, n)
+        v = rng.integers(1, 9, n)
+        c = mk_chunk([T_I64] * 2, np.zeros(n, np.uint8), [k, v])
+        wm += 20
+        for x in (a, b):
+            x.push(c)
+            x.watermark(0, wm)
+            x.flush(ep + 1)
+        ma = rows_multiset(a.poll_all())
+        mb = rows_multiset(b.poll_all())
+        assert ma == mb, f"epoch {ep}: restored EOWC agg diverged"
+        da = agg_checkpoint_drain_bytes(lib, a.h)
+        db = agg_checkpoint_drain_bytes(lib, b.h)
+        assert da == db, f"epoch {ep}: EOWC drain diverged"
+        outs.append((ma, da))
+    a.close()
+    b.close()
+    return outs
+
+
+def test_agg_eowc_restore_oracle():
+    _agg_eowc_restore_flow(oracle())
+
+
+@pytest.mark.gpu
+def test_agg_eowc_restore_gpu_and_parity():
+    import risingwave_amd
+
+    risingwave_amd.load_library()
+    glib = ffi.Lib(risingwave_amd.lib_path())
+    og = _agg_eowc_restore_flow(glib)
+    oo = _agg_eowc_restore_flow(oracle())
+    assert og == oo, "EOWC restore flow diverged from oracle"
+
+
 def test_topn_clean_restore_oracle():
     _topn_clean_restore_flow(oracle())
 
