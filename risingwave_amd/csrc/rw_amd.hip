@@ -4344,6 +4344,40 @@ __device__ __forceinline__ void jemit_row(JoinOutDev& out, const JoinMeta& m,
     }
 }
 
+// jemit_row over a REGISTER-RESIDENT match record (see the register walk
+// in join_probe_kernel): match values come from the register copy via
+// unrolled selects — a dynamic index into a private array would demote
+// it to scratch and defeat the single-fetch walk.
+__device__ __forceinline__ void jemit_row_rec(JoinOutDev& out,
+                                              const JoinMeta& m, int S,
+                                              uint8_t op,
+                                              const JoinBatchDev& b,
+                                              uint32_t r, uint32_t match_vb,
+                                              const long long* recv,
+                                              int form) {
+    uint32_t orow = atomicAdd(&out.counters[0], 1u);
+    if (orow >= out.cap) {
+        atomicExch(&out.counters[1], 1u);
+        return;
+    }
+    out.ops[orow] = op;
+    for (int c = 0; c < m.n_out; c++) {
+        bool from_probe = (int)m.out_src[c] == S;
+        uint8_t col = m.out_col[c];
+        int64_t v = 0;
+        uint8_t valid = 0;
+        if (from_probe && (form & 1)) {
+            valid = b.col_valid[col][r];
+            v = b.col_vals[col][r];
+        } else if (!from_probe && (form & 2)) {
+            valid = (match_vb >> col) & 1;
+            v = rec_sel(recv, col);
+        }
+        out.vals[(size_t)c * out.cap + orow] = valid ? v : 0;
+        out.nulls[(size_t)c * out.cap + orow] = !valid;
+    }
+}
+
 // own-side insert/delete (join/hash_join.rs:591-681 without the LRU tier)
 __device__ __forceinline__ void jown_insert(JoinSideDev own, const JoinMeta& m,
                                             int S, const JoinBatchDev& b,
@@ -4455,12 +4489,48 @@ __device__ void join_probe_row_noninner(const JoinBatchDev& b, JoinSideDev own,
         jbucket_head(match, hash_key(kw, nullmask, m.KW), false);
     uint32_t my_deg = 0;
     {
+        // REGISTER-RESIDENT walk for narrow aligned records (same refetch
+        // pathology and fix as the inner probe's walk);
+        // wide records keep the field walk. The degree update stays an
+        // atomic on the record itself (other lanes observe it through the
+        // atomic's return value, never through their register copies).
+        const bool regw =
+            match.row_stride <= 64 && !(match.row_stride & 15);
+        const int nw16 = (int)(match.row_stride >> 4);
+        long long rec[8];
         uint32_t row = mhead;
         while (row != UINT32_MAX) {
             JoinRowHdr* h = jrow(match, row);
-            if (h->alive &&
-                jhdr_key_eq(h, m.key_cols[1 - S], m.KW, kw, nullmask, false) &&
-                join_cond_ok(m, S, b, r, h->validbits, jvals(h))) {
+            uint32_t alive, nxt, vb;
+            if (regw) {
+                const long long* rp = (const long long*)h;
+#pragma unroll
+                for (int w = 0; w < 4; w++)
+                    if (w < nw16)
+                        *(longlong2*)&rec[2 * w] =
+                            *(const longlong2*)(rp + 2 * w);
+                alive = (uint32_t)(uint64_t)rec[0];
+                nxt = (uint32_t)((uint64_t)rec[0] >> 32);
+                vb = (uint32_t)(uint64_t)rec[1];
+            } else {
+                alive = h->alive;
+                nxt = h->next;
+                vb = h->validbits;
+            }
+            bool eq = alive != 0;
+            for (int i = 0; eq && i < m.KW; i++) {
+                uint8_t col = m.key_cols[1 - S][i];
+                bool valid = (vb >> col) & 1;
+                if (valid == (bool)((nullmask >> i) & 1)) {
+                    eq = false;
+                } else if (valid) {
+                    long long v =
+                        regw ? rec_sel(rec + 2, col) : jvals(h)[col];
+                    if (v != kw[i]) eq = false;
+                }
+            }
+            if (eq && (regw ? join_cond_ok_rec(m, S, b, r, vb, rec + 2)
+                            : join_cond_ok(m, S, b, r, vb, jvals(h)))) {
                 my_deg++;
                 bool zero = false;
                 if (m.need_deg[1 - S]) {
@@ -4478,37 +4548,38 @@ __device__ void join_probe_row_noninner(const JoinBatchDev& b, JoinSideDev own,
                         match.deg_dirty_list[di] = row;
                     }
                 }
-                const long long* mv = jvals(h);
-                uint32_t vb = h->validbits;
+                auto emit2 = [&](uint8_t op2, int form) {
+                    if (regw)
+                        jemit_row_rec(out, m, S, op2, b, r, vb, rec + 2,
+                                      form);
+                    else
+                        jemit_row(out, m, S, op2, b, r, vb, jvals(h), form);
+                };
                 if (is_insert) {
                     if (anti) {
-                        if (zero && only_fwd_m)
-                            jemit_row(out, m, S, RW_OP_DELETE, b, r, vb, mv, 2);
+                        if (zero && only_fwd_m) emit2(RW_OP_DELETE, 2);
                     } else if (semi) {
-                        if (zero && only_fwd_m)
-                            jemit_row(out, m, S, RW_OP_INSERT, b, r, vb, mv, 2);
+                        if (zero && only_fwd_m) emit2(RW_OP_INSERT, 2);
                     } else if (zero && outer_null) {
-                        jemit_row(out, m, S, RW_OP_DELETE, b, r, vb, mv, 2);
-                        jemit_row(out, m, S, RW_OP_INSERT, b, r, vb, mv, 3);
+                        emit2(RW_OP_DELETE, 2);
+                        emit2(RW_OP_INSERT, 3);
                     } else if (!fwd_once) {
-                        jemit_row(out, m, S, RW_OP_INSERT, b, r, vb, mv, 3);
+                        emit2(RW_OP_INSERT, 3);
                     }
                 } else {
                     if (anti) {
-                        if (zero && only_fwd_m)
-                            jemit_row(out, m, S, RW_OP_INSERT, b, r, vb, mv, 2);
+                        if (zero && only_fwd_m) emit2(RW_OP_INSERT, 2);
                     } else if (semi) {
-                        if (zero && only_fwd_m)
-                            jemit_row(out, m, S, RW_OP_DELETE, b, r, vb, mv, 2);
+                        if (zero && only_fwd_m) emit2(RW_OP_DELETE, 2);
                     } else if (zero && outer_null) {
-                        jemit_row(out, m, S, RW_OP_DELETE, b, r, vb, mv, 3);
-                        jemit_row(out, m, S, RW_OP_INSERT, b, r, vb, mv, 2);
+                        emit2(RW_OP_DELETE, 3);
+                        emit2(RW_OP_INSERT, 2);
                     } else if (!fwd_once) {
-                        jemit_row(out, m, S, RW_OP_DELETE, b, r, vb, mv, 3);
+                        emit2(RW_OP_DELETE, 3);
                     }
                 }
             }
-            row = h->next;
+            row = nxt;
         }
     }
     if (my_deg == 0) {
@@ -4552,6 +4623,36 @@ __global__ void join_count_emitted_kernel(const uint8_t* ops, uint32_t n,
         atomicAdd(out_count, (unsigned long long)(counters[0] - n));
 }
 
+// non-inner probe (outer/semi/anti): one thread per row over the shared
+// per-row prologue, then the matched-side walk + degree/emission tree in
+// join_probe_row_noninner. Separate kernel so its wider live range
+// (register-resident records + the emission decision tree) gets its own
+// launch bounds instead of spilling the inner kernel.
+__global__ __launch_bounds__(256, 4) void join_probe_noninner_kernel(
+    JoinBatchDev b, JoinSideDev own, JoinSideDev match, JoinMeta m, int S,
+    JoinOutDev out, uint32_t r0, uint32_t r1, int dbg_skip) {
+    uint32_t stride = gridDim.x * blockDim.x;
+    for (uint32_t r = r0 + blockIdx.x * blockDim.x + threadIdx.x; r < r1;
+         r += stride) {
+        if (b.vis && !b.vis[r]) continue;
+        uint8_t op_in = b.ops[r];
+        bool is_insert = (op_in == RW_OP_INSERT || op_in == RW_OP_UPDATE_INSERT);
+        uint8_t op = is_insert ? RW_OP_INSERT : RW_OP_DELETE;
+        int64_t kw[MAX_KW];
+        uint32_t nullmask = 0;
+        for (int i = 0; i < m.KW; i++) {
+            uint8_t col = m.key_cols[S][i];
+            bool valid = b.col_valid[col][r];
+            kw[i] = valid ? b.col_vals[col][r] : 0;
+            nullmask |= (!valid) << i;
+        }
+        bool never_match = (nullmask & ~(uint32_t)m.null_safe_mask) != 0;
+        if (!(dbg_skip & 2))
+            join_probe_row_noninner(b, own, match, m, S, out, r, is_insert,
+                                    op, kw, nullmask, never_match);
+    }
+}
+
 __global__ __launch_bounds__(256, 6) void join_probe_kernel(
     JoinBatchDev b, JoinSideDev own, JoinSideDev match, JoinMeta m, int S,
     JoinOutDev out, uint32_t r0, uint32_t r1, const uint32_t* row_base,
@@ -4583,12 +4684,9 @@ __global__ __launch_bounds__(256, 6) void join_probe_kernel(
             if (nullmask & ~(uint32_t)m.null_safe_mask) never_match = true;
         }
 
-        if (m.join_type != RW_JOIN_INNER) {
-            if (active && !(dbg_skip & 2))
-                join_probe_row_noninner(b, own, match, m, S, out, r, is_insert,
-                                        op, kw, nullmask, never_match);
-            continue;
-        }
+        // non-inner types run in join_probe_noninner_kernel (own launch
+        // bounds: its register walk + emission tree would spill this
+        // kernel's budget)
         if (never_match) active = false;
 
         uint32_t mhead = UINT32_MAX;
@@ -6209,6 +6307,9 @@ struct HashJoin {
                     side[s], side[1 - s], m, s, out, d_part_base, d_row_base,
                     out.counters + 1);
             }
+        } else if (m.join_type != RW_JOIN_INNER) {
+            join_probe_noninner_kernel<<<blocks, 256, 0, stream>>>(
+                b, side[s], side[1 - s], m, s, out, r0, r1, dbg_skip);
         } else {
             join_probe_kernel<<<blocks, 256, 0, stream>>>(
                 b, side[s], side[1 - s], m, s, out, r0, r1, d_probe_base,
