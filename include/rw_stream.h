@@ -298,6 +298,13 @@ int rw_agg_minput_restore(void* h, int mi, const uint8_t* buf, uint64_t len);
 int rw_hash_join_restore(void* h, int side, const uint8_t* buf, uint64_t len,
                          const uint8_t* deg_buf, uint64_t deg_len);
 int rw_join_checkpoint_drain(void* h, int side, uint8_t** buf, uint64_t* len);
+/* §8f-4 memory reclamation: rebuild one side's record store and buckets
+ * without dead records (retractions / watermark sweeps retire rows in
+ * place; the reference reclaims them in Hummock compaction). Call
+ * between epochs AFTER the side's checkpoint + degree drains (fails
+ * loudly on pending deltas or undrained rows). Logical state, drains and
+ * restore are unaffected; `reclaimed` (optional) reports freed bytes. */
+int rw_join_compact(void* h, int side, uint64_t* reclaimed);
 int rw_topn_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len);
 /* Replay concatenated rw_topn_checkpoint_drain streams into a freshly
  * created GroupTopN executor (PUT/DELETE frames net host-side; the

@@ -689,6 +689,14 @@ int rw_hash_join_update_vnode_bitmap(void* h, const uint8_t* bitmap,
     return ((HashJoinOracle*)h)->update_vnode_bitmap(bitmap, vnode_count);
 }
 
+int rw_join_compact(void* h, int side, uint64_t* reclaimed) {
+    // the oracle's maps hold no dead rows — compaction is a no-op with
+    // identical observable state (the GPU build reclaims retired records)
+    if (!h || (side != 0 && side != 1)) return RW_E_INVAL;
+    if (reclaimed) *reclaimed = 0;
+    return RW_OK;
+}
+
 int rw_hash_join_watermark(void* h, int side, uint32_t col_idx, int64_t val,
                            uint32_t* out_cols, int64_t* out_vals, int max_out) {
     return ((HashJoinOracle*)h)->watermark(side, col_idx, val, out_cols,

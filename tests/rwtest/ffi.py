@@ -656,6 +656,19 @@ def topn_checkpoint_drain(lib, h):
     return out
 
 
+def join_compact(lib, h, side):
+    """Reclaim one join side's dead records; returns freed bytes."""
+    L = lib.lib
+    L.rw_join_compact.restype = C.c_int
+    L.rw_join_compact.argtypes = [C.c_void_p, C.c_int,
+                                  C.POINTER(C.c_uint64)]
+    freed = C.c_uint64()
+    rc = L.rw_join_compact(h, side, C.byref(freed))
+    if rc != 0:
+        raise RuntimeError(f"join compact failed {rc}: {lib.last_error()}")
+    return freed.value
+
+
 def topn_restore(lib, h, buf):
     """Rebuild GroupTopN state from concatenated drain bytes (rw_stream.h)."""
     L = lib.lib

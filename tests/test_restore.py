@@ -530,6 +530,97 @@ def _join_clean_restore_flow(lib, join_type):
     return streams
 
 
+def _join_compact_flow(lib, join_type):
+    """rw_join_compact between epochs: logical state, emissions, drains
+    and restores are unchanged while dead records are reclaimed."""
+    from rwtest.ffi import join_compact
+
+    mk = lambda: ffi.HashJoin(lib, join_type, [T_I64, T_I64],
+                              [T_I64, T_I64], key_l=[0], key_r=[0],
+                              pk_l=[1], pk_r=[1], wm_jk=((0, True),))
+    rng = np.random.default_rng(2727)
+    pk = [0]
+    live = {SIDE_LEFT: [], SIDE_RIGHT: []}
+
+    def epoch_pushes(e):
+        pushes = []
+        for side in (SIDE_LEFT, SIDE_RIGHT):
+            n = 250
+            k = rng.integers(e * 10, e * 10 + 30, n)
+            v = np.arange(pk[0], pk[0] + n)
+            pk[0] += n
+            ops = np.zeros(n, np.uint8)
+            for r in range(n):
+                if live[side] and rng.random() < 0.3:
+                    jx = int(rng.integers(0, len(live[side])))
+                    k[r], v[r] = live[side].pop(jx)
+                    ops[r] = ffi.OP_DELETE
+                else:
+                    live[side].append((int(k[r]), int(v[r])))
+            pushes.append((side, mk_chunk([T_I64, T_I64], ops, [k, v])))
+        return pushes
+
+    # a compacts every epoch, c never does; both track the oracle flow
+    a, c = mk(), mk()
+    freed_total = 0
+    for e in range(5):
+        for side, ch in epoch_pushes(e):
+            for x in (a, c):
+                x.push(side, ch)
+            ma = rows_multiset(a.poll_all())
+            mc = rows_multiset(c.poll_all())
+            assert ma == mc, f"epoch {e}: compacted join diverged"
+        wm = e * 10
+        for x in (a, c):
+            x.watermark(SIDE_LEFT, 0, wm)
+            x.watermark(SIDE_RIGHT, 0, wm)
+        for s in (SIDE_LEFT, SIDE_RIGHT):
+            da = join_checkpoint_drain(lib, a.h, s)
+            dc = join_checkpoint_drain(lib, c.h, s)
+            assert da == dc, f"epoch {e} side {s}: drain diverged"
+            ga = join_degree_drain(lib, a.h, s)
+            gc = join_degree_drain(lib, c.h, s)
+            assert ga == gc, f"epoch {e} side {s}: degree drain diverged"
+            freed_total += join_compact(lib, a.h, s)
+    a.close()
+    c.close()
+    return freed_total
+
+
+def test_join_compact_oracle_noop():
+    assert _join_compact_flow(oracle(), JOIN_INNER) == 0
+
+
+@pytest.mark.gpu
+def test_join_compact_gpu():
+    import risingwave_amd
+
+    risingwave_amd.load_library()
+    glib = ffi.Lib(risingwave_amd.lib_path())
+    # deletes + watermark sweeps retire rows, so compaction must free
+    freed = _join_compact_flow(glib, JOIN_INNER)
+    assert freed > 0, "compaction reclaimed nothing despite retired rows"
+    _join_compact_flow(glib, JOIN_LEFT_SEMI)
+
+
+@pytest.mark.gpu
+def test_join_compact_requires_drained():
+    import risingwave_amd
+    from rwtest.ffi import join_compact
+
+    risingwave_amd.load_library()
+    glib = ffi.Lib(risingwave_amd.lib_path())
+    j = ffi.HashJoin(glib, JOIN_INNER, [T_I64, T_I64], [T_I64, T_I64],
+                     key_l=[0], key_r=[0], pk_l=[1], pk_r=[1])
+    j.push(SIDE_LEFT, mk_chunk([T_I64, T_I64], [0], [[1], [2]]))
+    j.poll_all()
+    with pytest.raises(RuntimeError, match="undrained|pending"):
+        join_compact(glib, j.h, SIDE_LEFT)
+    join_checkpoint_drain(glib, j.h, SIDE_LEFT)
+    join_compact(glib, j.h, SIDE_LEFT)  # drained: succeeds
+    j.close()
+
+
 def test_join_clean_restore_oracle():
     _join_clean_restore_flow(oracle(), JOIN_INNER)
 
