@@ -305,6 +305,11 @@ int rw_join_checkpoint_drain(void* h, int side, uint8_t** buf, uint64_t* len);
  * loudly on pending deltas or undrained rows). Logical state, drains and
  * restore are unaffected; `reclaimed` (optional) reports freed bytes. */
 int rw_join_compact(void* h, int side, uint64_t* reclaimed);
+/* Same reclamation for the agg's materialized-input row store and the
+ * GroupTopN record store (retractions retire rows in place). Same
+ * contract: call after the drains; fails loudly on pending deltas. */
+int rw_agg_minput_compact(void* h, uint64_t* reclaimed);
+int rw_topn_compact(void* h, uint64_t* reclaimed);
 int rw_topn_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len);
 /* Replay concatenated rw_topn_checkpoint_drain streams into a freshly
  * created GroupTopN executor (PUT/DELETE frames net host-side; the

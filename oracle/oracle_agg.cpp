@@ -1107,6 +1107,14 @@ int rw_agg_dedup_restore(void* h, int di, const uint8_t* buf, uint64_t len) {
     return ((HashAggOracle*)h)->dedup_restore(di, buf, len);
 }
 
+int rw_agg_minput_compact(void* h, uint64_t* reclaimed) {
+    // the oracle's ordered maps hold no dead rows — no-op (see
+    // rw_join_compact)
+    if (!h) return RW_E_INVAL;
+    if (reclaimed) *reclaimed = 0;
+    return RW_OK;
+}
+
 void rw_spill_free(uint8_t* buf) { free(buf); }
 RwChunk* rw_hash_agg_poll(void* h) { return ((HashAggOracle*)h)->poll(); }
 void rw_hash_agg_destroy(void* h) { delete (HashAggOracle*)h; }

@@ -365,6 +365,12 @@ int rw_topn_restore(void* h, const uint8_t* buf, uint64_t len) {
 int rw_group_top_n_watermark(void* h, uint32_t col_idx, int64_t val) {
     return ((GroupTopNOracle*)h)->watermark(col_idx, val);
 }
+int rw_topn_compact(void* h, uint64_t* reclaimed) {
+    // no-op: the oracle's maps hold no dead rows (see rw_join_compact)
+    if (!h) return RW_E_INVAL;
+    if (reclaimed) *reclaimed = 0;
+    return RW_OK;
+}
 int rw_topn_checkpoint_drain(void* h, uint8_t** buf, uint64_t* len) {
     std::vector<uint8_t> sp;
     ((GroupTopNOracle*)h)->checkpoint_drain(sp);

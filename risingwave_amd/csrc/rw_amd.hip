@@ -3146,6 +3146,111 @@ struct HashAgg {
         return RW_OK;
     }
 
+    // §8f-4 memory reclamation for the materialized-input row store:
+    // retractions retire rows in place (malive=0) and the store would
+    // otherwise grow without bound. Host-marshalled pack (maintenance
+    // cadence, same copy costs as a drain): alive rows compact down,
+    // chains rebuild host-side, heads re-upload. Requires drained minput
+    // tables (flush marks at the cursor, kill list consumed).
+    int minput_compact(uint64_t* reclaimed) {
+        if (!n_minput) {
+            if (reclaimed) *reclaimed = 0;
+            return RW_OK;
+        }
+        HIP_TRY(hipStreamSynchronize(stream));
+        uint32_t cur = 0, kcur = 0;
+        HIP_TRY(hipMemcpy(&cur, t.mcursor, 4, hipMemcpyDeviceToHost));
+        HIP_TRY(hipMemcpy(&kcur, t.mkilled_cursor, 4,
+                          hipMemcpyDeviceToHost));
+        if ((int)mflush_mark.size() < n_minput) {
+            mflush_mark.assign(n_minput, 0);
+            mkill_mark.assign(n_minput, 0);
+        }
+        for (int mi = 0; mi < n_minput; mi++)
+            if (mflush_mark[mi] != cur || mkill_mark[mi] != kcur)
+                FAIL(RW_E_INVAL, "minput compact requires drained tables "
+                                 "(table %d has undrained deltas)", mi);
+        if (cur > t.mrow_cap) cur = t.mrow_cap;
+        size_t cap = (size_t)t.cap_mask + 1;
+        std::vector<long long> mval(cur), sk((size_t)t.n_sk * cur);
+        std::vector<uint8_t> mnull(cur), skn((size_t)t.n_sk * cur),
+            mc(cur);
+        std::vector<uint32_t> mslot_h(cur), alive(cur);
+        if (cur) {
+            HIP_TRY(hipMemcpy(mval.data(), t.mval, (size_t)cur * 8,
+                              hipMemcpyDeviceToHost));
+            HIP_TRY(hipMemcpy(mnull.data(), t.mval_null, cur,
+                              hipMemcpyDeviceToHost));
+            HIP_TRY(hipMemcpy(mc.data(), t.mcall, cur,
+                              hipMemcpyDeviceToHost));
+            HIP_TRY(hipMemcpy(mslot_h.data(), t.mslot, (size_t)cur * 4,
+                              hipMemcpyDeviceToHost));
+            HIP_TRY(hipMemcpy(alive.data(), t.malive, (size_t)cur * 4,
+                              hipMemcpyDeviceToHost));
+            for (int k = 0; k < t.n_sk; k++) {
+                HIP_TRY(hipMemcpy(sk.data() + (size_t)k * cur,
+                                  t.msk + (size_t)k * t.mrow_cap,
+                                  (size_t)cur * 8, hipMemcpyDeviceToHost));
+                HIP_TRY(hipMemcpy(skn.data() + (size_t)k * cur,
+                                  t.msk_null + (size_t)k * t.mrow_cap,
+                                  (size_t)cur, hipMemcpyDeviceToHost));
+            }
+        }
+        uint32_t n_alive = 0;
+        std::vector<uint32_t> mnext_h;
+        std::vector<uint32_t> heads((size_t)n_minput * cap, UINT32_MAX);
+        for (uint32_t r = 0; r < cur; r++) {
+            if (!alive[r]) continue;
+            uint32_t nr = n_alive++;
+            mval[nr] = mval[r];
+            mnull[nr] = mnull[r];
+            mc[nr] = mc[r];
+            mslot_h[nr] = mslot_h[r];
+            for (int k = 0; k < t.n_sk; k++) {
+                sk[(size_t)k * cur + nr] = sk[(size_t)k * cur + r];
+                skn[(size_t)k * cur + nr] = skn[(size_t)k * cur + r];
+            }
+            size_t hi = (size_t)mc[nr] * cap + mslot_h[nr];
+            mnext_h.push_back(heads[hi]);
+            heads[hi] = nr;
+        }
+        if (n_alive) {
+            HIP_TRY(hipMemcpy(t.mval, mval.data(), (size_t)n_alive * 8,
+                              hipMemcpyHostToDevice));
+            HIP_TRY(hipMemcpy(t.mval_null, mnull.data(), n_alive,
+                              hipMemcpyHostToDevice));
+            HIP_TRY(hipMemcpy(t.mcall, mc.data(), n_alive,
+                              hipMemcpyHostToDevice));
+            HIP_TRY(hipMemcpy(t.mslot, mslot_h.data(), (size_t)n_alive * 4,
+                              hipMemcpyHostToDevice));
+            HIP_TRY(hipMemcpy(t.mnext, mnext_h.data(), (size_t)n_alive * 4,
+                              hipMemcpyHostToDevice));
+            std::vector<uint32_t> ones(n_alive, 1);
+            HIP_TRY(hipMemcpy(t.malive, ones.data(), (size_t)n_alive * 4,
+                              hipMemcpyHostToDevice));
+            for (int k = 0; k < t.n_sk; k++) {
+                HIP_TRY(hipMemcpy(t.msk + (size_t)k * t.mrow_cap,
+                                  sk.data() + (size_t)k * cur,
+                                  (size_t)n_alive * 8,
+                                  hipMemcpyHostToDevice));
+                HIP_TRY(hipMemcpy(t.msk_null + (size_t)k * t.mrow_cap,
+                                  skn.data() + (size_t)k * cur,
+                                  (size_t)n_alive, hipMemcpyHostToDevice));
+            }
+        }
+        HIP_TRY(hipMemcpy(t.mheads, heads.data(), heads.size() * 4,
+                          hipMemcpyHostToDevice));
+        HIP_TRY(hipMemcpy(t.mcursor, &n_alive, 4, hipMemcpyHostToDevice));
+        HIP_TRY(hipMemset(t.mkilled_cursor, 0, 4));
+        for (int mi = 0; mi < n_minput; mi++) {
+            mflush_mark[mi] = n_alive;
+            mkill_mark[mi] = 0;
+        }
+        if (reclaimed)
+            *reclaimed = (uint64_t)(cur - n_alive) * (8 + 1 + 9 * t.n_sk);
+        return RW_OK;
+    }
+
     int minput_restore(int mi, const uint8_t* buf, uint64_t len) {
         if (mi < 0 || mi >= n_minput) FAIL(RW_E_INVAL, "minput table %d", mi);
         int ci = -1, o = -1;
@@ -3866,6 +3971,10 @@ int rw_agg_minput_drain(void* h, int mi, uint8_t** buf, uint64_t* len) {
 }
 int rw_agg_minput_restore(void* h, int mi, const uint8_t* buf, uint64_t len) {
     return ((HashAgg*)h)->minput_restore(mi, buf, len);
+}
+
+int rw_agg_minput_compact(void* h, uint64_t* reclaimed) {
+    return ((HashAgg*)h)->minput_compact(reclaimed);
 }
 
 int rw_agg_n_dedup_tables(void* h) {
@@ -7965,6 +8074,43 @@ __global__ void topn_clean_kernel(JoinSideDev sd, int kpos, long long wm) {
     }
 }
 
+// §8f-4 memory reclamation (TopN): relink packed records into freshly
+// initialized group slots (jcompact_copy_kernel packs; this rebuilds the
+// chains — netted keys need not be unique here, concurrent pushes on one
+// group resolve by CAS).
+__global__ void topn_relink_kernel(JoinSideDev sd, TopMeta m, uint32_t n,
+                                   uint32_t* err) {
+    uint32_t stride = gridDim.x * blockDim.x;
+    for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < n;
+         r += stride) {
+        JoinRowHdr* h = jrow(sd, r);
+        uint32_t vb = h->validbits;
+        int64_t kw[MAX_KW];
+        uint32_t nm = 0;
+        for (int i = 0; i < m.KW; i++) {
+            uint8_t col = m.gk_cols[i];
+            bool valid = (vb >> col) & 1;
+            kw[i] = valid ? jvals(h)[col] : 0;
+            nm |= (uint32_t)(!valid) << i;
+        }
+        uint32_t slot =
+            jslot_find_or_insert(sd.slots, sd.cap_mask, kw, nm, m.KW);
+        if (slot == UINT32_MAX) {
+            atomicExch(err, 2u);
+            continue;
+        }
+        uint32_t* headp = &sd.slots[slot].head;
+        uint32_t old_head = ld_u32(headp);
+        for (;;) {
+            st_u32(&h->next, old_head);
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); // R1 drain
+            uint32_t prev = atomicCAS(headp, old_head, r);
+            if (prev == old_head) break;
+            old_head = prev;
+        }
+    }
+}
+
 // §8f-5 recovery: rebuild the TopN state table from a netted spill replay.
 // Each restored row is appended to the record store and chained onto its
 // group's slot (cache keys in the netted map are unique, so no same-key
@@ -8137,6 +8283,7 @@ struct GroupTopN {
         }
         if (touched_list) hipFree(touched_list);
         if (old_win) hipFree(old_win);
+        if (compact_scratch) hipFree(compact_scratch);
         free_stage();
         if (stream) hipStreamDestroy(stream);
         for (auto* c : outq) {
@@ -8329,6 +8476,62 @@ struct GroupTopN {
         return RW_OK;
     }
 
+    // §8f-4 memory reclamation: pack alive records, reinit the slot
+    // table, relink chains. Requires a drained table (kill list consumed,
+    // flush mark at the cursor); logical state is unchanged.
+    uint8_t* compact_scratch = nullptr;
+    int compact(uint64_t* reclaimed) {
+        HIP_TRY(hipStreamSynchronize(stream));
+        uint32_t cur = 0, kcur = 0;
+        HIP_TRY(hipMemcpy(&cur, sd.row_cursor, 4, hipMemcpyDeviceToHost));
+        HIP_TRY(hipMemcpy(&kcur, sd.killed_cursor, 4,
+                          hipMemcpyDeviceToHost));
+        if (kcur)
+            FAIL(RW_E_INVAL,
+                 "compact requires a drained kill list (%u pending)", kcur);
+        if (flush_mark != cur)
+            FAIL(RW_E_INVAL, "compact requires a drained state table "
+                             "(%u undrained fresh rows)", cur - flush_mark);
+        if (cur > sd.row_cap) cur = sd.row_cap;
+        if (!compact_scratch)
+            HIP_TRY(hipMalloc(&compact_scratch,
+                              (size_t)sd.row_cap * sd.row_stride));
+        uint32_t* ncur = nullptr;
+        HIP_TRY(hipMalloc(&ncur, 4));
+        HIP_TRY(hipMemsetAsync(ncur, 0, 4, stream));
+        jcompact_copy_kernel<<<4096, 256, 0, stream>>>(sd, compact_scratch,
+                                                       ncur, cur);
+        jslot_init_kernel<<<2048, 256, 0, stream>>>(
+            sd.slots, (size_t)sd.cap_mask + 1);
+        int rcs = hipStreamSynchronize(stream) == hipSuccess
+                      ? RW_OK : RW_E_INTERNAL;
+        uint32_t n_alive = 0;
+        if (rcs == RW_OK)
+            hipMemcpy(&n_alive, ncur, 4, hipMemcpyDeviceToHost);
+        hipFree(ncur);
+        if (rcs != RW_OK) FAIL(RW_E_INTERNAL, "topn compact copy failed");
+        uint8_t* old = sd.rows;
+        sd.rows = compact_scratch;
+        compact_scratch = old; // ping-pong
+        HIP_TRY(hipMemcpy(sd.row_cursor, &n_alive, 4,
+                          hipMemcpyHostToDevice));
+        HIP_TRY(hipMemsetAsync(tcounters, 0, 8, stream));
+        uint32_t blocks = (n_alive + 255) / 256;
+        if (blocks > 2048) blocks = 2048;
+        if (!blocks) blocks = 1;
+        topn_relink_kernel<<<blocks, 256, 0, stream>>>(sd, m, n_alive,
+                                                       tcounters + 1);
+        if (hipStreamSynchronize(stream) != hipSuccess)
+            FAIL(RW_E_INTERNAL, "topn compact relink failed");
+        uint32_t tc[2];
+        HIP_TRY(hipMemcpy(tc, tcounters, 8, hipMemcpyDeviceToHost));
+        if (tc[1]) FAIL(RW_E_INTERNAL, "topn compact overflow (%u)", tc[1]);
+        flush_mark = n_alive;
+        if (reclaimed)
+            *reclaimed = (uint64_t)(cur - n_alive) * sd.row_stride;
+        return RW_OK;
+    }
+
     // §8f-2 checkpoint spill: the TopN state table's per-epoch KV deltas —
     // key = memcomparable storage key (group cols ASC, then the cache-key
     // cols with their declared orders), value = value-encoded full row;
@@ -8503,6 +8706,10 @@ void* rw_group_top_n_create(const RwGroupTopNDesc* d) {
         return nullptr;
     }
     return t;
+}
+
+int rw_topn_compact(void* h, uint64_t* reclaimed) {
+    return ((GroupTopN*)h)->compact(reclaimed);
 }
 
 // handle_watermark (group_top_n.rs:266-273): a watermark on the FIRST

@@ -669,6 +669,28 @@ def join_compact(lib, h, side):
     return freed.value
 
 
+def _compact1(lib, h, sym):
+    L = lib.lib
+    fn = getattr(L, sym)
+    fn.restype = C.c_int
+    fn.argtypes = [C.c_void_p, C.POINTER(C.c_uint64)]
+    freed = C.c_uint64()
+    rc = fn(h, C.byref(freed))
+    if rc != 0:
+        raise RuntimeError(f"{sym} failed {rc}: {lib.last_error()}")
+    return freed.value
+
+
+def agg_minput_compact(lib, h):
+    """Reclaim the agg's dead materialized-input rows; returns bytes."""
+    return _compact1(lib, h, "rw_agg_minput_compact")
+
+
+def topn_compact(lib, h):
+    """Reclaim the GroupTopN store's dead records; returns bytes."""
+    return _compact1(lib, h, "rw_topn_compact")
+
+
 def topn_restore(lib, h, buf):
     """Rebuild GroupTopN state from concatenated drain bytes (rw_stream.h)."""
     L = lib.lib

@@ -587,6 +587,89 @@ def _join_compact_flow(lib, join_type):
     return freed_total
 
 
+def _minput_topn_compact_flow(lib):
+    """rw_agg_minput_compact / rw_topn_compact: reclamation with no
+    observable change (emissions + drains match a non-compacting twin)."""
+    from rwtest.ffi import (AGG_MIN, agg_minput_compact,
+                            agg_minput_drain_bytes, topn_checkpoint_drain,
+                            topn_compact)
+
+    calls = [(AGG_MIN, 1, T_I64), (AGG_COUNT_STAR, -1, T_I64)]
+    mka = lambda: ffi.HashAgg(lib, [T_I64, T_I64, T_I64], [0], calls, 1,
+                              stream_key=(2,))
+    mkt = lambda: ffi.GroupTopN(lib, [T_I64, T_I64, T_I64], [0],
+                                [(1, False)], [(2, False)], limit=2)
+    a1, a2 = mka(), mka()
+    t1, t2 = mkt(), mkt()
+    rng = np.random.default_rng(888)
+    live_a, live_t, rid = [], [], [0]
+    freed = 0
+    for e in range(4):
+        n = 200
+        g = rng.integers(0, 12, n)
+        v = rng.integers(-50, 50, n)
+        r2 = np.arange(rid[0], rid[0] + n)
+        rid[0] += n
+        ops = np.zeros(n, np.uint8)
+        for r in range(n):
+            if live_a and rng.random() < 0.4:
+                j = int(rng.integers(0, len(live_a)))
+                g[r], v[r], r2[r] = live_a.pop(j)
+                ops[r] = ffi.OP_DELETE
+            else:
+                live_a.append((int(g[r]), int(v[r]), int(r2[r])))
+        c = mk_chunk([T_I64] * 3, ops, [g, v, r2])
+        oa = _drive_agg(a1, [c], e + 1)
+        ob = _drive_agg(a2, [c], e + 1)
+        assert oa == ob, f"epoch {e}: compacted minput agg diverged"
+        for x in (a1, a2):
+            agg_checkpoint_drain_bytes(lib, x.h)
+        d1 = agg_minput_drain_bytes(lib, a1.h, 0)
+        d2 = agg_minput_drain_bytes(lib, a2.h, 0)
+        assert d1 == d2, f"epoch {e}: minput drain diverged"
+        freed += agg_minput_compact(lib, a1.h)
+        # TopN twin flow with deletes
+        gt = rng.integers(0, 10, n)
+        vt = rng.integers(0, 60, n)
+        rt = np.arange(rid[0], rid[0] + n)
+        rid[0] += n
+        opst = np.zeros(n, np.uint8)
+        for r in range(n):
+            if live_t and rng.random() < 0.4:
+                j = int(rng.integers(0, len(live_t)))
+                gt[r], vt[r], rt[r] = live_t.pop(j)
+                opst[r] = ffi.OP_DELETE
+            else:
+                live_t.append((int(gt[r]), int(vt[r]), int(rt[r])))
+        ct = mk_chunk([T_I64] * 3, opst, [gt, vt, rt])
+        t1.push(ct)
+        t2.push(ct)
+        m1 = rows_multiset(t1.poll_all())
+        m2 = rows_multiset(t2.poll_all())
+        assert m1 == m2, f"epoch {e}: compacted topn diverged"
+        s1 = topn_checkpoint_drain(lib, t1.h)
+        s2 = topn_checkpoint_drain(lib, t2.h)
+        assert s1 == s2, f"epoch {e}: topn drain diverged"
+        freed += topn_compact(lib, t1.h)
+    for x in (a1, a2, t1, t2):
+        x.close()
+    return freed
+
+
+def test_minput_topn_compact_oracle_noop():
+    assert _minput_topn_compact_flow(oracle()) == 0
+
+
+@pytest.mark.gpu
+def test_minput_topn_compact_gpu():
+    import risingwave_amd
+
+    risingwave_amd.load_library()
+    glib = ffi.Lib(risingwave_amd.lib_path())
+    freed = _minput_topn_compact_flow(glib)
+    assert freed > 0, "compaction reclaimed nothing despite retractions"
+
+
 def test_join_compact_oracle_noop():
     assert _join_compact_flow(oracle(), JOIN_INNER) == 0
 
