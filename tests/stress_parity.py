@@ -282,6 +282,9 @@ def stress_clean_restore(seed, n):
             state[side][1] += so
             degs[side][0] += dg
             degs[side][1] += do
+            if rng.random() < 0.5:  # randomized maintenance compaction
+                ffi.join_compact(GPU, g.h, side)
+                ffi.join_compact(oracle(), o.h, side)
         if i == 3:
             # crash: swap in executors restored from the drain streams
             g.close()
