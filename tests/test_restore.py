@@ -563,6 +563,9 @@ def _join_compact_flow(lib, join_type):
     # a compacts every epoch, c never does; both track the oracle flow
     a, c = mk(), mk()
     freed_total = 0
+    state = {s: b"" for s in (SIDE_LEFT, SIDE_RIGHT)}
+    degs = {s: b"" for s in (SIDE_LEFT, SIDE_RIGHT)}
+    wm = 0
     for e in range(5):
         for side, ch in epoch_pushes(e):
             for x in (a, c):
@@ -581,8 +584,28 @@ def _join_compact_flow(lib, join_type):
             ga = join_degree_drain(lib, a.h, s)
             gc = join_degree_drain(lib, c.h, s)
             assert ga == gc, f"epoch {e} side {s}: degree drain diverged"
+            state[s] += da
+            degs[s] += ga
             freed_total += join_compact(lib, a.h, s)
+    # a restore from streams spanning the compactions must continue in
+    # lockstep with the compacted executor
+    b = mk()
+    for s in (SIDE_LEFT, SIDE_RIGHT):
+        join_restore(lib, b.h, s, state[s], degs[s])
+    b.watermark(SIDE_LEFT, 0, wm)
+    b.watermark(SIDE_RIGHT, 0, wm)
+    for side, ch in epoch_pushes(5):
+        a.push(side, ch)
+        b.push(side, ch)
+        ma = rows_multiset(a.poll_all())
+        mb = rows_multiset(b.poll_all())
+        assert ma == mb, "restore across compactions diverged"
+    for s in (SIDE_LEFT, SIDE_RIGHT):
+        da = join_checkpoint_drain(lib, a.h, s)
+        db = join_checkpoint_drain(lib, b.h, s)
+        assert da == db, f"side {s}: post-compaction restore drain diverged"
     a.close()
+    b.close()
     c.close()
     return freed_total
 
