@@ -175,6 +175,18 @@ typedef struct RwHashJoinDesc {
     uint32_t n_wm_jk;
     const uint32_t* wm_jk_pos;
     const uint8_t* wm_jk_clean;
+    /* inequality pairs (hash_join.rs InequalityPairInfo / the non-equi
+     * condition's watermark derivation): per pair the left/right input
+     * column, whether the LEFT side is the larger one (left >= right),
+     * and whether the larger side's state is cleaned below the selected
+     * watermark. Watermarks on these columns buffer per side; when the
+     * min across sides advances it is emitted for the LARGER side's
+     * output columns and (if clean) sweeps that side's rows. */
+    uint32_t n_ineq;
+    const uint32_t* ineq_left_col;
+    const uint32_t* ineq_right_col;
+    const uint8_t* ineq_left_larger;
+    const uint8_t* ineq_clean;
 } RwHashJoinDesc;
 
 enum RwJoinSide { RW_SIDE_LEFT = 0, RW_SIDE_RIGHT = 1 };

@@ -142,6 +142,10 @@ struct HashJoinOracle {
     std::vector<uint8_t> wm_clean;
     std::vector<WmEntry> wm_side[2];
     std::vector<WmEntry> wm_out;
+    // inequality-pair watermarks (hash_join.rs:869-914)
+    std::vector<uint32_t> ineq_col[2];
+    std::vector<uint8_t> ineq_larger, ineq_do_clean;
+    std::vector<WmEntry> ineq_wm[2], ineq_out;
     uint8_t T;
     bool append_only;
     std::vector<uint8_t> null_safe;
@@ -169,6 +173,18 @@ struct HashJoinOracle {
         wm_side[0].resize(d->n_key);
         wm_side[1].resize(d->n_key);
         wm_out.resize(d->n_key);
+        if (d->n_ineq) {
+            ineq_col[0].assign(d->ineq_left_col,
+                               d->ineq_left_col + d->n_ineq);
+            ineq_col[1].assign(d->ineq_right_col,
+                               d->ineq_right_col + d->n_ineq);
+            ineq_larger.assign(d->ineq_left_larger,
+                               d->ineq_left_larger + d->n_ineq);
+            ineq_do_clean.assign(d->ineq_clean, d->ineq_clean + d->n_ineq);
+            ineq_wm[0].resize(d->n_ineq);
+            ineq_wm[1].resize(d->n_ineq);
+            ineq_out.resize(d->n_ineq);
+        }
 
         side[0].key_idx.assign(d->key_l, d->key_l + d->n_key);
         side[1].key_idx.assign(d->key_r, d->key_r + d->n_key);
@@ -616,6 +632,26 @@ struct HashJoinOracle {
         return RW_OK;
     }
 
+    // inequality-pair state cleaning: retire ROWS of side `s2` whose
+    // value column `col` sorts below the selected watermark (NULLs
+    // largest, kept); per-row spill DELETEs as clean_below
+    void clean_rows_below(int s2, uint32_t col, int64_t sel) {
+        auto& tab = side[s2].table;
+        for (auto it = tab.begin(); it != tab.end();) {
+            for (auto pit = it->second.begin(); pit != it->second.end();) {
+                const Datum& d = pit->second.row[col];
+                if (!d.null && d.i < sel) {
+                    delta_delete(s2, it->first, pit->first);
+                    pit = it->second.erase(pit);
+                } else {
+                    ++pit;
+                }
+            }
+            if (it->second.empty()) it = tab.erase(it);
+            else ++it;
+        }
+    }
+
     int watermark(int s, uint32_t col_idx, int64_t val, uint32_t* out_cols,
                   int64_t* out_vals, int max_out) {
         int n_out = 0;
@@ -639,6 +675,29 @@ struct HashJoinOracle {
                         out_vals[n_out] = sel;
                         n_out++;
                     }
+                }
+            }
+        }
+        // inequality-pair watermarks (hash_join.rs:869-914): min across
+        // sides; emit for the LARGER side's output columns; clean that
+        // side's rows below the selection when flagged
+        for (size_t p = 0; p < ineq_col[0].size(); p++) {
+            if (ineq_col[s][p] != col_idx) continue;
+            ineq_wm[s][p] = {true, val};
+            if (!ineq_wm[0][p].has || !ineq_wm[1][p].has) continue;
+            int64_t sel = std::min(ineq_wm[0][p].val, ineq_wm[1][p].val);
+            if (ineq_out[p].has && sel <= ineq_out[p].val) continue;
+            ineq_out[p] = {true, sel};
+            int larger = ineq_larger[p] ? 0 : 1;
+            if (ineq_do_clean[p])
+                clean_rows_below(larger, ineq_col[larger][p], sel);
+            uint32_t src = ineq_col[larger][p] +
+                           (larger == 1 ? (uint32_t)types_l.size() : 0);
+            for (uint32_t oi = 0; oi < output_indices.size(); oi++) {
+                if (output_indices[oi] == src && n_out < max_out) {
+                    out_cols[n_out] = oi;
+                    out_vals[n_out] = sel;
+                    n_out++;
                 }
             }
         }

@@ -6172,6 +6172,13 @@ struct HashJoin {
     std::vector<uint32_t> wm_pos;
     std::vector<uint8_t> wm_clean;
     std::vector<Wm> wm_side0, wm_side1, wm_out;
+    // inequality-pair watermarks (hash_join.rs:869-914): buffered per
+    // pair; the min across sides emits for the LARGER side's output
+    // columns and (clean flag) sweeps that side's rows by the pair's
+    // value column
+    std::vector<uint32_t> ineq_col[2]; // [pair] per side
+    std::vector<uint8_t> ineq_larger, ineq_do_clean;
+    std::vector<Wm> ineq_wm[2], ineq_out;
     // lazy event ring (same rationale as HashAgg: no per-step host sync)
     static constexpr int EV_RING = 256;
     hipEvent_t ev0[EV_RING] = {}, ev1[EV_RING] = {};
@@ -6273,6 +6280,18 @@ struct HashJoin {
         wm_side0.resize(d->n_key);
         wm_side1.resize(d->n_key);
         wm_out.resize(d->n_key);
+        if (d->n_ineq) {
+            ineq_col[0].assign(d->ineq_left_col,
+                               d->ineq_left_col + d->n_ineq);
+            ineq_col[1].assign(d->ineq_right_col,
+                               d->ineq_right_col + d->n_ineq);
+            ineq_larger.assign(d->ineq_left_larger,
+                               d->ineq_left_larger + d->n_ineq);
+            ineq_do_clean.assign(d->ineq_clean, d->ineq_clean + d->n_ineq);
+            ineq_wm[0].resize(d->n_ineq);
+            ineq_wm[1].resize(d->n_ineq);
+            ineq_out.resize(d->n_ineq);
+        }
 
         HIP_TRY(hipStreamCreate(&stream));
         uint64_t key_cap = d->state_capacity_hint ? d->state_capacity_hint : (1u << 20);
@@ -6967,6 +6986,36 @@ struct HashJoin {
                         out_vals[n_out] = sel;
                         n_out++;
                     }
+                }
+            }
+        }
+        // inequality-pair watermarks (hash_join.rs:869-914): buffer per
+        // side; the min across sides emits for the LARGER side's output
+        // columns and (clean flag) sweeps that side's rows whose pair
+        // column sorts below it — through the kill list, so the next
+        // drain nets the cleaned rows to DELETE frames
+        for (size_t p = 0; p < ineq_col[0].size(); p++) {
+            if (ineq_col[s][p] != col_idx) continue;
+            ineq_wm[s][p] = {true, val};
+            if (!ineq_wm[0][p].has || !ineq_wm[1][p].has) continue;
+            int64_t sel = std::min(ineq_wm[0][p].val, ineq_wm[1][p].val);
+            if (ineq_out[p].has && sel <= ineq_out[p].val) continue;
+            ineq_out[p] = {true, sel};
+            int larger = ineq_larger[p] ? 0 : 1;
+            if (ineq_do_clean[p]) {
+                join_clean_kernel<<<2048, 256, 0, stream>>>(
+                    side[larger], (int)ineq_col[larger][p], sel, m.KW);
+                HIP_TRY(hipStreamSynchronize(stream));
+            }
+            uint32_t src = ineq_col[larger][p] +
+                           (larger == 1 ? (uint32_t)m.n_cols[0] : 0);
+            for (int oi = 0; oi < m.n_out; oi++) {
+                uint32_t osrc = (uint32_t)m.out_col[oi] +
+                                (m.out_src[oi] ? (uint32_t)m.n_cols[0] : 0);
+                if (osrc == src && n_out < max_out) {
+                    out_cols[n_out] = (uint32_t)oi;
+                    out_vals[n_out] = sel;
+                    n_out++;
                 }
             }
         }

@@ -107,6 +107,11 @@ class RwHashJoinDesc(C.Structure):
         ("n_wm_jk", C.c_uint32),
         ("wm_jk_pos", C.POINTER(C.c_uint32)),
         ("wm_jk_clean", C.POINTER(C.c_uint8)),
+        ("n_ineq", C.c_uint32),
+        ("ineq_left_col", C.POINTER(C.c_uint32)),
+        ("ineq_right_col", C.POINTER(C.c_uint32)),
+        ("ineq_left_larger", C.POINTER(C.c_uint8)),
+        ("ineq_clean", C.POINTER(C.c_uint8)),
     ]
 
 
@@ -353,8 +358,10 @@ class HashJoin:
                  pk_l, pk_r, output_indices=None, null_safe=None, cond=None,
                  cond2=None,
                  chunk_size=1024, append_only=False, state_capacity_hint=0,
-                 row_capacity_hint=0, wm_jk=()):
-        """cond: (op, cond_l, cond_r) into the concatenated row, or None."""
+                 row_capacity_hint=0, wm_jk=(), wm_ineq=()):
+        """cond: (op, cond_l, cond_r) into the concatenated row, or None.
+        wm_ineq: (left_col, right_col, left_larger, clean) tuples (the
+        reference's InequalityPairInfo)."""
         self.lib = lib
         d = RwHashJoinDesc()
         d.join_type = join_type
@@ -411,6 +418,15 @@ class HashJoin:
         self._wc = _u8arr([1 if c else 0 for _, c in wm_jk])
         d.wm_jk_pos = self._wp
         d.wm_jk_clean = self._wc
+        d.n_ineq = len(wm_ineq)
+        self._il = _u32arr([l for l, _, _, _ in wm_ineq])
+        self._ir = _u32arr([r for _, r, _, _ in wm_ineq])
+        self._ig = _u8arr([1 if g else 0 for _, _, g, _ in wm_ineq])
+        self._ic = _u8arr([1 if c2 else 0 for _, _, _, c2 in wm_ineq])
+        d.ineq_left_col = self._il
+        d.ineq_right_col = self._ir
+        d.ineq_left_larger = self._ig
+        d.ineq_clean = self._ic
         self.h = lib.lib.rw_hash_join_create(C.byref(d))
         if not self.h:
             raise RuntimeError(f"rw_hash_join_create failed: {lib.last_error()}")

@@ -1386,3 +1386,34 @@ def test_join_epoch_ingest_mode_parity():
         assert d0 == d1 == d2, f"epoch {epoch}: left drains diverged"
     for x in (g_epoch, g_plain, o):
         x.close()
+
+
+@pytest.mark.gpu
+def test_inequality_join_watermark_gpu():
+    # hash_join.rs:1740-1829 on the GPU path, plus drain byte-parity with
+    # the oracle after the inequality state-cleaning sweep
+    import risingwave_amd
+    from rwtest.ffi import (CMP_GE, JOIN_INNER, SIDE_LEFT, SIDE_RIGHT,
+                            T_I64, from_pretty, join_checkpoint_drain,
+                            oracle, rows_ordered)
+
+    risingwave_amd.load_library()
+    glib = ffi.Lib(risingwave_amd.lib_path())
+    I2 = [T_I64, T_I64]
+    drains = {}
+    for name, lib in (("gpu", glib), ("orc", oracle())):
+        j = ffi.HashJoin(lib, JOIN_INNER, I2, I2, key_l=[0], key_r=[0],
+                         pk_l=[1], pk_r=[1], cond=(CMP_GE, 1, 3),
+                         wm_ineq=((1, 1, True, True),))
+        j.push(SIDE_LEFT, from_pretty(" I I\n + 2 4\n + 2 7\n + 3 8"))
+        assert rows_ordered(j.poll_all()) == []
+        assert j.watermark(SIDE_LEFT, 1, 10) == []
+        assert j.watermark(SIDE_RIGHT, 1, 6) == [(1, 6)]
+        j.push(SIDE_RIGHT, from_pretty(" I I\n + 2 6"))
+        assert rows_ordered(j.poll_all()) == [("+", (2, 7, 2, 6))]
+        j.push(SIDE_RIGHT, from_pretty(" I I\n + 2 3"))
+        assert rows_ordered(j.poll_all()) == [("+", (2, 7, 2, 3))]
+        drains[name] = tuple(join_checkpoint_drain(lib, j.h, s)
+                             for s in (SIDE_LEFT, SIDE_RIGHT))
+        j.close()
+    assert drains["gpu"] == drains["orc"]

@@ -446,3 +446,24 @@ def test_watermark_reference_fixture():
     assert o.watermark(SIDE_RIGHT, 0, 50) == [(2, 50), (0, 50)]
     assert o.watermark(SIDE_RIGHT, 0, 100) == [(2, 100), (0, 100)]
     o.close()
+
+
+def test_inequality_join_watermark():
+    # hash_join.rs:1740-1829 test_inequality_join_watermark: condition
+    # left.col1 >= right.col1 with InequalityPairInfo{left_idx:1,
+    # right_idx:1, clean_left_state} — value-column watermarks buffer per
+    # side, the min emits for the LARGER (left) side's output column, and
+    # the left rows below it are state-cleaned
+    from rwtest.ffi import CMP_GE
+
+    j = ffi.HashJoin(oracle(), JOIN_INNER, I2, I2, key_l=[0], key_r=[0],
+                     pk_l=[1], pk_r=[1], cond=(CMP_GE, 1, 3),
+                     wm_ineq=((1, 1, True, True),))
+    assert push(j, SIDE_LEFT, " I I\n + 2 4\n + 2 7\n + 3 8") == []
+    assert j.watermark(SIDE_LEFT, 1, 10) == []
+    # min(10, 6) = 6 emitted on the left col1 output position
+    assert j.watermark(SIDE_RIGHT, 1, 6) == [(1, 6)]
+    # (2,4) cleaned (4 < 6); (2,7) and (3,8) remain
+    assert push(j, SIDE_RIGHT, " I I\n + 2 6") == rows([("+", 2, 7, 2, 6)])
+    assert push(j, SIDE_RIGHT, " I I\n + 2 3") == rows([("+", 2, 7, 2, 3)])
+    j.close()
