@@ -1417,3 +1417,54 @@ def test_inequality_join_watermark_gpu():
                              for s in (SIDE_LEFT, SIDE_RIGHT))
         j.close()
     assert drains["gpu"] == drains["orc"]
+
+
+@pytest.mark.gpu
+def test_topn_recovery_fixtures_gpu():
+    # the reference's own recovery golden vectors
+    # (top_n_plain.rs:811-906 and :1113-1211) replayed on the GPU build
+    import risingwave_amd
+    from test_oracle_topn import (NEW_CHUNKS, NEW_EXPECT, TIES_CHUNKS,
+                                  TIES_EXPECT)
+    from rwtest.ffi import (T_I64, from_pretty, rows_multiset,
+                            topn_checkpoint_drain, topn_restore)
+
+    risingwave_amd.load_library()
+    glib = ffi.Lib(risingwave_amd.lib_path())
+
+    def expect(t, pretty):
+        got = rows_multiset(t.poll_all())
+        want = rows_multiset([from_pretty(pretty)]) if pretty else []
+        assert got == want, f"got {got}\nwant {want}"
+
+    I4 = [T_I64] * 4
+    mk4 = lambda: ffi.GroupTopN(glib, I4, [], [(0, False), (3, False)], [],
+                                offset=1, limit=3)
+    a = mk4()
+    for c, e in zip(NEW_CHUNKS[:2], NEW_EXPECT[:2]):
+        a.push(from_pretty(c))
+        expect(a, e)
+    sp = topn_checkpoint_drain(glib, a.h)
+    a.close()
+    b = mk4()
+    topn_restore(glib, b.h, sp)
+    for c, e in zip(NEW_CHUNKS[2:], NEW_EXPECT[2:]):
+        b.push(from_pretty(c))
+        expect(b, e)
+    b.close()
+
+    mkt = lambda: ffi.GroupTopN(glib, [T_I64, T_I64], [], [(0, False)],
+                                [(1, False)], offset=0, limit=3,
+                                with_ties=True)
+    a = mkt()
+    for c, e in zip(TIES_CHUNKS[:2], TIES_EXPECT[:2]):
+        a.push(from_pretty(c))
+        expect(a, e)
+    sp = topn_checkpoint_drain(glib, a.h)
+    a.close()
+    b = mkt()
+    topn_restore(glib, b.h, sp)
+    for c, e in zip(TIES_CHUNKS[2:], TIES_EXPECT[2:]):
+        b.push(from_pretty(c))
+        expect(b, e)
+    b.close()

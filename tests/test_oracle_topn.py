@@ -353,3 +353,94 @@ def test_plain_topn_no_group():
         " I I\n + 7 6\n - 3 2\n - 1 0\n + 5 7\n - 2 1\n + 9 4")])
     assert got == want, got
     t.close()
+
+
+I4 = [T_I64, T_I64, T_I64, T_I64]
+
+
+def make_plain4(offset, limit):
+    # plain TopN (no group), order_by (c0 asc, c3 asc) == storage key
+    # (top_n_plain.rs:732-741 "new" family)
+    return ffi.GroupTopN(oracle(), I4, [], [(0, False), (3, False)], [],
+                         offset=offset, limit=limit)
+
+
+NEW_CHUNKS = [
+    " I I I I\n + 1 1 4 1001",
+    " I I I I\n + 5 1 4 1002",
+    " I I I I\n + 1 9 1 1003\n + 9 8 1 1004\n + 0 2 3 1005",
+    " I I I I\n + 1 0 2 1006",
+]
+NEW_EXPECT = [
+    "",
+    " I I I I\n + 5 1 4 1002",
+    " I I I I\n + 1 9 1 1003\n + 1 1 4 1001",
+    " I I I I\n - 5 1 4 1002\n + 1 0 2 1006",
+]
+
+
+def test_plain_offset_and_limit_new():
+    # top_n_plain.rs:748-807 test_top_n_executor_with_offset_and_limit_new
+    # (offset 1, limit 3)
+    t = make_plain4(1, 3)
+    for c, e in zip(NEW_CHUNKS, NEW_EXPECT):
+        t.push(from_pretty(c))
+        expect(t, e)
+    t.close()
+
+
+def test_plain_offset_and_limit_new_after_recovery():
+    # top_n_plain.rs:811-906: chunks 0-1 before the crash, state drained,
+    # a fresh executor hydrated, chunks 2-3 after — same outputs as the
+    # uninterrupted fixture
+    from rwtest.ffi import topn_checkpoint_drain, topn_restore
+
+    a = make_plain4(1, 3)
+    for c, e in zip(NEW_CHUNKS[:2], NEW_EXPECT[:2]):
+        a.push(from_pretty(c))
+        expect(a, e)
+    sp = topn_checkpoint_drain(oracle(), a.h)
+    a.close()
+    b = make_plain4(1, 3)
+    topn_restore(oracle(), b.h, sp)
+    for c, e in zip(NEW_CHUNKS[2:], NEW_EXPECT[2:]):
+        b.push(from_pretty(c))
+        expect(b, e)
+    b.close()
+
+
+TIES_CHUNKS = [
+    "  I I\n + 1 0\n + 2 1\n + 3 2\n + 10 3\n + 9 4\n + 8 5",
+    "  I I\n + 3 6\n + 3 7\n + 1 8\n + 2 9\n + 10 10",
+    " I I\n - 1 0",
+    " I I\n - 1 8",
+]
+TIES_EXPECT = [
+    " I I\n + 1 0\n + 2 1\n + 3 2",
+    " I I\n - 3 2\n + 1 8\n + 2 9",
+    " I I\n - 1 0",
+    " I I\n - 1 8\n + 3 2\n + 3 6\n + 3 7",
+]
+
+
+def test_with_ties_recovery():
+    # top_n_plain.rs:1113-1211 test_with_ties_recovery: the ties fixture
+    # split at the barrier; a fresh executor hydrates from the drained
+    # state table and continues identically
+    from rwtest.ffi import topn_checkpoint_drain, topn_restore
+
+    mk = lambda: ffi.GroupTopN(oracle(), [T_I64, T_I64], [], [(0, False)],
+                               [(1, False)], offset=0, limit=3,
+                               with_ties=True)
+    a = mk()
+    for c, e in zip(TIES_CHUNKS[:2], TIES_EXPECT[:2]):
+        a.push(from_pretty(c))
+        expect(a, e)
+    sp = topn_checkpoint_drain(oracle(), a.h)
+    a.close()
+    b = mk()
+    topn_restore(oracle(), b.h, sp)
+    for c, e in zip(TIES_CHUNKS[2:], TIES_EXPECT[2:]):
+        b.push(from_pretty(c))
+        expect(b, e)
+    b.close()
