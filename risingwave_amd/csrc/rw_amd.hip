@@ -1028,7 +1028,11 @@ __global__ void agg_apply_kernel(AggBatch b, AggTableDev t, AggCallDev c0,
             long long ov[4];
             uint8_t ohas;
             for (int ci = 0; ci < n_calls; ci++) ov[ci] = __shfl_up(v[ci], d);
-            ohas = DENSE ? hasmask : (uint8_t)__shfl_up((int)hasmask, d);
+            // hasmask must ALWAYS be shuffled: even in a DENSE batch the
+            // per-lane masks differ when DISTINCT dedup hides rows (a
+            // dense own-mask shortcut dropped the run's sum/min/max here
+            // whenever the tail lane's own row was a hidden duplicate)
+            ohas = (uint8_t)__shfl_up((int)hasmask, d);
             if (run_pos >= d) {
                 for (int ci = 0; ci < n_calls; ci++) {
                     switch (calls[ci].kind) {

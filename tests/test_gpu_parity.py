@@ -1468,3 +1468,60 @@ def test_topn_recovery_fixtures_gpu():
         b.push(from_pretty(c))
         expect(b, e)
     b.close()
+
+
+@pytest.mark.gpu
+def test_distinct_dedup_invisible_rows_gpu():
+    # the distinct deduplicater fixture's chunks (distinct.rs:373-471)
+    # replayed at the executor level: the D (invisible) row must not
+    # touch the dedup tables; emissions and every drain byte-compare
+    # across builds, including a recovery replay between the chunks
+    import risingwave_amd
+    from rwtest.ffi import (AGG_COUNT, AGG_COUNT_STAR, AGG_SUM, T_I64,
+                            agg_checkpoint_drain_bytes, agg_dedup_drain_bytes,
+                            agg_dedup_restore, agg_restore, from_pretty,
+                            oracle, rows_multiset)
+
+    risingwave_amd.load_library()
+    glib = ffi.Lib(risingwave_amd.lib_path())
+    # count(distinct a), sum(distinct a), count(distinct b), count(*)
+    # (the fixture's plain count(a) dropped: 4-call kernel limit);
+    # group by c (single-group constant)
+    calls = [(AGG_COUNT, 0, T_I64, 1), (AGG_SUM, 0, T_I64, 1),
+             (AGG_COUNT, 1, T_I64, 1), (AGG_COUNT_STAR, -1, T_I64)]
+    chunks = [
+        " I  I  I\n + 1 10 0\n + 1 11 0",
+        " I  I  I\n + 1 11 0\n + 2 12 0 D\n + 2 12 0",
+    ]
+    results = {}
+    for name, lib in (("gpu", glib), ("orc", oracle())):
+        a = ffi.HashAgg(lib, [T_I64, T_I64, T_I64], [2], calls, 3)
+        outs, drains = [], []
+        for e, c in enumerate(chunks):
+            a.push(from_pretty(c))
+            a.flush(e + 1)
+            outs.append(rows_multiset(a.poll_all()))
+            inter = agg_checkpoint_drain_bytes(lib, a.h)
+            d0 = agg_dedup_drain_bytes(lib, a.h, 0)
+            d1 = agg_dedup_drain_bytes(lib, a.h, 1)
+            drains.append((inter, d0, d1))
+        a.close()
+        # recovery: replay all drains into a fresh executor, push one
+        # more chunk exercising both dedup tables
+        b = ffi.HashAgg(lib, [T_I64, T_I64, T_I64], [2], calls, 3)
+        agg_dedup_restore(lib, b.h, 0, b"".join(d[1] for d in drains))
+        agg_dedup_restore(lib, b.h, 1, b"".join(d[2] for d in drains))
+        agg_restore(lib, b.h, b"".join(d[0] for d in drains))
+        b.push(from_pretty(" I  I  I\n + 2 13 0\n + 3 10 0"))
+        b.flush(9)
+        outs.append(rows_multiset(b.poll_all()))
+        drains.append((agg_checkpoint_drain_bytes(lib, b.h),
+                       agg_dedup_drain_bytes(lib, b.h, 0),
+                       agg_dedup_drain_bytes(lib, b.h, 1)))
+        b.close()
+        results[name] = (outs, drains)
+    assert results["gpu"] == results["orc"]
+    # count(distinct a)=1, sum(distinct a)=1, count(distinct b)=2, n=2
+    first = results["gpu"][0][0]
+    assert first == rows_multiset([from_pretty(
+        " I I I I I\n + 0 1 1 2 2")])
