@@ -184,13 +184,18 @@ def stress_distinct(seed, n):
         keys = rng.integers(0, 40, n)
         vals = rng.integers(0, 6, n)
         ops = np.zeros(n, np.uint8)
-        for r in range(n):
+        # epoch 0 stays all-insert: DENSE batches take the template
+        # specialization whose has-mask handling a hidden duplicate once
+        # broke (the distinct.rs fixture bug) — keep it exercised
+        for r in range(n if ep else 0):
             if live and rng.random() < 0.4:
                 jx = int(rng.integers(0, len(live)))
                 keys[r], vals[r] = live.pop(jx)
                 ops[r] = ffi.OP_DELETE
             else:
                 live.append((int(keys[r]), int(vals[r])))
+        if ep == 0:
+            live += [(int(k), int(v)) for k, v in zip(keys, vals)]
         c = mk([T_I64, T_I64], ops, [keys, vals])
         outs = []
         for a in (g, o):
