@@ -776,22 +776,26 @@ def bench_q8(args, ffi, gpu_lib, rng, rank, world, dist):
 
     BUILD_KEYS = 10_000_000
     # PMC-measured HBM traffic of the probe kernel at this config
-    # (profiles/r02_q8_pmc_{fetch,write}_regwalk.txt: per-1M-row probe
+    # (profiles/r02_q8_pmc_{fetch,write}_hint25.txt: per-1M-row probe
     # launch averages with the 10 build launches subtracted out; raw
     # FETCH+WRITE, random narrow reads are uncalibrated on gfx950 so no
     # x2 correction is applied — MI355X_MICROARCH.md §HBM): the gap to
     # the 128-B/row algorithmic model is 128-B line granularity on the
     # random touches (match record + own insert RFO) plus the
     # pre-assigned layout's sparse-region writes. The register-resident
-    # walk removed the former eviction refetches (was 622 B/row fetched).
-    Q8_TRAFFIC_B_PER_ROW = (311_831 + 304_805) * 1024 / 1_048_576
+    # walk removed the former eviction refetches (was 622 B/row fetched;
+    # 602 at the LIC-resident hint-23 tables, 551 at the default 2^26
+    # low-load tables whose shorter chains re-walk less).
+    Q8_TRAFFIC_B_PER_ROW = (276_081 + 288_165) * 1024 / 1_048_576
     batch_rows = CHUNK_ROWS * CHUNKS_PER_BATCH
     t4 = [T_I64, T_TS, T_TS, T_I64]
     # hint 2^23 -> cap 2^24 slots x 8 B = 134 MB per side: BOTH slot tables
     # sit in the 256 MB Infinity Cache (load factor 0.6 at 10M keys)
-    # RW_Q8_HINT_LOG2 A/B hook: 23 (default) -> cap 2^24 slots/side; 22
-    # halves both tables so slots + records contend less for the LIC
-    hint_log2 = int(os.environ.get("RW_Q8_HINT_LOG2", "23"))
+    # RW_Q8_HINT_LOG2 A/B hook. Default 25 -> cap 2^26 slots/side
+    # (512 MB, load factor 0.15): shorter chains beat LIC residency —
+    # measured 0.238 vs 0.245 ms/step against hint 23's LIC-resident
+    # 2^24 tables (22/23/24/25 -> 0.262/0.245/0.242/0.238).
+    hint_log2 = int(os.environ.get("RW_Q8_HINT_LOG2", "25"))
     j = ffi.HashJoin(gpu_lib, JOIN_INNER, t4, t4, key_l=[0, 1, 2],
                      key_r=[0, 1, 2], pk_l=[3], pk_r=[3],
                      state_capacity_hint=1 << hint_log2,
