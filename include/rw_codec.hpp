@@ -63,8 +63,18 @@ inline void memcmp_encode_datum(std::vector<uint8_t>& buf, uint8_t type,
                 buf.push_back((uint8_t)dat.i);
                 break;
             case RW_T_F64: {
+                // ordered-float canonicalization before the sign-flip
+                // transform (memcmp_encoding.rs:543-590 via OrderedFloat):
+                // -0.0 encodes as +0.0 and every NaN (incl. -NaN) as one
+                // canonical largest pattern — they compare equal, so their
+                // storage keys must byte-compare equal too
+                double dv = dat.d;
+                if (dv == 0.0) dv = 0.0; // -0.0 -> +0.0
                 uint64_t bits;
-                std::memcpy(&bits, &dat.d, 8);
+                if (dv != dv)
+                    bits = 0x7ff8000000000000ULL; // canonical quiet NaN
+                else
+                    std::memcpy(&bits, &dv, 8);
                 bits = (bits & 0x8000000000000000ULL)
                            ? ~bits
                            : bits | 0x8000000000000000ULL;
@@ -73,8 +83,12 @@ inline void memcmp_encode_datum(std::vector<uint8_t>& buf, uint8_t type,
             }
             case RW_T_F32: {
                 float f = (float)dat.d;
+                if (f == 0.0f) f = 0.0f; // -0.0 -> +0.0 (see F64)
                 uint32_t bits;
-                std::memcpy(&bits, &f, 4);
+                if (f != f)
+                    bits = 0x7fc00000u; // canonical quiet NaN
+                else
+                    std::memcpy(&bits, &f, 4);
                 bits = (bits & 0x80000000u) ? ~bits : bits | 0x80000000u;
                 put_be(buf, bits, 4);
                 break;

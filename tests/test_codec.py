@@ -336,3 +336,57 @@ def test_join_degree_spill_bytes():
     ffi.join_checkpoint_drain(oracle(), o.h, SIDE_LEFT)
     assert ffi.join_degree_drain(oracle(), o.h, SIDE_LEFT) == b""
     o.close()
+
+
+def test_memcmp_f64_ordering_and_zero_identity():
+    # util/memcmp_encoding.rs:543-590 (the legacy-2057 ordered-float
+    # fixture, f64 projection): encoded keys sort -inf < -1 < 0 < 1 <
+    # inf < NaN (NaN largest), and -0.0 encodes identically to +0.0 —
+    # exercised through the GroupTopN storage key on an F64 order column
+    import numpy as np
+
+    from rwtest.ffi import T_F64, T_I64, topn_checkpoint_drain
+
+    t = ffi.GroupTopN(oracle(), [T_F64, T_I64], [], [(0, False)],
+                      [(1, False)], limit=10)
+    vals = [float("-inf"), -1.0, 0.0, 1.0, float("inf"), float("nan")]
+    n = len(vals)
+    t.push(ffi.Chunk([T_F64, T_I64], np.zeros(n, np.uint8),
+                     [np.array(vals, np.float64), np.arange(n)],
+                     [np.ones(n, np.uint8)] * 2))
+    t.poll_all()
+    sp = topn_checkpoint_drain(oracle(), t.h)
+    keys = []
+    off = 0
+    while off < len(sp):
+        klen = int.from_bytes(sp[off + 1:off + 5], "little")
+        keys.append(sp[off + 5:off + 5 + klen])
+        off += 5 + klen
+        vlen = int.from_bytes(sp[off:off + 4], "little")
+        off += 4 + vlen
+    # drains are memcmp-key sorted; recover the pk (second col) per key
+    # and check it matches the value order of `vals`
+    assert keys == sorted(keys)
+    order = [int.from_bytes(k[-8:], "big") ^ (1 << 63) for k in keys]
+    assert order == [0, 1, 2, 3, 4, 5], order  # -inf..nan ascending
+    t.close()
+
+    # -0.0 == +0.0: same storage key, the second insert upserts
+    t2 = ffi.GroupTopN(oracle(), [T_F64, T_I64], [], [(0, False)],
+                      [(1, False)], limit=10)
+    t2.push(ffi.Chunk([T_F64, T_I64], np.zeros(2, np.uint8),
+                      [np.array([0.0, -0.0], np.float64),
+                       np.array([7, 7], np.int64)],
+                      [np.ones(2, np.uint8)] * 2))
+    t2.poll_all()
+    sp2 = topn_checkpoint_drain(oracle(), t2.h)
+    n_frames = 0
+    off = 0
+    while off < len(sp2):
+        klen = int.from_bytes(sp2[off + 1:off + 5], "little")
+        off += 5 + klen
+        vlen = int.from_bytes(sp2[off:off + 4], "little")
+        off += 4 + vlen
+        n_frames += 1
+    assert n_frames == 1, "-0.0 and +0.0 must share one storage key"
+    t2.close()
