@@ -38,6 +38,28 @@ struct DatumC {
     int64_t i2 = 0; // decimal datums: high half of the 16-byte image
 };
 
+// Build a DatumC from a RAW 8-byte device word: float types reinterpret
+// the bits into .d (a plain {null, word, 0} initializer leaves .d = 0.0
+// and silently encodes every float as zero).
+inline DatumC datum_of_word(uint8_t type, bool null, int64_t w,
+                            int64_t w2 = 0) {
+    DatumC d{null, w, 0, w2};
+    if (!null && (type == RW_T_F64 || type == RW_T_F32))
+        std::memcpy(&d.d, &w, 8);
+    return d;
+}
+
+// Inverse: the raw 8-byte device word for a decoded datum (float types
+// take .d's bits; .i is 0 for them after value_decode_datum).
+inline int64_t word_of_datum(uint8_t type, const DatumC& d) {
+    if (type == RW_T_F64 || type == RW_T_F32) {
+        int64_t w;
+        std::memcpy(&w, &d.d, 8);
+        return w;
+    }
+    return d.i;
+}
+
 inline void put_be(std::vector<uint8_t>& buf, uint64_t v, int n) {
     for (int k = n - 1; k >= 0; k--) buf.push_back((uint8_t)(v >> (8 * k)));
 }

@@ -2818,8 +2818,9 @@ struct HashAgg {
                                                     {true, 0, 0});
                     } else {
                         size_t ix = (size_t)r * out_width + KW + ci;
-                        rwcodec::DatumC d{nulls[ix] != 0, vals[ix], 0,
-                                          vals2.empty() ? 0 : vals2[ix]};
+                        rwcodec::DatumC d = rwcodec::datum_of_word(
+                            calls[ci].ret_type, nulls[ix] != 0, vals[ix],
+                            vals2.empty() ? 0 : vals2[ix]);
                         rwcodec::value_encode_datum(v, calls[ci].ret_type, d);
                     }
                 }
@@ -2886,12 +2887,14 @@ struct HashAgg {
                 persisted[slot] = 0;
             }
             for (int c = 0; c <= KW; c++) {
-                rwcodec::DatumC d{((r.nulls >> c) & 1) != 0, r.key[c], 0};
+                rwcodec::DatumC d = rwcodec::datum_of_word(
+                    key_types[c], ((r.nulls >> c) & 1) != 0, r.key[c]);
                 rwcodec::memcmp_encode_datum(e.k, key_types[c], d, {});
             }
             if (e.put) {
                 for (int c = 0; c <= KW; c++) {
-                    rwcodec::DatumC d{((r.nulls >> c) & 1) != 0, r.key[c], 0};
+                    rwcodec::DatumC d = rwcodec::datum_of_word(
+                        key_types[c], ((r.nulls >> c) & 1) != 0, r.key[c]);
                     rwcodec::value_encode_datum(e.v, key_types[c], d);
                 }
                 for (int ci = 0; ci < n_calls; ci++)
@@ -2957,7 +2960,8 @@ struct HashAgg {
                     val.data() + off, val.size() - off, key_types[c], &d);
                 if (!got) FAIL(RW_E_INVAL, "dedup restore: bad key datum");
                 off += got;
-                keys[(size_t)i * KW1 + c] = d.null ? 0 : d.i;
+                keys[(size_t)i * KW1 + c] =
+                    d.null ? 0 : rwcodec::word_of_datum(key_types[c], d);
                 nulls[i] |= (uint32_t)(d.null != 0) << c;
             }
             rwcodec::DatumC d;
@@ -3083,7 +3087,8 @@ struct HashAgg {
                                   gkeys[(size_t)slot * KW + i2], 0};
                 rwcodec::memcmp_encode_datum(kb, out_types[i2], d, {});
             }
-            rwcodec::DatumC dv{mnull[row] != 0, mval[row], 0};
+            rwcodec::DatumC dv = rwcodec::datum_of_word(
+                input_types[calls[ci].arg], mnull[row] != 0, mval[row]);
             rwcodec::memcmp_encode_datum(
                 kb, input_types[calls[ci].arg], dv,
                 {calls[ci].kind == RW_AGG_MAX, true});
@@ -3100,7 +3105,8 @@ struct HashAgg {
                                       gkeys[(size_t)slot * KW + i2], 0};
                     rwcodec::value_encode_datum(*v, out_types[i2], d);
                 }
-                rwcodec::DatumC dv2{mnull[row] != 0, mval[row], 0};
+                rwcodec::DatumC dv2 = rwcodec::datum_of_word(
+                    input_types[calls[ci].arg], mnull[row] != 0, mval[row]);
                 rwcodec::value_encode_datum(*v, input_types[calls[ci].arg],
                                             dv2);
                 for (int k2 = 0; k2 < t.n_sk; k2++) {
@@ -3295,7 +3301,7 @@ struct HashAgg {
                     val.data() + off, val.size() - off, ty, &d);
                 if (!got) return false;
                 off += got;
-                *vv = d.null ? 0 : d.i;
+                *vv = d.null ? 0 : rwcodec::word_of_datum(ty, d);
                 *nn = d.null;
                 return true;
             };
@@ -3922,7 +3928,8 @@ int rw_hash_agg_restore(void* h, const uint8_t* buf, uint64_t len) {
                 keys[(size_t)i * KW + c] = d.null ? 0 : d.i;
                 knulls[i] |= (uint32_t)(d.null != 0) << c;
             } else {
-                vals[(size_t)(c - KW) * n + i] = d.i;
+                vals[(size_t)(c - KW) * n + i] =
+                    rwcodec::word_of_datum(ty, d);
                 vnulls[(size_t)(c - KW) * n + i] = d.null;
                 if (agg->n_dec) vals2[(size_t)(c - KW) * n + i] = d.i2;
             }
@@ -7786,6 +7793,11 @@ struct TopMeta {
     int KW;     // group key width
     int n_cols; // input schema width
     uint8_t gk_cols[MAX_KW];
+    uint8_t gk_float[MAX_KW]; // float group cols: canonicalize key words
+                              // (-0.0 -> +0.0, NaN -> one pattern) so the
+                              // raw-bit slot compare matches SQL equality
+                              // (the reference's HashKey normalizes,
+                              // common/src/hash/key.rs:517)
     int n_ck; // cache key = order_by then rest storage-key cols
     uint8_t ck_cols[TOPN_MAX_CK];
     uint8_t ck_desc[TOPN_MAX_CK];
@@ -7836,6 +7848,20 @@ __device__ __forceinline__ bool topn_ck_eq_batch(const TopMeta& m,
     return true;
 }
 
+// canonicalize a float group-key word: -0.0 -> +0.0, every NaN -> the
+// canonical quiet pattern — SQL equality for the raw-bit slot compare
+__device__ __forceinline__ int64_t fkey_canon(int64_t w) {
+    double d = __longlong_as_double((long long)w);
+    if (d == 0.0) return 0;
+    if (d != d) return 0x7ff8000000000000LL;
+    return w;
+}
+
+__device__ __forceinline__ int64_t topn_gkw(const TopMeta& m, int i,
+                                            int64_t v) {
+    return m.gk_float[i] ? fkey_canon(v) : v;
+}
+
 __global__ void topn_touch_kernel(JoinBatchDev b, JoinSideDev sd, TopMeta m,
                                   uint32_t* touched, uint32_t* touched_list,
                                   uint32_t* counters /*0=tcur 1=err*/) {
@@ -7848,7 +7874,7 @@ __global__ void topn_touch_kernel(JoinBatchDev b, JoinSideDev sd, TopMeta m,
         for (int i = 0; i < m.KW; i++) {
             uint8_t col = m.gk_cols[i];
             bool valid = b.col_valid[col][r];
-            kw[i] = valid ? b.col_vals[col][r] : 0;
+            kw[i] = valid ? topn_gkw(m, i, b.col_vals[col][r]) : 0;
             nullmask |= (!valid) << i;
         }
         uint32_t slot =
@@ -7951,7 +7977,7 @@ __global__ void topn_apply_kernel(JoinBatchDev b, JoinSideDev sd, TopMeta m,
         for (int i = 0; i < m.KW; i++) {
             uint8_t col = m.gk_cols[i];
             bool valid = b.col_valid[col][r];
-            kw[i] = valid ? b.col_vals[col][r] : 0;
+            kw[i] = valid ? topn_gkw(m, i, b.col_vals[col][r]) : 0;
             nullmask |= (!valid) << i;
         }
         uint32_t slot =
@@ -8143,7 +8169,7 @@ __global__ void topn_relink_kernel(JoinSideDev sd, TopMeta m, uint32_t n,
         for (int i = 0; i < m.KW; i++) {
             uint8_t col = m.gk_cols[i];
             bool valid = (vb >> col) & 1;
-            kw[i] = valid ? jvals(h)[col] : 0;
+            kw[i] = valid ? topn_gkw(m, i, jvals(h)[col]) : 0;
             nm |= (uint32_t)(!valid) << i;
         }
         uint32_t slot =
@@ -8204,7 +8230,7 @@ __global__ void topn_restore_kernel(JoinBatchDev b, JoinSideDev sd, TopMeta m,
         for (int i = 0; i < m.KW; i++) {
             uint8_t col = m.gk_cols[i];
             bool valid = (vb >> col) & 1;
-            kw[i] = valid ? hv[col] : 0;
+            kw[i] = valid ? topn_gkw(m, i, hv[col]) : 0;
             nm |= (uint32_t)(!valid) << i;
         }
         uint32_t slot =
@@ -8263,8 +8289,10 @@ struct GroupTopN {
                 FAIL(RW_E_INVAL, "GroupTopN supports 8-byte datum columns");
         m.KW = (int)d->n_group_by;
         m.n_cols = (int)d->n_cols;
-        for (uint32_t i = 0; i < d->n_group_by; i++)
+        for (uint32_t i = 0; i < d->n_group_by; i++) {
             m.gk_cols[i] = (uint8_t)d->group_by[i];
+            m.gk_float[i] = types[d->group_by[i]] == RW_T_F64;
+        }
         m.n_ck = (int)(d->n_order_by + d->n_rest);
         int k = 0;
         for (uint32_t i = 0; i < d->n_order_by; i++, k++) {
@@ -8615,12 +8643,14 @@ struct GroupTopN {
             std::vector<uint8_t> kb;
             for (int i = 0; i < m.KW; i++) {
                 uint8_t col = m.gk_cols[i];
-                rwcodec::DatumC d{!((vb >> col) & 1), vals[col], 0};
+                rwcodec::DatumC d = rwcodec::datum_of_word(
+                    types[col], !((vb >> col) & 1), vals[col]);
                 rwcodec::memcmp_encode_datum(kb, types[col], d, {});
             }
             for (int i = 0; i < m.n_ck; i++) {
                 uint8_t col = m.ck_cols[i];
-                rwcodec::DatumC d{!((vb >> col) & 1), vals[col], 0};
+                rwcodec::DatumC d = rwcodec::datum_of_word(
+                    types[col], !((vb >> col) & 1), vals[col]);
                 rwcodec::OrderType ot;
                 ot.desc = m.ck_desc[i] != 0;
                 rwcodec::memcmp_encode_datum(kb, types[col], d, ot);
@@ -8631,7 +8661,8 @@ struct GroupTopN {
             const uint32_t vb = ((const uint32_t*)rec)[2];
             const int64_t* vals = (const int64_t*)(rec + 16);
             for (int c = 0; c < m.n_cols; c++) {
-                rwcodec::DatumC d{!((vb >> c) & 1), vals[c], 0};
+                rwcodec::DatumC d = rwcodec::datum_of_word(
+                    types[c], !((vb >> c) & 1), vals[c]);
                 rwcodec::value_encode_datum(v, types[c], d);
             }
         };
@@ -8720,7 +8751,7 @@ int rw_topn_restore(void* h, const uint8_t* buf, uint64_t len) {
                 val.data() + off, val.size() - off, t->types[c], &d);
             if (!got) FAIL(RW_E_INVAL, "restore: bad row datum");
             off += got;
-            cols[c][i] = d.null ? 0 : d.i;
+            cols[c][i] = d.null ? 0 : rwcodec::word_of_datum(t->types[c], d);
             valid[c][i] = !d.null;
         }
         i++;

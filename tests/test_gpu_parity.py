@@ -1525,3 +1525,55 @@ def test_distinct_dedup_invisible_rows_gpu():
     first = results["gpu"][0][0]
     assert first == rows_multiset([from_pretty(
         " I I I I I\n + 0 1 1 2 2")])
+
+
+@pytest.mark.gpu
+def test_topn_float_group_key_zero_identity_gpu():
+    # SQL equality for float group keys: -0.0, +0.0 (and every NaN) are
+    # ONE group (the reference's HashKey normalizes floats,
+    # common/src/hash/key.rs:517); the GPU's raw-bit slot compare
+    # canonicalizes the key words. GPU vs oracle on emissions + drains +
+    # a restore replay.
+    import numpy as np
+
+    import risingwave_amd
+    from rwtest.ffi import (T_F64, T_I64, oracle, rows_multiset,
+                            topn_checkpoint_drain, topn_restore)
+
+    risingwave_amd.load_library()
+    glib = ffi.Lib(risingwave_amd.lib_path())
+    nan = float("nan")
+    g = [0.0, -0.0, nan, -nan, 1.5, 0.0]
+    v = [1.0, 2.0, 3.0, 4.0, 5.0, 6.0]
+    pk = np.arange(6)
+    results = {}
+    for name, lib in (("gpu", glib), ("orc", oracle())):
+        t = ffi.GroupTopN(lib, [T_F64, T_F64, T_I64], [0], [(1, False)],
+                          [(2, False)], limit=2)
+        t.push(ffi.Chunk([T_F64, T_F64, T_I64], np.zeros(6, np.uint8),
+                         [np.array(g, np.float64), np.array(v, np.float64),
+                          pk], [np.ones(6, np.uint8)] * 3))
+        out = rows_multiset(t.poll_all())
+        sp = topn_checkpoint_drain(lib, t.h)
+        t.close()
+        b = ffi.GroupTopN(lib, [T_F64, T_F64, T_I64], [0], [(1, False)],
+                          [(2, False)], limit=2)
+        topn_restore(lib, b.h, sp)
+        # a late row joins the zero group after restore
+        b.push(ffi.Chunk([T_F64, T_F64, T_I64], np.zeros(1, np.uint8),
+                         [np.array([-0.0], np.float64),
+                          np.array([0.5], np.float64),
+                          np.array([9], np.int64)],
+                         [np.ones(1, np.uint8)] * 3))
+        out2 = rows_multiset(b.poll_all())
+        sp2 = topn_checkpoint_drain(lib, b.h)
+        b.close()
+        def norm(ms):
+            return sorted((op, tuple(repr(c) for c in row))
+                          for op, row in ms)
+
+        results[name] = (norm(out), sp, norm(out2), sp2)
+    assert results["gpu"] == results["orc"]
+    # zero group limit 2: of v=1,2,6 only the two smallest stay
+    ops = [r[0] for r in results["gpu"][0]]
+    assert ops.count("+") >= 4  # zero group 2 + nan group 2 + 1.5 group 1
