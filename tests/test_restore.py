@@ -972,3 +972,79 @@ def test_agg_minput_drain_parity_cpu_noop():
     assert a.lib.lib.rw_agg_n_minput_tables(a.h) == 1
     assert agg_minput_drain_bytes(oracle(), a.h, 0) == b""
     a.close()
+
+
+def test_agg_float_state_restore_oracle():
+    # float SUM/MIN/MAX states round-trip the drain encoding (the raw-word
+    # datum bugs collapsed floats to 0.0 in drains and restores)
+    from rwtest.ffi import AGG_MAX, AGG_MIN, T_F64
+
+    calls = [(AGG_SUM, 1, T_F64), (AGG_COUNT_STAR, -1, T_I64)]
+    lib = oracle()
+    mk = lambda: ffi.HashAgg(lib, [T_I64, T_F64], [0], calls, 1)
+    a = mk()
+    rng = np.random.default_rng(11)
+    d = b""
+    for e in range(2):
+        n = 100
+        g = rng.integers(0, 8, n)
+        v = rng.uniform(-5, 5, n)
+        a.push(ffi.Chunk([T_I64, T_F64], np.zeros(n, np.uint8),
+                         [g, np.array(v, np.float64)],
+                         [np.ones(n, np.uint8)] * 2))
+        a.flush(e + 1)
+        a.poll_all()
+        d += agg_checkpoint_drain_bytes(lib, a.h)
+    b = mk()
+    agg_restore(lib, b.h, d)
+    n = 100
+    g = rng.integers(0, 8, n)
+    v = rng.uniform(-5, 5, n)
+    c = ffi.Chunk([T_I64, T_F64], np.zeros(n, np.uint8),
+                  [g, np.array(v, np.float64)], [np.ones(n, np.uint8)] * 2)
+    oa = _drive_agg(a, [c], 5)
+    ob = _drive_agg(b, [c], 5)
+    assert oa == ob, "restored float agg diverged"
+    assert agg_checkpoint_drain_bytes(lib, a.h) == \
+        agg_checkpoint_drain_bytes(lib, b.h)
+    a.close()
+    b.close()
+
+
+def test_topn_float_restore_oracle():
+    # float order/rest columns round-trip the TopN drain + restore
+    from rwtest.ffi import T_F64, topn_checkpoint_drain, topn_restore
+
+    lib = oracle()
+    mk = lambda: ffi.GroupTopN(lib, [T_I64, T_F64, T_I64], [0],
+                               [(1, True)], [(2, False)], limit=2)
+    a = mk()
+    rng = np.random.default_rng(13)
+    rid = [0]
+    d = b""
+    for e in range(2):
+        n = 80
+        g = rng.integers(0, 6, n)
+        v = rng.uniform(-3, 3, n)
+        r2 = np.arange(rid[0], rid[0] + n)
+        rid[0] += n
+        a.push(ffi.Chunk([T_I64, T_F64, T_I64], np.zeros(n, np.uint8),
+                         [g, np.array(v, np.float64), r2],
+                         [np.ones(n, np.uint8)] * 3))
+        a.poll_all()
+        d += topn_checkpoint_drain(lib, a.h)
+    b = mk()
+    topn_restore(lib, b.h, d)
+    n = 80
+    g = rng.integers(0, 6, n)
+    v = rng.uniform(-3, 3, n)
+    r2 = np.arange(rid[0], rid[0] + n)
+    c = ffi.Chunk([T_I64, T_F64, T_I64], np.zeros(n, np.uint8),
+                  [g, np.array(v, np.float64), r2],
+                  [np.ones(n, np.uint8)] * 3)
+    a.push(c)
+    b.push(c)
+    assert rows_multiset(a.poll_all()) == rows_multiset(b.poll_all())
+    assert topn_checkpoint_drain(lib, a.h) == topn_checkpoint_drain(lib, b.h)
+    a.close()
+    b.close()
