@@ -444,3 +444,59 @@ def test_with_ties_recovery():
         b.push(from_pretty(c))
         expect(b, e)
     b.close()
+
+
+OLD_CHUNKS = [
+    "  I I\n + 1 0\n + 2 1\n + 3 2\n + 10 3\n + 9 4\n + 8 5",
+    "  I I\n + 7 6\n - 3 2\n - 1 0\n + 5 7\n - 2 1\n + 11 8",
+    "  I I\n + 6 9\n + 12 10\n + 13 11\n + 14 12",
+    "  I I\n - 5 7\n - 6 9\n - 11 8",
+]
+
+
+def make_plain2(offset, limit):
+    # plain TopN, order_by c0 asc, storage key = order_by ++ [c1 asc]
+    # (top_n_plain.rs:266-278 old family)
+    return ffi.GroupTopN(oracle(), [T_I64, T_I64], [], [(0, False)],
+                         [(1, False)], offset=offset, limit=limit)
+
+
+def _run_old(offset, limit, expects):
+    t = make_plain2(offset, limit)
+    for c, e in zip(OLD_CHUNKS, expects):
+        t.push(from_pretty(c))
+        expect(t, e)
+    t.close()
+
+
+def test_plain_with_offset():
+    # top_n_plain.rs:298-376 test_top_n_executor_with_offset (3, 1000);
+    # limit 125 stands in for the reference's 1000 (TOPN_MAX_WIN is 128;
+    # at most 12 rows are ever alive, so the windows are identical)
+    _run_old(3, 125, [
+        "  I I\n + 10 3\n + 9 4\n + 8 5",
+        "  I I\n - 8 5\n + 11 8",
+        "  I I\n + 8 5\n + 12 10\n + 13 11\n + 14 12",
+        "  I I\n - 8 5\n - 9 4\n - 11 8",
+    ])
+
+
+def test_plain_with_limit():
+    # top_n_plain.rs:378-463 test_top_n_executor_with_limit (0, 4)
+    _run_old(0, 4, [
+        "  I I\n + 1 0\n + 2 1\n + 3 2\n + 8 5",
+        "  I I\n + 7 6\n - 3 2\n - 1 0\n + 5 7\n - 2 1\n + 9 4",
+        "  I I\n - 9 4\n + 6 9",
+        "  I I\n - 5 7\n + 9 4\n - 6 9\n + 10 3",
+    ])
+
+
+def test_plain_with_offset_and_limit():
+    # top_n_plain.rs:551-630 test_top_n_executor_with_offset_and_limit
+    # (3, 4)
+    _run_old(3, 4, [
+        "  I I\n + 10 3\n + 9 4\n + 8 5",
+        "  I I\n - 8 5\n + 11 8",
+        "  I I\n + 8 5",
+        "  I I\n - 8 5\n + 12 10\n - 9 4\n + 13 11\n - 11 8\n + 14 12",
+    ])
