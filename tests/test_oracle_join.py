@@ -467,3 +467,24 @@ def test_inequality_join_watermark():
     assert push(j, SIDE_RIGHT, " I I\n + 2 6") == rows([("+", 2, 7, 2, 6)])
     assert push(j, SIDE_RIGHT, " I I\n + 2 3") == rows([("+", 2, 7, 2, 3)])
     j.close()
+
+
+def test_inequality_join_watermark_right_larger():
+    # the mirror of hash_join.rs:1740-1829: condition left.col1 <=
+    # right.col1 (right side larger, clean_right_state) — the min emits
+    # for the RIGHT side's output column and cleans right rows below it
+    from rwtest.ffi import CMP_LE
+
+    j = ffi.HashJoin(oracle(), JOIN_INNER, I2, I2, key_l=[0], key_r=[0],
+                     pk_l=[1], pk_r=[1], cond=(CMP_LE, 1, 3),
+                     wm_ineq=((1, 1, False, True),))
+    assert push(j, SIDE_RIGHT, " I I\n + 2 4\n + 2 7\n + 3 8") == []
+    # arrival order right-then-left also buffers correctly
+    assert j.watermark(SIDE_RIGHT, 1, 10) == []
+    assert j.watermark(SIDE_LEFT, 1, 6) == [(3, 6)]
+    # right rows with col1 < 6 cleaned: (2,4) gone, (2,7)/(3,8) remain
+    assert push(j, SIDE_LEFT, " I I\n + 2 6") == rows([("+", 2, 6, 2, 7)])
+    # a stale watermark (not advancing the min) emits nothing
+    assert j.watermark(SIDE_LEFT, 1, 5) == []
+    assert push(j, SIDE_LEFT, " I I\n + 2 3") == rows([("+", 2, 3, 2, 7)])
+    j.close()
