@@ -1048,3 +1048,51 @@ def test_topn_float_restore_oracle():
     assert topn_checkpoint_drain(lib, a.h) == topn_checkpoint_drain(lib, b.h)
     a.close()
     b.close()
+
+
+def test_agg_float_distinct_and_minput_restore_oracle():
+    # float DISTINCT keys and float materialized-input values round-trip
+    # their dedup/minput drains (pins the raw-word float paths)
+    from rwtest.ffi import (AGG_COUNT, AGG_MIN, T_F64, agg_dedup_drain_bytes,
+                            agg_dedup_restore, agg_minput_drain_bytes,
+                            agg_minput_restore)
+
+    calls = [(AGG_COUNT, 1, T_I64, 1), (AGG_MIN, 1, T_F64),
+             (AGG_COUNT_STAR, -1, T_I64)]
+    lib = oracle()
+    mk = lambda: ffi.HashAgg(lib, [T_I64, T_F64, T_I64], [0], calls, 2,
+                             stream_key=(2,))
+    a = mk()
+    rng = np.random.default_rng(21)
+    rid = [0]
+
+    def chunk():
+        n = 120
+        g = rng.integers(0, 6, n)
+        v = np.round(rng.uniform(-2, 2, n), 1)  # repeats -> real dedup
+        r2 = np.arange(rid[0], rid[0] + n)
+        rid[0] += n
+        return ffi.Chunk([T_I64, T_F64, T_I64], np.zeros(n, np.uint8),
+                         [g, np.array(v, np.float64), r2],
+                         [np.ones(n, np.uint8)] * 3)
+
+    inter, ded, minp = b"", b"", b""
+    for e in range(2):
+        _drive_agg(a, [chunk()], e + 1)
+        inter += agg_checkpoint_drain_bytes(lib, a.h)
+        ded += agg_dedup_drain_bytes(lib, a.h, 0)
+        minp += agg_minput_drain_bytes(lib, a.h, 0)
+    b = mk()
+    agg_minput_restore(lib, b.h, 0, minp)
+    agg_dedup_restore(lib, b.h, 0, ded)
+    agg_restore(lib, b.h, inter)
+    c = chunk()
+    oa = _drive_agg(a, [c], 5)
+    ob = _drive_agg(b, [c], 5)
+    assert oa == ob, "restored float distinct/minput agg diverged"
+    for fn in (agg_checkpoint_drain_bytes,
+               lambda l, h: agg_dedup_drain_bytes(l, h, 0),
+               lambda l, h: agg_minput_drain_bytes(l, h, 0)):
+        assert fn(lib, a.h) == fn(lib, b.h)
+    a.close()
+    b.close()
